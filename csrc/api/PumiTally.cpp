@@ -1,0 +1,104 @@
+#include "PumiTally.h"
+
+#include "../core/engine.h"
+
+#include <chrono>
+#include <cstdio>
+#include <cstdlib>
+#include <string>
+
+namespace pumitally {
+
+namespace {
+double now_s() {
+  using clk = std::chrono::steady_clock;
+  return std::chrono::duration<double>(clk::now().time_since_epoch()).count();
+}
+} // namespace
+
+// Accumulated wall-clock phase timers; equivalent of the reference
+// TallyTimes (PumiTallyImpl.h:18-27) with working device fences.
+struct TallyTimes {
+  double initialization_time = 0.0;
+  double total_time_to_tally = 0.0;
+  double vtk_file_write_time = 0.0;
+  void print() const {
+    printf("\n");
+    printf("[TIME] Initialization time     : %f seconds\n", initialization_time);
+    printf("[TIME] Total time to tally     : %f seconds\n", total_time_to_tally);
+    printf("[TIME] VTK file write time     : %f seconds\n", vtk_file_write_time);
+    printf("[TIME] Total PUMI-Tally time   : %f seconds\n",
+           initialization_time + total_time_to_tally + vtk_file_write_time);
+  }
+};
+
+struct PumiTallyImpl {
+  std::unique_ptr<Engine> engine;
+  int32_t num_particles = 0;
+  TallyTimes times;
+  std::string output = "fluxresult.vtk";
+};
+
+PumiTally::PumiTally(const std::string &mesh_filename, int32_t num_particles,
+                     int &argc, char **&argv) {
+  (void)argc;
+  (void)argv;
+  pimpl_ = std::make_unique<PumiTallyImpl>();
+  pimpl_->num_particles = num_particles;
+  const double t0 = now_s();
+  Mesh mesh = read_mesh(mesh_filename);
+  printf("[INFO] pumitally loaded mesh %s with %lld elements\n",
+         mesh_filename.c_str(), (long long)mesh.nelems);
+
+  const char *dev = getenv("PUMITALLY_DEVICE");
+  std::unique_ptr<Engine> eng;
+  if (!dev || std::string(dev) != "cpu") {
+    const int ordinal = dev ? atoi(dev) : 0;
+    eng = make_gpu_engine(mesh, num_particles, ordinal);
+  }
+  if (!eng) eng = make_cpu_engine(std::move(mesh), num_particles);
+  pimpl_->engine = std::move(eng);
+  if (const char *out = getenv("PUMITALLY_OUTPUT")) pimpl_->output = out;
+  pimpl_->engine->synchronize();
+  pimpl_->times.initialization_time += now_s() - t0;
+}
+
+void PumiTally::CopyInitialPosition(double *init_particle_positions,
+                                    std::int32_t size) const {
+  const double t0 = now_s();
+  if (size != pimpl_->num_particles * 3)
+    throw std::runtime_error("CopyInitialPosition: size must be 3*num_particles");
+  pimpl_->engine->copy_initial_position(init_particle_positions,
+                                        pimpl_->num_particles);
+  pimpl_->engine->synchronize();
+  pimpl_->times.initialization_time += now_s() - t0;
+}
+
+void PumiTally::MoveToNextLocation(double *particle_origin,
+                                   double *particle_destinations,
+                                   int8_t *flying, double *weights,
+                                   int32_t size) const {
+  const double t0 = now_s();
+  if (size != pimpl_->num_particles * 3)
+    throw std::runtime_error("MoveToNextLocation: size must be 3*num_particles");
+  pimpl_->engine->move(particle_origin, particle_destinations, flying, weights,
+                       pimpl_->num_particles);
+  // Parity with the reference host contract: the flying array is consumed
+  // and zeroed after upload (PumiTallyImpl.cpp:169-172).
+  for (int32_t i = 0; i < pimpl_->num_particles; ++i) flying[i] = 0;
+  pimpl_->engine->synchronize();
+  pimpl_->times.total_time_to_tally += now_s() - t0;
+}
+
+void PumiTally::WriteTallyResults() const {
+  const double t0 = now_s();
+  pimpl_->engine->synchronize();
+  write_tally_vtk(pimpl_->output, pimpl_->engine->mesh(),
+                  pimpl_->engine->flux());
+  pimpl_->times.vtk_file_write_time += now_s() - t0;
+  pimpl_->times.print();
+}
+
+PumiTally::~PumiTally() = default;
+
+} // namespace pumitally

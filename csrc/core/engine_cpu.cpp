@@ -1,0 +1,131 @@
+// Serial CPU engine: the correctness oracle for the HIP engine and the
+// no-GPU plumbing path (BASELINE config 1).
+#include "engine.h"
+#include "walk.h"
+
+#include <cmath>
+#include <cstring>
+#include <stdexcept>
+
+namespace pumitally {
+
+int default_max_steps(const Mesh &m) {
+  const int c = (int)std::ceil(std::cbrt((double)m.nelems));
+  return std::max(1000, 32 * c + 64);
+}
+
+std::vector<double> normalize_flux(const Mesh &m, const std::vector<double> &flux) {
+  std::vector<double> out(m.nelems);
+  for (int64_t e = 0; e < m.nelems; ++e) out[e] = flux[e] / m.volumes[e];
+  return out;
+}
+
+namespace {
+
+class CpuEngine final : public Engine {
+public:
+  CpuEngine(Mesh mesh, int64_t n) : mesh_(std::move(mesh)), n_(n) {
+    flux_.assign(mesh_.nelems, 0.0);
+    pos_.resize(n_ * 3);
+    elem_.assign(n_, 0);
+    escaped_.assign(n_, 0);
+    const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
+    for (int64_t i = 0; i < n_; ++i) {
+      pos_[i * 3] = c0.x;
+      pos_[i * 3 + 1] = c0.y;
+      pos_[i * 3 + 2] = c0.z;
+    }
+    loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+  }
+
+  int64_t num_particles() const override { return n_; }
+  const Mesh &mesh() const override { return mesh_; }
+
+  void copy_initial_position(const double *p, int64_t n) override {
+    check_n(n);
+    for (int64_t i = 0; i < n; ++i) {
+      const Vec3 q{p[i * 3], p[i * 3 + 1], p[i * 3 + 2]};
+      elem_[i] = mesh_.locate(q, loc_tol_);
+      pos_[i * 3] = q.x;
+      pos_[i * 3 + 1] = q.y;
+      pos_[i * 3 + 2] = q.z;
+      escaped_[i] = 0;
+    }
+  }
+
+  void move(const double *origin, const double *dest, const int8_t *flying,
+            const double *weights, int64_t n) override {
+    check_n(n);
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    for (int64_t i = 0; i < n; ++i) {
+      if (!flying[i]) continue;
+      Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
+      // Phase A: relocate to the given origin (no tally).  Skipped for
+      // escaped particles (behavioral pin, see engine.h).
+      if (!escaped_[i]) {
+        const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
+        if (q.x != o.x || q.y != o.y || q.z != o.z) {
+          elem_[i] = mesh_.locate(q, loc_tol_);
+          o = q;
+          stats_.relocated++;
+        }
+      }
+      // Phase B: tallied walk to the destination.
+      if (elem_[i] < 0) {
+        // outside the mesh: nothing to tally; remember requested position
+        pos_[i * 3] = o.x; pos_[i * 3 + 1] = o.y; pos_[i * 3 + 2] = o.z;
+        continue;
+      }
+      const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+      int32_t out_elem;
+      Vec3 out_pos;
+      bool out_esc;
+      walk_segment(
+          mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d, weights[i],
+          steps, [&](int32_t e, double v) { flux_[e] += v; }, &out_elem,
+          &out_pos, &out_esc);
+      if (out_elem == kWalkLost) {
+        stats_.lost_particles++;
+        out_elem = elem_[i];
+      }
+      elem_[i] = out_elem;
+      pos_[i * 3] = out_pos.x;
+      pos_[i * 3 + 1] = out_pos.y;
+      pos_[i * 3 + 2] = out_pos.z;
+      escaped_[i] = out_esc ? 1 : 0;
+    }
+    stats_.moves++;
+  }
+
+  std::vector<double> flux() const override { return flux_; }
+  std::vector<int32_t> elem_ids() const override { return elem_; }
+  std::vector<double> positions() const override { return pos_; }
+  std::vector<uint8_t> escaped() const override { return escaped_; }
+  const EngineStats &stats() const override { return stats_; }
+
+  void set_flux(const double *f, int64_t ne) override {
+    if (ne != (int64_t)flux_.size()) throw std::runtime_error("set_flux size mismatch");
+    std::memcpy(flux_.data(), f, ne * sizeof(double));
+  }
+
+private:
+  void check_n(int64_t n) const {
+    if (n != n_) throw std::runtime_error("particle count mismatch");
+  }
+
+  Mesh mesh_;
+  int64_t n_;
+  double loc_tol_;
+  std::vector<double> flux_, pos_;
+  std::vector<int32_t> elem_;
+  std::vector<uint8_t> escaped_;
+  EngineStats stats_;
+};
+
+} // namespace
+
+std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles) {
+  return std::make_unique<CpuEngine>(std::move(mesh), num_particles);
+}
+
+} // namespace pumitally

@@ -1,0 +1,92 @@
+// First-party unstructured tetrahedral mesh core (CPU side).
+//
+// Replaces the used subset of Omega_h in the reference (PUMI-Tally,
+// /root/reference/src/pumitally/PumiTallyImpl.cpp:384-399,461-528,562):
+//   * mesh object with coords + tet->vert connectivity
+//   * derived tet->tet face adjacency (what Omega_h/pumipic derive for the
+//     particle walk)
+//   * per-tet volumes (Omega_h simplex_size_from_basis)
+//   * build_box test-mesh generator (Omega_h::build_box, used by the
+//     reference tests test/test_pumi_tally_impl_methods.cpp:34-35)
+//   * mesh file IO (.osh directory format read/write - reconstructed,
+//     see osh_io.cpp - plus Gmsh .msh and legacy VTK output)
+//
+// Unlike the reference stack there is no Kokkos view machinery: the mesh is
+// flat std::vectors on the host, uploaded once into flat HBM arrays by the
+// GPU engine.  The walk consumes only `planes` (4 canonically-oriented face
+// planes per tet, 128 B) and `nbr` (4 neighbor ids, 16 B): one contiguous
+// 144 B record per tet, no indirection to vertex coords in the hot loop.
+#pragma once
+
+#include "geom.h"
+
+#include <array>
+#include <cstdint>
+#include <string>
+#include <vector>
+
+namespace pumitally {
+
+// Local face f of a tet is the face OPPOSITE local vertex f.
+// For a positively-oriented tet (v0,v1,v2,v3) the inward-positive triangles:
+//   F0=(1,3,2)  F1=(0,2,3)  F2=(0,3,1)  F3=(0,1,2)
+constexpr int kFaceVerts[4][3] = {{1, 3, 2}, {0, 2, 3}, {0, 3, 1}, {0, 1, 2}};
+
+// Uniform-grid acceleration structure for point-in-mesh localization.
+// CSR lists of tets whose AABB overlaps each cell.
+struct LocGrid {
+  int nx = 0, ny = 0, nz = 0;
+  Vec3 lo{0, 0, 0};
+  Vec3 inv_h{0, 0, 0}; // 1/cell_size
+  std::vector<int32_t> cell_start; // size nx*ny*nz+1
+  std::vector<int32_t> cell_tets;  // CSR payload
+};
+
+struct Mesh {
+  int64_t nverts = 0;
+  int64_t nelems = 0;
+  std::vector<double> coords;   // nverts*3, xyz interleaved
+  std::vector<int32_t> tet2vert; // nelems*4, positively oriented
+
+  // Derived (built by finalize()):
+  std::vector<int32_t> nbr;     // nelems*4: neighbor tet across face f, -1 = boundary
+  std::vector<Plane> planes;    // nelems*4: inward-positive unit-normal face planes
+  std::vector<double> volumes;  // nelems
+  Vec3 bbox_lo{0, 0, 0}, bbox_hi{0, 0, 0};
+  LocGrid grid;
+
+  // Build adjacency, planes, volumes, bbox and the localization grid.
+  // Reorients negatively-oriented tets in place (swaps verts 2,3).
+  void finalize();
+
+  Vec3 vert(int32_t v) const { return {coords[v * 3], coords[v * 3 + 1], coords[v * 3 + 2]}; }
+  Vec3 centroid(int32_t t) const;
+
+  // Point-in-tet test used by localization (CPU path). tol: accepted signed
+  // distance below a face plane (>=0 means strictly inside).
+  bool contains(int32_t t, Vec3 p, double tol) const;
+  // Locate the tet containing p, or -1. CPU reference implementation of the
+  // GPU localization kernel.
+  int32_t locate(Vec3 p, double tol) const;
+};
+
+// Analytic box mesh generator: divisions (nx,ny,nz) over extents (lx,ly,lz).
+// Each grid cell is cut into 6 tets around its main diagonal using the same
+// element ordering as Omega_h::build_box for the unit cube, which the
+// reference tests pin (element ids 2,3,4 along the x-ray, centroid of
+// element 0 at (0.5,0.75,0.25); /root/reference/test/
+// test_pumi_tally_impl_methods.cpp:83,152-159,221-282).
+Mesh build_box(int nx, int ny, int nz, double lx, double ly, double lz);
+
+Mesh mesh_from_arrays(int64_t nverts, const double *coords, int64_t nelems,
+                      const int32_t *tets);
+
+// IO (implemented in mesh_io.cpp / osh_io.cpp)
+Mesh read_gmsh(const std::string &path);             // Gmsh .msh v2.2/v4.1 ASCII
+Mesh read_mesh(const std::string &path);             // dispatch on extension
+void write_vtk(const std::string &path, const Mesh &m,
+               const std::vector<std::pair<std::string, std::vector<double>>> &cell_data);
+Mesh read_osh(const std::string &dir);               // .osh directory
+void write_osh(const std::string &dir, const Mesh &m);
+
+} // namespace pumitally

@@ -1,0 +1,208 @@
+// Mesh IO: legacy VTK output, Gmsh .msh input, and the .osh directory
+// format (see osh_io.cpp for the binary reader/writer).
+//
+// Replaces the used subset of Omega_h file IO in the reference:
+//   Omega_h::binary::read  (PumiTallyImpl.cpp:562)
+//   Omega_h::vtk::write_parallel (PumiTallyImpl.cpp:415)
+// The reference writes "fluxresult.vtk" through Omega_h; we write a single
+// legacy VTK unstructured-grid file with cell data "flux" and "volume",
+// which ParaView reads directly.
+#include "engine.h"
+#include "mesh.h"
+
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <sstream>
+#include <stdexcept>
+
+namespace pumitally {
+
+void write_vtk(const std::string &path, const Mesh &m,
+               const std::vector<std::pair<std::string, std::vector<double>>> &cell_data) {
+  std::ofstream f(path);
+  if (!f) throw std::runtime_error("cannot open " + path + " for writing");
+  f << "# vtk DataFile Version 3.0\n";
+  f << "pumitally flux tally\n";
+  f << "ASCII\n";
+  f << "DATASET UNSTRUCTURED_GRID\n";
+  f << "POINTS " << m.nverts << " double\n";
+  char buf[128];
+  for (int64_t v = 0; v < m.nverts; ++v) {
+    snprintf(buf, sizeof buf, "%.17g %.17g %.17g\n", m.coords[v * 3],
+             m.coords[v * 3 + 1], m.coords[v * 3 + 2]);
+    f << buf;
+  }
+  f << "CELLS " << m.nelems << " " << m.nelems * 5 << "\n";
+  for (int64_t t = 0; t < m.nelems; ++t) {
+    f << "4 " << m.tet2vert[t * 4] << " " << m.tet2vert[t * 4 + 1] << " "
+      << m.tet2vert[t * 4 + 2] << " " << m.tet2vert[t * 4 + 3] << "\n";
+  }
+  f << "CELL_TYPES " << m.nelems << "\n";
+  for (int64_t t = 0; t < m.nelems; ++t) f << "10\n"; // VTK_TETRA
+  if (!cell_data.empty()) {
+    f << "CELL_DATA " << m.nelems << "\n";
+    for (const auto &cd : cell_data) {
+      f << "SCALARS " << cd.first << " double 1\nLOOKUP_TABLE default\n";
+      for (int64_t t = 0; t < m.nelems; ++t) {
+        snprintf(buf, sizeof buf, "%.17g\n", cd.second[t]);
+        f << buf;
+      }
+    }
+  }
+}
+
+void write_tally_vtk(const std::string &filename, const Mesh &m,
+                     const std::vector<double> &flux) {
+  std::vector<double> normalized = normalize_flux(m, flux);
+  write_vtk(filename, m, {{"flux", normalized}, {"volume", m.volumes}});
+}
+
+// ---------------------------------------------------------------------------
+// Gmsh .msh reader: ASCII v2.2 and v4.1, linear tets only (element type 4).
+// ---------------------------------------------------------------------------
+Mesh read_gmsh(const std::string &path) {
+  std::ifstream f(path);
+  if (!f) throw std::runtime_error("cannot open " + path);
+  std::string line;
+  double version = 0;
+  std::vector<double> coords;
+  std::vector<int64_t> node_tags;
+  std::vector<int32_t> tets;
+
+  auto expect_end = [&](const char *tag) {
+    while (std::getline(f, line)) {
+      if (line.rfind(tag, 0) == 0) return;
+    }
+    throw std::runtime_error(std::string("missing ") + tag + " in " + path);
+  };
+
+  while (std::getline(f, line)) {
+    if (line.rfind("$MeshFormat", 0) == 0) {
+      std::getline(f, line);
+      std::istringstream is(line);
+      int ftype, dsize;
+      is >> version >> ftype >> dsize;
+      if (ftype != 0) throw std::runtime_error("binary .msh not supported; export ASCII");
+      expect_end("$EndMeshFormat");
+    } else if (line.rfind("$Nodes", 0) == 0) {
+      if (version >= 4.0) {
+        std::getline(f, line);
+        std::istringstream is(line);
+        int64_t nblocks, nnodes, mintag, maxtag;
+        is >> nblocks >> nnodes >> mintag >> maxtag;
+        coords.reserve(nnodes * 3);
+        node_tags.reserve(nnodes);
+        for (int64_t b = 0; b < nblocks; ++b) {
+          std::getline(f, line);
+          std::istringstream bs(line);
+          int dim, etag, param;
+          int64_t nb;
+          bs >> dim >> etag >> param >> nb;
+          std::vector<int64_t> tags(nb);
+          for (int64_t i = 0; i < nb; ++i) { std::getline(f, line); tags[i] = std::stoll(line); }
+          for (int64_t i = 0; i < nb; ++i) {
+            std::getline(f, line);
+            std::istringstream cs(line);
+            double x, y, z;
+            cs >> x >> y >> z;
+            node_tags.push_back(tags[i]);
+            coords.push_back(x); coords.push_back(y); coords.push_back(z);
+          }
+        }
+      } else {
+        std::getline(f, line);
+        const int64_t nnodes = std::stoll(line);
+        coords.reserve(nnodes * 3);
+        node_tags.reserve(nnodes);
+        for (int64_t i = 0; i < nnodes; ++i) {
+          std::getline(f, line);
+          std::istringstream cs(line);
+          int64_t tag; double x, y, z;
+          cs >> tag >> x >> y >> z;
+          node_tags.push_back(tag);
+          coords.push_back(x); coords.push_back(y); coords.push_back(z);
+        }
+      }
+      expect_end("$EndNodes");
+    } else if (line.rfind("$Elements", 0) == 0) {
+      if (version >= 4.0) {
+        std::getline(f, line);
+        std::istringstream is(line);
+        int64_t nblocks, nelems, mintag, maxtag;
+        is >> nblocks >> nelems >> mintag >> maxtag;
+        for (int64_t b = 0; b < nblocks; ++b) {
+          std::getline(f, line);
+          std::istringstream bs(line);
+          int dim, etag, etype;
+          int64_t nb;
+          bs >> dim >> etag >> etype >> nb;
+          for (int64_t i = 0; i < nb; ++i) {
+            std::getline(f, line);
+            if (etype == 4) {
+              std::istringstream es(line);
+              int64_t tag, a, bb, c, d;
+              es >> tag >> a >> bb >> c >> d;
+              tets.push_back((int32_t)a); tets.push_back((int32_t)bb);
+              tets.push_back((int32_t)c); tets.push_back((int32_t)d);
+            }
+          }
+        }
+      } else {
+        std::getline(f, line);
+        const int64_t nelems = std::stoll(line);
+        for (int64_t i = 0; i < nelems; ++i) {
+          std::getline(f, line);
+          std::istringstream es(line);
+          int64_t tag; int etype, ntags;
+          es >> tag >> etype >> ntags;
+          int64_t skip;
+          for (int k = 0; k < ntags; ++k) es >> skip;
+          if (etype == 4) {
+            int64_t a, b, c, d;
+            es >> a >> b >> c >> d;
+            tets.push_back((int32_t)a); tets.push_back((int32_t)b);
+            tets.push_back((int32_t)c); tets.push_back((int32_t)d);
+          }
+        }
+      }
+      expect_end("$EndElements");
+    }
+  }
+  if (coords.empty() || tets.empty())
+    throw std::runtime_error("no tet mesh found in " + path);
+
+  // Remap gmsh node tags (1-based, possibly sparse) to dense 0-based ids.
+  std::vector<int64_t> remap;
+  int64_t max_tag = 0;
+  for (int64_t t : node_tags) max_tag = std::max(max_tag, t);
+  remap.assign(max_tag + 1, -1);
+  for (size_t i = 0; i < node_tags.size(); ++i) remap[node_tags[i]] = (int64_t)i;
+  for (auto &v : tets) {
+    const int64_t dense = remap[v];
+    if (dense < 0) throw std::runtime_error("bad node tag in .msh elements");
+    v = (int32_t)dense;
+  }
+
+  Mesh m;
+  m.nverts = (int64_t)node_tags.size();
+  m.nelems = (int64_t)tets.size() / 4;
+  m.coords = std::move(coords);
+  m.tet2vert = std::move(tets);
+  m.finalize();
+  return m;
+}
+
+Mesh read_mesh(const std::string &path) {
+  auto ends_with = [&](const char *s) {
+    const size_t n = strlen(s);
+    return path.size() >= n && path.compare(path.size() - n, n, s) == 0;
+  };
+  std::string p = path;
+  while (!p.empty() && p.back() == '/') p.pop_back();
+  if (p.size() >= 4 && p.compare(p.size() - 4, 4, ".osh") == 0) return read_osh(p);
+  if (ends_with(".msh")) return read_gmsh(path);
+  throw std::runtime_error("unknown mesh format (expected .osh or .msh): " + path);
+}
+
+} // namespace pumitally

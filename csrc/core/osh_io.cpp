@@ -1,0 +1,80 @@
+// .osh directory mesh format.
+//
+// The reference consumes Omega_h binary meshes ("mesh.osh" directories,
+// Omega_h::binary::read at PumiTallyImpl.cpp:562).  Omega_h's binary layout
+// is an undocumented, versioned, zlib-compressed stream and this build
+// environment has no network and no Omega_h sources, so byte-compatibility
+// cannot be implemented or verified here.  Instead we define a
+// self-describing directory format with the same shape (a `mesh.osh/`
+// directory containing an ASCII `nparts` file and per-rank data files) and
+// an explicit magic, plus converters from Gmsh .msh (the format msh2osh
+// starts from, README.md:115-126 of the reference).  A real Omega_h file is
+// detected by its magic and rejected with an actionable message.
+//
+// File layout of `<dir>/0.osh` (all little-endian):
+//   8 bytes  magic "PTOSH1\n\0"
+//   i64      nverts
+//   i64      nelems
+//   f64[nverts*3]   coords (xyz interleaved)
+//   i32[nelems*4]   tet2vert
+#include "mesh.h"
+
+#include <cerrno>
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <stdexcept>
+#include <sys/stat.h>
+
+namespace pumitally {
+
+static const char kOshMagic[8] = {'P', 'T', 'O', 'S', 'H', '1', '\n', '\0'};
+
+void write_osh(const std::string &dir, const Mesh &m) {
+  if (mkdir(dir.c_str(), 0755) != 0 && errno != EEXIST)
+    throw std::runtime_error("cannot create directory " + dir);
+  {
+    std::ofstream np(dir + "/nparts");
+    np << 1 << "\n";
+  }
+  std::ofstream f(dir + "/0.osh", std::ios::binary);
+  if (!f) throw std::runtime_error("cannot open " + dir + "/0.osh");
+  f.write(kOshMagic, 8);
+  const int64_t nv = m.nverts, ne = m.nelems;
+  f.write((const char *)&nv, 8);
+  f.write((const char *)&ne, 8);
+  f.write((const char *)m.coords.data(), nv * 3 * sizeof(double));
+  f.write((const char *)m.tet2vert.data(), ne * 4 * sizeof(int32_t));
+  if (!f) throw std::runtime_error("write failed: " + dir + "/0.osh");
+}
+
+Mesh read_osh(const std::string &dir) {
+  std::ifstream f(dir + "/0.osh", std::ios::binary);
+  if (!f) throw std::runtime_error("cannot open " + dir + "/0.osh (not a .osh directory?)");
+  char magic[8] = {0};
+  f.read(magic, 8);
+  if (memcmp(magic, kOshMagic, 8) != 0) {
+    throw std::runtime_error(
+        dir + " is not a pumitally .osh mesh (magic mismatch). If this is an "
+              "Omega_h binary mesh, convert it offline: export the mesh as "
+              "Gmsh ASCII .msh and load that, or use "
+              "pumiumtally_amd.mesh.convert(msh_path, osh_dir).");
+  }
+  int64_t nv = 0, ne = 0;
+  f.read((char *)&nv, 8);
+  f.read((char *)&ne, 8);
+  if (nv <= 0 || ne <= 0 || nv > (int64_t)1 << 40 || ne > (int64_t)1 << 40)
+    throw std::runtime_error("corrupt .osh header in " + dir);
+  Mesh m;
+  m.nverts = nv;
+  m.nelems = ne;
+  m.coords.resize(nv * 3);
+  m.tet2vert.resize(ne * 4);
+  f.read((char *)m.coords.data(), nv * 3 * sizeof(double));
+  f.read((char *)m.tet2vert.data(), ne * 4 * sizeof(int32_t));
+  if (!f) throw std::runtime_error("truncated .osh data in " + dir);
+  m.finalize();
+  return m;
+}
+
+} // namespace pumitally

@@ -1,0 +1,144 @@
+// The adjacency walk: the single hot algorithm of the framework.
+//
+// Re-implements from scratch the behavior of the external
+// pumipic::ParticleTracer element walk that the reference delegates to
+// (/root/reference/src/pumitally/PumiTallyImpl.cpp:454 and the handler
+// sequencing at :297-316): walk a straight segment origin->dest through the
+// tet mesh element by element, accumulate track_length*weight into the flux
+// tally of every element crossed (K9, :352-380), clip the destination at a
+// vacuum boundary (K6, :256-286), and advance the current element (K5,
+// :243-254) -- all fused into one loop per particle instead of one kernel
+// per walk iteration over all particles.
+//
+// The exit-face test is a signed-distance parametric test against the 4
+// precomputed inward-positive face planes: V(t) = plane_eval(p(t)) is linear
+// in t, the particle exits through the face with the smallest crossing
+// t = V_o / (V_o - V_d) among faces with V_d < 0.  Because shared-face
+// planes are bitwise-consistent across neighbor tets (mesh.cpp finalize()
+// step 3), t is monotone and the per-element intervals tile the segment
+// exactly.
+//
+// Header-only, host+device: the CPU engine is the oracle for the HIP kernel.
+#pragma once
+
+#include "geom.h"
+
+namespace pumitally {
+
+constexpr double kWalkTEps = 1e-12;   // tolerance on the segment parameter t
+constexpr int kWalkLost = -2;         // out_elem value when max_steps hit
+
+// FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
+// performs atomicAdd into the flux array; on the serial CPU path a plain +=.
+template <class FluxAdd>
+PT_HD void walk_segment(const Plane *__restrict__ planes,
+                        const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
+                        Vec3 d, double weight, int max_steps, FluxAdd &&add,
+                        int32_t *out_elem, Vec3 *out_pos, bool *out_escaped) {
+  const double seg_len = norm(d - o);
+  double t_cur = 0.0;
+  int32_t prev_elem = -1;
+  const bool tally = (weight != 0.0) && (seg_len > 0.0);
+
+  for (int step = 0; step < max_steps; ++step) {
+    // Evaluate the 4 face planes at both segment endpoints.
+    const Plane *pl = planes + (int64_t)elem * 4;
+    double t_exit = 2.0;
+    int exit_face = -1;
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
+    for (int f = 0; f < 4; ++f) {
+      const double vd = plane_eval(pl[f], d);
+      if (vd < 0.0) {
+        const double vo = plane_eval(pl[f], o);
+        const double denom = vo - vd; // > 0 since vd < 0 <= ~vo
+        if (denom > 0.0) {
+          double tf = vo / denom;
+          // Never step backwards (grazing entry can give tf slightly below
+          // t_cur); never consider the face we just came through.
+          if (nbr[(int64_t)elem * 4 + f] != prev_elem || prev_elem == -1) {
+            if (tf >= t_cur - kWalkTEps && tf < t_exit) {
+              t_exit = tf;
+              exit_face = f;
+            }
+          }
+        }
+      }
+    }
+
+    if (exit_face < 0 || t_exit >= 1.0) {
+      // Destination lies inside this element: tally the final partial
+      // segment and stop (reference: reached_destination, last_exit==-1).
+      if (tally) add(elem, (1.0 - t_cur) * seg_len * weight);
+      *out_elem = elem;
+      *out_pos = d;
+      *out_escaped = false;
+      return;
+    }
+
+    const double t_clamped = t_exit > t_cur ? t_exit : t_cur;
+    if (tally) add(elem, (t_clamped - t_cur) * seg_len * weight);
+
+    const int32_t next = nbr[(int64_t)elem * 4 + exit_face];
+    if (next < 0) {
+      // Vacuum boundary: clip the destination to the exit point; the
+      // particle keeps its last element id (reference K6 semantics,
+      // PumiTallyImpl.cpp:275-281 and the 1.0-not-1.1 test expectation).
+      *out_elem = elem;
+      *out_pos = o + t_clamped * (d - o);
+      *out_escaped = true;
+      return;
+    }
+    prev_elem = elem;
+    elem = next;
+    t_cur = t_clamped;
+  }
+
+  // Walk did not converge (numerically stuck / absurdly long chord): drop
+  // the particle at its current position, flag as lost.  The engine counts
+  // these (reference prints "Not all particles are found",
+  // PumiTallyImpl.cpp:455-458).
+  *out_elem = kWalkLost;
+  *out_pos = o + t_cur * (d - o);
+  *out_escaped = false;
+}
+
+// Point-in-tet test against the 4 planes (signed distances, unit normals).
+PT_HD bool tet_contains(const Plane *__restrict__ planes, int32_t t, Vec3 p,
+                        double tol) {
+  const Plane *pl = planes + (int64_t)t * 4;
+  double mn = plane_eval(pl[0], p);
+  mn = mn < plane_eval(pl[1], p) ? mn : plane_eval(pl[1], p);
+  mn = mn < plane_eval(pl[2], p) ? mn : plane_eval(pl[2], p);
+  mn = mn < plane_eval(pl[3], p) ? mn : plane_eval(pl[3], p);
+  return mn >= -tol;
+}
+
+// Grid-based localization shared by CPU and GPU.  Mirrors Mesh::locate but
+// takes flat arrays so the HIP kernel can use it directly.
+struct GridView {
+  int nx, ny, nz;
+  Vec3 lo, inv_h;
+  const int32_t *cell_start;
+  const int32_t *cell_tets;
+};
+
+PT_HD int32_t grid_locate(const GridView &g, const Plane *__restrict__ planes,
+                          Vec3 p, double tol) {
+  int cx = (int)((p.x - g.lo.x) * g.inv_h.x);
+  int cy = (int)((p.y - g.lo.y) * g.inv_h.y);
+  int cz = (int)((p.z - g.lo.z) * g.inv_h.z);
+  cx = cx < 0 ? 0 : (cx >= g.nx ? g.nx - 1 : cx);
+  cy = cy < 0 ? 0 : (cy >= g.ny ? g.ny - 1 : cy);
+  cz = cz < 0 ? 0 : (cz >= g.nz ? g.nz - 1 : cz);
+  const int64_t c = ((int64_t)cz * g.ny + cy) * g.nx + cx;
+  const int32_t b = g.cell_start[c], e = g.cell_start[c + 1];
+  for (int32_t i = b; i < e; ++i)
+    if (tet_contains(planes, g.cell_tets[i], p, tol)) return g.cell_tets[i];
+  for (int32_t i = b; i < e; ++i)
+    if (tet_contains(planes, g.cell_tets[i], p, tol * 1e4)) return g.cell_tets[i];
+  return -1;
+}
+
+} // namespace pumitally

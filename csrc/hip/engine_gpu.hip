@@ -1,0 +1,376 @@
+// MI355X (gfx950) tally engine.
+//
+// Design (MI355X-first, not a port of the reference's Kokkos structure):
+//   * The whole per-step pipeline of the reference -- H2D staging (M1-M3),
+//     buffer->particle copies (K1-K4), the external ParticleTracer search
+//     loop and the per-boundary handler kernels (K5-K9)
+//     (/root/reference/src/pumitally/PumiTallyImpl.cpp:66-149,243-380) --
+//     collapses into ONE fused kernel (k_move) that walks each particle's
+//     whole segment in registers and atomicAdds per-element contributions.
+//     No per-iteration kernel launches, no element-bucketed particle
+//     structure: a flat grid-stride loop over particles; walk-length
+//     variance is absorbed statistically by giving each lane many
+//     particles (grid capped at ~8 blocks/CU).
+//   * Mesh data is two flat HBM arrays consumed by the walk: 4 face planes
+//     (double4, inward-positive unit normals) + 4 neighbor ids per tet --
+//     144 B/tet, contiguous, no indirection to vertex coords in the hot
+//     loop.  A 1M-tet mesh (~150 MB) is fully resident in the 256 MiB
+//     Infinity Cache.
+//   * Host arrays from the caller are hipHostRegister'ed once (cached by
+//     pointer -- physics codes pass the same buffers every step) and then
+//     copied H2D in chunks on a copy stream that overlaps the walk kernel
+//     of the previous chunk on the compute stream, replacing the
+//     reference's fully synchronous deep_copy + fence staging.
+#include "../core/engine.h"
+#include "../core/walk.h"
+
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <array>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace pumitally {
+
+#define PT_HIP_CHECK(expr)                                                     \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    if (_e != hipSuccess)                                                      \
+      throw std::runtime_error(std::string("HIP error at " __FILE__ ":") +     \
+                               std::to_string(__LINE__) + ": " +               \
+                               hipGetErrorString(_e));                         \
+  } while (0)
+
+namespace {
+
+constexpr int kBlock = 256;
+
+__global__ void k_init_particles(double *__restrict__ pos,
+                                 int32_t *__restrict__ elem,
+                                 uint8_t *__restrict__ escaped, int64_t n,
+                                 double cx, double cy, double cz) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    pos[i * 3] = cx;
+    pos[i * 3 + 1] = cy;
+    pos[i * 3 + 2] = cz;
+    elem[i] = 0;
+    escaped[i] = 0;
+  }
+}
+
+__global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
+                         const double *__restrict__ q,
+                         double *__restrict__ pos, int32_t *__restrict__ elem,
+                         uint8_t *__restrict__ escaped, int64_t lo, int64_t hi,
+                         double tol) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = lo + blockIdx.x * blockDim.x + threadIdx.x; i < hi;
+       i += stride) {
+    const Vec3 p{q[i * 3], q[i * 3 + 1], q[i * 3 + 2]};
+    elem[i] = grid_locate(grid, planes, p, tol);
+    pos[i * 3] = p.x;
+    pos[i * 3 + 1] = p.y;
+    pos[i * 3 + 2] = p.z;
+    escaped[i] = 0;
+  }
+}
+
+// The fused move kernel: phase A (relocation of flying, non-escaped
+// particles whose origin changed) + phase B (tallied walk to destination).
+__global__ void k_move(const Plane *__restrict__ planes,
+                       const int32_t *__restrict__ nbr, GridView grid,
+                       const double *__restrict__ origin,
+                       const double *__restrict__ dest,
+                       const int8_t *__restrict__ flying,
+                       const double *__restrict__ weights,
+                       double *__restrict__ pos, int32_t *__restrict__ elem,
+                       uint8_t *__restrict__ escaped,
+                       double *__restrict__ flux,
+                       unsigned long long *__restrict__ lost, int64_t lo,
+                       int64_t hi, double loc_tol, int max_steps) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = lo + blockIdx.x * blockDim.x + threadIdx.x; i < hi;
+       i += stride) {
+    if (!flying[i]) continue;
+    Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
+    int32_t e = elem[i];
+    if (!escaped[i]) {
+      const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
+      if (q.x != o.x || q.y != o.y || q.z != o.z) {
+        e = grid_locate(grid, planes, q, loc_tol);
+        o = q;
+      }
+    }
+    if (e < 0) {
+      pos[i * 3] = o.x;
+      pos[i * 3 + 1] = o.y;
+      pos[i * 3 + 2] = o.z;
+      elem[i] = e;
+      continue;
+    }
+    const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+    int32_t out_elem;
+    Vec3 out_pos;
+    bool out_esc;
+    walk_segment(
+        planes, nbr, e, o, d, weights[i], max_steps,
+        [&](int32_t el, double v) {
+          atomicAdd(&flux[el], v);
+        },
+        &out_elem, &out_pos, &out_esc);
+    if (out_elem == kWalkLost) {
+      atomicAdd(lost, 1ull);
+      out_elem = e;
+    }
+    elem[i] = out_elem;
+    pos[i * 3] = out_pos.x;
+    pos[i * 3 + 1] = out_pos.y;
+    pos[i * 3 + 2] = out_pos.z;
+    escaped[i] = out_esc ? 1 : 0;
+  }
+}
+
+int grid_blocks(int64_t work) {
+  const int64_t blocks = (work + kBlock - 1) / kBlock;
+  // 256 CUs x 8 blocks/CU: cap and grid-stride the rest (guide G11).
+  return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
+}
+
+template <class T> T *dmalloc(int64_t count) {
+  void *p = nullptr;
+  PT_HIP_CHECK(hipMalloc(&p, count * sizeof(T)));
+  return (T *)p;
+}
+
+class GpuEngine final : public Engine {
+public:
+  GpuEngine(Mesh mesh, int64_t n, int device) : mesh_(std::move(mesh)), n_(n) {
+    PT_HIP_CHECK(hipSetDevice(device));
+    device_ = device;
+    PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
+    PT_HIP_CHECK(hipStreamCreateWithFlags(&s_comp_, hipStreamNonBlocking));
+    for (auto &ev : events_) PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+
+    // Mesh upload (once).
+    d_planes_ = dmalloc<Plane>(mesh_.nelems * 4);
+    d_nbr_ = dmalloc<int32_t>(mesh_.nelems * 4);
+    d_cell_start_ = dmalloc<int32_t>(mesh_.grid.cell_start.size());
+    d_cell_tets_ = dmalloc<int32_t>(mesh_.grid.cell_tets.size() + 1);
+    PT_HIP_CHECK(hipMemcpy(d_planes_, mesh_.planes.data(),
+                           mesh_.nelems * 4 * sizeof(Plane), hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_nbr_, mesh_.nbr.data(),
+                           mesh_.nelems * 4 * sizeof(int32_t), hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_cell_start_, mesh_.grid.cell_start.data(),
+                           mesh_.grid.cell_start.size() * sizeof(int32_t),
+                           hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_cell_tets_, mesh_.grid.cell_tets.data(),
+                           mesh_.grid.cell_tets.size() * sizeof(int32_t),
+                           hipMemcpyHostToDevice));
+    grid_view_ = GridView{mesh_.grid.nx, mesh_.grid.ny, mesh_.grid.nz,
+                          mesh_.grid.lo,  mesh_.grid.inv_h,
+                          d_cell_start_,  d_cell_tets_};
+
+    // Particle + staging arrays.
+    d_pos_ = dmalloc<double>(n_ * 3);
+    d_elem_ = dmalloc<int32_t>(n_);
+    d_escaped_ = dmalloc<uint8_t>(n_);
+    d_flux_ = dmalloc<double>(mesh_.nelems);
+    d_lost_ = dmalloc<unsigned long long>(1);
+    d_origin_ = dmalloc<double>(n_ * 3);
+    d_dest_ = dmalloc<double>(n_ * 3);
+    d_flying_ = dmalloc<int8_t>(n_);
+    d_weights_ = dmalloc<double>(n_);
+    PT_HIP_CHECK(hipMemset(d_flux_, 0, mesh_.nelems * sizeof(double)));
+    PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
+
+    loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+    const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
+    k_init_particles<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
+        d_pos_, d_elem_, d_escaped_, n_, c0.x, c0.y, c0.z);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+  }
+
+  ~GpuEngine() override {
+    (void)hipSetDevice(device_);
+    for (auto &r : registered_) (void)hipHostUnregister((void *)r.first);
+    for (void *p : {(void *)d_planes_, (void *)d_nbr_, (void *)d_cell_start_,
+                    (void *)d_cell_tets_, (void *)d_pos_, (void *)d_elem_,
+                    (void *)d_escaped_, (void *)d_flux_, (void *)d_lost_,
+                    (void *)d_origin_, (void *)d_dest_, (void *)d_flying_,
+                    (void *)d_weights_})
+      (void)hipFree(p);
+    for (auto &ev : events_) (void)hipEventDestroy(ev);
+    (void)hipStreamDestroy(s_copy_);
+    (void)hipStreamDestroy(s_comp_);
+  }
+
+  int64_t num_particles() const override { return n_; }
+  const Mesh &mesh() const override { return mesh_; }
+
+  void copy_initial_position(const double *p, int64_t n) override {
+    check_n(n);
+    PT_HIP_CHECK(hipSetDevice(device_));
+    stage(p, n * 3 * sizeof(double), d_origin_, s_copy_);
+    PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
+    k_locate<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
+        d_planes_, grid_view_, d_origin_, d_pos_, d_elem_, d_escaped_, 0, n_,
+        loc_tol_);
+    PT_HIP_CHECK(hipGetLastError());
+  }
+
+  void move(const double *origin, const double *dest, const int8_t *flying,
+            const double *weights, int64_t n) override {
+    check_n(n);
+    PT_HIP_CHECK(hipSetDevice(device_));
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    // Chunked H2D/compute pipeline: copy chunk k+1 while walking chunk k.
+    const int64_t chunk = std::max<int64_t>(1 << 20, (n + 7) / 8);
+    const int nchunks = (int)((n + chunk - 1) / chunk);
+    for (int k = 0; k < nchunks; ++k) {
+      const int64_t lo = k * chunk, hi = std::min(n, lo + chunk);
+      const int64_t m = hi - lo;
+      stage(origin + lo * 3, m * 3 * sizeof(double), d_origin_ + lo * 3, s_copy_);
+      stage(dest + lo * 3, m * 3 * sizeof(double), d_dest_ + lo * 3, s_copy_);
+      stage(flying + lo, m * sizeof(int8_t), d_flying_ + lo, s_copy_);
+      stage(weights + lo, m * sizeof(double), d_weights_ + lo, s_copy_);
+      hipEvent_t ev = events_[k % events_.size()];
+      PT_HIP_CHECK(hipEventRecord(ev, s_copy_));
+      PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, ev, 0));
+      k_move<<<grid_blocks(m), kBlock, 0, s_comp_>>>(
+          d_planes_, d_nbr_, grid_view_, d_origin_, d_dest_, d_flying_,
+          d_weights_, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi,
+          loc_tol_, steps);
+      PT_HIP_CHECK(hipGetLastError());
+    }
+    stats_.moves++;
+  }
+
+  std::vector<double> flux() const override {
+    sync();
+    std::vector<double> out(mesh_.nelems);
+    PT_HIP_CHECK(hipMemcpy(out.data(), d_flux_, mesh_.nelems * sizeof(double),
+                           hipMemcpyDeviceToHost));
+    return out;
+  }
+  std::vector<int32_t> elem_ids() const override {
+    sync();
+    std::vector<int32_t> out(n_);
+    PT_HIP_CHECK(hipMemcpy(out.data(), d_elem_, n_ * sizeof(int32_t),
+                           hipMemcpyDeviceToHost));
+    return out;
+  }
+  std::vector<double> positions() const override {
+    sync();
+    std::vector<double> out(n_ * 3);
+    PT_HIP_CHECK(hipMemcpy(out.data(), d_pos_, n_ * 3 * sizeof(double),
+                           hipMemcpyDeviceToHost));
+    return out;
+  }
+  std::vector<uint8_t> escaped() const override {
+    sync();
+    std::vector<uint8_t> out(n_);
+    PT_HIP_CHECK(hipMemcpy(out.data(), d_escaped_, n_, hipMemcpyDeviceToHost));
+    return out;
+  }
+
+  const EngineStats &stats() const override {
+    sync();
+    unsigned long long lost = 0;
+    PT_HIP_CHECK(hipMemcpy(&lost, d_lost_, sizeof lost, hipMemcpyDeviceToHost));
+    stats_.lost_particles = (int64_t)lost;
+    return stats_;
+  }
+
+  void set_flux(const double *f, int64_t ne) override {
+    if (ne != mesh_.nelems) throw std::runtime_error("set_flux size mismatch");
+    PT_HIP_CHECK(hipMemcpy(d_flux_, f, ne * sizeof(double), hipMemcpyHostToDevice));
+  }
+
+  void synchronize() override { sync(); }
+
+  // Direct device access for the Python/bench layer (zero-copy paths).
+  double *device_flux() const { return d_flux_; }
+
+private:
+  void check_n(int64_t n) const {
+    if (n != n_) throw std::runtime_error("particle count mismatch");
+  }
+
+  void sync() const {
+    PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+  }
+
+  // Async H2D from caller memory.  Registers (page-locks) the caller's
+  // buffer once per (pointer,length) so repeated per-step calls copy at
+  // full PCIe/xGMI rate with no host-side staging memcpy.
+  void stage(const void *src, size_t bytes, void *dst, hipStream_t s) {
+    ensure_registered(src, bytes);
+    PT_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, s));
+  }
+
+  void ensure_registered(const void *p, size_t bytes) {
+    auto it = registered_.lower_bound(p);
+    if (it != registered_.end() && it->first == p && it->second >= bytes) return;
+    // containment in a previously registered larger region
+    if (it != registered_.begin()) {
+      auto prev = std::prev(it);
+      if ((const char *)prev->first + prev->second >= (const char *)p + bytes)
+        return;
+    }
+    hipError_t e = hipHostRegister((void *)p, bytes, hipHostRegisterDefault);
+    if (e == hipSuccess) {
+      registered_[p] = bytes;
+    } else if (e != hipErrorHostMemoryAlreadyRegistered) {
+      (void)hipGetLastError(); // clear; fall back to pageable copy path
+    } else {
+      (void)hipGetLastError();
+    }
+  }
+
+  Mesh mesh_;
+  int64_t n_;
+  int device_ = 0;
+  double loc_tol_ = 1e-12;
+
+  hipStream_t s_copy_{}, s_comp_{};
+  std::array<hipEvent_t, 8> events_{};
+
+  Plane *d_planes_ = nullptr;
+  int32_t *d_nbr_ = nullptr;
+  int32_t *d_cell_start_ = nullptr;
+  int32_t *d_cell_tets_ = nullptr;
+  GridView grid_view_{};
+
+  double *d_pos_ = nullptr;
+  int32_t *d_elem_ = nullptr;
+  uint8_t *d_escaped_ = nullptr;
+  double *d_flux_ = nullptr;
+  unsigned long long *d_lost_ = nullptr;
+  double *d_origin_ = nullptr, *d_dest_ = nullptr, *d_weights_ = nullptr;
+  int8_t *d_flying_ = nullptr;
+
+  std::map<const void *, size_t> registered_;
+  mutable EngineStats stats_;
+};
+
+} // namespace
+
+std::unique_ptr<Engine> make_gpu_engine(Mesh mesh, int64_t num_particles,
+                                        int device) {
+  int count = 0;
+  if (hipGetDeviceCount(&count) != hipSuccess || count <= device) {
+    (void)hipGetLastError();
+    return nullptr;
+  }
+  return std::make_unique<GpuEngine>(std::move(mesh), num_particles, device);
+}
+
+} // namespace pumitally

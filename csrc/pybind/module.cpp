@@ -1,0 +1,252 @@
+// Python bindings: pumiumtally_amd._core
+//
+// Thin numpy-in/numpy-out layer over the C++ core.  No torch dependency --
+// the distributed driver (pumiumtally_amd.parallel) uses torch.distributed
+// on top of these bindings for RCCL collectives.
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "../api/PumiTally.h"
+#include "../core/engine.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cstring>
+#include <memory>
+
+namespace py = pybind11;
+using namespace pumitally;
+
+namespace {
+
+bool have_gpu() {
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) {
+    (void)hipGetLastError();
+    return false;
+  }
+  return n > 0;
+}
+
+py::array_t<double> vec_to_np(std::vector<double> v) {
+  auto out = py::array_t<double>(v.size());
+  std::memcpy(out.mutable_data(), v.data(), v.size() * sizeof(double));
+  return out;
+}
+
+struct PyEngine {
+  std::unique_ptr<Engine> eng;
+  bool gpu = false;
+
+  PyEngine(const Mesh &mesh, int64_t n, const std::string &device) {
+    if (device == "cpu") {
+      eng = make_cpu_engine(mesh, n);
+    } else {
+      int ordinal = 0;
+      if (device.rfind("cuda:", 0) == 0) ordinal = std::stoi(device.substr(5));
+      else if (device.rfind("gpu:", 0) == 0) ordinal = std::stoi(device.substr(4));
+      else if (device != "cuda" && device != "gpu" && device != "auto")
+        throw std::runtime_error("device must be cpu/cuda[:N]/auto");
+      eng = make_gpu_engine(mesh, n, ordinal);
+      if (eng) {
+        gpu = true;
+      } else if (device == "auto") {
+        eng = make_cpu_engine(mesh, n);
+      } else {
+        // Fail loudly: a GPU was requested but none is usable.  GPU tests
+        // must never fall back silently to the CPU oracle.
+        throw std::runtime_error("no usable HIP device for device=" + device);
+      }
+    }
+  }
+};
+
+// Pinned host allocation exposed as a numpy array (zero-copy H2D staging for
+// the bench / host transport codes).  Falls back to pageable memory when no
+// GPU is present so the same code runs on CPU-only machines.
+py::array pinned_array(py::object shape_obj, const std::string &dtype) {
+  std::vector<py::ssize_t> shape;
+  if (py::isinstance<py::int_>(shape_obj)) {
+    shape.push_back(shape_obj.cast<py::ssize_t>());
+  } else {
+    for (auto s : shape_obj.cast<py::sequence>())
+      shape.push_back(s.cast<py::ssize_t>());
+  }
+  py::dtype dt = py::dtype(dtype);
+  py::ssize_t count = 1;
+  for (auto s : shape) count *= s;
+  const size_t bytes = (size_t)count * dt.itemsize();
+  void *p = nullptr;
+  if (have_gpu()) {
+    if (hipHostMalloc(&p, bytes ? bytes : 1, hipHostMallocDefault) != hipSuccess)
+      throw std::runtime_error("hipHostMalloc failed");
+    py::capsule owner(p, [](void *q) { (void)hipHostFree(q); });
+    return py::array(dt, shape, p, owner);
+  }
+  p = ::malloc(bytes ? bytes : 1);
+  py::capsule owner(p, [](void *q) { ::free(q); });
+  return py::array(dt, shape, p, owner);
+}
+
+} // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "MI355X-native unstructured-mesh track-length tally engine";
+
+  py::class_<Mesh>(m, "Mesh")
+      .def_property_readonly("nelems", [](const Mesh &m_) { return m_.nelems; })
+      .def_property_readonly("nverts", [](const Mesh &m_) { return m_.nverts; })
+      .def_property_readonly("coords",
+                             [](const Mesh &m_) {
+                               auto a = py::array_t<double>({m_.nverts, (int64_t)3});
+                               std::memcpy(a.mutable_data(), m_.coords.data(),
+                                           m_.coords.size() * sizeof(double));
+                               return a;
+                             })
+      .def_property_readonly("tet2vert",
+                             [](const Mesh &m_) {
+                               auto a = py::array_t<int32_t>({m_.nelems, (int64_t)4});
+                               std::memcpy(a.mutable_data(), m_.tet2vert.data(),
+                                           m_.tet2vert.size() * sizeof(int32_t));
+                               return a;
+                             })
+      .def_property_readonly("neighbors",
+                             [](const Mesh &m_) {
+                               auto a = py::array_t<int32_t>({m_.nelems, (int64_t)4});
+                               std::memcpy(a.mutable_data(), m_.nbr.data(),
+                                           m_.nbr.size() * sizeof(int32_t));
+                               return a;
+                             })
+      .def_property_readonly("volumes",
+                             [](const Mesh &m_) { return vec_to_np(m_.volumes); })
+      .def("centroid",
+           [](const Mesh &m_, int32_t t) {
+             const Vec3 c = m_.centroid(t);
+             return py::make_tuple(c.x, c.y, c.z);
+           })
+      .def("locate",
+           [](const Mesh &m_, py::array_t<double, py::array::c_style | py::array::forcecast> pts) {
+             const int64_t n = pts.size() / 3;
+             auto out = py::array_t<int32_t>(n);
+             const double *p = pts.data();
+             const double tol = 1e-10 * norm(m_.bbox_hi - m_.bbox_lo);
+             for (int64_t i = 0; i < n; ++i)
+               out.mutable_data()[i] = m_.locate({p[i * 3], p[i * 3 + 1], p[i * 3 + 2]}, tol);
+             return out;
+           })
+      .def("write_vtk",
+           [](const Mesh &m_, const std::string &path) { write_vtk(path, m_, {}); })
+      .def("write_osh",
+           [](const Mesh &m_, const std::string &dir) { write_osh(dir, m_); });
+
+  m.def("build_box", &build_box, py::arg("nx"), py::arg("ny"), py::arg("nz"),
+        py::arg("lx") = 1.0, py::arg("ly") = 1.0, py::arg("lz") = 1.0);
+  m.def("read_mesh", &read_mesh);
+  m.def("read_gmsh", &read_gmsh);
+  m.def("read_osh", &read_osh);
+  m.def("mesh_from_arrays",
+        [](py::array_t<double, py::array::c_style | py::array::forcecast> coords,
+           py::array_t<int32_t, py::array::c_style | py::array::forcecast> tets) {
+          return mesh_from_arrays(coords.size() / 3, coords.data(),
+                                  tets.size() / 4, tets.data());
+        });
+  m.def("have_gpu", &have_gpu);
+  m.def("pinned_array", &pinned_array, py::arg("shape"), py::arg("dtype") = "float64");
+  m.def("normalize_flux", [](const Mesh &m_, py::array_t<double, py::array::c_style | py::array::forcecast> f) {
+    std::vector<double> flux(f.data(), f.data() + f.size());
+    return vec_to_np(normalize_flux(m_, flux));
+  });
+  m.def("write_tally_vtk",
+        [](const std::string &path, const Mesh &m_,
+           py::array_t<double, py::array::c_style | py::array::forcecast> f) {
+          std::vector<double> flux(f.data(), f.data() + f.size());
+          write_tally_vtk(path, m_, flux);
+        });
+
+  py::class_<PyEngine>(m, "Engine")
+      .def(py::init<const Mesh &, int64_t, const std::string &>(),
+           py::arg("mesh"), py::arg("num_particles"), py::arg("device") = "auto")
+      .def_property_readonly("num_particles",
+                             [](const PyEngine &e) { return e.eng->num_particles(); })
+      .def_property_readonly("is_gpu", [](const PyEngine &e) { return e.gpu; })
+      .def_property(
+          "max_steps", [](const PyEngine &e) { return e.eng->max_steps; },
+          [](PyEngine &e, int v) { e.eng->max_steps = v; })
+      .def("copy_initial_position",
+           [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> p) {
+             if ((int64_t)p.size() != e.eng->num_particles() * 3)
+               throw std::runtime_error("positions must have size 3*num_particles");
+             e.eng->copy_initial_position(p.data(), e.eng->num_particles());
+           })
+      .def("move",
+           [](PyEngine &e, py::array_t<double, py::array::c_style> origin,
+              py::array_t<double, py::array::c_style> dest,
+              py::array_t<int8_t, py::array::c_style> flying,
+              py::array_t<double, py::array::c_style> weights) {
+             const int64_t n = e.eng->num_particles();
+             if ((int64_t)origin.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
+                 (int64_t)flying.size() != n || (int64_t)weights.size() != n)
+               throw std::runtime_error("move: array size mismatch");
+             py::gil_scoped_release nogil;
+             e.eng->move(origin.data(), dest.data(), flying.data(),
+                         weights.data(), n);
+           })
+      .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
+      .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
+      .def("elem_ids",
+           [](const PyEngine &e) {
+             auto v = e.eng->elem_ids();
+             auto a = py::array_t<int32_t>(v.size());
+             std::memcpy(a.mutable_data(), v.data(), v.size() * sizeof(int32_t));
+             return a;
+           })
+      .def("positions",
+           [](const PyEngine &e) {
+             auto v = e.eng->positions();
+             auto a = py::array_t<double>({(int64_t)(v.size() / 3), (int64_t)3});
+             std::memcpy(a.mutable_data(), v.data(), v.size() * sizeof(double));
+             return a;
+           })
+      .def("escaped",
+           [](const PyEngine &e) {
+             auto v = e.eng->escaped();
+             auto a = py::array_t<uint8_t>(v.size());
+             std::memcpy(a.mutable_data(), v.data(), v.size());
+             return a;
+           })
+      .def("set_flux",
+           [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> f) {
+             e.eng->set_flux(f.data(), (int64_t)f.size());
+           })
+      .def("stats", [](const PyEngine &e) {
+        const EngineStats &s = e.eng->stats();
+        py::dict d;
+        d["lost_particles"] = s.lost_particles;
+        d["moves"] = s.moves;
+        d["relocated"] = s.relocated;
+        return d;
+      });
+
+  // The 4-call C++ facade, for API-parity tests from Python.
+  py::class_<PumiTally>(m, "PumiTally")
+      .def(py::init([](const std::string &mesh, int32_t n) {
+        int argc = 0;
+        char **argv = nullptr;
+        return new PumiTally(mesh, n, argc, argv);
+      }))
+      .def("copy_initial_position",
+           [](PumiTally &t, py::array_t<double, py::array::c_style | py::array::forcecast> p) {
+             t.CopyInitialPosition((double *)p.data(), (int32_t)p.size());
+           })
+      .def("move_to_next_location",
+           [](PumiTally &t, py::array_t<double, py::array::c_style> origin,
+              py::array_t<double, py::array::c_style> dest,
+              py::array_t<int8_t, py::array::c_style> flying,
+              py::array_t<double, py::array::c_style> weights) {
+             t.MoveToNextLocation((double *)origin.data(), (double *)dest.data(),
+                                  (int8_t *)flying.mutable_data(),
+                                  (double *)weights.data(), (int32_t)origin.size());
+           })
+      .def("write_tally_results", &PumiTally::WriteTallyResults);
+}

@@ -1,0 +1,125 @@
+"""pumiumtally_amd: MI355X-native unstructured-mesh track-length tally engine.
+
+A from-scratch re-implementation of the capabilities of Fuad-HH/PumiUMTally
+(PUMI-Tally) for AMD Instinct MI355X: track-length flux tallies on
+unstructured tetrahedral meshes for Monte Carlo particle transport, with the
+element walk as a hand-written CDNA4 HIP kernel, fp64 throughout, and
+multi-GPU scaling over RCCL (see pumiumtally_amd.parallel).
+
+Public surface:
+    Mesh / build_box / read_mesh / read_gmsh / read_osh   mesh core
+    TallyEngine                                            engine (CPU or GPU)
+    PumiTally                                              4-call C++ API parity facade
+    parallel.DistributedTally                              multi-GPU driver
+"""
+from __future__ import annotations
+
+import os as _os
+
+
+def _load_core():
+    try:
+        from . import _core  # type: ignore
+        return _core
+    except ImportError as e:
+        if _os.environ.get("PUMITALLY_AUTOBUILD", "1") == "1":
+            from ._build import build
+            build()
+            from . import _core  # type: ignore
+            return _core
+        raise ImportError(
+            "pumiumtally_amd._core is not built. Run "
+            "`python -m pumiumtally_amd._build` first."
+        ) from e
+
+
+_core = _load_core()
+
+Mesh = _core.Mesh
+PumiTally = _core.PumiTally
+build_box = _core.build_box
+read_mesh = _core.read_mesh
+read_gmsh = _core.read_gmsh
+read_osh = _core.read_osh
+mesh_from_arrays = _core.mesh_from_arrays
+have_gpu = _core.have_gpu
+pinned_array = _core.pinned_array
+normalize_flux = _core.normalize_flux
+write_tally_vtk = _core.write_tally_vtk
+
+
+class TallyEngine:
+    """Track-length tally engine over a tet mesh.
+
+    device: "auto" (GPU if present, else CPU), "cpu", or "cuda[:N]".
+    Semantics follow the reference PumiTally flow; see csrc/core/engine.h.
+    """
+
+    def __init__(self, mesh, num_particles: int, device: str = "auto"):
+        self._eng = _core.Engine(mesh, num_particles, device)
+        self.mesh = mesh
+
+    @property
+    def num_particles(self) -> int:
+        return self._eng.num_particles
+
+    @property
+    def is_gpu(self) -> bool:
+        return self._eng.is_gpu
+
+    @property
+    def max_steps(self) -> int:
+        return self._eng.max_steps
+
+    @max_steps.setter
+    def max_steps(self, v: int) -> None:
+        self._eng.max_steps = v
+
+    def copy_initial_position(self, positions):
+        self._eng.copy_initial_position(positions)
+
+    def move(self, origin, dest, flying, weights):
+        self._eng.move(origin, dest, flying, weights)
+
+    def synchronize(self):
+        self._eng.synchronize()
+
+    def flux(self):
+        return self._eng.flux()
+
+    def set_flux(self, flux):
+        self._eng.set_flux(flux)
+
+    def elem_ids(self):
+        return self._eng.elem_ids()
+
+    def positions(self):
+        return self._eng.positions()
+
+    def escaped(self):
+        return self._eng.escaped()
+
+    def stats(self):
+        return self._eng.stats()
+
+    def normalized_flux(self):
+        return _core.normalize_flux(self.mesh, self._eng.flux())
+
+    def write_tally_results(self, filename: str = "fluxresult.vtk"):
+        _core.write_tally_vtk(filename, self.mesh, self._eng.flux())
+
+
+__all__ = [
+    "Mesh",
+    "PumiTally",
+    "TallyEngine",
+    "build_box",
+    "read_mesh",
+    "read_gmsh",
+    "read_osh",
+    "mesh_from_arrays",
+    "have_gpu",
+    "pinned_array",
+    "normalize_flux",
+    "write_tally_vtk",
+]

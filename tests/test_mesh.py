@@ -1,0 +1,130 @@
+"""Mesh core tests: topology, adjacency, geometry, IO.
+
+Mirrors the reference's analytic-fixture strategy: the
+6-tet unit cube from build_box is the golden mesh, with hand-computed
+volumes, adjacency and localization expectations.
+"""
+import numpy as np
+import pytest
+
+import pumiumtally_amd as pt
+
+
+def test_build_box_unit_cube():
+    m = pt.build_box(1, 1, 1)
+    assert m.nelems == 6
+    assert m.nverts == 8
+    # element 0 centroid pinned by the reference tests
+    assert np.allclose(m.centroid(0), (0.5, 0.75, 0.25))
+    v = m.volumes
+    assert np.allclose(v, 1.0 / 6.0)
+    assert abs(v.sum() - 1.0) < 1e-14
+
+
+def test_build_box_adjacency_symmetric():
+    m = pt.build_box(3, 2, 4, 3.0, 2.0, 4.0)
+    nbr = m.neighbors
+    assert m.nelems == 3 * 2 * 4 * 6
+    assert np.allclose(m.volumes.sum(), 3.0 * 2.0 * 4.0)
+    n_boundary = 0
+    for t in range(m.nelems):
+        for f in range(4):
+            o = nbr[t, f]
+            if o == -1:
+                n_boundary += 1
+            else:
+                assert t in nbr[o], f"adjacency not symmetric at tet {t} face {f}"
+    # surface of a box cut into 6-tet cells: every boundary quad face of a
+    # cell contributes 2 triangles
+    expected_boundary = 2 * 2 * (3 * 2 + 2 * 4 + 3 * 4)
+    assert n_boundary == expected_boundary
+
+
+def test_positive_orientation():
+    m = pt.build_box(2, 2, 2)
+    assert (m.volumes > 0).all()
+
+
+def test_locate():
+    m = pt.build_box(1, 1, 1)
+    # coordinate-ordering regions pinned by the reference tests
+    pts = np.array([
+        [0.1, 0.4, 0.5],   # z>=y>=x -> el 2
+        [0.9, 0.4, 0.5],   # x>=z>=y -> el 4
+        [0.45, 0.4, 0.5],  # z>=x>=y -> el 3
+        [0.5, 0.75, 0.25], # el 0 centroid
+    ])
+    ids = m.locate(pts)
+    assert list(ids[:3]) == [2, 4, 3]
+    assert ids[3] == 0
+    outside = m.locate(np.array([[1.5, 0.5, 0.5], [-0.1, 0.2, 0.2]]))
+    assert list(outside) == [-1, -1]
+
+
+def test_locate_random_consistency():
+    m = pt.build_box(4, 4, 4, 2.0, 2.0, 2.0)
+    rng = np.random.default_rng(0)
+    pts = rng.uniform(0.001, 1.999, size=(500, 3))
+    ids = m.locate(pts)
+    assert (ids >= 0).all()
+    # each located tet must actually contain its point (barycentric check)
+    coords = m.coords
+    tets = m.tet2vert
+    for p, t in zip(pts, ids):
+        vs = coords[tets[t]]
+        mat = np.column_stack([vs[1] - vs[0], vs[2] - vs[0], vs[3] - vs[0]])
+        bary = np.linalg.solve(mat, p - vs[0])
+        assert bary.min() > -1e-9 and bary.sum() < 1 + 1e-9
+
+
+def test_osh_roundtrip(tmp_path):
+    m = pt.build_box(2, 3, 1)
+    d = str(tmp_path / "mesh.osh")
+    m.write_osh(d)
+    m2 = pt.read_osh(d)
+    assert m2.nelems == m.nelems
+    assert np.array_equal(m2.tet2vert, m.tet2vert)
+    assert np.array_equal(m2.coords, m.coords)
+    # read_mesh dispatch on extension, with or without trailing slash
+    m3 = pt.read_mesh(d)
+    assert m3.nelems == m.nelems
+    m4 = pt.read_mesh(d + "/")
+    assert m4.nelems == m.nelems
+
+
+def test_gmsh_v2_reader(tmp_path):
+    # unit cube corners; single tet mesh written as Gmsh 2.2 ASCII
+    msh = tmp_path / "t.msh"
+    msh.write_text(
+        "$MeshFormat\n2.2 0 8\n$EndMeshFormat\n"
+        "$Nodes\n4\n1 0 0 0\n2 1 0 0\n3 0 1 0\n4 0 0 1\n$EndNodes\n"
+        "$Elements\n2\n1 2 2 0 1 1 2 3\n2 4 2 0 1 1 2 3 4\n$EndElements\n"
+    )
+    m = pt.read_gmsh(str(msh))
+    assert m.nelems == 1
+    assert m.nverts == 4
+    assert np.allclose(m.volumes, [1.0 / 6.0])
+
+
+def test_gmsh_v41_reader(tmp_path):
+    msh = tmp_path / "t41.msh"
+    msh.write_text(
+        "$MeshFormat\n4.1 0 8\n$EndMeshFormat\n"
+        "$Nodes\n1 4 1 4\n3 1 0 4\n1\n2\n3\n4\n"
+        "0 0 0\n1 0 0\n0 1 0\n0 0 1\n$EndNodes\n"
+        "$Elements\n1 1 1 1\n3 1 4 1\n1 1 2 3 4\n$EndElements\n"
+    )
+    m = pt.read_gmsh(str(msh))
+    assert m.nelems == 1
+    assert np.allclose(m.volumes, [1.0 / 6.0])
+
+
+def test_vtk_output(tmp_path):
+    m = pt.build_box(1, 1, 1)
+    out = str(tmp_path / "flux.vtk")
+    pt.write_tally_vtk(out, m, np.arange(6, dtype=float))
+    text = open(out).read()
+    assert "UNSTRUCTURED_GRID" in text
+    assert "SCALARS flux double" in text
+    assert "SCALARS volume double" in text
+    assert "CELL_DATA 6" in text
