@@ -1,0 +1,158 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: track-length tally throughput (particle-steps/sec).
+
+BASELINE.json metric: particle-steps/sec (whole node) on a 1M-tet mesh.
+One "step" = one MoveToNextLocation over the full particle batch: H2D
+staging of origin/dest/flying/weights + the fused walk+tally kernel.
+Synthetic straight-line histories (no network for datasets), fp64
+positions/tallies (the reference's compute dtype).
+
+Single GPU:   python bench.py --steps 20 --warmup 5
+Multi-GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+Scaling mode is WEAK: each rank owns its own --particles batch on a
+replicated mesh; the only collective is one flux all-reduce at the end
+(outside the timed region, like the reference which reduces only at
+WriteTallyResults).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1, help="informational; actual world size comes from torchrun env")
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--particles", type=int, default=10_000_000, help="particles per GPU")
+    ap.add_argument("--mesh-tets", type=int, default=1_000_000)
+    ap.add_argument("--mean-chord", type=float, default=8.0, help="target mean element crossings per step")
+    ap.add_argument("--device", type=str, default=None, help="cpu / cuda:N (default: auto)")
+    ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--write-vtk", type=str, default=None)
+    args = ap.parse_args()
+
+    import numpy as np
+
+    import pumiumtally_amd as pt
+    from pumiumtally_amd.parallel import init_distributed
+    from pumiumtally_amd.utils import make_box_histories
+
+    rank, world, local = init_distributed()
+    on_gpu = pt.have_gpu() and args.device != "cpu"
+    if not on_gpu and args.device is None:
+        # CPU fallback (debug only): shrink to something a serial walk finishes
+        args.particles = min(args.particles, 20_000)
+        args.mesh_tets = min(args.mesh_tets, 50_000)
+
+    device = args.device or (f"cuda:{local}" if on_gpu else "cpu")
+
+    t0 = time.time()
+    from pumiumtally_amd.mesh import box_mesh_with_tets
+    mesh, cells = box_mesh_with_tets(args.mesh_tets, extent=1.0)
+    if rank == 0:
+        print(f"[bench] mesh: {mesh.nelems} tets ({cells}^3 cells), built in {time.time()-t0:.1f}s",
+              file=sys.stderr, flush=True)
+
+    eng = pt.TallyEngine(mesh, args.particles, device=device)
+    p0, p1, flying, weights = make_box_histories(
+        (1.0, 1.0, 1.0), args.particles, args.mean_chord, cells,
+        seed=args.seed + rank, pinned=eng.is_gpu)
+    eng.copy_initial_position(p0.reshape(-1))
+    eng.synchronize()
+    n_located = int((eng.elem_ids() >= 0).sum()) if args.particles <= 1_000_000 else args.particles
+
+    ends = (p0.reshape(-1), p1.reshape(-1))
+
+    def step(k):
+        # ping-pong: walk P0->P1, then P1->P0; origin == current position so
+        # phase A is a no-op compare, phase B walks the full segment set.
+        o, d = ends[k % 2], ends[(k + 1) % 2]
+        eng.move(o, d, flying, weights)
+
+    def barrier_sync():
+        eng.synchronize()
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+
+    for k in range(args.warmup):
+        step(k)
+    barrier_sync()
+
+    t_start = time.time()
+    for k in range(args.warmup, args.warmup + args.steps):
+        step(k)
+    eng.synchronize()
+    elapsed_local = time.time() - t_start
+    barrier_sync()
+
+    # MAX over ranks (slowest rank defines throughput)
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+        t = torch.tensor([elapsed_local], dtype=torch.float64)
+        if dist.get_backend() == "nccl":
+            t = t.cuda(local)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu().item())
+    else:
+        elapsed = elapsed_local
+
+    stats = eng.stats()
+    total_particle_steps = args.particles * args.steps * world
+    value = total_particle_steps / elapsed
+
+    # flux all-reduce + optional write (outside timed region, parity with
+    # the reference's single reduction at WriteTallyResults)
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+        f = torch.from_numpy(eng.flux())
+        if dist.get_backend() == "nccl":
+            f = f.cuda(local)
+        dist.all_reduce(f, op=dist.ReduceOp.SUM)
+        global_flux = f.cpu().numpy()
+    else:
+        global_flux = eng.flux()
+    if args.write_vtk and rank == 0:
+        pt.write_tally_vtk(args.write_vtk, mesh, global_flux)
+
+    if rank == 0:
+        result = {
+            "metric": "particle-steps/sec",
+            "value": value,
+            "unit": "particle-steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "track-length-tally-walk",
+                "mesh_tets": int(mesh.nelems),
+                "global_batch": args.particles * world,
+                "particles_per_gpu": args.particles,
+                "mean_chord_elems": args.mean_chord,
+                "seq_len": None,
+                "parallelism": f"dp{world}-replicated-mesh",
+                "device": "gpu" if eng.is_gpu else "cpu",
+                "lost_particles": stats["lost_particles"],
+                "flux_sum": float(global_flux.sum()),
+            },
+        }
+        print(json.dumps(result), flush=True)
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
