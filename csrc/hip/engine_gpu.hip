@@ -83,6 +83,14 @@ __global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
 
 // The fused move kernel: phase A (relocation of flying, non-escaped
 // particles whose origin changed) + phase B (tallied walk to destination).
+//
+// Block->particle mapping is XCD-aware: MI355X dispatches block b to XCD
+// b%8, and each XCD has a private 4 MiB L2.  We remap blocks so each XCD
+// owns one contiguous particle range; when the caller orders particles
+// spatially (Morton order -- pumiumtally_amd.utils.synthetic does), each
+// XCD's L2 then holds one compact mesh region instead of 8 interleaved
+// copies of the whole traffic.  Purely a speed lever (guide G/T1): any
+// placement is correct.
 __global__ void k_move(const Plane *__restrict__ planes,
                        const int32_t *__restrict__ nbr, GridView grid,
                        const double *__restrict__ origin,
@@ -94,9 +102,15 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        double *__restrict__ flux,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = lo + blockIdx.x * blockDim.x + threadIdx.x; i < hi;
-       i += stride) {
+  // gridDim.x is a multiple of 8 (grid_blocks).  Virtual block id: XCD x
+  // gets contiguous block slots [x*bpx, (x+1)*bpx).
+  const unsigned bpx = gridDim.x / 8u;
+  const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
+  const int64_t m = hi - lo;
+  const int64_t per_blk = (m + gridDim.x - 1) / gridDim.x;
+  const int64_t base = lo + (int64_t)vb * per_blk;
+  const int64_t end = base + per_blk < hi ? base + per_blk : hi;
+  for (int64_t i = base + threadIdx.x; i < end; i += blockDim.x) {
     if (!flying[i]) continue;
     Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
     int32_t e = elem[i];
@@ -137,9 +151,11 @@ __global__ void k_move(const Plane *__restrict__ planes,
 }
 
 int grid_blocks(int64_t work) {
-  const int64_t blocks = (work + kBlock - 1) / kBlock;
-  // 256 CUs x 8 blocks/CU: cap and grid-stride the rest (guide G11).
-  return (int)(blocks < 2048 ? (blocks < 1 ? 1 : blocks) : 2048);
+  int64_t blocks = (work + kBlock - 1) / kBlock;
+  // 256 CUs x 8 blocks/CU: cap and block-stride the rest (guide G11).
+  if (blocks > 2048) blocks = 2048;
+  // round up to a multiple of 8 for the XCD-aware remap in k_move
+  return (int)((blocks + 7) / 8 * 8);
 }
 
 template <class T> T *dmalloc(int64_t count) {

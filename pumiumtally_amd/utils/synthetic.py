@@ -12,8 +12,19 @@ from __future__ import annotations
 import numpy as np
 
 
+def _morton_keys(p, box, bits=10):
+    """Interleaved-bit (Morton/Z-order) keys of points p in box."""
+    scale = (1 << bits) - 1
+    q = (p / np.asarray(box) * scale).astype(np.uint64)
+    key = np.zeros(len(p), dtype=np.uint64)
+    for b in range(bits):
+        for axis in range(3):
+            key |= ((q[:, axis] >> np.uint64(b)) & np.uint64(1)) << np.uint64(3 * b + axis)
+    return key
+
+
 def make_box_histories(box, n: int, mean_chord_elems: float, cells_per_axis: int,
-                       seed: int = 0, pinned: bool = True):
+                       seed: int = 0, pinned: bool = True, sort: bool = True):
     """Returns (p0, p1, flying, weights) arrays for n particles in a box
     mesh of `cells_per_axis` cells per axis over extents `box` (3-tuple).
 
@@ -40,7 +51,14 @@ def make_box_histories(box, n: int, mean_chord_elems: float, cells_per_axis: int
     weights = alloc((n,), "float64")
 
     margin = 1e-6 * box
-    p0[:] = rng.uniform(margin, box - margin, size=(n, 3))
+    start = rng.uniform(margin, box - margin, size=(n, 3))
+    if sort:
+        # Spatial (Morton) ordering: adjacent particles walk adjacent mesh
+        # regions, so each wave/XCD touches a compact working set (pairs
+        # with the XCD-aware block remap in the walk kernel).  Event-based
+        # MC codes sort their event queues the same way.
+        start = start[np.argsort(_morton_keys(start, box))]
+    p0[:] = start
     # isotropic directions
     u = rng.uniform(-1.0, 1.0, n)
     phi = rng.uniform(0.0, 2 * np.pi, n)
