@@ -59,6 +59,14 @@ public:
   virtual void move(const double *origin, const double *dest,
                     const int8_t *flying, const double *weights, int64_t n) = 0;
 
+  // Fast path for callers that know no particle was resampled this step
+  // (origin == committed position for every particle): skips the origin
+  // upload and phase A entirely.  origin=nullptr in move() semantics.
+  virtual void move_continue(const double *dest, const int8_t *flying,
+                             const double *weights, int64_t n) {
+    move(nullptr, dest, flying, weights, n);
+  }
+
   // Read back state (host copies).
   virtual std::vector<double> flux() const = 0;           // nelems, raw tally
   virtual std::vector<int32_t> elem_ids() const = 0;      // n

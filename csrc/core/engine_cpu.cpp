@@ -62,7 +62,7 @@ public:
       Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
       // Phase A: relocate to the given origin (no tally).  Skipped for
       // escaped particles (behavioral pin, see engine.h).
-      if (!escaped_[i]) {
+      if (origin && !escaped_[i]) {
         const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
         if (q.x != o.x || q.y != o.y || q.z != o.z) {
           elem_[i] = mesh_.locate(q, loc_tol_);

@@ -114,7 +114,7 @@ __global__ void k_move(const Plane *__restrict__ planes,
     if (!flying[i]) continue;
     Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
     int32_t e = elem[i];
-    if (!escaped[i]) {
+    if (origin != nullptr && !escaped[i]) {
       const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
       if (q.x != o.x || q.y != o.y || q.z != o.z) {
         e = grid_locate(grid, planes, q, loc_tol);
@@ -252,7 +252,8 @@ public:
     for (int k = 0; k < nchunks; ++k) {
       const int64_t lo = k * chunk, hi = std::min(n, lo + chunk);
       const int64_t m = hi - lo;
-      stage(origin + lo * 3, m * 3 * sizeof(double), d_origin_ + lo * 3, s_copy_);
+      if (origin)
+        stage(origin + lo * 3, m * 3 * sizeof(double), d_origin_ + lo * 3, s_copy_);
       stage(dest + lo * 3, m * 3 * sizeof(double), d_dest_ + lo * 3, s_copy_);
       stage(flying + lo, m * sizeof(int8_t), d_flying_ + lo, s_copy_);
       stage(weights + lo, m * sizeof(double), d_weights_ + lo, s_copy_);
@@ -260,9 +261,9 @@ public:
       PT_HIP_CHECK(hipEventRecord(ev, s_copy_));
       PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, ev, 0));
       k_move<<<grid_blocks(m), kBlock, 0, s_comp_>>>(
-          d_planes_, d_nbr_, grid_view_, d_origin_, d_dest_, d_flying_,
-          d_weights_, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi,
-          loc_tol_, steps);
+          d_planes_, d_nbr_, grid_view_, origin ? d_origin_ : nullptr, d_dest_,
+          d_flying_, d_weights_, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
+          lo, hi, loc_tol_, steps);
       PT_HIP_CHECK(hipGetLastError());
     }
     stats_.moves++;

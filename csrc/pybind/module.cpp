@@ -192,6 +192,17 @@ PYBIND11_MODULE(_core, m) {
              e.eng->move(origin.data(), dest.data(), flying.data(),
                          weights.data(), n);
            })
+      .def("move_continue",
+           [](PyEngine &e, py::array_t<double, py::array::c_style> dest,
+              py::array_t<int8_t, py::array::c_style> flying,
+              py::array_t<double, py::array::c_style> weights) {
+             const int64_t n = e.eng->num_particles();
+             if ((int64_t)dest.size() != n * 3 || (int64_t)flying.size() != n ||
+                 (int64_t)weights.size() != n)
+               throw std::runtime_error("move_continue: array size mismatch");
+             py::gil_scoped_release nogil;
+             e.eng->move_continue(dest.data(), flying.data(), weights.data(), n);
+           })
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

@@ -81,6 +81,11 @@ class TallyEngine:
     def move(self, origin, dest, flying, weights):
         self._eng.move(origin, dest, flying, weights)
 
+    def move_continue(self, dest, flying, weights):
+        """move() without the phase-A origin upload: valid when no particle
+        was resampled this step (origin == committed position)."""
+        self._eng.move_continue(dest, flying, weights)
+
     def synchronize(self):
         self._eng.synchronize()
 
