@@ -67,6 +67,16 @@ public:
     move(nullptr, dest, flying, weights, n);
   }
 
+  // Raw batched walk for the partitioned driver: walks n independent
+  // segments (pos->dest) starting in the given elements, tallying into
+  // this engine's flux.  status: 0=reached dest, 1=escaped (vacuum),
+  // 2=handoff (out_elem = encoded foreign ref -(2+k)), 3=lost.
+  // Synchronous; arrays are plain host memory.
+  virtual void walk_raw(int64_t n, const double *pos, const double *dest,
+                        const int32_t *elem, const double *weights,
+                        double *out_pos, int32_t *out_elem,
+                        int8_t *out_status) = 0;
+
   // Read back state (host copies).
   virtual std::vector<double> flux() const = 0;           // nelems, raw tally
   virtual std::vector<int32_t> elem_ids() const = 0;      // n

@@ -97,6 +97,37 @@ public:
     stats_.moves++;
   }
 
+  void walk_raw(int64_t n, const double *pos, const double *dest,
+                const int32_t *elem, const double *weights, double *out_pos,
+                int32_t *out_elem, int8_t *out_status) override {
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    for (int64_t i = 0; i < n; ++i) {
+      const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
+      const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+      int32_t oe;
+      Vec3 op;
+      bool esc;
+      walk_segment(
+          mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d, weights[i],
+          steps, [&](int32_t e, double v) { flux_[e] += v; }, &oe, &op, &esc);
+      int8_t st = 0;
+      if (oe == kWalkLost) {
+        st = 3;
+        oe = elem[i];
+        stats_.lost_particles++;
+      } else if (esc) {
+        st = 1;
+      } else if (oe < -1) {
+        st = 2;
+      }
+      out_elem[i] = oe;
+      out_pos[i * 3] = op.x;
+      out_pos[i * 3 + 1] = op.y;
+      out_pos[i * 3 + 2] = op.z;
+      out_status[i] = st;
+    }
+  }
+
   std::vector<double> flux() const override { return flux_; }
   std::vector<int32_t> elem_ids() const override { return elem_; }
   std::vector<double> positions() const override { return pos_; }

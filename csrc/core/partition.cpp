@@ -1,0 +1,104 @@
+// Element-ownership domain decomposition (see mesh.h for the contract).
+#include "mesh.h"
+
+#include <algorithm>
+#include <cstdint>
+#include <numeric>
+#include <stdexcept>
+
+namespace pumitally {
+
+namespace {
+uint64_t morton3(uint32_t x, uint32_t y, uint32_t z) {
+  auto spread = [](uint64_t v) {
+    v &= 0x1fffff;
+    v = (v | v << 32) & 0x1f00000000ffffull;
+    v = (v | v << 16) & 0x1f0000ff0000ffull;
+    v = (v | v << 8) & 0x100f00f00f00f00full;
+    v = (v | v << 4) & 0x10c30c30c30c30c3ull;
+    v = (v | v << 2) & 0x1249249249249249ull;
+    return v;
+  };
+  return spread(x) | (spread(y) << 1) | (spread(z) << 2);
+}
+} // namespace
+
+std::vector<int32_t> partition_morton(const Mesh &m, int nparts) {
+  if (nparts < 1) throw std::runtime_error("nparts must be >= 1");
+  std::vector<int64_t> order(m.nelems);
+  std::iota(order.begin(), order.end(), 0);
+  if (nparts > 1) {
+    const Vec3 ext = m.bbox_hi - m.bbox_lo;
+    const double sx = ext.x > 0 ? 2097151.0 / ext.x : 0.0;
+    const double sy = ext.y > 0 ? 2097151.0 / ext.y : 0.0;
+    const double sz = ext.z > 0 ? 2097151.0 / ext.z : 0.0;
+    std::vector<uint64_t> key(m.nelems);
+    for (int64_t t = 0; t < m.nelems; ++t) {
+      const Vec3 c = m.centroid((int32_t)t);
+      key[t] = morton3((uint32_t)((c.x - m.bbox_lo.x) * sx),
+                       (uint32_t)((c.y - m.bbox_lo.y) * sy),
+                       (uint32_t)((c.z - m.bbox_lo.z) * sz));
+    }
+    std::sort(order.begin(), order.end(),
+              [&](int64_t a, int64_t b) { return key[a] < key[b]; });
+  }
+  std::vector<int32_t> owners(m.nelems);
+  for (int64_t i = 0; i < m.nelems; ++i)
+    owners[order[i]] = (int32_t)((i * nparts) / m.nelems);
+  return owners;
+}
+
+SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
+                        int part) {
+  if ((int64_t)owners.size() != m.nelems)
+    throw std::runtime_error("owners size mismatch");
+  SubMesh sub;
+  for (int64_t g = 0; g < m.nelems; ++g)
+    if (owners[g] == part) sub.elem_l2g.push_back(g);
+  const int64_t ne = (int64_t)sub.elem_l2g.size();
+
+  // local vertex numbering, ascending in global id (keeps canonical face
+  // planes bitwise-identical to the full mesh)
+  std::vector<int32_t> used;
+  used.reserve(ne * 4);
+  for (int64_t t = 0; t < ne; ++t)
+    for (int k = 0; k < 4; ++k) used.push_back(m.tet2vert[sub.elem_l2g[t] * 4 + k]);
+  std::sort(used.begin(), used.end());
+  used.erase(std::unique(used.begin(), used.end()), used.end());
+  std::vector<int32_t> g2l_vert(m.nverts, -1);
+  for (size_t i = 0; i < used.size(); ++i) g2l_vert[used[i]] = (int32_t)i;
+
+  Mesh &local = sub.local;
+  local.nverts = (int64_t)used.size();
+  local.coords.resize(local.nverts * 3);
+  for (size_t i = 0; i < used.size(); ++i)
+    for (int k = 0; k < 3; ++k)
+      local.coords[i * 3 + k] = m.coords[(int64_t)used[i] * 3 + k];
+  local.nelems = ne;
+  local.tet2vert.resize(ne * 4);
+  for (int64_t t = 0; t < ne; ++t)
+    for (int k = 0; k < 4; ++k)
+      local.tet2vert[t * 4 + k] = g2l_vert[m.tet2vert[sub.elem_l2g[t] * 4 + k]];
+  local.finalize(); // positive orientation preserved -> no vertex swaps
+
+  // Mark cross-part faces.  After finalize(), any local face with nbr==-1
+  // is either a true mesh boundary or a partition cut; consult the global
+  // adjacency to tell them apart.
+  std::vector<int64_t> g2l_elem(m.nelems, -1);
+  for (int64_t t = 0; t < ne; ++t) g2l_elem[sub.elem_l2g[t]] = t;
+  for (int64_t t = 0; t < ne; ++t) {
+    const int64_t g = sub.elem_l2g[t];
+    for (int f = 0; f < 4; ++f) {
+      if (local.nbr[t * 4 + f] != -1) continue;
+      const int32_t gn = m.nbr[g * 4 + f];
+      if (gn == -1) continue; // true boundary
+      const int32_t k = (int32_t)sub.foreign_gid.size();
+      sub.foreign_gid.push_back(gn);
+      sub.foreign_owner.push_back(owners[gn]);
+      local.nbr[t * 4 + f] = -(2 + k);
+    }
+  }
+  return sub;
+}
+
+} // namespace pumitally

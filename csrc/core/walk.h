@@ -26,7 +26,9 @@
 namespace pumitally {
 
 constexpr double kWalkTEps = 1e-12;   // tolerance on the segment parameter t
-constexpr int kWalkLost = -2;         // out_elem value when max_steps hit
+// out_elem value when max_steps hit.  Must not collide with the encoded
+// foreign-element refs -(2+k) used by partitioned submeshes.
+constexpr int32_t kWalkLost = INT32_MIN;
 
 // FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
 // performs atomicAdd into the flux array; on the serial CPU path a plain +=.
@@ -81,13 +83,22 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
     if (tally) add(elem, (t_clamped - t_cur) * seg_len * weight);
 
     const int32_t next = nbr[(int64_t)elem * 4 + exit_face];
-    if (next < 0) {
+    if (next == -1) {
       // Vacuum boundary: clip the destination to the exit point; the
       // particle keeps its last element id (reference K6 semantics,
       // PumiTallyImpl.cpp:275-281 and the 1.0-not-1.1 test expectation).
       *out_elem = elem;
       *out_pos = o + t_clamped * (d - o);
       *out_escaped = true;
+      return;
+    }
+    if (next < -1) {
+      // Partitioned submesh: the face crosses into an element owned by
+      // another rank.  Stop at the crossing; the caller decodes the
+      // foreign reference (k = -(next+2)) and ships the particle.
+      *out_elem = next;
+      *out_pos = o + t_clamped * (d - o);
+      *out_escaped = false;
       return;
     }
     prev_elem = elem;

@@ -150,6 +150,46 @@ __global__ void k_move(const Plane *__restrict__ planes,
   }
 }
 
+__global__ void k_walk_raw(const Plane *__restrict__ planes,
+                           const int32_t *__restrict__ nbr,
+                           const double *__restrict__ pos,
+                           const double *__restrict__ dest,
+                           const int32_t *__restrict__ elem,
+                           const double *__restrict__ weights,
+                           double *__restrict__ out_pos,
+                           int32_t *__restrict__ out_elem,
+                           int8_t *__restrict__ out_status,
+                           double *__restrict__ flux,
+                           unsigned long long *__restrict__ lost, int64_t n,
+                           int max_steps) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
+    const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+    int32_t oe;
+    Vec3 op;
+    bool esc;
+    walk_segment(
+        planes, nbr, elem[i], o, d, weights[i], max_steps,
+        [&](int32_t e, double v) { atomicAdd(&flux[e], v); }, &oe, &op, &esc);
+    int8_t st = 0;
+    if (oe == kWalkLost) {
+      st = 3;
+      oe = elem[i];
+      atomicAdd(lost, 1ull);
+    } else if (esc) {
+      st = 1;
+    } else if (oe < -1) {
+      st = 2;
+    }
+    out_elem[i] = oe;
+    out_pos[i * 3] = op.x;
+    out_pos[i * 3 + 1] = op.y;
+    out_pos[i * 3 + 2] = op.z;
+    out_status[i] = st;
+  }
+}
+
 int grid_blocks(int64_t work) {
   int64_t blocks = (work + kBlock - 1) / kBlock;
   // 256 CUs x 8 blocks/CU: cap and block-stride the rest (guide G11).
@@ -267,6 +307,34 @@ public:
       PT_HIP_CHECK(hipGetLastError());
     }
     stats_.moves++;
+  }
+
+  void walk_raw(int64_t n, const double *pos, const double *dest,
+                const int32_t *elem, const double *weights, double *out_pos,
+                int32_t *out_elem, int8_t *out_status) override {
+    if (n == 0) return;
+    PT_HIP_CHECK(hipSetDevice(device_));
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    double *dp = dmalloc<double>(n * 3), *dd = dmalloc<double>(n * 3),
+           *dw = dmalloc<double>(n);
+    int32_t *de = dmalloc<int32_t>(n), *doe = dmalloc<int32_t>(n);
+    double *dop = dmalloc<double>(n * 3);
+    int8_t *dst_ = dmalloc<int8_t>(n);
+    PT_HIP_CHECK(hipMemcpy(dp, pos, n * 3 * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(dd, dest, n * 3 * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(dw, weights, n * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(de, elem, n * 4, hipMemcpyHostToDevice));
+    k_walk_raw<<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+        d_planes_, d_nbr_, dp, dd, de, dw, dop, doe, dst_, d_flux_, d_lost_, n,
+        steps);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+    PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(out_elem, doe, n * 4, hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(out_status, dst_, n, hipMemcpyDeviceToHost));
+    for (void *p : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
+                    (void *)doe, (void *)dop, (void *)dst_})
+      (void)hipFree(p);
   }
 
   std::vector<double> flux() const override {

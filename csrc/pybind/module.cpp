@@ -140,6 +140,44 @@ PYBIND11_MODULE(_core, m) {
       .def("write_osh",
            [](const Mesh &m_, const std::string &dir) { write_osh(dir, m_); });
 
+  py::class_<SubMesh>(m, "SubMesh")
+      .def_property_readonly("local", [](const SubMesh &s) -> const Mesh & { return s.local; },
+                             py::return_value_policy::reference_internal)
+      .def_property_readonly("elem_l2g",
+                             [](const SubMesh &s) {
+                               auto a = py::array_t<int64_t>(s.elem_l2g.size());
+                               std::memcpy(a.mutable_data(), s.elem_l2g.data(),
+                                           s.elem_l2g.size() * 8);
+                               return a;
+                             })
+      .def_property_readonly("foreign_gid",
+                             [](const SubMesh &s) {
+                               auto a = py::array_t<int64_t>(s.foreign_gid.size());
+                               std::memcpy(a.mutable_data(), s.foreign_gid.data(),
+                                           s.foreign_gid.size() * 8);
+                               return a;
+                             })
+      .def_property_readonly("foreign_owner", [](const SubMesh &s) {
+        auto a = py::array_t<int32_t>(s.foreign_owner.size());
+        std::memcpy(a.mutable_data(), s.foreign_owner.data(),
+                    s.foreign_owner.size() * 4);
+        return a;
+      });
+
+  m.def("partition_morton",
+        [](const Mesh &m_, int nparts) {
+          auto owners = partition_morton(m_, nparts);
+          auto a = py::array_t<int32_t>(owners.size());
+          std::memcpy(a.mutable_data(), owners.data(), owners.size() * 4);
+          return a;
+        });
+  m.def("extract_submesh",
+        [](const Mesh &m_, py::array_t<int32_t, py::array::c_style | py::array::forcecast> owners,
+           int part) {
+          std::vector<int32_t> o(owners.data(), owners.data() + owners.size());
+          return extract_submesh(m_, o, part);
+        });
+
   m.def("build_box", &build_box, py::arg("nx"), py::arg("ny"), py::arg("nz"),
         py::arg("lx") = 1.0, py::arg("ly") = 1.0, py::arg("lz") = 1.0);
   m.def("read_mesh", &read_mesh);
@@ -202,6 +240,26 @@ PYBIND11_MODULE(_core, m) {
                throw std::runtime_error("move_continue: array size mismatch");
              py::gil_scoped_release nogil;
              e.eng->move_continue(dest.data(), flying.data(), weights.data(), n);
+           })
+      .def("walk_raw",
+           [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> pos,
+              py::array_t<double, py::array::c_style | py::array::forcecast> dest,
+              py::array_t<int32_t, py::array::c_style | py::array::forcecast> elem,
+              py::array_t<double, py::array::c_style | py::array::forcecast> weights) {
+             const int64_t n = (int64_t)elem.size();
+             if ((int64_t)pos.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
+                 (int64_t)weights.size() != n)
+               throw std::runtime_error("walk_raw: size mismatch");
+             auto out_pos = py::array_t<double>({n, (int64_t)3});
+             auto out_elem = py::array_t<int32_t>(n);
+             auto out_status = py::array_t<int8_t>(n);
+             {
+               py::gil_scoped_release nogil;
+               e.eng->walk_raw(n, pos.data(), dest.data(), elem.data(),
+                               weights.data(), out_pos.mutable_data(),
+                               out_elem.mutable_data(), out_status.mutable_data());
+             }
+             return py::make_tuple(out_pos, out_elem, out_status);
            })
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })

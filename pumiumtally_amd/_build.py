@@ -24,6 +24,7 @@ SOURCES = [
     "csrc/core/mesh_io.cpp",
     "csrc/core/osh_io.cpp",
     "csrc/core/engine_cpu.cpp",
+    "csrc/core/partition.cpp",
     "csrc/hip/engine_gpu.hip",
     "csrc/api/PumiTally.cpp",
     "csrc/pybind/module.cpp",

@@ -1,0 +1,155 @@
+"""Domain-decomposed tally: 8-way element partition + particle handoff.
+
+For meshes that exceed what you want to replicate per GPU (the replicated
+DistributedTally in .dist is the default and faster whenever the mesh
+fits -- 288 GB HBM3E per MI355X fits multi-billion-tet walk data), this
+mode partitions elements across ranks (Morton-balanced), builds per-rank
+submeshes whose cut faces carry encoded foreign-element references, and
+walks segments locally; particles crossing a cut are shipped to the
+owning rank and resume mid-segment.  Shared-face planes are
+bitwise-identical between submeshes (csrc/core/partition.cpp), so a
+cross-rank walk tallies exactly what the single-mesh walk would.
+
+Replaces the reference's pumipic picparts + ParticleTracer migration
+(PumiTallyImpl.cpp:433-459,530-539) -- which is degenerate there (full
+mesh on every rank, owners all rank 0) -- with a real decomposition.
+
+Exchange collective: torch.distributed all_to_all_single over RCCL/xGMI
+when on GPU (nccl backend), all_gather_object on gloo (CPU tests).
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from .dist import init_distributed
+
+
+class PartitionedTally:
+    def __init__(self, mesh, device=None, backend=None, max_rounds: int = 64):
+        from .. import TallyEngine, _core, have_gpu
+
+        self.rank, self.world, self.local = init_distributed(backend)
+        self.mesh = mesh
+        self.max_rounds = max_rounds
+        self.owners = _core.partition_morton(mesh, self.world)
+        self.sub = _core.extract_submesh(mesh, self.owners, self.rank)
+        self.l2g = self.sub.elem_l2g
+        self.g2l = -np.ones(mesh.nelems, dtype=np.int64)
+        self.g2l[self.l2g] = np.arange(len(self.l2g))
+        self.foreign_gid = self.sub.foreign_gid
+        self.foreign_owner = self.sub.foreign_owner
+        if device is None:
+            device = f"cuda:{self.local}" if have_gpu() else "cpu"
+        # Engine over the local submesh; used only through walk_raw + flux.
+        self.engine = TallyEngine(self.sub.local, 1, device=device)
+
+    # -- helpers -----------------------------------------------------------
+    def _exchange(self, records_per_rank):
+        """records_per_rank: list of world np.float64 arrays (k,8); returns
+        concatenated records received from all ranks."""
+        if self.world == 1:
+            return records_per_rank[0] if records_per_rank else np.zeros((0, 8))
+        import torch
+        import torch.distributed as dist
+
+        if dist.get_backend() == "nccl":
+            send = torch.cat([torch.from_numpy(np.ascontiguousarray(r)).view(-1)
+                              for r in records_per_rank]).cuda(self.local)
+            in_counts = [r.size for r in records_per_rank]
+            counts = torch.tensor(in_counts, dtype=torch.int64).cuda(self.local)
+            all_counts = torch.zeros(self.world * self.world, dtype=torch.int64,
+                                     device=counts.device)
+            dist.all_gather_into_tensor(all_counts, counts)
+            all_counts = all_counts.view(self.world, self.world).cpu().numpy()
+            out_counts = list(all_counts[:, self.rank])
+            recv = torch.empty(int(sum(out_counts)), dtype=torch.float64,
+                               device=counts.device)
+            dist.all_to_all_single(recv, send, out_counts, in_counts)
+            return recv.cpu().numpy().reshape(-1, 8)
+        # gloo: object all_gather
+        gathered = [None] * self.world
+        dist.all_gather_object(gathered, [np.asarray(r) for r in records_per_rank])
+        mine = [g[self.rank] for g in gathered if g[self.rank].size]
+        return np.concatenate(mine).reshape(-1, 8) if mine else np.zeros((0, 8))
+
+    # -- public API --------------------------------------------------------
+    def run_segments(self, origins, dests, weights):
+        """Walk one batch of global segments (origins->dests, weights),
+        tallying into the partitioned flux.  Each rank passes the SAME
+        global arrays (or its own shard -- ownership is resolved here);
+        segments starting outside this rank's elements are ignored locally
+        and handled by their owner."""
+        origins = np.asarray(origins, np.float64).reshape(-1, 3)
+        dests = np.asarray(dests, np.float64).reshape(-1, 3)
+        weights = np.asarray(weights, np.float64).reshape(-1)
+        gids = self.mesh.locate(origins)
+        mine = (gids >= 0) & (self.owners[np.maximum(gids, 0)] == self.rank)
+        pos = origins[mine]
+        dst = dests[mine]
+        wgt = weights[mine]
+        elem = self.g2l[gids[mine]].astype(np.int32)
+
+        for _round in range(self.max_rounds):
+            outbound = [np.zeros((0, 8)) for _ in range(self.world)]
+            if len(elem):
+                out_pos, out_elem, status = self.engine._eng.walk_raw(
+                    pos.ravel(), dst.ravel(), elem, wgt)
+                hand = status == 2
+                if hand.any():
+                    k = -(out_elem[hand].astype(np.int64) + 2)
+                    tgt_gid = self.foreign_gid[k]
+                    tgt_owner = self.foreign_owner[k]
+                    rec = np.concatenate(
+                        [out_pos[hand], dst[hand],
+                         wgt[hand, None], tgt_gid[:, None].astype(np.float64)],
+                        axis=1)
+                    for r in range(self.world):
+                        sel = tgt_owner == r
+                        if sel.any():
+                            outbound[r] = rec[sel]
+            # global termination: exchange; empty everywhere -> done
+            inbound = self._exchange(outbound)
+            if self.world > 1:
+                import torch
+                import torch.distributed as dist
+                n_in = torch.tensor([float(inbound.size)])
+                if dist.get_backend() == "nccl":
+                    n_in = n_in.cuda(self.local)
+                dist.all_reduce(n_in, op=dist.ReduceOp.SUM)
+                if float(n_in.item()) == 0.0:
+                    break
+                total_pending = float(n_in.item())
+            else:
+                total_pending = inbound.size
+            if total_pending == 0:
+                break
+            pos = inbound[:, 0:3]
+            dst = inbound[:, 3:6]
+            wgt = inbound[:, 6]
+            elem = self.g2l[inbound[:, 7].astype(np.int64)].astype(np.int32)
+        else:
+            raise RuntimeError("partitioned walk did not converge "
+                               f"in {self.max_rounds} handoff rounds")
+
+    def flux_global(self) -> np.ndarray:
+        """Scatter the local tally to global element ids and sum over ranks."""
+        local = self.engine.flux()
+        out = np.zeros(self.mesh.nelems)
+        out[self.l2g] = local
+        if self.world > 1:
+            import torch
+            import torch.distributed as dist
+            t = torch.from_numpy(out)
+            if dist.get_backend() == "nccl":
+                t = t.cuda(self.local)
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            out = t.cpu().numpy()
+        return out
+
+    def write_tally_results(self, filename="fluxresult.vtk"):
+        from .. import write_tally_vtk
+
+        f = self.flux_global()
+        if self.rank == 0:
+            write_tally_vtk(filename, self.mesh, f)
+        return f

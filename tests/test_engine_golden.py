@@ -146,3 +146,21 @@ def test_flux_accumulates_across_moves():
     b = np.tile([0.25, 0.4, 0.5], (NP_, 1)).ravel()
     e.move(a, b, np.ones(NP_, np.int8), w)
     assert abs(e.flux().sum() - 0.15 * NP_) < 1e-12
+
+
+@pytest.mark.gpu
+def test_move_continue_gpu_matches_cpu():
+    m = pt.build_box(6, 6, 6)
+    n = 10000
+    rng = np.random.default_rng(21)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    flux = {}
+    for dev in ("cpu", "cuda"):
+        e = pt.TallyEngine(m, n, device=dev)
+        e.copy_initial_position(o.ravel())
+        e.move_continue(d.ravel(), np.ones(n, np.int8), w)
+        e.synchronize()
+        flux[dev] = e.flux()
+    assert np.abs(flux["cpu"] - flux["cuda"]).max() < 1e-10

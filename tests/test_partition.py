@@ -1,0 +1,152 @@
+"""Domain-decomposition tests: submesh extraction and cross-part walks."""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+import pumiumtally_amd as pt
+from pumiumtally_amd import _core
+
+
+def test_partition_balanced():
+    m = pt.build_box(4, 4, 4)
+    owners = _core.partition_morton(m, 8)
+    counts = np.bincount(owners, minlength=8)
+    assert counts.sum() == m.nelems
+    assert counts.max() - counts.min() <= 1
+
+
+def test_submesh_structure():
+    m = pt.build_box(3, 3, 3)
+    owners = _core.partition_morton(m, 4)
+    total = 0
+    for p in range(4):
+        sub = _core.extract_submesh(m, owners, p)
+        total += sub.local.nelems
+        assert np.array_equal(owners[sub.elem_l2g], np.full(sub.local.nelems, p))
+        # local volumes match global volumes elementwise
+        assert np.allclose(sub.local.volumes, m.volumes[sub.elem_l2g])
+        # every foreign ref points to an element owned by another part
+        assert (owners[sub.foreign_gid] != p).all()
+        assert np.array_equal(owners[sub.foreign_gid], sub.foreign_owner)
+    assert total == m.nelems
+
+
+def test_single_rank_partitioned_equals_plain():
+    """world=1 PartitionedTally (no exchange) == plain engine flux."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(3, 3, 3)
+    n = 100
+    rng = np.random.default_rng(5)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = rng.uniform(0.05, 0.95, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+
+    ptal = PartitionedTally(m, device="cpu")
+    ptal.run_segments(o, d, w)
+    got = ptal.flux_global()
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(got, ref.flux(), atol=1e-12)
+
+
+def test_walk_raw_handoff_encoding():
+    """A walk on a submesh stops exactly at the partition cut."""
+    m = pt.build_box(2, 1, 1, 2.0, 1.0, 1.0)
+    # put x<1 cells in part 0, x>1 in part 1 (morton split does this for 2)
+    owners = _core.partition_morton(m, 2)
+    sub0 = _core.extract_submesh(m, owners, int(owners[m.locate(np.array([[0.2, 0.4, 0.5]]))[0]]))
+    eng = pt.TallyEngine(sub0.local, 1, device="cpu")
+    # segment crossing the whole domain
+    start = np.array([[0.1, 0.4, 0.5]])
+    lid = int(np.where(sub0.elem_l2g == m.locate(start)[0])[0][0])
+    out_pos, out_elem, status = eng._eng.walk_raw(
+        start.ravel(), np.array([1.9, 0.4, 0.5]), np.array([lid], np.int32),
+        np.ones(1))
+    assert status[0] == 2
+    k = -(int(out_elem[0]) + 2)
+    assert 0 <= k < len(sub0.foreign_gid)
+    # tally got only the inside part, and the handoff point is on the cut
+    assert abs(eng.flux().sum() - (out_pos[0, 0] - 0.1)) < 1e-12
+
+
+WORKER = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+from pumiumtally_amd.parallel.partition import PartitionedTally
+
+rank = int(os.environ["RANK"])
+mesh = pt.build_box(4, 4, 4)
+n = 300
+rng = np.random.default_rng(77)  # same segments on both ranks
+o = rng.uniform(0.05, 0.95, size=(n, 3))
+d = rng.uniform(0.05, 0.95, size=(n, 3))
+w = rng.uniform(0.1, 1.0, n)
+
+ptal = PartitionedTally(mesh, device="cpu", backend="gloo")
+ptal.run_segments(o, d, w)
+flux = ptal.flux_global()
+
+if rank == 0:
+    ref = pt.TallyEngine(mesh, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    err = np.abs(flux - ref.flux()).max()
+    assert err < 1e-10, err
+    print("PART_OK")
+import torch.distributed as dist
+dist.destroy_process_group()
+"""
+
+
+def test_gloo_world2_partitioned(tmp_path):
+    pytest.importorskip("torch")
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER)
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": "29919",
+        "WORLD_SIZE": "2",
+        "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "PART_OK" in outs[0]
+
+
+@pytest.mark.gpu
+def test_single_rank_partitioned_gpu():
+    """walk_raw on GPU submesh == CPU, including handoff stops."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(6, 6, 6)
+    n = 5000
+    rng = np.random.default_rng(9)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+
+    ptal = PartitionedTally(m, device="cuda:0")
+    assert ptal.engine.is_gpu
+    ptal.run_segments(o, d, w)
+    got = ptal.flux_global()
+    assert np.abs(got - ref.flux()).max() < 1e-10
