@@ -37,6 +37,9 @@ def main():
                     help="use move_continue (no origin upload); NOT the headline "
                          "reference-API config -- reported with a distinct metric name")
     ap.add_argument("--no-sort", action="store_true", help="disable Morton ordering of particles")
+    ap.add_argument("--device-resident", action="store_true",
+                    help="inputs pre-staged in device memory (GPU transport-code "
+                         "integration path); distinct metric name")
     args = ap.parse_args()
 
     import numpy as np
@@ -71,9 +74,20 @@ def main():
 
     ends = (p0.reshape(-1), p1.reshape(-1))
 
+    if args.device_resident:
+        import torch
+        dev = torch.device(f"cuda:{local}")
+        t_ends = (torch.from_numpy(np.asarray(ends[0])).to(dev),
+                  torch.from_numpy(np.asarray(ends[1])).to(dev))
+        t_flying = torch.from_numpy(np.asarray(flying)).to(dev)
+        t_weights = torch.from_numpy(np.asarray(weights)).to(dev)
+
     def step(k):
         # ping-pong: walk P0->P1, then P1->P0; origin == current position so
         # phase A is a no-op compare, phase B walks the full segment set.
+        if args.device_resident:
+            eng.move_from_device(t_ends[(k + 1) % 2], t_flying, t_weights)
+            return
         o, d = ends[k % 2], ends[(k + 1) % 2]
         if args.continue_mode:
             eng.move_continue(d, flying, weights)
@@ -131,7 +145,8 @@ def main():
     if rank == 0:
         result = {
             "metric": "particle-steps/sec"
-                      + ("-continue-mode" if args.continue_mode else ""),
+                      + ("-continue-mode" if args.continue_mode else "")
+                      + ("-device-resident" if args.device_resident else ""),
             "value": value,
             "unit": "particle-steps/s",
             "n_gpus": world,

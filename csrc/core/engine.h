@@ -67,6 +67,16 @@ public:
     move(nullptr, dest, flying, weights, n);
   }
 
+  // Device-resident move: all arrays already live in this engine's device
+  // memory (for GPU-side transport codes; no host staging at all).
+  // origin may be nullptr (continue semantics).  Throws on the CPU engine.
+  virtual void move_device(const double *d_origin, const double *d_dest,
+                           const int8_t *d_flying, const double *d_weights,
+                           int64_t n) {
+    (void)d_origin; (void)d_dest; (void)d_flying; (void)d_weights; (void)n;
+    throw std::runtime_error("move_device requires the GPU engine");
+  }
+
   // Raw batched walk for the partitioned driver: walks n independent
   // segments (pos->dest) starting in the given elements, tallying into
   // this engine's flux.  status: 0=reached dest, 1=escaped (vacuum),

@@ -309,6 +309,19 @@ public:
     stats_.moves++;
   }
 
+  void move_device(const double *d_origin, const double *d_dest,
+                   const int8_t *d_flying, const double *d_weights,
+                   int64_t n) override {
+    check_n(n);
+    PT_HIP_CHECK(hipSetDevice(device_));
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    k_move<<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+        d_planes_, d_nbr_, grid_view_, d_origin, d_dest, d_flying, d_weights,
+        d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, 0, n, loc_tol_, steps);
+    PT_HIP_CHECK(hipGetLastError());
+    stats_.moves++;
+  }
+
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status) override {

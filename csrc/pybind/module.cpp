@@ -12,6 +12,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdint>
 #include <cstring>
 #include <memory>
 
@@ -240,6 +241,19 @@ PYBIND11_MODULE(_core, m) {
                throw std::runtime_error("move_continue: array size mismatch");
              py::gil_scoped_release nogil;
              e.eng->move_continue(dest.data(), flying.data(), weights.data(), n);
+           })
+      .def("move_device",
+           // Raw device-pointer entry (integers as returned by
+           // torch.Tensor.data_ptr()).  origin_ptr=0 means continue
+           // semantics.  The Python wrapper validates device/dtype/shape.
+           [](PyEngine &e, uintptr_t origin_ptr, uintptr_t dest_ptr,
+              uintptr_t flying_ptr, uintptr_t weights_ptr) {
+             py::gil_scoped_release nogil;
+             e.eng->move_device((const double *)origin_ptr,
+                                (const double *)dest_ptr,
+                                (const int8_t *)flying_ptr,
+                                (const double *)weights_ptr,
+                                e.eng->num_particles());
            })
       .def("walk_raw",
            [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> pos,

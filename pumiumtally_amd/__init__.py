@@ -86,6 +86,40 @@ class TallyEngine:
         was resampled this step (origin == committed position)."""
         self._eng.move_continue(dest, flying, weights)
 
+    def move_from_device(self, dest, flying, weights, origin=None):
+        """Device-resident move: dest/flying/weights (and optionally origin)
+        are GPU tensors (torch CUDA tensors or anything with data_ptr()
+        semantics via __cuda_array_interface__) already on this engine's
+        device -- no host staging.  For GPU-side transport codes."""
+        def ptr(t, dtype, numel):
+            if t is None:
+                return 0
+            iface = getattr(t, "__cuda_array_interface__", None)
+            if iface is None:
+                raise TypeError("move_from_device expects CUDA tensors/arrays")
+            import math
+            n = math.prod(iface["shape"]) if iface["shape"] else 1
+            if iface["typestr"] != dtype or n != numel:
+                raise TypeError(
+                    f"expected {dtype} x{numel}, got {iface['typestr']} x{n}")
+            strides = iface.get("strides")
+            if strides is not None:
+                # C-contiguity required (None == C-contiguous per protocol)
+                itemsize = int(dtype[-1])
+                expect = []
+                acc = itemsize
+                for s in reversed(iface["shape"]):
+                    expect.append(acc)
+                    acc *= s
+                if list(strides) != list(reversed(expect)):
+                    raise TypeError("move_from_device requires contiguous tensors")
+            return iface["data"][0]
+
+        n = self.num_particles
+        self._eng.move_device(
+            ptr(origin, "<f8", n * 3), ptr(dest, "<f8", n * 3),
+            ptr(flying, "|i1", n), ptr(weights, "<f8", n))
+
     def synchronize(self):
         self._eng.synchronize()
 

@@ -164,3 +164,33 @@ def test_move_continue_gpu_matches_cpu():
         e.synchronize()
         flux[dev] = e.flux()
     assert np.abs(flux["cpu"] - flux["cuda"]).max() < 1e-10
+
+
+@pytest.mark.gpu
+def test_move_from_device_matches_host_path():
+    torch = pytest.importorskip("torch")
+    if not torch.cuda.is_available():
+        pytest.skip("no torch cuda")
+    m = pt.build_box(6, 6, 6)
+    n = 10000
+    rng = np.random.default_rng(33)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+
+    host = pt.TallyEngine(m, n, device="cuda")
+    host.copy_initial_position(o.ravel())
+    host.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    host.synchronize()
+
+    devE = pt.TallyEngine(m, n, device="cuda")
+    devE.copy_initial_position(o.ravel())
+    dev = torch.device("cuda:0")
+    t_d = torch.from_numpy(d.ravel()).to(dev)
+    t_f = torch.ones(n, dtype=torch.int8, device=dev)
+    t_w = torch.from_numpy(w).to(dev)
+    devE.move_from_device(t_d, t_f, t_w)
+    devE.synchronize()
+
+    assert np.array_equal(host.elem_ids(), devE.elem_ids())
+    assert np.abs(host.flux() - devE.flux()).max() < 1e-10
