@@ -16,6 +16,19 @@ from __future__ import annotations
 
 import os as _os
 
+# ROCm runtime ordering: torch wheels bundle their own libamdhip64.so.7
+# (same SONAME as /opt/rocm's).  Whichever loads FIRST owns the process;
+# loading ours first and torch later leaves torch's HIP dead ("No HIP GPUs
+# are available") and poisons subsequent HIP calls.  Importing torch first
+# is safe in both orders of use, so do it here, before _core loads, unless
+# explicitly disabled.  (Verified on MI355X: torch-first shares one runtime
+# and both torch.cuda and our engine work.)
+if _os.environ.get("PUMITALLY_NO_TORCH", "0") != "1":
+    try:
+        import torch as _torch  # noqa: F401
+    except ImportError:
+        pass
+
 
 def _load_core():
     try:
