@@ -30,6 +30,111 @@ constexpr double kWalkTEps = 1e-12;   // tolerance on the segment parameter t
 // foreign-element refs -(2+k) used by partitioned submeshes.
 constexpr int32_t kWalkLost = INT32_MIN;
 
+// Walk state for the incremental (one-crossing-at-a-time) API.  The GPU
+// move kernel uses this so a lane whose walk finished can immediately
+// acquire its next particle instead of idling behind the wave's slowest
+// lane (variable walk lengths gave 46% VALUUtilization with the
+// whole-segment-per-lane formulation).
+struct WalkState {
+  Vec3 o, d;
+  double seg_len, t_cur, weight;
+  int32_t elem, prev_elem;
+  int step;
+  bool tally;
+};
+
+PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
+  s.o = o;
+  s.d = d;
+  s.seg_len = norm(d - o);
+  s.t_cur = 0.0;
+  s.weight = w;
+  s.elem = elem;
+  s.prev_elem = -1;
+  s.step = 0;
+  s.tally = (w != 0.0) && (s.seg_len > 0.0);
+}
+
+// Advance one element crossing.  Returns true when the walk finished and
+// the out_* values are valid.  FluxAdd: void(int32_t elem, double v).
+template <class FluxAdd>
+PT_HD bool walk_advance(const Plane *__restrict__ planes,
+                        const int32_t *__restrict__ nbr, WalkState &s,
+                        int max_steps, FluxAdd &&add, int32_t *out_elem,
+                        Vec3 *out_pos, bool *out_escaped) {
+  if (s.step++ >= max_steps) {
+    // Did not converge (numerically stuck / absurd chord): drop here and
+    // flag as lost (reference prints "Not all particles are found",
+    // PumiTallyImpl.cpp:455-458).
+    *out_elem = kWalkLost;
+    *out_pos = s.o + s.t_cur * (s.d - s.o);
+    *out_escaped = false;
+    return true;
+  }
+  // Evaluate the 4 face planes at both segment endpoints.
+  const Plane *pl = planes + (int64_t)s.elem * 4;
+  double t_exit = 2.0;
+  int exit_face = -1;
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
+  for (int f = 0; f < 4; ++f) {
+    const double vd = plane_eval(pl[f], s.d);
+    if (vd < 0.0) {
+      const double vo = plane_eval(pl[f], s.o);
+      const double denom = vo - vd; // > 0 since vd < 0 <= ~vo
+      if (denom > 0.0) {
+        double tf = vo / denom;
+        // Never step backwards (grazing entry can give tf slightly below
+        // t_cur); never consider the face we just came through.
+        if (nbr[(int64_t)s.elem * 4 + f] != s.prev_elem || s.prev_elem == -1) {
+          if (tf >= s.t_cur - kWalkTEps && tf < t_exit) {
+            t_exit = tf;
+            exit_face = f;
+          }
+        }
+      }
+    }
+  }
+
+  if (exit_face < 0 || t_exit >= 1.0) {
+    // Destination lies inside this element: tally the final partial
+    // segment and stop (reference: reached_destination, last_exit==-1).
+    if (s.tally) add(s.elem, (1.0 - s.t_cur) * s.seg_len * s.weight);
+    *out_elem = s.elem;
+    *out_pos = s.d;
+    *out_escaped = false;
+    return true;
+  }
+
+  const double t_clamped = t_exit > s.t_cur ? t_exit : s.t_cur;
+  if (s.tally) add(s.elem, (t_clamped - s.t_cur) * s.seg_len * s.weight);
+
+  const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
+  if (next == -1) {
+    // Vacuum boundary: clip the destination to the exit point; the
+    // particle keeps its last element id (reference K6 semantics,
+    // PumiTallyImpl.cpp:275-281 and the 1.0-not-1.1 test expectation).
+    *out_elem = s.elem;
+    *out_pos = s.o + t_clamped * (s.d - s.o);
+    *out_escaped = true;
+    return true;
+  }
+  if (next < -1) {
+    // Partitioned submesh: the face crosses into an element owned by
+    // another rank.  Stop at the crossing; the caller decodes the
+    // foreign reference (k = -(next+2)) and ships the particle.
+    *out_elem = next;
+    *out_pos = s.o + t_clamped * (s.d - s.o);
+    *out_escaped = false;
+    return true;
+  }
+  s.prev_elem = s.elem;
+  s.elem = next;
+  s.t_cur = t_clamped;
+  return false;
+}
+
 // FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
 // performs atomicAdd into the flux array; on the serial CPU path a plain +=.
 template <class FluxAdd>
@@ -37,82 +142,11 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
                         Vec3 d, double weight, int max_steps, FluxAdd &&add,
                         int32_t *out_elem, Vec3 *out_pos, bool *out_escaped) {
-  const double seg_len = norm(d - o);
-  double t_cur = 0.0;
-  int32_t prev_elem = -1;
-  const bool tally = (weight != 0.0) && (seg_len > 0.0);
-
-  for (int step = 0; step < max_steps; ++step) {
-    // Evaluate the 4 face planes at both segment endpoints.
-    const Plane *pl = planes + (int64_t)elem * 4;
-    double t_exit = 2.0;
-    int exit_face = -1;
-#if defined(__HIP_DEVICE_COMPILE__)
-#pragma unroll
-#endif
-    for (int f = 0; f < 4; ++f) {
-      const double vd = plane_eval(pl[f], d);
-      if (vd < 0.0) {
-        const double vo = plane_eval(pl[f], o);
-        const double denom = vo - vd; // > 0 since vd < 0 <= ~vo
-        if (denom > 0.0) {
-          double tf = vo / denom;
-          // Never step backwards (grazing entry can give tf slightly below
-          // t_cur); never consider the face we just came through.
-          if (nbr[(int64_t)elem * 4 + f] != prev_elem || prev_elem == -1) {
-            if (tf >= t_cur - kWalkTEps && tf < t_exit) {
-              t_exit = tf;
-              exit_face = f;
-            }
-          }
-        }
-      }
-    }
-
-    if (exit_face < 0 || t_exit >= 1.0) {
-      // Destination lies inside this element: tally the final partial
-      // segment and stop (reference: reached_destination, last_exit==-1).
-      if (tally) add(elem, (1.0 - t_cur) * seg_len * weight);
-      *out_elem = elem;
-      *out_pos = d;
-      *out_escaped = false;
-      return;
-    }
-
-    const double t_clamped = t_exit > t_cur ? t_exit : t_cur;
-    if (tally) add(elem, (t_clamped - t_cur) * seg_len * weight);
-
-    const int32_t next = nbr[(int64_t)elem * 4 + exit_face];
-    if (next == -1) {
-      // Vacuum boundary: clip the destination to the exit point; the
-      // particle keeps its last element id (reference K6 semantics,
-      // PumiTallyImpl.cpp:275-281 and the 1.0-not-1.1 test expectation).
-      *out_elem = elem;
-      *out_pos = o + t_clamped * (d - o);
-      *out_escaped = true;
-      return;
-    }
-    if (next < -1) {
-      // Partitioned submesh: the face crosses into an element owned by
-      // another rank.  Stop at the crossing; the caller decodes the
-      // foreign reference (k = -(next+2)) and ships the particle.
-      *out_elem = next;
-      *out_pos = o + t_clamped * (d - o);
-      *out_escaped = false;
-      return;
-    }
-    prev_elem = elem;
-    elem = next;
-    t_cur = t_clamped;
+  WalkState s;
+  walk_init(s, elem, o, d, weight);
+  while (!walk_advance(planes, nbr, s, max_steps, add, out_elem, out_pos,
+                       out_escaped)) {
   }
-
-  // Walk did not converge (numerically stuck / absurdly long chord): drop
-  // the particle at its current position, flag as lost.  The engine counts
-  // these (reference prints "Not all particles are found",
-  // PumiTallyImpl.cpp:455-458).
-  *out_elem = kWalkLost;
-  *out_pos = o + t_cur * (d - o);
-  *out_escaped = false;
 }
 
 // Point-in-tet test against the 4 planes (signed distances, unit normals).
