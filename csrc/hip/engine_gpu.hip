@@ -111,67 +111,47 @@ __global__ void k_move(const Plane *__restrict__ planes,
   const int64_t base = lo + (int64_t)vb * per_blk;
   const int64_t end = base + per_blk < hi ? base + per_blk : hi;
 
-  // Per-lane work-refill state machine: walk lengths vary wildly, so a
-  // lane whose walk finished acquires its next particle IMMEDIATELY
-  // instead of idling (exec-masked) until the wave's slowest walk ends.
-  // Measured: the naive one-particle-per-lane loop ran at 46%
-  // VALUUtilization from exactly this tail divergence.
-  auto flux_add = [&](int32_t el, double v) { atomicAdd(&flux[el], v); };
-  WalkState st;
-  int32_t init_elem = -1; // walk's starting element (restored on lost)
-  int64_t cur = -1;       // particle index the lane is walking, -1 = idle
-  int64_t i = base + threadIdx.x - blockDim.x;
-  for (;;) {
-    if (cur < 0) {
-      // acquire: scan forward for a particle that needs walking, doing
-      // the cheap housekeeping (non-flying skip, phase A, outside-mesh)
-      // inline.
-      for (;;) {
-        i += blockDim.x;
-        if (i >= end) break;
-        if (!flying[i]) continue;
-        Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
-        int32_t e = elem[i];
-        if (origin != nullptr && !escaped[i]) {
-          const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
-          if (q.x != o.x || q.y != o.y || q.z != o.z) {
-            e = grid_locate(grid, planes, q, loc_tol);
-            o = q;
-          }
-        }
-        if (e < 0) {
-          pos[i * 3] = o.x;
-          pos[i * 3 + 1] = o.y;
-          pos[i * 3 + 2] = o.z;
-          elem[i] = e;
-          continue;
-        }
-        const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
-        walk_init(st, e, o, d, weights[i]);
-        init_elem = e;
-        cur = i;
-        break;
+  // One particle per lane, block-stride.  (A per-lane work-refill state
+  // machine was tried to attack tail divergence and measured STRICTLY
+  // worse -- 1297 vs 1600 M ps/s device-resident, 4x worse at chord 32:
+  // the kernel is latency-bound, not utilization-bound, and the extra
+  // per-crossing control flow + 7->5 waves/SIMD occupancy hurt more than
+  // the refill helped.  See profiles/README.md.)
+  for (int64_t i = base + threadIdx.x; i < end; i += blockDim.x) {
+    if (!flying[i]) continue;
+    Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
+    int32_t e = elem[i];
+    if (origin != nullptr && !escaped[i]) {
+      const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
+      if (q.x != o.x || q.y != o.y || q.z != o.z) {
+        e = grid_locate(grid, planes, q, loc_tol);
+        o = q;
       }
     }
-    if (!__any(cur >= 0)) break;
-    if (cur >= 0) {
-      int32_t out_elem;
-      Vec3 out_pos;
-      bool out_esc;
-      if (walk_advance(planes, nbr, st, max_steps, flux_add, &out_elem,
-                       &out_pos, &out_esc)) {
-        if (out_elem == kWalkLost) {
-          atomicAdd(lost, 1ull);
-          out_elem = init_elem;
-        }
-        elem[cur] = out_elem;
-        pos[cur * 3] = out_pos.x;
-        pos[cur * 3 + 1] = out_pos.y;
-        pos[cur * 3 + 2] = out_pos.z;
-        escaped[cur] = out_esc ? 1 : 0;
-        cur = -1;
-      }
+    if (e < 0) {
+      pos[i * 3] = o.x;
+      pos[i * 3 + 1] = o.y;
+      pos[i * 3 + 2] = o.z;
+      elem[i] = e;
+      continue;
     }
+    const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+    int32_t out_elem;
+    Vec3 out_pos;
+    bool out_esc;
+    walk_segment(
+        planes, nbr, e, o, d, weights[i], max_steps,
+        [&](int32_t el, double v) { atomicAdd(&flux[el], v); }, &out_elem,
+        &out_pos, &out_esc);
+    if (out_elem == kWalkLost) {
+      atomicAdd(lost, 1ull);
+      out_elem = e;
+    }
+    elem[i] = out_elem;
+    pos[i * 3] = out_pos.x;
+    pos[i * 3 + 1] = out_pos.y;
+    pos[i * 3 + 2] = out_pos.z;
+    escaped[i] = out_esc ? 1 : 0;
   }
 }
 
