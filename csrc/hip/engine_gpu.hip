@@ -311,6 +311,11 @@ public:
           lo, hi, loc_tol_, steps);
       PT_HIP_CHECK(hipGetLastError());
     }
+    // The caller may mutate or free its buffers as soon as move() returns
+    // (the facade zeroes the flying array; transients die).  Block until
+    // every H2D copy has consumed them; the walk keeps running async on
+    // the compute stream.
+    PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
     stats_.moves++;
   }
 
@@ -320,10 +325,19 @@ public:
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    k_move<<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-        d_planes_, d_nbr_, grid_view_, d_origin, d_dest, d_flying, d_weights,
-        d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, 0, n, loc_tol_, steps);
-    PT_HIP_CHECK(hipGetLastError());
+    // Chunked launches even with zero staging: a ~1M-particle launch keeps
+    // each XCD's Morton-contiguous particle range's mesh working set inside
+    // its private 4 MiB L2.  One 10M-particle launch spreads ~18 MB per XCD
+    // and thrashes L2: measured 4x slower at mean chord 32.
+    const int64_t chunk = std::max<int64_t>(1 << 20, (n + 7) / 8);
+    for (int64_t lo = 0; lo < n; lo += chunk) {
+      const int64_t hi = std::min(n, lo + chunk);
+      k_move<<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+          d_planes_, d_nbr_, grid_view_, d_origin, d_dest, d_flying, d_weights,
+          d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
+          steps);
+      PT_HIP_CHECK(hipGetLastError());
+    }
     stats_.moves++;
   }
 

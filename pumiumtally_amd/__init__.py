@@ -129,6 +129,14 @@ class TallyEngine:
             return iface["data"][0]
 
         n = self.num_particles
+        import sys
+        if "torch" in sys.modules:
+            # order against torch's stream: tensors produced by torch ops
+            # (.to(device), fills) must be materialized before our kernel,
+            # which runs on the engine's own HIP stream.
+            torch = sys.modules["torch"]
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
         self._eng.move_device(
             ptr(origin, "<f8", n * 3), ptr(dest, "<f8", n * 3),
             ptr(flying, "|i1", n), ptr(weights, "<f8", n))
