@@ -16,11 +16,12 @@
 //     144 B/tet, contiguous, no indirection to vertex coords in the hot
 //     loop.  A 1M-tet mesh (~150 MB) is fully resident in the 256 MiB
 //     Infinity Cache.
-//   * Host arrays from the caller are hipHostRegister'ed once (cached by
-//     pointer -- physics codes pass the same buffers every step) and then
-//     copied H2D in chunks on a copy stream that overlaps the walk kernel
-//     of the previous chunk on the compute stream, replacing the
-//     reference's fully synchronous deep_copy + fence staging.
+//   * Host arrays are copied H2D in chunks on a copy stream that overlaps
+//     the walk kernel of the previous chunk on the compute stream,
+//     replacing the reference's fully synchronous deep_copy + fence
+//     staging.  Pinned sources (pumiumtally_amd.pinned_array or
+//     app-registered buffers) run at full link rate; pageable sources take
+//     the runtime's internal staging path.
 #include "../core/engine.h"
 #include "../core/walk.h"
 
@@ -30,7 +31,6 @@
 #include <array>
 #include <cstdio>
 #include <cstring>
-#include <map>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -260,7 +260,6 @@ public:
 
   ~GpuEngine() override {
     (void)hipSetDevice(device_);
-    for (auto &r : registered_) (void)hipHostUnregister((void *)r.first);
     for (void *p : {(void *)d_planes_, (void *)d_nbr_, (void *)d_cell_start_,
                     (void *)d_cell_tets_, (void *)d_pos_, (void *)d_elem_,
                     (void *)d_escaped_, (void *)d_flux_, (void *)d_lost_,
@@ -425,31 +424,17 @@ private:
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
   }
 
-  // Async H2D from caller memory.  Registers (page-locks) the caller's
-  // buffer once per (pointer,length) so repeated per-step calls copy at
-  // full PCIe/xGMI rate with no host-side staging memcpy.
+  // Async H2D from caller memory.  No implicit hipHostRegister: a
+  // registration cache outliving freed caller buffers (numpy temporaries)
+  // poisons the runtime's pinning table for unrelated later copies
+  // ("pointer does not correspond to a registered memory region").
+  // Already-pinned sources (pumiumtally_amd.pinned_array, or app buffers
+  // the host code hipHostRegister'ed itself) copy at full link rate;
+  // pageable sources take the runtime's internal staging path - always
+  // correct, just slower.  The bench and any serious host integration use
+  // pinned buffers.
   void stage(const void *src, size_t bytes, void *dst, hipStream_t s) {
-    ensure_registered(src, bytes);
     PT_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, s));
-  }
-
-  void ensure_registered(const void *p, size_t bytes) {
-    auto it = registered_.lower_bound(p);
-    if (it != registered_.end() && it->first == p && it->second >= bytes) return;
-    // containment in a previously registered larger region
-    if (it != registered_.begin()) {
-      auto prev = std::prev(it);
-      if ((const char *)prev->first + prev->second >= (const char *)p + bytes)
-        return;
-    }
-    hipError_t e = hipHostRegister((void *)p, bytes, hipHostRegisterDefault);
-    if (e == hipSuccess) {
-      registered_[p] = bytes;
-    } else if (e != hipErrorHostMemoryAlreadyRegistered) {
-      (void)hipGetLastError(); // clear; fall back to pageable copy path
-    } else {
-      (void)hipGetLastError();
-    }
   }
 
   Mesh mesh_;
@@ -474,7 +459,6 @@ private:
   double *d_origin_ = nullptr, *d_dest_ = nullptr, *d_weights_ = nullptr;
   int8_t *d_flying_ = nullptr;
 
-  std::map<const void *, size_t> registered_;
   mutable EngineStats stats_;
 };
 
