@@ -99,7 +99,8 @@ class TallyEngine:
         was resampled this step (origin == committed position)."""
         self._eng.move_continue(dest, flying, weights)
 
-    def move_from_device(self, dest, flying, weights, origin=None):
+    def move_from_device(self, dest, flying, weights, origin=None,
+                         sync_torch=True):
         """Device-resident move: dest/flying/weights (and optionally origin)
         are GPU tensors (torch CUDA tensors or anything with data_ptr()
         semantics via __cuda_array_interface__) already on this engine's
@@ -130,7 +131,7 @@ class TallyEngine:
 
         n = self.num_particles
         import sys
-        if "torch" in sys.modules:
+        if sync_torch and "torch" in sys.modules:
             # order against torch's stream: tensors produced by torch ops
             # (.to(device), fills) must be materialized before our kernel,
             # which runs on the engine's own HIP stream.
