@@ -217,6 +217,7 @@ public:
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_comp_, hipStreamNonBlocking));
     for (auto &ev : events_) PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    for (auto &ev : comp_done_) PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
 
     // Mesh upload (once).
     d_planes_ = dmalloc<Plane>(mesh_.nelems * 4);
@@ -267,6 +268,7 @@ public:
                     (void *)d_weights_})
       (void)hipFree(p);
     for (auto &ev : events_) (void)hipEventDestroy(ev);
+    for (auto &ev : comp_done_) (void)hipEventDestroy(ev);
     (void)hipStreamDestroy(s_copy_);
     (void)hipStreamDestroy(s_comp_);
   }
@@ -277,6 +279,7 @@ public:
   void copy_initial_position(const double *p, int64_t n) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
+    sync(); // d_origin_ may still be read by in-flight move kernels
     stage(p, n * 3 * sizeof(double), d_origin_, s_copy_);
     PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
     k_locate<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
@@ -296,6 +299,13 @@ public:
     for (int k = 0; k < nchunks; ++k) {
       const int64_t lo = k * chunk, hi = std::min(n, lo + chunk);
       const int64_t m = hi - lo;
+      // WAR ordering across move() calls: this chunk's staging reuses the
+      // same device buffer region the PREVIOUS move's chunk-k kernel reads;
+      // the copy must wait for that kernel.  (Without this, a step whose
+      // walk outlives the next step's copies tallies corrupted segments --
+      // caught by flux-conservation checks at mean chord 32.)
+      hipEvent_t done = comp_done_[k % comp_done_.size()];
+      PT_HIP_CHECK(hipStreamWaitEvent(s_copy_, done, 0));
       if (origin)
         stage(origin + lo * 3, m * 3 * sizeof(double), d_origin_ + lo * 3, s_copy_);
       stage(dest + lo * 3, m * 3 * sizeof(double), d_dest_ + lo * 3, s_copy_);
@@ -309,6 +319,7 @@ public:
           d_flying_, d_weights_, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
           lo, hi, loc_tol_, steps);
       PT_HIP_CHECK(hipGetLastError());
+      PT_HIP_CHECK(hipEventRecord(done, s_comp_));
     }
     // The caller may mutate or free its buffers as soon as move() returns
     // (the facade zeroes the flying array; transients die).  Block until
@@ -444,6 +455,7 @@ private:
 
   hipStream_t s_copy_{}, s_comp_{};
   std::array<hipEvent_t, 8> events_{};
+  std::array<hipEvent_t, 8> comp_done_{};
 
   Plane *d_planes_ = nullptr;
   int32_t *d_nbr_ = nullptr;

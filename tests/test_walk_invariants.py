@@ -124,3 +124,29 @@ def test_gpu_matches_cpu_oracle():
     assert np.allclose(pc, pg, atol=0, rtol=0)  # walks are bitwise identical
     scale = np.abs(fc).max()
     assert np.abs(fc - fg).max() < 1e-10 * max(scale, 1.0)
+
+
+@pytest.mark.gpu
+def test_gpu_multistep_conservation_long_chords():
+    """Back-to-back moves with no intermediate sync at long mean chord:
+    regression test for the staging-vs-kernel WAR race across steps."""
+    import pumiumtally_amd as ptm
+    from pumiumtally_amd.mesh import box_mesh_with_tets
+    from pumiumtally_amd.utils import make_box_histories
+
+    mesh, cells = box_mesh_with_tets(200_000)
+    n = 2_000_000
+    steps = 6
+    e = ptm.TallyEngine(mesh, n, device="cuda")
+    p0, p1, fl, w = make_box_histories((1, 1, 1), n, 32.0, cells, pinned=True)
+    e.copy_initial_position(p0.reshape(-1))
+    seg = np.linalg.norm(np.asarray(p1) - np.asarray(p0), axis=1)
+    expected_per_step = float((seg * np.asarray(w)).sum())
+    ends = (p0.reshape(-1), p1.reshape(-1))
+    for k in range(steps):
+        e.move(ends[k % 2], ends[(k + 1) % 2], fl, w)
+    e.synchronize()
+    total = e.flux().sum()
+    assert e.stats()["lost_particles"] == 0
+    rel = abs(total - steps * expected_per_step) / (steps * expected_per_step)
+    assert rel < 1e-12, rel
