@@ -41,7 +41,6 @@ struct WalkState {
   int32_t elem, prev_elem;
   int step;
   bool tally;
-  double spec; // speculative-touch accumulator (device walk, never observable)
 };
 
 PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
@@ -54,7 +53,6 @@ PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
   s.prev_elem = -1;
   s.step = 0;
   s.tally = (w != 0.0) && (s.seg_len > 0.0);
-  s.spec = 0.0;
 }
 
 // Advance one element crossing.  Returns true when the walk finished and
@@ -75,26 +73,6 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
   }
   // Evaluate the 4 face planes at both segment endpoints.
   const Plane *pl = planes + (int64_t)s.elem * 4;
-  const int32_t *nrow = nbr + (int64_t)s.elem * 4;
-#if defined(__HIP_DEVICE_COMPILE__)
-  // Speculative warm-up: whichever face we exit, the next element is one of
-  // the 4 neighbors.  Touch both cache lines of every neighbor's 128-B
-  // plane row NOW (independent loads, bandwidth is idle in this
-  // latency-bound kernel) so the next crossing's dependent loads hit L1/L2.
-  // The accumulator is consumed under a branch that can never be taken, so
-  // the values are dead but the loads are not elided; vmcnt is in-order,
-  // so waiting for the real plane loads below also covers these.
-  {
-    double t = 0.0;
-#pragma unroll
-    for (int f = 0; f < 4; ++f) {
-      const int32_t nf = nrow[f];
-      const int64_t b = (int64_t)(nf > 0 ? nf : 0) * 4;
-      t += planes[b].nx + planes[b + 2].nx;
-    }
-    s.spec += t;
-  }
-#endif
   double t_exit = 2.0;
   int exit_face = -1;
 #if defined(__HIP_DEVICE_COMPILE__)
@@ -126,9 +104,6 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
     *out_elem = s.elem;
     *out_pos = s.d;
     *out_escaped = false;
-#if defined(__HIP_DEVICE_COMPILE__)
-    if (max_steps == -2147001001) out_pos->x = s.spec; // never taken; keeps spec live
-#endif
     return true;
   }
 
