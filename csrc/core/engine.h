@@ -99,6 +99,11 @@ public:
   // all-reduced global tally on rank 0 before writing).
   virtual void set_flux(const double *flux, int64_t nelems) = 0;
 
+  // Restore particle state (checkpoint/resume support -- the reference has
+  // none; a crash there loses the whole batch).
+  virtual void set_particle_state(const double *pos, const int32_t *elem,
+                                  const uint8_t *escaped, int64_t n) = 0;
+
   // Block until all queued device work is done (no-op on CPU).
   virtual void synchronize() {}
 

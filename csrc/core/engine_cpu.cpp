@@ -139,6 +139,14 @@ public:
     std::memcpy(flux_.data(), f, ne * sizeof(double));
   }
 
+  void set_particle_state(const double *pos, const int32_t *elem,
+                          const uint8_t *escaped, int64_t n) override {
+    check_n(n);
+    std::memcpy(pos_.data(), pos, n * 3 * sizeof(double));
+    std::memcpy(elem_.data(), elem, n * sizeof(int32_t));
+    std::memcpy(escaped_.data(), escaped, n);
+  }
+
 private:
   void check_n(int64_t n) const {
     if (n != n_) throw std::runtime_error("particle count mismatch");

@@ -420,6 +420,16 @@ public:
     PT_HIP_CHECK(hipMemcpy(d_flux_, f, ne * sizeof(double), hipMemcpyHostToDevice));
   }
 
+  void set_particle_state(const double *pos, const int32_t *elem,
+                          const uint8_t *escaped, int64_t n) override {
+    check_n(n);
+    PT_HIP_CHECK(hipSetDevice(device_));
+    sync();
+    PT_HIP_CHECK(hipMemcpy(d_pos_, pos, n * 3 * sizeof(double), hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_elem_, elem, n * sizeof(int32_t), hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_escaped_, escaped, n, hipMemcpyHostToDevice));
+  }
+
   void synchronize() override { sync(); }
 
   // Direct device access for the Python/bench layer (zero-copy paths).

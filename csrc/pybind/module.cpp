@@ -302,6 +302,16 @@ PYBIND11_MODULE(_core, m) {
            [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> f) {
              e.eng->set_flux(f.data(), (int64_t)f.size());
            })
+      .def("set_particle_state",
+           [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> pos,
+              py::array_t<int32_t, py::array::c_style | py::array::forcecast> elem,
+              py::array_t<uint8_t, py::array::c_style | py::array::forcecast> escaped) {
+             const int64_t n = e.eng->num_particles();
+             if ((int64_t)pos.size() != n * 3 || (int64_t)elem.size() != n ||
+                 (int64_t)escaped.size() != n)
+               throw std::runtime_error("set_particle_state: size mismatch");
+             e.eng->set_particle_state(pos.data(), elem.data(), escaped.data(), n);
+           })
       .def("stats", [](const PyEngine &e) {
         const EngineStats &s = e.eng->stats();
         py::dict d;

@@ -163,6 +163,36 @@ class TallyEngine:
     def stats(self):
         return self._eng.stats()
 
+    def save_checkpoint(self, path: str):
+        """Persist the full tally state (flux accumulator + particle
+        positions/elements/escaped flags) so a crashed batch can resume.
+        The reference has no checkpointing (a crash loses the batch)."""
+        import numpy as np
+
+        self.synchronize()
+        np.savez_compressed(
+            path,
+            flux=self.flux(),
+            positions=self.positions(),
+            elem_ids=self.elem_ids(),
+            escaped=self.escaped(),
+            num_particles=np.int64(self.num_particles),
+            nelems=np.int64(self.mesh.nelems),
+        )
+
+    def load_checkpoint(self, path: str):
+        import numpy as np
+
+        d = np.load(path)
+        if int(d["num_particles"]) != self.num_particles or \
+           int(d["nelems"]) != self.mesh.nelems:
+            raise ValueError("checkpoint does not match engine shape")
+        self._eng.set_flux(d["flux"])
+        self._eng.set_particle_state(
+            np.ascontiguousarray(d["positions"], dtype=np.float64).ravel(),
+            np.ascontiguousarray(d["elem_ids"], dtype=np.int32),
+            np.ascontiguousarray(d["escaped"], dtype=np.uint8))
+
     def normalized_flux(self):
         return _core.normalize_flux(self.mesh, self._eng.flux())
 
