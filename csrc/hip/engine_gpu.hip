@@ -30,6 +30,7 @@
 #include <algorithm>
 #include <array>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 #include <string>
@@ -195,10 +196,27 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
   }
 }
 
+int grid_cap() {
+  static int cap = [] {
+    const char *s = getenv("PUMITALLY_GRID_CAP");
+    return s ? atoi(s) : 2048;
+  }();
+  return cap;
+}
+
+int64_t chunk_particles(int64_t n) {
+  static int64_t c = [] {
+    const char *s = getenv("PUMITALLY_CHUNK");
+    return s ? (int64_t)atoll(s) : (int64_t)0;
+  }();
+  if (c > 0) return c;
+  return std::max<int64_t>(1 << 20, (n + 7) / 8);
+}
+
 int grid_blocks(int64_t work) {
   int64_t blocks = (work + kBlock - 1) / kBlock;
   // 256 CUs x 8 blocks/CU: cap and block-stride the rest (guide G11).
-  if (blocks > 2048) blocks = 2048;
+  if (blocks > grid_cap()) blocks = grid_cap();
   // round up to a multiple of 8 for the XCD-aware remap in k_move
   return (int)((blocks + 7) / 8 * 8);
 }
@@ -294,7 +312,7 @@ public:
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     // Chunked H2D/compute pipeline: copy chunk k+1 while walking chunk k.
-    const int64_t chunk = std::max<int64_t>(1 << 20, (n + 7) / 8);
+    const int64_t chunk = chunk_particles(n);
     const int nchunks = (int)((n + chunk - 1) / chunk);
     for (int k = 0; k < nchunks; ++k) {
       const int64_t lo = k * chunk, hi = std::min(n, lo + chunk);
@@ -339,7 +357,7 @@ public:
     // each XCD's Morton-contiguous particle range's mesh working set inside
     // its private 4 MiB L2.  One 10M-particle launch spreads ~18 MB per XCD
     // and thrashes L2: measured 4x slower at mean chord 32.
-    const int64_t chunk = std::max<int64_t>(1 << 20, (n + 7) / 8);
+    const int64_t chunk = chunk_particles(n);
     for (int64_t lo = 0; lo < n; lo += chunk) {
       const int64_t hi = std::min(n, lo + chunk);
       k_move<<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(

@@ -1,0 +1,122 @@
+"""Adversarial walk robustness: grazing rays, points exactly on faces,
+edges and vertices -- the degenerate geometry cases the reference
+outsources to pumipic_adjacency.tpp (SURVEY.md hard part 1)."""
+import numpy as np
+import pytest
+
+import pumiumtally_amd as pt
+
+
+def conservation(e, o, d, w):
+    e.copy_initial_position(o.ravel())
+    ok = e.elem_ids() >= 0
+    e.move(o.ravel(), d.ravel(), np.ones(len(w), np.int8), w)
+    stats = e.stats()
+    return e.flux().sum(), ok, stats
+
+
+def test_ray_along_internal_face_plane():
+    """Segments lying exactly in the x=y diagonal plane (shared faces of the
+    6-tet cells) must still tally their full length."""
+    m = pt.build_box(3, 3, 3)
+    n = 32
+    t = np.linspace(0.05, 0.95, n)
+    o = np.stack([t * 0 + 0.1, t * 0 + 0.1, t], axis=1)     # on x=y plane
+    d = np.stack([t * 0 + 0.9, t * 0 + 0.9, t], axis=1)     # stay on x=y
+    w = np.ones(n)
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, o, d, w)
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    expected = np.linalg.norm(d - o, axis=1).sum()
+    assert abs(total - expected) < 1e-9 * expected
+
+
+def test_origin_on_vertices():
+    """Origins exactly on mesh vertices localize and walk correctly."""
+    m = pt.build_box(4, 4, 4)
+    coords = m.coords
+    # interior vertices only
+    inner = coords[np.all((coords > 0.01) & (coords < 0.99), axis=1)][:20]
+    n = len(inner)
+    d = np.clip(inner + 0.13, 0.0, 0.97)
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, inner, d, np.ones(n))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    expected = np.linalg.norm(d - inner, axis=1).sum()
+    assert abs(total - expected) < 1e-9 * expected
+
+
+def test_dest_exactly_on_face_points():
+    """Destinations on cell-boundary planes (x=0.5 etc.)."""
+    m = pt.build_box(2, 2, 2)
+    n = 16
+    rng = np.random.default_rng(4)
+    o = rng.uniform(0.05, 0.45, size=(n, 3))
+    d = o.copy()
+    d[:, 0] = 0.5  # exactly on the internal grid plane
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, o, d, np.ones(n))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    expected = np.linalg.norm(d - o, axis=1).sum()
+    assert abs(total - expected) < 1e-9 * max(expected, 1e-300)
+
+
+def test_segment_along_mesh_edge():
+    """A segment running exactly along a cell edge (x=y=0.5 line)."""
+    m = pt.build_box(2, 2, 2)
+    e = pt.TallyEngine(m, 1, device="cpu")
+    o = np.array([0.5, 0.5, 0.1])
+    d = np.array([0.5, 0.5, 0.9])
+    total, ok, stats = conservation(e, o[None, :], d[None, :], np.ones(1))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    assert abs(total - 0.8) < 1e-9
+
+
+def test_zero_length_segments():
+    m = pt.build_box(2, 2, 2)
+    n = 8
+    rng = np.random.default_rng(5)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, o, o.copy(), np.ones(n))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    assert abs(total) < 1e-15
+    assert np.allclose(e.positions(), o)
+
+
+def test_long_thin_mesh_many_crossings():
+    """1000-crossing chords across an anisotropic mesh: monotone-t walk
+    must not get stuck."""
+    m = pt.build_box(200, 2, 2, 200.0, 1.0, 1.0)
+    n = 4
+    o = np.tile([0.05, 0.45, 0.55], (n, 1))
+    d = np.tile([199.95, 0.55, 0.45], (n, 1))
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, o, d, np.ones(n))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    expected = np.linalg.norm(d - o, axis=1).sum()
+    assert abs(total - expected) < 1e-8
+
+
+def test_random_stress_no_lost(seed=123):
+    """Dense random segments incl. near-degenerate short ones."""
+    m = pt.build_box(5, 5, 5)
+    n = 3000
+    rng = np.random.default_rng(seed)
+    o = rng.uniform(0.001, 0.999, size=(n, 3))
+    scale = 10.0 ** rng.uniform(-8, -0.3, n)  # lengths from 1e-8 to 0.5
+    dirv = rng.normal(size=(n, 3))
+    dirv /= np.linalg.norm(dirv, axis=1, keepdims=True)
+    d = np.clip(o + scale[:, None] * dirv, 1e-6, 1 - 1e-6)
+    e = pt.TallyEngine(m, n, device="cpu")
+    total, ok, stats = conservation(e, o, d, np.ones(n))
+    assert ok.all()
+    assert stats["lost_particles"] == 0
+    expected = np.linalg.norm(d - o, axis=1).sum()
+    assert abs(total - expected) < 1e-9 * max(1.0, expected)
