@@ -199,7 +199,7 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
 int grid_cap() {
   static int cap = [] {
     const char *s = getenv("PUMITALLY_GRID_CAP");
-    return s ? atoi(s) : 2048;
+    return s ? atoi(s) : 1024;  // swept on MI355X: 1024 > 2048 > 4096
   }();
   return cap;
 }
@@ -210,7 +210,8 @@ int64_t chunk_particles(int64_t n) {
     return s ? (int64_t)atoll(s) : (int64_t)0;
   }();
   if (c > 0) return c;
-  return std::max<int64_t>(1 << 20, (n + 7) / 8);
+  // swept on MI355X: ~2.6M-particle chunks beat 1.25M and whole-batch
+  return std::max<int64_t>((int64_t)2621440, (n + 7) / 8);
 }
 
 int grid_blocks(int64_t work) {
