@@ -37,6 +37,8 @@ def main():
                     help="use move_continue (no origin upload); NOT the headline "
                          "reference-API config -- reported with a distinct metric name")
     ap.add_argument("--no-sort", action="store_true", help="disable Morton ordering of particles")
+    ap.add_argument("--backend", type=str, default=None,
+                    help="torch.distributed backend override (nccl/gloo)")
     ap.add_argument("--device-resident", action="store_true",
                     help="inputs pre-staged in device memory (GPU transport-code "
                          "integration path); distinct metric name")
@@ -48,7 +50,7 @@ def main():
     from pumiumtally_amd.parallel import init_distributed
     from pumiumtally_amd.utils import make_box_histories
 
-    rank, world, local = init_distributed()
+    rank, world, local = init_distributed(args.backend)
     on_gpu = pt.have_gpu() and args.device != "cpu"
     if not on_gpu and args.device is None:
         # CPU fallback (debug only): shrink to something a serial walk finishes
