@@ -99,6 +99,12 @@ void PumiTally::WriteTallyResults() const {
   pimpl_->times.print();
 }
 
+double PumiTally::InitializationTime() const {
+  return pimpl_->times.initialization_time;
+}
+double PumiTally::TallyTime() const { return pimpl_->times.total_time_to_tally; }
+double PumiTally::WriteTime() const { return pimpl_->times.vtk_file_write_time; }
+
 PumiTally::~PumiTally() = default;
 
 } // namespace pumitally

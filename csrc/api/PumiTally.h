@@ -55,6 +55,14 @@ public:
 
   ~PumiTally();
 
+  // Accumulated wall-clock phase timings (seconds): initialization (mesh
+  // load + localization), tally (move calls, device-synchronized), vtk
+  // write.  The reference collects the same three but its device fences
+  // were dead code (macro-name mismatch); these are accurate.
+  double InitializationTime() const;
+  double TallyTime() const;
+  double WriteTime() const;
+
 private:
   std::unique_ptr<PumiTallyImpl> pimpl_;
 };

@@ -341,5 +341,12 @@ PYBIND11_MODULE(_core, m) {
                                   (int8_t *)flying.mutable_data(),
                                   (double *)weights.data(), (int32_t)origin.size());
            })
-      .def("write_tally_results", &PumiTally::WriteTallyResults);
+      .def("write_tally_results", &PumiTally::WriteTallyResults)
+      .def("tally_times", [](const PumiTally &t) {
+        py::dict d;
+        d["initialization_time"] = t.InitializationTime();
+        d["total_time_to_tally"] = t.TallyTime();
+        d["vtk_file_write_time"] = t.WriteTime();
+        return d;
+      });
 }

@@ -59,3 +59,17 @@ def test_facade_output_env(osh_mesh, tmp_path, monkeypatch):
     tally.copy_initial_position(init)
     tally.write_tally_results()
     assert os.path.exists(out)
+
+
+def test_facade_tally_times(osh_mesh, monkeypatch):
+    monkeypatch.setenv("PUMITALLY_DEVICE", "cpu")
+    n = 3
+    tally = pt.PumiTally(osh_mesh, n)
+    init = np.tile([0.1, 0.4, 0.5], (n, 1)).ravel()
+    tally.copy_initial_position(init)
+    dest = np.tile([0.9, 0.4, 0.5], (n, 1)).ravel()
+    tally.move_to_next_location(init, dest, np.ones(n, np.int8), np.ones(n))
+    t = tally.tally_times()
+    assert t["initialization_time"] > 0
+    assert t["total_time_to_tally"] > 0
+    assert t["vtk_file_write_time"] == 0.0
