@@ -33,6 +33,7 @@
 #include "mesh.h"
 
 #include <cstdint>
+#include <cstdlib>
 #include <memory>
 #include <string>
 #include <vector>
@@ -108,7 +109,18 @@ public:
   virtual void synchronize() {}
 
   int max_steps = 0; // 0 = auto (set by implementations from mesh size)
+
+  // fp32-traversal fast path (walk.h walk_advance32): candidate exit-face
+  // decisions in fp32, crossings/tallies in fp64.  Controlled by
+  // PUMITALLY_WALK=fp32|fp64; both engines honor it so CPU remains the
+  // bitwise oracle for the GPU in either mode.
+  bool walk_fp32 = false;
 };
+
+inline bool default_walk_fp32() {
+  const char *s = getenv("PUMITALLY_WALK");
+  return s && std::string(s) == "fp32";
+}
 
 std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles);
 

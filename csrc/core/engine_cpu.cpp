@@ -36,6 +36,7 @@ public:
       pos_[i * 3 + 2] = c0.z;
     }
     loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+    walk_fp32 = default_walk_fp32();
   }
 
   int64_t num_particles() const override { return n_; }
@@ -80,10 +81,14 @@ public:
       int32_t out_elem;
       Vec3 out_pos;
       bool out_esc;
-      walk_segment(
-          mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d, weights[i],
-          steps, [&](int32_t e, double v) { flux_[e] += v; }, &out_elem,
-          &out_pos, &out_esc);
+      auto add = [&](int32_t e, double v) { flux_[e] += v; };
+      if (walk_fp32)
+        walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
+                       mesh_.nbr.data(), elem_[i], o, d, weights[i], steps,
+                       add, &out_elem, &out_pos, &out_esc);
+      else
+        walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
+                     weights[i], steps, add, &out_elem, &out_pos, &out_esc);
       if (out_elem == kWalkLost) {
         stats_.lost_particles++;
         out_elem = elem_[i];
@@ -107,9 +112,14 @@ public:
       int32_t oe;
       Vec3 op;
       bool esc;
-      walk_segment(
-          mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d, weights[i],
-          steps, [&](int32_t e, double v) { flux_[e] += v; }, &oe, &op, &esc);
+      auto add = [&](int32_t e, double v) { flux_[e] += v; };
+      if (walk_fp32)
+        walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
+                       mesh_.nbr.data(), elem[i], o, d, weights[i], steps, add,
+                       &oe, &op, &esc);
+      else
+        walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d,
+                     weights[i], steps, add, &oe, &op, &esc);
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;

@@ -41,6 +41,21 @@ PT_HD double plane_eval(const Plane &pl, Vec3 p) {
   return pl.nx * p.x + pl.ny * p.y + pl.nz * p.z - pl.c;
 }
 
+// fp32 plane (float4 layout, 16 B) for the traversal fast path: exit-face
+// CANDIDATE decisions in fp32, the chosen face's crossing parameter and all
+// tallies in fp64 (walk.h).
+struct Plane32 {
+  float nx, ny, nz, c;
+};
+
+struct Vec3f {
+  float x, y, z;
+};
+
+PT_HD float plane_eval32(const Plane32 &pl, Vec3f p) {
+  return pl.nx * p.x + pl.ny * p.y + pl.nz * p.z - pl.c;
+}
+
 // Signed volume of tet (a,b,c,d): positive when d is on the positive side of
 // triangle (a,b,c) oriented by the right-hand rule.
 PT_HD double signed_volume(Vec3 a, Vec3 b, Vec3 c, Vec3 d) {

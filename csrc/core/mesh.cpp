@@ -104,6 +104,11 @@ void Mesh::finalize() {
     }
   }
 
+  planes32.resize(planes.size());
+  for (size_t i = 0; i < planes.size(); ++i)
+    planes32[i] = Plane32{(float)planes[i].nx, (float)planes[i].ny,
+                          (float)planes[i].nz, (float)planes[i].c};
+
   // 4. bbox
   bbox_lo = {1e300, 1e300, 1e300};
   bbox_hi = {-1e300, -1e300, -1e300};

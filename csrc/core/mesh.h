@@ -51,6 +51,7 @@ struct Mesh {
   // Derived (built by finalize()):
   std::vector<int32_t> nbr;     // nelems*4: neighbor tet across face f, -1 = boundary
   std::vector<Plane> planes;    // nelems*4: inward-positive unit-normal face planes
+  std::vector<Plane32> planes32; // fp32 copies (64 B/tet) for the traversal fast path
   std::vector<double> volumes;  // nelems
   Vec3 bbox_lo{0, 0, 0}, bbox_hi{0, 0, 0};
   LocGrid grid;

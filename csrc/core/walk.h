@@ -37,6 +37,7 @@ constexpr int32_t kWalkLost = INT32_MIN;
 // whole-segment-per-lane formulation).
 struct WalkState {
   Vec3 o, d;
+  Vec3f of, df; // fp32 copies for the fp32-traversal fast path
   double seg_len, t_cur, weight;
   int32_t elem, prev_elem;
   int step;
@@ -46,6 +47,8 @@ struct WalkState {
 PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
   s.o = o;
   s.d = d;
+  s.of = Vec3f{(float)o.x, (float)o.y, (float)o.z};
+  s.df = Vec3f{(float)d.x, (float)d.y, (float)d.z};
   s.seg_len = norm(d - o);
   s.t_cur = 0.0;
   s.weight = w;
@@ -135,6 +138,89 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
   return false;
 }
 
+// fp32-traversal variant: exit-face CANDIDATE selection runs on the fp32
+// planes (half the footprint, 2x the VALU rate -- CDNA4 fp64 vector rate is
+// half of fp32); the CHOSEN face's crossing parameter t is then recomputed
+// from the fp64 plane with the fp64 endpoints, so every tallied interval
+// and every committed position is fp64-exact and the intervals still
+// telescope to the full segment length (conservation holds to 1e-15).
+// A near-tie mis-ordering of candidate faces only re-routes the walk
+// through the neighbor for an O(1e-7)-long sliver, self-healed by the
+// monotone-t clamp.
+constexpr float kWalkTEps32 = 1e-6f;
+
+template <class FluxAdd>
+PT_HD bool walk_advance32(const Plane *__restrict__ planes,
+                          const Plane32 *__restrict__ planes32,
+                          const int32_t *__restrict__ nbr, WalkState &s,
+                          int max_steps, FluxAdd &&add, int32_t *out_elem,
+                          Vec3 *out_pos, bool *out_escaped) {
+  if (s.step++ >= max_steps) {
+    *out_elem = kWalkLost;
+    *out_pos = s.o + s.t_cur * (s.d - s.o);
+    *out_escaped = false;
+    return true;
+  }
+  const Plane32 *pl = planes32 + (int64_t)s.elem * 4;
+  const float tcur_f = (float)s.t_cur;
+  float t_exit = 2.0f;
+  int exit_face = -1;
+#if defined(__HIP_DEVICE_COMPILE__)
+#pragma unroll
+#endif
+  for (int f = 0; f < 4; ++f) {
+    const float vd = plane_eval32(pl[f], s.df);
+    if (vd < 0.0f) {
+      const float vo = plane_eval32(pl[f], s.of);
+      const float denom = vo - vd;
+      if (denom > 0.0f) {
+        const float tf = vo / denom;
+        if (nbr[(int64_t)s.elem * 4 + f] != s.prev_elem || s.prev_elem == -1) {
+          if (tf >= tcur_f - kWalkTEps32 && tf < t_exit) {
+            t_exit = tf;
+            exit_face = f;
+          }
+        }
+      }
+    }
+  }
+
+  if (exit_face < 0 || t_exit >= 1.0f) {
+    if (s.tally) add(s.elem, (1.0 - s.t_cur) * s.seg_len * s.weight);
+    *out_elem = s.elem;
+    *out_pos = s.d;
+    *out_escaped = false;
+    return true;
+  }
+
+  // Exact fp64 crossing of the chosen face.
+  const Plane &pe = planes[(int64_t)s.elem * 4 + exit_face];
+  const double vo64 = plane_eval(pe, s.o);
+  const double vd64 = plane_eval(pe, s.d);
+  double t64 = (vo64 - vd64) > 0.0 ? vo64 / (vo64 - vd64) : s.t_cur;
+  if (t64 > 1.0) t64 = 1.0;
+  const double t_clamped = t64 > s.t_cur ? t64 : s.t_cur;
+  if (s.tally) add(s.elem, (t_clamped - s.t_cur) * s.seg_len * s.weight);
+
+  const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
+  if (next == -1) {
+    *out_elem = s.elem;
+    *out_pos = s.o + t_clamped * (s.d - s.o);
+    *out_escaped = true;
+    return true;
+  }
+  if (next < -1) {
+    *out_elem = next;
+    *out_pos = s.o + t_clamped * (s.d - s.o);
+    *out_escaped = false;
+    return true;
+  }
+  s.prev_elem = s.elem;
+  s.elem = next;
+  s.t_cur = t_clamped;
+  return false;
+}
+
 // FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
 // performs atomicAdd into the flux array; on the serial CPU path a plain +=.
 template <class FluxAdd>
@@ -146,6 +232,20 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
   walk_init(s, elem, o, d, weight);
   while (!walk_advance(planes, nbr, s, max_steps, add, out_elem, out_pos,
                        out_escaped)) {
+  }
+}
+
+template <class FluxAdd>
+PT_HD void walk_segment32(const Plane *__restrict__ planes,
+                          const Plane32 *__restrict__ planes32,
+                          const int32_t *__restrict__ nbr, int32_t elem,
+                          Vec3 o, Vec3 d, double weight, int max_steps,
+                          FluxAdd &&add, int32_t *out_elem, Vec3 *out_pos,
+                          bool *out_escaped) {
+  WalkState s;
+  walk_init(s, elem, o, d, weight);
+  while (!walk_advance32(planes, planes32, nbr, s, max_steps, add, out_elem,
+                         out_pos, out_escaped)) {
   }
 }
 

@@ -92,7 +92,9 @@ __global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
 // XCD's L2 then holds one compact mesh region instead of 8 interleaved
 // copies of the whole traffic.  Purely a speed lever (guide G/T1): any
 // placement is correct.
+template <bool F32>
 __global__ void k_move(const Plane *__restrict__ planes,
+                       const Plane32 *__restrict__ planes32,
                        const int32_t *__restrict__ nbr, GridView grid,
                        const double *__restrict__ origin,
                        const double *__restrict__ dest,
@@ -140,10 +142,13 @@ __global__ void k_move(const Plane *__restrict__ planes,
     int32_t out_elem;
     Vec3 out_pos;
     bool out_esc;
-    walk_segment(
-        planes, nbr, e, o, d, weights[i], max_steps,
-        [&](int32_t el, double v) { atomicAdd(&flux[el], v); }, &out_elem,
-        &out_pos, &out_esc);
+    auto add = [&](int32_t el, double v) { atomicAdd(&flux[el], v); };
+    if constexpr (F32)
+      walk_segment32(planes, planes32, nbr, e, o, d, weights[i], max_steps,
+                     add, &out_elem, &out_pos, &out_esc);
+    else
+      walk_segment(planes, nbr, e, o, d, weights[i], max_steps, add,
+                   &out_elem, &out_pos, &out_esc);
     if (out_elem == kWalkLost) {
       atomicAdd(lost, 1ull);
       out_elem = e;
@@ -156,7 +161,9 @@ __global__ void k_move(const Plane *__restrict__ planes,
   }
 }
 
+template <bool F32>
 __global__ void k_walk_raw(const Plane *__restrict__ planes,
+                           const Plane32 *__restrict__ planes32,
                            const int32_t *__restrict__ nbr,
                            const double *__restrict__ pos,
                            const double *__restrict__ dest,
@@ -175,9 +182,13 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     int32_t oe;
     Vec3 op;
     bool esc;
-    walk_segment(
-        planes, nbr, elem[i], o, d, weights[i], max_steps,
-        [&](int32_t e, double v) { atomicAdd(&flux[e], v); }, &oe, &op, &esc);
+    auto add = [&](int32_t e, double v) { atomicAdd(&flux[e], v); };
+    if constexpr (F32)
+      walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
+                     max_steps, add, &oe, &op, &esc);
+    else
+      walk_segment(planes, nbr, elem[i], o, d, weights[i], max_steps, add,
+                   &oe, &op, &esc);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -240,11 +251,14 @@ public:
 
     // Mesh upload (once).
     d_planes_ = dmalloc<Plane>(mesh_.nelems * 4);
+    d_planes32_ = dmalloc<Plane32>(mesh_.nelems * 4);
     d_nbr_ = dmalloc<int32_t>(mesh_.nelems * 4);
     d_cell_start_ = dmalloc<int32_t>(mesh_.grid.cell_start.size());
     d_cell_tets_ = dmalloc<int32_t>(mesh_.grid.cell_tets.size() + 1);
     PT_HIP_CHECK(hipMemcpy(d_planes_, mesh_.planes.data(),
                            mesh_.nelems * 4 * sizeof(Plane), hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_planes32_, mesh_.planes32.data(),
+                           mesh_.nelems * 4 * sizeof(Plane32), hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(d_nbr_, mesh_.nbr.data(),
                            mesh_.nelems * 4 * sizeof(int32_t), hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(d_cell_start_, mesh_.grid.cell_start.data(),
@@ -271,6 +285,7 @@ public:
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
 
     loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+    walk_fp32 = default_walk_fp32();
     const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
     k_init_particles<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
         d_pos_, d_elem_, d_escaped_, n_, c0.x, c0.y, c0.z);
@@ -280,7 +295,8 @@ public:
 
   ~GpuEngine() override {
     (void)hipSetDevice(device_);
-    for (void *p : {(void *)d_planes_, (void *)d_nbr_, (void *)d_cell_start_,
+    for (void *p : {(void *)d_planes_, (void *)d_planes32_,
+                    (void *)d_nbr_, (void *)d_cell_start_,
                     (void *)d_cell_tets_, (void *)d_pos_, (void *)d_elem_,
                     (void *)d_escaped_, (void *)d_flux_, (void *)d_lost_,
                     (void *)d_origin_, (void *)d_dest_, (void *)d_flying_,
@@ -333,10 +349,18 @@ public:
       hipEvent_t ev = events_[k % events_.size()];
       PT_HIP_CHECK(hipEventRecord(ev, s_copy_));
       PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, ev, 0));
-      k_move<<<grid_blocks(m), kBlock, 0, s_comp_>>>(
-          d_planes_, d_nbr_, grid_view_, origin ? d_origin_ : nullptr, d_dest_,
-          d_flying_, d_weights_, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
-          lo, hi, loc_tol_, steps);
+      if (walk_fp32)
+        k_move<true><<<grid_blocks(m), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_,
+            origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
+            d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
+            steps);
+      else
+        k_move<false><<<grid_blocks(m), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_,
+            origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
+            d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
+            steps);
       PT_HIP_CHECK(hipGetLastError());
       PT_HIP_CHECK(hipEventRecord(done, s_comp_));
     }
@@ -361,10 +385,16 @@ public:
     const int64_t chunk = chunk_particles(n);
     for (int64_t lo = 0; lo < n; lo += chunk) {
       const int64_t hi = std::min(n, lo + chunk);
-      k_move<<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
-          d_planes_, d_nbr_, grid_view_, d_origin, d_dest, d_flying, d_weights,
-          d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
-          steps);
+      if (walk_fp32)
+        k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
+            d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
+            d_lost_, lo, hi, loc_tol_, steps);
+      else
+        k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
+            d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
+            d_lost_, lo, hi, loc_tol_, steps);
       PT_HIP_CHECK(hipGetLastError());
     }
     stats_.moves++;
@@ -385,9 +415,14 @@ public:
     PT_HIP_CHECK(hipMemcpy(dd, dest, n * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(dw, weights, n * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(de, elem, n * 4, hipMemcpyHostToDevice));
-    k_walk_raw<<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-        d_planes_, d_nbr_, dp, dd, de, dw, dop, doe, dst_, d_flux_, d_lost_, n,
-        steps);
+    if (walk_fp32)
+      k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
+          d_flux_, d_lost_, n, steps);
+    else
+      k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
+          d_flux_, d_lost_, n, steps);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
@@ -487,6 +522,7 @@ private:
   std::array<hipEvent_t, 8> comp_done_{};
 
   Plane *d_planes_ = nullptr;
+  Plane32 *d_planes32_ = nullptr;
   int32_t *d_nbr_ = nullptr;
   int32_t *d_cell_start_ = nullptr;
   int32_t *d_cell_tets_ = nullptr;
