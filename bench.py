@@ -72,7 +72,6 @@ def main():
         seed=args.seed + rank, pinned=eng.is_gpu, sort=not args.no_sort)
     eng.copy_initial_position(p0.reshape(-1))
     eng.synchronize()
-    n_located = int((eng.elem_ids() >= 0).sum()) if args.particles <= 1_000_000 else args.particles
 
     ends = (p0.reshape(-1), p1.reshape(-1))
 
