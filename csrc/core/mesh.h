@@ -97,13 +97,19 @@ struct SubMesh {
 std::vector<int32_t> partition_morton(const Mesh &m, int nparts);
 
 // Extract part `part`'s submesh.  Local faces whose global neighbor is
-// owned elsewhere get nbr = -(2+k) with foreign_gid[k]/foreign_owner[k]
+// not in the submesh get nbr = -(2+k) with foreign_gid[k]/foreign_owner[k]
 // describing the remote element; the walk stops there (walk.h).  Local
 // vertex ids ascend with global ids, so canonical face planes are
 // bitwise-identical to the full mesh's -- cross-rank walks tile segments
 // exactly like a single-mesh walk.
+//
+// ghost_rings: include that many rings of face-neighbor elements around
+// the owned set.  Particles keep walking through ghost elements (tallying
+// locally -- ghost-element tallies are summed by the flux all-reduce) and
+// hand off only when leaving the owned+ghost region, cutting the number
+// of exchange rounds for cut-hugging tracks.
 SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
-                        int part);
+                        int part, int ghost_rings = 0);
 
 // IO (implemented in mesh_io.cpp / osh_io.cpp)
 Mesh read_gmsh(const std::string &path);             // Gmsh .msh v2.2/v4.1 ASCII

@@ -49,12 +49,32 @@ std::vector<int32_t> partition_morton(const Mesh &m, int nparts) {
 }
 
 SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
-                        int part) {
+                        int part, int ghost_rings) {
   if ((int64_t)owners.size() != m.nelems)
     throw std::runtime_error("owners size mismatch");
   SubMesh sub;
+  std::vector<char> in_sub(m.nelems, 0);
   for (int64_t g = 0; g < m.nelems; ++g)
-    if (owners[g] == part) sub.elem_l2g.push_back(g);
+    if (owners[g] == part) {
+      sub.elem_l2g.push_back(g);
+      in_sub[g] = 1;
+    }
+  // grow ghost rings by face adjacency
+  std::vector<int64_t> frontier(sub.elem_l2g);
+  for (int r = 0; r < ghost_rings; ++r) {
+    std::vector<int64_t> next;
+    for (int64_t g : frontier)
+      for (int f = 0; f < 4; ++f) {
+        const int32_t nb = m.nbr[g * 4 + f];
+        if (nb >= 0 && !in_sub[nb]) {
+          in_sub[nb] = 1;
+          sub.elem_l2g.push_back(nb);
+          next.push_back(nb);
+        }
+      }
+    frontier.swap(next);
+  }
+  std::sort(sub.elem_l2g.begin(), sub.elem_l2g.end());
   const int64_t ne = (int64_t)sub.elem_l2g.size();
 
   // local vertex numbering, ascending in global id (keeps canonical face

@@ -174,10 +174,12 @@ PYBIND11_MODULE(_core, m) {
         });
   m.def("extract_submesh",
         [](const Mesh &m_, py::array_t<int32_t, py::array::c_style | py::array::forcecast> owners,
-           int part) {
+           int part, int ghost_rings) {
           std::vector<int32_t> o(owners.data(), owners.data() + owners.size());
-          return extract_submesh(m_, o, part);
-        });
+          return extract_submesh(m_, o, part, ghost_rings);
+        },
+        py::arg("mesh"), py::arg("owners"), py::arg("part"),
+        py::arg("ghost_rings") = 0);
 
   m.def("build_box", &build_box, py::arg("nx"), py::arg("ny"), py::arg("nz"),
         py::arg("lx") = 1.0, py::arg("ly") = 1.0, py::arg("lz") = 1.0);

@@ -25,14 +25,16 @@ from .dist import init_distributed
 
 
 class PartitionedTally:
-    def __init__(self, mesh, device=None, backend=None, max_rounds: int = 64):
+    def __init__(self, mesh, device=None, backend=None, max_rounds: int = 64,
+                 ghost_rings: int = 1):
         from .. import TallyEngine, _core, have_gpu
 
         self.rank, self.world, self.local = init_distributed(backend)
         self.mesh = mesh
         self.max_rounds = max_rounds
         self.owners = _core.partition_morton(mesh, self.world)
-        self.sub = _core.extract_submesh(mesh, self.owners, self.rank)
+        self.sub = _core.extract_submesh(mesh, self.owners, self.rank,
+                                         ghost_rings)
         self.l2g = self.sub.elem_l2g
         self.g2l = -np.ones(mesh.nelems, dtype=np.int64)
         self.g2l[self.l2g] = np.arange(len(self.l2g))

@@ -150,3 +150,24 @@ def test_single_rank_partitioned_gpu():
     ptal.run_segments(o, d, w)
     got = ptal.flux_global()
     assert np.abs(got - ref.flux()).max() < 1e-10
+
+
+def test_submesh_ghost_rings():
+    m = pt.build_box(4, 4, 4)
+    owners = _core.partition_morton(m, 4)
+    for p in range(4):
+        plain = _core.extract_submesh(m, owners, p)
+        ghosted = _core.extract_submesh(m, owners, p, ghost_rings=1)
+        n_owned = plain.local.nelems
+        assert ghosted.local.nelems > n_owned
+        # owned set is preserved
+        assert set(plain.elem_l2g).issubset(set(ghosted.elem_l2g))
+        # every ghost is a face-neighbor of the region grown so far and
+        # owned elsewhere
+        ghosts = set(ghosted.elem_l2g) - set(plain.elem_l2g)
+        assert all(owners[g] != p for g in ghosts)
+        # ghosting strictly reduces the cut surface per owned element
+        # (foreign refs now sit one ring further out)
+        assert (owners[ghosted.foreign_gid] != p).all()
+        # volumes still match global
+        assert np.allclose(ghosted.local.volumes, m.volumes[ghosted.elem_l2g])
