@@ -1,0 +1,82 @@
+// Latency-ceiling probe for the walk kernel's access pattern.
+//
+// The walk is a per-lane dependent chain: each element crossing needs the
+// current tet's 144-B record (4 planes + neighbors) before the next tet is
+// known.  This kernel strips away all geometry and measures the maximum
+// dependent-hop rate the chip sustains at the walk's own launch shape
+// (1024 blocks x 256 threads, ~7 waves/SIMD): one 32-B record read + one
+// 16-B next-pointer read per hop over an L3-resident table.
+// Build: hipcc --offload-arch=gfx950 -O3 -shared -fPIC tools/chase.hip -o chase.so
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+namespace {
+
+__global__ void k_fill(int32_t *next, double *rec, int64_t n, uint64_t seed) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n * 4;
+       i += stride) {
+    uint64_t x = (uint64_t)i * 6364136223846793005ull + seed;
+    x ^= x >> 33;
+    x *= 0xff51afd7ed558ccdull;
+    x ^= x >> 33;
+    next[i] = (int32_t)(x % (uint64_t)n);
+    rec[i * 4] = (double)(x & 0xffff);
+    rec[i * 4 + 1] = 1.0;
+    rec[i * 4 + 2] = 2.0;
+    rec[i * 4 + 3] = 3.0;
+  }
+}
+
+__global__ void k_chase(const int32_t *__restrict__ next,
+                        const double *__restrict__ rec, int64_t n, int hops,
+                        double *__restrict__ sink) {
+  const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t idx = (uint32_t)(((uint64_t)lane * 2654435761ull) % (uint64_t)n);
+  double acc = 0.0;
+  int f = 0;
+  for (int h = 0; h < hops; ++h) {
+    const int64_t base = (int64_t)idx * 4;
+    // one 32-B piece of the record (like one plane) + the 4 neighbor ids
+    acc += rec[base * 4 + f * 4] + rec[base * 4 + f * 4 + 1];
+    const int32_t n0 = next[base + ((f + 0) & 3)];
+    const int32_t n1 = next[base + ((f + 1) & 3)];
+    f = ((uint32_t)n0 ^ (uint32_t)h) & 3;
+    idx = (uint32_t)((f & 1) ? n1 : n0);
+  }
+  if (acc == 1e301) sink[lane] = acc;
+}
+
+} // namespace
+
+extern "C" double chase_bench(int64_t n, int hops, int blocks, int threads,
+                              int reps) {
+  int32_t *next = nullptr;
+  double *rec = nullptr, *sink = nullptr;
+  if (hipMalloc(&next, n * 4 * sizeof(int32_t)) != hipSuccess) return -1;
+  if (hipMalloc(&rec, n * 16 * sizeof(double)) != hipSuccess) return -1;
+  if (hipMalloc(&sink, (int64_t)blocks * threads * sizeof(double)) != hipSuccess)
+    return -1;
+  k_fill<<<1024, 256>>>(next, rec, n, 0x9e3779b97f4a7c15ull);
+  (void)hipDeviceSynchronize();
+  hipEvent_t a, b;
+  (void)hipEventCreate(&a);
+  (void)hipEventCreate(&b);
+  k_chase<<<blocks, threads>>>(next, rec, n, hops, sink); // warm
+  (void)hipDeviceSynchronize();
+  (void)hipEventRecord(a, 0);
+  for (int r = 0; r < reps; ++r) k_chase<<<blocks, threads>>>(next, rec, n, hops, sink);
+  (void)hipEventRecord(b, 0);
+  (void)hipEventSynchronize(b);
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, a, b);
+  const double hops_total = (double)blocks * threads * hops * reps;
+  (void)hipFree(next);
+  (void)hipFree(rec);
+  (void)hipFree(sink);
+  (void)hipEventDestroy(a);
+  (void)hipEventDestroy(b);
+  return hops_total / (ms * 1e-3); // dependent hops per second
+}

@@ -1,0 +1,29 @@
+"""Run the dependent-chase latency-ceiling probe (tools/chase.hip)."""
+import ctypes
+import subprocess
+from pathlib import Path
+
+import pumiumtally_amd  # noqa: F401  (loads the HIP runtime consistently)
+
+HERE = Path(__file__).resolve().parent
+
+
+def main():
+    so = HERE / "chase.so"
+    src = HERE / "chase.hip"
+    if not so.exists() or so.stat().st_mtime < src.stat().st_mtime:
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O3", "-shared", "-fPIC",
+             str(src), "-o", str(so)], check=True)
+    lib = ctypes.CDLL(str(so))
+    lib.chase_bench.restype = ctypes.c_double
+    lib.chase_bench.argtypes = [ctypes.c_int64, ctypes.c_int, ctypes.c_int,
+                                ctypes.c_int, ctypes.c_int]
+    n = 1_000_000  # same table size as the 1M-tet mesh
+    for blocks, threads in ((1024, 256), (2048, 256), (1024, 512)):
+        rate = lib.chase_bench(n, 64, blocks, threads, 5)
+        print(f"chase {blocks}x{threads}: {rate/1e9:.2f} G dependent hops/s")
+
+
+if __name__ == "__main__":
+    main()
