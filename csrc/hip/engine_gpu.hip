@@ -51,6 +51,20 @@ namespace {
 
 constexpr int kBlock = 256;
 
+__global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
+                                int slices) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nelems;
+       i += stride) {
+    double s = flux[i];
+    for (int k = 1; k < slices; ++k) {
+      s += flux[(int64_t)k * nelems + i];
+      flux[(int64_t)k * nelems + i] = 0.0;
+    }
+    flux[i] = s;
+  }
+}
+
 __global__ void k_init_particles(double *__restrict__ pos,
                                  int32_t *__restrict__ elem,
                                  uint8_t *__restrict__ escaped, int64_t n,
@@ -104,7 +118,12 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        uint8_t *__restrict__ escaped,
                        double *__restrict__ flux,
                        unsigned long long *__restrict__ lost, int64_t lo,
-                       int64_t hi, double loc_tol, int max_steps) {
+                       int64_t hi, double loc_tol, int max_steps,
+                       int64_t nelems, int slice_mask) {
+  // privatized tally slices: blocks in different groups accumulate into
+  // different copies of the flux array, reduced at read time -- cuts
+  // fp64-atomic pipe pressure (see profiles/README.md)
+  flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems;
   // gridDim.x is a multiple of 8 (grid_blocks).  Virtual block id: XCD x
   // gets contiguous block slots [x*bpx, (x+1)*bpx).
   const unsigned bpx = gridDim.x / 8u;
@@ -207,6 +226,19 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
   }
 }
 
+int flux_slices() {
+  static int v = [] {
+    const char *s = getenv("PUMITALLY_FLUX_SLICES");
+    int k = s ? atoi(s) : 1;
+    if (k < 1) k = 1;
+    if (k > 64) k = 64;
+    // power of two for the cheap in-kernel mask
+    while (k & (k - 1)) k--;
+    return k;
+  }();
+  return v;
+}
+
 int grid_cap() {
   static int cap = [] {
     const char *s = getenv("PUMITALLY_GRID_CAP");
@@ -275,13 +307,14 @@ public:
     d_pos_ = dmalloc<double>(n_ * 3);
     d_elem_ = dmalloc<int32_t>(n_);
     d_escaped_ = dmalloc<uint8_t>(n_);
-    d_flux_ = dmalloc<double>(mesh_.nelems);
+    slices_ = flux_slices();
+    d_flux_ = dmalloc<double>(mesh_.nelems * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
     d_origin_ = dmalloc<double>(n_ * 3);
     d_dest_ = dmalloc<double>(n_ * 3);
     d_flying_ = dmalloc<int8_t>(n_);
     d_weights_ = dmalloc<double>(n_);
-    PT_HIP_CHECK(hipMemset(d_flux_, 0, mesh_.nelems * sizeof(double)));
+    PT_HIP_CHECK(hipMemset(d_flux_, 0, mesh_.nelems * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
 
     loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
@@ -354,13 +387,13 @@ public:
             d_planes_, d_planes32_, d_nbr_, grid_view_,
             origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
             d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
-            steps);
+            steps, mesh_.nelems, slices_ - 1);
       else
         k_move<false><<<grid_blocks(m), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_,
             origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
             d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
-            steps);
+            steps, mesh_.nelems, slices_ - 1);
       PT_HIP_CHECK(hipGetLastError());
       PT_HIP_CHECK(hipEventRecord(done, s_comp_));
     }
@@ -389,12 +422,12 @@ public:
         k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
             d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
-            d_lost_, lo, hi, loc_tol_, steps);
+            d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
       else
         k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
             d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
-            d_lost_, lo, hi, loc_tol_, steps);
+            d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
       PT_HIP_CHECK(hipGetLastError());
     }
     stats_.moves++;
@@ -435,6 +468,12 @@ public:
 
   std::vector<double> flux() const override {
     sync();
+    if (slices_ > 1) {
+      k_reduce_slices<<<grid_blocks(mesh_.nelems), kBlock, 0, s_comp_>>>(
+          d_flux_, mesh_.nelems, slices_);
+      PT_HIP_CHECK(hipGetLastError());
+      PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+    }
     std::vector<double> out(mesh_.nelems);
     PT_HIP_CHECK(hipMemcpy(out.data(), d_flux_, mesh_.nelems * sizeof(double),
                            hipMemcpyDeviceToHost));
@@ -471,6 +510,7 @@ public:
 
   void set_flux(const double *f, int64_t ne) override {
     if (ne != mesh_.nelems) throw std::runtime_error("set_flux size mismatch");
+    PT_HIP_CHECK(hipMemset(d_flux_, 0, ne * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemcpy(d_flux_, f, ne * sizeof(double), hipMemcpyHostToDevice));
   }
 
@@ -528,6 +568,7 @@ private:
   int32_t *d_cell_tets_ = nullptr;
   GridView grid_view_{};
 
+  int slices_ = 1;
   double *d_pos_ = nullptr;
   int32_t *d_elem_ = nullptr;
   uint8_t *d_escaped_ = nullptr;
