@@ -3,9 +3,11 @@
 #include "engine.h"
 #include "walk.h"
 
+#include <atomic>
 #include <cmath>
 #include <cstring>
 #include <stdexcept>
+#include <thread>
 
 namespace pumitally {
 
@@ -58,48 +60,96 @@ public:
             const double *weights, int64_t n) override {
     check_n(n);
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    for (int64_t i = 0; i < n; ++i) {
-      if (!flying[i]) continue;
-      Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
-      // Phase A: relocate to the given origin (no tally).  Skipped for
-      // escaped particles (behavioral pin, see engine.h).
-      if (origin && !escaped_[i]) {
-        const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
-        if (q.x != o.x || q.y != o.y || q.z != o.z) {
-          elem_[i] = mesh_.locate(q, loc_tol_);
-          o = q;
-          stats_.relocated++;
-        }
-      }
-      // Phase B: tallied walk to the destination.
-      if (elem_[i] < 0) {
-        // outside the mesh: nothing to tally; remember requested position
-        pos_[i * 3] = o.x; pos_[i * 3 + 1] = o.y; pos_[i * 3 + 2] = o.z;
-        continue;
-      }
-      const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
-      int32_t out_elem;
-      Vec3 out_pos;
-      bool out_esc;
-      auto add = [&](int32_t e, double v) { flux_[e] += v; };
-      if (walk_fp32)
-        walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
-                       mesh_.nbr.data(), elem_[i], o, d, weights[i], steps,
-                       add, &out_elem, &out_pos, &out_esc);
-      else
-        walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
-                     weights[i], steps, add, &out_elem, &out_pos, &out_esc);
-      if (out_elem == kWalkLost) {
-        stats_.lost_particles++;
-        out_elem = elem_[i];
-      }
-      elem_[i] = out_elem;
-      pos_[i * 3] = out_pos.x;
-      pos_[i * 3 + 1] = out_pos.y;
-      pos_[i * 3 + 2] = out_pos.z;
-      escaped_[i] = out_esc ? 1 : 0;
+    // Thread-parallel over particles for large batches (the reference's CPU
+    // path is Kokkos OpenMP).  Each thread tallies into a private flux
+    // array; partials are summed in thread-index order, so results are
+    // DETERMINISTIC for a fixed thread count (and exactly serial-order for
+    // one thread).  Small batches stay serial for bitwise stability of the
+    // golden tests.
+    unsigned hw = std::thread::hardware_concurrency();
+    if (const char *env = getenv("PUMITALLY_CPU_THREADS")) {
+      const int v = atoi(env);
+      hw = v > 0 ? (unsigned)v : 1;
     }
+    const int nthreads =
+        (n >= 65536 && hw > 1) ? (int)std::min<unsigned>(hw, 64) : 1;
+    if (nthreads > 1) {
+      std::vector<std::vector<double>> partial(
+          nthreads, std::vector<double>(mesh_.nelems, 0.0));
+      std::atomic<int64_t> lost{0}, reloc{0};
+      std::vector<std::thread> workers;
+      const int64_t per = (n + nthreads - 1) / nthreads;
+      for (int t = 0; t < nthreads; ++t) {
+        workers.emplace_back([&, t] {
+          const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
+          int64_t my_lost = 0, my_reloc = 0;
+          for (int64_t i = lo; i < hi; ++i)
+            move_one(origin, dest, flying, weights, i, steps,
+                     partial[t].data(), my_lost, my_reloc);
+          lost += my_lost;
+          reloc += my_reloc;
+        });
+      }
+      for (auto &w : workers) w.join();
+      for (int t = 0; t < nthreads; ++t)
+        for (int64_t e = 0; e < mesh_.nelems; ++e) flux_[e] += partial[t][e];
+      stats_.lost_particles += lost.load();
+      stats_.relocated += reloc.load();
+      stats_.moves++;
+      return;
+    }
+    int64_t lost = 0, reloc = 0;
+    for (int64_t i = 0; i < n; ++i)
+      move_one(origin, dest, flying, weights, i, steps, flux_.data(), lost,
+               reloc);
+    stats_.lost_particles += lost;
+    stats_.relocated += reloc;
     stats_.moves++;
+  }
+
+  // One particle of a move(): phase A (relocation, skipped for escaped
+  // particles -- behavioral pin, see engine.h) + phase B tallied walk.
+  void move_one(const double *origin, const double *dest,
+                const int8_t *flying, const double *weights, int64_t i,
+                int steps, double *flux_out, int64_t &lost, int64_t &reloc) {
+    if (!flying[i]) return;
+    Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
+    if (origin && !escaped_[i]) {
+      const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
+      if (q.x != o.x || q.y != o.y || q.z != o.z) {
+        elem_[i] = mesh_.locate(q, loc_tol_);
+        o = q;
+        reloc++;
+      }
+    }
+    if (elem_[i] < 0) {
+      // outside the mesh: nothing to tally; remember requested position
+      pos_[i * 3] = o.x;
+      pos_[i * 3 + 1] = o.y;
+      pos_[i * 3 + 2] = o.z;
+      return;
+    }
+    const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+    int32_t out_elem;
+    Vec3 out_pos;
+    bool out_esc;
+    auto add = [&](int32_t e, double v) { flux_out[e] += v; };
+    if (walk_fp32)
+      walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
+                     mesh_.nbr.data(), elem_[i], o, d, weights[i], steps, add,
+                     &out_elem, &out_pos, &out_esc);
+    else
+      walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
+                   weights[i], steps, add, &out_elem, &out_pos, &out_esc);
+    if (out_elem == kWalkLost) {
+      lost++;
+      out_elem = elem_[i];
+    }
+    elem_[i] = out_elem;
+    pos_[i * 3] = out_pos.x;
+    pos_[i * 3 + 1] = out_pos.y;
+    pos_[i * 3 + 2] = out_pos.z;
+    escaped_[i] = out_esc ? 1 : 0;
   }
 
   void walk_raw(int64_t n, const double *pos, const double *dest,
