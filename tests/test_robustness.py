@@ -120,3 +120,48 @@ def test_random_stress_no_lost(seed=123):
     assert stats["lost_particles"] == 0
     expected = np.linalg.norm(d - o, axis=1).sum()
     assert abs(total - expected) < 1e-9 * max(1.0, expected)
+
+
+def test_nonconvex_mesh_void_clipping():
+    """Mesh with an interior void (notch): walks clip at the void's vacuum
+    boundary exactly like the outer boundary."""
+    full = pt.build_box(4, 4, 4)
+    coords = full.coords
+    tets = full.tet2vert
+    # remove the tets of the central 2x2x2-cell block -> interior void
+    centroids = coords[tets].mean(axis=1)
+    keep = ~np.all((centroids > 0.25) & (centroids < 0.75), axis=1)
+    m = pt.mesh_from_arrays(coords, tets[keep])
+    assert m.nelems < full.nelems
+
+    e = pt.TallyEngine(m, 1, device="cpu")
+    o = np.array([0.1, 0.4, 0.4])     # left of the void, aligned with it
+    d = np.array([0.9, 0.4, 0.4])     # would cross the void
+    e.copy_initial_position(o)
+    e.move(o, d, np.ones(1, np.int8), np.ones(1))
+    assert e.stats()["lost_particles"] == 0
+    assert e.escaped()[0] == 1
+    p = e.positions()[0]
+    assert abs(p[0] - 0.25) < 1e-9    # clipped at the void face
+    assert abs(e.flux().sum() - 0.15) < 1e-9  # only the pre-void run tallies
+
+
+def test_second_batch_after_reinitialization():
+    """CopyInitialPosition starts a new batch; flux keeps accumulating
+    (reference flow: one write at the very end)."""
+    m = pt.build_box(2, 2, 2)
+    n = 4
+    e = pt.TallyEngine(m, n, device="cpu")
+    o1 = np.tile([0.2, 0.3, 0.4], (n, 1))
+    d1 = o1 + [0.3, 0, 0]
+    e.copy_initial_position(o1.ravel())
+    e.move(o1.ravel(), d1.ravel(), np.ones(n, np.int8), np.ones(n))
+    f1 = e.flux().sum()
+    assert abs(f1 - 0.3 * n) < 1e-12
+    # new batch, new source positions
+    o2 = np.tile([0.6, 0.7, 0.2], (n, 1))
+    d2 = o2 + [0, 0.2, 0]
+    e.copy_initial_position(o2.ravel())
+    assert (e.escaped() == 0).all()
+    e.move(o2.ravel(), d2.ravel(), np.ones(n, np.int8), np.ones(n))
+    assert abs(e.flux().sum() - (0.3 + 0.2) * n) < 1e-12
