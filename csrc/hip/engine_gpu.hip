@@ -7,25 +7,28 @@
 //     (/root/reference/src/pumitally/PumiTallyImpl.cpp:66-149,243-380) --
 //     collapses into ONE fused kernel (k_move) that walks each particle's
 //     whole segment in registers and atomicAdds per-element contributions.
-//     No per-iteration kernel launches, no element-bucketed particle
-//     structure: a flat grid-stride loop over particles; walk-length
-//     variance is absorbed statistically by giving each lane many
-//     particles (grid capped at ~8 blocks/CU).
-//   * Mesh data is two flat HBM arrays consumed by the walk: 4 face planes
+//   * Mesh data is flat HBM arrays consumed by the walk: 4 face planes
 //     (double4, inward-positive unit normals) + 4 neighbor ids per tet --
 //     144 B/tet, contiguous, no indirection to vertex coords in the hot
-//     loop.  A 1M-tet mesh (~150 MB) is fully resident in the 256 MiB
+//     loop.  A 1M-tet mesh (~150 MB) is resident in the 256 MiB
 //     Infinity Cache.
-//   * Host arrays are copied H2D in chunks on a copy stream that overlaps
-//     the walk kernel of the previous chunk on the compute stream,
-//     replacing the reference's fully synchronous deep_copy + fence
-//     staging.  Pinned sources (pumiumtally_amd.pinned_array or
-//     app-registered buffers) run at full link rate; pageable sources take
-//     the runtime's internal staging path.
+//   * Particle state lives in SPATIAL (Morton) order: a device radix sort
+//     (hipCUB) orders particle slots by position at localization and
+//     periodically thereafter, so neighboring lanes walk neighboring
+//     tets and each XCD's private L2 sees one compact mesh region
+//     (measured 2.2x on the walk).  A slot->caller index map keeps the
+//     public API's particle indices unchanged; the caller's arrays are
+//     gathered through it on device.
+//   * Host input staging is parity double-buffered: step k+1's four H2D
+//     copies (on the copy stream) overlap step k's walk kernels (on the
+//     compute stream); buffer reuse is fenced with per-parity events.
+//     Pinned sources (pumiumtally_amd.pinned_array or app-registered
+//     buffers) run at full link rate.
 #include "../core/engine.h"
 #include "../core/walk.h"
 
 #include <hip/hip_runtime.h>
+#include <hipcub/hipcub.hpp>
 
 #include <algorithm>
 #include <array>
@@ -51,6 +54,93 @@ namespace {
 
 constexpr int kBlock = 256;
 
+__global__ void k_init_particles(double *__restrict__ pos,
+                                 int32_t *__restrict__ elem,
+                                 uint8_t *__restrict__ escaped,
+                                 int32_t *__restrict__ s2c, int64_t n,
+                                 double cx, double cy, double cz) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    pos[i * 3] = cx;
+    pos[i * 3 + 1] = cy;
+    pos[i * 3 + 2] = cz;
+    elem[i] = 0;
+    escaped[i] = 0;
+    s2c[i] = (int32_t)i;
+  }
+}
+
+__global__ void k_iota(int32_t *__restrict__ a, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    a[i] = (int32_t)i;
+}
+
+__global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
+                         const double *__restrict__ q,
+                         double *__restrict__ pos, int32_t *__restrict__ elem,
+                         uint8_t *__restrict__ escaped, int64_t n,
+                         double tol) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const Vec3 p{q[i * 3], q[i * 3 + 1], q[i * 3 + 2]};
+    elem[i] = grid_locate(grid, planes, p, tol);
+    pos[i * 3] = p.x;
+    pos[i * 3 + 1] = p.y;
+    pos[i * 3 + 2] = p.z;
+    escaped[i] = 0;
+  }
+}
+
+__device__ __forceinline__ uint64_t morton_spread(uint64_t v) {
+  v &= 0x1fffff;
+  v = (v | v << 32) & 0x1f00000000ffffull;
+  v = (v | v << 16) & 0x1f0000ff0000ffull;
+  v = (v | v << 8) & 0x100f00f00f00f00full;
+  v = (v | v << 4) & 0x10c30c30c30c30c3ull;
+  v = (v | v << 2) & 0x1249249249249249ull;
+  return v;
+}
+
+__global__ void k_morton_keys(const double *__restrict__ pos,
+                              uint64_t *__restrict__ keys,
+                              int32_t *__restrict__ vals, int64_t n, Vec3 lo,
+                              Vec3 scale) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    double x = (pos[i * 3] - lo.x) * scale.x;
+    double y = (pos[i * 3 + 1] - lo.y) * scale.y;
+    double z = (pos[i * 3 + 2] - lo.z) * scale.z;
+    x = x < 0 ? 0 : (x > 2097151.0 ? 2097151.0 : x);
+    y = y < 0 ? 0 : (y > 2097151.0 ? 2097151.0 : y);
+    z = z < 0 ? 0 : (z > 2097151.0 ? 2097151.0 : z);
+    keys[i] = morton_spread((uint64_t)x) | (morton_spread((uint64_t)y) << 1) |
+              (morton_spread((uint64_t)z) << 2);
+    vals[i] = (int32_t)i;
+  }
+}
+
+__global__ void k_permute_state(const int32_t *__restrict__ order,
+                                const double *__restrict__ pos_in,
+                                const int32_t *__restrict__ elem_in,
+                                const uint8_t *__restrict__ esc_in,
+                                const int32_t *__restrict__ s2c_in,
+                                double *__restrict__ pos_out,
+                                int32_t *__restrict__ elem_out,
+                                uint8_t *__restrict__ esc_out,
+                                int32_t *__restrict__ s2c_out, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = blockIdx.x * blockDim.x + threadIdx.x; j < n; j += stride) {
+    const int32_t i = order[j];
+    pos_out[j * 3] = pos_in[(int64_t)i * 3];
+    pos_out[j * 3 + 1] = pos_in[(int64_t)i * 3 + 1];
+    pos_out[j * 3 + 2] = pos_in[(int64_t)i * 3 + 2];
+    elem_out[j] = elem_in[i];
+    esc_out[j] = esc_in[i];
+    s2c_out[j] = s2c_in[i];
+  }
+}
+
 __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
                                 int slices) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -65,51 +155,19 @@ __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
   }
 }
 
-__global__ void k_init_particles(double *__restrict__ pos,
-                                 int32_t *__restrict__ elem,
-                                 uint8_t *__restrict__ escaped, int64_t n,
-                                 double cx, double cy, double cz) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-    pos[i * 3] = cx;
-    pos[i * 3 + 1] = cy;
-    pos[i * 3 + 2] = cz;
-    elem[i] = 0;
-    escaped[i] = 0;
-  }
-}
-
-__global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
-                         const double *__restrict__ q,
-                         double *__restrict__ pos, int32_t *__restrict__ elem,
-                         uint8_t *__restrict__ escaped, int64_t lo, int64_t hi,
-                         double tol) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = lo + blockIdx.x * blockDim.x + threadIdx.x; i < hi;
-       i += stride) {
-    const Vec3 p{q[i * 3], q[i * 3 + 1], q[i * 3 + 2]};
-    elem[i] = grid_locate(grid, planes, p, tol);
-    pos[i * 3] = p.x;
-    pos[i * 3 + 1] = p.y;
-    pos[i * 3 + 2] = p.z;
-    escaped[i] = 0;
-  }
-}
-
 // The fused move kernel: phase A (relocation of flying, non-escaped
 // particles whose origin changed) + phase B (tallied walk to destination).
 //
-// Block->particle mapping is XCD-aware: MI355X dispatches block b to XCD
-// b%8, and each XCD has a private 4 MiB L2.  We remap blocks so each XCD
-// owns one contiguous particle range; when the caller orders particles
-// spatially (Morton order -- pumiumtally_amd.utils.synthetic does), each
-// XCD's L2 then holds one compact mesh region instead of 8 interleaved
-// copies of the whole traffic.  Purely a speed lever (guide G/T1): any
-// placement is correct.
+// Iteration is over particle SLOTS (spatial order); the caller's arrays
+// are gathered through s2c.  Block->slot mapping is XCD-aware: MI355X
+// dispatches block b to XCD b%8 and each XCD has a private 4 MiB L2; we
+// remap blocks so each XCD owns one contiguous (hence spatially compact)
+// slot range.  Purely a speed lever: any placement is correct.
 template <bool F32>
 __global__ void k_move(const Plane *__restrict__ planes,
                        const Plane32 *__restrict__ planes32,
                        const int32_t *__restrict__ nbr, GridView grid,
+                       const int32_t *__restrict__ s2c,
                        const double *__restrict__ origin,
                        const double *__restrict__ dest,
                        const int8_t *__restrict__ flying,
@@ -120,12 +178,7 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
                        int64_t nelems, int slice_mask) {
-  // privatized tally slices: blocks in different groups accumulate into
-  // different copies of the flux array, reduced at read time -- cuts
-  // fp64-atomic pipe pressure (see profiles/README.md)
   flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems;
-  // gridDim.x is a multiple of 8 (grid_blocks).  Virtual block id: XCD x
-  // gets contiguous block slots [x*bpx, (x+1)*bpx).
   const unsigned bpx = gridDim.x / 8u;
   const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
   const int64_t m = hi - lo;
@@ -133,18 +186,13 @@ __global__ void k_move(const Plane *__restrict__ planes,
   const int64_t base = lo + (int64_t)vb * per_blk;
   const int64_t end = base + per_blk < hi ? base + per_blk : hi;
 
-  // One particle per lane, block-stride.  (A per-lane work-refill state
-  // machine was tried to attack tail divergence and measured STRICTLY
-  // worse -- 1297 vs 1600 M ps/s device-resident, 4x worse at chord 32:
-  // the kernel is latency-bound, not utilization-bound, and the extra
-  // per-crossing control flow + 7->5 waves/SIMD occupancy hurt more than
-  // the refill helped.  See profiles/README.md.)
   for (int64_t i = base + threadIdx.x; i < end; i += blockDim.x) {
-    if (!flying[i]) continue;
+    const int64_t c = s2c[i];
+    if (!flying[c]) continue;
     Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
     int32_t e = elem[i];
     if (origin != nullptr && !escaped[i]) {
-      const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
+      const Vec3 q{origin[c * 3], origin[c * 3 + 1], origin[c * 3 + 2]};
       if (q.x != o.x || q.y != o.y || q.z != o.z) {
         e = grid_locate(grid, planes, q, loc_tol);
         o = q;
@@ -157,16 +205,16 @@ __global__ void k_move(const Plane *__restrict__ planes,
       elem[i] = e;
       continue;
     }
-    const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
+    const Vec3 d{dest[c * 3], dest[c * 3 + 1], dest[c * 3 + 2]};
     int32_t out_elem;
     Vec3 out_pos;
     bool out_esc;
     auto add = [&](int32_t el, double v) { atomicAdd(&flux[el], v); };
     if constexpr (F32)
-      walk_segment32(planes, planes32, nbr, e, o, d, weights[i], max_steps,
+      walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
                      add, &out_elem, &out_pos, &out_esc);
     else
-      walk_segment(planes, nbr, e, o, d, weights[i], max_steps, add,
+      walk_segment(planes, nbr, e, o, d, weights[c], max_steps, add,
                    &out_elem, &out_pos, &out_esc);
     if (out_elem == kWalkLost) {
       atomicAdd(lost, 1ull);
@@ -232,8 +280,7 @@ int flux_slices() {
     int k = s ? atoi(s) : 1;
     if (k < 1) k = 1;
     if (k > 64) k = 64;
-    // power of two for the cheap in-kernel mask
-    while (k & (k - 1)) k--;
+    while (k & (k - 1)) k--; // power of two for the cheap in-kernel mask
     return k;
   }();
   return v;
@@ -242,9 +289,17 @@ int flux_slices() {
 int grid_cap() {
   static int cap = [] {
     const char *s = getenv("PUMITALLY_GRID_CAP");
-    return s ? atoi(s) : 1024;  // swept on MI355X: 1024 > 2048 > 4096
+    return s ? atoi(s) : 1024; // swept on MI355X: 1024 > 2048 > 4096
   }();
   return cap;
+}
+
+int sort_every() {
+  static int v = [] {
+    const char *s = getenv("PUMITALLY_SORT");
+    return s ? atoi(s) : 16; // re-sort cadence in moves; 0 disables
+  }();
+  return v;
 }
 
 int64_t chunk_particles(int64_t n) {
@@ -259,7 +314,6 @@ int64_t chunk_particles(int64_t n) {
 
 int grid_blocks(int64_t work) {
   int64_t blocks = (work + kBlock - 1) / kBlock;
-  // 256 CUs x 8 blocks/CU: cap and block-stride the rest (guide G11).
   if (blocks > grid_cap()) blocks = grid_cap();
   // round up to a multiple of 8 for the XCD-aware remap in k_move
   return (int)((blocks + 7) / 8 * 8);
@@ -278,8 +332,10 @@ public:
     device_ = device;
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_comp_, hipStreamNonBlocking));
-    for (auto &ev : events_) PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
-    for (auto &ev : comp_done_) PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    for (auto &ev : copy_ev_)
+      PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+    for (auto &ev : kernels_done_)
+      PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
 
     // Mesh upload (once).
     d_planes_ = dmalloc<Plane>(mesh_.nelems * 4);
@@ -303,40 +359,74 @@ public:
                           mesh_.grid.lo,  mesh_.grid.inv_h,
                           d_cell_start_,  d_cell_tets_};
 
-    // Particle + staging arrays.
+    // Particle state (slot order) + slot->caller map.
     d_pos_ = dmalloc<double>(n_ * 3);
     d_elem_ = dmalloc<int32_t>(n_);
     d_escaped_ = dmalloc<uint8_t>(n_);
+    d_s2c_ = dmalloc<int32_t>(n_);
     slices_ = flux_slices();
     d_flux_ = dmalloc<double>(mesh_.nelems * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
-    d_origin_ = dmalloc<double>(n_ * 3);
-    d_dest_ = dmalloc<double>(n_ * 3);
-    d_flying_ = dmalloc<int8_t>(n_);
-    d_weights_ = dmalloc<double>(n_);
     PT_HIP_CHECK(hipMemset(d_flux_, 0, mesh_.nelems * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
+
+    // Parity double-buffered input staging.
+    for (int p = 0; p < 2; ++p) {
+      d_origin_[p] = dmalloc<double>(n_ * 3);
+      d_dest_[p] = dmalloc<double>(n_ * 3);
+      d_flying_[p] = dmalloc<int8_t>(n_);
+      d_weights_[p] = dmalloc<double>(n_);
+    }
+
+    // Spatial-sort scratch.
+    sort_every_ = sort_every();
+    if (sort_every_ > 0) {
+      d_keys_ = dmalloc<uint64_t>(n_);
+      d_keys2_ = dmalloc<uint64_t>(n_);
+      d_vals_ = dmalloc<int32_t>(n_);
+      d_order_ = dmalloc<int32_t>(n_);
+      d_pos2_ = dmalloc<double>(n_ * 3);
+      d_elem2_ = dmalloc<int32_t>(n_);
+      d_esc2_ = dmalloc<uint8_t>(n_);
+      d_s2c2_ = dmalloc<int32_t>(n_);
+      size_t bytes = 0;
+      PT_HIP_CHECK(hipcub::DeviceRadixSort::SortPairs(
+          nullptr, bytes, d_keys_, d_keys2_, d_vals_, d_order_, (int)n_, 0, 64,
+          s_comp_));
+      sorttmp_bytes_ = bytes;
+      PT_HIP_CHECK(hipMalloc(&d_sorttmp_, bytes ? bytes : 1));
+      const Vec3 ext = mesh_.bbox_hi - mesh_.bbox_lo;
+      sort_scale_ = Vec3{ext.x > 0 ? 2097151.0 / ext.x : 0.0,
+                         ext.y > 0 ? 2097151.0 / ext.y : 0.0,
+                         ext.z > 0 ? 2097151.0 / ext.z : 0.0};
+    }
 
     loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
     walk_fp32 = default_walk_fp32();
     const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
     k_init_particles<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
-        d_pos_, d_elem_, d_escaped_, n_, c0.x, c0.y, c0.z);
+        d_pos_, d_elem_, d_escaped_, d_s2c_, n_, c0.x, c0.y, c0.z);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
   }
 
   ~GpuEngine() override {
     (void)hipSetDevice(device_);
-    for (void *p : {(void *)d_planes_, (void *)d_planes32_,
-                    (void *)d_nbr_, (void *)d_cell_start_,
-                    (void *)d_cell_tets_, (void *)d_pos_, (void *)d_elem_,
-                    (void *)d_escaped_, (void *)d_flux_, (void *)d_lost_,
-                    (void *)d_origin_, (void *)d_dest_, (void *)d_flying_,
-                    (void *)d_weights_})
-      (void)hipFree(p);
-    for (auto &ev : events_) (void)hipEventDestroy(ev);
-    for (auto &ev : comp_done_) (void)hipEventDestroy(ev);
+    (void)hipDeviceSynchronize();
+    for (void *p : {(void *)d_planes_, (void *)d_planes32_, (void *)d_nbr_,
+                    (void *)d_cell_start_, (void *)d_cell_tets_,
+                    (void *)d_pos_, (void *)d_elem_, (void *)d_escaped_,
+                    (void *)d_s2c_, (void *)d_flux_, (void *)d_lost_,
+                    (void *)d_origin_[0], (void *)d_origin_[1],
+                    (void *)d_dest_[0], (void *)d_dest_[1],
+                    (void *)d_flying_[0], (void *)d_flying_[1],
+                    (void *)d_weights_[0], (void *)d_weights_[1],
+                    (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
+                    (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
+                    (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
+      if (p) (void)hipFree(p);
+    for (auto &ev : copy_ev_) (void)hipEventDestroy(ev);
+    for (auto &ev : kernels_done_) (void)hipEventDestroy(ev);
     (void)hipStreamDestroy(s_copy_);
     (void)hipStreamDestroy(s_comp_);
   }
@@ -347,13 +437,16 @@ public:
   void copy_initial_position(const double *p, int64_t n) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
-    sync(); // d_origin_ may still be read by in-flight move kernels
-    stage(p, n * 3 * sizeof(double), d_origin_, s_copy_);
+    sync(); // staging buffers and state may be read by in-flight kernels
+    stage(p, n * 3 * sizeof(double), d_origin_[0], s_copy_);
     PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
+    k_iota<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(d_s2c_, n_);
     k_locate<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
-        d_planes_, grid_view_, d_origin_, d_pos_, d_elem_, d_escaped_, 0, n_,
+        d_planes_, grid_view_, d_origin_[0], d_pos_, d_elem_, d_escaped_, n_,
         loc_tol_);
     PT_HIP_CHECK(hipGetLastError());
+    if (sort_every_ > 0) spatial_sort();
+    moves_since_sort_ = 0;
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
@@ -361,46 +454,24 @@ public:
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    // Chunked H2D/compute pipeline: copy chunk k+1 while walking chunk k.
-    const int64_t chunk = chunk_particles(n);
-    const int nchunks = (int)((n + chunk - 1) / chunk);
-    for (int k = 0; k < nchunks; ++k) {
-      const int64_t lo = k * chunk, hi = std::min(n, lo + chunk);
-      const int64_t m = hi - lo;
-      // WAR ordering across move() calls: this chunk's staging reuses the
-      // same device buffer region the PREVIOUS move's chunk-k kernel reads;
-      // the copy must wait for that kernel.  (Without this, a step whose
-      // walk outlives the next step's copies tallies corrupted segments --
-      // caught by flux-conservation checks at mean chord 32.)
-      hipEvent_t done = comp_done_[k % comp_done_.size()];
-      PT_HIP_CHECK(hipStreamWaitEvent(s_copy_, done, 0));
-      if (origin)
-        stage(origin + lo * 3, m * 3 * sizeof(double), d_origin_ + lo * 3, s_copy_);
-      stage(dest + lo * 3, m * 3 * sizeof(double), d_dest_ + lo * 3, s_copy_);
-      stage(flying + lo, m * sizeof(int8_t), d_flying_ + lo, s_copy_);
-      stage(weights + lo, m * sizeof(double), d_weights_ + lo, s_copy_);
-      hipEvent_t ev = events_[k % events_.size()];
-      PT_HIP_CHECK(hipEventRecord(ev, s_copy_));
-      PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, ev, 0));
-      if (walk_fp32)
-        k_move<true><<<grid_blocks(m), kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_,
-            origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
-            d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
-            steps, mesh_.nelems, slices_ - 1);
-      else
-        k_move<false><<<grid_blocks(m), kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_,
-            origin ? d_origin_ : nullptr, d_dest_, d_flying_, d_weights_,
-            d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_,
-            steps, mesh_.nelems, slices_ - 1);
-      PT_HIP_CHECK(hipGetLastError());
-      PT_HIP_CHECK(hipEventRecord(done, s_comp_));
-    }
+    const int p = (int)(parity_++ & 1);
+    // WAR fence: buffer set p was last read by the kernels of the move
+    // two calls ago; its copies must wait for them.
+    PT_HIP_CHECK(hipStreamWaitEvent(s_copy_, kernels_done_[p], 0));
+    if (origin)
+      stage(origin, n * 3 * sizeof(double), d_origin_[p], s_copy_);
+    stage(dest, n * 3 * sizeof(double), d_dest_[p], s_copy_);
+    stage(flying, n * sizeof(int8_t), d_flying_[p], s_copy_);
+    stage(weights, n * sizeof(double), d_weights_[p], s_copy_);
+    PT_HIP_CHECK(hipEventRecord(copy_ev_[p], s_copy_));
+    PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, copy_ev_[p], 0));
+    launch_move_chunks(origin ? d_origin_[p] : nullptr, d_dest_[p],
+                       d_flying_[p], d_weights_[p], n, steps);
+    PT_HIP_CHECK(hipEventRecord(kernels_done_[p], s_comp_));
+    maybe_resort();
     // The caller may mutate or free its buffers as soon as move() returns
     // (the facade zeroes the flying array; transients die).  Block until
-    // every H2D copy has consumed them; the walk keeps running async on
-    // the compute stream.
+    // every H2D copy has consumed them; the walk keeps running async.
     PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
     stats_.moves++;
   }
@@ -411,25 +482,8 @@ public:
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    // Chunked launches even with zero staging: a ~1M-particle launch keeps
-    // each XCD's Morton-contiguous particle range's mesh working set inside
-    // its private 4 MiB L2.  One 10M-particle launch spreads ~18 MB per XCD
-    // and thrashes L2: measured 4x slower at mean chord 32.
-    const int64_t chunk = chunk_particles(n);
-    for (int64_t lo = 0; lo < n; lo += chunk) {
-      const int64_t hi = std::min(n, lo + chunk);
-      if (walk_fp32)
-        k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
-            d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
-            d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
-      else
-        k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_, d_origin, d_dest,
-            d_flying, d_weights, d_pos_, d_elem_, d_escaped_, d_flux_,
-            d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
-      PT_HIP_CHECK(hipGetLastError());
-    }
+    launch_move_chunks(d_origin, d_dest, d_flying, d_weights, n, steps);
+    maybe_resort();
     stats_.moves++;
   }
 
@@ -461,9 +515,9 @@ public:
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_elem, doe, n * 4, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_status, dst_, n, hipMemcpyDeviceToHost));
-    for (void *p : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
+    for (void *q : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
                     (void *)doe, (void *)dop, (void *)dst_})
-      (void)hipFree(p);
+      (void)hipFree(q);
   }
 
   std::vector<double> flux() const override {
@@ -479,24 +533,24 @@ public:
                            hipMemcpyDeviceToHost));
     return out;
   }
+
   std::vector<int32_t> elem_ids() const override {
-    sync();
+    auto [s2c, elem] = fetch<int32_t>(d_elem_, 1);
     std::vector<int32_t> out(n_);
-    PT_HIP_CHECK(hipMemcpy(out.data(), d_elem_, n_ * sizeof(int32_t),
-                           hipMemcpyDeviceToHost));
+    for (int64_t i = 0; i < n_; ++i) out[s2c[i]] = elem[i];
     return out;
   }
   std::vector<double> positions() const override {
-    sync();
+    auto [s2c, pos] = fetch<double>(d_pos_, 3);
     std::vector<double> out(n_ * 3);
-    PT_HIP_CHECK(hipMemcpy(out.data(), d_pos_, n_ * 3 * sizeof(double),
-                           hipMemcpyDeviceToHost));
+    for (int64_t i = 0; i < n_; ++i)
+      for (int k = 0; k < 3; ++k) out[(int64_t)s2c[i] * 3 + k] = pos[i * 3 + k];
     return out;
   }
   std::vector<uint8_t> escaped() const override {
-    sync();
+    auto [s2c, esc] = fetch<uint8_t>(d_escaped_, 1);
     std::vector<uint8_t> out(n_);
-    PT_HIP_CHECK(hipMemcpy(out.data(), d_escaped_, n_, hipMemcpyDeviceToHost));
+    for (int64_t i = 0; i < n_; ++i) out[s2c[i]] = esc[i];
     return out;
   }
 
@@ -510,6 +564,7 @@ public:
 
   void set_flux(const double *f, int64_t ne) override {
     if (ne != mesh_.nelems) throw std::runtime_error("set_flux size mismatch");
+    sync();
     PT_HIP_CHECK(hipMemset(d_flux_, 0, ne * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemcpy(d_flux_, f, ne * sizeof(double), hipMemcpyHostToDevice));
   }
@@ -519,15 +574,17 @@ public:
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     sync();
+    // caller order -> identity map; next move's cadence re-sorts
+    k_iota<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(d_s2c_, n_);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(d_pos_, pos, n * 3 * sizeof(double), hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(d_elem_, elem, n * sizeof(int32_t), hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(d_escaped_, escaped, n, hipMemcpyHostToDevice));
+    moves_since_sort_ = sort_every_; // re-sort on the next move
   }
 
   void synchronize() override { sync(); }
-
-  // Direct device access for the Python/bench layer (zero-copy paths).
-  double *device_flux() const { return d_flux_; }
 
 private:
   void check_n(int64_t n) const {
@@ -540,26 +597,90 @@ private:
   }
 
   // Async H2D from caller memory.  No implicit hipHostRegister: a
-  // registration cache outliving freed caller buffers (numpy temporaries)
-  // poisons the runtime's pinning table for unrelated later copies
-  // ("pointer does not correspond to a registered memory region").
-  // Already-pinned sources (pumiumtally_amd.pinned_array, or app buffers
-  // the host code hipHostRegister'ed itself) copy at full link rate;
-  // pageable sources take the runtime's internal staging path - always
-  // correct, just slower.  The bench and any serious host integration use
-  // pinned buffers.
+  // registration cache outliving freed caller buffers poisons the
+  // runtime's pinning table for unrelated later copies.  Pinned sources
+  // copy at full link rate; pageable sources take the runtime's internal
+  // staging path -- always correct, just slower.
   void stage(const void *src, size_t bytes, void *dst, hipStream_t s) {
     PT_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, s));
+  }
+
+  void launch_move_chunks(const double *origin, const double *dest,
+                          const int8_t *flying, const double *weights,
+                          int64_t n, int steps) {
+    // Chunked launches: a ~2.6M-slot launch keeps each XCD's Morton-
+    // contiguous slot range's mesh working set inside its private L2.
+    const int64_t chunk = chunk_particles(n);
+    for (int64_t lo = 0; lo < n; lo += chunk) {
+      const int64_t hi = std::min(n, lo + chunk);
+      if (walk_fp32)
+        k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
+            flying, weights, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
+            lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
+      else
+        k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+            d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
+            flying, weights, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
+            lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
+      PT_HIP_CHECK(hipGetLastError());
+    }
+  }
+
+  // Device Morton sort of particle slots by current position (hipCUB radix
+  // sort); measured 2.2x on the walk for spatially random particle order.
+  void spatial_sort() {
+    k_morton_keys<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
+        d_pos_, d_keys_, d_vals_, n_, mesh_.bbox_lo, sort_scale_);
+    PT_HIP_CHECK(hipGetLastError());
+    size_t bytes = sorttmp_bytes_;
+    PT_HIP_CHECK(hipcub::DeviceRadixSort::SortPairs(
+        d_sorttmp_, bytes, d_keys_, d_keys2_, d_vals_, d_order_, (int)n_, 0,
+        64, s_comp_));
+    k_permute_state<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
+        d_order_, d_pos_, d_elem_, d_escaped_, d_s2c_, d_pos2_, d_elem2_,
+        d_esc2_, d_s2c2_, n_);
+    PT_HIP_CHECK(hipGetLastError());
+    std::swap(d_pos_, d_pos2_);
+    std::swap(d_elem_, d_elem2_);
+    std::swap(d_escaped_, d_esc2_);
+    std::swap(d_s2c_, d_s2c2_);
+  }
+
+  void maybe_resort() {
+    if (sort_every_ > 0 && ++moves_since_sort_ >= sort_every_) {
+      spatial_sort();
+      moves_since_sort_ = 0;
+    }
+  }
+
+  // D2H of (s2c, state-array) for caller-order readback.
+  template <class T>
+  std::pair<std::vector<int32_t>, std::vector<T>> fetch(const T *dev,
+                                                        int comps) const {
+    sync();
+    std::vector<int32_t> s2c(n_);
+    std::vector<T> v(n_ * comps);
+    PT_HIP_CHECK(hipMemcpy(s2c.data(), d_s2c_, n_ * sizeof(int32_t),
+                           hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(v.data(), dev, n_ * comps * sizeof(T),
+                           hipMemcpyDeviceToHost));
+    return {std::move(s2c), std::move(v)};
   }
 
   Mesh mesh_;
   int64_t n_;
   int device_ = 0;
   double loc_tol_ = 1e-12;
+  int slices_ = 1;
+  int sort_every_ = 0;
+  int64_t moves_since_sort_ = 0;
+  uint64_t parity_ = 0;
+  Vec3 sort_scale_{0, 0, 0};
 
   hipStream_t s_copy_{}, s_comp_{};
-  std::array<hipEvent_t, 8> events_{};
-  std::array<hipEvent_t, 8> comp_done_{};
+  std::array<hipEvent_t, 2> copy_ev_{};
+  std::array<hipEvent_t, 2> kernels_done_{};
 
   Plane *d_planes_ = nullptr;
   Plane32 *d_planes32_ = nullptr;
@@ -568,14 +689,25 @@ private:
   int32_t *d_cell_tets_ = nullptr;
   GridView grid_view_{};
 
-  int slices_ = 1;
   double *d_pos_ = nullptr;
   int32_t *d_elem_ = nullptr;
   uint8_t *d_escaped_ = nullptr;
+  int32_t *d_s2c_ = nullptr;
   double *d_flux_ = nullptr;
   unsigned long long *d_lost_ = nullptr;
-  double *d_origin_ = nullptr, *d_dest_ = nullptr, *d_weights_ = nullptr;
-  int8_t *d_flying_ = nullptr;
+  double *d_origin_[2] = {nullptr, nullptr};
+  double *d_dest_[2] = {nullptr, nullptr};
+  int8_t *d_flying_[2] = {nullptr, nullptr};
+  double *d_weights_[2] = {nullptr, nullptr};
+
+  uint64_t *d_keys_ = nullptr, *d_keys2_ = nullptr;
+  int32_t *d_vals_ = nullptr, *d_order_ = nullptr;
+  double *d_pos2_ = nullptr;
+  int32_t *d_elem2_ = nullptr;
+  uint8_t *d_esc2_ = nullptr;
+  int32_t *d_s2c2_ = nullptr;
+  void *d_sorttmp_ = nullptr;
+  size_t sorttmp_bytes_ = 0;
 
   mutable EngineStats stats_;
 };
