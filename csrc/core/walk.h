@@ -74,10 +74,16 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
     *out_escaped = false;
     return true;
   }
-  // Evaluate the 4 face planes at both segment endpoints.
+  // Evaluate the 4 face planes at both segment endpoints.  Candidate
+  // comparisons use cross-multiplied forms (all denominators positive):
+  //   t_f >= t_cur - eps  <=>  vo >= (t_cur - eps) * den
+  //   t_f <  t_best       <=>  vo * den_best < num_best * den
+  // so the fp64 division (40-60 cycles) happens ONCE for the chosen face
+  // instead of once per candidate.
   const Plane *pl = planes + (int64_t)s.elem * 4;
-  double t_exit = 2.0;
+  double num_best = 2.0, den_best = 1.0; // t_best = 2 (no candidate yet)
   int exit_face = -1;
+  const double t_floor = s.t_cur - kWalkTEps;
 #if defined(__HIP_DEVICE_COMPILE__)
 #pragma unroll
 #endif
@@ -87,18 +93,19 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
       const double vo = plane_eval(pl[f], s.o);
       const double denom = vo - vd; // > 0 since vd < 0 <= ~vo
       if (denom > 0.0) {
-        double tf = vo / denom;
-        // Never step backwards (grazing entry can give tf slightly below
+        // Never step backwards (grazing entry can give t slightly below
         // t_cur); never consider the face we just came through.
         if (nbr[(int64_t)s.elem * 4 + f] != s.prev_elem || s.prev_elem == -1) {
-          if (tf >= s.t_cur - kWalkTEps && tf < t_exit) {
-            t_exit = tf;
+          if (vo >= t_floor * denom && vo * den_best < num_best * denom) {
+            num_best = vo;
+            den_best = denom;
             exit_face = f;
           }
         }
       }
     }
   }
+  const double t_exit = exit_face >= 0 ? num_best / den_best : 2.0;
 
   if (exit_face < 0 || t_exit >= 1.0) {
     // Destination lies inside this element: tally the final partial
