@@ -171,3 +171,34 @@ def test_submesh_ghost_rings():
         assert (owners[ghosted.foreign_gid] != p).all()
         # volumes still match global
         assert np.allclose(ghosted.local.volumes, m.volumes[ghosted.elem_l2g])
+
+
+WORKER_GPU = WORKER.replace('device="cpu", backend="gloo"',
+                            'device="cuda:0", backend="gloo"')
+
+
+@pytest.mark.gpu
+def test_gloo_world2_partitioned_gpu(tmp_path):
+    """Two ranks sharing one GPU (gloo rendezvous): the full partitioned
+    walk/handoff path on device submeshes."""
+    pytest.importorskip("torch")
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER_GPU)
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": "29923",
+        "WORLD_SIZE": "2",
+        "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = "0"
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "PART_OK" in outs[0]

@@ -150,3 +150,36 @@ def test_gpu_multistep_conservation_long_chords():
     assert e.stats()["lost_particles"] == 0
     rel = abs(total - steps * expected_per_step) / (steps * expected_per_step)
     assert rel < 1e-12, rel
+
+
+@pytest.mark.gpu
+def test_gpu_matches_cpu_on_irregular_meshes():
+    """Differential CPU-vs-GPU on anisotropic and notched (non-convex)
+    meshes, not just uniform boxes."""
+    rng = np.random.default_rng(17)
+
+    # anisotropic box
+    m1 = pt.build_box(20, 5, 3, 4.0, 1.0, 0.3)
+    # notched box (interior void)
+    full = pt.build_box(6, 6, 6)
+    cent = full.coords[full.tet2vert].mean(axis=1)
+    keep = ~np.all((cent > 0.33) & (cent < 0.67), axis=1)
+    m2 = pt.mesh_from_arrays(full.coords, full.tet2vert[keep])
+
+    for m, box in ((m1, (4.0, 1.0, 0.3)), (m2, (1.0, 1.0, 1.0))):
+        n = 8000
+        o = rng.uniform(0.01, 0.99, size=(n, 3)) * box
+        d = np.clip(o + rng.normal(0, 0.2, size=(n, 3)) * np.array(box),
+                    1e-5, np.array(box) - 1e-5)
+        w = rng.uniform(0.1, 1.0, n)
+        res = {}
+        for dev in ("cpu", "cuda"):
+            e = pt.TallyEngine(m, n, device=dev)
+            e.copy_initial_position(o.ravel())
+            e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+            e.synchronize()
+            res[dev] = (e.flux(), e.elem_ids(), e.positions())
+        assert np.array_equal(res["cpu"][1], res["cuda"][1])
+        assert np.allclose(res["cpu"][2], res["cuda"][2], atol=0, rtol=0)
+        scale = max(1.0, np.abs(res["cpu"][0]).max())
+        assert np.abs(res["cpu"][0] - res["cuda"][0]).max() < 1e-10 * scale
