@@ -60,7 +60,7 @@ def test_gloo_world2_flux_allreduce(tmp_path):
     env = dict(os.environ)
     env.update({
         "MASTER_ADDR": "127.0.0.1",
-        "MASTER_PORT": "29917",
+        "MASTER_PORT": str(20000 + (os.getpid() + 7) % 20000),
         "WORLD_SIZE": "2",
         "PT_TEST_OUT": str(tmp_path / "flux.vtk"),
         "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),

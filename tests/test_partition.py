@@ -112,7 +112,7 @@ def test_gloo_world2_partitioned(tmp_path):
     env = dict(os.environ)
     env.update({
         "MASTER_ADDR": "127.0.0.1",
-        "MASTER_PORT": "29919",
+        "MASTER_PORT": str(20000 + (os.getpid() + 9) % 20000),
         "WORLD_SIZE": "2",
         "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
     })
@@ -187,7 +187,7 @@ def test_gloo_world2_partitioned_gpu(tmp_path):
     env = dict(os.environ)
     env.update({
         "MASTER_ADDR": "127.0.0.1",
-        "MASTER_PORT": "29923",
+        "MASTER_PORT": str(20000 + (os.getpid() + 3) % 20000),
         "WORLD_SIZE": "2",
         "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
     })
