@@ -49,7 +49,83 @@ __global__ void k_chase(const int32_t *__restrict__ next,
   if (acc == 1e301) sink[lane] = acc;
 }
 
+// Variant doing the WALK's actual per-hop work: read the full 128-B plane
+// record + 16-B neighbor row, 8 fp64 plane evaluations (4 planes x 2
+// endpoints), cross-multiplied face selection, one fp64 division, and
+// (mode 1) one fp64 atomicAdd into a 1M-entry tally.
+__global__ void k_chase_walklike(const int32_t *__restrict__ next,
+                                 const double *__restrict__ rec, int64_t n,
+                                 int hops, int do_atomic,
+                                 double *__restrict__ tally,
+                                 double *__restrict__ sink) {
+  const int64_t lane = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  uint32_t idx = (uint32_t)(((uint64_t)lane * 2654435761ull) % (uint64_t)n);
+  const double ox = 0.1 + (lane & 7) * 0.01, oy = 0.2, oz = 0.3;
+  const double dx = 0.9, dy = 0.8, dz = 0.7;
+  double t_cur = 0.0;
+  double acc = 0.0;
+  for (int h = 0; h < hops; ++h) {
+    const int64_t base = (int64_t)idx * 4;
+    const double *pl = rec + base * 4;
+    double num_best = 2.0, den_best = 1.0;
+    int ef = 0;
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      const double vd =
+          pl[f * 4] * dx + pl[f * 4 + 1] * dy + pl[f * 4 + 2] * dz - pl[f * 4 + 3];
+      const double vo =
+          pl[f * 4] * ox + pl[f * 4 + 1] * oy + pl[f * 4 + 2] * oz - pl[f * 4 + 3];
+      const double den = vo - vd;
+      if (den > 0.0 && vo * den_best < num_best * den) {
+        num_best = vo < 0 ? -vo : vo;
+        den_best = den;
+        ef = f;
+      }
+    }
+    const double t = num_best / (den_best + 1.0);
+    t_cur = t > t_cur ? t_cur : t; // keep t bounded, data-dependent
+    if (do_atomic) atomicAdd(&tally[idx], t);
+    acc += t;
+    idx = (uint32_t)next[base + ef];
+  }
+  if (acc == 1e301) sink[lane] = acc + t_cur;
+}
+
 } // namespace
+
+extern "C" double chase_walklike_bench(int64_t n, int hops, int blocks,
+                                       int threads, int reps, int do_atomic) {
+  int32_t *next = nullptr;
+  double *rec = nullptr, *sink = nullptr, *tally = nullptr;
+  if (hipMalloc(&next, n * 4 * sizeof(int32_t)) != hipSuccess) return -1;
+  if (hipMalloc(&rec, n * 16 * sizeof(double)) != hipSuccess) return -1;
+  if (hipMalloc(&sink, (int64_t)blocks * threads * sizeof(double)) != hipSuccess)
+    return -1;
+  if (hipMalloc(&tally, n * sizeof(double)) != hipSuccess) return -1;
+  (void)hipMemset(tally, 0, n * sizeof(double));
+  k_fill<<<1024, 256>>>(next, rec, n, 0x9e3779b97f4a7c15ull);
+  (void)hipDeviceSynchronize();
+  hipEvent_t a, b;
+  (void)hipEventCreate(&a);
+  (void)hipEventCreate(&b);
+  k_chase_walklike<<<blocks, threads>>>(next, rec, n, hops, do_atomic, tally, sink);
+  (void)hipDeviceSynchronize();
+  (void)hipEventRecord(a, 0);
+  for (int r = 0; r < reps; ++r)
+    k_chase_walklike<<<blocks, threads>>>(next, rec, n, hops, do_atomic, tally, sink);
+  (void)hipEventRecord(b, 0);
+  (void)hipEventSynchronize(b);
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, a, b);
+  const double hops_total = (double)blocks * threads * hops * reps;
+  (void)hipFree(next);
+  (void)hipFree(rec);
+  (void)hipFree(sink);
+  (void)hipFree(tally);
+  (void)hipEventDestroy(a);
+  (void)hipEventDestroy(b);
+  return hops_total / (ms * 1e-3);
+}
 
 extern "C" double chase_bench(int64_t n, int hops, int blocks, int threads,
                               int reps) {

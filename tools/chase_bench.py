@@ -19,10 +19,17 @@ def main():
     lib.chase_bench.restype = ctypes.c_double
     lib.chase_bench.argtypes = [ctypes.c_int64, ctypes.c_int, ctypes.c_int,
                                 ctypes.c_int, ctypes.c_int]
+    lib.chase_walklike_bench.restype = ctypes.c_double
+    lib.chase_walklike_bench.argtypes = [ctypes.c_int64, ctypes.c_int,
+                                         ctypes.c_int, ctypes.c_int,
+                                         ctypes.c_int, ctypes.c_int]
     n = 1_000_000  # same table size as the 1M-tet mesh
-    for blocks, threads in ((1024, 256), (2048, 256), (1024, 512)):
+    for blocks, threads in ((1024, 256), (2048, 256)):
         rate = lib.chase_bench(n, 64, blocks, threads, 5)
-        print(f"chase {blocks}x{threads}: {rate/1e9:.2f} G dependent hops/s")
+        print(f"chase          {blocks}x{threads}: {rate/1e9:.2f} G hops/s")
+    for atom in (0, 1):
+        rate = lib.chase_walklike_bench(n, 64, 1024, 256, 5, atom)
+        print(f"chase+walkwork 1024x256 atomic={atom}: {rate/1e9:.2f} G hops/s")
 
 
 if __name__ == "__main__":
