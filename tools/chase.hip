@@ -84,7 +84,10 @@ __global__ void k_chase_walklike(const int32_t *__restrict__ next,
     }
     const double t = num_best / (den_best + 1.0);
     t_cur = t > t_cur ? t_cur : t; // keep t bounded, data-dependent
-    if (do_atomic) atomicAdd(&tally[idx], t);
+    if (do_atomic == 1) atomicAdd(&tally[idx], t);
+    if (do_atomic == 2)
+      atomicAdd((unsigned long long *)&tally[idx],
+                (unsigned long long)(t * 1048576.0));
     acc += t;
     idx = (uint32_t)next[base + ef];
   }

@@ -27,9 +27,9 @@ def main():
     for blocks, threads in ((1024, 256), (2048, 256)):
         rate = lib.chase_bench(n, 64, blocks, threads, 5)
         print(f"chase          {blocks}x{threads}: {rate/1e9:.2f} G hops/s")
-    for atom in (0, 1):
+    for atom in (0, 1, 2):
         rate = lib.chase_walklike_bench(n, 64, 1024, 256, 5, atom)
-        print(f"chase+walkwork 1024x256 atomic={atom}: {rate/1e9:.2f} G hops/s")
+        print(f"chase+walkwork 1024x256 atomic={atom} (0=none,1=f64,2=u64fixed): {rate/1e9:.2f} G hops/s")
 
 
 if __name__ == "__main__":
