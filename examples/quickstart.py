@@ -3,9 +3,13 @@
 Runs on CPU anywhere; add device="cuda:0" on an MI355X.
     python examples/quickstart.py
 """
+import os
+import sys
+
 import numpy as np
 
-import pumiumtally_amd as pt
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import pumiumtally_amd as pt  # noqa: E402
 
 # 1. a mesh: generated box here; pt.read_gmsh("reactor.msh") /
 #    pt.read_mesh("mesh.osh") for real geometry

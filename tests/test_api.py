@@ -73,3 +73,27 @@ def test_facade_tally_times(osh_mesh, monkeypatch):
     assert t["initialization_time"] > 0
     assert t["total_time_to_tally"] > 0
     assert t["vtk_file_write_time"] == 0.0
+
+
+def test_quickstart_example_runs(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    import runpy
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    runpy.run_path(os.path.join(repo, "examples", "quickstart.py"),
+                   run_name="__main__")
+    assert os.path.exists("quickstart_flux.vtk")
+
+
+def test_mesh_cli(tmp_path):
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, PYTHONPATH=repo)
+    osh = str(tmp_path / "b.osh")
+    r = subprocess.run([sys.executable, "-m", "pumiumtally_amd.mesh.cli",
+                        "box", osh, "--cells", "3"], env=env,
+                       capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    r = subprocess.run([sys.executable, "-m", "pumiumtally_amd.mesh.cli",
+                        "describe", osh], env=env, capture_output=True, text=True)
+    assert r.returncode == 0 and "elements : 162" in r.stdout
