@@ -122,6 +122,15 @@ inline bool default_walk_fp32() {
   return s && std::string(s) == "fp32";
 }
 
+// Localization tolerance (accepted signed distance below a face plane),
+// relative to the mesh bounding-box diagonal.  The reference hardcodes its
+// geometric tolerance (1e-8 at PumiTallyImpl.cpp:51); here it is runtime
+// configurable (SURVEY.md section 5 flags the hardcoding).
+inline double loc_tol_rel() {
+  const char *s = getenv("PUMITALLY_LOC_TOL");
+  return s ? atof(s) : 1e-10;
+}
+
 std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles);
 
 // Returns nullptr when no HIP device is available.

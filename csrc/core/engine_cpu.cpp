@@ -37,7 +37,7 @@ public:
       pos_[i * 3 + 1] = c0.y;
       pos_[i * 3 + 2] = c0.z;
     }
-    loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+    loc_tol_ = loc_tol_rel() * norm(mesh_.bbox_hi - mesh_.bbox_lo);
     walk_fp32 = default_walk_fp32();
   }
 

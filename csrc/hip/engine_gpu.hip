@@ -401,7 +401,7 @@ public:
                          ext.z > 0 ? 2097151.0 / ext.z : 0.0};
     }
 
-    loc_tol_ = 1e-10 * norm(mesh_.bbox_hi - mesh_.bbox_lo);
+    loc_tol_ = loc_tol_rel() * norm(mesh_.bbox_hi - mesh_.bbox_lo);
     walk_fp32 = default_walk_fp32();
     const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
     k_init_particles<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
