@@ -46,14 +46,29 @@ public:
 
   void copy_initial_position(const double *p, int64_t n) override {
     check_n(n);
-    for (int64_t i = 0; i < n; ++i) {
-      const Vec3 q{p[i * 3], p[i * 3 + 1], p[i * 3 + 2]};
-      elem_[i] = mesh_.locate(q, loc_tol_);
-      pos_[i * 3] = q.x;
-      pos_[i * 3 + 1] = q.y;
-      pos_[i * 3 + 2] = q.z;
-      escaped_[i] = 0;
+    auto locate_range = [&](int64_t lo, int64_t hi) {
+      for (int64_t i = lo; i < hi; ++i) {
+        const Vec3 q{p[i * 3], p[i * 3 + 1], p[i * 3 + 2]};
+        elem_[i] = mesh_.locate(q, loc_tol_);
+        pos_[i * 3] = q.x;
+        pos_[i * 3 + 1] = q.y;
+        pos_[i * 3 + 2] = q.z;
+        escaped_[i] = 0;
+      }
+    };
+    const unsigned hw = std::thread::hardware_concurrency();
+    if (n >= 65536 && hw > 1) {
+      const int nthreads = (int)std::min<unsigned>(hw, 64);
+      const int64_t per = (n + nthreads - 1) / nthreads;
+      std::vector<std::thread> workers;
+      for (int t = 0; t < nthreads; ++t)
+        workers.emplace_back([&, t] {
+          locate_range(t * per, std::min<int64_t>(n, (t + 1) * per));
+        });
+      for (auto &w : workers) w.join();
+      return;
     }
+    locate_range(0, n);
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
