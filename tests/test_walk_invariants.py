@@ -219,3 +219,37 @@ def test_gpu_matches_cpu_across_resorts():
     assert np.allclose(res["cpu"][2], res["cuda"][2], atol=0, rtol=0)
     scale = max(1.0, np.abs(res["cpu"][0]).max())
     assert np.abs(res["cpu"][0] - res["cuda"][0]).max() < 1e-9 * scale
+
+
+@pytest.mark.gpu
+def test_gpu_sort_disabled_config():
+    """PUMITALLY_SORT=0 (no device sorting) must stay correct."""
+    import subprocess
+    import sys
+    code = (
+        "import numpy as np, pumiumtally_amd as pt\n"
+        "m = pt.build_box(6, 6, 6)\n"
+        "n = 20000\n"
+        "rng = np.random.default_rng(11)\n"
+        "o = rng.uniform(0.02, 0.98, size=(n, 3))\n"
+        "d = np.clip(o + rng.normal(0, 0.2, size=(n, 3)), 1e-5, 1 - 1e-5)\n"
+        "w = rng.uniform(0.1, 1.0, n)\n"
+        "res = {}\n"
+        "for dev in ('cpu', 'cuda'):\n"
+        "    e = pt.TallyEngine(m, n, device=dev)\n"
+        "    e.copy_initial_position(o.ravel())\n"
+        "    for _ in range(3):\n"
+        "        e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)\n"
+        "        o, d = d, o\n"
+        "    e.synchronize()\n"
+        "    res[dev] = (e.flux(), e.elem_ids())\n"
+        "assert np.array_equal(res['cpu'][1], res['cuda'][1])\n"
+        "assert np.abs(res['cpu'][0] - res['cuda'][0]).max() < 1e-9 * max(1.0, np.abs(res['cpu'][0]).max())\n"
+        "print('SORT_OFF_OK')\n"
+    )
+    import os as _os
+    env = dict(_os.environ, PUMITALLY_SORT="0",
+               PYTHONPATH=_os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0 and "SORT_OFF_OK" in r.stdout, r.stdout + r.stderr
