@@ -236,11 +236,12 @@ def test_gpu_sort_disabled_config():
         "w = rng.uniform(0.1, 1.0, n)\n"
         "res = {}\n"
         "for dev in ('cpu', 'cuda'):\n"
+        "    a, b = o, d\n"
         "    e = pt.TallyEngine(m, n, device=dev)\n"
-        "    e.copy_initial_position(o.ravel())\n"
+        "    e.copy_initial_position(a.ravel())\n"
         "    for _ in range(3):\n"
-        "        e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)\n"
-        "        o, d = d, o\n"
+        "        e.move(a.ravel(), b.ravel(), np.ones(n, np.int8), w)\n"
+        "        a, b = b, a\n"
         "    e.synchronize()\n"
         "    res[dev] = (e.flux(), e.elem_ids())\n"
         "assert np.array_equal(res['cpu'][1], res['cuda'][1])\n"
