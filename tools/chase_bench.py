@@ -30,6 +30,11 @@ def main():
     for atom in (0, 1, 2):
         rate = lib.chase_walklike_bench(n, 64, 1024, 256, 5, atom)
         print(f"chase+walkwork 1024x256 atomic={atom} (0=none,1=f64,2=u64fixed): {rate/1e9:.2f} G hops/s")
+    lib.chase_walklike2_bench.restype = ctypes.c_double
+    lib.chase_walklike2_bench.argtypes = lib.chase_walklike_bench.argtypes
+    for atom in (0, 1):
+        rate = lib.chase_walklike2_bench(n, 64, 1024, 256, 5, atom)
+        print(f"chase+walkwork ILP2 1024x256 atomic={atom}: {rate/1e9:.2f} G hops/s")
 
 
 if __name__ == "__main__":
