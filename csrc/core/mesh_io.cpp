@@ -18,35 +18,83 @@
 
 namespace pumitally {
 
+namespace {
+// legacy binary VTK is big-endian
+void swap_write_f64(std::ofstream &f, const double *v, int64_t n) {
+  std::vector<uint64_t> buf(n);
+  for (int64_t i = 0; i < n; ++i) {
+    uint64_t x;
+    memcpy(&x, &v[i], 8);
+    x = __builtin_bswap64(x);
+    buf[i] = x;
+  }
+  f.write((const char *)buf.data(), n * 8);
+}
+void swap_write_i32(std::ofstream &f, const int32_t *v, int64_t n) {
+  std::vector<uint32_t> buf(n);
+  for (int64_t i = 0; i < n; ++i)
+    buf[i] = __builtin_bswap32((uint32_t)v[i]);
+  f.write((const char *)buf.data(), n * 4);
+}
+} // namespace
+
 void write_vtk(const std::string &path, const Mesh &m,
-               const std::vector<std::pair<std::string, std::vector<double>>> &cell_data) {
-  std::ofstream f(path);
+               const std::vector<std::pair<std::string, std::vector<double>>> &cell_data,
+               int binary /* -1 = auto (binary for big meshes) */) {
+  const bool bin = binary < 0 ? m.nelems > 200000 : binary != 0;
+  std::ofstream f(path, std::ios::binary);
   if (!f) throw std::runtime_error("cannot open " + path + " for writing");
   f << "# vtk DataFile Version 3.0\n";
   f << "pumitally flux tally\n";
-  f << "ASCII\n";
+  f << (bin ? "BINARY\n" : "ASCII\n");
   f << "DATASET UNSTRUCTURED_GRID\n";
   f << "POINTS " << m.nverts << " double\n";
   char buf[128];
-  for (int64_t v = 0; v < m.nverts; ++v) {
-    snprintf(buf, sizeof buf, "%.17g %.17g %.17g\n", m.coords[v * 3],
-             m.coords[v * 3 + 1], m.coords[v * 3 + 2]);
-    f << buf;
+  if (bin) {
+    swap_write_f64(f, m.coords.data(), m.nverts * 3);
+    f << "\n";
+  } else {
+    for (int64_t v = 0; v < m.nverts; ++v) {
+      snprintf(buf, sizeof buf, "%.17g %.17g %.17g\n", m.coords[v * 3],
+               m.coords[v * 3 + 1], m.coords[v * 3 + 2]);
+      f << buf;
+    }
   }
   f << "CELLS " << m.nelems << " " << m.nelems * 5 << "\n";
-  for (int64_t t = 0; t < m.nelems; ++t) {
-    f << "4 " << m.tet2vert[t * 4] << " " << m.tet2vert[t * 4 + 1] << " "
-      << m.tet2vert[t * 4 + 2] << " " << m.tet2vert[t * 4 + 3] << "\n";
+  if (bin) {
+    std::vector<int32_t> cells(m.nelems * 5);
+    for (int64_t t = 0; t < m.nelems; ++t) {
+      cells[t * 5] = 4;
+      for (int k = 0; k < 4; ++k) cells[t * 5 + 1 + k] = m.tet2vert[t * 4 + k];
+    }
+    swap_write_i32(f, cells.data(), m.nelems * 5);
+    f << "\n";
+  } else {
+    for (int64_t t = 0; t < m.nelems; ++t) {
+      f << "4 " << m.tet2vert[t * 4] << " " << m.tet2vert[t * 4 + 1] << " "
+        << m.tet2vert[t * 4 + 2] << " " << m.tet2vert[t * 4 + 3] << "\n";
+    }
   }
   f << "CELL_TYPES " << m.nelems << "\n";
-  for (int64_t t = 0; t < m.nelems; ++t) f << "10\n"; // VTK_TETRA
+  if (bin) {
+    std::vector<int32_t> types(m.nelems, 10); // VTK_TETRA
+    swap_write_i32(f, types.data(), m.nelems);
+    f << "\n";
+  } else {
+    for (int64_t t = 0; t < m.nelems; ++t) f << "10\n";
+  }
   if (!cell_data.empty()) {
     f << "CELL_DATA " << m.nelems << "\n";
     for (const auto &cd : cell_data) {
       f << "SCALARS " << cd.first << " double 1\nLOOKUP_TABLE default\n";
-      for (int64_t t = 0; t < m.nelems; ++t) {
-        snprintf(buf, sizeof buf, "%.17g\n", cd.second[t]);
-        f << buf;
+      if (bin) {
+        swap_write_f64(f, cd.second.data(), m.nelems);
+        f << "\n";
+      } else {
+        for (int64_t t = 0; t < m.nelems; ++t) {
+          snprintf(buf, sizeof buf, "%.17g\n", cd.second[t]);
+          f << buf;
+        }
       }
     }
   }

@@ -128,3 +128,30 @@ def test_vtk_output(tmp_path):
     assert "SCALARS flux double" in text
     assert "SCALARS volume double" in text
     assert "CELL_DATA 6" in text
+
+
+def test_vtk_binary_output(tmp_path):
+    import struct
+    m = pt.build_box(2, 2, 2)
+    from pumiumtally_amd import _core
+    out = str(tmp_path / "b.vtk")
+    # force binary mode through the facade-level writer by checking both
+    # via _core.write_tally_vtk on a big... use write_vtk binary via Mesh?
+    # The auto threshold is 200k elements; test the binary branch by
+    # parsing a file written with the C++ default auto for a small mesh
+    # (ASCII) and verify a large-mesh header says BINARY.
+    pt.write_tally_vtk(out, m, np.arange(m.nelems, dtype=float))
+    head = open(out, "rb").read(200).decode(errors="ignore")
+    assert "ASCII" in head  # small mesh -> ascii
+    big = pt.build_box(33, 33, 33)  # 215k tets -> auto binary
+    out2 = str(tmp_path / "big.vtk")
+    pt.write_tally_vtk(out2, big, np.ones(big.nelems))
+    data = open(out2, "rb").read()
+    assert b"BINARY" in data[:200]
+    # decode the first point (big-endian doubles) and check it's the origin
+    i = data.index(b"POINTS")
+    j = data.index(b"\n", i) + 1
+    x, y, z = struct.unpack(">3d", data[j:j + 24])
+    assert (x, y, z) == (0.0, 0.0, 0.0)
+    # file is much smaller than an ASCII equivalent would be
+    assert len(data) < 40_000_000
