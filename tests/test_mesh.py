@@ -132,14 +132,9 @@ def test_vtk_output(tmp_path):
 
 def test_vtk_binary_output(tmp_path):
     import struct
+    # auto mode: small meshes write ASCII, >200k-element meshes binary
     m = pt.build_box(2, 2, 2)
-    from pumiumtally_amd import _core
     out = str(tmp_path / "b.vtk")
-    # force binary mode through the facade-level writer by checking both
-    # via _core.write_tally_vtk on a big... use write_vtk binary via Mesh?
-    # The auto threshold is 200k elements; test the binary branch by
-    # parsing a file written with the C++ default auto for a small mesh
-    # (ASCII) and verify a large-mesh header says BINARY.
     pt.write_tally_vtk(out, m, np.arange(m.nelems, dtype=float))
     head = open(out, "rb").read(200).decode(errors="ignore")
     assert "ASCII" in head  # small mesh -> ascii
