@@ -10,9 +10,27 @@ import sys
 
 
 def parse_vtk_cell_data(path):
+    """Cell-data fields from a legacy VTK file, ASCII or BINARY."""
+    import struct
+
+    data = open(path, "rb").read()
+    if b"\nBINARY\n" in data[:200]:
+        fields = {}
+        pos = data.index(b"CELL_DATA")
+        ncells = int(data[pos:data.index(b"\n", pos)].split()[1])
+        while True:
+            pos = data.find(b"SCALARS", pos)
+            if pos < 0:
+                break
+            name = data[pos:data.index(b"\n", pos)].split()[1].decode()
+            pos = data.index(b"\n", data.index(b"LOOKUP_TABLE", pos)) + 1
+            fields[name] = list(struct.unpack(f">{ncells}d",
+                                              data[pos:pos + 8 * ncells]))
+            pos += 8 * ncells
+        return fields
+
     fields = {}
-    with open(path) as f:
-        lines = f.readlines()
+    lines = data.decode().splitlines()
     i = 0
     ncells = None
     while i < len(lines):
