@@ -94,7 +94,7 @@ class PartitionedTally:
         for _round in range(self.max_rounds):
             outbound = [np.zeros((0, 8)) for _ in range(self.world)]
             if len(elem):
-                out_pos, out_elem, status = self.engine._eng.walk_raw(
+                out_pos, out_elem, status = self.engine.walk_raw(
                     pos.ravel(), dst.ravel(), elem, wgt)
                 hand = status == 2
                 if hand.any():
