@@ -57,15 +57,21 @@ public:
   virtual void copy_initial_position(const double *positions, int64_t n) = 0;
 
   // origin/dest: n*3 doubles; flying: n int8; weights: n doubles.
+  // groups (optional, nullable): per-particle energy-group index in
+  // [0, ngroups); contributions land in flux[group*nelems + elem].  The
+  // reference has a single scalar tally; ngroups=1 (the default) matches
+  // it exactly.
   virtual void move(const double *origin, const double *dest,
-                    const int8_t *flying, const double *weights, int64_t n) = 0;
+                    const int8_t *flying, const double *weights, int64_t n,
+                    const uint16_t *groups = nullptr) = 0;
 
   // Fast path for callers that know no particle was resampled this step
   // (origin == committed position for every particle): skips the origin
   // upload and phase A entirely.  origin=nullptr in move() semantics.
   virtual void move_continue(const double *dest, const int8_t *flying,
-                             const double *weights, int64_t n) {
-    move(nullptr, dest, flying, weights, n);
+                             const double *weights, int64_t n,
+                             const uint16_t *groups = nullptr) {
+    move(nullptr, dest, flying, weights, n, groups);
   }
 
   // Device-resident move: all arrays already live in this engine's device
@@ -73,8 +79,9 @@ public:
   // origin may be nullptr (continue semantics).  Throws on the CPU engine.
   virtual void move_device(const double *d_origin, const double *d_dest,
                            const int8_t *d_flying, const double *d_weights,
-                           int64_t n) {
+                           int64_t n, const uint16_t *d_groups = nullptr) {
     (void)d_origin; (void)d_dest; (void)d_flying; (void)d_weights; (void)n;
+    (void)d_groups;
     throw std::runtime_error("move_device requires the GPU engine");
   }
 
@@ -89,7 +96,7 @@ public:
                         int8_t *out_status) = 0;
 
   // Read back state (host copies).
-  virtual std::vector<double> flux() const = 0;           // nelems, raw tally
+  virtual std::vector<double> flux() const = 0;           // nelems*ngroups, raw tally
   virtual std::vector<int32_t> elem_ids() const = 0;      // n
   virtual std::vector<double> positions() const = 0;      // n*3
   virtual std::vector<uint8_t> escaped() const = 0;       // n
@@ -109,6 +116,7 @@ public:
   virtual void synchronize() {}
 
   int max_steps = 0; // 0 = auto (set by implementations from mesh size)
+  int ngroups = 1;   // energy groups (flux is [ngroups x nelems])
 
   // fp32-traversal fast path (walk.h walk_advance32): candidate exit-face
   // decisions in fp32, crossings/tallies in fp64.  Controlled by
@@ -131,11 +139,12 @@ inline double loc_tol_rel() {
   return s ? atof(s) : 1e-10;
 }
 
-std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles);
+std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles,
+                                        int ngroups = 1);
 
 // Returns nullptr when no HIP device is available.
 std::unique_ptr<Engine> make_gpu_engine(Mesh mesh, int64_t num_particles,
-                                        int device);
+                                        int device, int ngroups = 1);
 
 // Normalized flux = flux / element volume (volume-only, matching the
 // reference implementation rather than its docstring:

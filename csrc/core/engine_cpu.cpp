@@ -26,8 +26,9 @@ namespace {
 
 class CpuEngine final : public Engine {
 public:
-  CpuEngine(Mesh mesh, int64_t n) : mesh_(std::move(mesh)), n_(n) {
-    flux_.assign(mesh_.nelems, 0.0);
+  CpuEngine(Mesh mesh, int64_t n, int groups) : mesh_(std::move(mesh)), n_(n) {
+    ngroups = groups < 1 ? 1 : groups;
+    flux_.assign(mesh_.nelems * ngroups, 0.0);
     pos_.resize(n_ * 3);
     elem_.assign(n_, 0);
     escaped_.assign(n_, 0);
@@ -72,7 +73,8 @@ public:
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
-            const double *weights, int64_t n) override {
+            const double *weights, int64_t n,
+            const uint16_t *groups = nullptr) override {
     check_n(n);
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     // Thread-parallel over particles for large batches (the reference's CPU
@@ -90,7 +92,7 @@ public:
         (n >= 65536 && hw > 1) ? (int)std::min<unsigned>(hw, 64) : 1;
     if (nthreads > 1) {
       std::vector<std::vector<double>> partial(
-          nthreads, std::vector<double>(mesh_.nelems, 0.0));
+          nthreads, std::vector<double>(mesh_.nelems * ngroups, 0.0));
       std::atomic<int64_t> lost{0}, reloc{0};
       std::vector<std::thread> workers;
       const int64_t per = (n + nthreads - 1) / nthreads;
@@ -99,7 +101,7 @@ public:
           const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
           int64_t my_lost = 0, my_reloc = 0;
           for (int64_t i = lo; i < hi; ++i)
-            move_one(origin, dest, flying, weights, i, steps,
+            move_one(origin, dest, flying, weights, groups, i, steps,
                      partial[t].data(), my_lost, my_reloc);
           lost += my_lost;
           reloc += my_reloc;
@@ -107,7 +109,8 @@ public:
       }
       for (auto &w : workers) w.join();
       for (int t = 0; t < nthreads; ++t)
-        for (int64_t e = 0; e < mesh_.nelems; ++e) flux_[e] += partial[t][e];
+        for (int64_t e = 0; e < mesh_.nelems * ngroups; ++e)
+          flux_[e] += partial[t][e];
       stats_.lost_particles += lost.load();
       stats_.relocated += reloc.load();
       stats_.moves++;
@@ -115,8 +118,8 @@ public:
     }
     int64_t lost = 0, reloc = 0;
     for (int64_t i = 0; i < n; ++i)
-      move_one(origin, dest, flying, weights, i, steps, flux_.data(), lost,
-               reloc);
+      move_one(origin, dest, flying, weights, groups, i, steps, flux_.data(),
+               lost, reloc);
     stats_.lost_particles += lost;
     stats_.relocated += reloc;
     stats_.moves++;
@@ -125,8 +128,9 @@ public:
   // One particle of a move(): phase A (relocation, skipped for escaped
   // particles -- behavioral pin, see engine.h) + phase B tallied walk.
   void move_one(const double *origin, const double *dest,
-                const int8_t *flying, const double *weights, int64_t i,
-                int steps, double *flux_out, int64_t &lost, int64_t &reloc) {
+                const int8_t *flying, const double *weights,
+                const uint16_t *groups, int64_t i, int steps,
+                double *flux_out, int64_t &lost, int64_t &reloc) {
     if (!flying[i]) return;
     Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
     if (origin && !escaped_[i]) {
@@ -148,7 +152,9 @@ public:
     int32_t out_elem;
     Vec3 out_pos;
     bool out_esc;
-    auto add = [&](int32_t e, double v) { flux_out[e] += v; };
+    const int64_t goff =
+        groups ? (int64_t)(groups[i] % ngroups) * mesh_.nelems : 0;
+    auto add = [&](int32_t e, double v) { flux_out[goff + e] += v; };
     if (walk_fp32)
       walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                      mesh_.nbr.data(), elem_[i], o, d, weights[i], steps, add,
@@ -238,8 +244,9 @@ private:
 
 } // namespace
 
-std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles) {
-  return std::make_unique<CpuEngine>(std::move(mesh), num_particles);
+std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles,
+                                        int ngroups) {
+  return std::make_unique<CpuEngine>(std::move(mesh), num_particles, ngroups);
 }
 
 } // namespace pumitally

@@ -172,13 +172,14 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        const double *__restrict__ dest,
                        const int8_t *__restrict__ flying,
                        const double *__restrict__ weights,
+                       const uint16_t *__restrict__ groups, int ngroups,
                        double *__restrict__ pos, int32_t *__restrict__ elem,
                        uint8_t *__restrict__ escaped,
                        double *__restrict__ flux,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
                        int64_t nelems, int slice_mask) {
-  flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems;
+  flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups;
   const unsigned bpx = gridDim.x / 8u;
   const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
   const int64_t m = hi - lo;
@@ -209,7 +210,9 @@ __global__ void k_move(const Plane *__restrict__ planes,
     int32_t out_elem;
     Vec3 out_pos;
     bool out_esc;
-    auto add = [&](int32_t el, double v) { atomicAdd(&flux[el], v); };
+    const int64_t goff =
+        groups ? (int64_t)(groups[c] % ngroups) * nelems : 0;
+    auto add = [&](int32_t el, double v) { atomicAdd(&flux[goff + el], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
                      add, &out_elem, &out_pos, &out_esc);
@@ -327,7 +330,9 @@ template <class T> T *dmalloc(int64_t count) {
 
 class GpuEngine final : public Engine {
 public:
-  GpuEngine(Mesh mesh, int64_t n, int device) : mesh_(std::move(mesh)), n_(n) {
+  GpuEngine(Mesh mesh, int64_t n, int device, int groups)
+      : mesh_(std::move(mesh)), n_(n) {
+    ngroups = groups < 1 ? 1 : groups;
     PT_HIP_CHECK(hipSetDevice(device));
     device_ = device;
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
@@ -365,9 +370,10 @@ public:
     d_escaped_ = dmalloc<uint8_t>(n_);
     d_s2c_ = dmalloc<int32_t>(n_);
     slices_ = flux_slices();
-    d_flux_ = dmalloc<double>(mesh_.nelems * slices_);
+    d_flux_ = dmalloc<double>(mesh_.nelems * ngroups * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
-    PT_HIP_CHECK(hipMemset(d_flux_, 0, mesh_.nelems * slices_ * sizeof(double)));
+    PT_HIP_CHECK(
+        hipMemset(d_flux_, 0, mesh_.nelems * ngroups * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
 
     // Parity double-buffered input staging.
@@ -376,6 +382,7 @@ public:
       d_dest_[p] = dmalloc<double>(n_ * 3);
       d_flying_[p] = dmalloc<int8_t>(n_);
       d_weights_[p] = dmalloc<double>(n_);
+      d_groups_[p] = ngroups > 1 ? dmalloc<uint16_t>(n_) : nullptr;
     }
 
     // Spatial-sort scratch.
@@ -421,6 +428,7 @@ public:
                     (void *)d_dest_[0], (void *)d_dest_[1],
                     (void *)d_flying_[0], (void *)d_flying_[1],
                     (void *)d_weights_[0], (void *)d_weights_[1],
+                    (void *)d_groups_[0], (void *)d_groups_[1],
                     (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
                     (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
@@ -450,7 +458,8 @@ public:
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
-            const double *weights, int64_t n) override {
+            const double *weights, int64_t n,
+            const uint16_t *groups = nullptr) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -463,10 +472,13 @@ public:
     stage(dest, n * 3 * sizeof(double), d_dest_[p], s_copy_);
     stage(flying, n * sizeof(int8_t), d_flying_[p], s_copy_);
     stage(weights, n * sizeof(double), d_weights_[p], s_copy_);
+    if (groups && d_groups_[p])
+      stage(groups, n * sizeof(uint16_t), d_groups_[p], s_copy_);
     PT_HIP_CHECK(hipEventRecord(copy_ev_[p], s_copy_));
     PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, copy_ev_[p], 0));
     launch_move_chunks(origin ? d_origin_[p] : nullptr, d_dest_[p],
-                       d_flying_[p], d_weights_[p], n, steps);
+                       d_flying_[p], d_weights_[p],
+                       groups ? d_groups_[p] : nullptr, n, steps);
     PT_HIP_CHECK(hipEventRecord(kernels_done_[p], s_comp_));
     maybe_resort();
     // The caller may mutate or free its buffers as soon as move() returns
@@ -477,12 +489,13 @@ public:
   }
 
   void move_device(const double *d_origin, const double *d_dest,
-                   const int8_t *d_flying, const double *d_weights,
-                   int64_t n) override {
+                   const int8_t *d_flying, const double *d_weights, int64_t n,
+                   const uint16_t *d_groups = nullptr) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    launch_move_chunks(d_origin, d_dest, d_flying, d_weights, n, steps);
+    launch_move_chunks(d_origin, d_dest, d_flying, d_weights, d_groups, n,
+                       steps);
     maybe_resort();
     stats_.moves++;
   }
@@ -522,14 +535,15 @@ public:
 
   std::vector<double> flux() const override {
     sync();
+    const int64_t fsz = mesh_.nelems * ngroups;
     if (slices_ > 1) {
-      k_reduce_slices<<<grid_blocks(mesh_.nelems), kBlock, 0, s_comp_>>>(
-          d_flux_, mesh_.nelems, slices_);
+      k_reduce_slices<<<grid_blocks(fsz), kBlock, 0, s_comp_>>>(d_flux_, fsz,
+                                                               slices_);
       PT_HIP_CHECK(hipGetLastError());
       PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     }
-    std::vector<double> out(mesh_.nelems);
-    PT_HIP_CHECK(hipMemcpy(out.data(), d_flux_, mesh_.nelems * sizeof(double),
+    std::vector<double> out(fsz);
+    PT_HIP_CHECK(hipMemcpy(out.data(), d_flux_, fsz * sizeof(double),
                            hipMemcpyDeviceToHost));
     return out;
   }
@@ -563,7 +577,8 @@ public:
   }
 
   void set_flux(const double *f, int64_t ne) override {
-    if (ne != mesh_.nelems) throw std::runtime_error("set_flux size mismatch");
+    if (ne != mesh_.nelems * ngroups)
+      throw std::runtime_error("set_flux size mismatch");
     sync();
     PT_HIP_CHECK(hipMemset(d_flux_, 0, ne * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemcpy(d_flux_, f, ne * sizeof(double), hipMemcpyHostToDevice));
@@ -607,7 +622,7 @@ private:
 
   void launch_move_chunks(const double *origin, const double *dest,
                           const int8_t *flying, const double *weights,
-                          int64_t n, int steps) {
+                          const uint16_t *groups, int64_t n, int steps) {
     // Chunked launches: a ~2.6M-slot launch keeps each XCD's Morton-
     // contiguous slot range's mesh working set inside its private L2.
     const int64_t chunk = chunk_particles(n);
@@ -616,13 +631,15 @@ private:
       if (walk_fp32)
         k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
-            flying, weights, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
-            lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
+            flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
+            d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
+            slices_ - 1);
       else
         k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
-            flying, weights, d_pos_, d_elem_, d_escaped_, d_flux_, d_lost_,
-            lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1);
+            flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
+            d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
+            slices_ - 1);
       PT_HIP_CHECK(hipGetLastError());
     }
   }
@@ -696,6 +713,7 @@ private:
   double *d_flux_ = nullptr;
   unsigned long long *d_lost_ = nullptr;
   double *d_origin_[2] = {nullptr, nullptr};
+  uint16_t *d_groups_[2] = {nullptr, nullptr};
   double *d_dest_[2] = {nullptr, nullptr};
   int8_t *d_flying_[2] = {nullptr, nullptr};
   double *d_weights_[2] = {nullptr, nullptr};
@@ -715,13 +733,14 @@ private:
 } // namespace
 
 std::unique_ptr<Engine> make_gpu_engine(Mesh mesh, int64_t num_particles,
-                                        int device) {
+                                        int device, int ngroups) {
   int count = 0;
   if (hipGetDeviceCount(&count) != hipSuccess || count <= device) {
     (void)hipGetLastError();
     return nullptr;
   }
-  return std::make_unique<GpuEngine>(std::move(mesh), num_particles, device);
+  return std::make_unique<GpuEngine>(std::move(mesh), num_particles, device,
+                                     ngroups);
 }
 
 } // namespace pumitally

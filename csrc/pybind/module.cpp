@@ -40,20 +40,21 @@ struct PyEngine {
   std::unique_ptr<Engine> eng;
   bool gpu = false;
 
-  PyEngine(const Mesh &mesh, int64_t n, const std::string &device) {
+  PyEngine(const Mesh &mesh, int64_t n, const std::string &device,
+           int ngroups = 1) {
     if (device == "cpu") {
-      eng = make_cpu_engine(mesh, n);
+      eng = make_cpu_engine(mesh, n, ngroups);
     } else {
       int ordinal = 0;
       if (device.rfind("cuda:", 0) == 0) ordinal = std::stoi(device.substr(5));
       else if (device.rfind("gpu:", 0) == 0) ordinal = std::stoi(device.substr(4));
       else if (device != "cuda" && device != "gpu" && device != "auto")
         throw std::runtime_error("device must be cpu/cuda[:N]/auto");
-      eng = make_gpu_engine(mesh, n, ordinal);
+      eng = make_gpu_engine(mesh, n, ordinal, ngroups);
       if (eng) {
         gpu = true;
       } else if (device == "auto") {
-        eng = make_cpu_engine(mesh, n);
+        eng = make_cpu_engine(mesh, n, ngroups);
       } else {
         // Fail loudly: a GPU was requested but none is usable.  GPU tests
         // must never fall back silently to the CPU oracle.
@@ -138,6 +139,17 @@ PYBIND11_MODULE(_core, m) {
            })
       .def("write_vtk",
            [](const Mesh &m_, const std::string &path) { write_vtk(path, m_, {}); })
+      .def("write_vtk_fields",
+           [](const Mesh &m_, const std::string &path, py::list fields) {
+             std::vector<std::pair<std::string, std::vector<double>>> cd;
+             for (auto item : fields) {
+               auto t = item.cast<py::tuple>();
+               auto name = t[0].cast<std::string>();
+               auto arr = t[1].cast<py::array_t<double, py::array::c_style | py::array::forcecast>>();
+               cd.emplace_back(name, std::vector<double>(arr.data(), arr.data() + arr.size()));
+             }
+             write_vtk(path, m_, cd);
+           })
       .def("write_osh",
            [](const Mesh &m_, const std::string &dir) { write_osh(dir, m_); });
 
@@ -206,8 +218,11 @@ PYBIND11_MODULE(_core, m) {
         });
 
   py::class_<PyEngine>(m, "Engine")
-      .def(py::init<const Mesh &, int64_t, const std::string &>(),
-           py::arg("mesh"), py::arg("num_particles"), py::arg("device") = "auto")
+      .def(py::init<const Mesh &, int64_t, const std::string &, int>(),
+           py::arg("mesh"), py::arg("num_particles"), py::arg("device") = "auto",
+           py::arg("ngroups") = 1)
+      .def_property_readonly("ngroups",
+                             [](const PyEngine &e) { return e.eng->ngroups; })
       .def_property_readonly("num_particles",
                              [](const PyEngine &e) { return e.eng->num_particles(); })
       .def_property_readonly("is_gpu", [](const PyEngine &e) { return e.gpu; })
@@ -224,15 +239,26 @@ PYBIND11_MODULE(_core, m) {
            [](PyEngine &e, py::array_t<double, py::array::c_style> origin,
               py::array_t<double, py::array::c_style> dest,
               py::array_t<int8_t, py::array::c_style> flying,
-              py::array_t<double, py::array::c_style> weights) {
+              py::array_t<double, py::array::c_style> weights,
+              py::object groups) {
              const int64_t n = e.eng->num_particles();
              if ((int64_t)origin.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
                  (int64_t)flying.size() != n || (int64_t)weights.size() != n)
                throw std::runtime_error("move: array size mismatch");
+             const uint16_t *gp = nullptr;
+             py::array_t<uint16_t, py::array::c_style | py::array::forcecast> garr;
+             if (!groups.is_none()) {
+               garr = groups.cast<py::array_t<uint16_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)garr.size() != n)
+                 throw std::runtime_error("move: groups size mismatch");
+               gp = garr.data();
+             }
              py::gil_scoped_release nogil;
              e.eng->move(origin.data(), dest.data(), flying.data(),
-                         weights.data(), n);
-           })
+                         weights.data(), n, gp);
+           },
+           py::arg("origin"), py::arg("dest"), py::arg("flying"),
+           py::arg("weights"), py::arg("groups") = py::none())
       .def("move_continue",
            [](PyEngine &e, py::array_t<double, py::array::c_style> dest,
               py::array_t<int8_t, py::array::c_style> flying,

@@ -68,9 +68,11 @@ class TallyEngine:
     Semantics follow the reference PumiTally flow; see csrc/core/engine.h.
     """
 
-    def __init__(self, mesh, num_particles: int, device: str = "auto"):
-        self._eng = _core.Engine(mesh, num_particles, device)
+    def __init__(self, mesh, num_particles: int, device: str = "auto",
+                 ngroups: int = 1):
+        self._eng = _core.Engine(mesh, num_particles, device, ngroups)
         self.mesh = mesh
+        self.ngroups = ngroups
 
     @property
     def num_particles(self) -> int:
@@ -91,8 +93,12 @@ class TallyEngine:
     def copy_initial_position(self, positions):
         self._eng.copy_initial_position(positions)
 
-    def move(self, origin, dest, flying, weights):
-        self._eng.move(origin, dest, flying, weights)
+    def move(self, origin, dest, flying, weights, groups=None):
+        """One transport step.  groups (optional): per-particle energy-group
+        indices (uint16, in [0, ngroups)); contributions land in
+        flux()[group, elem].  The reference has a single scalar tally;
+        ngroups=1 (default) matches it exactly."""
+        self._eng.move(origin, dest, flying, weights, groups)
 
     def move_continue(self, dest, flying, weights):
         """move() without the phase-A origin upload: valid when no particle
@@ -152,7 +158,11 @@ class TallyEngine:
         self._eng.synchronize()
 
     def flux(self):
-        return self._eng.flux()
+        """Raw tally: shape (nelems,) for ngroups=1, else (ngroups, nelems)."""
+        f = self._eng.flux()
+        if self.ngroups > 1:
+            return f.reshape(self.ngroups, self.mesh.nelems)
+        return f
 
     def set_flux(self, flux):
         self._eng.set_flux(flux)
@@ -203,6 +213,15 @@ class TallyEngine:
         return _core.normalize_flux(self.mesh, self._eng.flux())
 
     def write_tally_results(self, filename: str = "fluxresult.vtk"):
+        if self.ngroups > 1:
+            # one normalized field per energy group + the total
+            f = self.flux()
+            fields = [("flux", _core.normalize_flux(self.mesh, f.sum(axis=0)))]
+            fields += [(f"flux_g{g}", _core.normalize_flux(self.mesh, f[g]))
+                       for g in range(self.ngroups)]
+            import numpy as np
+            self.mesh.write_vtk_fields(filename, fields)
+            return
         _core.write_tally_vtk(filename, self.mesh, self._eng.flux())
 
 
