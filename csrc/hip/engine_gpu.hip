@@ -455,6 +455,11 @@ public:
     PT_HIP_CHECK(hipGetLastError());
     if (sort_every_ > 0) spatial_sort();
     moves_since_sort_ = 0;
+    // k_locate reads the staging buffer d_origin_[0]; an immediately
+    // following move() would re-stage into it with no recorded fence
+    // (kernels_done_[0] has never been recorded at that point).  This is
+    // a once-per-batch call: block until localization/sort finish.
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
