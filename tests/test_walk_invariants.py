@@ -254,3 +254,26 @@ def test_gpu_sort_disabled_config():
     r = subprocess.run([sys.executable, "-c", code], env=env,
                        capture_output=True, text=True, timeout=300)
     assert r.returncode == 0 and "SORT_OFF_OK" in r.stdout, r.stdout + r.stderr
+
+
+@pytest.mark.gpu
+def test_immediate_move_after_localization():
+    """copy_initial_position followed IMMEDIATELY by move (no sync):
+    regression test for the first-move staging race on d_origin_[0]."""
+    import pumiumtally_amd as ptm
+    from pumiumtally_amd.mesh import box_mesh_with_tets
+
+    mesh, _ = box_mesh_with_tets(500_000)
+    n = 2_000_000  # large enough that localization takes a while
+    rng = np.random.default_rng(41)
+    o = rng.uniform(0.01, 0.99, size=(n, 3))
+    d = np.clip(o + rng.normal(0, 0.05, size=(n, 3)), 1e-5, 1 - 1e-5)
+    w = rng.uniform(0.1, 1.0, n)
+    e = ptm.TallyEngine(mesh, n, device="cuda")
+    e.copy_initial_position(o.ravel())
+    e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)  # no sync between
+    e.synchronize()
+    expected = float((np.linalg.norm(d - o, axis=1) * w).sum())
+    got = float(e.flux().sum())
+    assert abs(got - expected) < 1e-10 * expected
+    assert e.stats()["lost_particles"] == 0
