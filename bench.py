@@ -37,6 +37,8 @@ def main():
                     help="use move_continue (no origin upload); NOT the headline "
                          "reference-API config -- reported with a distinct metric name")
     ap.add_argument("--no-sort", action="store_true", help="disable Morton ordering of particles")
+    ap.add_argument("--ngroups", type=int, default=1,
+                    help="energy groups (random per-particle group indices)")
     ap.add_argument("--backend", type=str, default=None,
                     help="torch.distributed backend override (nccl/gloo)")
     ap.add_argument("--device-resident", action="store_true",
@@ -66,10 +68,15 @@ def main():
         print(f"[bench] mesh: {mesh.nelems} tets ({cells}^3 cells), built in {time.time()-t0:.1f}s",
               file=sys.stderr, flush=True)
 
-    eng = pt.TallyEngine(mesh, args.particles, device=device)
+    eng = pt.TallyEngine(mesh, args.particles, device=device,
+                         ngroups=args.ngroups)
     p0, p1, flying, weights = make_box_histories(
         (1.0, 1.0, 1.0), args.particles, args.mean_chord, cells,
         seed=args.seed + rank, pinned=eng.is_gpu, sort=not args.no_sort)
+    groups = None
+    if args.ngroups > 1:
+        rng_g = np.random.default_rng(1234 + rank)
+        groups = rng_g.integers(0, args.ngroups, args.particles).astype(np.uint16)
     eng.copy_initial_position(p0.reshape(-1))
     eng.synchronize()
 
@@ -95,7 +102,7 @@ def main():
         if args.continue_mode:
             eng.move_continue(d, flying, weights)
         else:
-            eng.move(o, d, flying, weights)
+            eng.move(o, d, flying, weights, groups=groups)
 
     def barrier_sync():
         eng.synchronize()
@@ -169,6 +176,7 @@ def main():
                 "mean_chord_elems": args.mean_chord,
                 "seq_len": None,
                 "parallelism": f"dp{world}-replicated-mesh",
+                "ngroups": args.ngroups,
                 "device": "gpu" if eng.is_gpu else "cpu",
                 "lost_particles": stats["lost_particles"],
                 "flux_sum": float(global_flux.sum()),
