@@ -59,11 +59,15 @@ void Mesh::finalize() {
         auto it = open.find(key);
         if (it == open.end()) {
           open.emplace(key, t * 4 + f);
+        } else if (it->second < 0) {
+          throw std::runtime_error(
+              "non-manifold mesh: face shared by more than two tets (tet " +
+              std::to_string(t) + ")");
         } else {
           const int64_t other = it->second;
           nbr[t * 4 + f] = (int32_t)(other / 4);
           nbr[other] = (int32_t)t;
-          open.erase(it);
+          it->second = -1; // paired; a third occurrence is non-manifold
         }
       }
     }
@@ -83,6 +87,9 @@ void Mesh::finalize() {
     const Vec3 tv[4] = {vert(tet2vert[t * 4]), vert(tet2vert[t * 4 + 1]),
                         vert(tet2vert[t * 4 + 2]), vert(tet2vert[t * 4 + 3])};
     volumes[t] = signed_volume(tv[0], tv[1], tv[2], tv[3]);
+    if (!(volumes[t] > 0.0))
+      throw std::runtime_error("degenerate element " + std::to_string(t) +
+                               " (zero or NaN volume)");
     for (int f = 0; f < 4; ++f) {
       int32_t gv[3] = {tet2vert[t * 4 + kFaceVerts[f][0]],
                        tet2vert[t * 4 + kFaceVerts[f][1]],

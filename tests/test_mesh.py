@@ -150,3 +150,19 @@ def test_vtk_binary_output(tmp_path):
     assert (x, y, z) == (0.0, 0.0, 0.0)
     # file is much smaller than an ASCII equivalent would be
     assert len(data) < 40_000_000
+
+
+def test_degenerate_element_rejected():
+    coords = np.array([[0, 0, 0], [1, 0, 0], [0, 1, 0], [0, 0, 1]], float)
+    tets = np.array([[0, 1, 2, 2]], np.int32)  # repeated vertex -> zero volume
+    with pytest.raises(RuntimeError):
+        pt.mesh_from_arrays(coords, tets)
+
+
+def test_nonmanifold_mesh_rejected():
+    # three tets sharing the (0,1,2) face
+    coords = np.array([[0, 0, 0], [1, 0, 0], [0, 1, 0],
+                       [0, 0, 1], [0, 0, -1], [1, 1, 1]], float)
+    tets = np.array([[0, 1, 2, 3], [0, 1, 2, 4], [0, 1, 2, 5]], np.int32)
+    with pytest.raises(RuntimeError):
+        pt.mesh_from_arrays(coords, tets)
