@@ -166,3 +166,18 @@ def test_nonmanifold_mesh_rejected():
     tets = np.array([[0, 1, 2, 3], [0, 1, 2, 4], [0, 1, 2, 5]], np.int32)
     with pytest.raises(RuntimeError):
         pt.mesh_from_arrays(coords, tets)
+
+
+def test_gmsh_binary_rejected(tmp_path):
+    msh = tmp_path / "bin.msh"
+    msh.write_text("$MeshFormat\n2.2 1 8\n$EndMeshFormat\n")
+    with pytest.raises(RuntimeError):
+        pt.read_gmsh(str(msh))
+
+
+def test_osh_foreign_magic_rejected(tmp_path):
+    d = tmp_path / "foreign.osh"
+    d.mkdir()
+    (d / "0.osh").write_bytes(b"\xa1\x1a" + b"\x00" * 64)  # Omega_h-style magic
+    with pytest.raises(RuntimeError, match="not a pumitally"):
+        pt.read_osh(str(d))
