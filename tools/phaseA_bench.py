@@ -32,7 +32,12 @@ def main():
             o[:] = ends[par]
             if k_res:
                 idx = rng.choice(n, k_res, replace=False)
-                fresh = rng.uniform(1e-4, 1 - 1e-4, size=(k_res, 3))
+                # resample NEAR the destination so walk lengths stay
+                # comparable to the base workload; the measured delta then
+                # isolates the relocation (device localization) cost.
+                d_near = ends[1 - par][idx]
+                fresh = np.clip(d_near + rng.normal(0, 0.03, size=(k_res, 3)),
+                                1e-4, 1 - 1e-4)
                 o[idx] = fresh
                 # relocated particles walk fresh->dest instead of pos->dest
                 d = ends[1 - par]
