@@ -177,13 +177,48 @@ public:
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status) override {
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    for (int64_t i = 0; i < n; ++i) {
+    const unsigned hw = std::thread::hardware_concurrency();
+    if (n >= 65536 && hw > 1) {
+      const int nthreads = (int)std::min<unsigned>(hw, 64);
+      std::vector<std::vector<double>> partial(
+          nthreads, std::vector<double>(flux_.size(), 0.0));
+      std::atomic<int64_t> lost{0};
+      std::vector<std::thread> workers;
+      const int64_t per = (n + nthreads - 1) / nthreads;
+      for (int t = 0; t < nthreads; ++t) {
+        workers.emplace_back([&, t] {
+          const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
+          int64_t my_lost = 0;
+          for (int64_t i = lo; i < hi; ++i)
+            walk_raw_one(pos, dest, elem, weights, out_pos, out_elem,
+                         out_status, i, steps, partial[t].data(), my_lost);
+          lost += my_lost;
+        });
+      }
+      for (auto &w : workers) w.join();
+      for (int t = 0; t < nthreads; ++t)
+        for (size_t e = 0; e < flux_.size(); ++e) flux_[e] += partial[t][e];
+      stats_.lost_particles += lost.load();
+      return;
+    }
+    int64_t lost = 0;
+    for (int64_t i = 0; i < n; ++i)
+      walk_raw_one(pos, dest, elem, weights, out_pos, out_elem, out_status, i,
+                   steps, flux_.data(), lost);
+    stats_.lost_particles += lost;
+  }
+
+  void walk_raw_one(const double *pos, const double *dest,
+                    const int32_t *elem, const double *weights,
+                    double *out_pos, int32_t *out_elem, int8_t *out_status,
+                    int64_t i, int steps, double *flux_out, int64_t &lost) {
+    {
       const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
       const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
       int32_t oe;
       Vec3 op;
       bool esc;
-      auto add = [&](int32_t e, double v) { flux_[e] += v; };
+      auto add = [&](int32_t e, double v) { flux_out[e] += v; };
       if (walk_fp32)
         walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                        mesh_.nbr.data(), elem[i], o, d, weights[i], steps, add,
@@ -195,7 +230,7 @@ public:
       if (oe == kWalkLost) {
         st = 3;
         oe = elem[i];
-        stats_.lost_particles++;
+        lost++;
       } else if (esc) {
         st = 1;
       } else if (oe < -1) {
