@@ -118,6 +118,8 @@ public:
   int max_steps = 0; // 0 = auto (set by implementations from mesh size)
   int ngroups = 1;   // energy groups (flux is [ngroups x nelems])
 
+  bool reflective = false; // specular-reflect at boundaries (else vacuum)
+
   // fp32-traversal fast path (walk.h walk_advance32): candidate exit-face
   // decisions in fp32, crossings/tallies in fp64.  Controlled by
   // PUMITALLY_WALK=fp32|fp64; both engines honor it so CPU remains the
@@ -134,6 +136,14 @@ inline bool default_walk_fp32() {
 // relative to the mesh bounding-box diagonal.  The reference hardcodes its
 // geometric tolerance (1e-8 at PumiTallyImpl.cpp:51); here it is runtime
 // configurable (SURVEY.md section 5 flags the hardcoding).
+// Boundary condition: vacuum (reference parity, default) or specular
+// reflective on every boundary face (symmetry-plane models).  Env:
+// PUMITALLY_BC=vacuum|reflective.
+inline bool default_reflective() {
+  const char *s = getenv("PUMITALLY_BC");
+  return s && std::string(s) == "reflective";
+}
+
 inline double loc_tol_rel() {
   const char *s = getenv("PUMITALLY_LOC_TOL");
   return s ? atof(s) : 1e-10;

@@ -40,6 +40,7 @@ public:
     }
     loc_tol_ = loc_tol_rel() * norm(mesh_.bbox_hi - mesh_.bbox_lo);
     walk_fp32 = default_walk_fp32();
+    reflective = default_reflective();
   }
 
   int64_t num_particles() const override { return n_; }
@@ -158,10 +159,11 @@ public:
     if (walk_fp32)
       walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                      mesh_.nbr.data(), elem_[i], o, d, weights[i], steps, add,
-                     &out_elem, &out_pos, &out_esc);
+                     &out_elem, &out_pos, &out_esc, reflective);
     else
       walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
-                   weights[i], steps, add, &out_elem, &out_pos, &out_esc);
+                   weights[i], steps, add, &out_elem, &out_pos, &out_esc,
+                   reflective);
     if (out_elem == kWalkLost) {
       lost++;
       out_elem = elem_[i];
@@ -222,10 +224,10 @@ public:
       if (walk_fp32)
         walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                        mesh_.nbr.data(), elem[i], o, d, weights[i], steps, add,
-                       &oe, &op, &esc);
+                       &oe, &op, &esc, reflective);
       else
         walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d,
-                     weights[i], steps, add, &oe, &op, &esc);
+                     weights[i], steps, add, &oe, &op, &esc, reflective);
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;

@@ -60,11 +60,18 @@ PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
 
 // Advance one element crossing.  Returns true when the walk finished and
 // the out_* values are valid.  FluxAdd: void(int32_t elem, double v).
+// Specular reflection of point p across plane pl (unit normal).
+PT_HD Vec3 reflect_point(const Plane &pl, Vec3 p) {
+  const double v = plane_eval(pl, p);
+  return {p.x - 2.0 * v * pl.nx, p.y - 2.0 * v * pl.ny, p.z - 2.0 * v * pl.nz};
+}
+
 template <class FluxAdd>
 PT_HD bool walk_advance(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, WalkState &s,
                         int max_steps, FluxAdd &&add, int32_t *out_elem,
-                        Vec3 *out_pos, bool *out_escaped) {
+                        Vec3 *out_pos, bool *out_escaped,
+                        bool reflective = false) {
   if (s.step++ >= max_steps) {
     // Did not converge (numerically stuck / absurd chord): drop here and
     // flag as lost (reference prints "Not all particles are found",
@@ -122,6 +129,24 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
 
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
+    if (reflective) {
+      // Specular reflection: restart the segment at the crossing with the
+      // remaining part mirrored across the boundary plane (an isometry, so
+      // total tallied length is conserved).  The incoming face plane sees
+      // the mirrored destination on its positive side, so it cannot be
+      // re-selected.
+      const Vec3 hit = s.o + t_clamped * (s.d - s.o);
+      const Vec3 d2 = reflect_point(pl[exit_face], s.d);
+      const double remaining = (1.0 - t_clamped) * s.seg_len;
+      s.o = hit;
+      s.d = d2;
+      s.of = Vec3f{(float)hit.x, (float)hit.y, (float)hit.z};
+      s.df = Vec3f{(float)d2.x, (float)d2.y, (float)d2.z};
+      s.seg_len = remaining > 0.0 ? remaining : 0.0;
+      s.t_cur = 0.0;
+      s.prev_elem = -1;
+      return false;
+    }
     // Vacuum boundary: clip the destination to the exit point; the
     // particle keeps its last element id (reference K6 semantics,
     // PumiTallyImpl.cpp:275-281 and the 1.0-not-1.1 test expectation).
@@ -161,7 +186,8 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
                           const Plane32 *__restrict__ planes32,
                           const int32_t *__restrict__ nbr, WalkState &s,
                           int max_steps, FluxAdd &&add, int32_t *out_elem,
-                          Vec3 *out_pos, bool *out_escaped) {
+                          Vec3 *out_pos, bool *out_escaped,
+                          bool reflective = false) {
   if (s.step++ >= max_steps) {
     *out_elem = kWalkLost;
     *out_pos = s.o + s.t_cur * (s.d - s.o);
@@ -211,6 +237,20 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
 
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
+    if (reflective) {
+      const Plane &pe64 = planes[(int64_t)s.elem * 4 + exit_face];
+      const Vec3 hit = s.o + t_clamped * (s.d - s.o);
+      const Vec3 d2 = reflect_point(pe64, s.d);
+      const double remaining = (1.0 - t_clamped) * s.seg_len;
+      s.o = hit;
+      s.d = d2;
+      s.of = Vec3f{(float)hit.x, (float)hit.y, (float)hit.z};
+      s.df = Vec3f{(float)d2.x, (float)d2.y, (float)d2.z};
+      s.seg_len = remaining > 0.0 ? remaining : 0.0;
+      s.t_cur = 0.0;
+      s.prev_elem = -1;
+      return false;
+    }
     *out_elem = s.elem;
     *out_pos = s.o + t_clamped * (s.d - s.o);
     *out_escaped = true;
@@ -234,11 +274,12 @@ template <class FluxAdd>
 PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
                         Vec3 d, double weight, int max_steps, FluxAdd &&add,
-                        int32_t *out_elem, Vec3 *out_pos, bool *out_escaped) {
+                        int32_t *out_elem, Vec3 *out_pos, bool *out_escaped,
+                        bool reflective = false) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance(planes, nbr, s, max_steps, add, out_elem, out_pos,
-                       out_escaped)) {
+                       out_escaped, reflective)) {
   }
 }
 
@@ -248,11 +289,11 @@ PT_HD void walk_segment32(const Plane *__restrict__ planes,
                           const int32_t *__restrict__ nbr, int32_t elem,
                           Vec3 o, Vec3 d, double weight, int max_steps,
                           FluxAdd &&add, int32_t *out_elem, Vec3 *out_pos,
-                          bool *out_escaped) {
+                          bool *out_escaped, bool reflective = false) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance32(planes, planes32, nbr, s, max_steps, add, out_elem,
-                         out_pos, out_escaped)) {
+                         out_pos, out_escaped, reflective)) {
   }
 }
 

@@ -178,7 +178,7 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        double *__restrict__ flux,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
-                       int64_t nelems, int slice_mask) {
+                       int64_t nelems, int slice_mask, bool reflective) {
   flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups;
   const unsigned bpx = gridDim.x / 8u;
   const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
@@ -215,10 +215,10 @@ __global__ void k_move(const Plane *__restrict__ planes,
     auto add = [&](int32_t el, double v) { atomicAdd(&flux[goff + el], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
-                     add, &out_elem, &out_pos, &out_esc);
+                     add, &out_elem, &out_pos, &out_esc, reflective);
     else
       walk_segment(planes, nbr, e, o, d, weights[c], max_steps, add,
-                   &out_elem, &out_pos, &out_esc);
+                   &out_elem, &out_pos, &out_esc, reflective);
     if (out_elem == kWalkLost) {
       atomicAdd(lost, 1ull);
       out_elem = e;
@@ -244,7 +244,7 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            int8_t *__restrict__ out_status,
                            double *__restrict__ flux,
                            unsigned long long *__restrict__ lost, int64_t n,
-                           int max_steps) {
+                           int max_steps, bool reflective) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -255,10 +255,10 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     auto add = [&](int32_t e, double v) { atomicAdd(&flux[e], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
-                     max_steps, add, &oe, &op, &esc);
+                     max_steps, add, &oe, &op, &esc, reflective);
     else
       walk_segment(planes, nbr, elem[i], o, d, weights[i], max_steps, add,
-                   &oe, &op, &esc);
+                   &oe, &op, &esc, reflective);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -410,6 +410,7 @@ public:
 
     loc_tol_ = loc_tol_rel() * norm(mesh_.bbox_hi - mesh_.bbox_lo);
     walk_fp32 = default_walk_fp32();
+    reflective = default_reflective();
     const Vec3 c0 = mesh_.nelems > 0 ? mesh_.centroid(0) : Vec3{0, 0, 0};
     k_init_particles<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
         d_pos_, d_elem_, d_escaped_, d_s2c_, n_, c0.x, c0.y, c0.z);
@@ -523,11 +524,11 @@ public:
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps);
+          d_flux_, d_lost_, n, steps, reflective);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps);
+          d_flux_, d_lost_, n, steps, reflective);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
@@ -638,13 +639,13 @@ private:
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1);
+            slices_ - 1, reflective);
       else
         k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1);
+            slices_ - 1, reflective);
       PT_HIP_CHECK(hipGetLastError());
     }
   }
