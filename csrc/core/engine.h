@@ -103,6 +103,16 @@ public:
 
   virtual const EngineStats &stats() const = 0;
 
+  // Batch statistics (standard MC uncertainty accounting; the reference
+  // has a single accumulating tally with no variance).  end_batch()
+  // accumulates the current flux into running sum / sum-of-squares and
+  // zeroes the per-batch tally; batch_sum/batch_sum_sq read the
+  // accumulators (size nelems*ngroups); num_batches counts end_batch calls.
+  virtual void end_batch() = 0;
+  virtual std::vector<double> batch_sum() const = 0;
+  virtual std::vector<double> batch_sum_sq() const = 0;
+  virtual int64_t num_batches() const = 0;
+
   // Overwrite the flux tally (used by the distributed driver to install the
   // all-reduced global tally on rank 0 before writing).
   virtual void set_flux(const double *flux, int64_t nelems) = 0;

@@ -246,6 +246,26 @@ public:
     }
   }
 
+  void end_batch() override {
+    if (batch_sum_.empty()) {
+      batch_sum_.assign(flux_.size(), 0.0);
+      batch_sq_.assign(flux_.size(), 0.0);
+    }
+    for (size_t e = 0; e < flux_.size(); ++e) {
+      batch_sum_[e] += flux_[e];
+      batch_sq_[e] += flux_[e] * flux_[e];
+      flux_[e] = 0.0;
+    }
+    nbatches_++;
+  }
+  std::vector<double> batch_sum() const override {
+    return batch_sum_.empty() ? std::vector<double>(flux_.size(), 0.0) : batch_sum_;
+  }
+  std::vector<double> batch_sum_sq() const override {
+    return batch_sq_.empty() ? std::vector<double>(flux_.size(), 0.0) : batch_sq_;
+  }
+  int64_t num_batches() const override { return nbatches_; }
+
   std::vector<double> flux() const override { return flux_; }
   std::vector<int32_t> elem_ids() const override { return elem_; }
   std::vector<double> positions() const override { return pos_; }
@@ -273,7 +293,8 @@ private:
   Mesh mesh_;
   int64_t n_;
   double loc_tol_;
-  std::vector<double> flux_, pos_;
+  std::vector<double> flux_, pos_, batch_sum_, batch_sq_;
+  int64_t nbatches_ = 0;
   std::vector<int32_t> elem_;
   std::vector<uint8_t> escaped_;
   EngineStats stats_;

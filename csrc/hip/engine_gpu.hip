@@ -141,6 +141,18 @@ __global__ void k_permute_state(const int32_t *__restrict__ order,
   }
 }
 
+__global__ void k_accumulate_batch(double *__restrict__ flux,
+                                   double *__restrict__ bsum,
+                                   double *__restrict__ bsq, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const double f = flux[i];
+    bsum[i] += f;
+    bsq[i] += f * f;
+    flux[i] = 0.0;
+  }
+}
+
 __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
                                 int slices) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -430,6 +442,7 @@ public:
                     (void *)d_flying_[0], (void *)d_flying_[1],
                     (void *)d_weights_[0], (void *)d_weights_[1],
                     (void *)d_groups_[0], (void *)d_groups_[1],
+                    (void *)d_bsum_, (void *)d_bsq_,
                     (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
                     (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
@@ -537,6 +550,42 @@ public:
     for (void *q : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
                     (void *)doe, (void *)dop, (void *)dst_})
       (void)hipFree(q);
+  }
+
+  void end_batch() override {
+    PT_HIP_CHECK(hipSetDevice(device_));
+    sync();
+    const int64_t fsz = mesh_.nelems * ngroups;
+    if (!d_bsum_) {
+      d_bsum_ = dmalloc<double>(fsz);
+      d_bsq_ = dmalloc<double>(fsz);
+      PT_HIP_CHECK(hipMemset(d_bsum_, 0, fsz * sizeof(double)));
+      PT_HIP_CHECK(hipMemset(d_bsq_, 0, fsz * sizeof(double)));
+    }
+    if (slices_ > 1) {
+      k_reduce_slices<<<grid_blocks(fsz), kBlock, 0, s_comp_>>>(d_flux_, fsz,
+                                                               slices_);
+      PT_HIP_CHECK(hipGetLastError());
+    }
+    k_accumulate_batch<<<grid_blocks(fsz), kBlock, 0, s_comp_>>>(
+        d_flux_, d_bsum_, d_bsq_, fsz);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+    nbatches_++;
+  }
+  std::vector<double> batch_sum() const override { return fetch_acc(d_bsum_); }
+  std::vector<double> batch_sum_sq() const override { return fetch_acc(d_bsq_); }
+  int64_t num_batches() const override { return nbatches_; }
+
+  std::vector<double> fetch_acc(const double *p) const {
+    const int64_t fsz = mesh_.nelems * ngroups;
+    std::vector<double> out(fsz, 0.0);
+    if (p) {
+      sync();
+      PT_HIP_CHECK(hipMemcpy(out.data(), p, fsz * sizeof(double),
+                             hipMemcpyDeviceToHost));
+    }
+    return out;
   }
 
   std::vector<double> flux() const override {
@@ -720,6 +769,8 @@ private:
   unsigned long long *d_lost_ = nullptr;
   double *d_origin_[2] = {nullptr, nullptr};
   uint16_t *d_groups_[2] = {nullptr, nullptr};
+  double *d_bsum_ = nullptr, *d_bsq_ = nullptr;
+  int64_t nbatches_ = 0;
   double *d_dest_[2] = {nullptr, nullptr};
   int8_t *d_flying_[2] = {nullptr, nullptr};
   double *d_weights_[2] = {nullptr, nullptr};

@@ -340,6 +340,11 @@ PYBIND11_MODULE(_core, m) {
                throw std::runtime_error("set_particle_state: size mismatch");
              e.eng->set_particle_state(pos.data(), elem.data(), escaped.data(), n);
            })
+      .def("end_batch", [](PyEngine &e) { py::gil_scoped_release ng; e.eng->end_batch(); })
+      .def("batch_sum", [](const PyEngine &e) { return vec_to_np(e.eng->batch_sum()); })
+      .def("batch_sum_sq", [](const PyEngine &e) { return vec_to_np(e.eng->batch_sum_sq()); })
+      .def_property_readonly("num_batches",
+                             [](const PyEngine &e) { return e.eng->num_batches(); })
       .def("stats", [](const PyEngine &e) {
         const EngineStats &s = e.eng->stats();
         py::dict d;

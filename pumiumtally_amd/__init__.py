@@ -179,6 +179,33 @@ class TallyEngine:
     def stats(self):
         return self._eng.stats()
 
+    def end_batch(self):
+        """Close the current batch: accumulate the per-batch tally into
+        running sum / sum-of-squares and zero it (standard MC batch
+        statistics; the reference has no variance accounting)."""
+        self._eng.end_batch()
+
+    def batch_statistics(self):
+        """Returns (mean, rel_std_error) over the closed batches, shaped
+        like flux().  rel_std_error = sigma_of_mean / |mean| (0 where
+        mean == 0)."""
+        import numpy as np
+
+        nb = self._eng.num_batches
+        if nb < 1:
+            raise RuntimeError("no batches closed yet (call end_batch)")
+        s1 = self._eng.batch_sum()
+        s2 = self._eng.batch_sum_sq()
+        mean = s1 / nb
+        var = np.maximum(s2 / nb - mean * mean, 0.0)
+        sem = np.sqrt(var / max(nb - 1, 1))
+        rel = np.divide(sem, np.abs(mean), out=np.zeros_like(sem),
+                        where=mean != 0)
+        if self.ngroups > 1:
+            shape = (self.ngroups, self.mesh.nelems)
+            return mean.reshape(shape), rel.reshape(shape)
+        return mean, rel
+
     def save_checkpoint(self, path: str):
         """Persist the full tally state (flux accumulator + particle
         positions/elements/escaped flags) so a crashed batch can resume.
