@@ -156,14 +156,16 @@ public:
     const int64_t goff =
         groups ? (int64_t)(groups[i] % ngroups) * mesh_.nelems : 0;
     auto add = [&](int32_t e, double v) { flux_out[goff + e] += v; };
+    const uint32_t *bc =
+        mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
     if (walk_fp32)
       walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                      mesh_.nbr.data(), elem_[i], o, d, weights[i], steps, add,
-                     &out_elem, &out_pos, &out_esc, reflective);
+                     &out_elem, &out_pos, &out_esc, reflective, bc);
     else
       walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
                    weights[i], steps, add, &out_elem, &out_pos, &out_esc,
-                   reflective);
+                   reflective, bc);
     if (out_elem == kWalkLost) {
       lost++;
       out_elem = elem_[i];
@@ -221,13 +223,15 @@ public:
       Vec3 op;
       bool esc;
       auto add = [&](int32_t e, double v) { flux_out[e] += v; };
+      const uint32_t *bc =
+          mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
       if (walk_fp32)
         walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
                        mesh_.nbr.data(), elem[i], o, d, weights[i], steps, add,
-                       &oe, &op, &esc, reflective);
+                       &oe, &op, &esc, reflective, bc);
       else
         walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d,
-                     weights[i], steps, add, &oe, &op, &esc, reflective);
+                     weights[i], steps, add, &oe, &op, &esc, reflective, bc);
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;

@@ -7,6 +7,12 @@
 
 namespace pumitally {
 
+void Mesh::set_face_reflective(int64_t face_index) {
+  const size_t need = (size_t)((nelems * 4 + 31) / 32);
+  if (face_bc_bits.size() < need) face_bc_bits.resize(need, 0);
+  face_bc_bits[face_index >> 5] |= 1u << (face_index & 31);
+}
+
 Vec3 Mesh::centroid(int32_t t) const {
   Vec3 c{0, 0, 0};
   for (int k = 0; k < 4; ++k) c = c + vert(tet2vert[t * 4 + k]);

@@ -48,6 +48,19 @@ struct Mesh {
   std::vector<double> coords;   // nverts*3, xyz interleaved
   std::vector<int32_t> tet2vert; // nelems*4, positively oriented
 
+  // Optional per-face boundary conditions: bit set => that boundary face
+  // reflects (specular); unset/absent => vacuum.  Indexed by elem*4+f;
+  // only meaningful where nbr==-1.  Set via set_face_reflective BEFORE
+  // engine construction (engines snapshot it).  The global
+  // PUMITALLY_BC=reflective mode overrides everything to reflective.
+  std::vector<uint32_t> face_bc_bits;
+  void set_face_reflective(int64_t face_index);
+  bool face_is_reflective(int64_t face_index) const {
+    const size_t w = (size_t)(face_index >> 5);
+    return w < face_bc_bits.size() &&
+           ((face_bc_bits[w] >> (face_index & 31)) & 1u);
+  }
+
   // Derived (built by finalize()):
   std::vector<int32_t> nbr;     // nelems*4: neighbor tet across face f, -1 = boundary
   std::vector<Plane> planes;    // nelems*4: inward-positive unit-normal face planes

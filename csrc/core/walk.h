@@ -71,7 +71,8 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, WalkState &s,
                         int max_steps, FluxAdd &&add, int32_t *out_elem,
                         Vec3 *out_pos, bool *out_escaped,
-                        bool reflective = false) {
+                        bool reflective = false,
+                        const uint32_t *__restrict__ face_bc = nullptr) {
   if (s.step++ >= max_steps) {
     // Did not converge (numerically stuck / absurd chord): drop here and
     // flag as lost (reference prints "Not all particles are found",
@@ -129,7 +130,11 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
 
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
-    if (reflective) {
+    const int64_t fidx = (int64_t)s.elem * 4 + exit_face;
+    const bool refl_here =
+        reflective ||
+        (face_bc && ((face_bc[fidx >> 5] >> (fidx & 31)) & 1u));
+    if (refl_here) {
       // Specular reflection: restart the segment at the crossing with the
       // remaining part mirrored across the boundary plane (an isometry, so
       // total tallied length is conserved).  The incoming face plane sees
@@ -187,7 +192,8 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
                           const int32_t *__restrict__ nbr, WalkState &s,
                           int max_steps, FluxAdd &&add, int32_t *out_elem,
                           Vec3 *out_pos, bool *out_escaped,
-                          bool reflective = false) {
+                          bool reflective = false,
+                          const uint32_t *__restrict__ face_bc = nullptr) {
   if (s.step++ >= max_steps) {
     *out_elem = kWalkLost;
     *out_pos = s.o + s.t_cur * (s.d - s.o);
@@ -237,7 +243,11 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
 
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
-    if (reflective) {
+    const int64_t fidx = (int64_t)s.elem * 4 + exit_face;
+    const bool refl_here =
+        reflective ||
+        (face_bc && ((face_bc[fidx >> 5] >> (fidx & 31)) & 1u));
+    if (refl_here) {
       const Plane &pe64 = planes[(int64_t)s.elem * 4 + exit_face];
       const Vec3 hit = s.o + t_clamped * (s.d - s.o);
       const Vec3 d2 = reflect_point(pe64, s.d);
@@ -275,11 +285,12 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
                         Vec3 d, double weight, int max_steps, FluxAdd &&add,
                         int32_t *out_elem, Vec3 *out_pos, bool *out_escaped,
-                        bool reflective = false) {
+                        bool reflective = false,
+                        const uint32_t *__restrict__ face_bc = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance(planes, nbr, s, max_steps, add, out_elem, out_pos,
-                       out_escaped, reflective)) {
+                       out_escaped, reflective, face_bc)) {
   }
 }
 
@@ -289,11 +300,12 @@ PT_HD void walk_segment32(const Plane *__restrict__ planes,
                           const int32_t *__restrict__ nbr, int32_t elem,
                           Vec3 o, Vec3 d, double weight, int max_steps,
                           FluxAdd &&add, int32_t *out_elem, Vec3 *out_pos,
-                          bool *out_escaped, bool reflective = false) {
+                          bool *out_escaped, bool reflective = false,
+                          const uint32_t *__restrict__ face_bc = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance32(planes, planes32, nbr, s, max_steps, add, out_elem,
-                         out_pos, out_escaped, reflective)) {
+                         out_pos, out_escaped, reflective, face_bc)) {
   }
 }
 

@@ -190,7 +190,8 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        double *__restrict__ flux,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
-                       int64_t nelems, int slice_mask, bool reflective) {
+                       int64_t nelems, int slice_mask, bool reflective,
+                       const uint32_t *__restrict__ face_bc) {
   flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups;
   const unsigned bpx = gridDim.x / 8u;
   const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
@@ -227,10 +228,10 @@ __global__ void k_move(const Plane *__restrict__ planes,
     auto add = [&](int32_t el, double v) { atomicAdd(&flux[goff + el], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
-                     add, &out_elem, &out_pos, &out_esc, reflective);
+                     add, &out_elem, &out_pos, &out_esc, reflective, face_bc);
     else
       walk_segment(planes, nbr, e, o, d, weights[c], max_steps, add,
-                   &out_elem, &out_pos, &out_esc, reflective);
+                   &out_elem, &out_pos, &out_esc, reflective, face_bc);
     if (out_elem == kWalkLost) {
       atomicAdd(lost, 1ull);
       out_elem = e;
@@ -256,7 +257,8 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            int8_t *__restrict__ out_status,
                            double *__restrict__ flux,
                            unsigned long long *__restrict__ lost, int64_t n,
-                           int max_steps, bool reflective) {
+                           int max_steps, bool reflective,
+                           const uint32_t *__restrict__ face_bc) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -267,10 +269,10 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     auto add = [&](int32_t e, double v) { atomicAdd(&flux[e], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
-                     max_steps, add, &oe, &op, &esc, reflective);
+                     max_steps, add, &oe, &op, &esc, reflective, face_bc);
     else
       walk_segment(planes, nbr, elem[i], o, d, weights[i], max_steps, add,
-                   &oe, &op, &esc, reflective);
+                   &oe, &op, &esc, reflective, face_bc);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -372,6 +374,12 @@ public:
     PT_HIP_CHECK(hipMemcpy(d_cell_tets_, mesh_.grid.cell_tets.data(),
                            mesh_.grid.cell_tets.size() * sizeof(int32_t),
                            hipMemcpyHostToDevice));
+    if (!mesh_.face_bc_bits.empty()) {
+      d_face_bc_ = dmalloc<uint32_t>(mesh_.face_bc_bits.size());
+      PT_HIP_CHECK(hipMemcpy(d_face_bc_, mesh_.face_bc_bits.data(),
+                             mesh_.face_bc_bits.size() * sizeof(uint32_t),
+                             hipMemcpyHostToDevice));
+    }
     grid_view_ = GridView{mesh_.grid.nx, mesh_.grid.ny, mesh_.grid.nz,
                           mesh_.grid.lo,  mesh_.grid.inv_h,
                           d_cell_start_,  d_cell_tets_};
@@ -442,7 +450,7 @@ public:
                     (void *)d_flying_[0], (void *)d_flying_[1],
                     (void *)d_weights_[0], (void *)d_weights_[1],
                     (void *)d_groups_[0], (void *)d_groups_[1],
-                    (void *)d_bsum_, (void *)d_bsq_,
+                    (void *)d_bsum_, (void *)d_bsq_, (void *)d_face_bc_,
                     (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
                     (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
@@ -537,11 +545,11 @@ public:
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective);
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective);
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
@@ -688,13 +696,13 @@ private:
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1, reflective);
+            slices_ - 1, reflective, d_face_bc_);
       else
         k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1, reflective);
+            slices_ - 1, reflective, d_face_bc_);
       PT_HIP_CHECK(hipGetLastError());
     }
   }
@@ -770,6 +778,7 @@ private:
   double *d_origin_[2] = {nullptr, nullptr};
   uint16_t *d_groups_[2] = {nullptr, nullptr};
   double *d_bsum_ = nullptr, *d_bsq_ = nullptr;
+  uint32_t *d_face_bc_ = nullptr;
   int64_t nbatches_ = 0;
   double *d_dest_[2] = {nullptr, nullptr};
   int8_t *d_flying_[2] = {nullptr, nullptr};

@@ -151,7 +151,42 @@ PYBIND11_MODULE(_core, m) {
              write_vtk(path, m_, cd);
            })
       .def("write_osh",
-           [](const Mesh &m_, const std::string &dir) { write_osh(dir, m_); });
+           [](const Mesh &m_, const std::string &dir) { write_osh(dir, m_); })
+      .def("boundary_faces",
+           // (face_indices=elem*4+f, centroids (k,3), outward normals (k,3))
+           // of every boundary face, for marking per-face BCs.
+           [](const Mesh &m_) {
+             std::vector<int64_t> ids;
+             for (int64_t t = 0; t < m_.nelems; ++t)
+               for (int f = 0; f < 4; ++f)
+                 if (m_.nbr[t * 4 + f] == -1) ids.push_back(t * 4 + f);
+             const int64_t k = (int64_t)ids.size();
+             auto fid = py::array_t<int64_t>(k);
+             auto cen = py::array_t<double>({k, (int64_t)3});
+             auto nor = py::array_t<double>({k, (int64_t)3});
+             for (int64_t i = 0; i < k; ++i) {
+               const int64_t t = ids[i] / 4;
+               const int f = (int)(ids[i] % 4);
+               fid.mutable_data()[i] = ids[i];
+               Vec3 c{0, 0, 0};
+               for (int j = 0; j < 3; ++j)
+                 c = c + m_.vert(m_.tet2vert[t * 4 + kFaceVerts[f][j]]);
+               c = (1.0 / 3.0) * c;
+               const Plane &pl = m_.planes[ids[i]]; // inward-positive
+               cen.mutable_data()[i * 3] = c.x;
+               cen.mutable_data()[i * 3 + 1] = c.y;
+               cen.mutable_data()[i * 3 + 2] = c.z;
+               nor.mutable_data()[i * 3] = -pl.nx;
+               nor.mutable_data()[i * 3 + 1] = -pl.ny;
+               nor.mutable_data()[i * 3 + 2] = -pl.nz;
+             }
+             return py::make_tuple(fid, cen, nor);
+           })
+      .def("set_reflective_faces",
+           [](Mesh &m_, py::array_t<int64_t, py::array::c_style | py::array::forcecast> fids) {
+             for (py::ssize_t i = 0; i < fids.size(); ++i)
+               m_.set_face_reflective(fids.data()[i]);
+           });
 
   py::class_<SubMesh>(m, "SubMesh")
       .def_property_readonly("local", [](const SubMesh &s) -> const Mesh & { return s.local; },

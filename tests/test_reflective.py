@@ -12,6 +12,8 @@ import sys
 import numpy as np
 import pytest
 
+import pumiumtally_amd as pt
+
 CODE = r"""
 import numpy as np
 import pumiumtally_amd as pt
@@ -64,3 +66,59 @@ def test_reflective_cpu():
 @pytest.mark.gpu
 def test_reflective_gpu():
     run_case("cuda")
+
+
+def test_per_face_reflective_cpu():
+    """Quarter-model style: only the +x boundary face reflects; the other
+    five faces stay vacuum."""
+    m = pt.build_box(3, 3, 3)
+    fid, cen, nor = m.boundary_faces()
+    plus_x = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    assert len(plus_x) > 0
+    m.set_reflective_faces(plus_x)
+
+    n = 8
+    o = np.tile([0.5, 0.4, 0.45], (n, 1))
+    e = pt.TallyEngine(m, n, device="cpu")
+
+    # exits +x: reflected back inside, full length tallied, no escape
+    d = o.copy(); d[:, 0] = 1.4
+    e.copy_initial_position(o.ravel())
+    e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), np.ones(n))
+    assert (e.escaped() == 0).all()
+    assert abs(e.flux().sum() - 0.9 * n) < 1e-10
+    p = e.positions()
+    assert np.allclose(p[:, 0], 0.6, atol=1e-9)  # folded: 2.0 - 1.4
+
+    # exits -x: vacuum clip as usual
+    e2 = pt.TallyEngine(m, n, device="cpu")
+    d2 = o.copy(); d2[:, 0] = -0.4
+    e2.copy_initial_position(o.ravel())
+    e2.move(o.ravel(), d2.ravel(), np.ones(n, np.int8), np.ones(n))
+    assert (e2.escaped() == 1).all()
+    assert abs(e2.flux().sum() - 0.5 * n) < 1e-10
+    assert np.allclose(e2.positions()[:, 0], 0.0, atol=1e-9)
+
+
+@pytest.mark.gpu
+def test_per_face_reflective_gpu():
+    m = pt.build_box(3, 3, 3)
+    fid, cen, nor = m.boundary_faces()
+    m.set_reflective_faces(fid[np.abs(cen[:, 0] - 1.0) < 1e-12])
+    n = 2000
+    rng = np.random.default_rng(3)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = o + rng.normal(0, 0.4, size=(n, 3))
+    w = rng.uniform(0.2, 1.0, n)
+    res = {}
+    for dev in ("cpu", "cuda"):
+        e = pt.TallyEngine(m, n, device=dev)
+        e.copy_initial_position(o.ravel())
+        e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+        e.synchronize()
+        res[dev] = (e.flux(), e.elem_ids(), e.positions(), e.escaped())
+    assert np.array_equal(res["cpu"][1], res["cuda"][1])
+    assert np.array_equal(res["cpu"][3], res["cuda"][3])
+    assert np.allclose(res["cpu"][2], res["cuda"][2], atol=0, rtol=0)
+    assert np.abs(res["cpu"][0] - res["cuda"][0]).max() < 1e-10 * max(
+        1.0, np.abs(res["cpu"][0]).max())
