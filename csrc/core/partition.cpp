@@ -111,7 +111,12 @@ SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
     for (int f = 0; f < 4; ++f) {
       if (local.nbr[t * 4 + f] != -1) continue;
       const int32_t gn = m.nbr[g * 4 + f];
-      if (gn == -1) continue; // true boundary
+      if (gn == -1) {
+        // true boundary: carry the per-face boundary condition over
+        if (m.face_is_reflective(g * 4 + f))
+          local.set_face_reflective(t * 4 + f);
+        continue;
+      }
       const int32_t k = (int32_t)sub.foreign_gid.size();
       sub.foreign_gid.push_back(gn);
       sub.foreign_owner.push_back(owners[gn]);

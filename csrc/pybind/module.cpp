@@ -182,6 +182,8 @@ PYBIND11_MODULE(_core, m) {
              }
              return py::make_tuple(fid, cen, nor);
            })
+      .def("face_is_reflective",
+           [](const Mesh &m_, int64_t fid) { return m_.face_is_reflective(fid); })
       .def("set_reflective_faces",
            [](Mesh &m_, py::array_t<int64_t, py::array::c_style | py::array::forcecast> fids) {
              for (py::ssize_t i = 0; i < fids.size(); ++i)

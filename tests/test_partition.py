@@ -202,3 +202,24 @@ def test_gloo_world2_partitioned_gpu(tmp_path):
     for r, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0, f"rank {r} failed:\n{out}"
     assert "PART_OK" in outs[0]
+
+
+def test_submesh_carries_face_bc():
+    """Per-face reflective bits survive submesh extraction."""
+    m = pt.build_box(3, 3, 3)
+    fid, cen, nor = m.boundary_faces()
+    plus_x = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    m.set_reflective_faces(plus_x)
+    owners = _core.partition_morton(m, 2)
+    marked = 0
+    for p in range(2):
+        sub = _core.extract_submesh(m, owners, p, ghost_rings=1)
+        lfid, lcen, lnor = sub.local.boundary_faces()
+        for i in range(len(lfid)):
+            if abs(lcen[i, 0] - 1.0) < 1e-12:
+                assert sub.local.face_is_reflective(int(lfid[i]))
+                marked += 1
+            else:
+                assert not sub.local.face_is_reflective(int(lfid[i]))
+    # ghost rings can replicate a boundary face in both submeshes
+    assert marked >= len(plus_x)
