@@ -17,6 +17,10 @@ int default_max_steps(const Mesh &m) {
 }
 
 std::vector<double> normalize_flux(const Mesh &m, const std::vector<double> &flux) {
+  if ((int64_t)flux.size() != m.nelems)
+    throw std::runtime_error(
+        "normalize_flux: flux size != nelems (for grouped tallies normalize "
+        "each group slice)");
   std::vector<double> out(m.nelems);
   for (int64_t e = 0; e < m.nelems; ++e) out[e] = flux[e] / m.volumes[e];
   return out;

@@ -237,7 +237,14 @@ class TallyEngine:
             np.ascontiguousarray(d["escaped"], dtype=np.uint8))
 
     def normalized_flux(self):
-        return _core.normalize_flux(self.mesh, self._eng.flux())
+        """flux / element volume; shape matches flux() (per group when
+        ngroups > 1)."""
+        f = self.flux()
+        if self.ngroups > 1:
+            import numpy as np
+            return np.stack([_core.normalize_flux(self.mesh, f[g])
+                             for g in range(self.ngroups)])
+        return _core.normalize_flux(self.mesh, f)
 
     def write_tally_results(self, filename: str = "fluxresult.vtk"):
         if self.ngroups > 1:
