@@ -101,11 +101,13 @@ class DistributedTally:
         return t.cpu().numpy()
 
     def write_tally_results(self, filename: str = "fluxresult.vtk"):
-        """All-reduce, then rank 0 writes the normalized global tally."""
-        from .. import write_tally_vtk
-
+        """All-reduce, then rank 0 writes the normalized global tally
+        (grouped tallies get per-group fields, via the engine's writer)."""
         global_flux = self.allreduce_flux()
         if self.rank == 0:
-            write_tally_vtk(filename, self.mesh, global_flux)
+            local = self.engine.flux()
+            self.engine.set_flux(global_flux)
+            self.engine.write_tally_results(filename)
+            self.engine.set_flux(local)  # restore this rank's partial tally
         self.barrier()
         return global_flux
