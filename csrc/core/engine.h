@@ -34,6 +34,7 @@
 
 #include <cstdint>
 #include <cstdlib>
+#include <stdexcept>
 #include <memory>
 #include <string>
 #include <vector>
