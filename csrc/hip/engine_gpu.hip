@@ -52,7 +52,18 @@ namespace pumitally {
 
 namespace {
 
-constexpr int kBlock = 256;
+// Block size for all particle kernels; swept via PUMITALLY_BLOCK.
+inline int block_size() {
+  static int v = [] {
+    const char *s = getenv("PUMITALLY_BLOCK");
+    int k = s ? atoi(s) : 256;
+    if (k < 64) k = 64;
+    if (k > 1024) k = 1024;
+    return (k / 64) * 64; // multiple of the 64-wide wavefront
+  }();
+  return v;
+}
+#define kBlock block_size()
 
 __global__ void k_init_particles(double *__restrict__ pos,
                                  int32_t *__restrict__ elem,
