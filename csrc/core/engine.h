@@ -90,11 +90,14 @@ public:
   // segments (pos->dest) starting in the given elements, tallying into
   // this engine's flux.  status: 0=reached dest, 1=escaped (vacuum),
   // 2=handoff (out_elem = encoded foreign ref -(2+k)), 3=lost.
-  // Synchronous; arrays are plain host memory.
+  // groups (nullable): per-segment energy-group index, same semantics as
+  // move(); a handed-off particle keeps its group (the partitioned driver
+  // carries it in the exchange record).  Synchronous; host memory.
   virtual void walk_raw(int64_t n, const double *pos, const double *dest,
                         const int32_t *elem, const double *weights,
                         double *out_pos, int32_t *out_elem,
-                        int8_t *out_status) = 0;
+                        int8_t *out_status,
+                        const uint16_t *groups = nullptr) = 0;
 
   // Read back state (host copies).
   virtual std::vector<double> flux() const = 0;           // nelems*ngroups, raw tally

@@ -183,7 +183,8 @@ public:
 
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
-                int32_t *out_elem, int8_t *out_status) override {
+                int32_t *out_elem, int8_t *out_status,
+                const uint16_t *groups = nullptr) override {
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     const unsigned hw = std::thread::hardware_concurrency();
     if (n >= 65536 && hw > 1) {
@@ -198,7 +199,7 @@ public:
           const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
           int64_t my_lost = 0;
           for (int64_t i = lo; i < hi; ++i)
-            walk_raw_one(pos, dest, elem, weights, out_pos, out_elem,
+            walk_raw_one(pos, dest, elem, weights, groups, out_pos, out_elem,
                          out_status, i, steps, partial[t].data(), my_lost);
           lost += my_lost;
         });
@@ -211,14 +212,15 @@ public:
     }
     int64_t lost = 0;
     for (int64_t i = 0; i < n; ++i)
-      walk_raw_one(pos, dest, elem, weights, out_pos, out_elem, out_status, i,
-                   steps, flux_.data(), lost);
+      walk_raw_one(pos, dest, elem, weights, groups, out_pos, out_elem,
+                   out_status, i, steps, flux_.data(), lost);
     stats_.lost_particles += lost;
   }
 
   void walk_raw_one(const double *pos, const double *dest,
                     const int32_t *elem, const double *weights,
-                    double *out_pos, int32_t *out_elem, int8_t *out_status,
+                    const uint16_t *groups, double *out_pos,
+                    int32_t *out_elem, int8_t *out_status,
                     int64_t i, int steps, double *flux_out, int64_t &lost) {
     {
       const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -226,7 +228,9 @@ public:
       int32_t oe;
       Vec3 op;
       bool esc;
-      auto add = [&](int32_t e, double v) { flux_out[e] += v; };
+      const int64_t goff =
+          groups ? (int64_t)(groups[i] % ngroups) * mesh_.nelems : 0;
+      auto add = [&](int32_t e, double v) { flux_out[goff + e] += v; };
       const uint32_t *bc =
           mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
       if (walk_fp32)

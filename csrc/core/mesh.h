@@ -107,7 +107,12 @@ struct SubMesh {
 
 // Balanced spatial partition: elements sorted by Morton key of centroid,
 // split into nparts equal chunks.  Returns per-element owner ids.
-std::vector<int32_t> partition_morton(const Mesh &m, int nparts);
+// weights (nullable, nelems doubles): per-element work estimates; when
+// given, chunks equalize summed weight instead of element count (dynamic
+// load balance for partitioned tallies -- feed the previous batch's raw
+// flux back in and re-extract submeshes).
+std::vector<int32_t> partition_morton(const Mesh &m, int nparts,
+                                      const double *weights = nullptr);
 
 // Extract part `part`'s submesh.  Local faces whose global neighbor is
 // not in the submesh get nbr = -(2+k) with foreign_gid[k]/foreign_owner[k]

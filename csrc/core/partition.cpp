@@ -23,7 +23,8 @@ uint64_t morton3(uint32_t x, uint32_t y, uint32_t z) {
 }
 } // namespace
 
-std::vector<int32_t> partition_morton(const Mesh &m, int nparts) {
+std::vector<int32_t> partition_morton(const Mesh &m, int nparts,
+                                      const double *weights) {
   if (nparts < 1) throw std::runtime_error("nparts must be >= 1");
   std::vector<int64_t> order(m.nelems);
   std::iota(order.begin(), order.end(), 0);
@@ -43,8 +44,36 @@ std::vector<int32_t> partition_morton(const Mesh &m, int nparts) {
               [&](int64_t a, int64_t b) { return key[a] < key[b]; });
   }
   std::vector<int32_t> owners(m.nelems);
-  for (int64_t i = 0; i < m.nelems; ++i)
-    owners[order[i]] = (int32_t)((i * nparts) / m.nelems);
+  if (!weights) {
+    for (int64_t i = 0; i < m.nelems; ++i)
+      owners[order[i]] = (int32_t)((i * nparts) / m.nelems);
+    return owners;
+  }
+  // Work-weighted split: equal prefix-sum chunks along the Morton curve.
+  // Weights are per-element work estimates (e.g. the previous batch's raw
+  // flux + epsilon); zero/negative entries get a small floor so every
+  // element stays assignable and parts stay contiguous on the curve.
+  double total = 0.0;
+  double wmax = 0.0;
+  for (int64_t t = 0; t < m.nelems; ++t)
+    wmax = std::max(wmax, weights[t] > 0 ? weights[t] : 0.0);
+  const double floor_w = wmax > 0 ? wmax * 1e-6 : 1.0;
+  std::vector<double> w(m.nelems);
+  for (int64_t t = 0; t < m.nelems; ++t) {
+    w[t] = weights[t] > floor_w ? weights[t] : floor_w;
+    total += w[t];
+  }
+  double acc = 0.0;
+  int32_t part = 0;
+  const double per = total / nparts;
+  for (int64_t i = 0; i < m.nelems; ++i) {
+    // advance to the part whose [part*per, (part+1)*per) window holds the
+    // midpoint of this element's weight span
+    const double mid = acc + 0.5 * w[order[i]];
+    while (part + 1 < nparts && mid >= (part + 1) * per) part++;
+    owners[order[i]] = part;
+    acc += w[order[i]];
+  }
   return owners;
 }
 

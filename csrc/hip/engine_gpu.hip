@@ -263,13 +263,15 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            const double *__restrict__ dest,
                            const int32_t *__restrict__ elem,
                            const double *__restrict__ weights,
+                           const uint16_t *__restrict__ groups,
                            double *__restrict__ out_pos,
                            int32_t *__restrict__ out_elem,
                            int8_t *__restrict__ out_status,
                            double *__restrict__ flux,
                            unsigned long long *__restrict__ lost, int64_t n,
                            int max_steps, bool reflective,
-                           const uint32_t *__restrict__ face_bc) {
+                           const uint32_t *__restrict__ face_bc, int ngroups,
+                           int64_t nelems) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -277,7 +279,8 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     int32_t oe;
     Vec3 op;
     bool esc;
-    auto add = [&](int32_t e, double v) { atomicAdd(&flux[e], v); };
+    const int64_t goff = groups ? (int64_t)(groups[i] % ngroups) * nelems : 0;
+    auto add = [&](int32_t e, double v) { atomicAdd(&flux[goff + e], v); };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
                      max_steps, add, &oe, &op, &esc, reflective, face_bc);
@@ -540,7 +543,8 @@ public:
 
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
-                int32_t *out_elem, int8_t *out_status) override {
+                int32_t *out_elem, int8_t *out_status,
+                const uint16_t *groups = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -549,26 +553,31 @@ public:
     int32_t *de = dmalloc<int32_t>(n), *doe = dmalloc<int32_t>(n);
     double *dop = dmalloc<double>(n * 3);
     int8_t *dst_ = dmalloc<int8_t>(n);
+    uint16_t *dg = groups ? dmalloc<uint16_t>(n) : nullptr;
     PT_HIP_CHECK(hipMemcpy(dp, pos, n * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(dd, dest, n * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(dw, weights, n * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(de, elem, n * 4, hipMemcpyHostToDevice));
+    if (dg)
+      PT_HIP_CHECK(hipMemcpy(dg, groups, n * 2, hipMemcpyHostToDevice));
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_);
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dop, doe, dst_,
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_);
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dop, doe, dst_,
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_elem, doe, n * 4, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_status, dst_, n, hipMemcpyDeviceToHost));
     for (void *q : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
-                    (void *)doe, (void *)dop, (void *)dst_})
-      (void)hipFree(q);
+                    (void *)doe, (void *)dop, (void *)dst_, (void *)dg})
+      if (q) (void)hipFree(q);
   }
 
   void end_batch() override {

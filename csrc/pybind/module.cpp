@@ -215,12 +215,22 @@ PYBIND11_MODULE(_core, m) {
       });
 
   m.def("partition_morton",
-        [](const Mesh &m_, int nparts) {
-          auto owners = partition_morton(m_, nparts);
+        [](const Mesh &m_, int nparts, py::object weights_obj) {
+          const double *wp = nullptr;
+          py::array_t<double, py::array::c_style | py::array::forcecast> w;
+          if (!weights_obj.is_none()) {
+            w = weights_obj.cast<
+                py::array_t<double, py::array::c_style | py::array::forcecast>>();
+            if ((int64_t)w.size() != m_.nelems)
+              throw std::runtime_error("partition_morton: weights size != nelems");
+            wp = w.data();
+          }
+          auto owners = partition_morton(m_, nparts, wp);
           auto a = py::array_t<int32_t>(owners.size());
           std::memcpy(a.mutable_data(), owners.data(), owners.size() * 4);
           return a;
-        });
+        },
+        py::arg("mesh"), py::arg("nparts"), py::arg("weights") = py::none());
   m.def("extract_submesh",
         [](const Mesh &m_, py::array_t<int32_t, py::array::c_style | py::array::forcecast> owners,
            int part, int ghost_rings) {
@@ -324,11 +334,21 @@ PYBIND11_MODULE(_core, m) {
            [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> pos,
               py::array_t<double, py::array::c_style | py::array::forcecast> dest,
               py::array_t<int32_t, py::array::c_style | py::array::forcecast> elem,
-              py::array_t<double, py::array::c_style | py::array::forcecast> weights) {
+              py::array_t<double, py::array::c_style | py::array::forcecast> weights,
+              py::object groups_obj) {
              const int64_t n = (int64_t)elem.size();
              if ((int64_t)pos.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
                  (int64_t)weights.size() != n)
                throw std::runtime_error("walk_raw: size mismatch");
+             py::array_t<uint16_t, py::array::c_style | py::array::forcecast> groups;
+             const uint16_t *gp = nullptr;
+             if (!groups_obj.is_none()) {
+               groups = groups_obj.cast<
+                   py::array_t<uint16_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)groups.size() != n)
+                 throw std::runtime_error("walk_raw: groups size mismatch");
+               gp = groups.data();
+             }
              auto out_pos = py::array_t<double>({n, (int64_t)3});
              auto out_elem = py::array_t<int32_t>(n);
              auto out_status = py::array_t<int8_t>(n);
@@ -336,10 +356,13 @@ PYBIND11_MODULE(_core, m) {
                py::gil_scoped_release nogil;
                e.eng->walk_raw(n, pos.data(), dest.data(), elem.data(),
                                weights.data(), out_pos.mutable_data(),
-                               out_elem.mutable_data(), out_status.mutable_data());
+                               out_elem.mutable_data(), out_status.mutable_data(),
+                               gp);
              }
              return py::make_tuple(out_pos, out_elem, out_status);
-           })
+           },
+           py::arg("pos"), py::arg("dest"), py::arg("elem"),
+           py::arg("weights"), py::arg("groups") = py::none())
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

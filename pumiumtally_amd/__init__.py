@@ -148,11 +148,12 @@ class TallyEngine:
             ptr(origin, "<f8", n * 3), ptr(dest, "<f8", n * 3),
             ptr(flying, "|i1", n), ptr(weights, "<f8", n))
 
-    def walk_raw(self, pos, dest, elem, weights):
+    def walk_raw(self, pos, dest, elem, weights, groups=None):
         """Batched raw segment walk (domain-decomposition support): returns
         (out_pos, out_elem, status) with status 0=done 1=escaped 2=handoff
-        3=lost; tallies into this engine's flux."""
-        return self._eng.walk_raw(pos, dest, elem, weights)
+        3=lost; tallies into this engine's flux.  groups: optional uint16
+        per-segment energy-group indices (flux row group*nelems+elem)."""
+        return self._eng.walk_raw(pos, dest, elem, weights, groups)
 
     def synchronize(self):
         self._eng.synchronize()

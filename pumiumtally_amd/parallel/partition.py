@@ -16,6 +16,16 @@ mesh on every rank, owners all rank 0) -- with a real decomposition.
 
 Exchange collective: torch.distributed all_to_all_single over RCCL/xGMI
 when on GPU (nccl backend), all_gather_object on gloo (CPU tests).
+
+Exchange record (9 float64 per handed-off particle): position[3],
+destination[3], weight, target global element id, energy group.
+
+Load balance: pass elem_weights to the constructor (per-element work
+estimates -- e.g. a previous batch's raw flux) to split the Morton curve
+by equal summed work instead of equal element count, or call
+repartition(weights) between batches to rebuild the decomposition from
+measured work (Morton chunks only balance element counts; weighted
+splits balance particle work).
 """
 from __future__ import annotations
 
@@ -23,34 +33,58 @@ import numpy as np
 
 from .dist import init_distributed
 
+_REC = 9  # floats per exchange record
+
 
 class PartitionedTally:
     def __init__(self, mesh, device=None, backend=None, max_rounds: int = 64,
-                 ghost_rings: int = 1):
-        from .. import TallyEngine, _core, have_gpu
-
+                 ghost_rings: int = 1, ngroups: int = 1, elem_weights=None):
         self.rank, self.world, self.local = init_distributed(backend)
         self.mesh = mesh
         self.max_rounds = max_rounds
-        self.owners = _core.partition_morton(mesh, self.world)
+        self.ngroups = max(1, int(ngroups))
+        self.ghost_rings = ghost_rings
+        self._device = device
+        self._build(elem_weights)
+
+    def _build(self, elem_weights=None):
+        from .. import TallyEngine, _core, have_gpu
+
+        mesh = self.mesh
+        w = None if elem_weights is None else np.asarray(elem_weights, np.float64)
+        self.owners = _core.partition_morton(mesh, self.world, w)
         self.sub = _core.extract_submesh(mesh, self.owners, self.rank,
-                                         ghost_rings)
+                                         self.ghost_rings)
         self.l2g = self.sub.elem_l2g
         self.g2l = -np.ones(mesh.nelems, dtype=np.int64)
         self.g2l[self.l2g] = np.arange(len(self.l2g))
         self.foreign_gid = self.sub.foreign_gid
         self.foreign_owner = self.sub.foreign_owner
+        device = self._device
         if device is None:
             device = f"cuda:{self.local}" if have_gpu() else "cpu"
         # Engine over the local submesh; used only through walk_raw + flux.
-        self.engine = TallyEngine(self.sub.local, 1, device=device)
+        self.engine = TallyEngine(self.sub.local, 1, device=device,
+                                  ngroups=self.ngroups)
+
+    def repartition(self, elem_weights):
+        """Rebuild the decomposition from per-element work estimates
+        (global nelems array, identical on every rank -- e.g.
+        flux_global().sum(axis=0) of the previous batch).  Discards the
+        current local tally; call flux_global()/end-of-batch readout
+        first."""
+        elem_weights = np.asarray(elem_weights, np.float64).reshape(-1)
+        if elem_weights.size != self.mesh.nelems:
+            raise ValueError("elem_weights must have one entry per element")
+        self._build(elem_weights)
 
     # -- helpers -----------------------------------------------------------
     def _exchange(self, records_per_rank):
-        """records_per_rank: list of world np.float64 arrays (k,8); returns
-        concatenated records received from all ranks."""
+        """records_per_rank: list of world np.float64 arrays (k,_REC);
+        returns concatenated records received from all ranks."""
         if self.world == 1:
-            return records_per_rank[0] if records_per_rank else np.zeros((0, 8))
+            return (records_per_rank[0] if records_per_rank
+                    else np.zeros((0, _REC)))
         import torch
         import torch.distributed as dist
 
@@ -67,43 +101,53 @@ class PartitionedTally:
             recv = torch.empty(int(sum(out_counts)), dtype=torch.float64,
                                device=counts.device)
             dist.all_to_all_single(recv, send, out_counts, in_counts)
-            return recv.cpu().numpy().reshape(-1, 8)
+            return recv.cpu().numpy().reshape(-1, _REC)
         # gloo: object all_gather
         gathered = [None] * self.world
         dist.all_gather_object(gathered, [np.asarray(r) for r in records_per_rank])
         mine = [g[self.rank] for g in gathered if g[self.rank].size]
-        return np.concatenate(mine).reshape(-1, 8) if mine else np.zeros((0, 8))
+        return (np.concatenate(mine).reshape(-1, _REC) if mine
+                else np.zeros((0, _REC)))
 
     # -- public API --------------------------------------------------------
-    def run_segments(self, origins, dests, weights):
+    def run_segments(self, origins, dests, weights, groups=None):
         """Walk one batch of global segments (origins->dests, weights),
         tallying into the partitioned flux.  Each rank passes the SAME
         global arrays (or its own shard -- ownership is resolved here);
         segments starting outside this rank's elements are ignored locally
-        and handled by their owner."""
+        and handled by their owner.  groups: optional per-segment energy
+        group indices (requires ngroups>1 at construction)."""
         origins = np.asarray(origins, np.float64).reshape(-1, 3)
         dests = np.asarray(dests, np.float64).reshape(-1, 3)
         weights = np.asarray(weights, np.float64).reshape(-1)
+        if groups is not None:
+            groups = np.asarray(groups, np.uint16).reshape(-1)
+            if groups.size != weights.size:
+                raise ValueError("groups size mismatch")
         gids = self.mesh.locate(origins)
         mine = (gids >= 0) & (self.owners[np.maximum(gids, 0)] == self.rank)
         pos = origins[mine]
         dst = dests[mine]
         wgt = weights[mine]
+        grp = groups[mine] if groups is not None else None
         elem = self.g2l[gids[mine]].astype(np.int32)
 
         for _round in range(self.max_rounds):
-            outbound = [np.zeros((0, 8)) for _ in range(self.world)]
+            outbound = [np.zeros((0, _REC)) for _ in range(self.world)]
             if len(elem):
                 out_pos, out_elem, status = self.engine.walk_raw(
-                    pos.ravel(), dst.ravel(), elem, wgt)
+                    pos.ravel(), dst.ravel(), elem, wgt, grp)
                 hand = status == 2
                 if hand.any():
                     k = -(out_elem[hand].astype(np.int64) + 2)
                     tgt_gid = self.foreign_gid[k]
                     tgt_owner = self.foreign_owner[k]
+                    g_col = (grp[hand] if grp is not None
+                             else np.zeros(int(hand.sum()), np.uint16))
                     rec = np.concatenate(
                         [out_pos[hand], dst[hand],
-                         wgt[hand, None], tgt_gid[:, None].astype(np.float64)],
+                         wgt[hand, None], tgt_gid[:, None].astype(np.float64),
+                         g_col[:, None].astype(np.float64)],
                         axis=1)
                     for r in range(self.world):
                         sel = tgt_owner == r
@@ -129,15 +173,20 @@ class PartitionedTally:
             dst = inbound[:, 3:6]
             wgt = inbound[:, 6]
             elem = self.g2l[inbound[:, 7].astype(np.int64)].astype(np.int32)
+            grp = (inbound[:, 8].astype(np.uint16)
+                   if groups is not None else None)
         else:
             raise RuntimeError("partitioned walk did not converge "
                                f"in {self.max_rounds} handoff rounds")
 
     def flux_global(self) -> np.ndarray:
-        """Scatter the local tally to global element ids and sum over ranks."""
-        local = self.engine.flux()
-        out = np.zeros(self.mesh.nelems)
-        out[self.l2g] = local
+        """Scatter the local tally to global element ids and sum over ranks.
+        Shape (nelems,) for ngroups=1 else (ngroups, nelems)."""
+        local = np.asarray(self.engine.flux())
+        ng = self.ngroups
+        nloc = len(self.l2g)
+        out = np.zeros((ng, self.mesh.nelems))
+        out[:, self.l2g] = local.reshape(ng, nloc)
         if self.world > 1:
             import torch
             import torch.distributed as dist
@@ -146,12 +195,18 @@ class PartitionedTally:
                 t = t.cuda(self.local)
             dist.all_reduce(t, op=dist.ReduceOp.SUM)
             out = t.cpu().numpy()
-        return out
+        return out[0] if ng == 1 else out
 
     def write_tally_results(self, filename="fluxresult.vtk"):
-        from .. import write_tally_vtk
+        from .. import _core, write_tally_vtk
 
         f = self.flux_global()
         if self.rank == 0:
-            write_tally_vtk(filename, self.mesh, f)
+            if self.ngroups > 1:
+                fields = [("flux", _core.normalize_flux(self.mesh, f.sum(axis=0)))]
+                fields += [(f"flux_g{g}", _core.normalize_flux(self.mesh, f[g]))
+                           for g in range(self.ngroups)]
+                self.mesh.write_vtk_fields(filename, fields)
+            else:
+                write_tally_vtk(filename, self.mesh, f)
         return f

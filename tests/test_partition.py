@@ -204,6 +204,162 @@ def test_gloo_world2_partitioned_gpu(tmp_path):
     assert "PART_OK" in outs[0]
 
 
+def test_weighted_partition_balance():
+    """Work-weighted Morton split equalizes summed weight, not counts."""
+    m = pt.build_box(6, 6, 6)
+    rng = np.random.default_rng(3)
+    # strongly skewed work: elements in x<0.5 cost 100x more
+    cen = np.array([np.asarray(m.centroid(int(t))) for t in range(m.nelems)])
+    w = np.where(cen[:, 0] < 0.5, 100.0, 1.0) * rng.uniform(0.5, 1.5, m.nelems)
+    owners = _core.partition_morton(m, 4, w)
+    assert owners.min() == 0 and owners.max() == 3
+    sums = np.array([w[owners == p].sum() for p in range(4)])
+    counts = np.bincount(owners, minlength=4)
+    # weighted sums within 20% of each other; counts very unbalanced
+    assert sums.max() / sums.min() < 1.2, sums
+    assert counts.max() / counts.min() > 2, counts
+    # unweighted call still balances counts
+    owners0 = _core.partition_morton(m, 4)
+    c0 = np.bincount(owners0, minlength=4)
+    assert c0.max() - c0.min() <= 1
+
+
+def test_single_rank_partitioned_groups():
+    """world=1 PartitionedTally with energy groups == plain grouped engine."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(3, 3, 3)
+    n = 200
+    ng = 3
+    rng = np.random.default_rng(11)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = rng.uniform(0.05, 0.95, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, ng, n).astype(np.uint16)
+
+    ptal = PartitionedTally(m, device="cpu", ngroups=ng)
+    ptal.run_segments(o, d, w, groups=g)
+    got = ptal.flux_global()
+    assert got.shape == (ng, m.nelems)
+
+    ref = pt.TallyEngine(m, n, device="cpu", ngroups=ng)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g)
+    assert np.allclose(got, ref.flux(), atol=1e-12)
+    # group slices are genuinely distinct (skewed by construction)
+    assert not np.allclose(got[0], got[1])
+
+
+def test_repartition_weighted():
+    """repartition(weights) rebuilds the decomposition and the new walk
+    still matches the single-mesh oracle."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(3, 3, 3)
+    n = 150
+    rng = np.random.default_rng(21)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = rng.uniform(0.05, 0.95, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+
+    ptal = PartitionedTally(m, device="cpu")
+    ptal.run_segments(o, d, w)
+    batch1 = ptal.flux_global()
+
+    # feed measured work back in; decomposition changes, results don't
+    ptal.repartition(batch1 + 1e-9)
+    ptal.run_segments(o, d, w)
+    batch2 = ptal.flux_global()
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(batch1, ref.flux(), atol=1e-12)
+    assert np.allclose(batch2, ref.flux(), atol=1e-12)
+
+
+WORKER_GROUPS = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+from pumiumtally_amd.parallel.partition import PartitionedTally
+
+rank = int(os.environ["RANK"])
+mesh = pt.build_box(4, 4, 4)
+n = 300
+ng = 2
+rng = np.random.default_rng(78)  # same segments on both ranks
+o = rng.uniform(0.05, 0.95, size=(n, 3))
+d = rng.uniform(0.05, 0.95, size=(n, 3))
+w = rng.uniform(0.1, 1.0, n)
+g = rng.integers(0, ng, n).astype(np.uint16)
+
+ptal = PartitionedTally(mesh, device="cpu", backend="gloo", ngroups=ng)
+ptal.run_segments(o, d, w, groups=g)
+flux = ptal.flux_global()
+
+if rank == 0:
+    ref = pt.TallyEngine(mesh, n, device="cpu", ngroups=ng)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g)
+    err = np.abs(flux - ref.flux()).max()
+    assert err < 1e-10, err
+    print("PART_OK")
+import torch.distributed as dist
+dist.destroy_process_group()
+"""
+
+
+def test_gloo_world2_partitioned_groups(tmp_path):
+    """Cross-rank handoff carries the energy group (9-double record)."""
+    pytest.importorskip("torch")
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER_GROUPS)
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(20000 + (os.getpid() + 57) % 20000),
+        "WORLD_SIZE": "2",
+        "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "PART_OK" in outs[0]
+
+
+@pytest.mark.gpu
+def test_single_rank_partitioned_groups_gpu():
+    """Grouped walk_raw on a GPU submesh == grouped CPU oracle."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(6, 6, 6)
+    n = 5000
+    ng = 4
+    rng = np.random.default_rng(13)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, ng, n).astype(np.uint16)
+
+    ref = pt.TallyEngine(m, n, device="cpu", ngroups=ng)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g)
+
+    ptal = PartitionedTally(m, device="cuda:0", ngroups=ng)
+    assert ptal.engine.is_gpu
+    ptal.run_segments(o, d, w, groups=g)
+    got = ptal.flux_global()
+    assert np.abs(got - ref.flux()).max() < 1e-10
+
+
 def test_submesh_carries_face_bc():
     """Per-face reflective bits survive submesh extraction."""
     m = pt.build_box(3, 3, 3)
