@@ -62,17 +62,25 @@ public:
   // [0, ngroups); contributions land in flux[group*nelems + elem].  The
   // reference has a single scalar tally; ngroups=1 (the default) matches
   // it exactly.
+  // responses (optional, nullable): n*nscores per-particle response
+  // multipliers for simultaneous tally SCORES (e.g. flux + heating +
+  // fission from one walk).  Score k of a crossing tallies
+  // seg * weight * (responses ? responses[i*nscores+k] : 1.0) into
+  // flux[(k*ngroups + group)*nelems + elem].  nscores=1 with null
+  // responses (the default) is exactly the reference's single tally.
   virtual void move(const double *origin, const double *dest,
                     const int8_t *flying, const double *weights, int64_t n,
-                    const uint16_t *groups = nullptr) = 0;
+                    const uint16_t *groups = nullptr,
+                    const double *responses = nullptr) = 0;
 
   // Fast path for callers that know no particle was resampled this step
   // (origin == committed position for every particle): skips the origin
   // upload and phase A entirely.  origin=nullptr in move() semantics.
   virtual void move_continue(const double *dest, const int8_t *flying,
                              const double *weights, int64_t n,
-                             const uint16_t *groups = nullptr) {
-    move(nullptr, dest, flying, weights, n, groups);
+                             const uint16_t *groups = nullptr,
+                             const double *responses = nullptr) {
+    move(nullptr, dest, flying, weights, n, groups, responses);
   }
 
   // Device-resident move: all arrays already live in this engine's device
@@ -80,9 +88,10 @@ public:
   // origin may be nullptr (continue semantics).  Throws on the CPU engine.
   virtual void move_device(const double *d_origin, const double *d_dest,
                            const int8_t *d_flying, const double *d_weights,
-                           int64_t n, const uint16_t *d_groups = nullptr) {
+                           int64_t n, const uint16_t *d_groups = nullptr,
+                           const double *d_responses = nullptr) {
     (void)d_origin; (void)d_dest; (void)d_flying; (void)d_weights; (void)n;
-    (void)d_groups;
+    (void)d_groups; (void)d_responses;
     throw std::runtime_error("move_device requires the GPU engine");
   }
 
@@ -92,15 +101,18 @@ public:
   // 2=handoff (out_elem = encoded foreign ref -(2+k)), 3=lost.
   // groups (nullable): per-segment energy-group index, same semantics as
   // move(); a handed-off particle keeps its group (the partitioned driver
-  // carries it in the exchange record).  Synchronous; host memory.
+  // carries it in the exchange record).  responses (nullable): n*nscores
+  // per-segment score multipliers, same semantics as move().
+  // Synchronous; host memory.
   virtual void walk_raw(int64_t n, const double *pos, const double *dest,
                         const int32_t *elem, const double *weights,
                         double *out_pos, int32_t *out_elem,
                         int8_t *out_status,
-                        const uint16_t *groups = nullptr) = 0;
+                        const uint16_t *groups = nullptr,
+                        const double *responses = nullptr) = 0;
 
   // Read back state (host copies).
-  virtual std::vector<double> flux() const = 0;           // nelems*ngroups, raw tally
+  virtual std::vector<double> flux() const = 0;           // nelems*ngroups*nscores, raw tally
   virtual std::vector<int32_t> elem_ids() const = 0;      // n
   virtual std::vector<double> positions() const = 0;      // n*3
   virtual std::vector<uint8_t> escaped() const = 0;       // n
@@ -130,7 +142,8 @@ public:
   virtual void synchronize() {}
 
   int max_steps = 0; // 0 = auto (set by implementations from mesh size)
-  int ngroups = 1;   // energy groups (flux is [ngroups x nelems])
+  int ngroups = 1;   // energy groups
+  int nscores = 1;   // tally scores (flux is [nscores x ngroups x nelems])
 
   bool reflective = false; // specular-reflect at boundaries (else vacuum)
 
@@ -164,11 +177,12 @@ inline double loc_tol_rel() {
 }
 
 std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles,
-                                        int ngroups = 1);
+                                        int ngroups = 1, int nscores = 1);
 
 // Returns nullptr when no HIP device is available.
 std::unique_ptr<Engine> make_gpu_engine(Mesh mesh, int64_t num_particles,
-                                        int device, int ngroups = 1);
+                                        int device, int ngroups = 1,
+                                        int nscores = 1);
 
 // Normalized flux = flux / element volume (volume-only, matching the
 // reference implementation rather than its docstring:

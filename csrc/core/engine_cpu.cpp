@@ -30,9 +30,11 @@ namespace {
 
 class CpuEngine final : public Engine {
 public:
-  CpuEngine(Mesh mesh, int64_t n, int groups) : mesh_(std::move(mesh)), n_(n) {
+  CpuEngine(Mesh mesh, int64_t n, int groups, int scores)
+      : mesh_(std::move(mesh)), n_(n) {
     ngroups = groups < 1 ? 1 : groups;
-    flux_.assign(mesh_.nelems * ngroups, 0.0);
+    nscores = scores < 1 ? 1 : scores;
+    flux_.assign(mesh_.nelems * ngroups * nscores, 0.0);
     pos_.resize(n_ * 3);
     elem_.assign(n_, 0);
     escaped_.assign(n_, 0);
@@ -78,8 +80,8 @@ public:
   }
 
   void move(const double *origin, const double *dest, const int8_t *flying,
-            const double *weights, int64_t n,
-            const uint16_t *groups = nullptr) override {
+            const double *weights, int64_t n, const uint16_t *groups = nullptr,
+            const double *responses = nullptr) override {
     check_n(n);
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     // Thread-parallel over particles for large batches (the reference's CPU
@@ -106,8 +108,8 @@ public:
           const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
           int64_t my_lost = 0, my_reloc = 0;
           for (int64_t i = lo; i < hi; ++i)
-            move_one(origin, dest, flying, weights, groups, i, steps,
-                     partial[t].data(), my_lost, my_reloc);
+            move_one(origin, dest, flying, weights, groups, responses, i,
+                     steps, partial[t].data(), my_lost, my_reloc);
           lost += my_lost;
           reloc += my_reloc;
         });
@@ -123,8 +125,8 @@ public:
     }
     int64_t lost = 0, reloc = 0;
     for (int64_t i = 0; i < n; ++i)
-      move_one(origin, dest, flying, weights, groups, i, steps, flux_.data(),
-               lost, reloc);
+      move_one(origin, dest, flying, weights, groups, responses, i, steps,
+               flux_.data(), lost, reloc);
     stats_.lost_particles += lost;
     stats_.relocated += reloc;
     stats_.moves++;
@@ -134,8 +136,8 @@ public:
   // particles -- behavioral pin, see engine.h) + phase B tallied walk.
   void move_one(const double *origin, const double *dest,
                 const int8_t *flying, const double *weights,
-                const uint16_t *groups, int64_t i, int steps,
-                double *flux_out, int64_t &lost, int64_t &reloc) {
+                const uint16_t *groups, const double *responses, int64_t i,
+                int steps, double *flux_out, int64_t &lost, int64_t &reloc) {
     if (!flying[i]) return;
     Vec3 o{pos_[i * 3], pos_[i * 3 + 1], pos_[i * 3 + 2]};
     if (origin && !escaped_[i]) {
@@ -159,7 +161,16 @@ public:
     bool out_esc;
     const int64_t goff =
         groups ? (int64_t)(groups[i] % ngroups) * mesh_.nelems : 0;
-    auto add = [&](int32_t e, double v) { flux_out[goff + e] += v; };
+    const int64_t gsz = (int64_t)ngroups * mesh_.nelems;
+    const double *resp = responses ? responses + i * nscores : nullptr;
+    auto add = [&](int32_t e, double v) {
+      if (!resp) {
+        flux_out[goff + e] += v;
+        return;
+      }
+      for (int k = 0; k < nscores; ++k)
+        flux_out[k * gsz + goff + e] += v * resp[k];
+    };
     const uint32_t *bc =
         mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
     if (walk_fp32)
@@ -184,7 +195,8 @@ public:
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status,
-                const uint16_t *groups = nullptr) override {
+                const uint16_t *groups = nullptr,
+                const double *responses = nullptr) override {
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     const unsigned hw = std::thread::hardware_concurrency();
     if (n >= 65536 && hw > 1) {
@@ -199,8 +211,9 @@ public:
           const int64_t lo = t * per, hi = std::min<int64_t>(n, lo + per);
           int64_t my_lost = 0;
           for (int64_t i = lo; i < hi; ++i)
-            walk_raw_one(pos, dest, elem, weights, groups, out_pos, out_elem,
-                         out_status, i, steps, partial[t].data(), my_lost);
+            walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
+                         out_elem, out_status, i, steps, partial[t].data(),
+                         my_lost);
           lost += my_lost;
         });
       }
@@ -212,15 +225,15 @@ public:
     }
     int64_t lost = 0;
     for (int64_t i = 0; i < n; ++i)
-      walk_raw_one(pos, dest, elem, weights, groups, out_pos, out_elem,
-                   out_status, i, steps, flux_.data(), lost);
+      walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
+                   out_elem, out_status, i, steps, flux_.data(), lost);
     stats_.lost_particles += lost;
   }
 
   void walk_raw_one(const double *pos, const double *dest,
                     const int32_t *elem, const double *weights,
-                    const uint16_t *groups, double *out_pos,
-                    int32_t *out_elem, int8_t *out_status,
+                    const uint16_t *groups, const double *responses,
+                    double *out_pos, int32_t *out_elem, int8_t *out_status,
                     int64_t i, int steps, double *flux_out, int64_t &lost) {
     {
       const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -230,7 +243,16 @@ public:
       bool esc;
       const int64_t goff =
           groups ? (int64_t)(groups[i] % ngroups) * mesh_.nelems : 0;
-      auto add = [&](int32_t e, double v) { flux_out[goff + e] += v; };
+      const int64_t gsz = (int64_t)ngroups * mesh_.nelems;
+      const double *resp = responses ? responses + i * nscores : nullptr;
+      auto add = [&](int32_t e, double v) {
+        if (!resp) {
+          flux_out[goff + e] += v;
+          return;
+        }
+        for (int k = 0; k < nscores; ++k)
+          flux_out[k * gsz + goff + e] += v * resp[k];
+      };
       const uint32_t *bc =
           mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
       if (walk_fp32)
@@ -315,8 +337,9 @@ private:
 } // namespace
 
 std::unique_ptr<Engine> make_cpu_engine(Mesh mesh, int64_t num_particles,
-                                        int ngroups) {
-  return std::make_unique<CpuEngine>(std::move(mesh), num_particles, ngroups);
+                                        int ngroups, int nscores) {
+  return std::make_unique<CpuEngine>(std::move(mesh), num_particles, ngroups,
+                                     nscores);
 }
 
 } // namespace pumitally

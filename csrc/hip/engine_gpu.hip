@@ -186,7 +186,7 @@ __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
 // dispatches block b to XCD b%8 and each XCD has a private 4 MiB L2; we
 // remap blocks so each XCD owns one contiguous (hence spatially compact)
 // slot range.  Purely a speed lever: any placement is correct.
-template <bool F32>
+template <bool F32, bool Scored = false>
 __global__ void k_move(const Plane *__restrict__ planes,
                        const Plane32 *__restrict__ planes32,
                        const int32_t *__restrict__ nbr, GridView grid,
@@ -202,8 +202,14 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        unsigned long long *__restrict__ lost, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
                        int64_t nelems, int slice_mask, bool reflective,
-                       const uint32_t *__restrict__ face_bc) {
-  flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups;
+                       const uint32_t *__restrict__ face_bc,
+                       const double *__restrict__ resp = nullptr,
+                       int nscores = 1) {
+  // Scored=false is the headline instantiation: resp/nscores are unused and
+  // the slice stride stays nelems*ngroups -- codegen identical to before
+  // multi-score existed.
+  flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups *
+          (Scored ? nscores : 1);
   const unsigned bpx = gridDim.x / 8u;
   const unsigned vb = (blockIdx.x % 8u) * bpx + blockIdx.x / 8u;
   const int64_t m = hi - lo;
@@ -236,7 +242,19 @@ __global__ void k_move(const Plane *__restrict__ planes,
     bool out_esc;
     const int64_t goff =
         groups ? (int64_t)(groups[c] % ngroups) * nelems : 0;
-    auto add = [&](int32_t el, double v) { atomicAdd(&flux[goff + el], v); };
+    auto add = [&](int32_t el, double v) {
+      if constexpr (Scored) {
+        const int64_t gsz = (int64_t)ngroups * nelems;
+        if (resp) {
+          for (int k = 0; k < nscores; ++k)
+            atomicAdd(&flux[k * gsz + goff + el], v * resp[c * nscores + k]);
+        } else {
+          atomicAdd(&flux[goff + el], v);
+        }
+      } else {
+        atomicAdd(&flux[goff + el], v);
+      }
+    };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
                      add, &out_elem, &out_pos, &out_esc, reflective, face_bc);
@@ -264,6 +282,7 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            const int32_t *__restrict__ elem,
                            const double *__restrict__ weights,
                            const uint16_t *__restrict__ groups,
+                           const double *__restrict__ resp,
                            double *__restrict__ out_pos,
                            int32_t *__restrict__ out_elem,
                            int8_t *__restrict__ out_status,
@@ -271,7 +290,7 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            unsigned long long *__restrict__ lost, int64_t n,
                            int max_steps, bool reflective,
                            const uint32_t *__restrict__ face_bc, int ngroups,
-                           int64_t nelems) {
+                           int64_t nelems, int nscores) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -280,7 +299,15 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     Vec3 op;
     bool esc;
     const int64_t goff = groups ? (int64_t)(groups[i] % ngroups) * nelems : 0;
-    auto add = [&](int32_t e, double v) { atomicAdd(&flux[goff + e], v); };
+    const int64_t gsz = (int64_t)ngroups * nelems;
+    auto add = [&](int32_t e, double v) {
+      if (resp) {
+        for (int k = 0; k < nscores; ++k)
+          atomicAdd(&flux[k * gsz + goff + e], v * resp[i * nscores + k]);
+      } else {
+        atomicAdd(&flux[goff + e], v);
+      }
+    };
     if constexpr (F32)
       walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
                      max_steps, add, &oe, &op, &esc, reflective, face_bc);
@@ -358,9 +385,10 @@ template <class T> T *dmalloc(int64_t count) {
 
 class GpuEngine final : public Engine {
 public:
-  GpuEngine(Mesh mesh, int64_t n, int device, int groups)
+  GpuEngine(Mesh mesh, int64_t n, int device, int groups, int scores)
       : mesh_(std::move(mesh)), n_(n) {
     ngroups = groups < 1 ? 1 : groups;
+    nscores = scores < 1 ? 1 : scores;
     PT_HIP_CHECK(hipSetDevice(device));
     device_ = device;
     PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
@@ -404,10 +432,10 @@ public:
     d_escaped_ = dmalloc<uint8_t>(n_);
     d_s2c_ = dmalloc<int32_t>(n_);
     slices_ = flux_slices();
-    d_flux_ = dmalloc<double>(mesh_.nelems * ngroups * slices_);
+    fsz_ = mesh_.nelems * ngroups * nscores;
+    d_flux_ = dmalloc<double>(fsz_ * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
-    PT_HIP_CHECK(
-        hipMemset(d_flux_, 0, mesh_.nelems * ngroups * slices_ * sizeof(double)));
+    PT_HIP_CHECK(hipMemset(d_flux_, 0, fsz_ * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
 
     // Parity double-buffered input staging.
@@ -417,6 +445,9 @@ public:
       d_flying_[p] = dmalloc<int8_t>(n_);
       d_weights_[p] = dmalloc<double>(n_);
       d_groups_[p] = ngroups > 1 ? dmalloc<uint16_t>(n_) : nullptr;
+      // d_resp_ is lazy-allocated on the first move() that passes
+      // responses (valid even with nscores==1: a single per-particle
+      // response multiplier).
     }
 
     // Spatial-sort scratch.
@@ -464,6 +495,7 @@ public:
                     (void *)d_flying_[0], (void *)d_flying_[1],
                     (void *)d_weights_[0], (void *)d_weights_[1],
                     (void *)d_groups_[0], (void *)d_groups_[1],
+                    (void *)d_resp_[0], (void *)d_resp_[1],
                     (void *)d_bsum_, (void *)d_bsq_, (void *)d_face_bc_,
                     (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
@@ -500,7 +532,8 @@ public:
 
   void move(const double *origin, const double *dest, const int8_t *flying,
             const double *weights, int64_t n,
-            const uint16_t *groups = nullptr) override {
+            const uint16_t *groups = nullptr,
+            const double *responses = nullptr) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -515,11 +548,16 @@ public:
     stage(weights, n * sizeof(double), d_weights_[p], s_copy_);
     if (groups && d_groups_[p])
       stage(groups, n * sizeof(uint16_t), d_groups_[p], s_copy_);
+    if (responses) {
+      if (!d_resp_[p]) d_resp_[p] = dmalloc<double>(n_ * nscores);
+      stage(responses, n * nscores * sizeof(double), d_resp_[p], s_copy_);
+    }
     PT_HIP_CHECK(hipEventRecord(copy_ev_[p], s_copy_));
     PT_HIP_CHECK(hipStreamWaitEvent(s_comp_, copy_ev_[p], 0));
     launch_move_chunks(origin ? d_origin_[p] : nullptr, d_dest_[p],
                        d_flying_[p], d_weights_[p],
-                       groups ? d_groups_[p] : nullptr, n, steps);
+                       groups ? d_groups_[p] : nullptr,
+                       responses ? d_resp_[p] : nullptr, n, steps);
     PT_HIP_CHECK(hipEventRecord(kernels_done_[p], s_comp_));
     maybe_resort();
     // The caller may mutate or free its buffers as soon as move() returns
@@ -531,12 +569,13 @@ public:
 
   void move_device(const double *d_origin, const double *d_dest,
                    const int8_t *d_flying, const double *d_weights, int64_t n,
-                   const uint16_t *d_groups = nullptr) override {
+                   const uint16_t *d_groups = nullptr,
+                   const double *d_responses = nullptr) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    launch_move_chunks(d_origin, d_dest, d_flying, d_weights, d_groups, n,
-                       steps);
+    launch_move_chunks(d_origin, d_dest, d_flying, d_weights, d_groups,
+                       d_responses, n, steps);
     maybe_resort();
     stats_.moves++;
   }
@@ -544,7 +583,8 @@ public:
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status,
-                const uint16_t *groups = nullptr) override {
+                const uint16_t *groups = nullptr,
+                const double *responses = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -554,36 +594,41 @@ public:
     double *dop = dmalloc<double>(n * 3);
     int8_t *dst_ = dmalloc<int8_t>(n);
     uint16_t *dg = groups ? dmalloc<uint16_t>(n) : nullptr;
+    double *dr = responses ? dmalloc<double>(n * nscores) : nullptr;
     PT_HIP_CHECK(hipMemcpy(dp, pos, n * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(dd, dest, n * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(dw, weights, n * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(de, elem, n * 4, hipMemcpyHostToDevice));
     if (dg)
       PT_HIP_CHECK(hipMemcpy(dg, groups, n * 2, hipMemcpyHostToDevice));
+    if (dr)
+      PT_HIP_CHECK(
+          hipMemcpy(dr, responses, n * nscores * 8, hipMemcpyHostToDevice));
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems);
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
+          dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems, nscores);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dop, doe, dst_,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems);
+          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
+          dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems, nscores);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_elem, doe, n * 4, hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_status, dst_, n, hipMemcpyDeviceToHost));
     for (void *q : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
-                    (void *)doe, (void *)dop, (void *)dst_, (void *)dg})
+                    (void *)doe, (void *)dop, (void *)dst_, (void *)dg,
+                    (void *)dr})
       if (q) (void)hipFree(q);
   }
 
   void end_batch() override {
     PT_HIP_CHECK(hipSetDevice(device_));
     sync();
-    const int64_t fsz = mesh_.nelems * ngroups;
+    const int64_t fsz = fsz_;
     if (!d_bsum_) {
       d_bsum_ = dmalloc<double>(fsz);
       d_bsq_ = dmalloc<double>(fsz);
@@ -606,7 +651,7 @@ public:
   int64_t num_batches() const override { return nbatches_; }
 
   std::vector<double> fetch_acc(const double *p) const {
-    const int64_t fsz = mesh_.nelems * ngroups;
+    const int64_t fsz = fsz_;
     std::vector<double> out(fsz, 0.0);
     if (p) {
       sync();
@@ -618,7 +663,7 @@ public:
 
   std::vector<double> flux() const override {
     sync();
-    const int64_t fsz = mesh_.nelems * ngroups;
+    const int64_t fsz = fsz_;
     if (slices_ > 1) {
       k_reduce_slices<<<grid_blocks(fsz), kBlock, 0, s_comp_>>>(d_flux_, fsz,
                                                                slices_);
@@ -660,7 +705,7 @@ public:
   }
 
   void set_flux(const double *f, int64_t ne) override {
-    if (ne != mesh_.nelems * ngroups)
+    if (ne != fsz_)
       throw std::runtime_error("set_flux size mismatch");
     sync();
     PT_HIP_CHECK(hipMemset(d_flux_, 0, ne * slices_ * sizeof(double)));
@@ -705,20 +750,39 @@ private:
 
   void launch_move_chunks(const double *origin, const double *dest,
                           const int8_t *flying, const double *weights,
-                          const uint16_t *groups, int64_t n, int steps) {
+                          const uint16_t *groups, const double *resp,
+                          int64_t n, int steps) {
     // Chunked launches: a ~2.6M-slot launch keeps each XCD's Morton-
     // contiguous slot range's mesh working set inside its private L2.
+    // Scored moves take the k_move<_,true> instantiation; the unscored
+    // (headline) instantiation is compiled without any score machinery.
     const int64_t chunk = chunk_particles(n);
     for (int64_t lo = 0; lo < n; lo += chunk) {
       const int64_t hi = std::min(n, lo + chunk);
-      if (walk_fp32)
-        k_move<true><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+      const int blocks = grid_blocks(hi - lo);
+      if (resp) {
+        if (walk_fp32)
+          k_move<true, true><<<blocks, kBlock, 0, s_comp_>>>(
+              d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin,
+              dest, flying, weights, groups, ngroups, d_pos_, d_elem_,
+              d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_, steps,
+              mesh_.nelems, slices_ - 1, reflective, d_face_bc_, resp,
+              nscores);
+        else
+          k_move<false, true><<<blocks, kBlock, 0, s_comp_>>>(
+              d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin,
+              dest, flying, weights, groups, ngroups, d_pos_, d_elem_,
+              d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_, steps,
+              mesh_.nelems, slices_ - 1, reflective, d_face_bc_, resp,
+              nscores);
+      } else if (walk_fp32)
+        k_move<true><<<blocks, kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
             slices_ - 1, reflective, d_face_bc_);
       else
-        k_move<false><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+        k_move<false><<<blocks, kBlock, 0, s_comp_>>>(
             d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
             flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
             d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
@@ -770,6 +834,7 @@ private:
 
   Mesh mesh_;
   int64_t n_;
+  int64_t fsz_ = 0; // nelems * ngroups * nscores
   int device_ = 0;
   double loc_tol_ = 1e-12;
   int slices_ = 1;
@@ -797,6 +862,7 @@ private:
   unsigned long long *d_lost_ = nullptr;
   double *d_origin_[2] = {nullptr, nullptr};
   uint16_t *d_groups_[2] = {nullptr, nullptr};
+  double *d_resp_[2] = {nullptr, nullptr};
   double *d_bsum_ = nullptr, *d_bsq_ = nullptr;
   uint32_t *d_face_bc_ = nullptr;
   int64_t nbatches_ = 0;
@@ -819,14 +885,14 @@ private:
 } // namespace
 
 std::unique_ptr<Engine> make_gpu_engine(Mesh mesh, int64_t num_particles,
-                                        int device, int ngroups) {
+                                        int device, int ngroups, int nscores) {
   int count = 0;
   if (hipGetDeviceCount(&count) != hipSuccess || count <= device) {
     (void)hipGetLastError();
     return nullptr;
   }
   return std::make_unique<GpuEngine>(std::move(mesh), num_particles, device,
-                                     ngroups);
+                                     ngroups, nscores);
 }
 
 } // namespace pumitally

@@ -41,20 +41,20 @@ struct PyEngine {
   bool gpu = false;
 
   PyEngine(const Mesh &mesh, int64_t n, const std::string &device,
-           int ngroups = 1) {
+           int ngroups = 1, int nscores = 1) {
     if (device == "cpu") {
-      eng = make_cpu_engine(mesh, n, ngroups);
+      eng = make_cpu_engine(mesh, n, ngroups, nscores);
     } else {
       int ordinal = 0;
       if (device.rfind("cuda:", 0) == 0) ordinal = std::stoi(device.substr(5));
       else if (device.rfind("gpu:", 0) == 0) ordinal = std::stoi(device.substr(4));
       else if (device != "cuda" && device != "gpu" && device != "auto")
         throw std::runtime_error("device must be cpu/cuda[:N]/auto");
-      eng = make_gpu_engine(mesh, n, ordinal, ngroups);
+      eng = make_gpu_engine(mesh, n, ordinal, ngroups, nscores);
       if (eng) {
         gpu = true;
       } else if (device == "auto") {
-        eng = make_cpu_engine(mesh, n, ngroups);
+        eng = make_cpu_engine(mesh, n, ngroups, nscores);
       } else {
         // Fail loudly: a GPU was requested but none is usable.  GPU tests
         // must never fall back silently to the CPU oracle.
@@ -265,11 +265,13 @@ PYBIND11_MODULE(_core, m) {
         });
 
   py::class_<PyEngine>(m, "Engine")
-      .def(py::init<const Mesh &, int64_t, const std::string &, int>(),
+      .def(py::init<const Mesh &, int64_t, const std::string &, int, int>(),
            py::arg("mesh"), py::arg("num_particles"), py::arg("device") = "auto",
-           py::arg("ngroups") = 1)
+           py::arg("ngroups") = 1, py::arg("nscores") = 1)
       .def_property_readonly("ngroups",
                              [](const PyEngine &e) { return e.eng->ngroups; })
+      .def_property_readonly("nscores",
+                             [](const PyEngine &e) { return e.eng->nscores; })
       .def_property_readonly("num_particles",
                              [](const PyEngine &e) { return e.eng->num_particles(); })
       .def_property_readonly("is_gpu", [](const PyEngine &e) { return e.gpu; })
@@ -287,7 +289,7 @@ PYBIND11_MODULE(_core, m) {
               py::array_t<double, py::array::c_style> dest,
               py::array_t<int8_t, py::array::c_style> flying,
               py::array_t<double, py::array::c_style> weights,
-              py::object groups) {
+              py::object groups, py::object responses) {
              const int64_t n = e.eng->num_particles();
              if ((int64_t)origin.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
                  (int64_t)flying.size() != n || (int64_t)weights.size() != n)
@@ -300,42 +302,78 @@ PYBIND11_MODULE(_core, m) {
                  throw std::runtime_error("move: groups size mismatch");
                gp = garr.data();
              }
+             const double *rp = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast> rarr;
+             if (!responses.is_none()) {
+               rarr = responses.cast<py::array_t<double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)rarr.size() != n * e.eng->nscores)
+                 throw std::runtime_error("move: responses size must be n*nscores");
+               rp = rarr.data();
+             }
              py::gil_scoped_release nogil;
              e.eng->move(origin.data(), dest.data(), flying.data(),
-                         weights.data(), n, gp);
+                         weights.data(), n, gp, rp);
            },
            py::arg("origin"), py::arg("dest"), py::arg("flying"),
-           py::arg("weights"), py::arg("groups") = py::none())
+           py::arg("weights"), py::arg("groups") = py::none(),
+           py::arg("responses") = py::none())
       .def("move_continue",
            [](PyEngine &e, py::array_t<double, py::array::c_style> dest,
               py::array_t<int8_t, py::array::c_style> flying,
-              py::array_t<double, py::array::c_style> weights) {
+              py::array_t<double, py::array::c_style> weights,
+              py::object groups, py::object responses) {
              const int64_t n = e.eng->num_particles();
              if ((int64_t)dest.size() != n * 3 || (int64_t)flying.size() != n ||
                  (int64_t)weights.size() != n)
                throw std::runtime_error("move_continue: array size mismatch");
+             const uint16_t *gp = nullptr;
+             py::array_t<uint16_t, py::array::c_style | py::array::forcecast> garr;
+             if (!groups.is_none()) {
+               garr = groups.cast<py::array_t<uint16_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)garr.size() != n)
+                 throw std::runtime_error("move_continue: groups size mismatch");
+               gp = garr.data();
+             }
+             const double *rp = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast> rarr;
+             if (!responses.is_none()) {
+               rarr = responses.cast<py::array_t<double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)rarr.size() != n * e.eng->nscores)
+                 throw std::runtime_error("move_continue: responses size must be n*nscores");
+               rp = rarr.data();
+             }
              py::gil_scoped_release nogil;
-             e.eng->move_continue(dest.data(), flying.data(), weights.data(), n);
-           })
+             e.eng->move_continue(dest.data(), flying.data(), weights.data(),
+                                  n, gp, rp);
+           },
+           py::arg("dest"), py::arg("flying"), py::arg("weights"),
+           py::arg("groups") = py::none(), py::arg("responses") = py::none())
       .def("move_device",
            // Raw device-pointer entry (integers as returned by
            // torch.Tensor.data_ptr()).  origin_ptr=0 means continue
-           // semantics.  The Python wrapper validates device/dtype/shape.
+           // semantics; groups_ptr/responses_ptr=0 means none.  The Python
+           // wrapper validates device/dtype/shape.
            [](PyEngine &e, uintptr_t origin_ptr, uintptr_t dest_ptr,
-              uintptr_t flying_ptr, uintptr_t weights_ptr) {
+              uintptr_t flying_ptr, uintptr_t weights_ptr,
+              uintptr_t groups_ptr, uintptr_t responses_ptr) {
              py::gil_scoped_release nogil;
              e.eng->move_device((const double *)origin_ptr,
                                 (const double *)dest_ptr,
                                 (const int8_t *)flying_ptr,
                                 (const double *)weights_ptr,
-                                e.eng->num_particles());
-           })
+                                e.eng->num_particles(),
+                                (const uint16_t *)groups_ptr,
+                                (const double *)responses_ptr);
+           },
+           py::arg("origin_ptr"), py::arg("dest_ptr"), py::arg("flying_ptr"),
+           py::arg("weights_ptr"), py::arg("groups_ptr") = 0,
+           py::arg("responses_ptr") = 0)
       .def("walk_raw",
            [](PyEngine &e, py::array_t<double, py::array::c_style | py::array::forcecast> pos,
               py::array_t<double, py::array::c_style | py::array::forcecast> dest,
               py::array_t<int32_t, py::array::c_style | py::array::forcecast> elem,
               py::array_t<double, py::array::c_style | py::array::forcecast> weights,
-              py::object groups_obj) {
+              py::object groups_obj, py::object responses_obj) {
              const int64_t n = (int64_t)elem.size();
              if ((int64_t)pos.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
                  (int64_t)weights.size() != n)
@@ -349,6 +387,15 @@ PYBIND11_MODULE(_core, m) {
                  throw std::runtime_error("walk_raw: groups size mismatch");
                gp = groups.data();
              }
+             py::array_t<double, py::array::c_style | py::array::forcecast> resp;
+             const double *rp = nullptr;
+             if (!responses_obj.is_none()) {
+               resp = responses_obj.cast<
+                   py::array_t<double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)resp.size() != n * e.eng->nscores)
+                 throw std::runtime_error("walk_raw: responses size must be n*nscores");
+               rp = resp.data();
+             }
              auto out_pos = py::array_t<double>({n, (int64_t)3});
              auto out_elem = py::array_t<int32_t>(n);
              auto out_status = py::array_t<int8_t>(n);
@@ -357,12 +404,13 @@ PYBIND11_MODULE(_core, m) {
                e.eng->walk_raw(n, pos.data(), dest.data(), elem.data(),
                                weights.data(), out_pos.mutable_data(),
                                out_elem.mutable_data(), out_status.mutable_data(),
-                               gp);
+                               gp, rp);
              }
              return py::make_tuple(out_pos, out_elem, out_status);
            },
            py::arg("pos"), py::arg("dest"), py::arg("elem"),
-           py::arg("weights"), py::arg("groups") = py::none())
+           py::arg("weights"), py::arg("groups") = py::none(),
+           py::arg("responses") = py::none())
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

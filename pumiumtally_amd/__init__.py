@@ -69,10 +69,12 @@ class TallyEngine:
     """
 
     def __init__(self, mesh, num_particles: int, device: str = "auto",
-                 ngroups: int = 1):
-        self._eng = _core.Engine(mesh, num_particles, device, ngroups)
+                 ngroups: int = 1, nscores: int = 1):
+        self._eng = _core.Engine(mesh, num_particles, device, ngroups,
+                                 nscores)
         self.mesh = mesh
         self.ngroups = ngroups
+        self.nscores = nscores
 
     @property
     def num_particles(self) -> int:
@@ -93,24 +95,30 @@ class TallyEngine:
     def copy_initial_position(self, positions):
         self._eng.copy_initial_position(positions)
 
-    def move(self, origin, dest, flying, weights, groups=None):
+    def move(self, origin, dest, flying, weights, groups=None,
+             responses=None):
         """One transport step.  groups (optional): per-particle energy-group
         indices (uint16, in [0, ngroups)); contributions land in
-        flux()[group, elem].  The reference has a single scalar tally;
-        ngroups=1 (default) matches it exactly."""
-        self._eng.move(origin, dest, flying, weights, groups)
+        flux()[group, elem].  responses (optional, n x nscores float64):
+        per-particle score multipliers -- score k tallies
+        seg*weight*responses[i,k] (flux + heating + ... from one walk).
+        The reference has a single scalar tally; ngroups=nscores=1
+        (default) matches it exactly."""
+        self._eng.move(origin, dest, flying, weights, groups, responses)
 
-    def move_continue(self, dest, flying, weights):
+    def move_continue(self, dest, flying, weights, groups=None,
+                      responses=None):
         """move() without the phase-A origin upload: valid when no particle
         was resampled this step (origin == committed position)."""
-        self._eng.move_continue(dest, flying, weights)
+        self._eng.move_continue(dest, flying, weights, groups, responses)
 
     def move_from_device(self, dest, flying, weights, origin=None,
-                         sync_torch=True):
-        """Device-resident move: dest/flying/weights (and optionally origin)
-        are GPU tensors (torch CUDA tensors or anything with data_ptr()
-        semantics via __cuda_array_interface__) already on this engine's
-        device -- no host staging.  For GPU-side transport codes."""
+                         sync_torch=True, groups=None, responses=None):
+        """Device-resident move: dest/flying/weights (and optionally origin,
+        groups, responses) are GPU tensors (torch CUDA tensors or anything
+        with data_ptr() semantics via __cuda_array_interface__) already on
+        this engine's device -- no host staging.  For GPU-side transport
+        codes."""
         def ptr(t, dtype, numel):
             if t is None:
                 return 0
@@ -146,21 +154,32 @@ class TallyEngine:
                 torch.cuda.synchronize()
         self._eng.move_device(
             ptr(origin, "<f8", n * 3), ptr(dest, "<f8", n * 3),
-            ptr(flying, "|i1", n), ptr(weights, "<f8", n))
+            ptr(flying, "|i1", n), ptr(weights, "<f8", n),
+            ptr(groups, "<u2", n), ptr(responses, "<f8", n * self.nscores))
 
-    def walk_raw(self, pos, dest, elem, weights, groups=None):
+    def walk_raw(self, pos, dest, elem, weights, groups=None,
+                 responses=None):
         """Batched raw segment walk (domain-decomposition support): returns
         (out_pos, out_elem, status) with status 0=done 1=escaped 2=handoff
         3=lost; tallies into this engine's flux.  groups: optional uint16
-        per-segment energy-group indices (flux row group*nelems+elem)."""
-        return self._eng.walk_raw(pos, dest, elem, weights, groups)
+        per-segment energy-group indices (flux row group*nelems+elem);
+        responses: optional n x nscores score multipliers."""
+        return self._eng.walk_raw(pos, dest, elem, weights, groups,
+                                  responses)
 
     def synchronize(self):
         self._eng.synchronize()
 
     def flux(self):
-        """Raw tally: shape (nelems,) for ngroups=1, else (ngroups, nelems)."""
+        """Raw tally.  Shape: (nelems,) for ngroups=nscores=1;
+        (ngroups, nelems) for grouped single-score;
+        (nscores, nelems) for scored single-group;
+        (nscores, ngroups, nelems) for both."""
         f = self._eng.flux()
+        if self.nscores > 1 and self.ngroups > 1:
+            return f.reshape(self.nscores, self.ngroups, self.mesh.nelems)
+        if self.nscores > 1:
+            return f.reshape(self.nscores, self.mesh.nelems)
         if self.ngroups > 1:
             return f.reshape(self.ngroups, self.mesh.nelems)
         return f
@@ -202,10 +221,15 @@ class TallyEngine:
         sem = np.sqrt(var / max(nb - 1, 1))
         rel = np.divide(sem, np.abs(mean), out=np.zeros_like(sem),
                         where=mean != 0)
-        if self.ngroups > 1:
+        if self.nscores > 1 and self.ngroups > 1:
+            shape = (self.nscores, self.ngroups, self.mesh.nelems)
+        elif self.nscores > 1:
+            shape = (self.nscores, self.mesh.nelems)
+        elif self.ngroups > 1:
             shape = (self.ngroups, self.mesh.nelems)
-            return mean.reshape(shape), rel.reshape(shape)
-        return mean, rel
+        else:
+            return mean, rel
+        return mean.reshape(shape), rel.reshape(shape)
 
     def save_checkpoint(self, path: str):
         """Persist the full tally state (flux accumulator + particle
@@ -248,13 +272,28 @@ class TallyEngine:
         return _core.normalize_flux(self.mesh, f)
 
     def write_tally_results(self, filename: str = "fluxresult.vtk"):
+        if self.nscores > 1:
+            # one field per score (score 0 keeps the name "flux"),
+            # plus per-group fields when also grouped
+            f = self.flux().reshape(self.nscores, self.ngroups,
+                                    self.mesh.nelems)
+            fields = []
+            for k in range(self.nscores):
+                name = "flux" if k == 0 else f"score{k}"
+                fields.append(
+                    (name, _core.normalize_flux(self.mesh, f[k].sum(axis=0))))
+                if self.ngroups > 1:
+                    fields += [(f"{name}_g{g}",
+                                _core.normalize_flux(self.mesh, f[k, g]))
+                               for g in range(self.ngroups)]
+            self.mesh.write_vtk_fields(filename, fields)
+            return
         if self.ngroups > 1:
             # one normalized field per energy group + the total
             f = self.flux()
             fields = [("flux", _core.normalize_flux(self.mesh, f.sum(axis=0)))]
             fields += [(f"flux_g{g}", _core.normalize_flux(self.mesh, f[g]))
                        for g in range(self.ngroups)]
-            import numpy as np
             self.mesh.write_vtk_fields(filename, fields)
             return
         _core.write_tally_vtk(filename, self.mesh, self._eng.flux())

@@ -17,8 +17,9 @@ mesh on every rank, owners all rank 0) -- with a real decomposition.
 Exchange collective: torch.distributed all_to_all_single over RCCL/xGMI
 when on GPU (nccl backend), all_gather_object on gloo (CPU tests).
 
-Exchange record (9 float64 per handed-off particle): position[3],
-destination[3], weight, target global element id, energy group.
+Exchange record (9+nscores float64 per handed-off particle when
+responses are used, else 9): position[3], destination[3], weight,
+target global element id, energy group[, response multipliers].
 
 Load balance: pass elem_weights to the constructor (per-element work
 estimates -- e.g. a previous batch's raw flux) to split the Morton curve
@@ -33,16 +34,18 @@ import numpy as np
 
 from .dist import init_distributed
 
-_REC = 9  # floats per exchange record
+_BASE_REC = 9  # floats per exchange record (before response columns)
 
 
 class PartitionedTally:
     def __init__(self, mesh, device=None, backend=None, max_rounds: int = 64,
-                 ghost_rings: int = 1, ngroups: int = 1, elem_weights=None):
+                 ghost_rings: int = 1, ngroups: int = 1, nscores: int = 1,
+                 elem_weights=None):
         self.rank, self.world, self.local = init_distributed(backend)
         self.mesh = mesh
         self.max_rounds = max_rounds
         self.ngroups = max(1, int(ngroups))
+        self.nscores = max(1, int(nscores))
         self.ghost_rings = ghost_rings
         self._device = device
         self._build(elem_weights)
@@ -65,7 +68,8 @@ class PartitionedTally:
             device = f"cuda:{self.local}" if have_gpu() else "cpu"
         # Engine over the local submesh; used only through walk_raw + flux.
         self.engine = TallyEngine(self.sub.local, 1, device=device,
-                                  ngroups=self.ngroups)
+                                  ngroups=self.ngroups,
+                                  nscores=self.nscores)
 
     def repartition(self, elem_weights):
         """Rebuild the decomposition from per-element work estimates
@@ -79,12 +83,12 @@ class PartitionedTally:
         self._build(elem_weights)
 
     # -- helpers -----------------------------------------------------------
-    def _exchange(self, records_per_rank):
-        """records_per_rank: list of world np.float64 arrays (k,_REC);
+    def _exchange(self, records_per_rank, rec_w):
+        """records_per_rank: list of world np.float64 arrays (k, rec_w);
         returns concatenated records received from all ranks."""
         if self.world == 1:
             return (records_per_rank[0] if records_per_rank
-                    else np.zeros((0, _REC)))
+                    else np.zeros((0, rec_w)))
         import torch
         import torch.distributed as dist
 
@@ -101,22 +105,25 @@ class PartitionedTally:
             recv = torch.empty(int(sum(out_counts)), dtype=torch.float64,
                                device=counts.device)
             dist.all_to_all_single(recv, send, out_counts, in_counts)
-            return recv.cpu().numpy().reshape(-1, _REC)
+            return recv.cpu().numpy().reshape(-1, rec_w)
         # gloo: object all_gather
         gathered = [None] * self.world
         dist.all_gather_object(gathered, [np.asarray(r) for r in records_per_rank])
         mine = [g[self.rank] for g in gathered if g[self.rank].size]
-        return (np.concatenate(mine).reshape(-1, _REC) if mine
-                else np.zeros((0, _REC)))
+        return (np.concatenate(mine).reshape(-1, rec_w) if mine
+                else np.zeros((0, rec_w)))
 
     # -- public API --------------------------------------------------------
-    def run_segments(self, origins, dests, weights, groups=None):
+    def run_segments(self, origins, dests, weights, groups=None,
+                     responses=None):
         """Walk one batch of global segments (origins->dests, weights),
         tallying into the partitioned flux.  Each rank passes the SAME
         global arrays (or its own shard -- ownership is resolved here);
         segments starting outside this rank's elements are ignored locally
         and handled by their owner.  groups: optional per-segment energy
-        group indices (requires ngroups>1 at construction)."""
+        group indices (requires ngroups>1 at construction).  responses:
+        optional (n, nscores) per-segment score multipliers (they ride the
+        exchange record across ranks)."""
         origins = np.asarray(origins, np.float64).reshape(-1, 3)
         dests = np.asarray(dests, np.float64).reshape(-1, 3)
         weights = np.asarray(weights, np.float64).reshape(-1)
@@ -124,19 +131,26 @@ class PartitionedTally:
             groups = np.asarray(groups, np.uint16).reshape(-1)
             if groups.size != weights.size:
                 raise ValueError("groups size mismatch")
+        if responses is not None:
+            responses = np.asarray(responses, np.float64).reshape(
+                -1, self.nscores)
+            if responses.shape[0] != weights.size:
+                raise ValueError("responses size mismatch")
+        rec_w = _BASE_REC + (self.nscores if responses is not None else 0)
         gids = self.mesh.locate(origins)
         mine = (gids >= 0) & (self.owners[np.maximum(gids, 0)] == self.rank)
         pos = origins[mine]
         dst = dests[mine]
         wgt = weights[mine]
         grp = groups[mine] if groups is not None else None
+        rsp = responses[mine] if responses is not None else None
         elem = self.g2l[gids[mine]].astype(np.int32)
 
         for _round in range(self.max_rounds):
-            outbound = [np.zeros((0, _REC)) for _ in range(self.world)]
+            outbound = [np.zeros((0, rec_w)) for _ in range(self.world)]
             if len(elem):
                 out_pos, out_elem, status = self.engine.walk_raw(
-                    pos.ravel(), dst.ravel(), elem, wgt, grp)
+                    pos.ravel(), dst.ravel(), elem, wgt, grp, rsp)
                 hand = status == 2
                 if hand.any():
                     k = -(out_elem[hand].astype(np.int64) + 2)
@@ -144,17 +158,18 @@ class PartitionedTally:
                     tgt_owner = self.foreign_owner[k]
                     g_col = (grp[hand] if grp is not None
                              else np.zeros(int(hand.sum()), np.uint16))
-                    rec = np.concatenate(
-                        [out_pos[hand], dst[hand],
-                         wgt[hand, None], tgt_gid[:, None].astype(np.float64),
-                         g_col[:, None].astype(np.float64)],
-                        axis=1)
+                    cols = [out_pos[hand], dst[hand],
+                            wgt[hand, None], tgt_gid[:, None].astype(np.float64),
+                            g_col[:, None].astype(np.float64)]
+                    if rsp is not None:
+                        cols.append(rsp[hand])
+                    rec = np.concatenate(cols, axis=1)
                     for r in range(self.world):
                         sel = tgt_owner == r
                         if sel.any():
                             outbound[r] = rec[sel]
             # global termination: exchange; empty everywhere -> done
-            inbound = self._exchange(outbound)
+            inbound = self._exchange(outbound, rec_w)
             if self.world > 1:
                 import torch
                 import torch.distributed as dist
@@ -175,18 +190,21 @@ class PartitionedTally:
             elem = self.g2l[inbound[:, 7].astype(np.int64)].astype(np.int32)
             grp = (inbound[:, 8].astype(np.uint16)
                    if groups is not None else None)
+            rsp = (np.ascontiguousarray(inbound[:, 9:9 + self.nscores])
+                   if responses is not None else None)
         else:
             raise RuntimeError("partitioned walk did not converge "
                                f"in {self.max_rounds} handoff rounds")
 
     def flux_global(self) -> np.ndarray:
         """Scatter the local tally to global element ids and sum over ranks.
-        Shape (nelems,) for ngroups=1 else (ngroups, nelems)."""
+        Shape mirrors TallyEngine.flux(): (nelems,) plain, (ngroups, nelems)
+        grouped, (nscores, nelems) scored, (nscores, ngroups, nelems) both."""
         local = np.asarray(self.engine.flux())
-        ng = self.ngroups
+        ns, ng = self.nscores, self.ngroups
         nloc = len(self.l2g)
-        out = np.zeros((ng, self.mesh.nelems))
-        out[:, self.l2g] = local.reshape(ng, nloc)
+        out = np.zeros((ns, ng, self.mesh.nelems))
+        out[:, :, self.l2g] = local.reshape(ns, ng, nloc)
         if self.world > 1:
             import torch
             import torch.distributed as dist
@@ -195,14 +213,33 @@ class PartitionedTally:
                 t = t.cuda(self.local)
             dist.all_reduce(t, op=dist.ReduceOp.SUM)
             out = t.cpu().numpy()
-        return out[0] if ng == 1 else out
+        if ns == 1 and ng == 1:
+            return out[0, 0]
+        if ns == 1:
+            return out[0]
+        if ng == 1:
+            return out[:, 0]
+        return out
 
     def write_tally_results(self, filename="fluxresult.vtk"):
         from .. import _core, write_tally_vtk
 
         f = self.flux_global()
         if self.rank == 0:
-            if self.ngroups > 1:
+            if self.nscores > 1:
+                f2 = np.asarray(f).reshape(self.nscores, self.ngroups,
+                                           self.mesh.nelems)
+                fields = []
+                for k in range(self.nscores):
+                    name = "flux" if k == 0 else f"score{k}"
+                    fields.append((name, _core.normalize_flux(
+                        self.mesh, f2[k].sum(axis=0))))
+                    if self.ngroups > 1:
+                        fields += [(f"{name}_g{g}", _core.normalize_flux(
+                                        self.mesh, f2[k, g]))
+                                   for g in range(self.ngroups)]
+                self.mesh.write_vtk_fields(filename, fields)
+            elif self.ngroups > 1:
                 fields = [("flux", _core.normalize_flux(self.mesh, f.sum(axis=0)))]
                 fields += [(f"flux_g{g}", _core.normalize_flux(self.mesh, f[g]))
                            for g in range(self.ngroups)]
