@@ -173,14 +173,20 @@ public:
     };
     const uint32_t *bc =
         mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
+    const int32_t *pix =
+        mesh_.periodic_idx.empty() ? nullptr : mesh_.periodic_idx.data();
     if (walk_fp32)
-      walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
-                     mesh_.nbr.data(), elem_[i], o, d, weights[i], steps, add,
-                     &out_elem, &out_pos, &out_esc, reflective, bc);
+      walk_segment32<true>(mesh_.planes.data(), mesh_.planes32.data(),
+                           mesh_.nbr.data(), elem_[i], o, d, weights[i],
+                           steps, add, &out_elem, &out_pos, &out_esc,
+                           reflective, bc, pix, mesh_.periodic_elem.data(),
+                           mesh_.periodic_shift.data());
     else
-      walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o, d,
-                   weights[i], steps, add, &out_elem, &out_pos, &out_esc,
-                   reflective, bc);
+      walk_segment<true>(mesh_.planes.data(), mesh_.nbr.data(), elem_[i], o,
+                         d, weights[i], steps, add, &out_elem, &out_pos,
+                         &out_esc, reflective, bc, pix,
+                         mesh_.periodic_elem.data(),
+                         mesh_.periodic_shift.data());
     if (out_elem == kWalkLost) {
       lost++;
       out_elem = elem_[i];
@@ -255,13 +261,19 @@ public:
       };
       const uint32_t *bc =
           mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
+      const int32_t *pix =
+          mesh_.periodic_idx.empty() ? nullptr : mesh_.periodic_idx.data();
       if (walk_fp32)
-        walk_segment32(mesh_.planes.data(), mesh_.planes32.data(),
-                       mesh_.nbr.data(), elem[i], o, d, weights[i], steps, add,
-                       &oe, &op, &esc, reflective, bc);
+        walk_segment32<true>(mesh_.planes.data(), mesh_.planes32.data(),
+                             mesh_.nbr.data(), elem[i], o, d, weights[i],
+                             steps, add, &oe, &op, &esc, reflective, bc, pix,
+                             mesh_.periodic_elem.data(),
+                             mesh_.periodic_shift.data());
       else
-        walk_segment(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o, d,
-                     weights[i], steps, add, &oe, &op, &esc, reflective, bc);
+        walk_segment<true>(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o,
+                           d, weights[i], steps, add, &oe, &op, &esc,
+                           reflective, bc, pix, mesh_.periodic_elem.data(),
+                           mesh_.periodic_shift.data());
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;

@@ -19,6 +19,77 @@ Vec3 Mesh::centroid(int32_t t) const {
   return 0.25 * c;
 }
 
+namespace {
+inline Vec3 face_centroid(const Mesh &m, int64_t fidx) {
+  const int32_t t = (int32_t)(fidx / 4);
+  const int f = (int)(fidx % 4);
+  Vec3 c{0, 0, 0};
+  for (int k = 0; k < 3; ++k)
+    c = c + m.vert(m.tet2vert[(int64_t)t * 4 + kFaceVerts[f][k]]);
+  return (1.0 / 3.0) * c;
+}
+} // namespace
+
+void Mesh::set_periodic_faces(const std::vector<int64_t> &faces_a,
+                              const std::vector<int64_t> &faces_b,
+                              Vec3 translation, double tol) {
+  if (nbr.empty())
+    throw std::runtime_error("set_periodic_faces: call finalize() first");
+  if (faces_a.size() != faces_b.size())
+    throw std::runtime_error("set_periodic_faces: face lists differ in size");
+  if (tol <= 0.0) tol = 1e-8 * norm(bbox_hi - bbox_lo);
+  for (int64_t fidx : faces_a)
+    if (fidx < 0 || fidx >= nelems * 4 || nbr[fidx] != -1)
+      throw std::runtime_error("set_periodic_faces: face in A is not a "
+                               "boundary face: " + std::to_string(fidx));
+  for (int64_t fidx : faces_b)
+    if (fidx < 0 || fidx >= nelems * 4 || nbr[fidx] != -1)
+      throw std::runtime_error("set_periodic_faces: face in B is not a "
+                               "boundary face: " + std::to_string(fidx));
+  // Match face a + T to face b by centroid: sort B by x and scan a window.
+  struct Entry { double x; Vec3 c; int64_t fidx; };
+  std::vector<Entry> bs;
+  bs.reserve(faces_b.size());
+  for (int64_t fidx : faces_b) {
+    const Vec3 c = face_centroid(*this, fidx);
+    bs.push_back({c.x, c, fidx});
+  }
+  std::sort(bs.begin(), bs.end(),
+            [](const Entry &l, const Entry &r) { return l.x < r.x; });
+  if (periodic_idx.empty()) periodic_idx.assign(nelems * 4, -1);
+  std::vector<char> taken(bs.size(), 0);
+  for (int64_t fa : faces_a) {
+    const Vec3 want = face_centroid(*this, fa) + translation;
+    auto lo = std::lower_bound(
+        bs.begin(), bs.end(), want.x - tol,
+        [](const Entry &e, double v) { return e.x < v; });
+    int64_t fb = -1;
+    for (auto it = lo; it != bs.end() && it->x <= want.x + tol; ++it) {
+      if (taken[it - bs.begin()]) continue;
+      if (norm(it->c - want) <= tol) {
+        fb = it->fidx;
+        taken[it - bs.begin()] = 1;
+        break;
+      }
+    }
+    if (fb < 0)
+      throw std::runtime_error(
+          "set_periodic_faces: no face in B matches face " +
+          std::to_string(fa) + " translated by T (geometry mismatch?)");
+    // a -> elem(b) with +T; b -> elem(a) with -T
+    const int32_t ka = (int32_t)periodic_elem.size();
+    periodic_elem.push_back((int32_t)(fb / 4));
+    periodic_shift.insert(periodic_shift.end(),
+                          {translation.x, translation.y, translation.z});
+    periodic_idx[fa] = ka;
+    const int32_t kb = (int32_t)periodic_elem.size();
+    periodic_elem.push_back((int32_t)(fa / 4));
+    periodic_shift.insert(periodic_shift.end(),
+                          {-translation.x, -translation.y, -translation.z});
+    periodic_idx[fb] = kb;
+  }
+}
+
 bool Mesh::contains(int32_t t, Vec3 p, double tol) const {
   for (int f = 0; f < 4; ++f)
     if (plane_eval(planes[t * 4 + f], p) < -tol) return false;

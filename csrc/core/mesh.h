@@ -61,6 +61,22 @@ struct Mesh {
            ((face_bc_bits[w] >> (face_index & 31)) & 1u);
   }
 
+  // Optional periodic boundary pairing: boundary face a (elem*4+f) maps to
+  // an entry element + translation vector; a walk hitting the face
+  // teleports its remaining segment by the translation and resumes there
+  // (walk.h periodic_restart).  Built by set_periodic_faces(a, b, T):
+  // every face in `a`, translated by T, must coincide with a face in `b`
+  // (matched by face centroid to tol); pairing is installed both ways
+  // (b gets -T).  Call AFTER finalize() and BEFORE engine construction.
+  // Not supported in partitioned submeshes (PartitionedTally raises).
+  std::vector<int32_t> periodic_idx;   // nelems*4, -1 = no pair; empty = none
+  std::vector<int32_t> periodic_elem;  // pair entry -> entry element
+  std::vector<double> periodic_shift;  // pair entry -> translation (x,y,z)
+  void set_periodic_faces(const std::vector<int64_t> &faces_a,
+                          const std::vector<int64_t> &faces_b,
+                          Vec3 translation, double tol = -1.0);
+  bool has_periodic() const { return !periodic_elem.empty(); }
+
   // Derived (built by finalize()):
   std::vector<int32_t> nbr;     // nelems*4: neighbor tet across face f, -1 = boundary
   std::vector<Plane> planes;    // nelems*4: inward-positive unit-normal face planes

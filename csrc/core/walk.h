@@ -66,13 +66,39 @@ PT_HD Vec3 reflect_point(const Plane &pl, Vec3 p) {
   return {p.x - 2.0 * v * pl.nx, p.y - 2.0 * v * pl.ny, p.z - 2.0 * v * pl.nz};
 }
 
-template <class FluxAdd>
+// Periodic-translation helper shared by both advance variants: teleport the
+// remaining segment by the pair's translation vector and resume in the
+// paired element (a translation is an isometry, so total tallied length is
+// conserved, exactly like the reflective restart).
+PT_HD inline void periodic_restart(WalkState &s, double t_clamped,
+                                   int32_t pair_elem, const double *shift) {
+  const Vec3 hit = s.o + t_clamped * (s.d - s.o);
+  const Vec3 T{shift[0], shift[1], shift[2]};
+  const double remaining = (1.0 - t_clamped) * s.seg_len;
+  s.o = hit + T;
+  s.d = s.d + T;
+  s.of = Vec3f{(float)s.o.x, (float)s.o.y, (float)s.o.z};
+  s.df = Vec3f{(float)s.d.x, (float)s.d.y, (float)s.d.z};
+  s.seg_len = remaining > 0.0 ? remaining : 0.0;
+  s.t_cur = 0.0;
+  s.prev_elem = -1;
+  s.elem = pair_elem;
+}
+
+// Periodic=false (the default, and the headline GPU instantiation) compiles
+// the pairing arguments away entirely -- codegen identical to the
+// pre-periodic kernel.  Periodic=true callers may still pass pidx=nullptr
+// (no pairs on this mesh); the check is null-safe.
+template <bool Periodic = false, class FluxAdd>
 PT_HD bool walk_advance(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, WalkState &s,
                         int max_steps, FluxAdd &&add, int32_t *out_elem,
                         Vec3 *out_pos, bool *out_escaped,
                         bool reflective = false,
-                        const uint32_t *__restrict__ face_bc = nullptr) {
+                        const uint32_t *__restrict__ face_bc = nullptr,
+                        const int32_t *__restrict__ pidx = nullptr,
+                        const int32_t *__restrict__ pelem = nullptr,
+                        const double *__restrict__ pshift = nullptr) {
   if (s.step++ >= max_steps) {
     // Did not converge (numerically stuck / absurd chord): drop here and
     // flag as lost (reference prints "Not all particles are found",
@@ -131,6 +157,13 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
     const int64_t fidx = (int64_t)s.elem * 4 + exit_face;
+    if constexpr (Periodic) {
+      if (pidx && pidx[fidx] >= 0) {
+        const int32_t pk = pidx[fidx];
+        periodic_restart(s, t_clamped, pelem[pk], pshift + (int64_t)pk * 3);
+        return false;
+      }
+    }
     const bool refl_here =
         reflective ||
         (face_bc && ((face_bc[fidx >> 5] >> (fidx & 31)) & 1u));
@@ -186,14 +219,17 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
 // monotone-t clamp.
 constexpr float kWalkTEps32 = 1e-6f;
 
-template <class FluxAdd>
+template <bool Periodic = false, class FluxAdd>
 PT_HD bool walk_advance32(const Plane *__restrict__ planes,
                           const Plane32 *__restrict__ planes32,
                           const int32_t *__restrict__ nbr, WalkState &s,
                           int max_steps, FluxAdd &&add, int32_t *out_elem,
                           Vec3 *out_pos, bool *out_escaped,
                           bool reflective = false,
-                          const uint32_t *__restrict__ face_bc = nullptr) {
+                          const uint32_t *__restrict__ face_bc = nullptr,
+                          const int32_t *__restrict__ pidx = nullptr,
+                          const int32_t *__restrict__ pelem = nullptr,
+                          const double *__restrict__ pshift = nullptr) {
   if (s.step++ >= max_steps) {
     *out_elem = kWalkLost;
     *out_pos = s.o + s.t_cur * (s.d - s.o);
@@ -244,6 +280,13 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
   const int32_t next = nbr[(int64_t)s.elem * 4 + exit_face];
   if (next == -1) {
     const int64_t fidx = (int64_t)s.elem * 4 + exit_face;
+    if constexpr (Periodic) {
+      if (pidx && pidx[fidx] >= 0) {
+        const int32_t pk = pidx[fidx];
+        periodic_restart(s, t_clamped, pelem[pk], pshift + (int64_t)pk * 3);
+        return false;
+      }
+    }
     const bool refl_here =
         reflective ||
         (face_bc && ((face_bc[fidx >> 5] >> (fidx & 31)) & 1u));
@@ -280,32 +323,40 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
 
 // FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
 // performs atomicAdd into the flux array; on the serial CPU path a plain +=.
-template <class FluxAdd>
+template <bool Periodic = false, class FluxAdd>
 PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
                         Vec3 d, double weight, int max_steps, FluxAdd &&add,
                         int32_t *out_elem, Vec3 *out_pos, bool *out_escaped,
                         bool reflective = false,
-                        const uint32_t *__restrict__ face_bc = nullptr) {
+                        const uint32_t *__restrict__ face_bc = nullptr,
+                        const int32_t *__restrict__ pidx = nullptr,
+                        const int32_t *__restrict__ pelem = nullptr,
+                        const double *__restrict__ pshift = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
-  while (!walk_advance(planes, nbr, s, max_steps, add, out_elem, out_pos,
-                       out_escaped, reflective, face_bc)) {
+  while (!walk_advance<Periodic>(planes, nbr, s, max_steps, add, out_elem,
+                                 out_pos, out_escaped, reflective, face_bc,
+                                 pidx, pelem, pshift)) {
   }
 }
 
-template <class FluxAdd>
+template <bool Periodic = false, class FluxAdd>
 PT_HD void walk_segment32(const Plane *__restrict__ planes,
                           const Plane32 *__restrict__ planes32,
                           const int32_t *__restrict__ nbr, int32_t elem,
                           Vec3 o, Vec3 d, double weight, int max_steps,
                           FluxAdd &&add, int32_t *out_elem, Vec3 *out_pos,
                           bool *out_escaped, bool reflective = false,
-                          const uint32_t *__restrict__ face_bc = nullptr) {
+                          const uint32_t *__restrict__ face_bc = nullptr,
+                          const int32_t *__restrict__ pidx = nullptr,
+                          const int32_t *__restrict__ pelem = nullptr,
+                          const double *__restrict__ pshift = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
-  while (!walk_advance32(planes, planes32, nbr, s, max_steps, add, out_elem,
-                         out_pos, out_escaped, reflective, face_bc)) {
+  while (!walk_advance32<Periodic>(planes, planes32, nbr, s, max_steps, add,
+                                   out_elem, out_pos, out_escaped, reflective,
+                                   face_bc, pidx, pelem, pshift)) {
   }
 }
 

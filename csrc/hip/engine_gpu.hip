@@ -31,6 +31,7 @@
 #include <hipcub/hipcub.hpp>
 
 #include <algorithm>
+#include <type_traits>
 #include <array>
 #include <cstdio>
 #include <cstdlib>
@@ -186,7 +187,7 @@ __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
 // dispatches block b to XCD b%8 and each XCD has a private 4 MiB L2; we
 // remap blocks so each XCD owns one contiguous (hence spatially compact)
 // slot range.  Purely a speed lever: any placement is correct.
-template <bool F32, bool Scored = false>
+template <bool F32, bool Scored = false, bool Periodic = false>
 __global__ void k_move(const Plane *__restrict__ planes,
                        const Plane32 *__restrict__ planes32,
                        const int32_t *__restrict__ nbr, GridView grid,
@@ -204,10 +205,14 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        int64_t nelems, int slice_mask, bool reflective,
                        const uint32_t *__restrict__ face_bc,
                        const double *__restrict__ resp = nullptr,
-                       int nscores = 1) {
-  // Scored=false is the headline instantiation: resp/nscores are unused and
-  // the slice stride stays nelems*ngroups -- codegen identical to before
-  // multi-score existed.
+                       int nscores = 1,
+                       const int32_t *__restrict__ pidx = nullptr,
+                       const int32_t *__restrict__ pelem = nullptr,
+                       const double *__restrict__ pshift = nullptr) {
+  // Scored=false, Periodic=false is the headline instantiation:
+  // resp/nscores/pidx are unused, the slice stride stays nelems*ngroups,
+  // and the walk compiles without the score loop or the periodic branch --
+  // codegen identical to before those features existed.
   flux += (int64_t)(blockIdx.x & (unsigned)slice_mask) * nelems * ngroups *
           (Scored ? nscores : 1);
   const unsigned bpx = gridDim.x / 8u;
@@ -256,11 +261,13 @@ __global__ void k_move(const Plane *__restrict__ planes,
       }
     };
     if constexpr (F32)
-      walk_segment32(planes, planes32, nbr, e, o, d, weights[c], max_steps,
-                     add, &out_elem, &out_pos, &out_esc, reflective, face_bc);
+      walk_segment32<Periodic>(planes, planes32, nbr, e, o, d, weights[c],
+                               max_steps, add, &out_elem, &out_pos, &out_esc,
+                               reflective, face_bc, pidx, pelem, pshift);
     else
-      walk_segment(planes, nbr, e, o, d, weights[c], max_steps, add,
-                   &out_elem, &out_pos, &out_esc, reflective, face_bc);
+      walk_segment<Periodic>(planes, nbr, e, o, d, weights[c], max_steps,
+                             add, &out_elem, &out_pos, &out_esc, reflective,
+                             face_bc, pidx, pelem, pshift);
     if (out_elem == kWalkLost) {
       atomicAdd(lost, 1ull);
       out_elem = e;
@@ -290,7 +297,10 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            unsigned long long *__restrict__ lost, int64_t n,
                            int max_steps, bool reflective,
                            const uint32_t *__restrict__ face_bc, int ngroups,
-                           int64_t nelems, int nscores) {
+                           int64_t nelems, int nscores,
+                           const int32_t *__restrict__ pidx = nullptr,
+                           const int32_t *__restrict__ pelem = nullptr,
+                           const double *__restrict__ pshift = nullptr) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
@@ -309,11 +319,13 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
       }
     };
     if constexpr (F32)
-      walk_segment32(planes, planes32, nbr, elem[i], o, d, weights[i],
-                     max_steps, add, &oe, &op, &esc, reflective, face_bc);
+      walk_segment32<true>(planes, planes32, nbr, elem[i], o, d, weights[i],
+                           max_steps, add, &oe, &op, &esc, reflective,
+                           face_bc, pidx, pelem, pshift);
     else
-      walk_segment(planes, nbr, elem[i], o, d, weights[i], max_steps, add,
-                   &oe, &op, &esc, reflective, face_bc);
+      walk_segment<true>(planes, nbr, elem[i], o, d, weights[i], max_steps,
+                         add, &oe, &op, &esc, reflective, face_bc, pidx,
+                         pelem, pshift);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -422,6 +434,20 @@ public:
                              mesh_.face_bc_bits.size() * sizeof(uint32_t),
                              hipMemcpyHostToDevice));
     }
+    if (mesh_.has_periodic()) {
+      d_pidx_ = dmalloc<int32_t>(mesh_.periodic_idx.size());
+      d_pelem_ = dmalloc<int32_t>(mesh_.periodic_elem.size());
+      d_pshift_ = dmalloc<double>(mesh_.periodic_shift.size());
+      PT_HIP_CHECK(hipMemcpy(d_pidx_, mesh_.periodic_idx.data(),
+                             mesh_.periodic_idx.size() * 4,
+                             hipMemcpyHostToDevice));
+      PT_HIP_CHECK(hipMemcpy(d_pelem_, mesh_.periodic_elem.data(),
+                             mesh_.periodic_elem.size() * 4,
+                             hipMemcpyHostToDevice));
+      PT_HIP_CHECK(hipMemcpy(d_pshift_, mesh_.periodic_shift.data(),
+                             mesh_.periodic_shift.size() * 8,
+                             hipMemcpyHostToDevice));
+    }
     grid_view_ = GridView{mesh_.grid.nx, mesh_.grid.ny, mesh_.grid.nz,
                           mesh_.grid.lo,  mesh_.grid.inv_h,
                           d_cell_start_,  d_cell_tets_};
@@ -497,6 +523,7 @@ public:
                     (void *)d_groups_[0], (void *)d_groups_[1],
                     (void *)d_resp_[0], (void *)d_resp_[1],
                     (void *)d_bsum_, (void *)d_bsq_, (void *)d_face_bc_,
+                    (void *)d_pidx_, (void *)d_pelem_, (void *)d_pshift_,
                     (void *)d_keys_, (void *)d_keys2_, (void *)d_vals_,
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
                     (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
@@ -608,12 +635,12 @@ public:
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
           dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores);
+          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
           dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores);
+          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
@@ -748,46 +775,45 @@ private:
     PT_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, s));
   }
 
+  template <bool F32, bool Scored, bool Periodic>
+  void launch_move_one(const double *origin, const double *dest,
+                       const int8_t *flying, const double *weights,
+                       const uint16_t *groups, const double *resp,
+                       int64_t lo, int64_t hi, int steps) {
+    k_move<F32, Scored, Periodic><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+        d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
+        flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
+        d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1,
+        reflective, d_face_bc_, resp, nscores, d_pidx_, d_pelem_, d_pshift_);
+    PT_HIP_CHECK(hipGetLastError());
+  }
+
   void launch_move_chunks(const double *origin, const double *dest,
                           const int8_t *flying, const double *weights,
                           const uint16_t *groups, const double *resp,
                           int64_t n, int steps) {
     // Chunked launches: a ~2.6M-slot launch keeps each XCD's Morton-
     // contiguous slot range's mesh working set inside its private L2.
-    // Scored moves take the k_move<_,true> instantiation; the unscored
-    // (headline) instantiation is compiled without any score machinery.
+    // Scored / periodic moves take dedicated k_move instantiations; the
+    // plain (headline) instantiation compiles without either feature.
     const int64_t chunk = chunk_particles(n);
+    const bool per = d_pidx_ != nullptr;
     for (int64_t lo = 0; lo < n; lo += chunk) {
       const int64_t hi = std::min(n, lo + chunk);
-      const int blocks = grid_blocks(hi - lo);
-      if (resp) {
-        if (walk_fp32)
-          k_move<true, true><<<blocks, kBlock, 0, s_comp_>>>(
-              d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin,
-              dest, flying, weights, groups, ngroups, d_pos_, d_elem_,
-              d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_, steps,
-              mesh_.nelems, slices_ - 1, reflective, d_face_bc_, resp,
-              nscores);
-        else
-          k_move<false, true><<<blocks, kBlock, 0, s_comp_>>>(
-              d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin,
-              dest, flying, weights, groups, ngroups, d_pos_, d_elem_,
-              d_escaped_, d_flux_, d_lost_, lo, hi, loc_tol_, steps,
-              mesh_.nelems, slices_ - 1, reflective, d_face_bc_, resp,
-              nscores);
-      } else if (walk_fp32)
-        k_move<true><<<blocks, kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
-            flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
-            d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1, reflective, d_face_bc_);
-      else
-        k_move<false><<<blocks, kBlock, 0, s_comp_>>>(
-            d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
-            flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
-            d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems,
-            slices_ - 1, reflective, d_face_bc_);
-      PT_HIP_CHECK(hipGetLastError());
+      auto go = [&](auto f32c, auto scoredc, auto perc) {
+        launch_move_one<decltype(f32c)::value, decltype(scoredc)::value,
+                        decltype(perc)::value>(origin, dest, flying, weights,
+                                               groups, resp, lo, hi, steps);
+      };
+      using T = std::true_type;
+      using F = std::false_type;
+      if (walk_fp32) {
+        if (resp) per ? go(T{}, T{}, T{}) : go(T{}, T{}, F{});
+        else      per ? go(T{}, F{}, T{}) : go(T{}, F{}, F{});
+      } else {
+        if (resp) per ? go(F{}, T{}, T{}) : go(F{}, T{}, F{});
+        else      per ? go(F{}, F{}, T{}) : go(F{}, F{}, F{});
+      }
     }
   }
 
@@ -865,6 +891,8 @@ private:
   double *d_resp_[2] = {nullptr, nullptr};
   double *d_bsum_ = nullptr, *d_bsq_ = nullptr;
   uint32_t *d_face_bc_ = nullptr;
+  int32_t *d_pidx_ = nullptr, *d_pelem_ = nullptr;
+  double *d_pshift_ = nullptr;
   int64_t nbatches_ = 0;
   double *d_dest_[2] = {nullptr, nullptr};
   int8_t *d_flying_[2] = {nullptr, nullptr};

@@ -188,7 +188,26 @@ PYBIND11_MODULE(_core, m) {
            [](Mesh &m_, py::array_t<int64_t, py::array::c_style | py::array::forcecast> fids) {
              for (py::ssize_t i = 0; i < fids.size(); ++i)
                m_.set_face_reflective(fids.data()[i]);
-           });
+           })
+      .def("set_periodic_faces",
+           // Pair every boundary face in A with the boundary face in B at
+           // centroid_A + translation; walks leaving through one re-enter
+           // through the other (remaining segment translated).
+           [](Mesh &m_, py::array_t<int64_t, py::array::c_style | py::array::forcecast> a,
+              py::array_t<int64_t, py::array::c_style | py::array::forcecast> b,
+              py::array_t<double, py::array::c_style | py::array::forcecast> t,
+              double tol) {
+             if (t.size() != 3)
+               throw std::runtime_error("translation must have 3 components");
+             m_.set_periodic_faces(
+                 std::vector<int64_t>(a.data(), a.data() + a.size()),
+                 std::vector<int64_t>(b.data(), b.data() + b.size()),
+                 Vec3{t.data()[0], t.data()[1], t.data()[2]}, tol);
+           },
+           py::arg("faces_a"), py::arg("faces_b"), py::arg("translation"),
+           py::arg("tol") = -1.0)
+      .def_property_readonly("has_periodic",
+                             [](const Mesh &m_) { return m_.has_periodic(); });
 
   py::class_<SubMesh>(m, "SubMesh")
       .def_property_readonly("local", [](const SubMesh &s) -> const Mesh & { return s.local; },

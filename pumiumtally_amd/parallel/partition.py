@@ -54,6 +54,11 @@ class PartitionedTally:
         from .. import TallyEngine, _core, have_gpu
 
         mesh = self.mesh
+        if getattr(mesh, "has_periodic", False):
+            raise NotImplementedError(
+                "periodic BCs are not supported in partitioned mode: the "
+                "handoff record does not carry the accumulated translation. "
+                "Use the replicated DistributedTally for periodic meshes.")
         w = None if elem_weights is None else np.asarray(elem_weights, np.float64)
         self.owners = _core.partition_morton(mesh, self.world, w)
         self.sub = _core.extract_submesh(mesh, self.owners, self.rank,

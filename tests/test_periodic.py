@@ -1,0 +1,182 @@
+"""Periodic boundary conditions: paired boundary faces with a translation.
+
+The reference supports only vacuum boundaries (ApplyVacuumBC,
+/root/reference/src/pumitally/PumiTallyImpl.cpp:256-286); reflective and
+periodic BCs are extensions.  A walk that exits through a periodic face
+teleports its remaining segment by the pair's translation vector and
+resumes in the paired element (walk.h periodic_restart) -- an isometry,
+so total tallied track length is conserved exactly.
+
+Oracle: a wrapped walk equals the concatenation of plain sub-walks of the
+same sub-segments (vacuum walk up to the boundary + fresh walk from the
+translated entry point).
+"""
+import numpy as np
+import pytest
+
+import pumiumtally_amd as pt
+
+
+def _periodic_x_box(nx=3, ny=3, nz=3):
+    m = pt.build_box(nx, ny, nz)
+    fid, cen, nor = m.boundary_faces()
+    hi = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    lo = fid[np.abs(cen[:, 0] - 0.0) < 1e-12]
+    m.set_periodic_faces(hi, lo, np.array([-1.0, 0.0, 0.0]))
+    assert m.has_periodic
+    return m
+
+
+def test_periodic_single_wrap_matches_split_oracle():
+    m = _periodic_x_box()
+    n = 1
+    o = np.array([[0.8, 0.41, 0.57]])
+    d = np.array([[1.3, 0.41, 0.57]])
+    w = np.array([2.0])
+
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    got = eng.flux()
+    # conservation: the whole 0.5-long segment stays inside the box
+    assert abs(got.sum() - 0.5 * w[0]) < 1e-12
+
+    # oracle: vacuum walk 0.8->1.3 (tallies [0.8, 1.0]) + plain walk of the
+    # wrapped remainder 0.0->0.3 on the plain mesh
+    plain = pt.build_box(3, 3, 3)
+    ref = pt.TallyEngine(plain, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    o2 = np.array([[0.0, 0.41, 0.57]])
+    d2 = np.array([[0.3, 0.41, 0.57]])
+    ref2 = pt.TallyEngine(plain, n, device="cpu")
+    ref2.copy_initial_position(o2.ravel())
+    ref2.move(o2.ravel(), d2.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(got, ref.flux() + ref2.flux(), atol=1e-12)
+    # final particle position is the translated destination
+    p = eng.positions().reshape(-1, 3)
+    assert np.allclose(p[0], [0.3, 0.41, 0.57], atol=1e-12)
+    # particle did NOT escape (it wrapped)
+    assert eng._eng.escaped()[0] == 0
+
+
+def test_periodic_multi_wrap_conservation():
+    m = _periodic_x_box(4, 4, 4)
+    n = 200
+    rng = np.random.default_rng(3)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    # long x-rays wrapping up to 3 times; y/z stay interior
+    d = o + np.column_stack([rng.uniform(1.5, 3.2, n),
+                             rng.uniform(-0.02, 0.02, n),
+                             rng.uniform(-0.02, 0.02, n)])
+    d[:, 1:] = np.clip(d[:, 1:], 0.05, 0.95)
+    w = rng.uniform(0.1, 1.0, n)
+    seg = np.linalg.norm(d - o, axis=1)
+
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    total = eng.flux().sum()
+    assert eng.stats()["lost_particles"] == 0
+    assert abs(total - (seg * w).sum()) < 1e-10 * (seg * w).sum()
+
+
+def test_periodic_fp32_traversal(monkeypatch):
+    monkeypatch.setenv("PUMITALLY_WALK", "fp32")
+    m = _periodic_x_box()
+    n = 50
+    rng = np.random.default_rng(8)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = o + np.column_stack([rng.uniform(0.5, 1.5, n), np.zeros(n), np.zeros(n)])
+    w = rng.uniform(0.5, 1.0, n)
+    seg = np.linalg.norm(d - o, axis=1)
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert abs(eng.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
+
+
+def test_periodic_validation_errors():
+    m = pt.build_box(2, 2, 2)
+    fid, cen, nor = m.boundary_faces()
+    hi = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    lo = fid[np.abs(cen[:, 0] - 0.0) < 1e-12]
+    # wrong translation: no geometric match
+    with pytest.raises(RuntimeError, match="no face in B matches"):
+        m.set_periodic_faces(hi, lo, np.array([-0.5, 0.0, 0.0]))
+    # non-boundary face
+    interior = -1
+    for f in range(m.nelems * 4):
+        if f not in set(fid.tolist()):
+            interior = f
+            break
+    with pytest.raises(RuntimeError, match="not a .*boundary face"):
+        m.set_periodic_faces(np.array([interior]), lo[:1],
+                             np.array([-1.0, 0.0, 0.0]))
+    # size mismatch
+    with pytest.raises(RuntimeError, match="differ in size"):
+        m.set_periodic_faces(hi, lo[:-1], np.array([-1.0, 0.0, 0.0]))
+
+
+def test_periodic_partitioned_raises():
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = _periodic_x_box()
+    with pytest.raises(NotImplementedError, match="periodic"):
+        PartitionedTally(m, device="cpu")
+
+
+def test_periodic_two_axis():
+    """x and y both periodic: diagonal rays wrap in both axes."""
+    m = pt.build_box(3, 3, 3)
+    fid, cen, nor = m.boundary_faces()
+    xhi = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    xlo = fid[np.abs(cen[:, 0] - 0.0) < 1e-12]
+    yhi = fid[np.abs(cen[:, 1] - 1.0) < 1e-12]
+    ylo = fid[np.abs(cen[:, 1] - 0.0) < 1e-12]
+    m.set_periodic_faces(xhi, xlo, np.array([-1.0, 0.0, 0.0]))
+    m.set_periodic_faces(yhi, ylo, np.array([0.0, -1.0, 0.0]))
+
+    n = 100
+    rng = np.random.default_rng(17)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = o + np.column_stack([rng.uniform(0.5, 1.8, n),
+                             rng.uniform(0.5, 1.8, n),
+                             rng.uniform(-0.05, 0.05, n)])
+    d[:, 2] = np.clip(d[:, 2], 0.05, 0.95)
+    w = rng.uniform(0.1, 1.0, n)
+    seg = np.linalg.norm(d - o, axis=1)
+
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert eng.stats()["lost_particles"] == 0
+    assert abs(eng.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
+
+
+@pytest.mark.gpu
+def test_periodic_gpu_matches_cpu():
+    m = _periodic_x_box(5, 5, 5)
+    n = 20000
+    rng = np.random.default_rng(77)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = o + np.column_stack([rng.uniform(0.5, 2.5, n),
+                             rng.uniform(-0.02, 0.02, n),
+                             rng.uniform(-0.02, 0.02, n)])
+    d[:, 1:] = np.clip(d[:, 1:], 0.05, 0.95)
+    w = rng.uniform(0.1, 1.0, n)
+    seg = np.linalg.norm(d - o, axis=1)
+
+    cpu = pt.TallyEngine(m, n, device="cpu")
+    cpu.copy_initial_position(o.ravel())
+    cpu.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+
+    gpu = pt.TallyEngine(m, n, device="cuda:0")
+    assert gpu.is_gpu
+    gpu.copy_initial_position(o.ravel())
+    gpu.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    gpu.synchronize()
+
+    assert abs(gpu.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
+    assert np.abs(cpu.flux() - gpu.flux()).max() < 1e-9
+    assert gpu.stats()["lost_particles"] == 0
