@@ -70,7 +70,7 @@ PT_HD Vec3 reflect_point(const Plane &pl, Vec3 p) {
 // remaining segment by the pair's translation vector and resume in the
 // paired element (a translation is an isometry, so total tallied length is
 // conserved, exactly like the reflective restart).
-PT_HD inline void periodic_restart(WalkState &s, double t_clamped,
+PT_HD void periodic_restart(WalkState &s, double t_clamped,
                                    int32_t pair_elem, const double *shift) {
   const Vec3 hit = s.o + t_clamped * (s.d - s.o);
   const Vec3 T{shift[0], shift[1], shift[2]};

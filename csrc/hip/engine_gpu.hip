@@ -528,6 +528,7 @@ public:
                     (void *)d_order_, (void *)d_pos2_, (void *)d_elem2_,
                     (void *)d_esc2_, (void *)d_s2c2_, d_sorttmp_})
       if (p) (void)hipFree(p);
+    free_wr();
     for (auto &ev : copy_ev_) (void)hipEventDestroy(ev);
     for (auto &ev : kernels_done_) (void)hipEventDestroy(ev);
     (void)hipStreamDestroy(s_copy_);
@@ -607,6 +608,44 @@ public:
     stats_.moves++;
   }
 
+  // Persistent scratch for walk_raw.  The partitioned driver calls
+  // walk_raw once per exchange round; per-call hipMalloc/hipFree pairs
+  // would device-sync every round, so the buffers are grown once and
+  // reused (freed in the dtor).
+  struct WalkRawScratch {
+    int64_t cap = 0;
+    double *pos = nullptr, *dest = nullptr, *w = nullptr, *out_pos = nullptr,
+           *resp = nullptr;
+    int32_t *elem = nullptr, *out_elem = nullptr;
+    int8_t *status = nullptr;
+    uint16_t *groups = nullptr;
+  };
+
+  void ensure_wr_cap(int64_t n, bool need_groups, bool need_resp) {
+    if (n > wr_.cap) {
+      free_wr();
+      wr_.cap = n + n / 4; // headroom against round-to-round growth
+      wr_.pos = dmalloc<double>(wr_.cap * 3);
+      wr_.dest = dmalloc<double>(wr_.cap * 3);
+      wr_.w = dmalloc<double>(wr_.cap);
+      wr_.out_pos = dmalloc<double>(wr_.cap * 3);
+      wr_.elem = dmalloc<int32_t>(wr_.cap);
+      wr_.out_elem = dmalloc<int32_t>(wr_.cap);
+      wr_.status = dmalloc<int8_t>(wr_.cap);
+    }
+    if (need_groups && !wr_.groups) wr_.groups = dmalloc<uint16_t>(wr_.cap);
+    if (need_resp && !wr_.resp) wr_.resp = dmalloc<double>(wr_.cap * nscores);
+  }
+
+  void free_wr() {
+    for (void *q : {(void *)wr_.pos, (void *)wr_.dest, (void *)wr_.w,
+                    (void *)wr_.out_pos, (void *)wr_.elem,
+                    (void *)wr_.out_elem, (void *)wr_.status,
+                    (void *)wr_.groups, (void *)wr_.resp})
+      if (q) (void)hipFree(q);
+    wr_ = WalkRawScratch{};
+  }
+
   void walk_raw(int64_t n, const double *pos, const double *dest,
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status,
@@ -615,41 +654,39 @@ public:
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
-    double *dp = dmalloc<double>(n * 3), *dd = dmalloc<double>(n * 3),
-           *dw = dmalloc<double>(n);
-    int32_t *de = dmalloc<int32_t>(n), *doe = dmalloc<int32_t>(n);
-    double *dop = dmalloc<double>(n * 3);
-    int8_t *dst_ = dmalloc<int8_t>(n);
-    uint16_t *dg = groups ? dmalloc<uint16_t>(n) : nullptr;
-    double *dr = responses ? dmalloc<double>(n * nscores) : nullptr;
-    PT_HIP_CHECK(hipMemcpy(dp, pos, n * 3 * 8, hipMemcpyHostToDevice));
-    PT_HIP_CHECK(hipMemcpy(dd, dest, n * 3 * 8, hipMemcpyHostToDevice));
-    PT_HIP_CHECK(hipMemcpy(dw, weights, n * 8, hipMemcpyHostToDevice));
-    PT_HIP_CHECK(hipMemcpy(de, elem, n * 4, hipMemcpyHostToDevice));
-    if (dg)
-      PT_HIP_CHECK(hipMemcpy(dg, groups, n * 2, hipMemcpyHostToDevice));
-    if (dr)
-      PT_HIP_CHECK(
-          hipMemcpy(dr, responses, n * nscores * 8, hipMemcpyHostToDevice));
+    ensure_wr_cap(n, groups != nullptr, responses != nullptr);
+    // Previous round's kernel has been synchronized below before this
+    // call returns, so the scratch is free for reuse here.
+    PT_HIP_CHECK(hipMemcpy(wr_.pos, pos, n * 3 * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(wr_.dest, dest, n * 3 * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(wr_.w, weights, n * 8, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(wr_.elem, elem, n * 4, hipMemcpyHostToDevice));
+    const uint16_t *dg = groups ? wr_.groups : nullptr;
+    const double *dr = responses ? wr_.resp : nullptr;
+    if (groups)
+      PT_HIP_CHECK(hipMemcpy(wr_.groups, groups, n * 2, hipMemcpyHostToDevice));
+    if (responses)
+      PT_HIP_CHECK(hipMemcpy(wr_.resp, responses, n * nscores * 8,
+                             hipMemcpyHostToDevice));
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
-          dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+          d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_, n,
+          steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
+          d_pidx_, d_pelem_, d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
-          d_planes_, d_planes32_, d_nbr_, dp, dd, de, dw, dg, dr, dop, doe,
-          dst_, d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+          d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_, n,
+          steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
+          d_pidx_, d_pelem_, d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
-    PT_HIP_CHECK(hipMemcpy(out_pos, dop, n * 3 * 8, hipMemcpyDeviceToHost));
-    PT_HIP_CHECK(hipMemcpy(out_elem, doe, n * 4, hipMemcpyDeviceToHost));
-    PT_HIP_CHECK(hipMemcpy(out_status, dst_, n, hipMemcpyDeviceToHost));
-    for (void *q : {(void *)dp, (void *)dd, (void *)dw, (void *)de,
-                    (void *)doe, (void *)dop, (void *)dst_, (void *)dg,
-                    (void *)dr})
-      if (q) (void)hipFree(q);
+    PT_HIP_CHECK(hipMemcpy(out_pos, wr_.out_pos, n * 3 * 8,
+                           hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(out_elem, wr_.out_elem, n * 4,
+                           hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(out_status, wr_.status, n, hipMemcpyDeviceToHost));
   }
 
   void end_batch() override {
@@ -893,6 +930,7 @@ private:
   uint32_t *d_face_bc_ = nullptr;
   int32_t *d_pidx_ = nullptr, *d_pelem_ = nullptr;
   double *d_pshift_ = nullptr;
+  WalkRawScratch wr_;
   int64_t nbatches_ = 0;
   double *d_dest_[2] = {nullptr, nullptr};
   int8_t *d_flying_[2] = {nullptr, nullptr};
