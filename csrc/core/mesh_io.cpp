@@ -38,9 +38,85 @@ void swap_write_i32(std::ofstream &f, const int32_t *v, int64_t n) {
 }
 } // namespace
 
+// Modern XML .vtu (UnstructuredGrid) writer: raw appended binary blocks
+// (little-endian, UInt64 block-size headers), the format ParaView/VTK
+// readers consume natively.  The reference's Omega_h writes .pvtu/.vtu
+// directories (vtk::write_parallel, PumiTallyImpl.cpp:415); a single .vtu
+// covers the rank-0-gathered output this framework produces.
+static void write_vtu(const std::string &path, const Mesh &m,
+                      const std::vector<std::pair<std::string, std::vector<double>>> &cell_data) {
+  std::ofstream f(path, std::ios::binary);
+  if (!f) throw std::runtime_error("cannot open " + path + " for writing");
+  // appended blocks: points, connectivity, offsets, types, then cell data
+  uint64_t off = 0;
+  const uint64_t b_points = m.nverts * 3 * 8;
+  const uint64_t b_conn = m.nelems * 4 * 8;
+  const uint64_t b_offs = m.nelems * 8;
+  const uint64_t b_types = m.nelems;
+  std::vector<uint64_t> offsets;
+  auto next = [&](uint64_t bytes) {
+    offsets.push_back(off);
+    off += 8 + bytes; // UInt64 size header + payload
+  };
+  next(b_points);
+  next(b_conn);
+  next(b_offs);
+  next(b_types);
+  for (const auto &cd : cell_data) next((uint64_t)cd.second.size() * 8);
+
+  f << "<?xml version=\"1.0\"?>\n"
+    << "<VTKFile type=\"UnstructuredGrid\" version=\"1.0\" "
+       "byte_order=\"LittleEndian\" header_type=\"UInt64\">\n"
+    << "<UnstructuredGrid>\n"
+    << "<Piece NumberOfPoints=\"" << m.nverts << "\" NumberOfCells=\""
+    << m.nelems << "\">\n";
+  size_t bi = 0;
+  f << "<Points>\n<DataArray type=\"Float64\" NumberOfComponents=\"3\" "
+       "format=\"appended\" offset=\"" << offsets[bi++] << "\"/>\n</Points>\n";
+  f << "<Cells>\n"
+    << "<DataArray type=\"Int64\" Name=\"connectivity\" format=\"appended\" "
+       "offset=\"" << offsets[bi++] << "\"/>\n"
+    << "<DataArray type=\"Int64\" Name=\"offsets\" format=\"appended\" "
+       "offset=\"" << offsets[bi++] << "\"/>\n"
+    << "<DataArray type=\"UInt8\" Name=\"types\" format=\"appended\" "
+       "offset=\"" << offsets[bi++] << "\"/>\n"
+    << "</Cells>\n";
+  f << "<CellData>\n";
+  for (const auto &cd : cell_data)
+    f << "<DataArray type=\"Float64\" Name=\"" << cd.first
+      << "\" format=\"appended\" offset=\"" << offsets[bi++] << "\"/>\n";
+  f << "</CellData>\n</Piece>\n</UnstructuredGrid>\n"
+    << "<AppendedData encoding=\"raw\">_";
+  auto block = [&](const void *data, uint64_t bytes) {
+    f.write((const char *)&bytes, 8);
+    f.write((const char *)data, (std::streamsize)bytes);
+  };
+  block(m.coords.data(), b_points);
+  {
+    std::vector<int64_t> conn(m.nelems * 4);
+    for (int64_t i = 0; i < m.nelems * 4; ++i) conn[i] = m.tet2vert[i];
+    block(conn.data(), b_conn);
+  }
+  {
+    std::vector<int64_t> offs(m.nelems);
+    for (int64_t t = 0; t < m.nelems; ++t) offs[t] = (t + 1) * 4;
+    block(offs.data(), b_offs);
+  }
+  {
+    std::vector<uint8_t> types(m.nelems, 10); // VTK_TETRA
+    block(types.data(), b_types);
+  }
+  for (const auto &cd : cell_data)
+    block(cd.second.data(), (uint64_t)cd.second.size() * 8);
+  f << "</AppendedData>\n</VTKFile>\n";
+  if (!f) throw std::runtime_error("write failed: " + path);
+}
+
 void write_vtk(const std::string &path, const Mesh &m,
                const std::vector<std::pair<std::string, std::vector<double>>> &cell_data,
                int binary /* -1 = auto (binary for big meshes) */) {
+  if (path.size() > 4 && path.compare(path.size() - 4, 4, ".vtu") == 0)
+    return write_vtu(path, m, cell_data);
   const bool bin = binary < 0 ? m.nelems > 200000 : binary != 0;
   std::ofstream f(path, std::ios::binary);
   if (!f) throw std::runtime_error("cannot open " + path + " for writing");

@@ -181,3 +181,65 @@ def test_osh_foreign_magic_rejected(tmp_path):
     (d / "0.osh").write_bytes(b"\xa1\x1a" + b"\x00" * 64)  # Omega_h-style magic
     with pytest.raises(RuntimeError, match="not a pumitally"):
         pt.read_osh(str(d))
+
+
+def test_vtu_writer_roundtrip(tmp_path):
+    """XML .vtu output: self-parse the raw appended blocks and verify the
+    geometry and cell fields round-trip bitwise."""
+    m = pt.build_box(2, 2, 2)
+    n = 30
+    rng = np.random.default_rng(6)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = rng.uniform(0.1, 0.9, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    out = tmp_path / "tally.vtu"
+    eng.write_tally_results(str(out))
+
+    raw = out.read_bytes()
+    header = raw.split(b"<AppendedData", 1)[0].decode()
+    assert 'type="UnstructuredGrid"' in header
+    assert 'Name="flux"' in header and 'Name="volume"' in header
+    payload = raw.split(b'encoding="raw">_', 1)[1]
+
+    def block(off):
+        size = int.from_bytes(payload[off:off + 8], "little")
+        return payload[off + 8:off + 8 + size], off + 8 + size
+
+    import re
+    offsets = [int(x) for x in re.findall(r'offset="(\d+)"', header)]
+    pts, _ = block(offsets[0])
+    conn, _ = block(offsets[1])
+    offs, _ = block(offsets[2])
+    types, _ = block(offsets[3])
+    flux, _ = block(offsets[4])
+    vol, _ = block(offsets[5])
+
+    assert np.frombuffer(pts, np.float64).size == m.nverts * 3
+    c = np.frombuffer(conn, np.int64).reshape(-1, 4)
+    assert c.shape[0] == m.nelems and c.max() == m.nverts - 1
+    assert np.frombuffer(offs, np.int64)[-1] == m.nelems * 4
+    assert set(np.frombuffer(types, np.uint8)) == {10}
+    expected = pt.normalize_flux(m, eng.flux())
+    assert np.array_equal(np.frombuffer(flux, np.float64), expected)
+    assert np.allclose(np.frombuffer(vol, np.float64), np.asarray(m.volumes))
+
+
+def test_vtu_grouped_fields(tmp_path):
+    m = pt.build_box(2, 2, 2)
+    n, G = 30, 2
+    rng = np.random.default_rng(7)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = rng.uniform(0.1, 0.9, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, G, n).astype(np.uint16)
+    eng = pt.TallyEngine(m, n, device="cpu", ngroups=G)
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g)
+    out = tmp_path / "grouped.vtu"
+    eng.write_tally_results(str(out))
+    head = out.read_bytes().split(b"<AppendedData", 1)[0].decode()
+    assert 'Name="flux"' in head
+    assert 'Name="flux_g0"' in head and 'Name="flux_g1"' in head
