@@ -32,16 +32,35 @@ static void fuzz() {
   std::mt19937_64 rng(42);
   for (int trial = 0; trial < 40; ++trial) {
     std::uniform_int_distribution<int> ci(1, 5);
-    Mesh m = build_box(ci(rng), ci(rng), ci(rng), 1.0, 1.0, 1.0);
+    const int nx = ci(rng), ny = ci(rng), nz = ci(rng);
+    Mesh m = build_box(nx, ny, nz, 1.0, 1.0, 1.0);
+    if (trial % 3 == 1) {
+      // exercise the periodic pairing + wrap path
+      std::vector<int64_t> hi, lo;
+      for (int64_t t = 0; t < m.nelems; ++t)
+        for (int f = 0; f < 4; ++f) {
+          if (m.nbr[t * 4 + f] != -1) continue;
+          double cx = 0;
+          for (int k = 0; k < 3; ++k)
+            cx += m.coords[(int64_t)m.tet2vert[t * 4 + kFaceVerts[f][k]] * 3];
+          cx /= 3.0;
+          if (std::fabs(cx - 1.0) < 1e-12) hi.push_back(t * 4 + f);
+          else if (std::fabs(cx) < 1e-12) lo.push_back(t * 4 + f);
+        }
+      m.set_periodic_faces(hi, lo, Vec3{-1.0, 0.0, 0.0});
+    }
     const int64_t n = 1 + (int64_t)(rng() % 50);
-    auto e = make_cpu_engine(m, n);
+    const int nscores = 1 + (int)(rng() % 3);
+    auto e = make_cpu_engine(m, n, 1, nscores);
     std::uniform_real_distribution<double> u(0.01, 0.99);
-    std::vector<double> o(n * 3), d(n * 3), w(n);
+    std::vector<double> o(n * 3), d(n * 3), w(n), resp(n * nscores);
     std::vector<int8_t> fly(n, 1);
     for (int64_t i = 0; i < n * 3; ++i) { o[i] = u(rng); d[i] = u(rng); }
     for (int64_t i = 0; i < n; ++i) w[i] = u(rng);
+    for (int64_t i = 0; i < n * nscores; ++i) resp[i] = u(rng);
     e->copy_initial_position(o.data(), n);
-    e->move(o.data(), d.data(), fly.data(), w.data(), n);
+    e->move(o.data(), d.data(), fly.data(), w.data(), n, nullptr,
+            nscores > 1 ? resp.data() : nullptr);
     e->move_continue(d.data(), fly.data(), w.data(), n);
     e->end_batch();
     (void)e->batch_sum();
