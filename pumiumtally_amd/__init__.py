@@ -262,13 +262,16 @@ class TallyEngine:
             np.ascontiguousarray(d["escaped"], dtype=np.uint8))
 
     def normalized_flux(self):
-        """flux / element volume; shape matches flux() (per group when
-        ngroups > 1)."""
+        """flux / element volume; shape matches flux() (per group/score
+        when ngroups/nscores > 1)."""
+        import numpy as np
+
         f = self.flux()
-        if self.ngroups > 1:
-            import numpy as np
-            return np.stack([_core.normalize_flux(self.mesh, f[g])
-                             for g in range(self.ngroups)])
+        if self.nscores > 1 or self.ngroups > 1:
+            flat = np.asarray(f).reshape(-1, self.mesh.nelems)
+            out = np.stack([_core.normalize_flux(self.mesh, row)
+                            for row in flat])
+            return out.reshape(np.asarray(f).shape)
         return _core.normalize_flux(self.mesh, f)
 
     def write_tally_results(self, filename: str = "fluxresult.vtk"):

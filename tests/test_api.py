@@ -97,3 +97,37 @@ def test_mesh_cli(tmp_path):
     r = subprocess.run([sys.executable, "-m", "pumiumtally_amd.mesh.cli",
                         "describe", osh], env=env, capture_output=True, text=True)
     assert r.returncode == 0 and "elements : 162" in r.stdout
+
+
+def test_move_from_device_validates_groups_and_responses():
+    """ptr() accepts uint16 groups / float64 responses interfaces and
+    rejects mismatched dtypes; the CPU engine then refuses move_device
+    (the GPU-only path), proving validation ran first."""
+    import numpy as np
+    import pytest
+    import pumiumtally_amd as pt
+
+    class Fake:
+        def __init__(self, shape, typestr):
+            self.__cuda_array_interface__ = {
+                "shape": shape, "typestr": typestr, "strides": None,
+                "data": (16, False), "version": 2,
+            }
+
+    m = pt.build_box(1, 1, 1)
+    n = 4
+    eng = pt.TallyEngine(m, n, device="cpu", nscores=2)
+    dest = Fake((n * 3,), "<f8")
+    fly = Fake((n,), "|i1")
+    w = Fake((n,), "<f8")
+    g = Fake((n,), "<u2")
+    r = Fake((n * 2,), "<f8")
+    with pytest.raises(RuntimeError, match="GPU engine"):
+        eng.move_from_device(dest, fly, w, sync_torch=False, groups=g,
+                             responses=r)
+    with pytest.raises(TypeError):
+        eng.move_from_device(dest, fly, w, sync_torch=False,
+                             groups=Fake((n,), "<i2"))
+    with pytest.raises(TypeError):
+        eng.move_from_device(dest, fly, w, sync_torch=False,
+                             responses=Fake((n,), "<f8"))  # wrong size

@@ -180,3 +180,32 @@ def test_periodic_gpu_matches_cpu():
     assert abs(gpu.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
     assert np.abs(cpu.flux() - gpu.flux()).max() < 1e-9
     assert gpu.stats()["lost_particles"] == 0
+
+
+def test_periodic_grouped_scored_combined():
+    """Combined stress: periodic wrap + energy groups + multi-score in one
+    move loop (CPU mini version of tools/burnin.py --extended)."""
+    m = _periodic_x_box(4, 4, 4)
+    n, G, S = 300, 2, 2
+    rng = np.random.default_rng(99)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    p1 = rng.uniform(0.05, 0.95, size=(n, 3))
+    wrapsel = rng.random(n) < 0.3
+    p1[wrapsel, 0] += rng.uniform(0.05, 0.5, int(wrapsel.sum()))
+    w = rng.uniform(0.1, 1.0, n)
+    groups = rng.integers(0, G, n).astype(np.uint16)
+    resp = np.column_stack([np.ones(n), rng.uniform(0.2, 3.0, n)])
+    seg = np.linalg.norm(p1 - p0, axis=1)
+    per_move = (seg[:, None] * w[:, None] * resp).sum(axis=0)
+
+    eng = pt.TallyEngine(m, n, device="cpu", ngroups=G, nscores=S)
+    eng.copy_initial_position(p0.ravel())
+    moves = 5
+    for _ in range(moves):
+        eng.move(p0.ravel(), p1.ravel(), np.ones(n, np.int8), w,
+                 groups=groups, responses=resp)
+    assert eng.stats()["lost_particles"] == 0
+    totals = eng.flux().reshape(S, -1).sum(axis=1)
+    assert np.allclose(totals, moves * per_move, rtol=1e-12)
+    by_group = eng.flux().sum(axis=2)
+    assert np.allclose(by_group.sum(axis=1), totals, rtol=1e-12)
