@@ -3,11 +3,14 @@ periodic conservation checks.  Exercises full-API moves, continue moves,
 device-resident moves, re-localizations, checkpoint save/load, flux reads
 and group tallies in one long loop."""
 import argparse
+import os
+import sys
 import time
 
 import numpy as np
 
-import pumiumtally_amd as pt
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import pumiumtally_amd as pt  # noqa: E402
 from pumiumtally_amd.mesh import box_mesh_with_tets
 from pumiumtally_amd.utils import make_box_histories
 
