@@ -80,3 +80,64 @@ def test_gloo_world2_flux_allreduce(tmp_path):
         assert p.returncode == 0, f"rank {r} failed:\n{out}"
     assert "DIST_OK" in outs[0]
     assert (tmp_path / "flux.vtk").exists()
+
+
+def test_bench_single_process_cpu(tmp_path):
+    """bench.py driver contract: one JSON line with the required keys."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--steps", "2",
+         "--warmup", "1", "--particles", "2000", "--mesh-tets", "3000",
+         "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=root)
+    assert out.returncode == 0, out.stderr
+    line = out.stdout.strip().splitlines()[-1]
+    r = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in r, key
+    assert r["n_gpus"] == 1 and r["steps"] == 2
+    assert r["value"] > 0
+    assert r["config"]["lost_particles"] == 0
+
+
+def test_bench_gloo_world2(tmp_path):
+    """bench.py under torchrun-style env (world 2, gloo): the exact launch
+    contract the driver's SCALE run uses, minus the GPUs."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(20000 + (os.getpid() + 123) % 20000),
+        "WORLD_SIZE": "2",
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(root, "bench.py"), "--steps", "2",
+             "--warmup", "1", "--particles", "1500", "--mesh-tets", "3000",
+             "--device", "cpu", "--backend", "gloo", "--gpus", "2"],
+            env=e, cwd=root, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    # rank 0 prints the single JSON result line
+    line = [l for l in outs[0].splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["config"]["global_batch"] == 3000
+    # rank 1 prints nothing JSON
+    assert not any(l.startswith("{") for l in outs[1].splitlines())
