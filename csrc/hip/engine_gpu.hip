@@ -179,6 +179,53 @@ __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
   }
 }
 
+// Wave-level tally aggregation: Morton-sorted slots put neighboring lanes
+// in neighboring tets, so at a given walk iteration many of a wave's 64
+// lanes tally into the SAME element.  The fp64-atomic pipe is the measured
+// ceiling of the walk (profiles/README.md: chase probe 19.1 G atomics/s;
+// walk at 84% of it), while VALU sits ~12% busy -- so spend idle VALU on a
+// segmented wave reduction over runs of equal element id and issue ONE
+// atomicAdd per run (the run tail carries the total).  Runs are maximal
+// CONTIGUOUS ACTIVE lane spans with equal el: a gap in the active mask
+// starts a new run, so shuffles never read across inactive lanes.
+__device__ __forceinline__ void wave_agg_atomic_add(double *__restrict__ flux,
+                                                    int32_t el, double v) {
+  const unsigned long long mask = __ballot(1);
+  const int lane = (int)(threadIdx.x & 63u);
+  const int32_t prev_el = __shfl_up(el, 1);
+  const bool prev_active = lane > 0 && ((mask >> (lane - 1)) & 1ull);
+  const bool head = !prev_active || prev_el != el;
+  // inclusive max-scan of head lane indices: every lane learns its run head
+  int head_lane = head ? lane : 0;
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) {
+    const int up = __shfl_up(head_lane, off);
+    if (lane >= off && ((mask >> (lane - off)) & 1ull) && up > head_lane)
+      head_lane = up;
+  }
+  // inclusive segmented sum over the run (all source lanes inside the run
+  // are active and contiguous by construction)
+  double acc = v;
+#pragma unroll
+  for (int off = 1; off < 64; off <<= 1) {
+    const double up = __shfl_up(acc, off);
+    if (lane - off >= head_lane) acc += up;
+  }
+  const int32_t next_el = __shfl_down(el, 1);
+  const bool next_active = lane < 63 && ((mask >> (lane + 1)) & 1ull);
+  if (!next_active || next_el != el) atomicAdd(&flux[el], acc);
+}
+
+inline bool wave_agg() {
+  static bool v = [] {
+    const char *s = getenv("PUMITALLY_WAVE_AGG");
+    // measured on MI355X: device-resident walk 1776 -> 1831M ps/s
+    // (chord 8, sorted slots), flux bit-identical; default on
+    return s ? atoi(s) != 0 : true;
+  }();
+  return v;
+}
+
 // The fused move kernel: phase A (relocation of flying, non-escaped
 // particles whose origin changed) + phase B (tallied walk to destination).
 //
@@ -187,7 +234,8 @@ __global__ void k_reduce_slices(double *__restrict__ flux, int64_t nelems,
 // dispatches block b to XCD b%8 and each XCD has a private 4 MiB L2; we
 // remap blocks so each XCD owns one contiguous (hence spatially compact)
 // slot range.  Purely a speed lever: any placement is correct.
-template <bool F32, bool Scored = false, bool Periodic = false>
+template <bool F32, bool Scored = false, bool Periodic = false,
+          bool Agg = false>
 __global__ void k_move(const Plane *__restrict__ planes,
                        const Plane32 *__restrict__ planes32,
                        const int32_t *__restrict__ nbr, GridView grid,
@@ -256,6 +304,10 @@ __global__ void k_move(const Plane *__restrict__ planes,
         } else {
           atomicAdd(&flux[goff + el], v);
         }
+      } else if constexpr (Agg) {
+        // dispatched only when groups==nullptr, so goff==0 for every lane
+        // and el alone is the aggregation key
+        wave_agg_atomic_add(flux, el, v);
       } else {
         atomicAdd(&flux[goff + el], v);
       }
@@ -812,12 +864,13 @@ private:
     PT_HIP_CHECK(hipMemcpyAsync(dst, src, bytes, hipMemcpyHostToDevice, s));
   }
 
-  template <bool F32, bool Scored, bool Periodic>
+  template <bool F32, bool Scored, bool Periodic, bool Agg = false>
   void launch_move_one(const double *origin, const double *dest,
                        const int8_t *flying, const double *weights,
                        const uint16_t *groups, const double *resp,
                        int64_t lo, int64_t hi, int steps) {
-    k_move<F32, Scored, Periodic><<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
+    k_move<F32, Scored, Periodic, Agg>
+        <<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
         d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
         flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
         d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1,
@@ -831,25 +884,31 @@ private:
                           int64_t n, int steps) {
     // Chunked launches: a ~2.6M-slot launch keeps each XCD's Morton-
     // contiguous slot range's mesh working set inside its private L2.
-    // Scored / periodic moves take dedicated k_move instantiations; the
-    // plain (headline) instantiation compiles without either feature.
+    // Scored / periodic / wave-aggregated moves take dedicated k_move
+    // instantiations; the plain (headline) instantiation compiles without
+    // any of those features.
     const int64_t chunk = chunk_particles(n);
     const bool per = d_pidx_ != nullptr;
+    // wave aggregation needs a single flat tally key per lane: plain
+    // (unscored, ungrouped) moves only
+    const bool agg = wave_agg() && !resp && !groups;
     for (int64_t lo = 0; lo < n; lo += chunk) {
       const int64_t hi = std::min(n, lo + chunk);
-      auto go = [&](auto f32c, auto scoredc, auto perc) {
+      auto go = [&](auto f32c, auto scoredc, auto perc, auto aggc) {
         launch_move_one<decltype(f32c)::value, decltype(scoredc)::value,
-                        decltype(perc)::value>(origin, dest, flying, weights,
-                                               groups, resp, lo, hi, steps);
+                        decltype(perc)::value, decltype(aggc)::value>(
+            origin, dest, flying, weights, groups, resp, lo, hi, steps);
       };
       using T = std::true_type;
       using F = std::false_type;
       if (walk_fp32) {
-        if (resp) per ? go(T{}, T{}, T{}) : go(T{}, T{}, F{});
-        else      per ? go(T{}, F{}, T{}) : go(T{}, F{}, F{});
+        if (resp)     per ? go(T{}, T{}, T{}, F{}) : go(T{}, T{}, F{}, F{});
+        else if (agg) per ? go(T{}, F{}, T{}, T{}) : go(T{}, F{}, F{}, T{});
+        else          per ? go(T{}, F{}, T{}, F{}) : go(T{}, F{}, F{}, F{});
       } else {
-        if (resp) per ? go(F{}, T{}, T{}) : go(F{}, T{}, F{});
-        else      per ? go(F{}, F{}, T{}) : go(F{}, F{}, F{});
+        if (resp)     per ? go(F{}, T{}, T{}, F{}) : go(F{}, T{}, F{}, F{});
+        else if (agg) per ? go(F{}, F{}, T{}, T{}) : go(F{}, F{}, F{}, T{});
+        else          per ? go(F{}, F{}, T{}, F{}) : go(F{}, F{}, F{}, F{});
       }
     }
   }
