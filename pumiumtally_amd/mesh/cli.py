@@ -3,13 +3,25 @@ workflow depends on (msh2osh / osh_describe / osh_scale,
 reference README.md:115-126).
 
     python -m pumiumtally_amd.mesh.cli convert  in.msh  out.osh
+    python -m pumiumtally_amd.mesh.cli convert  in.msh  out.vtu   # viz export
     python -m pumiumtally_amd.mesh.cli describe mesh.osh
     python -m pumiumtally_amd.mesh.cli scale    in.osh out.osh --factor 0.01
     python -m pumiumtally_amd.mesh.cli box      out.osh --cells 55 --extent 1.0
+
+convert/scale/box write the format the output extension names: .osh
+(directory, engine input), or .vtk/.vtu (viewer export with a "volume"
+cell field).
 """
 import argparse
 
 import numpy as np
+
+
+def _write_any(m, path):
+    if path.endswith(".vtk") or path.endswith(".vtu"):
+        m.write_vtk_fields(path, [("volume", np.asarray(m.volumes))])
+    else:
+        m.write_osh(path)
 
 
 def main():
@@ -38,7 +50,7 @@ def main():
 
     if args.cmd == "convert":
         m = pt.read_gmsh(args.msh)
-        m.write_osh(args.osh)
+        _write_any(m, args.osh)
         print(f"wrote {args.osh}: {m.nelems} tets, {m.nverts} verts")
     elif args.cmd == "describe":
         m = pt.read_mesh(args.mesh)
@@ -55,12 +67,12 @@ def main():
         m = pt.read_mesh(args.inp)
         coords = np.asarray(m.coords) * args.factor
         m2 = pt.mesh_from_arrays(coords, m.tet2vert)
-        m2.write_osh(args.out)
+        _write_any(m2, args.out)
         print(f"wrote {args.out} scaled by {args.factor}")
     elif args.cmd == "box":
         n = args.cells
         m = pt.build_box(n, n, n, args.extent, args.extent, args.extent)
-        m.write_osh(args.osh)
+        _write_any(m, args.osh)
         print(f"wrote {args.osh}: {m.nelems} tets")
 
 

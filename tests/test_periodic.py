@@ -209,3 +209,22 @@ def test_periodic_grouped_scored_combined():
     assert np.allclose(totals, moves * per_move, rtol=1e-12)
     by_group = eng.flux().sum(axis=2)
     assert np.allclose(by_group.sum(axis=1), totals, rtol=1e-12)
+
+
+def test_periodic_walk_raw():
+    """walk_raw on a full periodic mesh wraps like move() (the partitioned
+    driver refuses periodic, but direct raw walks are supported)."""
+    m = _periodic_x_box(3, 3, 3)
+    n = 60
+    rng = np.random.default_rng(41)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = o + np.column_stack([rng.uniform(0.5, 1.5, n),
+                             np.zeros(n), np.zeros(n)])
+    w = rng.uniform(0.1, 1.0, n)
+    elem = m.locate(o).astype(np.int32)
+    seg = np.linalg.norm(d - o, axis=1)
+
+    eng = pt.TallyEngine(m, 1, device="cpu")
+    out_pos, out_elem, status = eng.walk_raw(o.ravel(), d.ravel(), elem, w)
+    assert (status == 0).all()  # wrapped walks reach their (translated) dest
+    assert abs(eng.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()

@@ -3,6 +3,8 @@ import ctypes
 import subprocess
 from pathlib import Path
 
+import os as _os, sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import pumiumtally_amd  # noqa: F401  (loads the HIP runtime consistently)
 
 HERE = Path(__file__).resolve().parent

@@ -2,6 +2,8 @@
 import ctypes
 import time
 
+import os as _os, sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import pumiumtally_amd as pt  # loads the runtime
 
 

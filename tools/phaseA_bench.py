@@ -10,6 +10,8 @@ import time
 
 import numpy as np
 
+import os as _os, sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 import pumiumtally_amd as pt
 from pumiumtally_amd.mesh import box_mesh_with_tets
 from pumiumtally_amd.utils import make_box_histories
