@@ -19,8 +19,9 @@
 //     (measured 2.2x on the walk).  A slot->caller index map keeps the
 //     public API's particle indices unchanged; the caller's arrays are
 //     gathered through it on device.
-//   * Host input staging is parity double-buffered: step k+1's four H2D
-//     copies (on the copy stream) overlap step k's walk kernels (on the
+//   * Host input staging is parity double-buffered: step k+1's H2D
+//     copies (origin/dest/flying/weights, plus groups/responses when
+//     used; on the copy stream) overlap step k's walk kernels (on the
 //     compute stream); buffer reuse is fenced with per-parity events.
 //     Pinned sources (pumiumtally_amd.pinned_array or app-registered
 //     buffers) run at full link rate.
