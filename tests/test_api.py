@@ -131,3 +131,43 @@ def test_move_from_device_validates_groups_and_responses():
     with pytest.raises(TypeError):
         eng.move_from_device(dest, fly, w, sync_torch=False,
                              responses=Fake((n,), "<f8"))  # wrong size
+
+
+def test_new_api_size_validation():
+    """Bindings reject mis-sized groups/responses/weights arrays loudly."""
+    import numpy as np
+    import pytest
+    import pumiumtally_amd as pt
+    from pumiumtally_amd import _core
+
+    m = pt.build_box(2, 2, 2)
+    n = 10
+    eng = pt.TallyEngine(m, n, device="cpu", ngroups=2, nscores=2)
+    o = np.full(n * 3, 0.5)
+    fly = np.ones(n, np.int8)
+    w = np.ones(n)
+    eng.copy_initial_position(o)
+    with pytest.raises(RuntimeError, match="groups size"):
+        eng.move(o, o, fly, w, groups=np.zeros(n - 1, np.uint16))
+    with pytest.raises(RuntimeError, match="responses size"):
+        eng.move(o, o, fly, w, responses=np.zeros((n, 3)))
+    with pytest.raises(RuntimeError, match="responses size"):
+        eng.move_continue(o, fly, w, responses=np.zeros(n - 1))
+    with pytest.raises(RuntimeError, match="groups size"):
+        eng.walk_raw(o, o, np.zeros(n, np.int32), w,
+                     groups=np.zeros(2 * n, np.uint16))
+    with pytest.raises(RuntimeError, match="responses size"):
+        eng.walk_raw(o, o, np.zeros(n, np.int32), w,
+                     responses=np.zeros(n))
+    with pytest.raises(RuntimeError, match="weights size"):
+        _core.partition_morton(m, 2, np.zeros(m.nelems - 1))
+
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+    ptal = PartitionedTally(m, device="cpu", ngroups=2, nscores=2)
+    oo = np.full((n, 3), 0.5)
+    with pytest.raises(ValueError, match="groups size"):
+        ptal.run_segments(oo, oo, w, groups=np.zeros(n - 1, np.uint16))
+    with pytest.raises(ValueError, match="responses size"):
+        ptal.run_segments(oo, oo, w, responses=np.zeros((n - 1, 2)))
+    with pytest.raises(ValueError, match="one entry per element"):
+        ptal.repartition(np.zeros(3))
