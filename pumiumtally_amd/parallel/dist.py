@@ -62,7 +62,8 @@ class DistributedTally:
     """
 
     def __init__(self, mesh, particles_per_rank: int, device: Optional[str] = None,
-                 backend: Optional[str] = None, ngroups: int = 1):
+                 backend: Optional[str] = None, ngroups: int = 1,
+                 nscores: int = 1):
         from .. import TallyEngine, have_gpu
 
         self.rank, self.world, self.local = init_distributed(backend)
@@ -70,14 +71,16 @@ class DistributedTally:
             device = f"cuda:{self.local}" if have_gpu() else "cpu"
         self.device = device
         self.engine = TallyEngine(mesh, particles_per_rank, device=device,
-                                  ngroups=ngroups)
+                                  ngroups=ngroups, nscores=nscores)
         self.mesh = mesh
 
     def copy_initial_position(self, positions):
         self.engine.copy_initial_position(positions)
 
-    def move(self, origin, dest, flying, weights, groups=None):
-        self.engine.move(origin, dest, flying, weights, groups=groups)
+    def move(self, origin, dest, flying, weights, groups=None,
+             responses=None):
+        self.engine.move(origin, dest, flying, weights, groups=groups,
+                         responses=responses)
 
     def barrier(self):
         if self.world > 1:

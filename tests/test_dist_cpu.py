@@ -141,3 +141,72 @@ def test_bench_gloo_world2(tmp_path):
     assert res["config"]["global_batch"] == 3000
     # rank 1 prints nothing JSON
     assert not any(l.startswith("{") for l in outs[1].splitlines())
+
+
+WORKER_SCORED = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+from pumiumtally_amd.parallel import DistributedTally
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+mesh = pt.build_box(4, 4, 4)
+n_total = 200
+n_rank = n_total // world
+G, S = 2, 2
+
+rng = np.random.default_rng(321)  # same stream everywhere
+o = rng.uniform(0.05, 0.95, size=(n_total, 3))
+d = rng.uniform(0.05, 0.95, size=(n_total, 3))
+w = rng.uniform(0.1, 1.0, n_total)
+g = rng.integers(0, G, n_total).astype(np.uint16)
+r = rng.uniform(0.2, 2.0, size=(n_total, S))
+
+lo, hi = rank * n_rank, (rank + 1) * n_rank
+dt = DistributedTally(mesh, n_rank, device="cpu", backend="gloo",
+                      ngroups=G, nscores=S)
+dt.copy_initial_position(o[lo:hi].ravel())
+dt.move(o[lo:hi].ravel(), d[lo:hi].ravel(), np.ones(n_rank, np.int8),
+        w[lo:hi], groups=g[lo:hi], responses=np.ascontiguousarray(r[lo:hi]))
+global_flux = dt.allreduce_flux()
+
+if rank == 0:
+    ref = pt.TallyEngine(mesh, n_total, device="cpu", ngroups=G, nscores=S)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n_total, np.int8), w, groups=g,
+             responses=r)
+    assert global_flux.shape == (S, G, mesh.nelems)
+    assert np.allclose(global_flux, ref.flux(), atol=1e-12), \
+        np.abs(global_flux - ref.flux()).max()
+    print("DIST_SCORED_OK")
+import torch.distributed as dist
+dist.destroy_process_group()
+"""
+
+
+def test_gloo_world2_scored_grouped(tmp_path):
+    """Replicated driver with groups+scores: all-reduced (S,G,nelems)
+    tally equals the single-engine oracle."""
+    pytest.importorskip("torch")
+    script = tmp_path / "worker.py"
+    script.write_text(WORKER_SCORED)
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(20000 + (os.getpid() + 201) % 20000),
+        "WORLD_SIZE": "2",
+        "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "DIST_SCORED_OK" in outs[0]
