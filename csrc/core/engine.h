@@ -111,6 +111,22 @@ public:
                         const uint16_t *groups = nullptr,
                         const double *responses = nullptr) = 0;
 
+  // Device-resident walk_raw: every array already lives in this engine's
+  // device memory (no staging at all -- the partitioned driver keeps its
+  // round loop on device and exchanges records over RCCL directly).
+  // Synchronous like walk_raw.  Throws on the CPU engine.
+  virtual void walk_raw_device(int64_t n, const double *d_pos,
+                               const double *d_dest, const int32_t *d_elem,
+                               const double *d_weights, double *d_out_pos,
+                               int32_t *d_out_elem, int8_t *d_out_status,
+                               const uint16_t *d_groups = nullptr,
+                               const double *d_responses = nullptr) {
+    (void)n; (void)d_pos; (void)d_dest; (void)d_elem; (void)d_weights;
+    (void)d_out_pos; (void)d_out_elem; (void)d_out_status; (void)d_groups;
+    (void)d_responses;
+    throw std::runtime_error("walk_raw_device requires the GPU engine");
+  }
+
   // Read back state (host copies).
   virtual std::vector<double> flux() const = 0;           // nelems*ngroups*nscores, raw tally
   virtual std::vector<int32_t> elem_ids() const = 0;      // n

@@ -742,6 +742,31 @@ public:
     PT_HIP_CHECK(hipMemcpy(out_status, wr_.status, n, hipMemcpyDeviceToHost));
   }
 
+  void walk_raw_device(int64_t n, const double *d_pos, const double *d_dest,
+                       const int32_t *d_elem, const double *d_weights,
+                       double *d_out_pos, int32_t *d_out_elem,
+                       int8_t *d_out_status,
+                       const uint16_t *d_groups = nullptr,
+                       const double *d_responses = nullptr) override {
+    if (n == 0) return;
+    PT_HIP_CHECK(hipSetDevice(device_));
+    const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
+    if (walk_fp32)
+      k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+          d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
+          d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+    else
+      k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
+          d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
+          d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
+          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
+  }
+
   void end_batch() override {
     PT_HIP_CHECK(hipSetDevice(device_));
     sync();

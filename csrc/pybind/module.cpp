@@ -430,6 +430,24 @@ PYBIND11_MODULE(_core, m) {
            py::arg("pos"), py::arg("dest"), py::arg("elem"),
            py::arg("weights"), py::arg("groups") = py::none(),
            py::arg("responses") = py::none())
+      .def("walk_raw_device",
+           // Raw device-pointer variant for the device-resident partitioned
+           // round loop (pointers as from torch.Tensor.data_ptr()).
+           [](PyEngine &e, int64_t n, uintptr_t pos, uintptr_t dest,
+              uintptr_t elem, uintptr_t weights, uintptr_t out_pos,
+              uintptr_t out_elem, uintptr_t out_status, uintptr_t groups,
+              uintptr_t responses) {
+             py::gil_scoped_release nogil;
+             e.eng->walk_raw_device(
+                 n, (const double *)pos, (const double *)dest,
+                 (const int32_t *)elem, (const double *)weights,
+                 (double *)out_pos, (int32_t *)out_elem, (int8_t *)out_status,
+                 (const uint16_t *)groups, (const double *)responses);
+           },
+           py::arg("n"), py::arg("pos"), py::arg("dest"), py::arg("elem"),
+           py::arg("weights"), py::arg("out_pos"), py::arg("out_elem"),
+           py::arg("out_status"), py::arg("groups") = 0,
+           py::arg("responses") = 0)
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

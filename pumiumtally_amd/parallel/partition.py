@@ -75,6 +75,7 @@ class PartitionedTally:
         self.engine = TallyEngine(self.sub.local, 1, device=device,
                                   ngroups=self.ngroups,
                                   nscores=self.nscores)
+        self._dev_tables = None  # lazy per-decomposition device lookup tables
 
     def repartition(self, elem_weights):
         """Rebuild the decomposition from per-element work estimates
@@ -151,6 +152,10 @@ class PartitionedTally:
         rsp = responses[mine] if responses is not None else None
         elem = self.g2l[gids[mine]].astype(np.int32)
 
+        if self._use_device_rounds():
+            return self._run_rounds_device(pos, dst, wgt, grp, rsp, elem,
+                                           rec_w)
+
         for _round in range(self.max_rounds):
             outbound = [np.zeros((0, rec_w)) for _ in range(self.world)]
             if len(elem):
@@ -197,6 +202,108 @@ class PartitionedTally:
                    if groups is not None else None)
             rsp = (np.ascontiguousarray(inbound[:, 9:9 + self.nscores])
                    if responses is not None else None)
+        else:
+            raise RuntimeError("partitioned walk did not converge "
+                               f"in {self.max_rounds} handoff rounds")
+
+    def _use_device_rounds(self) -> bool:
+        """Device-resident rounds: walk + record building + unpack stay in
+        HBM; only per-round record counts touch the host.  Active on GPU
+        engines when the exchange (if any) runs over RCCL (nccl) --
+        gloo-coordinated GPU tests keep the host path.  PUMITALLY_PART_DEVICE=0
+        disables."""
+        import os
+        if os.environ.get("PUMITALLY_PART_DEVICE", "1") == "0":
+            return False
+        if not self.engine.is_gpu:
+            return False
+        if self.world == 1:
+            return True
+        import torch.distributed as dist
+        return dist.get_backend() == "nccl"
+
+    def _run_rounds_device(self, pos, dst, wgt, grp, rsp, elem, rec_w):
+        import torch
+
+        dev = torch.device(f"cuda:{self.local}")
+        to = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+        pos_t = to(pos)
+        dst_t = to(dst)
+        wgt_t = to(wgt)
+        elem_t = to(elem)
+        # uint16 group indices travel as bit-identical int16 tensors
+        grp_t = to(grp.view(np.int16)) if grp is not None else None
+        rsp_t = to(rsp) if rsp is not None else None
+        if self._dev_tables is None:
+            self._dev_tables = (
+                to(self.g2l),
+                to(np.asarray(self.foreign_gid, np.int64)),
+                to(np.asarray(self.foreign_owner, np.int64)),
+            )
+        g2l_t, fg_t, fo_t = self._dev_tables
+
+        for _round in range(self.max_rounds):
+            k = int(elem_t.numel())
+            if k:
+                out_pos = torch.empty((k, 3), dtype=torch.float64, device=dev)
+                out_elem = torch.empty(k, dtype=torch.int32, device=dev)
+                status = torch.empty(k, dtype=torch.int8, device=dev)
+                torch.cuda.synchronize()  # inputs/outputs materialized
+                self.engine._eng.walk_raw_device(
+                    k, pos_t.data_ptr(), dst_t.data_ptr(), elem_t.data_ptr(),
+                    wgt_t.data_ptr(), out_pos.data_ptr(), out_elem.data_ptr(),
+                    status.data_ptr(),
+                    grp_t.data_ptr() if grp_t is not None else 0,
+                    rsp_t.data_ptr() if rsp_t is not None else 0)
+                hand = status == 2
+                nh = int(hand.sum())
+            else:
+                nh = 0
+            if nh:
+                idx = -(out_elem[hand].long() + 2)
+                tgt_owner = fo_t[idx]
+                order = torch.argsort(tgt_owner)
+                cols = [out_pos[hand], dst_t[hand], wgt_t[hand, None],
+                        fg_t[idx][:, None].double(),
+                        (grp_t[hand][:, None].double() if grp_t is not None
+                         else torch.zeros((nh, 1), dtype=torch.float64,
+                                          device=dev))]
+                if rsp_t is not None:
+                    cols.append(rsp_t[hand])
+                rec = torch.cat(cols, dim=1)[order].contiguous()
+                counts = torch.bincount(tgt_owner, minlength=self.world)
+            else:
+                rec = torch.zeros((0, rec_w), dtype=torch.float64, device=dev)
+                counts = torch.zeros(self.world, dtype=torch.int64,
+                                     device=dev)
+
+            if self.world == 1:
+                inbound = rec
+                if inbound.numel() == 0:
+                    break
+            else:
+                import torch.distributed as dist
+                all_counts = torch.zeros(self.world * self.world,
+                                         dtype=torch.int64, device=dev)
+                dist.all_gather_into_tensor(all_counts, counts)
+                all_counts = all_counts.view(self.world, self.world)
+                if int(all_counts.sum().item()) == 0:
+                    break
+                in_counts = [int(c) * rec_w for c in all_counts[:, self.rank]]
+                out_counts = [int(c) * rec_w for c in counts]
+                recv = torch.empty(sum(in_counts), dtype=torch.float64,
+                                   device=dev)
+                dist.all_to_all_single(recv, rec.view(-1), in_counts,
+                                       out_counts)
+                inbound = recv.view(-1, rec_w)
+            pos_t = inbound[:, 0:3].contiguous()
+            dst_t = inbound[:, 3:6].contiguous()
+            wgt_t = inbound[:, 6].contiguous()
+            elem_t = g2l_t[inbound[:, 7].long()].int().contiguous()
+            if grp_t is not None:
+                grp_t = inbound[:, 8].to(torch.int16).contiguous()
+            if rsp_t is not None:
+                rsp_t = inbound[:, 9:9 + self.nscores].contiguous()
         else:
             raise RuntimeError("partitioned walk did not converge "
                                f"in {self.max_rounds} handoff rounds")

@@ -379,3 +379,62 @@ def test_submesh_carries_face_bc():
                 assert not sub.local.face_is_reflective(int(lfid[i]))
     # ghost rings can replicate a boundary face in both submeshes
     assert marked >= len(plus_x)
+
+
+@pytest.mark.gpu
+def test_partitioned_device_rounds_gpu():
+    """Device-resident round loop (walk_raw_device + on-device record
+    building) == the host-staged loop == the single-mesh oracle."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(6, 6, 6)
+    n, G, S = 8000, 2, 2
+    rng = np.random.default_rng(61)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, G, n).astype(np.uint16)
+    r = rng.uniform(0.2, 2.0, size=(n, S))
+
+    ref = pt.TallyEngine(m, n, device="cpu", ngroups=G, nscores=S)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+
+    ptal = PartitionedTally(m, device="cuda:0", ngroups=G, nscores=S)
+    assert ptal.engine.is_gpu and ptal._use_device_rounds()
+    ptal.run_segments(o, d, w, groups=g, responses=r)
+    got = ptal.flux_global()
+    assert np.abs(got - ref.flux()).max() < 1e-10
+
+    # host-staged loop agrees (PUMITALLY_PART_DEVICE=0 route)
+    import os
+    os.environ["PUMITALLY_PART_DEVICE"] = "0"
+    try:
+        ptal2 = PartitionedTally(m, device="cuda:0", ngroups=G, nscores=S)
+        assert not ptal2._use_device_rounds()
+        ptal2.run_segments(o, d, w, groups=g, responses=r)
+        assert np.abs(ptal2.flux_global() - ref.flux()).max() < 1e-10
+    finally:
+        del os.environ["PUMITALLY_PART_DEVICE"]
+
+
+@pytest.mark.gpu
+def test_partitioned_device_rounds_plain_gpu():
+    """Plain (ungrouped, unscored) device rounds vs oracle."""
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+
+    m = pt.build_box(5, 5, 5)
+    n = 6000
+    rng = np.random.default_rng(62)
+    o = rng.uniform(0.02, 0.98, size=(n, 3))
+    d = rng.uniform(0.02, 0.98, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+
+    ptal = PartitionedTally(m, device="cuda:0")
+    ptal.run_segments(o, d, w)
+    assert np.abs(ptal.flux_global() - ref.flux()).max() < 1e-10
