@@ -15,6 +15,8 @@
 #include <cstdint>
 #include <cstring>
 #include <memory>
+#include <thread>
+#include <vector>
 
 namespace py = pybind11;
 using namespace pumitally;
@@ -132,9 +134,26 @@ PYBIND11_MODULE(_core, m) {
              const int64_t n = pts.size() / 3;
              auto out = py::array_t<int32_t>(n);
              const double *p = pts.data();
+             int32_t *o = out.mutable_data();
              const double tol = 1e-10 * norm(m_.bbox_hi - m_.bbox_lo);
-             for (int64_t i = 0; i < n; ++i)
-               out.mutable_data()[i] = m_.locate({p[i * 3], p[i * 3 + 1], p[i * 3 + 2]}, tol);
+             py::gil_scoped_release nogil;
+             auto range = [&](int64_t lo, int64_t hi) {
+               for (int64_t i = lo; i < hi; ++i)
+                 o[i] = m_.locate({p[i * 3], p[i * 3 + 1], p[i * 3 + 2]}, tol);
+             };
+             const unsigned hw = std::thread::hardware_concurrency();
+             if (n >= 16384 && hw > 1) {
+               const int nthreads = (int)std::min<unsigned>(hw, 64);
+               const int64_t per = (n + nthreads - 1) / nthreads;
+               std::vector<std::thread> workers;
+               for (int t = 0; t < nthreads; ++t)
+                 workers.emplace_back([&, t] {
+                   range(t * per, std::min<int64_t>(n, (t + 1) * per));
+                 });
+               for (auto &w : workers) w.join();
+             } else {
+               range(0, n);
+             }
              return out;
            })
       .def("write_vtk",
