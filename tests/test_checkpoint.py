@@ -67,3 +67,46 @@ def test_checkpoint_shape_mismatch(tmp_path):
     e2 = pt.TallyEngine(m, 11, device="cpu")
     with pytest.raises(ValueError):
         e2.load_checkpoint(p)
+
+
+def test_checkpoint_scored_grouped(tmp_path):
+    """Checkpoint round-trip preserves the full (S,G,nelems) tally."""
+    import numpy as np
+    import pumiumtally_amd as pt
+
+    m = pt.build_box(2, 2, 2)
+    n, G, S = 40, 2, 2
+    rng = np.random.default_rng(14)
+    o = rng.uniform(0.1, 0.9, size=(n, 3))
+    d = rng.uniform(0.1, 0.9, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, G, n).astype(np.uint16)
+    r = rng.uniform(0.2, 2.0, size=(n, S))
+
+    eng = pt.TallyEngine(m, n, device="cpu", ngroups=G, nscores=S)
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+    f0 = eng.flux().copy()
+    p0 = eng.positions().copy()
+    path = str(tmp_path / "scored.npz")
+    eng.save_checkpoint(path)
+
+    # walk further, then restore: state must be exactly the checkpoint
+    eng.move(d.ravel(), o.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+    assert not np.allclose(eng.flux(), f0)
+    eng.load_checkpoint(path)
+    assert np.array_equal(eng.flux(), f0)
+    assert np.array_equal(eng.positions(), p0)
+
+    # resumed walk from the restored state matches a never-checkpointed run
+    eng.move(d.ravel(), o.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+    ref = pt.TallyEngine(m, n, device="cpu", ngroups=G, nscores=S)
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+    ref.move(d.ravel(), o.ravel(), np.ones(n, np.int8), w, groups=g,
+             responses=r)
+    assert np.allclose(eng.flux(), ref.flux(), atol=1e-14)
