@@ -22,6 +22,92 @@ import sys
 import time
 
 
+def run_partitioned(args, mesh, cells, rank, world, local, device):
+    """BASELINE config 3: Morton element partition across ranks, ghost
+    rings, cross-rank handoff over RCCL all-to-all.  Every rank holds the
+    same global segment batch (identical seed); ownership is resolved by
+    the driver.  One timed step = one full run_segments over the global
+    batch (localize + walk + handoff rounds) -- unlike the replicated
+    engine there is no persistent particle state, so this measures the
+    partitioned driver's whole batch pipeline."""
+    import numpy as np
+
+    from pumiumtally_amd.parallel.partition import PartitionedTally
+    from pumiumtally_amd.utils import make_box_histories
+
+    n_global = args.particles * world
+    p0, p1, flying, weights = make_box_histories(
+        (1.0, 1.0, 1.0), n_global, args.mean_chord, cells,
+        seed=args.seed, pinned=False, sort=not args.no_sort)
+    o = np.asarray(p0).reshape(-1, 3)
+    d = np.asarray(p1).reshape(-1, 3)
+    w = np.asarray(weights)
+
+    ptal = PartitionedTally(mesh, device=device, ngroups=args.ngroups)
+
+    def barrier_sync():
+        ptal.engine.synchronize()
+        if world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+
+    for _ in range(args.warmup):
+        ptal.run_segments(o, d, w)
+    barrier_sync()
+    t_start = time.time()
+    for _ in range(args.steps):
+        ptal.run_segments(o, d, w)
+    ptal.engine.synchronize()
+    elapsed_local = time.time() - t_start
+    barrier_sync()
+    if world > 1:
+        import torch
+        import torch.distributed as dist
+        t = torch.tensor([elapsed_local], dtype=torch.float64)
+        if dist.get_backend() == "nccl":
+            t = t.cuda(local)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.cpu().item())
+    else:
+        elapsed = elapsed_local
+
+    global_flux = ptal.flux_global()
+    if args.write_vtk:
+        ptal.write_tally_results(args.write_vtk)
+    if rank == 0:
+        result = {
+            "metric": "particle-steps/sec-partitioned",
+            "value": n_global * args.steps / elapsed,
+            "unit": "particle-steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "track-length-tally-walk",
+                "mesh_tets": int(mesh.nelems),
+                "global_batch": n_global,
+                "particles_per_gpu": args.particles,
+                "mean_chord_elems": args.mean_chord,
+                "seq_len": None,
+                "parallelism": f"partitioned{world}-morton-ghost1",
+                "ngroups": args.ngroups,
+                "device": "gpu" if ptal.engine.is_gpu else "cpu",
+                "lost_particles": ptal.engine.stats()["lost_particles"],
+                "flux_sum": float(np.asarray(global_flux).sum()),
+            },
+        }
+        print(json.dumps(result), flush=True)
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1, help="informational; actual world size comes from torchrun env")
@@ -44,6 +130,10 @@ def main():
     ap.add_argument("--device-resident", action="store_true",
                     help="inputs pre-staged in device memory (GPU transport-code "
                          "integration path); distinct metric name")
+    ap.add_argument("--partitioned", action="store_true",
+                    help="domain-decomposed mode (BASELINE config 3: Morton "
+                         "element partition + ghost rings + cross-rank "
+                         "particle handoff); distinct metric name")
     args = ap.parse_args()
 
     import numpy as np
@@ -67,6 +157,9 @@ def main():
     if rank == 0:
         print(f"[bench] mesh: {mesh.nelems} tets ({cells}^3 cells), built in {time.time()-t0:.1f}s",
               file=sys.stderr, flush=True)
+
+    if args.partitioned:
+        return run_partitioned(args, mesh, cells, rank, world, local, device)
 
     eng = pt.TallyEngine(mesh, args.particles, device=device,
                          ngroups=args.ngroups)

@@ -210,3 +210,37 @@ def test_gloo_world2_scored_grouped(tmp_path):
     for r, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0, f"rank {r} failed:\n{out}"
     assert "DIST_SCORED_OK" in outs[0]
+
+
+def test_bench_partitioned_gloo_world2(tmp_path):
+    """bench.py --partitioned (BASELINE config 3 shape) under world-2 gloo."""
+    import json
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(20000 + (os.getpid() + 307) % 20000),
+        "WORLD_SIZE": "2",
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen(
+            [sys.executable, os.path.join(root, "bench.py"), "--partitioned",
+             "--steps", "2", "--warmup", "1", "--particles", "800",
+             "--mesh-tets", "3000", "--device", "cpu", "--backend", "gloo"],
+            env=e, cwd=root, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    line = [l for l in outs[0].splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["metric"] == "particle-steps/sec-partitioned"
+    assert res["n_gpus"] == 2 and res["config"]["global_batch"] == 1600
+    assert res["config"]["lost_particles"] == 0
