@@ -107,20 +107,63 @@ def one_trial(rng, trial):
             del os.environ["PUMITALLY_WALK"]
 
 
+def one_snapped_trial(rng, trial):
+    """Adversarial geometry: origins/destinations snapped exactly onto
+    mesh vertices, edge midpoints and face centroids.  Under an isometry
+    BC (reflective everywhere) total track length must STILL be conserved
+    exactly and nothing may be lost -- this probes the walk's tolerance
+    discipline at measure-zero configurations."""
+    nx, ny, nz = rng.integers(1, 5, 3)
+    m = pt.build_box(int(nx), int(ny), int(nz))
+    fid, cen, nor = m.boundary_faces()
+    m.set_reflective_faces(fid)
+    verts = np.asarray(m.coords).reshape(-1, 3)
+    n = int(rng.integers(4, 64))
+
+    def snapped_points(k):
+        kind = rng.integers(0, 3, k)
+        pts = np.empty((k, 3))
+        vi = rng.integers(0, len(verts), k)
+        vj = rng.integers(0, len(verts), k)
+        vk = rng.integers(0, len(verts), k)
+        pts[kind == 0] = verts[vi[kind == 0]]                       # vertex
+        pts[kind == 1] = 0.5 * (verts[vi[kind == 1]] + verts[vj[kind == 1]])
+        pts[kind == 2] = (verts[vi[kind == 2]] + verts[vj[kind == 2]]
+                          + verts[vk[kind == 2]]) / 3.0
+        return pts
+
+    o = snapped_points(n)
+    d = snapped_points(n)
+    w = rng.uniform(0.1, 2.0, n)
+    seg = np.linalg.norm(d - o, axis=1)
+
+    eng = pt.TallyEngine(m, n, device="cpu")
+    eng.copy_initial_position(o.ravel())
+    assert (np.asarray(eng.elem_ids()) >= 0).all(), (trial, "snap locate")
+    eng.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert eng.stats()["lost_particles"] == 0, (trial, "snap lost")
+    expected = (seg * w).sum()
+    got = float(np.asarray(eng.flux()).sum())
+    assert abs(got - expected) <= 1e-9 * max(1.0, expected), \
+        (trial, "snap conservation", got, expected)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--trials", type=int, default=300)
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--snapped", action="store_true",
+                    help="adversarial on-vertex/edge/face geometry instead")
     args = ap.parse_args()
     rng = np.random.default_rng(args.seed)
     t0 = time.time()
     for trial in range(args.trials):
-        one_trial(rng, trial)
+        (one_snapped_trial if args.snapped else one_trial)(rng, trial)
         if (trial + 1) % 50 == 0:
             print(f"{trial + 1}/{args.trials} trials OK "
                   f"({time.time() - t0:.0f}s)", flush=True)
-    print(f"fuzz_differential: {args.trials} trials PASS "
-          f"in {time.time() - t0:.0f}s")
+    print(f"fuzz_differential{' (snapped)' if args.snapped else ''}: "
+          f"{args.trials} trials PASS in {time.time() - t0:.0f}s")
 
 
 if __name__ == "__main__":
