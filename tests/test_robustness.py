@@ -165,3 +165,25 @@ def test_second_batch_after_reinitialization():
     assert (e.escaped() == 0).all()
     e.move(o2.ravel(), d2.ravel(), np.ones(n, np.int8), np.ones(n))
     assert abs(e.flux().sum() - (0.3 + 0.2) * n) < 1e-12
+
+
+def test_extreme_anisotropy_and_scale():
+    """Conservation and localization across 10^7:1 aspect ratios and
+    mesh scales from 1e-4 to 1e6 (relative-tolerance discipline: loc_tol
+    scales with the bbox diagonal, walk t-tolerance is dimensionless)."""
+    rng = np.random.default_rng(7)
+    for ex, ey, ez in [(1000.0, 1000.0, 0.01), (1e-3, 1.0, 1e3),
+                       (5e4, 2.0, 3.0), (1e-4, 1e-4, 1e-4),
+                       (1e6, 1e6, 1e6)]:
+        m = pt.build_box(4, 4, 4, ex, ey, ez)
+        n = 300
+        o = rng.uniform(1e-4, 1 - 1e-4, (n, 3)) * [ex, ey, ez]
+        d = rng.uniform(1e-4, 1 - 1e-4, (n, 3)) * [ex, ey, ez]
+        w = rng.uniform(0.1, 1.0, n)
+        e = pt.TallyEngine(m, n, device="cpu")
+        e.copy_initial_position(o.ravel())
+        assert (np.asarray(e.elem_ids()) >= 0).all(), (ex, ey, ez)
+        e.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+        expected = (np.linalg.norm(d - o, axis=1) * w).sum()
+        assert e.stats()["lost_particles"] == 0, (ex, ey, ez)
+        assert abs(e.flux().sum() - expected) < 1e-9 * expected, (ex, ey, ez)
