@@ -4,7 +4,10 @@
 // straight-line histories.  This is the C++ twin of bench.py.
 //
 // Usage: pumitally_driver [mesh.osh|mesh.msh] [num_particles] [steps]
+//        pumitally_driver --ohMesh mesh.osh [num_particles] [steps]
 //        (no mesh argument: generates a 1M-tet box)
+// `--ohMesh` mirrors the reference host-app CLI (`openmc --ohMesh mesh.osh`,
+// reference README.md:131).
 #include "PumiTally.h"
 
 #include "../csrc/core/mesh.h"
@@ -18,9 +21,21 @@
 #include <vector>
 
 int main(int argc, char **argv) {
-  std::string mesh_path = argc > 1 ? argv[1] : "";
-  const int n = argc > 2 ? atoi(argv[2]) : 1'000'000;
-  const int steps = argc > 3 ? atoi(argv[3]) : 10;
+  // accept the reference's `--ohMesh <path>` spelling transparently
+  int a = 1;
+  std::string mesh_path;
+  if (a < argc && std::string(argv[a]) == "--ohMesh") {
+    if (a + 1 >= argc) {
+      fprintf(stderr, "--ohMesh requires a mesh path\n");
+      return 2;
+    }
+    mesh_path = argv[a + 1];
+    a += 2;
+  } else if (a < argc) {
+    mesh_path = argv[a++];
+  }
+  const int n = a < argc ? atoi(argv[a++]) : 1'000'000;
+  const int steps = a < argc ? atoi(argv[a++]) : 10;
 
   if (mesh_path.empty()) {
     printf("[driver] generating 1M-tet box mesh...\n");
