@@ -16,8 +16,9 @@
 //   * Timing (TallyTimes equivalent) uses real device synchronization; the
 //     reference's phase fences were dead code due to a macro-name mismatch
 //     (PUMI_MEASURE_TIME vs PUMITALLY_MEASURE_TIME, CMakeLists.txt:68-75).
-//   * Output is a single legacy VTK file (default "fluxresult.vtk",
-//     override with PUMITALLY_OUTPUT).
+//   * Output is a single file (default "fluxresult.vtk", override with
+//     PUMITALLY_OUTPUT; a path ending in .vtu writes modern XML
+//     UnstructuredGrid instead of legacy VTK).
 #ifndef PUMITALLY_PUMITALLY_H
 #define PUMITALLY_PUMITALLY_H
 
