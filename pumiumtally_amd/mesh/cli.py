@@ -45,6 +45,17 @@ def main():
     b.add_argument("--cells", type=int, default=10)
     b.add_argument("--extent", type=float, default=1.0)
 
+    p = sub.add_parser("partition",
+                       help="preview a domain decomposition (Morton split)")
+    p.add_argument("mesh")
+    p.add_argument("--parts", type=int, default=8)
+    p.add_argument("--ghost-rings", type=int, default=1)
+    p.add_argument("--weights", default=None,
+                   help=".npy per-element work estimates (weighted split)")
+    p.add_argument("--out", default=None,
+                   help="write owners as .npy / a .vtk|.vtu with an "
+                        "'owner' cell field")
+
     args = ap.parse_args()
     import pumiumtally_amd as pt
 
@@ -74,6 +85,34 @@ def main():
         m = pt.build_box(n, n, n, args.extent, args.extent, args.extent)
         _write_any(m, args.osh)
         print(f"wrote {args.osh}: {m.nelems} tets")
+    elif args.cmd == "partition":
+        from pumiumtally_amd import _core
+
+        m = pt.read_mesh(args.mesh)
+        w = np.load(args.weights) if args.weights else None
+        owners = _core.partition_morton(m, args.parts, w)
+        counts = np.bincount(owners, minlength=args.parts)
+        print(f"elements : {m.nelems} into {args.parts} parts")
+        print(f"counts   : min {counts.min()}, max {counts.max()}, "
+              f"imbalance {counts.max() / max(1, counts.min()):.3f}")
+        if w is not None:
+            sums = np.array([w[owners == r].sum() for r in range(args.parts)])
+            print(f"work     : min {sums.min():.4g}, max {sums.max():.4g}, "
+                  f"imbalance {sums.max() / max(1e-300, sums.min()):.3f}")
+        ncut = nghost = 0
+        for r in range(args.parts):
+            sub = _core.extract_submesh(m, owners, r, args.ghost_rings)
+            ncut += len(sub.foreign_gid)
+            nghost += sub.local.nelems - int(counts[r])
+        print(f"cut      : {ncut} handoff faces, {nghost} ghost elements "
+              f"(rings={args.ghost_rings})")
+        if args.out:
+            if args.out.endswith(".npy"):
+                np.save(args.out, owners)
+            else:
+                m.write_vtk_fields(args.out,
+                                   [("owner", owners.astype(np.float64))])
+            print(f"wrote {args.out}")
 
 
 if __name__ == "__main__":

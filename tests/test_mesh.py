@@ -4,6 +4,8 @@ Mirrors the reference's analytic-fixture strategy: the
 6-tet unit cube from build_box is the golden mesh, with hand-computed
 volumes, adjacency and localization expectations.
 """
+import os
+
 import numpy as np
 import pytest
 
@@ -243,3 +245,29 @@ def test_vtu_grouped_fields(tmp_path):
     head = out.read_bytes().split(b"<AppendedData", 1)[0].decode()
     assert 'Name="flux"' in head
     assert 'Name="flux_g0"' in head and 'Name="flux_g1"' in head
+
+
+def test_cli_partition(tmp_path):
+    """Mesh CLI partition preview: counts, weighted split, owner export."""
+    import subprocess
+    import sys
+
+    osh = tmp_path / "m.osh"
+    m = pt.build_box(4, 4, 4)
+    m.write_osh(str(osh))
+    w = np.linspace(1.0, 10.0, m.nelems)
+    wpath = tmp_path / "w.npy"
+    np.save(wpath, w)
+    out = tmp_path / "owners.npy"
+    r = subprocess.run(
+        [sys.executable, "-m", "pumiumtally_amd.mesh.cli", "partition",
+         str(osh), "--parts", "4", "--weights", str(wpath),
+         "--out", str(out)],
+        capture_output=True, text=True, timeout=120,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr
+    assert "work" in r.stdout and "cut" in r.stdout
+    owners = np.load(out)
+    assert owners.shape == (m.nelems,)
+    sums = np.array([w[owners == p].sum() for p in range(4)])
+    assert sums.max() / sums.min() < 1.3
