@@ -7,6 +7,8 @@ reference README.md:115-126).
     python -m pumiumtally_amd.mesh.cli describe mesh.osh
     python -m pumiumtally_amd.mesh.cli scale    in.osh out.osh --factor 0.01
     python -m pumiumtally_amd.mesh.cli box      out.osh --cells 55 --extent 1.0
+    python -m pumiumtally_amd.mesh.cli partition mesh.osh --parts 8 \
+        [--weights w.npy] [--out owners.vtu]
 
 convert/scale/box write the format the output extension names: .osh
 (directory, engine input), or .vtk/.vtu (viewer export with a "volume"
