@@ -45,7 +45,16 @@ struct EngineStats {
   int64_t lost_particles = 0;   // walks that hit max_steps
   int64_t moves = 0;            // move() calls
   int64_t relocated = 0;        // phase-A relocations performed
+  // Localizations that only succeeded at the loosened tolerance (tol*1e4
+  // retry in grid_locate / Mesh::locate).  A nonzero count near partition
+  // cuts can mean mis-assigned particles; surfaced so it is never silent.
+  int64_t loose_localizations = 0;
 };
+
+// First-K lost-particle capture (walks that hit max_steps): enough to find
+// and reproduce the offending histories the day a real mesh produces a
+// nonzero lost count, without any steady-state cost.
+constexpr int kMaxLostRecords = 16;
 
 class Engine {
 public:
@@ -64,10 +73,13 @@ public:
   // it exactly.
   // responses (optional, nullable): n*nscores per-particle response
   // multipliers for simultaneous tally SCORES (e.g. flux + heating +
-  // fission from one walk).  Score k of a crossing tallies
-  // seg * weight * (responses ? responses[i*nscores+k] : 1.0) into
-  // flux[(k*ngroups + group)*nelems + elem].  nscores=1 with null
-  // responses (the default) is exactly the reference's single tally.
+  // fission from one walk).  With responses given, score k of a crossing
+  // tallies seg * weight * responses[i*nscores+k] into
+  // flux[(k*ngroups + group)*nelems + elem].  With responses == nullptr
+  // only score 0 is tallied (plain seg * weight); scores 1..nscores-1
+  // stay zero (see test_scored_null_responses_is_plain_flux).  nscores=1
+  // with null responses (the default) is exactly the reference's single
+  // tally.
   virtual void move(const double *origin, const double *dest,
                     const int8_t *flying, const double *weights, int64_t n,
                     const uint16_t *groups = nullptr,
@@ -134,6 +146,11 @@ public:
   virtual std::vector<uint8_t> escaped() const = 0;       // n
 
   virtual const EngineStats &stats() const = 0;
+
+  // First min(lost_particles, kMaxLostRecords) lost-walk records, 4 doubles
+  // each: (caller particle index -- or segment index for walk_raw --,
+  // drop x, drop y, drop z).  Empty when nothing was lost.
+  virtual std::vector<double> lost_records() const { return {}; }
 
   // Batch statistics (standard MC uncertainty accounting; the reference
   // has a single accumulating tally with no variance).  end_batch()

@@ -55,14 +55,18 @@ public:
   void copy_initial_position(const double *p, int64_t n) override {
     check_n(n);
     auto locate_range = [&](int64_t lo, int64_t hi) {
+      int64_t my_loose = 0;
       for (int64_t i = lo; i < hi; ++i) {
         const Vec3 q{p[i * 3], p[i * 3 + 1], p[i * 3 + 2]};
-        elem_[i] = mesh_.locate(q, loc_tol_);
+        bool loose = false;
+        elem_[i] = mesh_.locate(q, loc_tol_, &loose);
+        my_loose += loose;
         pos_[i * 3] = q.x;
         pos_[i * 3 + 1] = q.y;
         pos_[i * 3 + 2] = q.z;
         escaped_[i] = 0;
       }
+      loose_ += my_loose;
     };
     const unsigned hw = std::thread::hardware_concurrency();
     if (n >= 65536 && hw > 1) {
@@ -143,7 +147,9 @@ public:
     if (origin && !escaped_[i]) {
       const Vec3 q{origin[i * 3], origin[i * 3 + 1], origin[i * 3 + 2]};
       if (q.x != o.x || q.y != o.y || q.z != o.z) {
-        elem_[i] = mesh_.locate(q, loc_tol_);
+        bool loose = false;
+        elem_[i] = mesh_.locate(q, loc_tol_, &loose);
+        if (loose) loose_++; // rare; atomic is fine in the threaded path
         o = q;
         reloc++;
       }
@@ -189,6 +195,7 @@ public:
                          mesh_.periodic_shift.data());
     if (out_elem == kWalkLost) {
       lost++;
+      record_lost(i, out_pos);
       out_elem = elem_[i];
     }
     elem_[i] = out_elem;
@@ -279,6 +286,7 @@ public:
         st = 3;
         oe = elem[i];
         lost++;
+        record_lost(i, op);
       } else if (esc) {
         st = 1;
       } else if (oe < -1) {
@@ -316,7 +324,16 @@ public:
   std::vector<int32_t> elem_ids() const override { return elem_; }
   std::vector<double> positions() const override { return pos_; }
   std::vector<uint8_t> escaped() const override { return escaped_; }
-  const EngineStats &stats() const override { return stats_; }
+  const EngineStats &stats() const override {
+    stats_.loose_localizations = loose_.load();
+    return stats_;
+  }
+
+  std::vector<double> lost_records() const override {
+    const int64_t k =
+        std::min<int64_t>(lost_rec_n_.load(), kMaxLostRecords);
+    return {lost_rec_.begin(), lost_rec_.begin() + k * 4};
+  }
 
   void set_flux(const double *f, int64_t ne) override {
     if (ne != (int64_t)flux_.size()) throw std::runtime_error("set_flux size mismatch");
@@ -336,6 +353,16 @@ private:
     if (n != n_) throw std::runtime_error("particle count mismatch");
   }
 
+  void record_lost(int64_t i, Vec3 p) {
+    const int64_t k = lost_rec_n_.fetch_add(1);
+    if (k < kMaxLostRecords) {
+      lost_rec_[k * 4] = (double)i;
+      lost_rec_[k * 4 + 1] = p.x;
+      lost_rec_[k * 4 + 2] = p.y;
+      lost_rec_[k * 4 + 3] = p.z;
+    }
+  }
+
   Mesh mesh_;
   int64_t n_;
   double loc_tol_;
@@ -343,7 +370,11 @@ private:
   int64_t nbatches_ = 0;
   std::vector<int32_t> elem_;
   std::vector<uint8_t> escaped_;
-  EngineStats stats_;
+  mutable EngineStats stats_;
+  std::atomic<int64_t> loose_{0};
+  std::vector<double> lost_rec_ =
+      std::vector<double>((size_t)kMaxLostRecords * 4, 0.0);
+  std::atomic<int64_t> lost_rec_n_{0};
 };
 
 } // namespace

@@ -260,7 +260,7 @@ void Mesh::finalize() {
   }
 }
 
-int32_t Mesh::locate(Vec3 p, double tol) const {
+int32_t Mesh::locate(Vec3 p, double tol, bool *used_loose) const {
   const int cx = std::clamp((int)((p.x - grid.lo.x) * grid.inv_h.x), 0, grid.nx - 1);
   const int cy = std::clamp((int)((p.y - grid.lo.y) * grid.inv_h.y), 0, grid.ny - 1);
   const int cz = std::clamp((int)((p.z - grid.lo.z) * grid.inv_h.z), 0, grid.nz - 1);
@@ -271,7 +271,10 @@ int32_t Mesh::locate(Vec3 p, double tol) const {
   // exactly on cell-boundary faces can fail the strict test in every listed
   // tet by a few ulps.
   for (int32_t i = grid.cell_start[c]; i < grid.cell_start[c + 1]; ++i)
-    if (contains(grid.cell_tets[i], p, tol * 1e4)) return grid.cell_tets[i];
+    if (contains(grid.cell_tets[i], p, tol * 1e4)) {
+      if (used_loose) *used_loose = true;
+      return grid.cell_tets[i];
+    }
   return -1;
 }
 

@@ -97,7 +97,9 @@ struct Mesh {
   bool contains(int32_t t, Vec3 p, double tol) const;
   // Locate the tet containing p, or -1. CPU reference implementation of the
   // GPU localization kernel.
-  int32_t locate(Vec3 p, double tol) const;
+  // used_loose (optional): set to true when only the tol*1e4 retry pass
+  // succeeded (see EngineStats::loose_localizations).
+  int32_t locate(Vec3 p, double tol, bool *used_loose = nullptr) const;
 };
 
 // Analytic box mesh generator: divisions (nx,ny,nz) over extents (lx,ly,lz).

@@ -45,6 +45,16 @@ void swap_write_i32(std::ofstream &f, const int32_t *v, int64_t n) {
 // covers the rank-0-gathered output this framework produces.
 static void write_vtu(const std::string &path, const Mesh &m,
                       const std::vector<std::pair<std::string, std::vector<double>>> &cell_data) {
+  // The appended blocks are raw host memory declared LittleEndian in the
+  // header; on a big-endian host that would silently corrupt the file
+  // (the legacy .vtk path byte-swaps; this one does not), so refuse.
+  {
+    const uint32_t probe = 1;
+    if (*(const uint8_t *)&probe != 1)
+      throw std::runtime_error(
+          "write_vtu: big-endian hosts are not supported (raw blocks are "
+          "written little-endian); use the legacy .vtk writer");
+  }
   std::ofstream f(path, std::ios::binary);
   if (!f) throw std::runtime_error("cannot open " + path + " for writing");
   // appended blocks: points, connectivity, offsets, types, then cell data

@@ -26,6 +26,11 @@ uint64_t morton3(uint32_t x, uint32_t y, uint32_t z) {
 std::vector<int32_t> partition_morton(const Mesh &m, int nparts,
                                       const double *weights) {
   if (nparts < 1) throw std::runtime_error("nparts must be >= 1");
+  if (m.nelems < nparts)
+    throw std::runtime_error(
+        "partition_morton: mesh has " + std::to_string(m.nelems) +
+        " elements but " + std::to_string(nparts) +
+        " parts were requested; every part must own at least one element");
   std::vector<int64_t> order(m.nelems);
   std::iota(order.begin(), order.end(), 0);
   if (nparts > 1) {
@@ -63,17 +68,34 @@ std::vector<int32_t> partition_morton(const Mesh &m, int nparts,
     w[t] = weights[t] > floor_w ? weights[t] : floor_w;
     total += w[t];
   }
-  double acc = 0.0;
-  int32_t part = 0;
+  // Split the curve where the prefix weight-midpoint crosses k*per: element
+  // i belongs to part k iff k*per <= mid_i < (k+1)*per (mid_i increasing).
+  // The clamp passes then guarantee every part owns at least one element
+  // even under extreme skew (one element holding > total/nparts weight),
+  // so no rank ever receives an empty submesh.
   const double per = total / nparts;
-  for (int64_t i = 0; i < m.nelems; ++i) {
-    // advance to the part whose [part*per, (part+1)*per) window holds the
-    // midpoint of this element's weight span
-    const double mid = acc + 0.5 * w[order[i]];
-    while (part + 1 < nparts && mid >= (part + 1) * per) part++;
-    owners[order[i]] = part;
-    acc += w[order[i]];
+  std::vector<int64_t> split((size_t)nparts + 1);
+  split[0] = 0;
+  split[nparts] = m.nelems;
+  {
+    double acc = 0.0;
+    int64_t i = 0;
+    for (int k = 1; k < nparts; ++k) {
+      const double target = k * per;
+      while (i < m.nelems && acc + 0.5 * w[order[i]] < target) {
+        acc += w[order[i]];
+        ++i;
+      }
+      split[k] = i;
+    }
   }
+  for (int k = 1; k < nparts; ++k)
+    split[k] = std::max(split[k], (int64_t)k);
+  for (int k = nparts - 1; k >= 1; --k)
+    split[k] = std::min(split[k], split[k + 1] - 1);
+  for (int k = 0; k < nparts; ++k)
+    for (int64_t j = split[k]; j < split[k + 1]; ++j)
+      owners[order[j]] = (int32_t)k;
   return owners;
 }
 

@@ -41,8 +41,18 @@ struct WalkState {
   double seg_len, t_cur, weight;
   int32_t elem, prev_elem;
   int step;
+  int wraps; // periodic-translation restarts taken (see kMaxWraps)
   bool tally;
 };
+
+// A periodic restart resets the per-wrap step budget (a long segment
+// wrapping a small periodic box many times is geometrically valid and must
+// not be dropped as lost), but the number of wraps itself is capped so a
+// numerically stuck zero-progress bounce between periodic faces still
+// terminates.  4096 wraps is far beyond any physical chord; combined with
+// max_steps per wrap the worst-case per-particle iteration count stays
+// bounded.
+constexpr int kMaxWraps = 4096;
 
 PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
   s.o = o;
@@ -55,6 +65,7 @@ PT_HD void walk_init(WalkState &s, int32_t elem, Vec3 o, Vec3 d, double w) {
   s.elem = elem;
   s.prev_elem = -1;
   s.step = 0;
+  s.wraps = 0;
   s.tally = (w != 0.0) && (s.seg_len > 0.0);
 }
 
@@ -83,6 +94,8 @@ PT_HD void periodic_restart(WalkState &s, double t_clamped,
   s.t_cur = 0.0;
   s.prev_elem = -1;
   s.elem = pair_elem;
+  s.step = 0; // fresh per-wrap budget; total wraps capped by kMaxWraps
+  s.wraps++;
 }
 
 // Periodic=false (the default, and the headline GPU instantiation) compiles
@@ -99,7 +112,9 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ pidx = nullptr,
                         const int32_t *__restrict__ pelem = nullptr,
                         const double *__restrict__ pshift = nullptr) {
-  if (s.step++ >= max_steps) {
+  bool budget_exhausted = s.step++ >= max_steps;
+  if constexpr (Periodic) budget_exhausted |= s.wraps >= kMaxWraps;
+  if (budget_exhausted) {
     // Did not converge (numerically stuck / absurd chord): drop here and
     // flag as lost (reference prints "Not all particles are found",
     // PumiTallyImpl.cpp:455-458).
@@ -230,7 +245,9 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
                           const int32_t *__restrict__ pidx = nullptr,
                           const int32_t *__restrict__ pelem = nullptr,
                           const double *__restrict__ pshift = nullptr) {
-  if (s.step++ >= max_steps) {
+  bool budget_exhausted = s.step++ >= max_steps;
+  if constexpr (Periodic) budget_exhausted |= s.wraps >= kMaxWraps;
+  if (budget_exhausted) {
     *out_elem = kWalkLost;
     *out_pos = s.o + s.t_cur * (s.d - s.o);
     *out_escaped = false;
@@ -380,8 +397,11 @@ struct GridView {
   const int32_t *cell_tets;
 };
 
+// used_loose (optional): set to true when only the tol*1e4 retry pass
+// succeeded (the strict pass can fail by a few ulps for points exactly on
+// faces); callers count these into EngineStats::loose_localizations.
 PT_HD int32_t grid_locate(const GridView &g, const Plane *__restrict__ planes,
-                          Vec3 p, double tol) {
+                          Vec3 p, double tol, bool *used_loose = nullptr) {
   int cx = (int)((p.x - g.lo.x) * g.inv_h.x);
   int cy = (int)((p.y - g.lo.y) * g.inv_h.y);
   int cz = (int)((p.z - g.lo.z) * g.inv_h.z);
@@ -393,7 +413,10 @@ PT_HD int32_t grid_locate(const GridView &g, const Plane *__restrict__ planes,
   for (int32_t i = b; i < e; ++i)
     if (tet_contains(planes, g.cell_tets[i], p, tol)) return g.cell_tets[i];
   for (int32_t i = b; i < e; ++i)
-    if (tet_contains(planes, g.cell_tets[i], p, tol * 1e4)) return g.cell_tets[i];
+    if (tet_contains(planes, g.cell_tets[i], p, tol * 1e4)) {
+      if (used_loose) *used_loose = true;
+      return g.cell_tets[i];
+    }
   return -1;
 }
 

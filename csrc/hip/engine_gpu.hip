@@ -92,12 +92,15 @@ __global__ void k_iota(int32_t *__restrict__ a, int64_t n) {
 __global__ void k_locate(const Plane *__restrict__ planes, GridView grid,
                          const double *__restrict__ q,
                          double *__restrict__ pos, int32_t *__restrict__ elem,
-                         uint8_t *__restrict__ escaped, int64_t n,
+                         uint8_t *__restrict__ escaped,
+                         unsigned long long *__restrict__ loose, int64_t n,
                          double tol) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const Vec3 p{q[i * 3], q[i * 3 + 1], q[i * 3 + 2]};
-    elem[i] = grid_locate(grid, planes, p, tol);
+    bool used_loose = false;
+    elem[i] = grid_locate(grid, planes, p, tol, &used_loose);
+    if (used_loose) atomicAdd(loose, 1ull); // rare by construction
     pos[i * 3] = p.x;
     pos[i * 3 + 1] = p.y;
     pos[i * 3 + 2] = p.z;
@@ -249,7 +252,9 @@ __global__ void k_move(const Plane *__restrict__ planes,
                        double *__restrict__ pos, int32_t *__restrict__ elem,
                        uint8_t *__restrict__ escaped,
                        double *__restrict__ flux,
-                       unsigned long long *__restrict__ lost, int64_t lo,
+                       unsigned long long *__restrict__ lost,
+                       double *__restrict__ lostrec,
+                       unsigned long long *__restrict__ loose, int64_t lo,
                        int64_t hi, double loc_tol, int max_steps,
                        int64_t nelems, int slice_mask, bool reflective,
                        const uint32_t *__restrict__ face_bc,
@@ -279,7 +284,9 @@ __global__ void k_move(const Plane *__restrict__ planes,
     if (origin != nullptr && !escaped[i]) {
       const Vec3 q{origin[c * 3], origin[c * 3 + 1], origin[c * 3 + 2]};
       if (q.x != o.x || q.y != o.y || q.z != o.z) {
-        e = grid_locate(grid, planes, q, loc_tol);
+        bool used_loose = false;
+        e = grid_locate(grid, planes, q, loc_tol, &used_loose);
+        if (used_loose) atomicAdd(loose, 1ull);
         o = q;
       }
     }
@@ -322,7 +329,13 @@ __global__ void k_move(const Plane *__restrict__ planes,
                              add, &out_elem, &out_pos, &out_esc, reflective,
                              face_bc, pidx, pelem, pshift);
     if (out_elem == kWalkLost) {
-      atomicAdd(lost, 1ull);
+      const unsigned long long k = atomicAdd(lost, 1ull);
+      if (k < (unsigned long long)kMaxLostRecords) {
+        lostrec[k * 4] = (double)c;
+        lostrec[k * 4 + 1] = out_pos.x;
+        lostrec[k * 4 + 2] = out_pos.y;
+        lostrec[k * 4 + 3] = out_pos.z;
+      }
       out_elem = e;
     }
     elem[i] = out_elem;
@@ -347,7 +360,8 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            int32_t *__restrict__ out_elem,
                            int8_t *__restrict__ out_status,
                            double *__restrict__ flux,
-                           unsigned long long *__restrict__ lost, int64_t n,
+                           unsigned long long *__restrict__ lost,
+                           double *__restrict__ lostrec, int64_t n,
                            int max_steps, bool reflective,
                            const uint32_t *__restrict__ face_bc, int ngroups,
                            int64_t nelems, int nscores,
@@ -383,7 +397,13 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     if (oe == kWalkLost) {
       st = 3;
       oe = elem[i];
-      atomicAdd(lost, 1ull);
+      const unsigned long long k = atomicAdd(lost, 1ull);
+      if (k < (unsigned long long)kMaxLostRecords) {
+        lostrec[k * 4] = (double)i;
+        lostrec[k * 4 + 1] = op.x;
+        lostrec[k * 4 + 2] = op.y;
+        lostrec[k * 4 + 3] = op.z;
+      }
     } else if (esc) {
       st = 1;
     } else if (oe < -1) {
@@ -514,8 +534,13 @@ public:
     fsz_ = mesh_.nelems * ngroups * nscores;
     d_flux_ = dmalloc<double>(fsz_ * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
+    d_loose_ = dmalloc<unsigned long long>(1);
+    d_lostrec_ = dmalloc<double>((int64_t)kMaxLostRecords * 4);
     PT_HIP_CHECK(hipMemset(d_flux_, 0, fsz_ * slices_ * sizeof(double)));
     PT_HIP_CHECK(hipMemset(d_lost_, 0, sizeof(unsigned long long)));
+    PT_HIP_CHECK(hipMemset(d_loose_, 0, sizeof(unsigned long long)));
+    PT_HIP_CHECK(
+        hipMemset(d_lostrec_, 0, kMaxLostRecords * 4 * sizeof(double)));
 
     // Parity double-buffered input staging.
     for (int p = 0; p < 2; ++p) {
@@ -569,6 +594,7 @@ public:
                     (void *)d_cell_start_, (void *)d_cell_tets_,
                     (void *)d_pos_, (void *)d_elem_, (void *)d_escaped_,
                     (void *)d_s2c_, (void *)d_flux_, (void *)d_lost_,
+                    (void *)d_loose_, (void *)d_lostrec_,
                     (void *)d_origin_[0], (void *)d_origin_[1],
                     (void *)d_dest_[0], (void *)d_dest_[1],
                     (void *)d_flying_[0], (void *)d_flying_[1],
@@ -599,8 +625,8 @@ public:
     PT_HIP_CHECK(hipStreamSynchronize(s_copy_));
     k_iota<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(d_s2c_, n_);
     k_locate<<<grid_blocks(n_), kBlock, 0, s_comp_>>>(
-        d_planes_, grid_view_, d_origin_[0], d_pos_, d_elem_, d_escaped_, n_,
-        loc_tol_);
+        d_planes_, grid_view_, d_origin_[0], d_pos_, d_elem_, d_escaped_,
+        d_loose_, n_, loc_tol_);
     PT_HIP_CHECK(hipGetLastError());
     if (sort_every_ > 0) spatial_sort();
     moves_since_sort_ = 0;
@@ -724,13 +750,15 @@ public:
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
-          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_, n,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_,
+          d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
           d_pidx_, d_pelem_, d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
-          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_, n,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_,
+          d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
           d_pidx_, d_pelem_, d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
@@ -755,13 +783,13 @@ public:
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          d_flux_, d_lost_, d_lostrec_, n, steps, reflective, d_face_bc_, ngroups,
           mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
-          d_flux_, d_lost_, n, steps, reflective, d_face_bc_, ngroups,
+          d_flux_, d_lost_, d_lostrec_, n, steps, reflective, d_face_bc_, ngroups,
           mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
@@ -840,10 +868,25 @@ public:
 
   const EngineStats &stats() const override {
     sync();
+    unsigned long long lost = 0, loose = 0;
+    PT_HIP_CHECK(hipMemcpy(&lost, d_lost_, sizeof lost, hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(
+        hipMemcpy(&loose, d_loose_, sizeof loose, hipMemcpyDeviceToHost));
+    stats_.lost_particles = (int64_t)lost;
+    stats_.loose_localizations = (int64_t)loose;
+    return stats_;
+  }
+
+  std::vector<double> lost_records() const override {
+    sync();
     unsigned long long lost = 0;
     PT_HIP_CHECK(hipMemcpy(&lost, d_lost_, sizeof lost, hipMemcpyDeviceToHost));
-    stats_.lost_particles = (int64_t)lost;
-    return stats_;
+    const int64_t k = std::min<int64_t>((int64_t)lost, kMaxLostRecords);
+    std::vector<double> out((size_t)k * 4);
+    if (k)
+      PT_HIP_CHECK(hipMemcpy(out.data(), d_lostrec_, k * 4 * sizeof(double),
+                             hipMemcpyDeviceToHost));
+    return out;
   }
 
   void set_flux(const double *f, int64_t ne) override {
@@ -899,8 +942,9 @@ private:
         <<<grid_blocks(hi - lo), kBlock, 0, s_comp_>>>(
         d_planes_, d_planes32_, d_nbr_, grid_view_, d_s2c_, origin, dest,
         flying, weights, groups, ngroups, d_pos_, d_elem_, d_escaped_,
-        d_flux_, d_lost_, lo, hi, loc_tol_, steps, mesh_.nelems, slices_ - 1,
-        reflective, d_face_bc_, resp, nscores, d_pidx_, d_pelem_, d_pshift_);
+        d_flux_, d_lost_, d_lostrec_, d_loose_, lo, hi, loc_tol_, steps,
+        mesh_.nelems, slices_ - 1, reflective, d_face_bc_, resp, nscores,
+        d_pidx_, d_pelem_, d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
   }
 
@@ -1008,6 +1052,8 @@ private:
   int32_t *d_s2c_ = nullptr;
   double *d_flux_ = nullptr;
   unsigned long long *d_lost_ = nullptr;
+  unsigned long long *d_loose_ = nullptr;
+  double *d_lostrec_ = nullptr;
   double *d_origin_[2] = {nullptr, nullptr};
   uint16_t *d_groups_[2] = {nullptr, nullptr};
   double *d_resp_[2] = {nullptr, nullptr};

@@ -515,7 +515,15 @@ PYBIND11_MODULE(_core, m) {
         d["lost_particles"] = s.lost_particles;
         d["moves"] = s.moves;
         d["relocated"] = s.relocated;
+        d["loose_localizations"] = s.loose_localizations;
         return d;
+      })
+      // First-K lost-walk capture: (k, 4) array of
+      // (particle index, drop x, drop y, drop z); empty when nothing lost.
+      .def("lost_records", [](const PyEngine &e) {
+        auto v = e.eng->lost_records();
+        auto arr = vec_to_np(std::move(v));
+        return arr.reshape({(py::ssize_t)(arr.size() / 4), (py::ssize_t)4});
       });
 
   // The 4-call C++ facade, for API-parity tests from Python.

@@ -199,6 +199,13 @@ class TallyEngine:
     def stats(self):
         return self._eng.stats()
 
+    def lost_records(self):
+        """First-K records of walks dropped at max_steps: (k, 4) array of
+        (particle index, drop x, drop y, drop z).  Pairs with
+        stats()["lost_particles"] so a nonzero lost count on a real mesh is
+        reproducible, not just counted (the reference only printfs)."""
+        return self._eng.lost_records()
+
     def end_batch(self):
         """Close the current batch: accumulate the per-batch tally into
         running sum / sum-of-squares and zero it (standard MC batch

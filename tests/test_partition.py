@@ -438,3 +438,20 @@ def test_partitioned_device_rounds_plain_gpu():
     ptal = PartitionedTally(m, device="cuda:0")
     ptal.run_segments(o, d, w)
     assert np.abs(ptal.flux_global() - ref.flux()).max() < 1e-10
+
+
+def test_weighted_partition_never_empty_under_skew():
+    """One element holding almost all the weight must not starve other
+    parts: the split clamp guarantees every part owns >= 1 element
+    (ADVICE round-1, partition.cpp weighted split)."""
+    m = pt.build_box(3, 3, 3)
+    w = np.full(m.nelems, 1e-12)
+    w[0] = 1e12  # extreme skew: element 0 dominates the total
+    for nparts in (2, 4, 8):
+        owners = np.asarray(pt._core.partition_morton(m, nparts, w))
+        counts = np.bincount(owners, minlength=nparts)
+        assert counts.min() >= 1, counts
+    # more parts than elements is a clear error, not an empty submesh
+    tiny = pt.build_box(1, 1, 1)  # 6 tets
+    with pytest.raises(RuntimeError):
+        pt._core.partition_morton(tiny, 7, None)

@@ -228,3 +228,23 @@ def test_periodic_walk_raw():
     out_pos, out_elem, status = eng.walk_raw(o.ravel(), d.ravel(), elem, w)
     assert (status == 0).all()  # wrapped walks reach their (translated) dest
     assert abs(eng.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
+
+
+def test_periodic_many_wraps_not_lost():
+    """A long segment wrapping the box more times than max_steps worth of
+    element crossings is geometrically valid and must not be dropped: a
+    periodic restart resets the per-wrap step budget (walk.h), bounded by
+    the kMaxWraps wrap cap."""
+    m = _periodic_x_box()
+    eng = pt.TallyEngine(m, 1, device="cpu")
+    eng.max_steps = 40  # ~13 crossings per unit in x at 3 cells/axis
+    wraps = 50          # 50 domain crossings >> one max_steps budget
+    o = np.array([[0.5, 0.41, 0.57]])
+    d = np.array([[0.5 + wraps, 0.41, 0.57]])
+    eng.copy_initial_position(o.ravel())
+    eng.move(o.ravel(), d.ravel(), np.ones(1, np.int8), np.ones(1))
+    assert eng.stats()["lost_particles"] == 0
+    # conservation: the whole segment length stays inside the box
+    assert abs(eng.flux().sum() - wraps) < 1e-9
+    p = eng.positions().reshape(-1, 3)
+    assert np.allclose(p[0], [0.5, 0.41, 0.57], atol=1e-7)
