@@ -22,6 +22,70 @@ import sys
 import time
 
 
+class CommShim:
+    """One surface for the two comm stacks: torch.distributed (RCCL via
+    torch) or the library's own csrc/comm layer (RCCL via rcclComm, TCP
+    fallback on CPU) -- bench runs identically on either."""
+
+    def __init__(self, native: bool, world: int, local: int, on_gpu: bool):
+        self.native = native
+        self.world = world
+        self.local = local
+        self.on_gpu = on_gpu
+        self._comm = None
+        if native and world > 1:
+            import pumiumtally_amd as pt
+            self._comm = pt._core.make_native_comm(want_gpu=on_gpu,
+                                                   device=local)
+
+    def barrier(self):
+        if self.world == 1:
+            return
+        if self.native:
+            self._comm.barrier()
+        else:
+            import torch.distributed as dist
+            dist.barrier()
+
+    def max_scalar(self, v: float) -> float:
+        if self.world == 1:
+            return v
+        if self.native:
+            import numpy as np
+            a = np.array([v])
+            self._comm.allreduce_max(a)
+            return float(a[0])
+        import torch
+        import torch.distributed as dist
+        t = torch.tensor([v], dtype=torch.float64)
+        if dist.get_backend() == "nccl":
+            t = t.cuda(self.local)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        return float(t.cpu().item())
+
+    def sum_array(self, a):
+        if self.world == 1:
+            return a
+        if self.native:
+            import numpy as np
+            out = np.ascontiguousarray(a, dtype=np.float64)
+            self._comm.allreduce_sum(out)
+            return out
+        import torch
+        import torch.distributed as dist
+        t = torch.from_numpy(a)
+        if dist.get_backend() == "nccl":
+            t = t.cuda(self.local)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        return t.cpu().numpy()
+
+    def finalize(self):
+        if self.world > 1 and not self.native:
+            import torch.distributed as dist
+            dist.destroy_process_group()
+        self._comm = None
+
+
 def run_partitioned(args, mesh, cells, rank, world, local, device):
     """BASELINE config 3: Morton element partition across ranks, ghost
     rings, cross-rank handoff over RCCL all-to-all.  Every rank holds the
@@ -134,6 +198,11 @@ def main():
                     help="domain-decomposed mode (BASELINE config 3: Morton "
                          "element partition + ghost rings + cross-rank "
                          "particle handoff); distinct metric name")
+    ap.add_argument("--native-comm", action="store_true",
+                    help="use the library's own comm layer (csrc/comm: "
+                         "rcclComm over xGMI / TCP on CPU) instead of "
+                         "torch.distributed; same metric, config notes the "
+                         "stack")
     args = ap.parse_args()
 
     import numpy as np
@@ -142,7 +211,15 @@ def main():
     from pumiumtally_amd.parallel import init_distributed
     from pumiumtally_amd.utils import make_box_histories
 
-    rank, world, local = init_distributed(args.backend)
+    if args.native_comm:
+        if args.partitioned:
+            raise SystemExit("--native-comm --partitioned: use the torch "
+                             "stack for the partitioned bench")
+        rank = int(os.environ.get("RANK", "0"))
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        local = int(os.environ.get("LOCAL_RANK", rank))
+    else:
+        rank, world, local = init_distributed(args.backend)
     on_gpu = pt.have_gpu() and args.device != "cpu"
     if not on_gpu and args.device is None:
         # CPU fallback (debug only): shrink to something a serial walk finishes
@@ -197,11 +274,11 @@ def main():
         else:
             eng.move(o, d, flying, weights, groups=groups)
 
+    comm = CommShim(args.native_comm, world, local, on_gpu)
+
     def barrier_sync():
         eng.synchronize()
-        if world > 1:
-            import torch.distributed as dist
-            dist.barrier()
+        comm.barrier()
 
     for k in range(args.warmup):
         step(k)
@@ -215,16 +292,7 @@ def main():
     barrier_sync()
 
     # MAX over ranks (slowest rank defines throughput)
-    if world > 1:
-        import torch
-        import torch.distributed as dist
-        t = torch.tensor([elapsed_local], dtype=torch.float64)
-        if dist.get_backend() == "nccl":
-            t = t.cuda(local)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.cpu().item())
-    else:
-        elapsed = elapsed_local
+    elapsed = comm.max_scalar(elapsed_local)
 
     stats = eng.stats()
     total_particle_steps = args.particles * args.steps * world
@@ -232,16 +300,7 @@ def main():
 
     # flux all-reduce + optional write (outside timed region, parity with
     # the reference's single reduction at WriteTallyResults)
-    if world > 1:
-        import torch
-        import torch.distributed as dist
-        f = torch.from_numpy(eng.flux())
-        if dist.get_backend() == "nccl":
-            f = f.cuda(local)
-        dist.all_reduce(f, op=dist.ReduceOp.SUM)
-        global_flux = f.cpu().numpy()
-    else:
-        global_flux = eng.flux()
+    global_flux = comm.sum_array(eng.flux())
     if args.write_vtk and rank == 0:
         pt.write_tally_vtk(args.write_vtk, mesh, global_flux)
 
@@ -269,6 +328,7 @@ def main():
                 "mean_chord_elems": args.mean_chord,
                 "seq_len": None,
                 "parallelism": f"dp{world}-replicated-mesh",
+                "comm": "native-rccl" if args.native_comm else "torch-rccl",
                 "ngroups": args.ngroups,
                 "device": "gpu" if eng.is_gpu else "cpu",
                 "lost_particles": stats["lost_particles"],
@@ -276,9 +336,7 @@ def main():
             },
         }
         print(json.dumps(result), flush=True)
-    if world > 1:
-        import torch.distributed as dist
-        dist.destroy_process_group()
+    comm.finalize()
 
 
 if __name__ == "__main__":

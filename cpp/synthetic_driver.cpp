@@ -13,6 +13,7 @@
 #include "../csrc/core/mesh.h"
 
 #include <chrono>
+#include <thread>
 #include <cmath>
 #include <cstdio>
 #include <cstring>
@@ -37,16 +38,38 @@ int main(int argc, char **argv) {
   const int n = a < argc ? atoi(argv[a++]) : 1'000'000;
   const int steps = a < argc ? atoi(argv[a++]) : 10;
 
+  // Multi-process launch (library-held comm, csrc/comm): RANK/WORLD_SIZE
+  // in the env make each process walk its own particle stream on its own
+  // GPU; the facade all-reduces the flux at WriteTallyResults and rank 0
+  // writes the single fluxresult.vtk.  e.g.
+  //   for r in 0..7: RANK=$r WORLD_SIZE=8 pumitally_driver --ohMesh m.osh &
+  const char *rank_env = getenv("RANK");
+  const int rank = rank_env ? atoi(rank_env) : 0;
+
   if (mesh_path.empty()) {
-    printf("[driver] generating 1M-tet box mesh...\n");
-    pumitally::Mesh box = pumitally::build_box(55, 55, 55, 1.0, 1.0, 1.0);
-    pumitally::write_osh("driver_mesh.osh", box);
+    if (rank == 0) {
+      printf("[driver] generating 1M-tet box mesh...\n");
+      pumitally::Mesh box = pumitally::build_box(55, 55, 55, 1.0, 1.0, 1.0);
+      pumitally::write_osh("driver_mesh.osh", box);
+    }
     mesh_path = "driver_mesh.osh";
+  }
+  if (rank > 0) {
+    // wait for rank 0's generated mesh to appear (no comm exists yet)
+    for (int w = 0; w < 600; ++w) {
+      FILE *probe = fopen((mesh_path + "/nparts").c_str(), "rb");
+      if (!probe) probe = fopen(mesh_path.c_str(), "rb");
+      if (probe) {
+        fclose(probe);
+        break;
+      }
+      std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    }
   }
 
   pumitally::PumiTally tally(mesh_path, n, argc, argv);
 
-  std::mt19937_64 rng(12345);
+  std::mt19937_64 rng(12345 + 7919ull * rank); // disjoint per-rank streams
   std::uniform_real_distribution<double> upos(0.01, 0.99);
   std::uniform_real_distribution<double> u01(0.0, 1.0);
 

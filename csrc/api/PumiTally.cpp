@@ -1,5 +1,6 @@
 #include "PumiTally.h"
 
+#include "../comm/comm.h"
 #include "../core/engine.h"
 
 #include <chrono>
@@ -34,6 +35,7 @@ struct TallyTimes {
 
 struct PumiTallyImpl {
   std::unique_ptr<Engine> engine;
+  std::unique_ptr<Comm> comm; // null in single-process runs
   int32_t num_particles = 0;
   TallyTimes times;
   std::string output = "fluxresult.vtk";
@@ -47,17 +49,31 @@ PumiTally::PumiTally(const std::string &mesh_filename, int32_t num_particles,
   pimpl_->num_particles = num_particles;
   const double t0 = now_s();
   Mesh mesh = read_mesh(mesh_filename);
-  printf("[INFO] pumitally loaded mesh %s with %lld elements\n",
-         mesh_filename.c_str(), (long long)mesh.nelems);
 
+  // Library-held multi-process support (the reference's pumipic::Library
+  // holds the MPI world comm, PumiTallyImpl.cpp:238-241): rank/world come
+  // from torchrun-compatible env; each rank walks its own particle batch
+  // on its own GPU (LOCAL_RANK) against the replicated mesh, and
+  // WriteTallyResults all-reduces the flux over RCCL before rank 0 writes.
+  const EnvComm env = comm_env();
   const char *dev = getenv("PUMITALLY_DEVICE");
   std::unique_ptr<Engine> eng;
   if (!dev || std::string(dev) != "cpu") {
-    const int ordinal = dev ? atoi(dev) : 0;
-    eng = make_gpu_engine(mesh, num_particles, ordinal);
+    const int ordinal = dev ? atoi(dev) : env.local_rank;
+    eng = make_gpu_engine(mesh, num_particles, ordinal,
+                          /*ngroups=*/1, /*nscores=*/1);
   }
+  const bool on_gpu = eng != nullptr;
   if (!eng) eng = make_cpu_engine(std::move(mesh), num_particles);
   pimpl_->engine = std::move(eng);
+  pimpl_->comm = make_comm_from_env(on_gpu, env.local_rank);
+  if (!pimpl_->comm || pimpl_->comm->rank() == 0)
+    printf("[INFO] pumitally loaded mesh %s with %lld elements%s\n",
+           mesh_filename.c_str(), (long long)pimpl_->engine->mesh().nelems,
+           pimpl_->comm
+               ? (" (world " + std::to_string(pimpl_->comm->world()) + ")")
+                     .c_str()
+               : "");
   if (const char *out = getenv("PUMITALLY_OUTPUT")) pimpl_->output = out;
   pimpl_->engine->synchronize();
   pimpl_->times.initialization_time += now_s() - t0;
@@ -93,10 +109,17 @@ void PumiTally::MoveToNextLocation(double *particle_origin,
 void PumiTally::WriteTallyResults() const {
   const double t0 = now_s();
   pimpl_->engine->synchronize();
-  write_tally_vtk(pimpl_->output, pimpl_->engine->mesh(),
-                  pimpl_->engine->flux());
+  std::vector<double> flux = pimpl_->engine->flux();
+  if (pimpl_->comm) {
+    // the one collective of the whole run: sum the per-rank tallies
+    // (RCCL over xGMI on GPU engines, TCP fallback on CPU)
+    pimpl_->comm->allreduce_sum(flux.data(), (int64_t)flux.size());
+  }
+  if (!pimpl_->comm || pimpl_->comm->rank() == 0)
+    write_tally_vtk(pimpl_->output, pimpl_->engine->mesh(), flux);
+  if (pimpl_->comm) pimpl_->comm->barrier();
   pimpl_->times.vtk_file_write_time += now_s() - t0;
-  pimpl_->times.print();
+  if (!pimpl_->comm || pimpl_->comm->rank() == 0) pimpl_->times.print();
 }
 
 double PumiTally::InitializationTime() const {

@@ -8,6 +8,7 @@
 #include <pybind11/stl.h>
 
 #include "../api/PumiTally.h"
+#include "../comm/comm.h"
 #include "../core/engine.h"
 
 #include <hip/hip_runtime.h>
@@ -290,6 +291,56 @@ PYBIND11_MODULE(_core, m) {
                                   tets.size() / 4, tets.data());
         });
   m.def("have_gpu", &have_gpu);
+
+  // The library's own communication layer (csrc/comm): RCCL over xGMI on
+  // GPU, TCP fallback on CPU -- no torch, no MPI.  Python surface for
+  // bench.py --native-comm and the torch-free distributed path.
+  py::class_<Comm>(m, "NativeComm")
+      .def_property_readonly("rank", &Comm::rank)
+      .def_property_readonly("world", &Comm::world)
+      .def("barrier", [](Comm &c) { py::gil_scoped_release ng; c.barrier(); })
+      .def("allreduce_sum",
+           [](Comm &c, py::array_t<double, py::array::c_style |
+                                               py::array::forcecast> a) {
+             py::gil_scoped_release ng;
+             c.allreduce_sum(a.mutable_data(), (int64_t)a.size());
+           })
+      .def("allreduce_max",
+           [](Comm &c, py::array_t<double, py::array::c_style |
+                                               py::array::forcecast> a) {
+             py::gil_scoped_release ng;
+             c.allreduce_max(a.mutable_data(), (int64_t)a.size());
+           })
+      .def("allgather",
+           [](Comm &c, int64_t v) {
+             py::gil_scoped_release ng;
+             return c.allgather(v);
+           })
+      .def("alltoallv",
+           [](Comm &c, py::array_t<double, py::array::c_style |
+                                               py::array::forcecast> send,
+              const std::vector<int64_t> &counts) {
+             std::vector<double> out;
+             {
+               py::gil_scoped_release ng;
+               out = c.alltoallv(send.data(), counts);
+             }
+             return vec_to_np(std::move(out));
+           })
+      // device-pointer variants for torch-tensor callers (GPU memory)
+      .def("allreduce_sum_device",
+           [](Comm &c, uintptr_t ptr, int64_t n) {
+             py::gil_scoped_release ng;
+             c.allreduce_sum_device((double *)ptr, n);
+           });
+  m.def(
+      "make_native_comm",
+      [](bool want_gpu, int device) {
+        auto c = make_comm_from_env(want_gpu, device);
+        return c ? c.release() : nullptr; // nullptr => world 1
+      },
+      py::arg("want_gpu") = true, py::arg("device") = 0,
+      py::return_value_policy::take_ownership);
   m.def("pinned_array", &pinned_array, py::arg("shape"), py::arg("dtype") = "float64");
   m.def("normalize_flux", [](const Mesh &m_, py::array_t<double, py::array::c_style | py::array::forcecast> f) {
     std::vector<double> flux(f.data(), f.data() + f.size());

@@ -25,6 +25,8 @@ SOURCES = [
     "csrc/core/osh_io.cpp",
     "csrc/core/engine_cpu.cpp",
     "csrc/core/partition.cpp",
+    "csrc/comm/comm_tcp.cpp",
+    "csrc/comm/comm_rccl.hip",
     "csrc/hip/engine_gpu.hip",
     "csrc/api/PumiTally.cpp",
     "csrc/pybind/module.cpp",
@@ -34,6 +36,7 @@ HEADERS = [
     "csrc/core/mesh.h",
     "csrc/core/walk.h",
     "csrc/core/engine.h",
+    "csrc/comm/comm.h",
     "csrc/api/PumiTally.h",
 ]
 
@@ -75,6 +78,8 @@ def build(force: bool = False, verbose: bool = True) -> Path:
          "-parallel-jobs=8"]
         + inc
         + [str(REPO / s) for s in SOURCES]
+        # RCCL for the library-held multi-GPU comm (csrc/comm)
+        + ["-L/opt/rocm/lib", "-lrccl", "-Wl,-rpath,/opt/rocm/lib"]
         + ["-o", str(SO_PATH)]
     )
     if verbose:
