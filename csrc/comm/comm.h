@@ -59,6 +59,9 @@ public:
 
   // Device-resident variants (RcclComm only; others throw).  d_data is
   // device memory on this rank's GPU; synchronous on return.
+  // has_device_collectives() tells device-side callers whether to use
+  // them directly or stage through host buffers.
+  virtual bool has_device_collectives() const { return false; }
   virtual void allreduce_sum_device(double *d_data, int64_t n);
   // Device all-to-all-v: d_send as in alltoallv; recv_counts[r] doubles
   // are received from rank r into *d_recv (device buffer owned by the

@@ -31,6 +31,7 @@
 #pragma once
 
 #include "mesh.h"
+#include "walk.h"
 
 #include <cstdint>
 #include <cstdlib>
@@ -173,6 +174,19 @@ public:
 
   // Block until all queued device work is done (no-op on CPU).
   virtual void synchronize() {}
+
+  // Device-resident mesh views (GPU engines only): lets sibling device
+  // code (the partitioned engine's localization kernels) reuse the
+  // engine's uploaded planes/grid instead of duplicating them in HBM.
+  struct DeviceMeshView {
+    const Plane *planes = nullptr; // nelems*4
+    const int32_t *nbr = nullptr;  // nelems*4
+    GridView grid{};
+  };
+  virtual bool device_mesh(DeviceMeshView *out) const {
+    (void)out;
+    return false;
+  }
 
   int max_steps = 0; // 0 = auto (set by implementations from mesh size)
   int ngroups = 1;   // energy groups

@@ -914,6 +914,13 @@ public:
 
   void synchronize() override { sync(); }
 
+  bool device_mesh(DeviceMeshView *out) const override {
+    out->planes = d_planes_;
+    out->nbr = d_nbr_;
+    out->grid = grid_view_;
+    return true;
+  }
+
 private:
   void check_n(int64_t n) const {
     if (n != n_) throw std::runtime_error("particle count mismatch");

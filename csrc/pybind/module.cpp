@@ -10,6 +10,7 @@
 #include "../api/PumiTally.h"
 #include "../comm/comm.h"
 #include "../core/engine.h"
+#include "../core/partition_engine.h"
 
 #include <hip/hip_runtime.h>
 
@@ -292,6 +293,142 @@ PYBIND11_MODULE(_core, m) {
         });
   m.def("have_gpu", &have_gpu);
 
+  // Stateful domain-decomposed engine (csrc/hip/partition_engine.hip):
+  // particles stay resident on their owner rank between steps; the comm
+  // (when world > 1) is the library's own TCP/RCCL layer, created from
+  // the torchrun-compatible env.
+  struct PyPartEngine {
+    std::unique_ptr<Comm> comm;
+    std::unique_ptr<PartitionedEngine> pe;
+  };
+  py::class_<PyPartEngine>(m, "PartitionedEngine")
+      .def(py::init([](const Mesh &mesh, int64_t n_global,
+                       const std::string &device, int ngroups,
+                       py::object owners, int ghost_rings) {
+             auto self = std::make_unique<PyPartEngine>();
+             const EnvComm env = comm_env();
+             self->comm =
+                 make_comm_from_env(device != "cpu", env.local_rank);
+             const int rank = self->comm ? self->comm->rank() : 0;
+             const int world = self->comm ? self->comm->world() : 1;
+             const int32_t *op = nullptr;
+             py::array_t<int32_t, py::array::c_style | py::array::forcecast>
+                 oarr;
+             if (!owners.is_none()) {
+               oarr = owners.cast<py::array_t<
+                   int32_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)oarr.size() != mesh.nelems)
+                 throw std::runtime_error("owners size != nelems");
+               op = oarr.data();
+             }
+             std::string dev = device;
+             if (dev == "auto" || dev.empty())
+               dev = have_gpu() ? ("cuda:" + std::to_string(env.local_rank))
+                                : "cpu";
+             self->pe = make_partitioned_engine(mesh, n_global,
+                                                self->comm.get(), rank,
+                                                world, dev, ngroups, op,
+                                                ghost_rings);
+             return self;
+           }),
+           py::arg("mesh"), py::arg("n_global"), py::arg("device") = "auto",
+           py::arg("ngroups") = 1, py::arg("owners") = py::none(),
+           py::arg("ghost_rings") = 1)
+      .def_property_readonly("rank",
+                             [](const PyPartEngine &s) { return s.pe->rank(); })
+      .def_property_readonly(
+          "world", [](const PyPartEngine &s) { return s.pe->world(); })
+      .def_property_readonly(
+          "num_particles",
+          [](const PyPartEngine &s) { return s.pe->num_particles(); })
+      .def_property_readonly(
+          "resident", [](const PyPartEngine &s) { return s.pe->resident(); })
+      .def("localize",
+           [](PyPartEngine &s,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  origins) {
+             if ((int64_t)origins.size() != s.pe->num_particles() * 3)
+               throw std::runtime_error("localize: size must be 3*n_global");
+             py::gil_scoped_release ng;
+             s.pe->localize(origins.data(), s.pe->num_particles());
+           })
+      .def("step",
+           [](PyPartEngine &s,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  dest,
+              py::array_t<int8_t, py::array::c_style | py::array::forcecast>
+                  flying,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  weights,
+              py::object origin, py::object groups) {
+             const int64_t n = s.pe->num_particles();
+             if ((int64_t)dest.size() != n * 3 ||
+                 (int64_t)flying.size() != n ||
+                 (int64_t)weights.size() != n)
+               throw std::runtime_error("step: array size mismatch");
+             const double *op = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast>
+                 oarr;
+             if (!origin.is_none()) {
+               oarr = origin.cast<py::array_t<
+                   double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)oarr.size() != n * 3)
+                 throw std::runtime_error("step: origin size mismatch");
+               op = oarr.data();
+             }
+             const uint16_t *gp = nullptr;
+             py::array_t<uint16_t, py::array::c_style | py::array::forcecast>
+                 garr;
+             if (!groups.is_none()) {
+               garr = groups.cast<py::array_t<
+                   uint16_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)garr.size() != n)
+                 throw std::runtime_error("step: groups size mismatch");
+               gp = garr.data();
+             }
+             py::gil_scoped_release ng;
+             s.pe->step(dest.data(), flying.data(), weights.data(), n, op,
+                        gp);
+           },
+           py::arg("dest"), py::arg("flying"), py::arg("weights"),
+           py::arg("origin") = py::none(), py::arg("groups") = py::none())
+      .def("flux_global",
+           [](PyPartEngine &s) {
+             std::vector<double> f;
+             {
+               py::gil_scoped_release ng;
+               f = s.pe->flux_global();
+             }
+             return vec_to_np(std::move(f));
+           })
+      .def("stats",
+           [](const PyPartEngine &s) {
+             const EngineStats &st = s.pe->stats();
+             py::dict d;
+             d["lost_particles"] = st.lost_particles;
+             d["moves"] = st.moves;
+             d["relocated"] = st.relocated;
+             d["loose_localizations"] = st.loose_localizations;
+             return d;
+           })
+      .def("resident_mask",
+           [](const PyPartEngine &s) {
+             auto v = s.pe->resident_mask();
+             py::array_t<uint8_t> out(v.size());
+             std::memcpy(out.mutable_data(), v.data(), v.size());
+             return out;
+           })
+      .def("positions",
+           [](const PyPartEngine &s) { return vec_to_np(s.pe->positions()); })
+      .def("elem_ids",
+           [](const PyPartEngine &s) {
+             auto v = s.pe->elem_ids();
+             py::array_t<int32_t> out(v.size());
+             std::memcpy(out.mutable_data(), v.data(), v.size() * 4);
+             return out;
+           })
+      .def("synchronize", [](PyPartEngine &s) { s.pe->synchronize(); });
+
   // The library's own communication layer (csrc/comm): RCCL over xGMI on
   // GPU, TCP fallback on CPU -- no torch, no MPI.  Python surface for
   // bench.py --native-comm and the torch-free distributed path.
@@ -341,6 +478,18 @@ PYBIND11_MODULE(_core, m) {
       },
       py::arg("want_gpu") = true, py::arg("device") = 0,
       py::return_value_policy::take_ownership);
+  // Explicit RcclComm construction (any world incl. 1): lets GPU tests
+  // exercise the rcclCommInitRank + collective paths without a multi-GPU
+  // box.  Returns None when no HIP device is present.
+  m.def(
+      "make_rccl_comm",
+      [](int rank, int world, const std::string &addr, int port, int device) {
+        auto c = make_rccl_comm(rank, world, addr, port, device);
+        return c ? c.release() : nullptr;
+      },
+      py::arg("rank") = 0, py::arg("world") = 1,
+      py::arg("addr") = "127.0.0.1", py::arg("port") = 29871,
+      py::arg("device") = 0, py::return_value_policy::take_ownership);
   m.def("pinned_array", &pinned_array, py::arg("shape"), py::arg("dtype") = "float64");
   m.def("normalize_flux", [](const Mesh &m_, py::array_t<double, py::array::c_style | py::array::forcecast> f) {
     std::vector<double> flux(f.data(), f.data() + f.size());

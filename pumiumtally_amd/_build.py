@@ -28,6 +28,7 @@ SOURCES = [
     "csrc/comm/comm_tcp.cpp",
     "csrc/comm/comm_rccl.hip",
     "csrc/hip/engine_gpu.hip",
+    "csrc/hip/partition_engine.hip",
     "csrc/api/PumiTally.cpp",
     "csrc/pybind/module.cpp",
 ]
@@ -37,6 +38,7 @@ HEADERS = [
     "csrc/core/walk.h",
     "csrc/core/engine.h",
     "csrc/comm/comm.h",
+    "csrc/core/partition_engine.h",
     "csrc/api/PumiTally.h",
 ]
 

@@ -33,8 +33,11 @@ import numpy as np
 def init_distributed(backend: Optional[str] = None):
     """Initialize torch.distributed from torchrun env vars; returns
     (rank, world_size, local_rank).  Safe to call in single-process mode
-    (returns (0, 1, 0) without initializing)."""
-    if "RANK" not in os.environ or int(os.environ.get("WORLD_SIZE", "1")) == 1:
+    (returns (0, 1, 0) without initializing).  An EXPLICIT backend forces
+    initialization even at world_size 1 (when RANK is set) so the
+    nccl-only code paths can be executed live on a single GPU."""
+    if "RANK" not in os.environ or (
+            backend is None and int(os.environ.get("WORLD_SIZE", "1")) == 1):
         return 0, 1, 0  # single process: torch never imported
     import torch
     import torch.distributed as dist

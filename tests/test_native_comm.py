@@ -187,3 +187,53 @@ def test_bench_native_comm_world2(tmp_path):
     assert rec["n_gpus"] == 2
     assert rec["config"]["comm"] == "native-rccl"
     assert rec["value"] > 0
+
+
+@pytest.mark.gpu
+def test_rccl_comm_world1_device_paths():
+    """rcclCommInitRank + ncclAllReduce + grouped ncclSend/ncclRecv
+    (self-exchange) executed live on one GPU: the world-1 rehearsal of
+    the 8-GPU RCCL paths inside the C++ library (csrc/comm/comm_rccl)."""
+    import pumiumtally_amd as pt
+
+    comm = pt._core.make_rccl_comm(0, 1, "127.0.0.1", 29871, 0)
+    assert comm is not None and comm.world == 1 and comm.rank == 0
+    # host-staged ncclAllReduce (the flux-reduction path)
+    a = np.arange(4096, dtype=np.float64) * 0.5
+    comm.allreduce_sum(a)
+    assert np.array_equal(a, np.arange(4096) * 0.5)
+    mx = np.array([2.5, -1.0])
+    comm.allreduce_max(mx)
+    assert np.array_equal(mx, [2.5, -1.0])
+    # all-to-all-v: world-1 self send/recv over the RCCL p2p path (the
+    # partitioned particle-record exchange shape)
+    send = np.array([3.0, 1.0, 4.0, 1.0, 5.0])
+    got = comm.alltoallv(send, [5])
+    assert np.array_equal(got, send)
+    # empty exchange round (the termination case)
+    got0 = comm.alltoallv(np.zeros(0), [0])
+    assert got0.size == 0
+    assert comm.allgather(42) == [42]
+    comm.barrier()
+
+
+@pytest.mark.gpu
+def test_facade_world2_tcp_gpu_engines(tmp_path):
+    """Two processes sharing GPU 0 with engines on device and the comm
+    forced to TCP (two ranks on one GPU is outside RCCL's support):
+    exercises the facade's multi-process GPU path end-to-end on real
+    hardware without needing an 8-GPU box."""
+    import pumiumtally_amd as pt
+
+    body = FACADE_WORKER.replace(
+        'ref = pt.TallyEngine(mesh, n_total, device="cpu")',
+        'ref = pt.TallyEngine(mesh, n_total, device="cpu")')
+    outs = _spawn_world2(
+        body, tmp_path,
+        extra_env={
+            "PT_MESH": str(tmp_path / "mesh.osh"),
+            "PUMITALLY_OUTPUT": str(tmp_path / "flux.vtk"),
+            "PUMITALLY_DEVICE": "0",     # GPU engines on device 0
+            "PUMITALLY_COMM": "tcp",     # comm stays off the GPU
+        })
+    assert "FACADE_WORLD2_OK" in outs[0]

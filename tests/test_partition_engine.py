@@ -1,0 +1,224 @@
+"""Stateful partitioned engine (csrc/hip/partition_engine.hip) vs the
+replicated single-engine oracle.
+
+Residency semantics under test (matching Engine::move, engine.h):
+non-flying particles stay put; escaped particles keep clipped pos/elem
+and are never phase-A relocated; resampled origins relocate (possibly
+across ranks); out-of-mesh particles tally nothing and track their
+requested position.  Flux equality vs the oracle is the full-system
+check: any residency/handoff/localization bug shows up as a missing or
+double-counted track segment.
+"""
+import os
+import subprocess
+import sys
+
+import numpy as np
+import pytest
+
+import pumiumtally_amd as pt
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _histories(mesh, n, seed, steps):
+    """Per-step (origin, dest, flying, weights) with resamples/stops."""
+    rng = np.random.default_rng(seed)
+    out = []
+    pos = rng.uniform(0.05, 0.95, size=(n, 3))
+    for s in range(steps):
+        dest = np.clip(pos + rng.normal(0, 0.25, size=(n, 3)), -0.2, 1.2)
+        flying = (rng.random(n) > 0.15).astype(np.int8)
+        w = rng.uniform(0.1, 1.0, n)
+        # resample ~10% of particles to brand-new origins
+        res = rng.random(n) < 0.10
+        origin = pos.copy()
+        origin[res] = rng.uniform(-0.05, 1.05, size=(int(res.sum()), 3))
+        out.append((origin, dest, flying, w))
+        # committed position approximation for the NEXT step's origin: the
+        # oracle tracks the true one; we just need plausible inputs, so
+        # use dest clipped into the box for flying particles
+        pos = np.where(flying[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
+    return out
+
+
+def _run_oracle(mesh, n, hist):
+    eng = pt.TallyEngine(mesh, n, device="cpu")
+    eng.copy_initial_position(hist[0][0].ravel())
+    for origin, dest, flying, w in hist:
+        eng.move(origin.ravel(), dest.ravel(), flying.copy(), w)
+    return eng
+
+
+def test_partition_engine_world1_matches_oracle():
+    mesh = pt.build_box(4, 4, 4)
+    n = 300
+    hist = _histories(mesh, n, seed=11, steps=5)
+
+    pe = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    assert pe.world == 1
+    pe.localize(hist[0][0].ravel())
+    for origin, dest, flying, w in hist:
+        pe.step(dest.ravel(), flying, w, origin=origin.ravel())
+    got = pe.flux_global()
+
+    eng = _run_oracle(mesh, n, hist)
+    want = eng.flux()
+    assert np.allclose(got, want, atol=1e-12), np.abs(got - want).max()
+
+    # per-particle state parity (positions + escaped bookkeeping): at
+    # world 1 every particle is resident here
+    assert pe.resident == n
+    assert np.allclose(np.asarray(pe.positions()).ravel(),
+                       np.asarray(eng.positions()).ravel(), atol=1e-12)
+
+
+def test_partition_engine_world1_continue_steps():
+    """origin=None (continue) steps: no relocation scans at all."""
+    mesh = pt.build_box(4, 4, 4)
+    n = 200
+    rng = np.random.default_rng(4)
+    p0 = rng.uniform(0.1, 0.9, size=(n, 3))
+    p1 = rng.uniform(0.1, 0.9, size=(n, 3))
+    w = rng.uniform(0.5, 1.5, n)
+    fly = np.ones(n, np.int8)
+
+    pe = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    pe.localize(p0.ravel())
+    for k in range(4):
+        dest = p1 if k % 2 == 0 else p0
+        pe.step(dest.ravel(), fly, w)  # continue semantics
+
+    eng = pt.TallyEngine(mesh, n, device="cpu")
+    eng.copy_initial_position(p0.ravel())
+    for k in range(4):
+        dest = p1 if k % 2 == 0 else p0
+        eng.move_continue(dest.ravel(), fly.copy(), w)
+    assert np.allclose(pe.flux_global(), eng.flux(), atol=1e-12)
+    assert pe.stats()["relocated"] == 0
+
+
+def test_partition_engine_groups_world1():
+    mesh = pt.build_box(3, 3, 3)
+    n = 150
+    rng = np.random.default_rng(9)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    p1 = rng.uniform(0.05, 0.95, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    g = rng.integers(0, 3, n).astype(np.uint16)
+    fly = np.ones(n, np.int8)
+
+    pe = pt._core.PartitionedEngine(mesh, n, device="cpu", ngroups=3)
+    pe.localize(p0.ravel())
+    pe.step(p1.ravel(), fly, w, groups=g)
+
+    eng = pt.TallyEngine(mesh, n, device="cpu", ngroups=3)
+    eng.copy_initial_position(p0.ravel())
+    eng.move(p0.ravel(), p1.ravel(), fly.copy(), w, groups=g)
+    assert np.allclose(pe.flux_global(),
+                       np.asarray(eng.flux()).ravel(), atol=1e-12)
+
+
+WORKER = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+device = os.environ.get("PT_DEVICE", "cpu")
+
+mesh = pt.build_box(4, 4, 4)
+n = 400
+steps = 5
+rng = np.random.default_rng(21)  # same stream on all ranks
+pos = rng.uniform(0.05, 0.95, size=(n, 3))
+hist = []
+for s in range(steps):
+    dest = np.clip(pos + rng.normal(0, 0.3, size=(n, 3)), -0.1, 1.1)
+    flying = (rng.random(n) > 0.1).astype(np.int8)
+    w = rng.uniform(0.1, 1.0, n)
+    res = rng.random(n) < 0.08
+    origin = pos.copy()
+    origin[res] = rng.uniform(0.0, 1.0, size=(int(res.sum()), 3))
+    hist.append((origin, dest, flying, w))
+    pos = np.where(flying[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
+
+pe = pt._core.PartitionedEngine(mesh, n, device=device)
+assert pe.world == world, pe.world
+pe.localize(hist[0][0].ravel())
+# residency is a partition of the batch across ranks
+import numpy as _np
+for origin, dest, flying, w in hist:
+    pe.step(dest.ravel(), flying, w, origin=origin.ravel())
+got = pe.flux_global()
+
+if rank == 0:
+    eng = pt.TallyEngine(mesh, n, device="cpu")
+    eng.copy_initial_position(hist[0][0].ravel())
+    for origin, dest, flying, w in hist:
+        eng.move(origin.ravel(), dest.ravel(), flying.copy(), w)
+    want = eng.flux()
+    assert np.allclose(got, want, atol=1e-11), np.abs(got - want).max()
+    print("PART_ENGINE_WORLD2_OK resident=", pe.resident)
+"""
+
+
+def _spawn2(tmp_path, device):
+    script = tmp_path / "w.py"
+    script.write_text(WORKER)
+    env = dict(os.environ)
+    env.update({
+        "WORLD_SIZE": "2",
+        "MASTER_ADDR": "127.0.0.1",
+        "PUMITALLY_PORT": str(25000 + (os.getpid() + 31) % 15000),
+        "PUMITALLY_NO_TORCH": "1",
+        "PUMITALLY_COMM": "tcp" if device != "cpu" else "",
+        "PT_DEVICE": device,
+        "PYTHONPATH": ROOT,
+    })
+    if not env["PUMITALLY_COMM"]:
+        del env["PUMITALLY_COMM"]
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = "0"
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "PART_ENGINE_WORLD2_OK" in outs[0]
+
+
+def test_partition_engine_world2_cpu(tmp_path):
+    """Real 2-part decomposition with cross-rank handoffs and cross-rank
+    resampling, against the single-engine oracle."""
+    _spawn2(tmp_path, "cpu")
+
+
+@pytest.mark.gpu
+def test_partition_engine_world1_gpu_matches_oracle():
+    mesh = pt.build_box(4, 4, 4)
+    n = 300
+    hist = _histories(mesh, n, seed=11, steps=5)
+    pe = pt._core.PartitionedEngine(mesh, n, device="cuda:0")
+    pe.localize(hist[0][0].ravel())
+    for origin, dest, flying, w in hist:
+        pe.step(dest.ravel(), flying, w, origin=origin.ravel())
+    got = pe.flux_global()
+    eng = _run_oracle(mesh, n, hist)
+    assert np.allclose(got, eng.flux(), atol=1e-11)
+    assert pe.resident == n
+    assert np.allclose(np.asarray(pe.positions()).ravel(),
+                       np.asarray(eng.positions()).ravel(), atol=1e-12)
+
+
+@pytest.mark.gpu
+def test_partition_engine_world2_gpu_shared(tmp_path):
+    """Two ranks share GPU 0 (TCP comm): the whole device pipeline --
+    prepare/gather/walk/collect/pack/unpack kernels + host-staged
+    exchange -- runs on real hardware with real handoffs."""
+    _spawn2(tmp_path, "cuda:0")
