@@ -86,14 +86,102 @@ class CommShim:
         self._comm = None
 
 
+def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
+    """BASELINE config 3 (default partitioned mode): Morton element
+    partition across ranks, ghost rings, cross-rank handoff over RCCL,
+    with PERSISTENT particle residency (the native C++ PartitionedEngine,
+    csrc/hip/partition_engine.hip).  One timed step = one step() over the
+    global batch: upload dest/flying/weights, walk resident particles,
+    exchange cut-crossers, repeat until converged.  All communication is
+    the library's own comm layer (no torch in the data path)."""
+    import numpy as np
+
+    import pumiumtally_amd as pt
+    from pumiumtally_amd.utils import make_box_histories
+
+    n_global = args.particles * world
+    p0, p1, flying, weights = make_box_histories(
+        (1.0, 1.0, 1.0), n_global, args.mean_chord, cells,
+        seed=args.seed, pinned=False, sort=not args.no_sort)
+    o = np.ascontiguousarray(np.asarray(p0, np.float64).reshape(-1))
+    d = np.ascontiguousarray(np.asarray(p1, np.float64).reshape(-1))
+    w = np.ascontiguousarray(np.asarray(weights, np.float64))
+    fly = np.ascontiguousarray(np.asarray(flying, np.int8))
+    groups = None
+    if args.ngroups > 1:
+        rng_g = np.random.default_rng(1234)
+        groups = rng_g.integers(0, args.ngroups, n_global).astype(np.uint16)
+
+    pe = pt._core.PartitionedEngine(mesh, n_global, device=device,
+                                    ngroups=args.ngroups)
+    pe.localize(o)
+    ends = (o, d)
+
+    def step(k):
+        # ping-pong continue-mode: origin == committed position
+        pe.step(ends[(k + 1) % 2], fly, w, groups=groups)
+
+    def barrier_sync():
+        pe.synchronize()
+        pe.barrier()
+
+    for k in range(args.warmup):
+        step(k)
+    barrier_sync()
+    t_start = time.time()
+    for k in range(args.warmup, args.warmup + args.steps):
+        step(k)
+    pe.synchronize()
+    elapsed_local = time.time() - t_start
+    barrier_sync()
+    elapsed = float(pe.allreduce_max(
+        np.array([elapsed_local], dtype=np.float64))[0])
+
+    global_flux = pe.flux_global()
+    st = pe.stats()
+    if args.write_vtk and rank == 0:
+        f = np.asarray(global_flux)
+        if args.ngroups > 1:
+            f = f.reshape(args.ngroups, mesh.nelems).sum(axis=0)
+        pt.write_tally_vtk(args.write_vtk, mesh, f)
+    if rank == 0:
+        result = {
+            "metric": "particle-steps/sec-partitioned",
+            "value": n_global * args.steps / elapsed,
+            "unit": "particle-steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "track-length-tally-walk",
+                "mesh_tets": int(mesh.nelems),
+                "global_batch": n_global,
+                "particles_per_gpu": args.particles,
+                "mean_chord_elems": args.mean_chord,
+                "seq_len": None,
+                "parallelism": f"partitioned{world}-morton-ghost1-stateful",
+                "comm": "native-rccl",
+                "ngroups": args.ngroups,
+                "device": "gpu" if device != "cpu" else "cpu",
+                "resident_rank0": int(pe.resident),
+                "lost_particles": st["lost_particles"],
+                "flux_sum": float(np.asarray(global_flux).sum()),
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
 def run_partitioned(args, mesh, cells, rank, world, local, device):
-    """BASELINE config 3: Morton element partition across ranks, ghost
-    rings, cross-rank handoff over RCCL all-to-all.  Every rank holds the
-    same global segment batch (identical seed); ownership is resolved by
-    the driver.  One timed step = one full run_segments over the global
-    batch (localize + walk + handoff rounds) -- unlike the replicated
-    engine there is no persistent particle state, so this measures the
-    partitioned driver's whole batch pipeline."""
+    """Stateless legacy partitioned mode (--partitioned-stateless): the
+    round-1 torch-based driver; every step re-localizes and re-uploads
+    the whole global batch (kept for comparison; the stateful native
+    engine above is the default --partitioned path)."""
     import numpy as np
 
     from pumiumtally_amd.parallel.partition import PartitionedTally
@@ -197,7 +285,12 @@ def main():
     ap.add_argument("--partitioned", action="store_true",
                     help="domain-decomposed mode (BASELINE config 3: Morton "
                          "element partition + ghost rings + cross-rank "
-                         "particle handoff); distinct metric name")
+                         "particle handoff, persistent residency via the "
+                         "native C++ PartitionedEngine); distinct metric name")
+    ap.add_argument("--partitioned-stateless", action="store_true",
+                    help="round-1 stateless partitioned driver (torch "
+                         "collectives, per-step re-localization); kept for "
+                         "comparison")
     ap.add_argument("--native-comm", action="store_true",
                     help="use the library's own comm layer (csrc/comm: "
                          "rcclComm over xGMI / TCP on CPU) instead of "
@@ -212,9 +305,9 @@ def main():
     from pumiumtally_amd.utils import make_box_histories
 
     if args.native_comm:
-        if args.partitioned:
-            raise SystemExit("--native-comm --partitioned: use the torch "
-                             "stack for the partitioned bench")
+        if args.partitioned_stateless:
+            raise SystemExit("--native-comm requires the stateful "
+                             "partitioned mode (plain --partitioned)")
         rank = int(os.environ.get("RANK", "0"))
         world = int(os.environ.get("WORLD_SIZE", "1"))
         local = int(os.environ.get("LOCAL_RANK", rank))
@@ -235,8 +328,11 @@ def main():
         print(f"[bench] mesh: {mesh.nelems} tets ({cells}^3 cells), built in {time.time()-t0:.1f}s",
               file=sys.stderr, flush=True)
 
-    if args.partitioned:
+    if args.partitioned_stateless:
         return run_partitioned(args, mesh, cells, rank, world, local, device)
+    if args.partitioned:
+        return run_partitioned_stateful(args, mesh, cells, rank, world, local,
+                                        device)
 
     eng = pt.TallyEngine(mesh, args.particles, device=device,
                          ngroups=args.ngroups)

@@ -427,7 +427,26 @@ PYBIND11_MODULE(_core, m) {
              std::memcpy(out.mutable_data(), v.data(), v.size() * 4);
              return out;
            })
-      .def("synchronize", [](PyPartEngine &s) { s.pe->synchronize(); });
+      .def("synchronize", [](PyPartEngine &s) { s.pe->synchronize(); })
+      // comm passthroughs so a driver (bench.py --partitioned) needs no
+      // second communication stack
+      .def("barrier",
+           [](PyPartEngine &s) {
+             if (s.comm) {
+               py::gil_scoped_release ng;
+               s.comm->barrier();
+             }
+           })
+      .def("allreduce_max",
+           [](PyPartEngine &s,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  a) {
+             if (s.comm) {
+               py::gil_scoped_release ng;
+               s.comm->allreduce_max(a.mutable_data(), (int64_t)a.size());
+             }
+             return a; // forcecast may have copied; reduced values are here
+           });
 
   // The library's own communication layer (csrc/comm): RCCL over xGMI on
   // GPU, TCP fallback on CPU -- no torch, no MPI.  Python surface for
