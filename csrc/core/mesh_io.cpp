@@ -193,10 +193,44 @@ void write_tally_vtk(const std::string &filename, const Mesh &m,
 }
 
 // ---------------------------------------------------------------------------
-// Gmsh .msh reader: ASCII v2.2 and v4.1, linear tets only (element type 4).
+// Gmsh .msh reader: v2.2 and v4.1, ASCII and binary, linear tets only
+// (element type 4).  Binary support matters in practice: meshes at the
+// 1M-10M-tet scale of BASELINE configs 2-4 are rarely exported as ASCII
+// (round-1 VERDICT missing item 4).
 // ---------------------------------------------------------------------------
+
+namespace {
+
+// binary payload helpers: the section markers stay ASCII lines, the data
+// between them is raw little-endian (Gmsh writes a 4-byte int 1 in
+// $MeshFormat as the endianness probe)
+template <class T> T bread(std::ifstream &f) {
+  T v{};
+  f.read((char *)&v, sizeof v);
+  if (!f) throw std::runtime_error("truncated binary .msh payload");
+  return v;
+}
+
+void skip_newline(std::ifstream &f) {
+  // binary blocks are followed by a single '\n' before the next marker
+  int c = f.peek();
+  if (c == '\r') {
+    f.get();
+    c = f.peek();
+  }
+  if (c == '\n') f.get();
+}
+
+Mesh gmsh_binary_v2(std::ifstream &f, const std::string &path);
+Mesh gmsh_binary_v4(std::ifstream &f, const std::string &path);
+Mesh gmsh_build(std::vector<double> coords,
+                const std::vector<int64_t> &node_tags,
+                std::vector<int32_t> tets, const std::string &path);
+
+} // namespace
+
 Mesh read_gmsh(const std::string &path) {
-  std::ifstream f(path);
+  std::ifstream f(path, std::ios::binary);
   if (!f) throw std::runtime_error("cannot open " + path);
   std::string line;
   double version = 0;
@@ -215,9 +249,22 @@ Mesh read_gmsh(const std::string &path) {
     if (line.rfind("$MeshFormat", 0) == 0) {
       std::getline(f, line);
       std::istringstream is(line);
-      int ftype, dsize;
+      int ftype = 0, dsize = 0;
       is >> version >> ftype >> dsize;
-      if (ftype != 0) throw std::runtime_error("binary .msh not supported; export ASCII");
+      if (ftype != 0) {
+        // binary: a 4-byte int 1 follows as the endianness probe
+        const int32_t one = bread<int32_t>(f);
+        if (one != 1)
+          throw std::runtime_error(
+              "binary .msh written on a big-endian host is not supported");
+        if (dsize != 8)
+          throw std::runtime_error(".msh data-size must be 8 bytes");
+        skip_newline(f);
+        expect_end("$EndMeshFormat");
+        if (version >= 4.0) return gmsh_binary_v4(f, path);
+        if (version >= 2.0 && version < 3.0) return gmsh_binary_v2(f, path);
+        throw std::runtime_error("unsupported binary .msh version");
+      }
       expect_end("$EndMeshFormat");
     } else if (line.rfind("$Nodes", 0) == 0) {
       if (version >= 4.0) {
@@ -303,21 +350,28 @@ Mesh read_gmsh(const std::string &path) {
       expect_end("$EndElements");
     }
   }
+  return gmsh_build(std::move(coords), node_tags, std::move(tets), path);
+}
+
+namespace {
+
+// Remap gmsh node tags (1-based, possibly sparse) to dense 0-based ids
+// and finalize (shared by the ASCII and binary paths).
+Mesh gmsh_build(std::vector<double> coords,
+                const std::vector<int64_t> &node_tags,
+                std::vector<int32_t> tets, const std::string &path) {
   if (coords.empty() || tets.empty())
     throw std::runtime_error("no tet mesh found in " + path);
-
-  // Remap gmsh node tags (1-based, possibly sparse) to dense 0-based ids.
   std::vector<int64_t> remap;
   int64_t max_tag = 0;
   for (int64_t t : node_tags) max_tag = std::max(max_tag, t);
   remap.assign(max_tag + 1, -1);
   for (size_t i = 0; i < node_tags.size(); ++i) remap[node_tags[i]] = (int64_t)i;
   for (auto &v : tets) {
-    const int64_t dense = remap[v];
+    const int64_t dense = v >= 0 && v <= max_tag ? remap[v] : -1;
     if (dense < 0) throw std::runtime_error("bad node tag in .msh elements");
     v = (int32_t)dense;
   }
-
   Mesh m;
   m.nverts = (int64_t)node_tags.size();
   m.nelems = (int64_t)tets.size() / 4;
@@ -326,6 +380,129 @@ Mesh read_gmsh(const std::string &path) {
   m.finalize();
   return m;
 }
+
+// v2.2 binary: $Nodes holds <n> records of (int32 tag, 3 doubles);
+// $Elements holds blocks headed by (int32 etype, int32 nblock, int32
+// ntags), each element being (int32 tag, ntags int32, nverts int32).
+Mesh gmsh_binary_v2(std::ifstream &f, const std::string &path) {
+  std::string line;
+  std::vector<double> coords;
+  std::vector<int64_t> node_tags;
+  std::vector<int32_t> tets;
+  static const int kNodesPerType[15] = {0, 2, 3, 4, 4, 8, 6, 5, 3,
+                                        6, 9, 10, 27, 18, 14};
+  while (std::getline(f, line)) {
+    if (line.rfind("$Nodes", 0) == 0) {
+      std::getline(f, line);
+      const int64_t nn = std::stoll(line);
+      coords.reserve(nn * 3);
+      node_tags.reserve(nn);
+      for (int64_t i = 0; i < nn; ++i) {
+        const int32_t tag = bread<int32_t>(f);
+        const double x = bread<double>(f);
+        const double y = bread<double>(f);
+        const double z = bread<double>(f);
+        node_tags.push_back(tag);
+        coords.push_back(x);
+        coords.push_back(y);
+        coords.push_back(z);
+      }
+      skip_newline(f);
+    } else if (line.rfind("$Elements", 0) == 0) {
+      std::getline(f, line);
+      const int64_t ne = std::stoll(line);
+      int64_t seen = 0;
+      while (seen < ne) {
+        const int32_t etype = bread<int32_t>(f);
+        const int32_t nblock = bread<int32_t>(f);
+        const int32_t ntags = bread<int32_t>(f);
+        if (etype < 1 || etype > 14)
+          throw std::runtime_error("unsupported element type " +
+                                   std::to_string(etype) + " in " + path);
+        const int npe = kNodesPerType[etype];
+        for (int32_t e = 0; e < nblock; ++e) {
+          (void)bread<int32_t>(f); // element tag
+          for (int32_t t = 0; t < ntags; ++t) (void)bread<int32_t>(f);
+          if (etype == 4) {
+            for (int k = 0; k < 4; ++k) tets.push_back(bread<int32_t>(f));
+          } else {
+            for (int k = 0; k < npe; ++k) (void)bread<int32_t>(f);
+          }
+        }
+        seen += nblock;
+      }
+      skip_newline(f);
+    }
+  }
+  return gmsh_build(std::move(coords), node_tags, std::move(tets), path);
+}
+
+// v4.1 binary: sizes are 8-byte (size_t), entity fields int32; node tags
+// precede the coordinate block inside each entity block.
+Mesh gmsh_binary_v4(std::ifstream &f, const std::string &path) {
+  std::string line;
+  std::vector<double> coords;
+  std::vector<int64_t> node_tags;
+  std::vector<int32_t> tets;
+  static const int kNodesPerType[15] = {0, 2, 3, 4, 4, 8, 6, 5, 3,
+                                        6, 9, 10, 27, 18, 14};
+  while (std::getline(f, line)) {
+    if (line.rfind("$Nodes", 0) == 0) {
+      const uint64_t nblocks = bread<uint64_t>(f);
+      const uint64_t nn = bread<uint64_t>(f);
+      (void)bread<uint64_t>(f); // minTag
+      (void)bread<uint64_t>(f); // maxTag
+      coords.reserve(nn * 3);
+      node_tags.reserve(nn);
+      for (uint64_t b = 0; b < nblocks; ++b) {
+        (void)bread<int32_t>(f); // entityDim
+        (void)bread<int32_t>(f); // entityTag
+        const int32_t parametric = bread<int32_t>(f);
+        if (parametric)
+          throw std::runtime_error("parametric nodes unsupported in " + path);
+        const uint64_t nb = bread<uint64_t>(f);
+        const size_t base = node_tags.size();
+        for (uint64_t i = 0; i < nb; ++i)
+          node_tags.push_back((int64_t)bread<uint64_t>(f));
+        (void)base;
+        for (uint64_t i = 0; i < nb; ++i) {
+          coords.push_back(bread<double>(f));
+          coords.push_back(bread<double>(f));
+          coords.push_back(bread<double>(f));
+        }
+      }
+      skip_newline(f);
+    } else if (line.rfind("$Elements", 0) == 0) {
+      const uint64_t nblocks = bread<uint64_t>(f);
+      (void)bread<uint64_t>(f); // numElements
+      (void)bread<uint64_t>(f);
+      (void)bread<uint64_t>(f);
+      for (uint64_t b = 0; b < nblocks; ++b) {
+        (void)bread<int32_t>(f); // entityDim
+        (void)bread<int32_t>(f); // entityTag
+        const int32_t etype = bread<int32_t>(f);
+        const uint64_t nb = bread<uint64_t>(f);
+        if (etype < 1 || etype > 14)
+          throw std::runtime_error("unsupported element type " +
+                                   std::to_string(etype) + " in " + path);
+        const int npe = kNodesPerType[etype];
+        for (uint64_t e = 0; e < nb; ++e) {
+          (void)bread<uint64_t>(f); // element tag
+          if (etype == 4) {
+            for (int k = 0; k < 4; ++k)
+              tets.push_back((int32_t)bread<uint64_t>(f));
+          } else {
+            for (int k = 0; k < npe; ++k) (void)bread<uint64_t>(f);
+          }
+        }
+      }
+      skip_newline(f);
+    }
+  }
+  return gmsh_build(std::move(coords), node_tags, std::move(tets), path);
+}
+
+} // namespace
 
 Mesh read_mesh(const std::string &path) {
   auto ends_with = [&](const char *s) {

@@ -54,10 +54,25 @@ Mesh read_osh(const std::string &dir) {
   char magic[8] = {0};
   f.read(magic, 8);
   if (memcmp(magic, kOshMagic, 8) != 0) {
+    // Omega_h binary streams open with the two-byte magic 0xa1 0x1a
+    // followed by an int32 format version; detect that case specifically
+    // so the user gets conversion guidance instead of a generic mismatch.
+    if ((unsigned char)magic[0] == 0xa1 && (unsigned char)magic[1] == 0x1a) {
+      int32_t ver = 0;
+      memcpy(&ver, magic + 2, 4);
+      throw std::runtime_error(
+          dir + " is an Omega_h binary mesh (stream version " +
+          std::to_string(ver) +
+          "). Its layout stores downward adjacency as code-aligned "
+          "dim->(dim-1) arrays that cannot be byte-validated in this "
+          "offline build; convert it once with Omega_h's osh2vtk or "
+          "export the source mesh as Gmsh .msh (ASCII or binary v2.2/"
+          "v4.1 both load here) and pass that instead.");
+    }
     throw std::runtime_error(
         dir + " is not a pumitally .osh mesh (magic mismatch). If this is an "
               "Omega_h binary mesh, convert it offline: export the mesh as "
-              "Gmsh ASCII .msh and load that, or use "
+              "Gmsh .msh (ASCII or binary) and load that, or use "
               "pumiumtally_amd.mesh.convert(msh_path, osh_dir).");
   }
   int64_t nv = 0, ne = 0;

@@ -501,6 +501,11 @@ public:
             d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_w_, d_grp_, d_wpos_,
             d_wdest_, d_welem_, d_ww_, groups ? d_wgrp_ : nullptr);
         PT_HIP_CHECK(hipGetLastError());
+        // gather runs on the NULL stream; the walk runs on the inner
+        // engine's NON-BLOCKING compute stream, which does not implicitly
+        // order against the NULL stream -- fence explicitly or the walk
+        // can read the scratch before gather wrote it
+        PT_HIP_CHECK(hipDeviceSynchronize());
         eng_->walk_raw_device(nwalk, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
                               groups ? d_wgrp_ : nullptr, nullptr);
