@@ -181,7 +181,7 @@ def test_osh_foreign_magic_rejected(tmp_path):
     d = tmp_path / "foreign.osh"
     d.mkdir()
     (d / "0.osh").write_bytes(b"\xa1\x1a" + b"\x00" * 64)  # Omega_h-style magic
-    with pytest.raises(RuntimeError, match="not a pumitally"):
+    with pytest.raises(RuntimeError, match="Omega_h binary mesh"):
         pt.read_osh(str(d))
 
 
