@@ -275,6 +275,12 @@ def main():
                     help="use move_continue (no origin upload); NOT the headline "
                          "reference-API config -- reported with a distinct metric name")
     ap.add_argument("--no-sort", action="store_true", help="disable Morton ordering of particles")
+    ap.add_argument("--source-frac", type=float, default=1.0,
+                    help="origins sampled inside the central cube of this "
+                         "axis fraction; small values (0.02) are the "
+                         "BASELINE config-4 atomic-contention stress "
+                         "(point source, hot elements); distinct metric "
+                         "suffix when < 1")
     ap.add_argument("--ngroups", type=int, default=1,
                     help="energy groups (random per-particle group indices)")
     ap.add_argument("--backend", type=str, default=None,
@@ -338,7 +344,8 @@ def main():
                          ngroups=args.ngroups)
     p0, p1, flying, weights = make_box_histories(
         (1.0, 1.0, 1.0), args.particles, args.mean_chord, cells,
-        seed=args.seed + rank, pinned=eng.is_gpu, sort=not args.no_sort)
+        seed=args.seed + rank, pinned=eng.is_gpu, sort=not args.no_sort,
+        source_frac=args.source_frac)
     groups = None
     if args.ngroups > 1:
         rng_g = np.random.default_rng(1234 + rank)
@@ -404,7 +411,8 @@ def main():
         result = {
             "metric": "particle-steps/sec"
                       + ("-continue-mode" if args.continue_mode else "")
-                      + ("-device-resident" if args.device_resident else ""),
+                      + ("-device-resident" if args.device_resident else "")
+                      + ("-point-source" if args.source_frac < 1.0 else ""),
             "value": value,
             "unit": "particle-steps/s",
             "n_gpus": world,
@@ -424,6 +432,7 @@ def main():
                 "mean_chord_elems": args.mean_chord,
                 "seq_len": None,
                 "parallelism": f"dp{world}-replicated-mesh",
+                "source_frac": args.source_frac,
                 "comm": "native-rccl" if args.native_comm else "torch-rccl",
                 "ngroups": args.ngroups,
                 "device": "gpu" if eng.is_gpu else "cpu",

@@ -24,13 +24,21 @@ def _morton_keys(p, box, bits=10):
 
 
 def make_box_histories(box, n: int, mean_chord_elems: float, cells_per_axis: int,
-                       seed: int = 0, pinned: bool = True, sort: bool = True):
+                       seed: int = 0, pinned: bool = True, sort: bool = True,
+                       source_frac: float = 1.0):
     """Returns (p0, p1, flying, weights) arrays for n particles in a box
     mesh of `cells_per_axis` cells per axis over extents `box` (3-tuple).
 
     mean_chord_elems: target mean number of element crossings per segment
     (an element is ~1/6 of a grid cell; a chord of k cells crosses ~2.2*k
     tets for this 6-tet cell cut, measured empirically).
+
+    source_frac: origins are sampled inside the central cube covering this
+    fraction of each axis.  1.0 = whole box (spread sources, the default
+    steady-state load); small values (e.g. 0.02 = a couple of grid cells)
+    put every particle through the same few elements -- the BASELINE
+    config-4 atomic-contention stress, where thousands of lanes
+    atomicAdd into the handful of tets around the source every step.
     """
     from .. import pinned_array
 
@@ -51,7 +59,13 @@ def make_box_histories(box, n: int, mean_chord_elems: float, cells_per_axis: int
     weights = alloc((n,), "float64")
 
     margin = 1e-6 * box
-    start = rng.uniform(margin, box - margin, size=(n, 3))
+    if source_frac < 1.0:
+        lo = box * (0.5 - source_frac / 2.0)
+        hi = box * (0.5 + source_frac / 2.0)
+        start = rng.uniform(np.maximum(lo, margin),
+                            np.minimum(hi, box - margin), size=(n, 3))
+    else:
+        start = rng.uniform(margin, box - margin, size=(n, 3))
     if sort:
         # Spatial (Morton) ordering: adjacent particles walk adjacent mesh
         # regions, so each wave/XCD touches a compact working set (pairs
