@@ -112,14 +112,20 @@ def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
         rng_g = np.random.default_rng(1234)
         groups = rng_g.integers(0, args.ngroups, n_global).astype(np.uint16)
 
+    responses = None
+    if args.nscores > 1:
+        rng_r = np.random.default_rng(4321)
+        responses = rng_r.uniform(0.5, 2.0, size=(n_global, args.nscores))
     pe = pt._core.PartitionedEngine(mesh, n_global, device=device,
-                                    ngroups=args.ngroups)
+                                    ngroups=args.ngroups,
+                                    nscores=args.nscores)
     pe.localize(o)
     ends = (o, d)
 
     def step(k):
         # ping-pong continue-mode: origin == committed position
-        pe.step(ends[(k + 1) % 2], fly, w, groups=groups)
+        pe.step(ends[(k + 1) % 2], fly, w, groups=groups,
+                responses=responses)
 
     def barrier_sync():
         pe.synchronize()
@@ -168,6 +174,7 @@ def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
                 "parallelism": f"partitioned{world}-morton-ghost1-stateful",
                 "comm": "native-rccl",
                 "ngroups": args.ngroups,
+                "nscores": args.nscores,
                 "device": "gpu" if device != "cpu" else "cpu",
                 "resident_rank0": int(pe.resident),
                 "lost_particles": st["lost_particles"],
@@ -283,6 +290,10 @@ def main():
                          "suffix when < 1")
     ap.add_argument("--ngroups", type=int, default=1,
                     help="energy groups (random per-particle group indices)")
+    ap.add_argument("--nscores", type=int, default=1,
+                    help="simultaneous tally scores (random per-particle "
+                         "response multipliers); supported in replicated AND "
+                         "partitioned modes")
     ap.add_argument("--backend", type=str, default=None,
                     help="torch.distributed backend override (nccl/gloo)")
     ap.add_argument("--device-resident", action="store_true",
@@ -341,7 +352,7 @@ def main():
                                         device)
 
     eng = pt.TallyEngine(mesh, args.particles, device=device,
-                         ngroups=args.ngroups)
+                         ngroups=args.ngroups, nscores=args.nscores)
     p0, p1, flying, weights = make_box_histories(
         (1.0, 1.0, 1.0), args.particles, args.mean_chord, cells,
         seed=args.seed + rank, pinned=eng.is_gpu, sort=not args.no_sort,
@@ -350,6 +361,11 @@ def main():
     if args.ngroups > 1:
         rng_g = np.random.default_rng(1234 + rank)
         groups = rng_g.integers(0, args.ngroups, args.particles).astype(np.uint16)
+    responses = None
+    if args.nscores > 1:
+        rng_r = np.random.default_rng(4321 + rank)
+        responses = rng_r.uniform(0.5, 2.0,
+                                  size=(args.particles, args.nscores))
     eng.copy_initial_position(p0.reshape(-1))
     eng.synchronize()
 
@@ -373,9 +389,10 @@ def main():
             return
         o, d = ends[k % 2], ends[(k + 1) % 2]
         if args.continue_mode:
-            eng.move_continue(d, flying, weights)
+            eng.move_continue(d, flying, weights, responses=responses)
         else:
-            eng.move(o, d, flying, weights, groups=groups)
+            eng.move(o, d, flying, weights, groups=groups,
+                     responses=responses)
 
     comm = CommShim(args.native_comm, world, local, on_gpu)
 
@@ -433,6 +450,7 @@ def main():
                 "seq_len": None,
                 "parallelism": f"dp{world}-replicated-mesh",
                 "source_frac": args.source_frac,
+                "nscores": args.nscores,
                 "comm": "native-rccl" if args.native_comm else "torch-rccl",
                 "ngroups": args.ngroups,
                 "device": "gpu" if eng.is_gpu else "cpu",

@@ -59,14 +59,18 @@ public:
   // One transport step over the global batch.  Every rank passes the
   // same global arrays.  origin == nullptr means no particle was
   // resampled (continue from committed positions).  groups optional
-  // (requires ngroups > 1 at construction).
+  // (requires ngroups > 1 at construction); responses optional
+  // (n_global * nscores score multipliers, engine.h semantics) -- like
+  // dest/weights they are gathered by gid on whichever rank walks the
+  // particle, so they never ride the exchange records.
   virtual void step(const double *dest, const int8_t *flying,
                     const double *weights, int64_t n_global,
                     const double *origin = nullptr,
-                    const uint16_t *groups = nullptr) = 0;
+                    const uint16_t *groups = nullptr,
+                    const double *responses = nullptr) = 0;
 
   // Local tally scattered to global element ids and summed over ranks
-  // (nelems * ngroups doubles).
+  // (nscores * ngroups * nelems doubles).
   virtual std::vector<double> flux_global() = 0;
 
   // Residency / diagnostics.
@@ -87,7 +91,7 @@ public:
 // extract_submesh.
 std::unique_ptr<PartitionedEngine> make_partitioned_engine(
     const Mesh &full, int64_t n_global, Comm *comm, int rank, int world,
-    const std::string &device, int ngroups = 1,
+    const std::string &device, int ngroups = 1, int nscores = 1,
     const int32_t *owners = nullptr, int ghost_rings = 1);
 
 } // namespace pumitally

@@ -166,11 +166,13 @@ __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
                               const double *__restrict__ dest,
                               const double *__restrict__ w,
                               const uint16_t *__restrict__ grp,
+                              const double *__restrict__ resp, int nscores,
                               double *__restrict__ wpos,
                               double *__restrict__ wdest,
                               int32_t *__restrict__ welem,
                               double *__restrict__ ww,
-                              uint16_t *__restrict__ wgrp) {
+                              uint16_t *__restrict__ wgrp,
+                              double *__restrict__ wresp) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m; j += stride) {
     const int64_t g = list[j];
@@ -183,6 +185,9 @@ __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
     welem[j] = elem[g];
     ww[j] = w[g];
     if (wgrp) wgrp[j] = grp[g];
+    if (wresp)
+      for (int k = 0; k < nscores; ++k)
+        wresp[j * nscores + k] = resp[g * nscores + k];
   }
 }
 
@@ -329,15 +334,16 @@ Decomp build_decomp(const Mesh &full, int rank, int world,
 class GpuPartitionedEngine final : public PartitionedEngine {
 public:
   GpuPartitionedEngine(const Mesh &full, int64_t n, Comm *comm, int rank,
-                       int world, int device, int ngroups,
+                       int world, int device, int ngroups, int nscores,
                        const int32_t *owners, int ghost_rings)
       : n_(n), comm_(comm), rank_(rank), world_(world), device_(device),
-        ngroups_(ngroups < 1 ? 1 : ngroups), nelems_global_(full.nelems),
+        ngroups_(ngroups < 1 ? 1 : ngroups),
+        nscores_(nscores < 1 ? 1 : nscores), nelems_global_(full.nelems),
         dec_(build_decomp(full, rank, world, owners, ghost_rings)) {
     if (world_ > 1 && !comm_)
       throw std::runtime_error("PartitionedEngine: world > 1 needs a comm");
     PT_HIP_CHECK(hipSetDevice(device_));
-    eng_ = make_gpu_engine(dec_.sub.local, 1, device_, ngroups_, 1);
+    eng_ = make_gpu_engine(dec_.sub.local, 1, device_, ngroups_, nscores_);
     if (!eng_) throw std::runtime_error("no HIP device for PartitionedEngine");
     if (!eng_->device_mesh(&dmesh_))
       throw std::runtime_error("GPU engine did not expose its device mesh");
@@ -405,7 +411,8 @@ public:
           (void *)d_eject_, (void *)d_ctr_, (void *)d_wpos_, (void *)d_wdest_,
           (void *)d_welem_, (void *)d_ww_, (void *)d_wgrp_,
           (void *)d_wout_pos_, (void *)d_wout_elem_, (void *)d_wstatus_,
-          (void *)d_offs_, (void *)d_send_, (void *)d_recv_})
+          (void *)d_offs_, (void *)d_send_, (void *)d_recv_,
+          (void *)d_resp_, (void *)d_wresp_})
       if (p) (void)hipFree(p);
   }
 
@@ -451,11 +458,20 @@ public:
   }
 
   void step(const double *dest, const int8_t *flying, const double *weights,
-            int64_t n, const double *origin, const uint16_t *groups) override {
+            int64_t n, const double *origin, const uint16_t *groups,
+            const double *responses) override {
     check_n(n);
     PT_HIP_CHECK(hipSetDevice(device_));
     if (groups && ngroups_ <= 1)
       throw std::runtime_error("groups passed but ngroups == 1");
+    if (responses) {
+      if (!d_resp_) {
+        d_resp_ = pdmalloc<double>(n_ * nscores_);
+        d_wresp_ = pdmalloc<double>(n_ * nscores_);
+      }
+      PT_HIP_CHECK(hipMemcpy(d_resp_, responses, n_ * nscores_ * 8,
+                             hipMemcpyHostToDevice));
+    }
     eng_->synchronize();
     PT_HIP_CHECK(hipMemcpy(d_dest_, dest, n_ * 3 * 8, hipMemcpyHostToDevice));
     PT_HIP_CHECK(hipMemcpy(d_fly_, flying, n_, hipMemcpyHostToDevice));
@@ -498,8 +514,9 @@ public:
     for (int round = 0; round < max_rounds_; ++round) {
       if (nwalk > 0) {
         k_part_gather<<<pgrid(nwalk), kPBlock>>>(
-            d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_w_, d_grp_, d_wpos_,
-            d_wdest_, d_welem_, d_ww_, groups ? d_wgrp_ : nullptr);
+            d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_w_, d_grp_, d_resp_,
+            nscores_, d_wpos_, d_wdest_, d_welem_, d_ww_,
+            groups ? d_wgrp_ : nullptr, responses ? d_wresp_ : nullptr);
         PT_HIP_CHECK(hipGetLastError());
         // gather runs on the NULL stream; the walk runs on the inner
         // engine's NON-BLOCKING compute stream, which does not implicitly
@@ -508,7 +525,8 @@ public:
         PT_HIP_CHECK(hipDeviceSynchronize());
         eng_->walk_raw_device(nwalk, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
-                              groups ? d_wgrp_ : nullptr, nullptr);
+                              groups ? d_wgrp_ : nullptr,
+                              responses ? d_wresp_ : nullptr);
         k_part_collect<<<pgrid(nwalk), kPBlock>>>(
             d_list_, nwalk, d_wout_pos_, d_wout_elem_, d_wstatus_, d_pos_,
             d_elem_, d_esc_, d_res_, d_fgid_, d_fowner_, d_dep_, d_ctr_);
@@ -611,13 +629,14 @@ public:
 
   std::vector<double> flux_global() override {
     eng_->synchronize();
-    const std::vector<double> local = eng_->flux(); // nlocal*ngroups
+    const std::vector<double> local = eng_->flux(); // nscores*ngroups*nlocal
     const int64_t nl = dec_.sub.local.nelems;
-    std::vector<double> out((int64_t)nelems_global_ * ngroups_, 0.0);
-    for (int g = 0; g < ngroups_; ++g)
+    const int64_t slabs = (int64_t)nscores_ * ngroups_;
+    std::vector<double> out(slabs * nelems_global_, 0.0);
+    for (int64_t s = 0; s < slabs; ++s)
       for (int64_t t = 0; t < nl; ++t)
-        out[(int64_t)g * nelems_global_ + dec_.sub.elem_l2g[t]] +=
-            local[(int64_t)g * nl + t];
+        out[s * nelems_global_ + dec_.sub.elem_l2g[t]] +=
+            local[s * nl + t];
     if (world_ > 1)
       comm_->allreduce_sum(out.data(), (int64_t)out.size());
     return out;
@@ -732,7 +751,7 @@ private:
   int64_t n_;
   Comm *comm_;
   int rank_, world_, device_;
-  int ngroups_;
+  int ngroups_, nscores_ = 1;
   int64_t nelems_global_;
   Decomp dec_;
   std::unique_ptr<Engine> eng_;
@@ -746,6 +765,7 @@ private:
   int32_t *d_elem_ = nullptr;
   uint8_t *d_res_ = nullptr, *d_esc_ = nullptr;
   double *d_dest_ = nullptr, *d_w_ = nullptr, *d_orig_ = nullptr;
+  double *d_resp_ = nullptr, *d_wresp_ = nullptr;
   int8_t *d_fly_ = nullptr;
   uint16_t *d_grp_ = nullptr, *d_wgrp_ = nullptr;
   int32_t *d_list_ = nullptr, *d_eject_ = nullptr;
@@ -767,14 +787,15 @@ private:
 class CpuPartitionedEngine final : public PartitionedEngine {
 public:
   CpuPartitionedEngine(const Mesh &full, int64_t n, Comm *comm, int rank,
-                       int world, int ngroups, const int32_t *owners,
-                       int ghost_rings)
+                       int world, int ngroups, int nscores,
+                       const int32_t *owners, int ghost_rings)
       : n_(n), comm_(comm), rank_(rank), world_(world),
-        ngroups_(ngroups < 1 ? 1 : ngroups), nelems_global_(full.nelems),
+        ngroups_(ngroups < 1 ? 1 : ngroups),
+        nscores_(nscores < 1 ? 1 : nscores), nelems_global_(full.nelems),
         dec_(build_decomp(full, rank, world, owners, ghost_rings)) {
     if (world_ > 1 && !comm_)
       throw std::runtime_error("PartitionedEngine: world > 1 needs a comm");
-    eng_ = make_cpu_engine(dec_.sub.local, 1, ngroups_, 1);
+    eng_ = make_cpu_engine(dec_.sub.local, 1, ngroups_, nscores_);
     loc_tol_ = loc_tol_rel() * norm(full.bbox_hi - full.bbox_lo);
     pos_.assign(n_ * 3, 0.0);
     elem_.assign(n_, -1);
@@ -822,7 +843,8 @@ public:
   }
 
   void step(const double *dest, const int8_t *flying, const double *weights,
-            int64_t n, const double *origin, const uint16_t *groups) override {
+            int64_t n, const double *origin, const uint16_t *groups,
+            const double *responses) override {
     check_n(n);
     if (groups && ngroups_ <= 1)
       throw std::runtime_error("groups passed but ngroups == 1");
@@ -881,6 +903,7 @@ public:
         std::vector<int32_t> welem(m), wout_elem(m);
         std::vector<int8_t> wstatus(m);
         std::vector<uint16_t> wgrp(groups ? m : 0);
+        std::vector<double> wresp(responses ? m * nscores_ : 0);
         for (int64_t j = 0; j < m; ++j) {
           const int64_t g = list[j];
           for (int k = 0; k < 3; ++k) {
@@ -890,10 +913,14 @@ public:
           welem[j] = elem_[g];
           ww[j] = weights[g];
           if (groups) wgrp[j] = groups[g];
+          if (responses)
+            for (int k = 0; k < nscores_; ++k)
+              wresp[j * nscores_ + k] = responses[g * nscores_ + k];
         }
         eng_->walk_raw(m, wpos.data(), wdest.data(), welem.data(), ww.data(),
                        wout_pos.data(), wout_elem.data(), wstatus.data(),
-                       groups ? wgrp.data() : nullptr, nullptr);
+                       groups ? wgrp.data() : nullptr,
+                       responses ? wresp.data() : nullptr);
         for (int64_t j = 0; j < m; ++j) {
           const int64_t g = list[j];
           if (wstatus[j] == 2) {
@@ -971,11 +998,12 @@ public:
   std::vector<double> flux_global() override {
     const std::vector<double> local = eng_->flux();
     const int64_t nl = dec_.sub.local.nelems;
-    std::vector<double> out((int64_t)nelems_global_ * ngroups_, 0.0);
-    for (int g = 0; g < ngroups_; ++g)
+    const int64_t slabs = (int64_t)nscores_ * ngroups_;
+    std::vector<double> out(slabs * nelems_global_, 0.0);
+    for (int64_t s = 0; s < slabs; ++s)
       for (int64_t t = 0; t < nl; ++t)
-        out[(int64_t)g * nelems_global_ + dec_.sub.elem_l2g[t]] +=
-            local[(int64_t)g * nl + t];
+        out[s * nelems_global_ + dec_.sub.elem_l2g[t]] +=
+            local[s * nl + t];
     if (world_ > 1)
       comm_->allreduce_sum(out.data(), (int64_t)out.size());
     return out;
@@ -1011,7 +1039,7 @@ private:
   int64_t n_;
   Comm *comm_;
   int rank_, world_;
-  int ngroups_;
+  int ngroups_, nscores_ = 1;
   int64_t nelems_global_;
   Decomp dec_;
   std::unique_ptr<Engine> eng_;
@@ -1026,8 +1054,8 @@ private:
 
 std::unique_ptr<PartitionedEngine> make_partitioned_engine(
     const Mesh &full, int64_t n_global, Comm *comm, int rank, int world,
-    const std::string &device, int ngroups, const int32_t *owners,
-    int ghost_rings) {
+    const std::string &device, int ngroups, int nscores,
+    const int32_t *owners, int ghost_rings) {
   // the rare global-resolve path keeps a host copy of the full mesh
   auto full_copy = std::make_shared<Mesh>(full);
   auto locate = [full_copy](Vec3 q, double tol, bool *lo) {
@@ -1040,8 +1068,8 @@ std::unique_ptr<PartitionedEngine> make_partitioned_engine(
     int count = 0;
     if (hipGetDeviceCount(&count) == hipSuccess && count > ordinal) {
       auto e = std::make_unique<GpuPartitionedEngine>(
-          full, n_global, comm, rank, world, ordinal, ngroups, owners,
-          ghost_rings);
+          full, n_global, comm, rank, world, ordinal, ngroups, nscores,
+          owners, ghost_rings);
       e->full_locate_ = locate;
       return e;
     }
@@ -1050,8 +1078,8 @@ std::unique_ptr<PartitionedEngine> make_partitioned_engine(
       throw std::runtime_error("PartitionedEngine: HIP device unavailable");
   }
   auto e = std::make_unique<CpuPartitionedEngine>(full, n_global, comm, rank,
-                                                  world, ngroups, owners,
-                                                  ghost_rings);
+                                                  world, ngroups, nscores,
+                                                  owners, ghost_rings);
   e->full_locate_ = locate;
   return e;
 }

@@ -303,7 +303,7 @@ PYBIND11_MODULE(_core, m) {
   };
   py::class_<PyPartEngine>(m, "PartitionedEngine")
       .def(py::init([](const Mesh &mesh, int64_t n_global,
-                       const std::string &device, int ngroups,
+                       const std::string &device, int ngroups, int nscores,
                        py::object owners, int ghost_rings) {
              auto self = std::make_unique<PyPartEngine>();
              const EnvComm env = comm_env();
@@ -327,13 +327,13 @@ PYBIND11_MODULE(_core, m) {
                                 : "cpu";
              self->pe = make_partitioned_engine(mesh, n_global,
                                                 self->comm.get(), rank,
-                                                world, dev, ngroups, op,
-                                                ghost_rings);
+                                                world, dev, ngroups, nscores,
+                                                op, ghost_rings);
              return self;
            }),
            py::arg("mesh"), py::arg("n_global"), py::arg("device") = "auto",
-           py::arg("ngroups") = 1, py::arg("owners") = py::none(),
-           py::arg("ghost_rings") = 1)
+           py::arg("ngroups") = 1, py::arg("nscores") = 1,
+           py::arg("owners") = py::none(), py::arg("ghost_rings") = 1)
       .def_property_readonly("rank",
                              [](const PyPartEngine &s) { return s.pe->rank(); })
       .def_property_readonly(
@@ -360,7 +360,7 @@ PYBIND11_MODULE(_core, m) {
                   flying,
               py::array_t<double, py::array::c_style | py::array::forcecast>
                   weights,
-              py::object origin, py::object groups) {
+              py::object origin, py::object groups, py::object responses) {
              const int64_t n = s.pe->num_particles();
              if ((int64_t)dest.size() != n * 3 ||
                  (int64_t)flying.size() != n ||
@@ -386,12 +386,21 @@ PYBIND11_MODULE(_core, m) {
                  throw std::runtime_error("step: groups size mismatch");
                gp = garr.data();
              }
+             const double *rp = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast>
+                 rarr;
+             if (!responses.is_none()) {
+               rarr = responses.cast<py::array_t<
+                   double, py::array::c_style | py::array::forcecast>>();
+               rp = rarr.data();
+             }
              py::gil_scoped_release ng;
              s.pe->step(dest.data(), flying.data(), weights.data(), n, op,
-                        gp);
+                        gp, rp);
            },
            py::arg("dest"), py::arg("flying"), py::arg("weights"),
-           py::arg("origin") = py::none(), py::arg("groups") = py::none())
+           py::arg("origin") = py::none(), py::arg("groups") = py::none(),
+           py::arg("responses") = py::none())
       .def("flux_global",
            [](PyPartEngine &s) {
              std::vector<double> f;

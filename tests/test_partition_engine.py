@@ -222,3 +222,26 @@ def test_partition_engine_world2_gpu_shared(tmp_path):
     prepare/gather/walk/collect/pack/unpack kernels + host-staged
     exchange -- runs on real hardware with real handoffs."""
     _spawn2(tmp_path, "cuda:0")
+
+
+def test_partition_engine_scored_world1():
+    """nscores + responses through the stateful engine: responses are
+    gathered by gid like dest/weights (never shipped in records)."""
+    mesh = pt.build_box(3, 3, 3)
+    n = 120
+    rng = np.random.default_rng(17)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    p1 = rng.uniform(0.05, 0.95, size=(n, 3))
+    w = rng.uniform(0.1, 1.0, n)
+    resp = rng.uniform(0.5, 2.0, size=(n, 2))
+    fly = np.ones(n, np.int8)
+
+    pe = pt._core.PartitionedEngine(mesh, n, device="cpu", nscores=2)
+    pe.localize(p0.ravel())
+    pe.step(p1.ravel(), fly, w, responses=resp)
+
+    eng = pt.TallyEngine(mesh, n, device="cpu", nscores=2)
+    eng.copy_initial_position(p0.ravel())
+    eng.move(p0.ravel(), p1.ravel(), fly.copy(), w, responses=resp)
+    assert np.allclose(pe.flux_global(),
+                       np.asarray(eng.flux()).ravel(), atol=1e-12)
