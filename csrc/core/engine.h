@@ -117,12 +117,17 @@ public:
   // carries it in the exchange record).  responses (nullable): n*nscores
   // per-segment score multipliers, same semantics as move().
   // Synchronous; host memory.
+  // out_dest (optional, n*3): the walk's CURRENT destination at
+  // termination -- reflective restarts mirror it, periodic restarts
+  // translate it, so a handoff (status 2) must resume toward out_dest,
+  // not the original dest.
   virtual void walk_raw(int64_t n, const double *pos, const double *dest,
                         const int32_t *elem, const double *weights,
                         double *out_pos, int32_t *out_elem,
                         int8_t *out_status,
                         const uint16_t *groups = nullptr,
-                        const double *responses = nullptr) = 0;
+                        const double *responses = nullptr,
+                        double *out_dest = nullptr) = 0;
 
   // Device-resident walk_raw: every array already lives in this engine's
   // device memory (no staging at all -- the partitioned driver keeps its
@@ -133,10 +138,11 @@ public:
                                const double *d_weights, double *d_out_pos,
                                int32_t *d_out_elem, int8_t *d_out_status,
                                const uint16_t *d_groups = nullptr,
-                               const double *d_responses = nullptr) {
+                               const double *d_responses = nullptr,
+                               double *d_out_dest = nullptr) {
     (void)n; (void)d_pos; (void)d_dest; (void)d_elem; (void)d_weights;
     (void)d_out_pos; (void)d_out_elem; (void)d_out_status; (void)d_groups;
-    (void)d_responses;
+    (void)d_responses; (void)d_out_dest;
     throw std::runtime_error("walk_raw_device requires the GPU engine");
   }
 

@@ -209,7 +209,8 @@ public:
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status,
                 const uint16_t *groups = nullptr,
-                const double *responses = nullptr) override {
+                const double *responses = nullptr,
+                double *out_dest = nullptr) override {
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     const unsigned hw = std::thread::hardware_concurrency();
     if (n >= 65536 && hw > 1) {
@@ -225,8 +226,8 @@ public:
           int64_t my_lost = 0;
           for (int64_t i = lo; i < hi; ++i)
             walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
-                         out_elem, out_status, i, steps, partial[t].data(),
-                         my_lost);
+                         out_elem, out_status, out_dest, i, steps,
+                         partial[t].data(), my_lost);
           lost += my_lost;
         });
       }
@@ -239,7 +240,8 @@ public:
     int64_t lost = 0;
     for (int64_t i = 0; i < n; ++i)
       walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
-                   out_elem, out_status, i, steps, flux_.data(), lost);
+                   out_elem, out_status, out_dest, i, steps, flux_.data(),
+                   lost);
     stats_.lost_particles += lost;
   }
 
@@ -247,7 +249,8 @@ public:
                     const int32_t *elem, const double *weights,
                     const uint16_t *groups, const double *responses,
                     double *out_pos, int32_t *out_elem, int8_t *out_status,
-                    int64_t i, int steps, double *flux_out, int64_t &lost) {
+                    double *out_dest, int64_t i, int steps, double *flux_out,
+                    int64_t &lost) {
     {
       const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
       const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
@@ -270,17 +273,18 @@ public:
           mesh_.face_bc_bits.empty() ? nullptr : mesh_.face_bc_bits.data();
       const int32_t *pix =
           mesh_.periodic_idx.empty() ? nullptr : mesh_.periodic_idx.data();
+      Vec3 od{d.x, d.y, d.z};
       if (walk_fp32)
         walk_segment32<true>(mesh_.planes.data(), mesh_.planes32.data(),
                              mesh_.nbr.data(), elem[i], o, d, weights[i],
                              steps, add, &oe, &op, &esc, reflective, bc, pix,
                              mesh_.periodic_elem.data(),
-                             mesh_.periodic_shift.data());
+                             mesh_.periodic_shift.data(), &od);
       else
         walk_segment<true>(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o,
                            d, weights[i], steps, add, &oe, &op, &esc,
                            reflective, bc, pix, mesh_.periodic_elem.data(),
-                           mesh_.periodic_shift.data());
+                           mesh_.periodic_shift.data(), &od);
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;
@@ -297,6 +301,11 @@ public:
       out_pos[i * 3 + 1] = op.y;
       out_pos[i * 3 + 2] = op.z;
       out_status[i] = st;
+      if (out_dest) {
+        out_dest[i * 3] = od.x;
+        out_dest[i * 3 + 1] = od.y;
+        out_dest[i * 3 + 2] = od.z;
+      }
     }
   }
 

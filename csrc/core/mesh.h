@@ -121,6 +121,13 @@ struct SubMesh {
   std::vector<int64_t> elem_l2g;     // local elem -> global elem
   std::vector<int64_t> foreign_gid;  // handoff table k -> global elem id
   std::vector<int32_t> foreign_owner; // handoff table k -> owning part
+  // handoff table k -> periodic translation to apply to the particle's
+  // position AND destination when shipping through this face (zero for
+  // plain partition cuts; nonzero when the global face is a periodic
+  // pair whose partner element lives on another part).  Local periodic
+  // pairs (partner inside this submesh) are wired straight into
+  // local.periodic_* and never reach the handoff table.
+  std::vector<double> foreign_shift; // k*3
 };
 
 // Balanced spatial partition: elements sorted by Morton key of centroid,

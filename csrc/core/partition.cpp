@@ -163,14 +163,41 @@ SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
       if (local.nbr[t * 4 + f] != -1) continue;
       const int32_t gn = m.nbr[g * 4 + f];
       if (gn == -1) {
-        // true boundary: carry the per-face boundary condition over
-        if (m.face_is_reflective(g * 4 + f))
+        // true boundary: reflective bit, periodic pair, or vacuum
+        const int64_t gface = g * 4 + f;
+        if (m.face_is_reflective(gface))
           local.set_face_reflective(t * 4 + f);
+        if (!m.periodic_idx.empty() && m.periodic_idx[gface] >= 0) {
+          const int32_t kg = m.periodic_idx[gface];
+          const int32_t ge = m.periodic_elem[kg];
+          const double *sh = m.periodic_shift.data() + (int64_t)kg * 3;
+          if (g2l_elem[ge] >= 0) {
+            // partner element is in this submesh (owned or ghost):
+            // local periodic restart, exactly as on the full mesh
+            if (local.periodic_idx.empty())
+              local.periodic_idx.assign(ne * 4, -1);
+            const int32_t kl = (int32_t)local.periodic_elem.size();
+            local.periodic_elem.push_back((int32_t)g2l_elem[ge]);
+            local.periodic_shift.insert(local.periodic_shift.end(),
+                                        {sh[0], sh[1], sh[2]});
+            local.periodic_idx[t * 4 + f] = kl;
+          } else {
+            // partner lives on another part: hand off with the
+            // translation (the driver applies it to pos AND dest)
+            const int32_t k = (int32_t)sub.foreign_gid.size();
+            sub.foreign_gid.push_back(ge);
+            sub.foreign_owner.push_back(owners[ge]);
+            sub.foreign_shift.insert(sub.foreign_shift.end(),
+                                     {sh[0], sh[1], sh[2]});
+            local.nbr[t * 4 + f] = -(2 + k);
+          }
+        }
         continue;
       }
       const int32_t k = (int32_t)sub.foreign_gid.size();
       sub.foreign_gid.push_back(gn);
       sub.foreign_owner.push_back(owners[gn]);
+      sub.foreign_shift.insert(sub.foreign_shift.end(), {0.0, 0.0, 0.0});
       local.nbr[t * 4 + f] = -(2 + k);
     }
   }

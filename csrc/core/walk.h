@@ -340,6 +340,11 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
 
 // FluxAdd: functor void(int32_t elem, double contribution).  On the GPU this
 // performs atomicAdd into the flux array; on the serial CPU path a plain +=.
+// out_dest (optional): the walk's CURRENT destination at termination.
+// Reflective restarts mirror the destination and periodic restarts
+// translate it, so after either a handoff record must ship s.d -- not
+// the caller's original d -- or the receiving rank walks to a stale
+// point (the round-2 periodic-partitioned ping-pong bug).
 template <bool Periodic = false, class FluxAdd>
 PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
@@ -349,13 +354,15 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const uint32_t *__restrict__ face_bc = nullptr,
                         const int32_t *__restrict__ pidx = nullptr,
                         const int32_t *__restrict__ pelem = nullptr,
-                        const double *__restrict__ pshift = nullptr) {
+                        const double *__restrict__ pshift = nullptr,
+                        Vec3 *out_dest = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance<Periodic>(planes, nbr, s, max_steps, add, out_elem,
                                  out_pos, out_escaped, reflective, face_bc,
                                  pidx, pelem, pshift)) {
   }
+  if (out_dest) *out_dest = s.d;
 }
 
 template <bool Periodic = false, class FluxAdd>
@@ -368,13 +375,15 @@ PT_HD void walk_segment32(const Plane *__restrict__ planes,
                           const uint32_t *__restrict__ face_bc = nullptr,
                           const int32_t *__restrict__ pidx = nullptr,
                           const int32_t *__restrict__ pelem = nullptr,
-                          const double *__restrict__ pshift = nullptr) {
+                          const double *__restrict__ pshift = nullptr,
+                          Vec3 *out_dest = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
   while (!walk_advance32<Periodic>(planes, planes32, nbr, s, max_steps, add,
                                    out_elem, out_pos, out_escaped, reflective,
                                    face_bc, pidx, pelem, pshift)) {
   }
+  if (out_dest) *out_dest = s.d;
 }
 
 // Point-in-tet test against the 4 planes (signed distances, unit normals).

@@ -359,6 +359,7 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            double *__restrict__ out_pos,
                            int32_t *__restrict__ out_elem,
                            int8_t *__restrict__ out_status,
+                           double *__restrict__ out_dest,
                            double *__restrict__ flux,
                            unsigned long long *__restrict__ lost,
                            double *__restrict__ lostrec, int64_t n,
@@ -385,14 +386,15 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
         atomicAdd(&flux[goff + e], v);
       }
     };
+    Vec3 od{d.x, d.y, d.z};
     if constexpr (F32)
       walk_segment32<true>(planes, planes32, nbr, elem[i], o, d, weights[i],
                            max_steps, add, &oe, &op, &esc, reflective,
-                           face_bc, pidx, pelem, pshift);
+                           face_bc, pidx, pelem, pshift, &od);
     else
       walk_segment<true>(planes, nbr, elem[i], o, d, weights[i], max_steps,
                          add, &oe, &op, &esc, reflective, face_bc, pidx,
-                         pelem, pshift);
+                         pelem, pshift, &od);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -414,19 +416,34 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
     out_pos[i * 3 + 1] = op.y;
     out_pos[i * 3 + 2] = op.z;
     out_status[i] = st;
+    if (out_dest) {
+      out_dest[i * 3] = od.x;
+      out_dest[i * 3 + 1] = od.y;
+      out_dest[i * 3 + 2] = od.z;
+    }
   }
 }
 
-int flux_slices() {
-  static int v = [] {
-    const char *s = getenv("PUMITALLY_FLUX_SLICES");
-    int k = s ? atoi(s) : 1;
-    if (k < 1) k = 1;
-    if (k > 64) k = 64;
-    while (k & (k - 1)) k--; // power of two for the cheap in-kernel mask
-    return k;
-  }();
-  return v;
+// Tally slices: S independent copies of the flux array, selected by
+// blockIdx & (S-1), reduced at readout.  The hot-element (point-source)
+// mitigation: measured on MI355X (profiles/README.md round 2), the
+// config-4 contention stress at 10M particles through a 2%-cube source
+// runs 74.2 ms/step with 1 slice and 7.8 ms with 64 -- 9.5x -- while the
+// spread-source case is unchanged (5.48 -> 5.58 ms at 16).  Default is
+// adaptive: 64 slices when the extra copies fit a modest HBM budget
+// (trivial against 288 GB for 1M-tet meshes), halved until they do.
+int flux_slices(int64_t flux_doubles) {
+  const char *s = getenv("PUMITALLY_FLUX_SLICES");
+  int k = s ? atoi(s) : 64;
+  if (k < 1) k = 1;
+  if (k > 64) k = 64;
+  while (k & (k - 1)) k--; // power of two for the cheap in-kernel mask
+  if (!s) {
+    // adaptive: keep the slice copies under ~1 GiB
+    const int64_t budget = (int64_t)1 << 30;
+    while (k > 1 && flux_doubles * k * 8 > budget) k >>= 1;
+  }
+  return k;
 }
 
 int grid_cap() {
@@ -530,8 +547,8 @@ public:
     d_elem_ = dmalloc<int32_t>(n_);
     d_escaped_ = dmalloc<uint8_t>(n_);
     d_s2c_ = dmalloc<int32_t>(n_);
-    slices_ = flux_slices();
     fsz_ = mesh_.nelems * ngroups * nscores;
+    slices_ = flux_slices(fsz_);
     d_flux_ = dmalloc<double>(fsz_ * slices_);
     d_lost_ = dmalloc<unsigned long long>(1);
     d_loose_ = dmalloc<unsigned long long>(1);
@@ -694,7 +711,7 @@ public:
   struct WalkRawScratch {
     int64_t cap = 0;
     double *pos = nullptr, *dest = nullptr, *w = nullptr, *out_pos = nullptr,
-           *resp = nullptr;
+           *resp = nullptr, *out_dest = nullptr;
     int32_t *elem = nullptr, *out_elem = nullptr;
     int8_t *status = nullptr;
     uint16_t *groups = nullptr;
@@ -720,7 +737,8 @@ public:
     for (void *q : {(void *)wr_.pos, (void *)wr_.dest, (void *)wr_.w,
                     (void *)wr_.out_pos, (void *)wr_.elem,
                     (void *)wr_.out_elem, (void *)wr_.status,
-                    (void *)wr_.groups, (void *)wr_.resp})
+                    (void *)wr_.groups, (void *)wr_.resp,
+                    (void *)wr_.out_dest})
       if (q) (void)hipFree(q);
     wr_ = WalkRawScratch{};
   }
@@ -729,7 +747,8 @@ public:
                 const int32_t *elem, const double *weights, double *out_pos,
                 int32_t *out_elem, int8_t *out_status,
                 const uint16_t *groups = nullptr,
-                const double *responses = nullptr) override {
+                const double *responses = nullptr,
+                double *out_dest = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -747,17 +766,20 @@ public:
     if (responses)
       PT_HIP_CHECK(hipMemcpy(wr_.resp, responses, n * nscores * 8,
                              hipMemcpyHostToDevice));
+    if (out_dest && !wr_.out_dest) wr_.out_dest = dmalloc<double>(wr_.cap * 3);
     if (walk_fp32)
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
-          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status,
+          out_dest ? wr_.out_dest : nullptr, d_flux_, d_lost_,
           d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
           d_pidx_, d_pelem_, d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
-          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status, d_flux_, d_lost_,
+          dg, dr, wr_.out_pos, wr_.out_elem, wr_.status,
+          out_dest ? wr_.out_dest : nullptr, d_flux_, d_lost_,
           d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
           d_pidx_, d_pelem_, d_pshift_);
@@ -768,6 +790,9 @@ public:
     PT_HIP_CHECK(hipMemcpy(out_elem, wr_.out_elem, n * 4,
                            hipMemcpyDeviceToHost));
     PT_HIP_CHECK(hipMemcpy(out_status, wr_.status, n, hipMemcpyDeviceToHost));
+    if (out_dest)
+      PT_HIP_CHECK(hipMemcpy(out_dest, wr_.out_dest, n * 3 * 8,
+                             hipMemcpyDeviceToHost));
   }
 
   void walk_raw_device(int64_t n, const double *d_pos, const double *d_dest,
@@ -775,7 +800,8 @@ public:
                        double *d_out_pos, int32_t *d_out_elem,
                        int8_t *d_out_status,
                        const uint16_t *d_groups = nullptr,
-                       const double *d_responses = nullptr) override {
+                       const double *d_responses = nullptr,
+                       double *d_out_dest = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -783,14 +809,16 @@ public:
       k_walk_raw<true><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
-          d_flux_, d_lost_, d_lostrec_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+          d_out_dest, d_flux_, d_lost_, d_lostrec_, n, steps, reflective,
+          d_face_bc_, ngroups, mesh_.nelems, nscores, d_pidx_, d_pelem_,
+          d_pshift_);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
-          d_flux_, d_lost_, d_lostrec_, n, steps, reflective, d_face_bc_, ngroups,
-          mesh_.nelems, nscores, d_pidx_, d_pelem_, d_pshift_);
+          d_out_dest, d_flux_, d_lost_, d_lostrec_, n, steps, reflective,
+          d_face_bc_, ngroups, mesh_.nelems, nscores, d_pidx_, d_pelem_,
+          d_pshift_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
   }

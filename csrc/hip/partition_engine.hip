@@ -41,7 +41,12 @@ namespace {
                                hipGetErrorString(_e));                         \
   } while (0)
 
-constexpr int kRecW = 5; // [gid, px, py, pz, target_gid]
+// record: [gid, px, py, pz, target_gid, dx, dy, dz] -- the destination
+// travels with the handoff because reflective/periodic restarts inside
+// the walk MUTATE it (walk.h out_dest); the receiver's first continued
+// walk uses the record's dest, while weights/groups/responses are still
+// gathered by gid.
+constexpr int kRecW = 8;
 constexpr int kPBlock = 256;
 
 inline int pgrid(int64_t n) {
@@ -106,7 +111,7 @@ __global__ void k_part_claim_rest(const unsigned long long *__restrict__ claim,
   }
 }
 
-// dep entry: 6 doubles [gid, px, py, pz, target_gid, owner]
+// dep entry: 9 doubles [gid, px, py, pz, target_gid, owner, dx, dy, dz]
 __global__ void k_part_prepare(
     int64_t n, uint8_t *res /* read+clear, no restrict */,
     const uint8_t *__restrict__ esc, const int8_t *__restrict__ fly,
@@ -114,8 +119,9 @@ __global__ void k_part_prepare(
     int32_t *__restrict__ elem, const Plane *__restrict__ planes,
     GridView grid, const int32_t *__restrict__ lowner,
     const int32_t *__restrict__ l2g, int myrank, double tol,
-    int32_t *__restrict__ list, double *__restrict__ dep,
-    int32_t *__restrict__ eject, unsigned long long *__restrict__ ctr) {
+    const double *__restrict__ dest, int32_t *__restrict__ list,
+    double *__restrict__ dep, int32_t *__restrict__ eject,
+    unsigned long long *__restrict__ ctr) {
   // ctr: [0]=nwalk [1]=ndep [2]=neject [3]=relocated [4]=loose
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n; g += stride) {
@@ -137,12 +143,15 @@ __global__ void k_part_prepare(
           } else {
             // resampled into a ghost element: reroute to its owner
             const unsigned long long k = atomicAdd(&ctr[1], 1ull);
-            dep[k * 6] = (double)g;
-            dep[k * 6 + 1] = q.x;
-            dep[k * 6 + 2] = q.y;
-            dep[k * 6 + 3] = q.z;
-            dep[k * 6 + 4] = (double)l2g[le];
-            dep[k * 6 + 5] = (double)lowner[le];
+            dep[k * 9] = (double)g;
+            dep[k * 9 + 1] = q.x;
+            dep[k * 9 + 2] = q.y;
+            dep[k * 9 + 3] = q.z;
+            dep[k * 9 + 4] = (double)l2g[le];
+            dep[k * 9 + 5] = (double)lowner[le];
+            dep[k * 9 + 6] = dest[g * 3];
+            dep[k * 9 + 7] = dest[g * 3 + 1];
+            dep[k * 9 + 8] = dest[g * 3 + 2];
             res[g] = 0;
             continue;
           }
@@ -160,10 +169,15 @@ __global__ void k_part_prepare(
   }
 }
 
+// use_ovr: round >= 1, every list entry is a fresh arrival whose
+// destination came in its record (dest_ovr); round 0 reads the global
+// dest by gid.
 __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
                               const double *__restrict__ pos,
                               const int32_t *__restrict__ elem,
                               const double *__restrict__ dest,
+                              const double *__restrict__ dest_ovr,
+                              bool use_ovr,
                               const double *__restrict__ w,
                               const uint16_t *__restrict__ grp,
                               const double *__restrict__ resp, int nscores,
@@ -179,9 +193,10 @@ __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
     wpos[j * 3] = pos[g * 3];
     wpos[j * 3 + 1] = pos[g * 3 + 1];
     wpos[j * 3 + 2] = pos[g * 3 + 2];
-    wdest[j * 3] = dest[g * 3];
-    wdest[j * 3 + 1] = dest[g * 3 + 1];
-    wdest[j * 3 + 2] = dest[g * 3 + 2];
+    const double *dsrc = use_ovr ? dest_ovr : dest;
+    wdest[j * 3] = dsrc[g * 3];
+    wdest[j * 3 + 1] = dsrc[g * 3 + 1];
+    wdest[j * 3 + 2] = dsrc[g * 3 + 2];
     welem[j] = elem[g];
     ww[j] = w[g];
     if (wgrp) wgrp[j] = grp[g];
@@ -195,6 +210,7 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
                                const double *__restrict__ wout_pos,
                                const int32_t *__restrict__ wout_elem,
                                const int8_t *__restrict__ wstatus,
+                               const double *__restrict__ wout_dest,
                                double *__restrict__ pos,
                                int32_t *__restrict__ elem,
                                uint8_t *__restrict__ esc,
@@ -210,12 +226,15 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
     if (st == 2) {
       const int32_t k = -(wout_elem[j] + 2);
       const unsigned long long d = atomicAdd(&ctr[1], 1ull);
-      dep[d * 6] = (double)g;
-      dep[d * 6 + 1] = wout_pos[j * 3];
-      dep[d * 6 + 2] = wout_pos[j * 3 + 1];
-      dep[d * 6 + 3] = wout_pos[j * 3 + 2];
-      dep[d * 6 + 4] = (double)fgid[k];
-      dep[d * 6 + 5] = (double)fowner[k];
+      dep[d * 9] = (double)g;
+      dep[d * 9 + 1] = wout_pos[j * 3];
+      dep[d * 9 + 2] = wout_pos[j * 3 + 1];
+      dep[d * 9 + 3] = wout_pos[j * 3 + 2];
+      dep[d * 9 + 4] = (double)fgid[k];
+      dep[d * 9 + 5] = (double)fowner[k];
+      dep[d * 9 + 6] = wout_dest[j * 3];
+      dep[d * 9 + 7] = wout_dest[j * 3 + 1];
+      dep[d * 9 + 8] = wout_dest[j * 3 + 2];
       res[g] = 0;
     } else {
       pos[g * 3] = wout_pos[j * 3];
@@ -231,7 +250,7 @@ __global__ void k_part_count(const double *__restrict__ dep, int64_t m,
                              unsigned long long *__restrict__ dcnt) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride)
-    atomicAdd(&dcnt[(int)dep[i * 6 + 5]], 1ull);
+    atomicAdd(&dcnt[(int)dep[i * 9 + 5]], 1ull);
 }
 
 __global__ void k_part_pack(const double *__restrict__ dep, int64_t m,
@@ -240,13 +259,16 @@ __global__ void k_part_pack(const double *__restrict__ dep, int64_t m,
                             double *__restrict__ send) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride) {
-    const int o = (int)dep[i * 6 + 5];
+    const int o = (int)dep[i * 9 + 5];
     const int64_t s = offs[o] + (int64_t)atomicAdd(&cur[o], 1ull);
-    send[s * kRecW] = dep[i * 6];
-    send[s * kRecW + 1] = dep[i * 6 + 1];
-    send[s * kRecW + 2] = dep[i * 6 + 2];
-    send[s * kRecW + 3] = dep[i * 6 + 3];
-    send[s * kRecW + 4] = dep[i * 6 + 4];
+    send[s * kRecW] = dep[i * 9];
+    send[s * kRecW + 1] = dep[i * 9 + 1];
+    send[s * kRecW + 2] = dep[i * 9 + 2];
+    send[s * kRecW + 3] = dep[i * 9 + 3];
+    send[s * kRecW + 4] = dep[i * 9 + 4];
+    send[s * kRecW + 5] = dep[i * 9 + 6];
+    send[s * kRecW + 6] = dep[i * 9 + 7];
+    send[s * kRecW + 7] = dep[i * 9 + 8];
   }
 }
 
@@ -256,6 +278,7 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
                               int32_t *__restrict__ elem,
                               uint8_t *__restrict__ res,
                               uint8_t *__restrict__ esc,
+                              double *__restrict__ dest_ovr,
                               int32_t *__restrict__ list,
                               unsigned long long *__restrict__ ctr) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -267,6 +290,9 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
     pos[g * 3 + 1] = recv[i * kRecW + 2];
     pos[g * 3 + 2] = recv[i * kRecW + 3];
     elem[g] = g2l[(int64_t)recv[i * kRecW + 4]];
+    dest_ovr[g * 3] = recv[i * kRecW + 5];
+    dest_ovr[g * 3 + 1] = recv[i * kRecW + 6];
+    dest_ovr[g * 3 + 2] = recv[i * kRecW + 7];
     const unsigned long long k = atomicAdd(&ctr[0], 1ull);
     list[k] = (int32_t)g;
   }
@@ -310,6 +336,14 @@ struct Decomp {
 
 Decomp build_decomp(const Mesh &full, int rank, int world,
                     const int32_t *owners_in, int ghost_rings) {
+  if (full.has_periodic() && world > 1)
+    throw std::runtime_error(
+        "PartitionedEngine: periodic BCs with world > 1 are unsupported in "
+        "the stateful engine (a cross-rank wrap would need a translated "
+        "destination, but destinations are gathered by gid from the "
+        "untranslated global arrays); use the stateless "
+        "pumiumtally_amd.parallel.PartitionedTally driver, which ships the "
+        "translated destination in its exchange records");
   Decomp d;
   d.owners = owners_in
                  ? std::vector<int32_t>(owners_in, owners_in + full.nelems)
@@ -387,7 +421,8 @@ public:
 
     // work buffers
     d_list_ = pdmalloc<int32_t>(n_);
-    d_dep_ = pdmalloc<double>(n_ * 6);
+    d_dep_ = pdmalloc<double>(n_ * 9);
+    d_dest_ovr_ = pdmalloc<double>(n_ * 3);
     d_eject_ = pdmalloc<int32_t>(n_);
     d_ctr_ = pdmalloc<unsigned long long>(8 + 2 * world_);
     d_wpos_ = pdmalloc<double>(n_ * 3);
@@ -395,6 +430,7 @@ public:
     d_welem_ = pdmalloc<int32_t>(n_);
     d_ww_ = pdmalloc<double>(n_);
     d_wout_pos_ = pdmalloc<double>(n_ * 3);
+    d_wout_dest_ = pdmalloc<double>(n_ * 3);
     d_wout_elem_ = pdmalloc<int32_t>(n_);
     d_wstatus_ = pdmalloc<int8_t>(n_);
     d_offs_ = pdmalloc<int64_t>(world_);
@@ -407,10 +443,11 @@ public:
          {(void *)d_lowner_, (void *)d_l2g_, (void *)d_g2l_, (void *)d_fgid_,
           (void *)d_fowner_, (void *)d_pos_, (void *)d_elem_, (void *)d_res_,
           (void *)d_esc_, (void *)d_dest_, (void *)d_fly_, (void *)d_w_,
-          (void *)d_grp_, (void *)d_orig_, (void *)d_list_, (void *)d_dep_,
+          (void *)d_grp_, (void *)d_orig_, (void *)d_list_, (void *)d_dep_, (void *)d_dest_ovr_,
           (void *)d_eject_, (void *)d_ctr_, (void *)d_wpos_, (void *)d_wdest_,
           (void *)d_welem_, (void *)d_ww_, (void *)d_wgrp_,
-          (void *)d_wout_pos_, (void *)d_wout_elem_, (void *)d_wstatus_,
+          (void *)d_wout_pos_, (void *)d_wout_dest_,
+          (void *)d_wout_elem_, (void *)d_wstatus_,
           (void *)d_offs_, (void *)d_send_, (void *)d_recv_,
           (void *)d_resp_, (void *)d_wresp_})
       if (p) (void)hipFree(p);
@@ -493,7 +530,7 @@ public:
     k_part_prepare<<<pgrid(n_), kPBlock>>>(
         n_, d_res_, d_esc_, d_fly_, origin ? d_orig_ : nullptr, d_pos_,
         d_elem_, dmesh_.planes, dmesh_.grid, d_lowner_, d_l2g_, rank_,
-        loc_tol_, d_list_, d_dep_, d_eject_, d_ctr_);
+        loc_tol_, d_dest_, d_list_, d_dep_, d_eject_, d_ctr_);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipDeviceSynchronize());
     unsigned long long hctr[5];
@@ -506,7 +543,7 @@ public:
     // the gid+origin to ALL ranks?  No: we keep the full mesh on the
     // host (the caller constructed us with it) -- resolved in
     // host_resolve_ejects().
-    if (hctr[2] > 0) host_resolve_ejects((int64_t)hctr[2], origin);
+    if (hctr[2] > 0) host_resolve_ejects((int64_t)hctr[2], origin, dest);
 
     stats_.relocated += (int64_t)hctr[3];
     stats_.loose_localizations += (int64_t)hctr[4];
@@ -514,7 +551,8 @@ public:
     for (int round = 0; round < max_rounds_; ++round) {
       if (nwalk > 0) {
         k_part_gather<<<pgrid(nwalk), kPBlock>>>(
-            d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_w_, d_grp_, d_resp_,
+            d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
+            /*use_ovr=*/round > 0, d_w_, d_grp_, d_resp_,
             nscores_, d_wpos_, d_wdest_, d_welem_, d_ww_,
             groups ? d_wgrp_ : nullptr, responses ? d_wresp_ : nullptr);
         PT_HIP_CHECK(hipGetLastError());
@@ -526,10 +564,11 @@ public:
         eng_->walk_raw_device(nwalk, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
                               groups ? d_wgrp_ : nullptr,
-                              responses ? d_wresp_ : nullptr);
+                              responses ? d_wresp_ : nullptr, d_wout_dest_);
         k_part_collect<<<pgrid(nwalk), kPBlock>>>(
-            d_list_, nwalk, d_wout_pos_, d_wout_elem_, d_wstatus_, d_pos_,
-            d_elem_, d_esc_, d_res_, d_fgid_, d_fowner_, d_dep_, d_ctr_);
+            d_list_, nwalk, d_wout_pos_, d_wout_elem_, d_wstatus_,
+            d_wout_dest_, d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
+            d_fowner_, d_dep_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
         PT_HIP_CHECK(hipDeviceSynchronize());
       }
@@ -614,7 +653,8 @@ public:
       if (nrecv > 0) {
         k_part_unpack<<<pgrid(nrecv), kPBlock>>>(recv_ptr, nrecv, d_g2l_,
                                                  d_pos_, d_elem_, d_res_,
-                                                 d_esc_, d_list_, d_ctr_);
+                                                 d_esc_, d_dest_ovr_,
+                                                 d_list_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
         PT_HIP_CHECK(hipDeviceSynchronize());
       }
@@ -692,11 +732,12 @@ private:
   // the pieces needed): resolve globally, then inject reroute records
   // into the departure list (self-routes included -- they come back
   // through the exchange uniformly).
-  void host_resolve_ejects(int64_t ne, const double *origin) {
+  void host_resolve_ejects(int64_t ne, const double *origin,
+                           const double *dest_host) {
     std::vector<int32_t> gids(ne);
     PT_HIP_CHECK(
         hipMemcpy(gids.data(), d_eject_, ne * 4, hipMemcpyDeviceToHost));
-    std::vector<double> dep6;
+    std::vector<double> dep9;
     std::vector<int32_t> outside;
     for (int64_t i = 0; i < ne; ++i) {
       const int64_t g = gids[i];
@@ -705,27 +746,28 @@ private:
       const int32_t ge = full_locate_ ? full_locate_(q, loc_tol_, &lo) : -1;
       if (lo) stats_.loose_localizations++;
       if (ge >= 0) {
-        dep6.insert(dep6.end(),
+        dep9.insert(dep9.end(),
                     {(double)g, q.x, q.y, q.z, (double)ge,
-                     (double)dec_.owners[ge]});
+                     (double)dec_.owners[ge], dest_host[g * 3],
+                     dest_host[g * 3 + 1], dest_host[g * 3 + 2]});
       } else {
         outside.push_back((int32_t)g);
       }
     }
-    if (!dep6.empty()) {
-      // append to the device dep list (capacity n_*6 is plenty: ejects
+    if (!dep9.empty()) {
+      // append to the device dep list (capacity n_*9 is plenty: ejects
       // are a subset of residents)
       unsigned long long ndep = 0;
       PT_HIP_CHECK(hipMemcpy(&ndep, &d_ctr_[1], 8, hipMemcpyDeviceToHost));
-      const int64_t m = (int64_t)dep6.size() / 6;
-      PT_HIP_CHECK(hipMemcpy(d_dep_ + (int64_t)ndep * 6, dep6.data(),
-                             dep6.size() * 8, hipMemcpyHostToDevice));
+      const int64_t m = (int64_t)dep9.size() / 9;
+      PT_HIP_CHECK(hipMemcpy(d_dep_ + (int64_t)ndep * 9, dep9.data(),
+                             dep9.size() * 8, hipMemcpyHostToDevice));
       ndep += (unsigned long long)m;
       PT_HIP_CHECK(hipMemcpy(&d_ctr_[1], &ndep, 8, hipMemcpyHostToDevice));
       // the particle left this rank
       std::vector<uint8_t> zero(1, 0);
       for (int64_t i = 0; i < m; ++i) {
-        const int64_t g = (int64_t)dep6[i * 6];
+        const int64_t g = (int64_t)dep9[i * 9];
         PT_HIP_CHECK(
             hipMemcpy(d_res_ + g, zero.data(), 1, hipMemcpyHostToDevice));
       }
@@ -773,7 +815,8 @@ private:
   unsigned long long *d_ctr_ = nullptr;
   double *d_wpos_ = nullptr, *d_wdest_ = nullptr, *d_ww_ = nullptr;
   int32_t *d_welem_ = nullptr, *d_wout_elem_ = nullptr;
-  double *d_wout_pos_ = nullptr;
+  double *d_wout_pos_ = nullptr, *d_wout_dest_ = nullptr;
+  double *d_dest_ovr_ = nullptr;
   int8_t *d_wstatus_ = nullptr;
   int64_t *d_offs_ = nullptr;
   double *d_send_ = nullptr, *d_recv_ = nullptr;
@@ -850,7 +893,8 @@ public:
       throw std::runtime_error("groups passed but ngroups == 1");
     const Mesh &lm = dec_.sub.local;
     std::vector<int32_t> list;
-    std::vector<double> dep; // 6 doubles per entry
+    std::vector<double> dep; // 9 doubles per entry (see kRecW comment)
+    std::vector<double> ovr; // per-arrival destination overrides
     for (int64_t g = 0; g < n_; ++g) {
       if (!res_[g] || !flying[g]) continue;
       if (origin && !esc_[g]) {
@@ -864,7 +908,8 @@ public:
           if (le >= 0 && dec_.lowner[le] != rank_) {
             dep.insert(dep.end(), {(double)g, q.x, q.y, q.z,
                                    (double)dec_.l2g32[le],
-                                   (double)dec_.lowner[le]});
+                                   (double)dec_.lowner[le], dest[g * 3],
+                                   dest[g * 3 + 1], dest[g * 3 + 2]});
             res_[g] = 0;
             continue;
           }
@@ -876,7 +921,8 @@ public:
             if (lo2) stats_.loose_localizations++;
             if (ge >= 0) {
               dep.insert(dep.end(), {(double)g, q.x, q.y, q.z, (double)ge,
-                                     (double)dec_.owners[ge]});
+                                     (double)dec_.owners[ge], dest[g * 3],
+                                     dest[g * 3 + 1], dest[g * 3 + 2]});
               res_[g] = 0;
             } else {
               elem_[g] = -1;
@@ -900,15 +946,17 @@ public:
       if (!list.empty()) {
         const int64_t m = (int64_t)list.size();
         std::vector<double> wpos(m * 3), wdest(m * 3), ww(m), wout_pos(m * 3);
+        std::vector<double> wout_dest(m * 3);
         std::vector<int32_t> welem(m), wout_elem(m);
         std::vector<int8_t> wstatus(m);
         std::vector<uint16_t> wgrp(groups ? m : 0);
         std::vector<double> wresp(responses ? m * nscores_ : 0);
+        const double *dsrc = round > 0 ? ovr.data() : dest;
         for (int64_t j = 0; j < m; ++j) {
           const int64_t g = list[j];
           for (int k = 0; k < 3; ++k) {
             wpos[j * 3 + k] = pos_[g * 3 + k];
-            wdest[j * 3 + k] = dest[g * 3 + k];
+            wdest[j * 3 + k] = dsrc[g * 3 + k];
           }
           welem[j] = elem_[g];
           ww[j] = weights[g];
@@ -920,7 +968,7 @@ public:
         eng_->walk_raw(m, wpos.data(), wdest.data(), welem.data(), ww.data(),
                        wout_pos.data(), wout_elem.data(), wstatus.data(),
                        groups ? wgrp.data() : nullptr,
-                       responses ? wresp.data() : nullptr);
+                       responses ? wresp.data() : nullptr, wout_dest.data());
         for (int64_t j = 0; j < m; ++j) {
           const int64_t g = list[j];
           if (wstatus[j] == 2) {
@@ -929,7 +977,9 @@ public:
                        {(double)g, wout_pos[j * 3], wout_pos[j * 3 + 1],
                         wout_pos[j * 3 + 2],
                         (double)dec_.sub.foreign_gid[k],
-                        (double)dec_.sub.foreign_owner[k]});
+                        (double)dec_.sub.foreign_owner[k],
+                        wout_dest[j * 3], wout_dest[j * 3 + 1],
+                        wout_dest[j * 3 + 2]});
             res_[g] = 0;
           } else {
             for (int k = 0; k < 3; ++k) pos_[g * 3 + k] = wout_pos[j * 3 + k];
@@ -941,21 +991,23 @@ public:
       list.clear();
 
       // bucket by destination, exchange, unpack
-      const int64_t m = (int64_t)dep.size() / 6;
+      const int64_t m = (int64_t)dep.size() / 9;
       std::vector<int64_t> scounts(world_, 0);
       std::vector<double> send(m * kRecW);
       {
         std::vector<int64_t> offs(world_, 0), cur(world_, 0);
-        for (int64_t i = 0; i < m; ++i) scounts[(int)dep[i * 6 + 5]]++;
+        for (int64_t i = 0; i < m; ++i) scounts[(int)dep[i * 9 + 5]]++;
         int64_t acc = 0;
         for (int r = 0; r < world_; ++r) {
           offs[r] = acc;
           acc += scounts[r];
         }
         for (int64_t i = 0; i < m; ++i) {
-          const int o = (int)dep[i * 6 + 5];
+          const int o = (int)dep[i * 9 + 5];
           const int64_t s = offs[o] + cur[o]++;
-          for (int k = 0; k < kRecW; ++k) send[s * kRecW + k] = dep[i * 6 + k];
+          for (int k = 0; k < 5; ++k) send[s * kRecW + k] = dep[i * 9 + k];
+          for (int k = 0; k < 3; ++k)
+            send[s * kRecW + 5 + k] = dep[i * 9 + 6 + k];
         }
       }
       dep.clear();
@@ -977,6 +1029,7 @@ public:
         recv = comm_->alltoallv(send.data(), sc);
       }
       const int64_t nr = (int64_t)recv.size() / kRecW;
+      if (nr && ovr.empty()) ovr.assign(n_ * 3, 0.0);
       for (int64_t i = 0; i < nr; ++i) {
         const int64_t g = (int64_t)recv[i * kRecW];
         res_[g] = 1;
@@ -985,6 +1038,9 @@ public:
         pos_[g * 3 + 1] = recv[i * kRecW + 2];
         pos_[g * 3 + 2] = recv[i * kRecW + 3];
         elem_[g] = dec_.g2l[(int64_t)recv[i * kRecW + 4]];
+        ovr[g * 3] = recv[i * kRecW + 5];
+        ovr[g * 3 + 1] = recv[i * kRecW + 6];
+        ovr[g * 3 + 2] = recv[i * kRecW + 7];
         list.push_back((int32_t)g);
       }
       if (round == max_rounds_ - 1)

@@ -247,6 +247,16 @@ PYBIND11_MODULE(_core, m) {
                                            s.foreign_gid.size() * 8);
                                return a;
                              })
+      .def_property_readonly("foreign_shift",
+                             [](const SubMesh &s) {
+                               auto a = py::array_t<double>(
+                                   {(py::ssize_t)(s.foreign_shift.size() / 3),
+                                    (py::ssize_t)3});
+                               std::memcpy(a.mutable_data(),
+                                           s.foreign_shift.data(),
+                                           s.foreign_shift.size() * 8);
+                               return a;
+                             })
       .def_property_readonly("foreign_owner", [](const SubMesh &s) {
         auto a = py::array_t<int32_t>(s.foreign_owner.size());
         std::memcpy(a.mutable_data(), s.foreign_owner.data(),
@@ -665,14 +675,18 @@ PYBIND11_MODULE(_core, m) {
              auto out_pos = py::array_t<double>({n, (int64_t)3});
              auto out_elem = py::array_t<int32_t>(n);
              auto out_status = py::array_t<int8_t>(n);
+             auto out_dest = py::array_t<double>({n, (int64_t)3});
              {
                py::gil_scoped_release nogil;
                e.eng->walk_raw(n, pos.data(), dest.data(), elem.data(),
                                weights.data(), out_pos.mutable_data(),
                                out_elem.mutable_data(), out_status.mutable_data(),
-                               gp, rp);
+                               gp, rp, out_dest.mutable_data());
              }
-             return py::make_tuple(out_pos, out_elem, out_status);
+             // out_dest: the walk's final destination (mutated by
+             // reflective/periodic restarts); handoffs must resume
+             // toward it, not the original dest
+             return py::make_tuple(out_pos, out_elem, out_status, out_dest);
            },
            py::arg("pos"), py::arg("dest"), py::arg("elem"),
            py::arg("weights"), py::arg("groups") = py::none(),
@@ -683,18 +697,19 @@ PYBIND11_MODULE(_core, m) {
            [](PyEngine &e, int64_t n, uintptr_t pos, uintptr_t dest,
               uintptr_t elem, uintptr_t weights, uintptr_t out_pos,
               uintptr_t out_elem, uintptr_t out_status, uintptr_t groups,
-              uintptr_t responses) {
+              uintptr_t responses, uintptr_t out_dest) {
              py::gil_scoped_release nogil;
              e.eng->walk_raw_device(
                  n, (const double *)pos, (const double *)dest,
                  (const int32_t *)elem, (const double *)weights,
                  (double *)out_pos, (int32_t *)out_elem, (int8_t *)out_status,
-                 (const uint16_t *)groups, (const double *)responses);
+                 (const uint16_t *)groups, (const double *)responses,
+                 (double *)out_dest);
            },
            py::arg("n"), py::arg("pos"), py::arg("dest"), py::arg("elem"),
            py::arg("weights"), py::arg("out_pos"), py::arg("out_elem"),
            py::arg("out_status"), py::arg("groups") = 0,
-           py::arg("responses") = 0)
+           py::arg("responses") = 0, py::arg("out_dest") = 0)
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

@@ -160,10 +160,13 @@ class TallyEngine:
     def walk_raw(self, pos, dest, elem, weights, groups=None,
                  responses=None):
         """Batched raw segment walk (domain-decomposition support): returns
-        (out_pos, out_elem, status) with status 0=done 1=escaped 2=handoff
-        3=lost; tallies into this engine's flux.  groups: optional uint16
-        per-segment energy-group indices (flux row group*nelems+elem);
-        responses: optional n x nscores score multipliers."""
+        (out_pos, out_elem, status, out_dest) with status 0=done 1=escaped
+        2=handoff 3=lost; tallies into this engine's flux.  out_dest is the
+        walk's final destination -- reflective/periodic restarts mutate it,
+        and a handoff must resume toward out_dest, not the original dest.
+        groups: optional uint16 per-segment energy-group indices (flux row
+        group*nelems+elem); responses: optional n x nscores score
+        multipliers."""
         return self._eng.walk_raw(pos, dest, elem, weights, groups,
                                   responses)
 

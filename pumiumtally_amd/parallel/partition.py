@@ -54,11 +54,6 @@ class PartitionedTally:
         from .. import TallyEngine, _core, have_gpu
 
         mesh = self.mesh
-        if getattr(mesh, "has_periodic", False):
-            raise NotImplementedError(
-                "periodic BCs are not supported in partitioned mode: the "
-                "handoff record does not carry the accumulated translation. "
-                "Use the replicated DistributedTally for periodic meshes.")
         w = None if elem_weights is None else np.asarray(elem_weights, np.float64)
         self.owners = _core.partition_morton(mesh, self.world, w)
         self.sub = _core.extract_submesh(mesh, self.owners, self.rank,
@@ -68,6 +63,10 @@ class PartitionedTally:
         self.g2l[self.l2g] = np.arange(len(self.l2g))
         self.foreign_gid = self.sub.foreign_gid
         self.foreign_owner = self.sub.foreign_owner
+        # periodic cross-part faces carry a translation applied to BOTH the
+        # shipped position and destination (zero rows for plain cuts)
+        self.foreign_shift = np.asarray(self.sub.foreign_shift,
+                                        np.float64).reshape(-1, 3)
         device = self._device
         if device is None:
             device = f"cuda:{self.local}" if have_gpu() else "cpu"
@@ -159,16 +158,19 @@ class PartitionedTally:
         for _round in range(self.max_rounds):
             outbound = [np.zeros((0, rec_w)) for _ in range(self.world)]
             if len(elem):
-                out_pos, out_elem, status = self.engine.walk_raw(
+                out_pos, out_elem, status, out_dest = self.engine.walk_raw(
                     pos.ravel(), dst.ravel(), elem, wgt, grp, rsp)
                 hand = status == 2
                 if hand.any():
                     k = -(out_elem[hand].astype(np.int64) + 2)
                     tgt_gid = self.foreign_gid[k]
                     tgt_owner = self.foreign_owner[k]
+                    shift = self.foreign_shift[k]
                     g_col = (grp[hand] if grp is not None
                              else np.zeros(int(hand.sum()), np.uint16))
-                    cols = [out_pos[hand], dst[hand],
+                    # out_dest, not dst: reflective/periodic restarts
+                    # inside the walk mutate the destination
+                    cols = [out_pos[hand] + shift, out_dest[hand] + shift,
                             wgt[hand, None], tgt_gid[:, None].astype(np.float64),
                             g_col[:, None].astype(np.float64)]
                     if rsp is not None:
@@ -239,8 +241,9 @@ class PartitionedTally:
                 to(self.g2l),
                 to(np.asarray(self.foreign_gid, np.int64)),
                 to(np.asarray(self.foreign_owner, np.int64)),
+                to(self.foreign_shift),
             )
-        g2l_t, fg_t, fo_t = self._dev_tables
+        g2l_t, fg_t, fo_t, fs_t = self._dev_tables
 
         for _round in range(self.max_rounds):
             k = int(elem_t.numel())
@@ -248,13 +251,16 @@ class PartitionedTally:
                 out_pos = torch.empty((k, 3), dtype=torch.float64, device=dev)
                 out_elem = torch.empty(k, dtype=torch.int32, device=dev)
                 status = torch.empty(k, dtype=torch.int8, device=dev)
+                out_dest = torch.empty((k, 3), dtype=torch.float64,
+                                       device=dev)
                 torch.cuda.synchronize()  # inputs/outputs materialized
                 self.engine._eng.walk_raw_device(
                     k, pos_t.data_ptr(), dst_t.data_ptr(), elem_t.data_ptr(),
                     wgt_t.data_ptr(), out_pos.data_ptr(), out_elem.data_ptr(),
                     status.data_ptr(),
                     grp_t.data_ptr() if grp_t is not None else 0,
-                    rsp_t.data_ptr() if rsp_t is not None else 0)
+                    rsp_t.data_ptr() if rsp_t is not None else 0,
+                    out_dest.data_ptr())
                 hand = status == 2
                 nh = int(hand.sum())
             else:
@@ -262,8 +268,12 @@ class PartitionedTally:
             if nh:
                 idx = -(out_elem[hand].long() + 2)
                 tgt_owner = fo_t[idx]
+                shift = fs_t[idx]
                 order = torch.argsort(tgt_owner)
-                cols = [out_pos[hand], dst_t[hand], wgt_t[hand, None],
+                # out_dest, not dst: in-walk reflective/periodic restarts
+                # mutate the destination
+                cols = [out_pos[hand] + shift, out_dest[hand] + shift,
+                        wgt_t[hand, None],
                         fg_t[idx][:, None].double(),
                         (grp_t[hand][:, None].double() if grp_t is not None
                          else torch.zeros((nh, 1), dtype=torch.float64,

@@ -118,12 +118,111 @@ def test_periodic_validation_errors():
         m.set_periodic_faces(hi, lo[:-1], np.array([-1.0, 0.0, 0.0]))
 
 
-def test_periodic_partitioned_raises():
+def test_periodic_partitioned_world1():
+    """Periodic BCs in the stateless partitioned driver (round-1 closed
+    the feature out; now cross-part periodic faces carry the pair's
+    translation in the exchange record -- SubMesh.foreign_shift).  At
+    world 1 all pairs are local restarts; oracle = the replicated
+    periodic engine."""
     from pumiumtally_amd.parallel.partition import PartitionedTally
 
     m = _periodic_x_box()
-    with pytest.raises(NotImplementedError, match="periodic"):
-        PartitionedTally(m, device="cpu")
+    n = 60
+    rng = np.random.default_rng(7)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = o + rng.normal(0, 0.6, size=(n, 3))  # many segments wrap in x
+    d[:, 1:] = np.clip(d[:, 1:], 0.02, 0.98)
+    w = rng.uniform(0.5, 1.5, n)
+
+    ptal = PartitionedTally(m, device="cpu")
+    ptal.run_segments(o, d, w)
+    got = ptal.flux_global()
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(got, ref.flux(), atol=1e-12), \
+        np.abs(got - ref.flux()).max()
+
+
+def test_periodic_partitioned_cross_part_shift():
+    """Cross-part periodic handoff: partition the periodic box so the
+    x=0 and x=1 boundary elements land on DIFFERENT parts, then verify a
+    wrapping walk against the replicated periodic oracle.  Runs both
+    part engines in one process (world-1 semantics don't apply: we drive
+    the submeshes by hand through walk_raw and route the records
+    ourselves, exactly what the world-2 driver does)."""
+    from pumiumtally_amd import _core
+
+    m = _periodic_x_box()
+    # explicit owners: split by x (element centroid): left half part 0
+    cents = np.array([m.centroid(t) for t in range(m.nelems)])
+    owners = (cents[:, 0] > 0.5).astype(np.int32)
+    subs = [_core.extract_submesh(m, owners, p, 1) for p in range(2)]
+    shifts = [np.asarray(s.foreign_shift).reshape(-1, 3) for s in subs]
+    # the periodic pair faces must appear as foreign entries with a
+    # nonzero shift on at least one side
+    assert any(np.abs(sh).max() > 0.5 for sh in shifts if sh.size)
+
+    engines = [pt.TallyEngine(s.local, 1, device="cpu") for s in subs]
+    g2l = [np.full(m.nelems, -1, np.int64) for _ in range(2)]
+    for p in range(2):
+        g2l[p][np.asarray(subs[p].elem_l2g)] = np.arange(
+            len(subs[p].elem_l2g))
+
+    n = 40
+    rng = np.random.default_rng(3)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = o.copy()
+    d[:, 0] += rng.uniform(0.3, 1.4, n)  # all wrap through x=1
+    d[:, 1:] = np.clip(d[:, 1:] + rng.normal(0, 0.1, size=(n, 2)),
+                       0.02, 0.98)
+    w = rng.uniform(0.5, 1.5, n)
+
+    gids = m.locate(o)
+    assert (gids >= 0).all()
+    work = {p: [] for p in range(2)}
+    for i in range(n):
+        p = owners[gids[i]]
+        work[p].append((o[i], d[i], w[i], g2l[p][gids[i]]))
+    for _ in range(64):
+        moved = 0
+        next_work = {p: [] for p in range(2)}
+        for p in range(2):
+            if not work[p]:
+                continue
+            arr = work[p]
+            pos = np.array([a[0] for a in arr])
+            dst = np.array([a[1] for a in arr])
+            wg = np.array([a[2] for a in arr])
+            el = np.array([a[3] for a in arr], np.int32)
+            out_pos, out_elem, status, out_dest = engines[p].walk_raw(
+                pos.ravel(), dst.ravel(), el, wg)
+            for j in range(len(arr)):
+                if status[j] == 2:
+                    k = -(int(out_elem[j]) + 2)
+                    tg = int(np.asarray(subs[p].foreign_gid)[k])
+                    to = int(np.asarray(subs[p].foreign_owner)[k])
+                    sh = shifts[p][k]
+                    next_work[to].append(
+                        (out_pos[j] + sh, out_dest[j] + sh, wg[j],
+                         g2l[to][tg]))
+                    moved += 1
+        work = next_work
+        if moved == 0:
+            break
+    else:
+        raise AssertionError("handoff did not converge")
+
+    got = np.zeros(m.nelems)
+    for p in range(2):
+        got[np.asarray(subs[p].elem_l2g)] += np.asarray(engines[p].flux())
+
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(got, ref.flux(), atol=1e-12), \
+        np.abs(got - ref.flux()).max()
 
 
 def test_periodic_two_axis():
@@ -225,7 +324,7 @@ def test_periodic_walk_raw():
     seg = np.linalg.norm(d - o, axis=1)
 
     eng = pt.TallyEngine(m, 1, device="cpu")
-    out_pos, out_elem, status = eng.walk_raw(o.ravel(), d.ravel(), elem, w)
+    out_pos, out_elem, status, _ = eng.walk_raw(o.ravel(), d.ravel(), elem, w)
     assert (status == 0).all()  # wrapped walks reach their (translated) dest
     assert abs(eng.flux().sum() - (seg * w).sum()) < 1e-10 * (seg * w).sum()
 
