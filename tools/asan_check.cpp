@@ -2,11 +2,16 @@
 // partition), compiled with plain g++ (no HIP).  Runs the golden sequence
 // plus randomized fuzz through every core path.
 // Build/run: tools/asan_check.sh
+#include "../csrc/comm/comm.h"
+
+#include <sys/wait.h>
+#include <unistd.h>
 #include "../csrc/core/engine.h"
 
 #include <cassert>
 #include <cmath>
 #include <cstdio>
+#include <cstring>
 #include <random>
 #include <vector>
 
@@ -84,9 +89,49 @@ static void fuzz() {
   }
 }
 
+// world-2 TCP comm under the sanitizers: fork two ranks, run every
+// collective (ASan/UBSan follow the forked children).
+static void comm_world2() {
+  setenv("PUMITALLY_PORT", "29793", 1);
+  pid_t kid = fork();
+  const int rank = kid == 0 ? 1 : 0;
+  {
+    auto comm = make_tcp_comm(rank, 2, "127.0.0.1", 29793);
+    double v[4] = {1.0 * rank, 2.0, 3.0, 4.0};
+    comm->allreduce_sum(v, 4);
+    if (v[0] != 1.0 || v[1] != 4.0) abort();
+    comm->allreduce_max(v, 4);
+    int64_t c[2] = {rank, 7};
+    comm->allreduce_sum(c, 2);
+    if (c[0] != 1 || c[1] != 14) abort();
+    auto g = comm->allgather(10 + rank);
+    if (g[0] != 10 || g[1] != 11) abort();
+    char buf[8] = {0};
+    if (rank == 1) memcpy(buf, "ok", 3);
+    comm->bcast(buf, 8, 1);
+    if (buf[0] != 'o') abort();
+    std::vector<double> send = rank == 0
+        ? std::vector<double>{1.0, 2.0, 3.0}   // 1 to r0, 2 to r1
+        : std::vector<double>{9.0};            // 1 to r0, 0 to r1
+    std::vector<int64_t> cnt = rank == 0 ? std::vector<int64_t>{1, 2}
+                                         : std::vector<int64_t>{1, 0};
+    auto got = comm->alltoallv(send.data(), cnt);
+    if (rank == 0 && (got.size() != 2 || got[0] != 1.0 || got[1] != 9.0))
+      abort();
+    if (rank == 1 && (got.size() != 2 || got[0] != 2.0 || got[1] != 3.0))
+      abort();
+    comm->barrier();
+  }
+  if (kid == 0) _exit(0);
+  int st = 0;
+  waitpid(kid, &st, 0);
+  if (!WIFEXITED(st) || WEXITSTATUS(st) != 0) abort();
+}
+
 int main() {
   golden();
   fuzz();
+  comm_world2();
   printf("asan_check: PASS\n");
   return 0;
 }
