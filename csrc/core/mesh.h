@@ -154,6 +154,11 @@ std::vector<int32_t> partition_morton(const Mesh &m, int nparts,
 SubMesh extract_submesh(const Mesh &m, const std::vector<int32_t> &owners,
                         int part, int ghost_rings = 0);
 
+// Best-effort Omega_h binary stream reader (osh_omegah.cpp): probed
+// layout + orientation-free reconstruction, refuses anything it cannot
+// prove consistent.  Called by read_osh when the Omega_h magic is seen.
+Mesh read_osh_omegah_stream(const std::string &stream_path);
+
 // IO (implemented in mesh_io.cpp / osh_io.cpp)
 Mesh read_gmsh(const std::string &path);             // Gmsh .msh v2.2/v4.1 ASCII
 Mesh read_mesh(const std::string &path);             // dispatch on extension

@@ -58,16 +58,22 @@ Mesh read_osh(const std::string &dir) {
     // followed by an int32 format version; detect that case specifically
     // so the user gets conversion guidance instead of a generic mismatch.
     if ((unsigned char)magic[0] == 0xa1 && (unsigned char)magic[1] == 0x1a) {
-      int32_t ver = 0;
-      memcpy(&ver, magic + 2, 4);
-      throw std::runtime_error(
-          dir + " is an Omega_h binary mesh (stream version " +
-          std::to_string(ver) +
-          "). Its layout stores downward adjacency as code-aligned "
-          "dim->(dim-1) arrays that cannot be byte-validated in this "
-          "offline build; convert it once with Omega_h's osh2vtk or "
-          "export the source mesh as Gmsh .msh (ASCII or binary v2.2/"
-          "v4.1 both load here) and pass that instead.");
+      // Omega_h binary stream: attempt the best-effort probed reader
+      // (osh_omegah.cpp); it validates aggressively and throws with a
+      // precise diagnostic when the layout cannot be proven consistent.
+      f.close();
+      try {
+        return read_osh_omegah_stream(dir + "/0.osh");
+      } catch (const std::exception &e) {
+        int32_t ver = 0;
+        memcpy(&ver, magic + 2, 4);
+        throw std::runtime_error(
+            dir + " is an Omega_h binary mesh (stream version " +
+            std::to_string(ver) + ") and the best-effort reader could "
+            "not validate it: " + e.what() +
+            "\nFallback: export the source mesh as Gmsh .msh (ASCII or "
+            "binary v2.2/v4.1 both load here) and pass that instead.");
+      }
     }
     throw std::runtime_error(
         dir + " is not a pumitally .osh mesh (magic mismatch). If this is an "

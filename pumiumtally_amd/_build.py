@@ -23,6 +23,7 @@ SOURCES = [
     "csrc/core/mesh.cpp",
     "csrc/core/mesh_io.cpp",
     "csrc/core/osh_io.cpp",
+    "csrc/core/osh_omegah.cpp",
     "csrc/core/engine_cpu.cpp",
     "csrc/core/partition.cpp",
     "csrc/comm/comm_tcp.cpp",
@@ -81,7 +82,7 @@ def build(force: bool = False, verbose: bool = True) -> Path:
         + inc
         + [str(REPO / s) for s in SOURCES]
         # RCCL for the library-held multi-GPU comm (csrc/comm)
-        + ["-L/opt/rocm/lib", "-lrccl", "-Wl,-rpath,/opt/rocm/lib"]
+        + ["-L/opt/rocm/lib", "-lrccl", "-lz", "-Wl,-rpath,/opt/rocm/lib"]
         + ["-o", str(SO_PATH)]
     )
     if verbose:
