@@ -188,6 +188,10 @@ public:
     const Plane *planes = nullptr; // nelems*4
     const int32_t *nbr = nullptr;  // nelems*4
     GridView grid{};
+    // the engine's compute stream (hipStream_t; void* keeps this header
+    // HIP-free): sibling kernels launched here order naturally against
+    // walk_raw_device without device-wide fences
+    void *stream = nullptr;
   };
   virtual bool device_mesh(DeviceMeshView *out) const {
     (void)out;

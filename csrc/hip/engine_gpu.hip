@@ -946,6 +946,7 @@ public:
     out->planes = d_planes_;
     out->nbr = d_nbr_;
     out->grid = grid_view_;
+    out->stream = (void *)s_comp_;
     return true;
   }
 

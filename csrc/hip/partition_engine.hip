@@ -48,12 +48,28 @@ namespace {
 // gathered by gid.
 constexpr int kRecW = 8;
 constexpr int kPBlock = 256;
+constexpr int kMaxChunks = 8; // step() copy/walk pipeline depth
 
 inline int pgrid(int64_t n) {
   int64_t b = (n + kPBlock - 1) / kPBlock;
   if (b > 2048) b = 2048;
   if (b < 1) b = 1;
   return (int)b;
+}
+
+// Wave-aggregated append slot: one atomicAdd per 64-wide wavefront
+// instead of one per lane (the single global counter was 1.9 ms/step of
+// the prepare pass at 10M particles -- measured, profiles/README.md).
+// Call from exactly the lanes that append; the ballot captures them.
+__device__ __forceinline__ int64_t wave_append(unsigned long long *ctr) {
+  const unsigned long long mask = __ballot(1);
+  const int lane = (int)(threadIdx.x & 63u);
+  const int leader = __ffsll((long long)mask) - 1;
+  unsigned long long base = 0;
+  if (lane == leader)
+    base = atomicAdd(ctr, (unsigned long long)__popcll(mask));
+  base = (unsigned long long)__shfl((long long)base, leader);
+  return (int64_t)(base + __popcll(mask & ((1ull << lane) - 1ull)));
 }
 
 // ---------------------------------------------------------------------------
@@ -112,8 +128,12 @@ __global__ void k_part_claim_rest(const unsigned long long *__restrict__ claim,
 }
 
 // dep entry: 9 doubles [gid, px, py, pz, target_gid, owner, dx, dy, dz]
+// Operates on the gid range [g_lo, g_hi): step() pipelines chunks of the
+// batch so chunk c's prepare+walk overlap chunk c+1's H2D copies.  The
+// chunk's walk list lives in list[g_lo ...] with its own counter
+// (nwalk_ctr), so segments never collide.
 __global__ void k_part_prepare(
-    int64_t n, uint8_t *res /* read+clear, no restrict */,
+    int64_t g_lo, int64_t g_hi, uint8_t *res /* read+clear, no restrict */,
     const uint8_t *__restrict__ esc, const int8_t *__restrict__ fly,
     const double *__restrict__ orig, double *__restrict__ pos,
     int32_t *__restrict__ elem, const Plane *__restrict__ planes,
@@ -121,10 +141,12 @@ __global__ void k_part_prepare(
     const int32_t *__restrict__ l2g, int myrank, double tol,
     const double *__restrict__ dest, int32_t *__restrict__ list,
     double *__restrict__ dep, int32_t *__restrict__ eject,
-    unsigned long long *__restrict__ ctr) {
-  // ctr: [0]=nwalk [1]=ndep [2]=neject [3]=relocated [4]=loose
+    unsigned long long *__restrict__ ctr,
+    unsigned long long *__restrict__ nwalk_ctr) {
+  // ctr: [1]=ndep [2]=neject [3]=relocated [4]=loose
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n; g += stride) {
+  for (int64_t g = g_lo + blockIdx.x * blockDim.x + threadIdx.x; g < g_hi;
+       g += stride) {
     if (!res[g] || !fly[g]) continue;
     if (orig != nullptr && !esc[g]) {
       const Vec3 q{orig[g * 3], orig[g * 3 + 1], orig[g * 3 + 2]};
@@ -164,8 +186,7 @@ __global__ void k_part_prepare(
       }
     }
     if (elem[g] < 0) continue; // outside mesh: nothing to walk
-    const unsigned long long k = atomicAdd(&ctr[0], 1ull);
-    list[k] = (int32_t)g;
+    list[g_lo + wave_append(nwalk_ctr)] = (int32_t)g;
   }
 }
 
@@ -293,8 +314,7 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
     dest_ovr[g * 3] = recv[i * kRecW + 5];
     dest_ovr[g * 3 + 1] = recv[i * kRecW + 6];
     dest_ovr[g * 3 + 2] = recv[i * kRecW + 7];
-    const unsigned long long k = atomicAdd(&ctr[0], 1ull);
-    list[k] = (int32_t)g;
+    list[wave_append(&ctr[0])] = (int32_t)g;
   }
 }
 
@@ -381,6 +401,10 @@ public:
     if (!eng_) throw std::runtime_error("no HIP device for PartitionedEngine");
     if (!eng_->device_mesh(&dmesh_))
       throw std::runtime_error("GPU engine did not expose its device mesh");
+    cs_ = (hipStream_t)dmesh_.stream; // the engine's compute stream
+    PT_HIP_CHECK(hipStreamCreateWithFlags(&s_copy_, hipStreamNonBlocking));
+    for (auto &ev : ev_in_)
+      PT_HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
     loc_tol_ = loc_tol_rel() * norm(full.bbox_hi - full.bbox_lo);
 
     // lookup tables
@@ -424,7 +448,10 @@ public:
     d_dep_ = pdmalloc<double>(n_ * 9);
     d_dest_ovr_ = pdmalloc<double>(n_ * 3);
     d_eject_ = pdmalloc<int32_t>(n_);
-    d_ctr_ = pdmalloc<unsigned long long>(8 + 2 * world_);
+    // counters: [0..4] round/step counters, [8 .. 8+2*world) bucket
+    // counts+cursors, then kMaxChunks per-chunk walk counters
+    ctr_chunk0_ = 8 + 2 * world_;
+    d_ctr_ = pdmalloc<unsigned long long>(ctr_chunk0_ + kMaxChunks);
     d_wpos_ = pdmalloc<double>(n_ * 3);
     d_wdest_ = pdmalloc<double>(n_ * 3);
     d_welem_ = pdmalloc<int32_t>(n_);
@@ -439,6 +466,8 @@ public:
   ~GpuPartitionedEngine() override {
     (void)hipSetDevice(device_);
     (void)hipDeviceSynchronize();
+    for (auto &ev : ev_in_) (void)hipEventDestroy(ev);
+    (void)hipStreamDestroy(s_copy_);
     for (void *p :
          {(void *)d_lowner_, (void *)d_l2g_, (void *)d_g2l_, (void *)d_fgid_,
           (void *)d_fowner_, (void *)d_pos_, (void *)d_elem_, (void *)d_res_,
@@ -501,76 +530,118 @@ public:
     PT_HIP_CHECK(hipSetDevice(device_));
     if (groups && ngroups_ <= 1)
       throw std::runtime_error("groups passed but ngroups == 1");
-    if (responses) {
-      if (!d_resp_) {
-        d_resp_ = pdmalloc<double>(n_ * nscores_);
-        d_wresp_ = pdmalloc<double>(n_ * nscores_);
-      }
-      PT_HIP_CHECK(hipMemcpy(d_resp_, responses, n_ * nscores_ * 8,
-                             hipMemcpyHostToDevice));
+    if (responses && !d_resp_) {
+      d_resp_ = pdmalloc<double>(n_ * nscores_);
+      d_wresp_ = pdmalloc<double>(n_ * nscores_);
     }
     eng_->synchronize();
-    PT_HIP_CHECK(hipMemcpy(d_dest_, dest, n_ * 3 * 8, hipMemcpyHostToDevice));
-    PT_HIP_CHECK(hipMemcpy(d_fly_, flying, n_, hipMemcpyHostToDevice));
-    PT_HIP_CHECK(hipMemcpy(d_w_, weights, n_ * 8, hipMemcpyHostToDevice));
-    if (origin) {
-      if (!d_orig_) d_orig_ = pdmalloc<double>(n_ * 3);
-      PT_HIP_CHECK(
-          hipMemcpy(d_orig_, origin, n_ * 3 * 8, hipMemcpyHostToDevice));
-    }
-    if (groups) {
-      if (!d_grp_) {
-        d_grp_ = pdmalloc<uint16_t>(n_);
-        d_wgrp_ = pdmalloc<uint16_t>(n_);
-      }
-      PT_HIP_CHECK(hipMemcpy(d_grp_, groups, n_ * 2, hipMemcpyHostToDevice));
+    if (origin && !d_orig_) d_orig_ = pdmalloc<double>(n_ * 3);
+    if (groups && !d_grp_) {
+      d_grp_ = pdmalloc<uint16_t>(n_);
+      d_wgrp_ = pdmalloc<uint16_t>(n_);
     }
 
-    PT_HIP_CHECK(hipMemset(d_ctr_, 0, (8 + 2 * world_) * 8));
-    k_part_prepare<<<pgrid(n_), kPBlock>>>(
-        n_, d_res_, d_esc_, d_fly_, origin ? d_orig_ : nullptr, d_pos_,
-        d_elem_, dmesh_.planes, dmesh_.grid, d_lowner_, d_l2g_, rank_,
-        loc_tol_, d_dest_, d_list_, d_dep_, d_eject_, d_ctr_);
-    PT_HIP_CHECK(hipGetLastError());
-    PT_HIP_CHECK(hipDeviceSynchronize());
+    // Chunked copy/walk pipeline: all chunks' H2D copies are enqueued up
+    // front on the copy stream; chunk c's prepare/gather/walk on the
+    // compute stream waits only for chunk c's event, so the walk of
+    // chunk c overlaps the copies of chunks c+1..  (measured: the
+    // monolithic step serialized ~5.9 ms of H2D against ~6.5 ms of
+    // kernels).
+    const int C = (int)std::min<int64_t>(
+        kMaxChunks, std::max<int64_t>(1, n_ >> 21)); // >=2M particles: 4+
+    PT_HIP_CHECK(hipMemsetAsync(d_ctr_, 0,
+                                (ctr_chunk0_ + kMaxChunks) * 8, cs_));
+    std::vector<int64_t> clo(C + 1);
+    for (int c = 0; c <= C; ++c) clo[c] = n_ * c / C;
+    for (int c = 0; c < C; ++c) {
+      const int64_t lo = clo[c], hi = clo[c + 1];
+      PT_HIP_CHECK(hipMemcpyAsync(d_fly_ + lo, flying + lo, hi - lo,
+                                  hipMemcpyHostToDevice, s_copy_));
+      PT_HIP_CHECK(hipMemcpyAsync(d_w_ + lo, weights + lo, (hi - lo) * 8,
+                                  hipMemcpyHostToDevice, s_copy_));
+      PT_HIP_CHECK(hipMemcpyAsync(d_dest_ + lo * 3, dest + lo * 3,
+                                  (hi - lo) * 24, hipMemcpyHostToDevice,
+                                  s_copy_));
+      if (origin)
+        PT_HIP_CHECK(hipMemcpyAsync(d_orig_ + lo * 3, origin + lo * 3,
+                                    (hi - lo) * 24, hipMemcpyHostToDevice,
+                                    s_copy_));
+      if (groups)
+        PT_HIP_CHECK(hipMemcpyAsync(d_grp_ + lo, groups + lo, (hi - lo) * 2,
+                                    hipMemcpyHostToDevice, s_copy_));
+      if (responses)
+        PT_HIP_CHECK(hipMemcpyAsync(d_resp_ + lo * nscores_,
+                                    responses + lo * nscores_,
+                                    (hi - lo) * nscores_ * 8,
+                                    hipMemcpyHostToDevice, s_copy_));
+      PT_HIP_CHECK(hipEventRecord(ev_in_[c], s_copy_));
+    }
+
+    int64_t nwalk = 0; // round-0 total across chunks (for stats only)
+    for (int c = 0; c < C; ++c) {
+      const int64_t lo = clo[c], hi = clo[c + 1];
+      PT_HIP_CHECK(hipStreamWaitEvent(cs_, ev_in_[c], 0));
+      k_part_prepare<<<pgrid(hi - lo), kPBlock, 0, cs_>>>(
+          lo, hi, d_res_, d_esc_, d_fly_, origin ? d_orig_ : nullptr,
+          d_pos_, d_elem_, dmesh_.planes, dmesh_.grid, d_lowner_, d_l2g_,
+          rank_, loc_tol_, d_dest_, d_list_, d_dep_, d_eject_, d_ctr_,
+          &d_ctr_[ctr_chunk0_ + c]);
+      PT_HIP_CHECK(hipGetLastError());
+      PT_HIP_CHECK(hipStreamSynchronize(cs_));
+      unsigned long long cw = 0;
+      PT_HIP_CHECK(hipMemcpy(&cw, &d_ctr_[ctr_chunk0_ + c], 8,
+                             hipMemcpyDeviceToHost));
+      if (cw == 0) continue;
+      nwalk += (int64_t)cw;
+      k_part_gather<<<pgrid((int64_t)cw), kPBlock, 0, cs_>>>(
+          d_list_ + lo, (int64_t)cw, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
+          /*use_ovr=*/false, d_w_, d_grp_, d_resp_, nscores_, d_wpos_,
+          d_wdest_, d_welem_, d_ww_, groups ? d_wgrp_ : nullptr,
+          responses ? d_wresp_ : nullptr);
+      PT_HIP_CHECK(hipGetLastError());
+      eng_->walk_raw_device((int64_t)cw, d_wpos_, d_wdest_, d_welem_,
+                            d_ww_, d_wout_pos_, d_wout_elem_, d_wstatus_,
+                            groups ? d_wgrp_ : nullptr,
+                            responses ? d_wresp_ : nullptr, d_wout_dest_);
+      k_part_collect<<<pgrid((int64_t)cw), kPBlock, 0, cs_>>>(
+          d_list_ + lo, (int64_t)cw, d_wout_pos_, d_wout_elem_, d_wstatus_,
+          d_wout_dest_, d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
+          d_fowner_, d_dep_, d_ctr_);
+      PT_HIP_CHECK(hipGetLastError());
+    }
+    PT_HIP_CHECK(hipStreamSynchronize(cs_));
+    (void)nwalk;
+
     unsigned long long hctr[5];
     PT_HIP_CHECK(hipMemcpy(hctr, d_ctr_, 5 * 8, hipMemcpyDeviceToHost));
-    int64_t nwalk = (int64_t)hctr[0];
-
-    // host-resolve ejected relocations (rare): global locate on the full
-    // mesh is not available here -- route by asking every rank.  The
-    // canonical resolution: the mover does not know the target, so ship
-    // the gid+origin to ALL ranks?  No: we keep the full mesh on the
-    // host (the caller constructed us with it) -- resolved in
-    // host_resolve_ejects().
+    // host-resolve ejected relocations (rare: resampled origins that
+    // missed the local submesh grid); their reroute records join the dep
+    // list before the first exchange below
     if (hctr[2] > 0) host_resolve_ejects((int64_t)hctr[2], origin, dest);
-
     stats_.relocated += (int64_t)hctr[3];
     stats_.loose_localizations += (int64_t)hctr[4];
 
-    for (int round = 0; round < max_rounds_; ++round) {
-      if (nwalk > 0) {
-        k_part_gather<<<pgrid(nwalk), kPBlock>>>(
-            d_list_, nwalk, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
-            /*use_ovr=*/round > 0, d_w_, d_grp_, d_resp_,
+    // rounds 1+: walk lists come from exchanged records (round 0's walks
+    // already ran chunk-pipelined above)
+    int64_t nwalk_r = 0;
+    for (int round = 1; round <= max_rounds_; ++round) {
+      if (nwalk_r > 0) {
+        k_part_gather<<<pgrid(nwalk_r), kPBlock, 0, cs_>>>(
+            d_list_, nwalk_r, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
+            /*use_ovr=*/true, d_w_, d_grp_, d_resp_,
             nscores_, d_wpos_, d_wdest_, d_welem_, d_ww_,
             groups ? d_wgrp_ : nullptr, responses ? d_wresp_ : nullptr);
         PT_HIP_CHECK(hipGetLastError());
-        // gather runs on the NULL stream; the walk runs on the inner
-        // engine's NON-BLOCKING compute stream, which does not implicitly
-        // order against the NULL stream -- fence explicitly or the walk
-        // can read the scratch before gather wrote it
-        PT_HIP_CHECK(hipDeviceSynchronize());
-        eng_->walk_raw_device(nwalk, d_wpos_, d_wdest_, d_welem_, d_ww_,
+        eng_->walk_raw_device(nwalk_r, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
                               groups ? d_wgrp_ : nullptr,
                               responses ? d_wresp_ : nullptr, d_wout_dest_);
-        k_part_collect<<<pgrid(nwalk), kPBlock>>>(
-            d_list_, nwalk, d_wout_pos_, d_wout_elem_, d_wstatus_,
+        k_part_collect<<<pgrid(nwalk_r), kPBlock, 0, cs_>>>(
+            d_list_, nwalk_r, d_wout_pos_, d_wout_elem_, d_wstatus_,
             d_wout_dest_, d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
             d_fowner_, d_dep_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
-        PT_HIP_CHECK(hipDeviceSynchronize());
+        PT_HIP_CHECK(hipStreamSynchronize(cs_));
       }
       unsigned long long ndep = 0;
       PT_HIP_CHECK(
@@ -580,10 +651,10 @@ public:
       // per-destination counts -> offsets -> packed send buffer
       std::vector<int64_t> scounts(world_, 0);
       if (m > 0) {
-        PT_HIP_CHECK(hipMemset(&d_ctr_[8], 0, world_ * 8));
-        k_part_count<<<pgrid(m), kPBlock>>>(d_dep_, m, &d_ctr_[8]);
+        PT_HIP_CHECK(hipMemsetAsync(&d_ctr_[8], 0, world_ * 8, cs_));
+        k_part_count<<<pgrid(m), kPBlock, 0, cs_>>>(d_dep_, m, &d_ctr_[8]);
         PT_HIP_CHECK(hipGetLastError());
-        PT_HIP_CHECK(hipDeviceSynchronize());
+        PT_HIP_CHECK(hipStreamSynchronize(cs_));
         std::vector<unsigned long long> dc(world_);
         PT_HIP_CHECK(hipMemcpy(dc.data(), &d_ctr_[8], world_ * 8,
                                hipMemcpyDeviceToHost));
@@ -597,11 +668,13 @@ public:
         ensure_cap(&d_send_, &cap_send_, m * kRecW);
         PT_HIP_CHECK(hipMemcpy(d_offs_, offs.data(), world_ * 8,
                                hipMemcpyHostToDevice));
-        PT_HIP_CHECK(hipMemset(&d_ctr_[8 + world_], 0, world_ * 8));
-        k_part_pack<<<pgrid(m), kPBlock>>>(d_dep_, m, d_offs_,
-                                           &d_ctr_[8 + world_], d_send_);
+        PT_HIP_CHECK(hipMemsetAsync(&d_ctr_[8 + world_], 0, world_ * 8,
+                                    cs_));
+        k_part_pack<<<pgrid(m), kPBlock, 0, cs_>>>(d_dep_, m, d_offs_,
+                                                   &d_ctr_[8 + world_],
+                                                   d_send_);
         PT_HIP_CHECK(hipGetLastError());
-        PT_HIP_CHECK(hipDeviceSynchronize());
+        PT_HIP_CHECK(hipStreamSynchronize(cs_));
       }
 
       // global termination + recv counts
@@ -649,17 +722,16 @@ public:
       }
 
       // unpack received records; they form the next round's walk list
-      PT_HIP_CHECK(hipMemset(d_ctr_, 0, 2 * 8)); // nwalk, ndep
+      PT_HIP_CHECK(hipMemsetAsync(d_ctr_, 0, 2 * 8, cs_)); // nwalk, ndep
       if (nrecv > 0) {
-        k_part_unpack<<<pgrid(nrecv), kPBlock>>>(recv_ptr, nrecv, d_g2l_,
-                                                 d_pos_, d_elem_, d_res_,
-                                                 d_esc_, d_dest_ovr_,
-                                                 d_list_, d_ctr_);
+        k_part_unpack<<<pgrid(nrecv), kPBlock, 0, cs_>>>(
+            recv_ptr, nrecv, d_g2l_, d_pos_, d_elem_, d_res_, d_esc_,
+            d_dest_ovr_, d_list_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
-        PT_HIP_CHECK(hipDeviceSynchronize());
       }
-      nwalk = nrecv;
-      if (round == max_rounds_ - 1)
+      PT_HIP_CHECK(hipStreamSynchronize(cs_));
+      nwalk_r = nrecv;
+      if (round == max_rounds_)
         throw std::runtime_error("partitioned step did not converge in " +
                                  std::to_string(max_rounds_) +
                                  " handoff rounds");
@@ -798,6 +870,10 @@ private:
   Decomp dec_;
   std::unique_ptr<Engine> eng_;
   Engine::DeviceMeshView dmesh_{};
+  hipStream_t cs_ = nullptr;     // the inner engine's compute stream
+  hipStream_t s_copy_ = nullptr; // step-input H2D pipeline
+  hipEvent_t ev_in_[kMaxChunks] = {};
+  int ctr_chunk0_ = 8;
   double loc_tol_ = 0.0;
   mutable EngineStats stats_;
 
