@@ -548,7 +548,7 @@ public:
     // monolithic step serialized ~5.9 ms of H2D against ~6.5 ms of
     // kernels).
     const int C = (int)std::min<int64_t>(
-        kMaxChunks, std::max<int64_t>(1, n_ >> 21)); // >=2M particles: 4+
+        kMaxChunks, std::max<int64_t>(1, n_ >> 20)); // >=1M particles/chunk
     PT_HIP_CHECK(hipMemsetAsync(d_ctr_, 0,
                                 (ctr_chunk0_ + kMaxChunks) * 8, cs_));
     std::vector<int64_t> clo(C + 1);

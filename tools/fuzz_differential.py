@@ -81,9 +81,9 @@ def one_trial(rng, trial):
         rf = np.asarray(ref.flux()).reshape(G, m.nelems)
         assert np.allclose(f[k], rf, rtol=1e-9, atol=1e-12), (trial, "score")
 
-    # partitioned == plain (vacuum/reflective only; partitioned rejects
-    # periodic)
-    if bc != "periodic" and int(rng.integers(0, 2)):
+    # partitioned == plain (since round 2 this includes periodic: the
+    # exchange record carries the pair translation + walk destination)
+    if int(rng.integers(0, 2)):
         from pumiumtally_amd.parallel.partition import PartitionedTally
         ptal = PartitionedTally(m, device="cpu", ngroups=G, nscores=S,
                                 ghost_rings=int(rng.integers(0, 2)))
@@ -91,6 +91,19 @@ def one_trial(rng, trial):
         pf = np.asarray(ptal.flux_global()).reshape(S, G, m.nelems)
         assert np.allclose(pf, f, atol=1e-11), (trial, "partitioned",
                                                 np.abs(pf - f).max())
+
+    # stateful PartitionedEngine == plain (world-1; multi-step with the
+    # same segments replayed through localize+step; periodic is world-1
+    # legal there too)
+    if int(rng.integers(0, 2)):
+        pe = pt._core.PartitionedEngine(m, n, device="cpu", ngroups=G,
+                                        nscores=S)
+        pe.localize(o.ravel())
+        pe.step(d.ravel(), np.ones(n, np.int8), w, origin=o.ravel(),
+                groups=g, responses=r)
+        sf = np.asarray(pe.flux_global()).reshape(S, G, m.nelems)
+        assert np.allclose(sf, f, atol=1e-11), (trial, "stateful",
+                                                np.abs(sf - f).max())
 
     # fp32 traversal conserves the same totals
     if int(rng.integers(0, 2)):
