@@ -31,6 +31,7 @@ SOURCES = [
     "csrc/hip/engine_gpu.hip",
     "csrc/hip/partition_engine.hip",
     "csrc/api/PumiTally.cpp",
+    "csrc/api/pumitally_c.cpp",
     "csrc/pybind/module.cpp",
 ]
 HEADERS = [
@@ -41,6 +42,7 @@ HEADERS = [
     "csrc/comm/comm.h",
     "csrc/core/partition_engine.h",
     "csrc/api/PumiTally.h",
+    "csrc/api/pumitally_c.h",
 ]
 
 GFX_ARCH = os.environ.get("PUMITALLY_GFX_ARCH", "gfx950")
