@@ -1,8 +1,14 @@
-"""Multi-GPU example: replicated-mesh data parallelism and true domain
-decomposition, runnable under torchrun.
+"""Multi-GPU example: replicated-mesh data parallelism, stateless domain
+decomposition, and the stateful native partitioned engine.
 
     python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
         --master-addr 127.0.0.1 examples/distributed.py [--partitioned]
+
+Stateful native engine (the library's OWN comm -- no torchrun needed,
+plain processes with RANK/WORLD_SIZE env):
+
+    for r in 0 1; do RANK=$r WORLD_SIZE=2 \
+        python examples/distributed.py --stateful --device cpu & done; wait
 
 CPU rehearsal (no GPUs, gloo):
     python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
@@ -23,7 +29,10 @@ from pumiumtally_amd.parallel.partition import PartitionedTally  # noqa: E402
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--partitioned", action="store_true",
-                    help="domain decomposition instead of mesh replication")
+                    help="stateless domain decomposition (torch collectives)")
+    ap.add_argument("--stateful", action="store_true",
+                    help="stateful native PartitionedEngine (library comm, "
+                         "persistent particle residency)")
     ap.add_argument("--device", default=None)
     ap.add_argument("--particles", type=int, default=100_000)
     args = ap.parse_args()
@@ -31,6 +40,24 @@ def main():
     mesh = pt.build_box(12, 12, 12)
     rng = np.random.default_rng(0)
 
+    if args.stateful:
+        # Persistent residency: localize once, then step per transport
+        # step; cut-crossers migrate over the library's own RCCL/TCP comm.
+        n = args.particles
+        pe = pt._core.PartitionedEngine(mesh, n,
+                                        device=args.device or "auto")
+        o = rng.uniform(0.02, 0.98, size=(n, 3))
+        d = rng.uniform(0.02, 0.98, size=(n, 3))
+        w = rng.uniform(0.2, 1.0, n)
+        pe.localize(o.ravel())
+        for step in range(3):
+            dest = d if step % 2 == 0 else o
+            pe.step(dest.ravel(), np.ones(n, np.int8), w)
+        flux = pe.flux_global()
+        if pe.rank == 0:
+            print(f"[distributed] stateful world={pe.world} "
+                  f"resident(rank0)={pe.resident} total flux {flux.sum():.4f}")
+        return
     if args.partitioned:
         # True decomposition: every rank owns a Morton chunk of elements
         # (plus a ghost ring); all ranks pass the same global segment set
