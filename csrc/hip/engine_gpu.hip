@@ -369,8 +369,17 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            const int32_t *__restrict__ pidx = nullptr,
                            const int32_t *__restrict__ pelem = nullptr,
                            const double *__restrict__ pshift = nullptr) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+  // XCD-aware block->range remap, same as k_move: MI355X dispatches
+  // block b to XCD b%8; giving each XCD one contiguous (spatially
+  // compact, since callers keep lists near-Morton-ordered) index range
+  // keeps its private L2 on one mesh region.
+  const unsigned bpx = gridDim.x / 8u;
+  const unsigned vb = bpx ? (blockIdx.x % 8u) * bpx + blockIdx.x / 8u
+                          : blockIdx.x;
+  const int64_t per_blk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t base = (int64_t)vb * per_blk;
+  const int64_t end = base + per_blk < n ? base + per_blk : n;
+  for (int64_t i = base + threadIdx.x; i < end; i += blockDim.x) {
     const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
     const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
     int32_t oe;
