@@ -176,6 +176,11 @@ def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
                 "mean_chord_elems": args.mean_chord,
                 "seq_len": None,
                 "parallelism": f"partitioned{world}-morton-ghost1-stateful",
+                # steady-state steps use continue semantics (origin=None:
+                # no particle resampled, 33 B/particle uploaded); the
+                # replicated headline metric pays the reference-API origin
+                # leg (57 B) -- metric names differ accordingly
+                "step_mode": "continue",
                 "comm": "native-rccl",
                 "ngroups": args.ngroups,
                 "nscores": args.nscores,
