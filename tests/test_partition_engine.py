@@ -144,21 +144,24 @@ for s in range(steps):
     hist.append((origin, dest, flying, w))
     pos = np.where(flying[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
 
-pe = pt._core.PartitionedEngine(mesh, n, device=device)
+grp = rng.integers(0, 2, n).astype(np.uint16)
+rsp = rng.uniform(0.5, 2.0, size=(n, 2))
+pe = pt._core.PartitionedEngine(mesh, n, device=device, ngroups=2,
+                                nscores=2)
 assert pe.world == world, pe.world
 pe.localize(hist[0][0].ravel())
-# residency is a partition of the batch across ranks
-import numpy as _np
 for origin, dest, flying, w in hist:
-    pe.step(dest.ravel(), flying, w, origin=origin.ravel())
+    pe.step(dest.ravel(), flying, w, origin=origin.ravel(), groups=grp,
+            responses=rsp)
 got = pe.flux_global()
 
 if rank == 0:
-    eng = pt.TallyEngine(mesh, n, device="cpu")
+    eng = pt.TallyEngine(mesh, n, device="cpu", ngroups=2, nscores=2)
     eng.copy_initial_position(hist[0][0].ravel())
     for origin, dest, flying, w in hist:
-        eng.move(origin.ravel(), dest.ravel(), flying.copy(), w)
-    want = eng.flux()
+        eng.move(origin.ravel(), dest.ravel(), flying.copy(), w,
+                 groups=grp, responses=rsp)
+    want = np.asarray(eng.flux()).ravel()
     assert np.allclose(got, want, atol=1e-11), np.abs(got - want).max()
     print("PART_ENGINE_WORLD2_OK resident=", pe.resident)
 """
