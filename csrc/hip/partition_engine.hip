@@ -756,6 +756,7 @@ public:
 
   int64_t resident() const override {
     PT_HIP_CHECK(hipSetDevice(device_));
+    eng_->synchronize(); // order the NULL-stream copy behind cs_ work
     std::vector<uint8_t> r(n_);
     PT_HIP_CHECK(hipMemcpy(r.data(), d_res_, n_, hipMemcpyDeviceToHost));
     int64_t c = 0;
@@ -772,16 +773,19 @@ public:
   void synchronize() override { eng_->synchronize(); }
 
   std::vector<uint8_t> resident_mask() const override {
+    eng_->synchronize();
     std::vector<uint8_t> r(n_);
     PT_HIP_CHECK(hipMemcpy(r.data(), d_res_, n_, hipMemcpyDeviceToHost));
     return r;
   }
   std::vector<double> positions() const override {
+    eng_->synchronize();
     std::vector<double> p(n_ * 3);
     PT_HIP_CHECK(hipMemcpy(p.data(), d_pos_, n_ * 24, hipMemcpyDeviceToHost));
     return p;
   }
   std::vector<int32_t> elem_ids() const override {
+    eng_->synchronize();
     std::vector<int32_t> e(n_);
     PT_HIP_CHECK(hipMemcpy(e.data(), d_elem_, n_ * 4, hipMemcpyDeviceToHost));
     return e;
