@@ -69,6 +69,23 @@ public:
                     const uint16_t *groups = nullptr,
                     const double *responses = nullptr) = 0;
 
+  // ------ coupled-host path (rank-parallel transport codes) ---------
+  // When the host app is decomposed the same way (the config-5 multi-
+  // GPU coupling), it produces step inputs only for the particles IT
+  // owns.  resident_list() snapshots this rank's resident particles and
+  // returns their global ids; the NEXT step_local call passes arrays in
+  // exactly that order (n_local entries).  Unlike step(), nothing
+  // global-sized crosses PCIe, so the upload cost scales with the local
+  // batch -- the weak-scaling-correct form.  Handoff records carry
+  // weight/group/responses along with the destination (the receiver
+  // cannot gather them from arrays it never saw).
+  virtual std::vector<int64_t> resident_list() = 0;
+  virtual void step_local(const double *dest, const int8_t *flying,
+                          const double *weights, int64_t n_local,
+                          const double *origin = nullptr,
+                          const uint16_t *groups = nullptr,
+                          const double *responses = nullptr) = 0;
+
   // Local tally scattered to global element ids and summed over ranks
   // (nscores * ngroups * nelems doubles).
   virtual std::vector<double> flux_global() = 0;

@@ -41,18 +41,33 @@ namespace {
                                hipGetErrorString(_e));                         \
   } while (0)
 
-// record: [gid, px, py, pz, target_gid, dx, dy, dz] -- the destination
-// travels with the handoff because reflective/periodic restarts inside
-// the walk MUTATE it (walk.h out_dest); the receiver's first continued
-// walk uses the record's dest, while weights/groups/responses are still
-// gathered by gid.
-constexpr int kRecW = 8;
+// Record layout (runtime width rec_w = 9 + carry_grp + nscores):
+//   [0]=gid [1..3]=pos [4]=target_gid [5..7]=dest [8]=weight
+//   [9]=group (iff ngroups>1) [9+cg ..]=nscores response multipliers
+// The destination travels because reflective/periodic restarts inside
+// the walk MUTATE it (walk.h out_dest); weight/group/responses travel so
+// the coupled-host path (step_local) works -- the receiver never saw
+// the sender's input arrays.  In global-array mode the receiver scatters
+// the carried values into its own global-indexed buffers (identical
+// values, so both modes share one code path).
+// dep entries append the destination OWNER rank at column [rec_w].
 constexpr int kPBlock = 256;
 constexpr int kMaxChunks = 8; // step() copy/walk pipeline depth
 
 inline int pgrid(int64_t n) {
   int64_t b = (n + kPBlock - 1) / kPBlock;
   if (b > 2048) b = 2048;
+  if (b < 1) b = 1;
+  return (int)b;
+}
+
+// Streaming one-pass kernels (prepare/gather/collect/unpack) want one
+// item per thread: the grid-stride cap that suits the walk launches
+// leaves them latency-bound (~0.24 ms per 1.25M-item prepare measured
+// at 2048 blocks).
+inline int pgrid_flat(int64_t n) {
+  int64_t b = (n + kPBlock - 1) / kPBlock;
+  if (b > 65535) b = 65535;
   if (b < 1) b = 1;
   return (int)b;
 }
@@ -132,6 +147,16 @@ __global__ void k_part_claim_rest(const unsigned long long *__restrict__ claim,
 // batch so chunk c's prepare+walk overlap chunk c+1's H2D copies.  The
 // chunk's walk list lives in list[g_lo ...] with its own counter
 // (nwalk_ctr), so segments never collide.
+__device__ __forceinline__ void dep_fill_tail(
+    double *e, int rec_w, double weight, bool carry_grp, uint16_t grp,
+    const double *resp, int nscores) {
+  e[8] = weight;
+  int at = 9;
+  if (carry_grp) e[at++] = (double)grp;
+  for (int k = 0; k < nscores; ++k) e[at + k] = resp ? resp[k] : 1.0;
+  (void)rec_w;
+}
+
 __global__ void k_part_prepare(
     int64_t g_lo, int64_t g_hi, uint8_t *res /* read+clear, no restrict */,
     const uint8_t *__restrict__ esc, const int8_t *__restrict__ fly,
@@ -139,7 +164,9 @@ __global__ void k_part_prepare(
     int32_t *__restrict__ elem, const Plane *__restrict__ planes,
     GridView grid, const int32_t *__restrict__ lowner,
     const int32_t *__restrict__ l2g, int myrank, double tol,
-    const double *__restrict__ dest, int32_t *__restrict__ list,
+    const double *__restrict__ dest, const double *__restrict__ w,
+    const uint16_t *__restrict__ grp, const double *__restrict__ resp,
+    int nscores, bool carry_grp, int rec_w, int32_t *__restrict__ list,
     double *__restrict__ dep, int32_t *__restrict__ eject,
     unsigned long long *__restrict__ ctr,
     unsigned long long *__restrict__ nwalk_ctr) {
@@ -165,15 +192,20 @@ __global__ void k_part_prepare(
           } else {
             // resampled into a ghost element: reroute to its owner
             const unsigned long long k = atomicAdd(&ctr[1], 1ull);
-            dep[k * 9] = (double)g;
-            dep[k * 9 + 1] = q.x;
-            dep[k * 9 + 2] = q.y;
-            dep[k * 9 + 3] = q.z;
-            dep[k * 9 + 4] = (double)l2g[le];
-            dep[k * 9 + 5] = (double)lowner[le];
-            dep[k * 9 + 6] = dest[g * 3];
-            dep[k * 9 + 7] = dest[g * 3 + 1];
-            dep[k * 9 + 8] = dest[g * 3 + 2];
+            double *e = dep + k * (rec_w + 1);
+            e[0] = (double)g;
+            e[1] = q.x;
+            e[2] = q.y;
+            e[3] = q.z;
+            e[4] = (double)l2g[le];
+            e[5] = dest[g * 3];
+            e[6] = dest[g * 3 + 1];
+            e[7] = dest[g * 3 + 2];
+            dep_fill_tail(e, rec_w, w[g], carry_grp,
+                          grp ? grp[g] : (uint16_t)0,
+                          resp ? resp + (int64_t)g * nscores : nullptr,
+                          nscores);
+            e[rec_w] = (double)lowner[le];
             res[g] = 0;
             continue;
           }
@@ -232,6 +264,10 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
                                const int32_t *__restrict__ wout_elem,
                                const int8_t *__restrict__ wstatus,
                                const double *__restrict__ wout_dest,
+                               const double *__restrict__ ww,
+                               const uint16_t *__restrict__ wgrp,
+                               const double *__restrict__ wresp,
+                               int nscores, bool carry_grp, int rec_w,
                                double *__restrict__ pos,
                                int32_t *__restrict__ elem,
                                uint8_t *__restrict__ esc,
@@ -247,15 +283,19 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
     if (st == 2) {
       const int32_t k = -(wout_elem[j] + 2);
       const unsigned long long d = atomicAdd(&ctr[1], 1ull);
-      dep[d * 9] = (double)g;
-      dep[d * 9 + 1] = wout_pos[j * 3];
-      dep[d * 9 + 2] = wout_pos[j * 3 + 1];
-      dep[d * 9 + 3] = wout_pos[j * 3 + 2];
-      dep[d * 9 + 4] = (double)fgid[k];
-      dep[d * 9 + 5] = (double)fowner[k];
-      dep[d * 9 + 6] = wout_dest[j * 3];
-      dep[d * 9 + 7] = wout_dest[j * 3 + 1];
-      dep[d * 9 + 8] = wout_dest[j * 3 + 2];
+      double *e = dep + d * (rec_w + 1);
+      e[0] = (double)g;
+      e[1] = wout_pos[j * 3];
+      e[2] = wout_pos[j * 3 + 1];
+      e[3] = wout_pos[j * 3 + 2];
+      e[4] = (double)fgid[k];
+      e[5] = wout_dest[j * 3];
+      e[6] = wout_dest[j * 3 + 1];
+      e[7] = wout_dest[j * 3 + 2];
+      dep_fill_tail(e, rec_w, ww[j], carry_grp,
+                    wgrp ? wgrp[j] : (uint16_t)0,
+                    wresp ? wresp + (int64_t)j * nscores : nullptr, nscores);
+      e[rec_w] = (double)fowner[k];
       res[g] = 0;
     } else {
       pos[g * 3] = wout_pos[j * 3];
@@ -268,52 +308,61 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
 }
 
 __global__ void k_part_count(const double *__restrict__ dep, int64_t m,
+                             int rec_w,
                              unsigned long long *__restrict__ dcnt) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride)
-    atomicAdd(&dcnt[(int)dep[i * 9 + 5]], 1ull);
+    atomicAdd(&dcnt[(int)dep[i * (rec_w + 1) + rec_w]], 1ull);
 }
 
 __global__ void k_part_pack(const double *__restrict__ dep, int64_t m,
-                            const int64_t *__restrict__ offs,
+                            int rec_w, const int64_t *__restrict__ offs,
                             unsigned long long *__restrict__ cur,
                             double *__restrict__ send) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride) {
-    const int o = (int)dep[i * 9 + 5];
+    const double *e = dep + i * (rec_w + 1);
+    const int o = (int)e[rec_w];
     const int64_t s = offs[o] + (int64_t)atomicAdd(&cur[o], 1ull);
-    send[s * kRecW] = dep[i * 9];
-    send[s * kRecW + 1] = dep[i * 9 + 1];
-    send[s * kRecW + 2] = dep[i * 9 + 2];
-    send[s * kRecW + 3] = dep[i * 9 + 3];
-    send[s * kRecW + 4] = dep[i * 9 + 4];
-    send[s * kRecW + 5] = dep[i * 9 + 6];
-    send[s * kRecW + 6] = dep[i * 9 + 7];
-    send[s * kRecW + 7] = dep[i * 9 + 8];
+    for (int k = 0; k < rec_w; ++k) send[s * rec_w + k] = e[k];
   }
 }
 
 __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
+                              int rec_w, bool carry_grp, int nscores,
                               const int32_t *__restrict__ g2l,
                               double *__restrict__ pos,
                               int32_t *__restrict__ elem,
                               uint8_t *__restrict__ res,
                               uint8_t *__restrict__ esc,
                               double *__restrict__ dest_ovr,
+                              double *__restrict__ w_out,
+                              uint16_t *__restrict__ grp_out,
+                              double *__restrict__ resp_out,
                               int32_t *__restrict__ list,
                               unsigned long long *__restrict__ ctr) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride) {
-    const int64_t g = (int64_t)recv[i * kRecW];
+    const double *e = recv + i * rec_w;
+    const int64_t g = (int64_t)e[0];
     res[g] = 1;
     esc[g] = 0;
-    pos[g * 3] = recv[i * kRecW + 1];
-    pos[g * 3 + 1] = recv[i * kRecW + 2];
-    pos[g * 3 + 2] = recv[i * kRecW + 3];
-    elem[g] = g2l[(int64_t)recv[i * kRecW + 4]];
-    dest_ovr[g * 3] = recv[i * kRecW + 5];
-    dest_ovr[g * 3 + 1] = recv[i * kRecW + 6];
-    dest_ovr[g * 3 + 2] = recv[i * kRecW + 7];
+    pos[g * 3] = e[1];
+    pos[g * 3 + 1] = e[2];
+    pos[g * 3 + 2] = e[3];
+    elem[g] = g2l[(int64_t)e[4]];
+    dest_ovr[g * 3] = e[5];
+    dest_ovr[g * 3 + 1] = e[6];
+    dest_ovr[g * 3 + 2] = e[7];
+    // scatter the carried step inputs so later rounds (and the shared
+    // gather-by-gid path) see them regardless of which mode sent them
+    w_out[g] = e[8];
+    int at = 9;
+    if (carry_grp && grp_out) grp_out[g] = (uint16_t)e[at];
+    at += carry_grp ? 1 : 0;
+    if (resp_out)
+      for (int k = 0; k < nscores; ++k)
+        resp_out[(int64_t)g * nscores + k] = e[at + k];
     list[wave_append(&ctr[0])] = (int32_t)g;
   }
 }
@@ -332,6 +381,98 @@ __global__ void k_part_apply_outside(const int32_t *__restrict__ gids,
     pos[g * 3] = orig[g * 3];
     pos[g * 3 + 1] = orig[g * 3 + 1];
     pos[g * 3 + 2] = orig[g * 3 + 2];
+  }
+}
+
+// compact the resident mask into an ordered gid list (coupled-host
+// frame); wave_append keeps rough gid order
+__global__ void k_part_compact(const uint8_t *__restrict__ res, int64_t n,
+                               int32_t *__restrict__ frame,
+                               unsigned long long *__restrict__ ctr) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n; g += stride)
+    if (res[g]) frame[wave_append(ctr)] = (int32_t)g;
+}
+
+// coupled-host round 0: inputs indexed by frame position j, state by the
+// frame's gid; the gather into the walk scratch is fused (there is no
+// gid-indexed input array to gather from)
+__global__ void k_part_prepare_local(
+    int64_t nloc, const int32_t *__restrict__ frame, uint8_t *res,
+    const uint8_t *__restrict__ esc, const int8_t *__restrict__ fly,
+    const double *__restrict__ orig, double *__restrict__ pos,
+    int32_t *__restrict__ elem, const Plane *__restrict__ planes,
+    GridView grid, const int32_t *__restrict__ lowner,
+    const int32_t *__restrict__ l2g, int myrank, double tol,
+    const double *__restrict__ dest, const double *__restrict__ w,
+    const uint16_t *__restrict__ grp, const double *__restrict__ resp,
+    int nscores, bool carry_grp, int rec_w, int32_t *__restrict__ list,
+    double *__restrict__ wpos, double *__restrict__ wdest,
+    int32_t *__restrict__ welem, double *__restrict__ ww,
+    uint16_t *__restrict__ wgrp, double *__restrict__ wresp,
+    double *__restrict__ dep, int32_t *__restrict__ eject,
+    unsigned long long *__restrict__ ctr,
+    unsigned long long *__restrict__ nwalk_ctr) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = blockIdx.x * blockDim.x + threadIdx.x; j < nloc;
+       j += stride) {
+    const int64_t g = frame[j];
+    if (!res[g] || !fly[j]) continue;
+    Vec3 p{pos[g * 3], pos[g * 3 + 1], pos[g * 3 + 2]};
+    if (orig != nullptr && !esc[g]) {
+      const Vec3 q{orig[j * 3], orig[j * 3 + 1], orig[j * 3 + 2]};
+      if (q.x != p.x || q.y != p.y || q.z != p.z) {
+        atomicAdd(&ctr[3], 1ull);
+        bool lo = false;
+        const int32_t le = grid_locate(grid, planes, q, tol, &lo);
+        if (lo) atomicAdd(&ctr[4], 1ull);
+        if (le >= 0) {
+          if (lowner[le] == myrank) {
+            elem[g] = le;
+            pos[g * 3] = q.x;
+            pos[g * 3 + 1] = q.y;
+            pos[g * 3 + 2] = q.z;
+            p = q;
+          } else {
+            const unsigned long long k = atomicAdd(&ctr[1], 1ull);
+            double *e = dep + k * (rec_w + 1);
+            e[0] = (double)g;
+            e[1] = q.x;
+            e[2] = q.y;
+            e[3] = q.z;
+            e[4] = (double)l2g[le];
+            e[5] = dest[j * 3];
+            e[6] = dest[j * 3 + 1];
+            e[7] = dest[j * 3 + 2];
+            dep_fill_tail(e, rec_w, w[j], carry_grp,
+                          grp ? grp[j] : (uint16_t)0,
+                          resp ? resp + j * nscores : nullptr, nscores);
+            e[rec_w] = (double)lowner[le];
+            res[g] = 0;
+            continue;
+          }
+        } else {
+          const unsigned long long k = atomicAdd(&ctr[2], 1ull);
+          eject[k] = (int32_t)j; // frame index: host inputs are by j
+          continue;
+        }
+      }
+    }
+    if (elem[g] < 0) continue;
+    const int64_t slot = wave_append(nwalk_ctr);
+    list[slot] = (int32_t)g;
+    wpos[slot * 3] = p.x;
+    wpos[slot * 3 + 1] = p.y;
+    wpos[slot * 3 + 2] = p.z;
+    wdest[slot * 3] = dest[j * 3];
+    wdest[slot * 3 + 1] = dest[j * 3 + 1];
+    wdest[slot * 3 + 2] = dest[j * 3 + 2];
+    welem[slot] = elem[g];
+    ww[slot] = w[j];
+    if (wgrp) wgrp[slot] = grp[j];
+    if (wresp)
+      for (int k = 0; k < nscores; ++k)
+        wresp[slot * nscores + k] = resp[j * nscores + k];
   }
 }
 
@@ -443,9 +584,13 @@ public:
     d_fly_ = pdmalloc<int8_t>(n_);
     d_w_ = pdmalloc<double>(n_);
 
+    // record widths (see the layout comment at the top)
+    carry_grp_ = ngroups_ > 1;
+    rec_w_ = 9 + (carry_grp_ ? 1 : 0) + nscores_;
+
     // work buffers
     d_list_ = pdmalloc<int32_t>(n_);
-    d_dep_ = pdmalloc<double>(n_ * 9);
+    d_dep_ = pdmalloc<double>(n_ * (rec_w_ + 1));
     d_dest_ovr_ = pdmalloc<double>(n_ * 3);
     d_eject_ = pdmalloc<int32_t>(n_);
     // counters: [0..4] round/step counters, [8 .. 8+2*world) bucket
@@ -478,7 +623,9 @@ public:
           (void *)d_wout_pos_, (void *)d_wout_dest_,
           (void *)d_wout_elem_, (void *)d_wstatus_,
           (void *)d_offs_, (void *)d_send_, (void *)d_recv_,
-          (void *)d_resp_, (void *)d_wresp_})
+          (void *)d_resp_, (void *)d_wresp_, (void *)d_frame_,
+          (void *)d_ldest_, (void *)d_lw_, (void *)d_lorig_,
+          (void *)d_lresp_, (void *)d_lfly_, (void *)d_lgrp_})
       if (p) (void)hipFree(p);
   }
 
@@ -496,7 +643,7 @@ public:
     unsigned long long *d_claim = pdmalloc<unsigned long long>(nwords);
     PT_HIP_CHECK(hipMemset(d_claim, 0, nwords * 8));
     PT_HIP_CHECK(hipMemset(d_ctr_, 0, 8 * 8));
-    k_part_localize<<<pgrid(n_), kPBlock>>>(
+    k_part_localize<<<pgrid_flat(n_), kPBlock>>>(
         dmesh_.planes, dmesh_.grid, d_lowner_, rank_, d_dest_, n_, loc_tol_,
         d_pos_, d_elem_, d_res_, d_esc_, d_claim, &d_ctr_[4]);
     PT_HIP_CHECK(hipGetLastError());
@@ -512,7 +659,7 @@ public:
                              hipMemcpyHostToDevice));
     }
     if (rank_ == 0) {
-      k_part_claim_rest<<<pgrid(n_), kPBlock>>>(d_claim, d_dest_, n_, d_pos_,
+      k_part_claim_rest<<<pgrid_flat(n_), kPBlock>>>(d_claim, d_dest_, n_, d_pos_,
                                                 d_elem_, d_res_, d_esc_);
       PT_HIP_CHECK(hipGetLastError());
     }
@@ -581,11 +728,12 @@ public:
     for (int c = 0; c < C; ++c) {
       const int64_t lo = clo[c], hi = clo[c + 1];
       PT_HIP_CHECK(hipStreamWaitEvent(cs_, ev_in_[c], 0));
-      k_part_prepare<<<pgrid(hi - lo), kPBlock, 0, cs_>>>(
+      k_part_prepare<<<pgrid_flat(hi - lo), kPBlock, 0, cs_>>>(
           lo, hi, d_res_, d_esc_, d_fly_, origin ? d_orig_ : nullptr,
           d_pos_, d_elem_, dmesh_.planes, dmesh_.grid, d_lowner_, d_l2g_,
-          rank_, loc_tol_, d_dest_, d_list_, d_dep_, d_eject_, d_ctr_,
-          &d_ctr_[ctr_chunk0_ + c]);
+          rank_, loc_tol_, d_dest_, d_w_, groups ? d_grp_ : nullptr,
+          responses ? d_resp_ : nullptr, nscores_, carry_grp_, rec_w_,
+          d_list_, d_dep_, d_eject_, d_ctr_, &d_ctr_[ctr_chunk0_ + c]);
       PT_HIP_CHECK(hipGetLastError());
       PT_HIP_CHECK(hipStreamSynchronize(cs_));
       unsigned long long cw = 0;
@@ -593,7 +741,7 @@ public:
                              hipMemcpyDeviceToHost));
       if (cw == 0) continue;
       nwalk += (int64_t)cw;
-      k_part_gather<<<pgrid((int64_t)cw), kPBlock, 0, cs_>>>(
+      k_part_gather<<<pgrid_flat((int64_t)cw), kPBlock, 0, cs_>>>(
           d_list_ + lo, (int64_t)cw, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
           /*use_ovr=*/false, d_w_, d_grp_, d_resp_, nscores_, d_wpos_,
           d_wdest_, d_welem_, d_ww_, groups ? d_wgrp_ : nullptr,
@@ -603,9 +751,11 @@ public:
                             d_ww_, d_wout_pos_, d_wout_elem_, d_wstatus_,
                             groups ? d_wgrp_ : nullptr,
                             responses ? d_wresp_ : nullptr, d_wout_dest_);
-      k_part_collect<<<pgrid((int64_t)cw), kPBlock, 0, cs_>>>(
+      k_part_collect<<<pgrid_flat((int64_t)cw), kPBlock, 0, cs_>>>(
           d_list_ + lo, (int64_t)cw, d_wout_pos_, d_wout_elem_, d_wstatus_,
-          d_wout_dest_, d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
+          d_wout_dest_, d_ww_, groups ? d_wgrp_ : nullptr,
+          responses ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
+          d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
           d_fowner_, d_dep_, d_ctr_);
       PT_HIP_CHECK(hipGetLastError());
     }
@@ -621,24 +771,31 @@ public:
     stats_.relocated += (int64_t)hctr[3];
     stats_.loose_localizations += (int64_t)hctr[4];
 
-    // rounds 1+: walk lists come from exchanged records (round 0's walks
-    // already ran chunk-pipelined above)
+    run_exchange_rounds(groups != nullptr, responses != nullptr);
+    stats_.moves++;
+  }
+
+  // rounds 1+ of a step: walk lists come from exchanged records (round
+  // 0's walks already ran); shared by step() and step_local().
+  void run_exchange_rounds(bool groups_used, bool resp_used) {
     int64_t nwalk_r = 0;
     for (int round = 1; round <= max_rounds_; ++round) {
       if (nwalk_r > 0) {
-        k_part_gather<<<pgrid(nwalk_r), kPBlock, 0, cs_>>>(
+        k_part_gather<<<pgrid_flat(nwalk_r), kPBlock, 0, cs_>>>(
             d_list_, nwalk_r, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
             /*use_ovr=*/true, d_w_, d_grp_, d_resp_,
             nscores_, d_wpos_, d_wdest_, d_welem_, d_ww_,
-            groups ? d_wgrp_ : nullptr, responses ? d_wresp_ : nullptr);
+            groups_used ? d_wgrp_ : nullptr, resp_used ? d_wresp_ : nullptr);
         PT_HIP_CHECK(hipGetLastError());
         eng_->walk_raw_device(nwalk_r, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
-                              groups ? d_wgrp_ : nullptr,
-                              responses ? d_wresp_ : nullptr, d_wout_dest_);
-        k_part_collect<<<pgrid(nwalk_r), kPBlock, 0, cs_>>>(
+                              groups_used ? d_wgrp_ : nullptr,
+                              resp_used ? d_wresp_ : nullptr, d_wout_dest_);
+        k_part_collect<<<pgrid_flat(nwalk_r), kPBlock, 0, cs_>>>(
             d_list_, nwalk_r, d_wout_pos_, d_wout_elem_, d_wstatus_,
-            d_wout_dest_, d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
+            d_wout_dest_, d_ww_, groups_used ? d_wgrp_ : nullptr,
+            resp_used ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
+            d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
             d_fowner_, d_dep_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
         PT_HIP_CHECK(hipStreamSynchronize(cs_));
@@ -652,7 +809,8 @@ public:
       std::vector<int64_t> scounts(world_, 0);
       if (m > 0) {
         PT_HIP_CHECK(hipMemsetAsync(&d_ctr_[8], 0, world_ * 8, cs_));
-        k_part_count<<<pgrid(m), kPBlock, 0, cs_>>>(d_dep_, m, &d_ctr_[8]);
+        k_part_count<<<pgrid_flat(m), kPBlock, 0, cs_>>>(d_dep_, m, rec_w_,
+                                                       &d_ctr_[8]);
         PT_HIP_CHECK(hipGetLastError());
         PT_HIP_CHECK(hipStreamSynchronize(cs_));
         std::vector<unsigned long long> dc(world_);
@@ -665,14 +823,13 @@ public:
           scounts[r] = (int64_t)dc[r];
           acc += (int64_t)dc[r];
         }
-        ensure_cap(&d_send_, &cap_send_, m * kRecW);
+        ensure_cap(&d_send_, &cap_send_, m * rec_w_);
         PT_HIP_CHECK(hipMemcpy(d_offs_, offs.data(), world_ * 8,
                                hipMemcpyHostToDevice));
         PT_HIP_CHECK(hipMemsetAsync(&d_ctr_[8 + world_], 0, world_ * 8,
                                     cs_));
-        k_part_pack<<<pgrid(m), kPBlock, 0, cs_>>>(d_dep_, m, d_offs_,
-                                                   &d_ctr_[8 + world_],
-                                                   d_send_);
+        k_part_pack<<<pgrid_flat(m), kPBlock, 0, cs_>>>(
+            d_dep_, m, rec_w_, d_offs_, &d_ctr_[8 + world_], d_send_);
         PT_HIP_CHECK(hipGetLastError());
         PT_HIP_CHECK(hipStreamSynchronize(cs_));
       }
@@ -698,8 +855,8 @@ public:
         // counts are in records; collectives speak doubles
         std::vector<int64_t> sc(world_), rc(world_);
         for (int r = 0; r < world_; ++r) {
-          sc[r] = scounts[r] * kRecW;
-          rc[r] = rcounts[r] * kRecW;
+          sc[r] = scounts[r] * rec_w_;
+          rc[r] = rcounts[r] * rec_w_;
         }
         for (int64_t c : rcounts) nrecv += c;
         if (comm_->has_device_collectives()) {
@@ -708,9 +865,9 @@ public:
           recv_ptr = d_recv;
         } else {
           // host-staged exchange (TCP fallback): D2H -> alltoallv -> H2D
-          std::vector<double> hsend(m * kRecW);
+          std::vector<double> hsend(m * rec_w_);
           if (m)
-            PT_HIP_CHECK(hipMemcpy(hsend.data(), d_send_, m * kRecW * 8,
+            PT_HIP_CHECK(hipMemcpy(hsend.data(), d_send_, m * rec_w_ * 8,
                                    hipMemcpyDeviceToHost));
           std::vector<double> hrecv = comm_->alltoallv(hsend.data(), sc);
           ensure_cap(&d_recv_, &cap_recv_, (int64_t)hrecv.size());
@@ -724,9 +881,10 @@ public:
       // unpack received records; they form the next round's walk list
       PT_HIP_CHECK(hipMemsetAsync(d_ctr_, 0, 2 * 8, cs_)); // nwalk, ndep
       if (nrecv > 0) {
-        k_part_unpack<<<pgrid(nrecv), kPBlock, 0, cs_>>>(
-            recv_ptr, nrecv, d_g2l_, d_pos_, d_elem_, d_res_, d_esc_,
-            d_dest_ovr_, d_list_, d_ctr_);
+        k_part_unpack<<<pgrid_flat(nrecv), kPBlock, 0, cs_>>>(
+            recv_ptr, nrecv, rec_w_, carry_grp_, nscores_, d_g2l_, d_pos_,
+            d_elem_, d_res_, d_esc_, d_dest_ovr_, d_w_, d_grp_, d_resp_,
+            d_list_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
       }
       PT_HIP_CHECK(hipStreamSynchronize(cs_));
@@ -736,7 +894,183 @@ public:
                                  std::to_string(max_rounds_) +
                                  " handoff rounds");
     }
+  }
+
+  std::vector<int64_t> resident_list() override {
+    PT_HIP_CHECK(hipSetDevice(device_));
+    eng_->synchronize();
+    if (!d_frame_) d_frame_ = pdmalloc<int32_t>(n_);
+    PT_HIP_CHECK(hipMemsetAsync(&d_ctr_[5], 0, 8, cs_));
+    k_part_compact<<<pgrid_flat(n_), kPBlock, 0, cs_>>>(d_res_, n_, d_frame_,
+                                                       &d_ctr_[5]);
+    PT_HIP_CHECK(hipGetLastError());
+    PT_HIP_CHECK(hipStreamSynchronize(cs_));
+    unsigned long long nf = 0;
+    PT_HIP_CHECK(hipMemcpy(&nf, &d_ctr_[5], 8, hipMemcpyDeviceToHost));
+    n_frame_ = (int64_t)nf;
+    std::vector<int32_t> f32(n_frame_);
+    if (n_frame_)
+      PT_HIP_CHECK(hipMemcpy(f32.data(), d_frame_, n_frame_ * 4,
+                             hipMemcpyDeviceToHost));
+    h_frame_.assign(f32.begin(), f32.end());
+    return h_frame_;
+  }
+
+  void step_local(const double *dest, const int8_t *flying,
+                  const double *weights, int64_t n_local,
+                  const double *origin, const uint16_t *groups,
+                  const double *responses) override {
+    PT_HIP_CHECK(hipSetDevice(device_));
+    if (n_frame_ < 0)
+      throw std::runtime_error(
+          "step_local: call resident_list() first (it defines the input "
+          "order this call consumes)");
+    if (n_local != n_frame_)
+      throw std::runtime_error("step_local: n_local " +
+                               std::to_string(n_local) +
+                               " != resident_list size " +
+                               std::to_string(n_frame_));
+    if (groups && ngroups_ <= 1)
+      throw std::runtime_error("groups passed but ngroups == 1");
+    eng_->synchronize();
+    if (!d_ldest_) {
+      d_ldest_ = pdmalloc<double>(n_ * 3);
+      d_lw_ = pdmalloc<double>(n_);
+      d_lfly_ = pdmalloc<int8_t>(n_);
+    }
+    if (origin && !d_lorig_) d_lorig_ = pdmalloc<double>(n_ * 3);
+    if (groups && !d_lgrp_) d_lgrp_ = pdmalloc<uint16_t>(n_);
+    if (responses && !d_lresp_) d_lresp_ = pdmalloc<double>(n_ * nscores_);
+    if (responses && !d_resp_) {
+      d_resp_ = pdmalloc<double>(n_ * nscores_);
+      d_wresp_ = pdmalloc<double>(n_ * nscores_);
+    }
+    if (groups && !d_grp_) {
+      d_grp_ = pdmalloc<uint16_t>(n_);
+      d_wgrp_ = pdmalloc<uint16_t>(n_);
+    }
+    if (n_local) {
+      PT_HIP_CHECK(hipMemcpy(d_ldest_, dest, n_local * 24,
+                             hipMemcpyHostToDevice));
+      PT_HIP_CHECK(hipMemcpy(d_lfly_, flying, n_local,
+                             hipMemcpyHostToDevice));
+      PT_HIP_CHECK(hipMemcpy(d_lw_, weights, n_local * 8,
+                             hipMemcpyHostToDevice));
+      if (origin)
+        PT_HIP_CHECK(hipMemcpy(d_lorig_, origin, n_local * 24,
+                               hipMemcpyHostToDevice));
+      if (groups)
+        PT_HIP_CHECK(hipMemcpy(d_lgrp_, groups, n_local * 2,
+                               hipMemcpyHostToDevice));
+      if (responses)
+        PT_HIP_CHECK(hipMemcpy(d_lresp_, responses,
+                               n_local * nscores_ * 8,
+                               hipMemcpyHostToDevice));
+    }
+    PT_HIP_CHECK(hipMemsetAsync(d_ctr_, 0,
+                                (ctr_chunk0_ + kMaxChunks) * 8, cs_));
+    if (n_local) {
+      k_part_prepare_local<<<pgrid_flat(n_local), kPBlock, 0, cs_>>>(
+          n_local, d_frame_, d_res_, d_esc_, d_lfly_,
+          origin ? d_lorig_ : nullptr, d_pos_, d_elem_, dmesh_.planes,
+          dmesh_.grid, d_lowner_, d_l2g_, rank_, loc_tol_, d_ldest_, d_lw_,
+          groups ? d_lgrp_ : nullptr, responses ? d_lresp_ : nullptr,
+          nscores_, carry_grp_, rec_w_, d_list_, d_wpos_, d_wdest_,
+          d_welem_, d_ww_, groups ? d_wgrp_ : nullptr,
+          responses ? d_wresp_ : nullptr, d_dep_, d_eject_, d_ctr_,
+          &d_ctr_[ctr_chunk0_]);
+      PT_HIP_CHECK(hipGetLastError());
+    }
+    PT_HIP_CHECK(hipStreamSynchronize(cs_));
+    unsigned long long hctr[5] = {0, 0, 0, 0, 0}, cw = 0;
+    PT_HIP_CHECK(hipMemcpy(hctr, d_ctr_, 5 * 8, hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(&cw, &d_ctr_[ctr_chunk0_], 8,
+                           hipMemcpyDeviceToHost));
+    if (hctr[2] > 0)
+      host_resolve_ejects_local((int64_t)hctr[2], origin, dest, weights,
+                                groups, responses);
+    stats_.relocated += (int64_t)hctr[3];
+    stats_.loose_localizations += (int64_t)hctr[4];
+    if (cw > 0) {
+      eng_->walk_raw_device((int64_t)cw, d_wpos_, d_wdest_, d_welem_, d_ww_,
+                            d_wout_pos_, d_wout_elem_, d_wstatus_,
+                            groups ? d_wgrp_ : nullptr,
+                            responses ? d_wresp_ : nullptr, d_wout_dest_);
+      k_part_collect<<<pgrid_flat((int64_t)cw), kPBlock, 0, cs_>>>(
+          d_list_, (int64_t)cw, d_wout_pos_, d_wout_elem_, d_wstatus_,
+          d_wout_dest_, d_ww_, groups ? d_wgrp_ : nullptr,
+          responses ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
+          d_pos_, d_elem_, d_esc_, d_res_, d_fgid_, d_fowner_, d_dep_,
+          d_ctr_);
+      PT_HIP_CHECK(hipGetLastError());
+      PT_HIP_CHECK(hipStreamSynchronize(cs_));
+    }
+    run_exchange_rounds(groups != nullptr, responses != nullptr);
+    n_frame_ = -1; // residency changed; a new snapshot is required
     stats_.moves++;
+  }
+
+  // step_local's eject resolution: inputs are frame-indexed
+  void host_resolve_ejects_local(int64_t ne, const double *origin,
+                                 const double *dest, const double *weights,
+                                 const uint16_t *groups,
+                                 const double *responses) {
+    std::vector<int32_t> js(ne);
+    PT_HIP_CHECK(
+        hipMemcpy(js.data(), d_eject_, ne * 4, hipMemcpyDeviceToHost));
+    const int dw = rec_w_ + 1;
+    std::vector<double> depx;
+    std::vector<int32_t> out_j;
+    for (int64_t i = 0; i < ne; ++i) {
+      const int64_t j = js[i];
+      const int64_t g = h_frame_[j];
+      const Vec3 q{origin[j * 3], origin[j * 3 + 1], origin[j * 3 + 2]};
+      bool lo = false;
+      const int32_t ge = full_locate_ ? full_locate_(q, loc_tol_, &lo) : -1;
+      if (lo) stats_.loose_localizations++;
+      if (ge >= 0) {
+        std::vector<double> e(dw, 0.0);
+        e[0] = (double)g;
+        e[1] = q.x;
+        e[2] = q.y;
+        e[3] = q.z;
+        e[4] = (double)ge;
+        e[5] = dest[j * 3];
+        e[6] = dest[j * 3 + 1];
+        e[7] = dest[j * 3 + 2];
+        e[8] = weights[j];
+        int at = 9;
+        if (carry_grp_) e[at++] = groups ? (double)groups[j] : 0.0;
+        for (int k = 0; k < nscores_; ++k)
+          e[at + k] = responses ? responses[j * nscores_ + k] : 1.0;
+        e[rec_w_] = (double)dec_.owners[ge];
+        depx.insert(depx.end(), e.begin(), e.end());
+        // the particle left this rank
+        const uint8_t zero = 0;
+        PT_HIP_CHECK(
+            hipMemcpy(d_res_ + g, &zero, 1, hipMemcpyHostToDevice));
+      } else {
+        out_j.push_back((int32_t)j);
+      }
+    }
+    if (!depx.empty()) {
+      unsigned long long ndep = 0;
+      PT_HIP_CHECK(hipMemcpy(&ndep, &d_ctr_[1], 8, hipMemcpyDeviceToHost));
+      PT_HIP_CHECK(hipMemcpy(d_dep_ + (int64_t)ndep * dw, depx.data(),
+                             depx.size() * 8, hipMemcpyHostToDevice));
+      ndep += (unsigned long long)(depx.size() / dw);
+      PT_HIP_CHECK(hipMemcpy(&d_ctr_[1], &ndep, 8, hipMemcpyHostToDevice));
+    }
+    for (int32_t j : out_j) {
+      // outside the mesh: keep resident here at the requested origin
+      const int64_t g = h_frame_[j];
+      const double q[3] = {origin[j * 3], origin[j * 3 + 1],
+                           origin[j * 3 + 2]};
+      const int32_t none = -1;
+      PT_HIP_CHECK(hipMemcpy(d_pos_ + g * 3, q, 24, hipMemcpyHostToDevice));
+      PT_HIP_CHECK(
+          hipMemcpy(d_elem_ + g, &none, 4, hipMemcpyHostToDevice));
+    }
   }
 
   std::vector<double> flux_global() override {
@@ -878,6 +1212,15 @@ private:
   hipStream_t s_copy_ = nullptr; // step-input H2D pipeline
   hipEvent_t ev_in_[kMaxChunks] = {};
   int ctr_chunk0_ = 8;
+  int rec_w_ = 10;
+  bool carry_grp_ = false;
+  int32_t *d_frame_ = nullptr;   // coupled-host frame (resident gids)
+  int64_t n_frame_ = -1;         // -1: no snapshot taken
+  std::vector<int64_t> h_frame_; // host copy of the frame
+  double *d_ldest_ = nullptr, *d_lw_ = nullptr, *d_lorig_ = nullptr,
+         *d_lresp_ = nullptr;
+  int8_t *d_lfly_ = nullptr;
+  uint16_t *d_lgrp_ = nullptr;
   double loc_tol_ = 0.0;
   mutable EngineStats stats_;
 
@@ -920,6 +1263,8 @@ public:
       throw std::runtime_error("PartitionedEngine: world > 1 needs a comm");
     eng_ = make_cpu_engine(dec_.sub.local, 1, ngroups_, nscores_);
     loc_tol_ = loc_tol_rel() * norm(full.bbox_hi - full.bbox_lo);
+    carry_grp_ = ngroups_ > 1;
+    rec_w_ = 9 + (carry_grp_ ? 1 : 0) + nscores_;
     pos_.assign(n_ * 3, 0.0);
     elem_.assign(n_, -1);
     res_.assign(n_, 0);
@@ -965,169 +1310,73 @@ public:
     }
   }
 
+  // A resolved unit of walk work: every input the particle needs travels
+  // with it, so global-array steps, coupled-host steps and exchanged
+  // arrivals all feed the same loop (the CPU mirror of the GPU wire
+  // format).
+  struct Item {
+    int64_t gid;
+    Vec3 pos, dest;
+    double w;
+    uint16_t grp;
+    std::vector<double> resp; // nscores entries when used, else empty
+    int32_t lelem;
+  };
+
   void step(const double *dest, const int8_t *flying, const double *weights,
             int64_t n, const double *origin, const uint16_t *groups,
             const double *responses) override {
     check_n(n);
     if (groups && ngroups_ <= 1)
       throw std::runtime_error("groups passed but ngroups == 1");
-    const Mesh &lm = dec_.sub.local;
-    std::vector<int32_t> list;
-    std::vector<double> dep; // 9 doubles per entry (see kRecW comment)
-    std::vector<double> ovr; // per-arrival destination overrides
+    std::vector<Item> items;
+    std::vector<double> dep;
     for (int64_t g = 0; g < n_; ++g) {
       if (!res_[g] || !flying[g]) continue;
-      if (origin && !esc_[g]) {
-        const Vec3 q{origin[g * 3], origin[g * 3 + 1], origin[g * 3 + 2]};
-        if (q.x != pos_[g * 3] || q.y != pos_[g * 3 + 1] ||
-            q.z != pos_[g * 3 + 2]) {
-          stats_.relocated++;
-          bool lo = false;
-          int32_t le = lm.locate(q, loc_tol_, &lo);
-          if (lo) stats_.loose_localizations++;
-          if (le >= 0 && dec_.lowner[le] != rank_) {
-            dep.insert(dep.end(), {(double)g, q.x, q.y, q.z,
-                                   (double)dec_.l2g32[le],
-                                   (double)dec_.lowner[le], dest[g * 3],
-                                   dest[g * 3 + 1], dest[g * 3 + 2]});
-            res_[g] = 0;
-            continue;
-          }
-          if (le < 0) {
-            // global resolve via full-mesh locate
-            bool lo2 = false;
-            const int32_t ge =
-                full_locate_ ? full_locate_(q, loc_tol_, &lo2) : -1;
-            if (lo2) stats_.loose_localizations++;
-            if (ge >= 0) {
-              dep.insert(dep.end(), {(double)g, q.x, q.y, q.z, (double)ge,
-                                     (double)dec_.owners[ge], dest[g * 3],
-                                     dest[g * 3 + 1], dest[g * 3 + 2]});
-              res_[g] = 0;
-            } else {
-              elem_[g] = -1;
-              pos_[g * 3] = q.x;
-              pos_[g * 3 + 1] = q.y;
-              pos_[g * 3 + 2] = q.z;
-            }
-            continue;
-          }
-          elem_[g] = le;
-          pos_[g * 3] = q.x;
-          pos_[g * 3 + 1] = q.y;
-          pos_[g * 3 + 2] = q.z;
-        }
-      }
-      if (elem_[g] < 0) continue;
-      list.push_back((int32_t)g);
+      enter_particle(g, origin ? origin + g * 3 : nullptr, dest + g * 3,
+                     weights[g], groups ? groups[g] : (uint16_t)0,
+                     responses ? responses + g * nscores_ : nullptr,
+                     responses != nullptr, items, dep);
     }
+    run_items(std::move(items), std::move(dep), responses != nullptr);
+    stats_.moves++;
+  }
 
-    for (int round = 0; round < max_rounds_; ++round) {
-      if (!list.empty()) {
-        const int64_t m = (int64_t)list.size();
-        std::vector<double> wpos(m * 3), wdest(m * 3), ww(m), wout_pos(m * 3);
-        std::vector<double> wout_dest(m * 3);
-        std::vector<int32_t> welem(m), wout_elem(m);
-        std::vector<int8_t> wstatus(m);
-        std::vector<uint16_t> wgrp(groups ? m : 0);
-        std::vector<double> wresp(responses ? m * nscores_ : 0);
-        const double *dsrc = round > 0 ? ovr.data() : dest;
-        for (int64_t j = 0; j < m; ++j) {
-          const int64_t g = list[j];
-          for (int k = 0; k < 3; ++k) {
-            wpos[j * 3 + k] = pos_[g * 3 + k];
-            wdest[j * 3 + k] = dsrc[g * 3 + k];
-          }
-          welem[j] = elem_[g];
-          ww[j] = weights[g];
-          if (groups) wgrp[j] = groups[g];
-          if (responses)
-            for (int k = 0; k < nscores_; ++k)
-              wresp[j * nscores_ + k] = responses[g * nscores_ + k];
-        }
-        eng_->walk_raw(m, wpos.data(), wdest.data(), welem.data(), ww.data(),
-                       wout_pos.data(), wout_elem.data(), wstatus.data(),
-                       groups ? wgrp.data() : nullptr,
-                       responses ? wresp.data() : nullptr, wout_dest.data());
-        for (int64_t j = 0; j < m; ++j) {
-          const int64_t g = list[j];
-          if (wstatus[j] == 2) {
-            const int32_t k = -(wout_elem[j] + 2);
-            dep.insert(dep.end(),
-                       {(double)g, wout_pos[j * 3], wout_pos[j * 3 + 1],
-                        wout_pos[j * 3 + 2],
-                        (double)dec_.sub.foreign_gid[k],
-                        (double)dec_.sub.foreign_owner[k],
-                        wout_dest[j * 3], wout_dest[j * 3 + 1],
-                        wout_dest[j * 3 + 2]});
-            res_[g] = 0;
-          } else {
-            for (int k = 0; k < 3; ++k) pos_[g * 3 + k] = wout_pos[j * 3 + k];
-            elem_[g] = wout_elem[j];
-            esc_[g] = (wstatus[j] == 1) ? 1 : 0;
-          }
-        }
-      }
-      list.clear();
+  std::vector<int64_t> resident_list() override {
+    frame_.clear();
+    for (int64_t g = 0; g < n_; ++g)
+      if (res_[g]) frame_.push_back(g);
+    have_frame_ = true;
+    return frame_;
+  }
 
-      // bucket by destination, exchange, unpack
-      const int64_t m = (int64_t)dep.size() / 9;
-      std::vector<int64_t> scounts(world_, 0);
-      std::vector<double> send(m * kRecW);
-      {
-        std::vector<int64_t> offs(world_, 0), cur(world_, 0);
-        for (int64_t i = 0; i < m; ++i) scounts[(int)dep[i * 9 + 5]]++;
-        int64_t acc = 0;
-        for (int r = 0; r < world_; ++r) {
-          offs[r] = acc;
-          acc += scounts[r];
-        }
-        for (int64_t i = 0; i < m; ++i) {
-          const int o = (int)dep[i * 9 + 5];
-          const int64_t s = offs[o] + cur[o]++;
-          for (int k = 0; k < 5; ++k) send[s * kRecW + k] = dep[i * 9 + k];
-          for (int k = 0; k < 3; ++k)
-            send[s * kRecW + 5 + k] = dep[i * 9 + 6 + k];
-        }
-      }
-      dep.clear();
-
-      std::vector<double> recv;
-      if (world_ == 1) {
-        if (m == 0) break;
-        recv = std::move(send);
-      } else {
-        std::vector<int64_t> flat((int64_t)world_ * world_, 0);
-        for (int r = 0; r < world_; ++r)
-          flat[(int64_t)rank_ * world_ + r] = scounts[r];
-        comm_->allreduce_sum(flat.data(), (int64_t)world_ * world_);
-        int64_t total = 0;
-        for (int64_t c : flat) total += c;
-        if (total == 0) break;
-        std::vector<int64_t> sc(world_);
-        for (int r = 0; r < world_; ++r) sc[r] = scounts[r] * kRecW;
-        recv = comm_->alltoallv(send.data(), sc);
-      }
-      const int64_t nr = (int64_t)recv.size() / kRecW;
-      if (nr && ovr.empty()) ovr.assign(n_ * 3, 0.0);
-      for (int64_t i = 0; i < nr; ++i) {
-        const int64_t g = (int64_t)recv[i * kRecW];
-        res_[g] = 1;
-        esc_[g] = 0;
-        pos_[g * 3] = recv[i * kRecW + 1];
-        pos_[g * 3 + 1] = recv[i * kRecW + 2];
-        pos_[g * 3 + 2] = recv[i * kRecW + 3];
-        elem_[g] = dec_.g2l[(int64_t)recv[i * kRecW + 4]];
-        ovr[g * 3] = recv[i * kRecW + 5];
-        ovr[g * 3 + 1] = recv[i * kRecW + 6];
-        ovr[g * 3 + 2] = recv[i * kRecW + 7];
-        list.push_back((int32_t)g);
-      }
-      if (round == max_rounds_ - 1)
-        throw std::runtime_error("partitioned step did not converge in " +
-                                 std::to_string(max_rounds_) +
-                                 " handoff rounds");
+  void step_local(const double *dest, const int8_t *flying,
+                  const double *weights, int64_t n_local,
+                  const double *origin, const uint16_t *groups,
+                  const double *responses) override {
+    if (!have_frame_)
+      throw std::runtime_error(
+          "step_local: call resident_list() first (it defines the input "
+          "order this call consumes)");
+    if (n_local != (int64_t)frame_.size())
+      throw std::runtime_error("step_local: n_local " +
+                               std::to_string(n_local) +
+                               " != resident_list size " +
+                               std::to_string(frame_.size()));
+    if (groups && ngroups_ <= 1)
+      throw std::runtime_error("groups passed but ngroups == 1");
+    std::vector<Item> items;
+    std::vector<double> dep;
+    for (int64_t j = 0; j < n_local; ++j) {
+      const int64_t g = frame_[j];
+      if (!res_[g] || !flying[j]) continue;
+      enter_particle(g, origin ? origin + j * 3 : nullptr, dest + j * 3,
+                     weights[j], groups ? groups[j] : (uint16_t)0,
+                     responses ? responses + j * nscores_ : nullptr,
+                     responses != nullptr, items, dep);
     }
+    have_frame_ = false; // residency changes below
+    run_items(std::move(items), std::move(dep), responses != nullptr);
     stats_.moves++;
   }
 
@@ -1136,10 +1385,10 @@ public:
     const int64_t nl = dec_.sub.local.nelems;
     const int64_t slabs = (int64_t)nscores_ * ngroups_;
     std::vector<double> out(slabs * nelems_global_, 0.0);
-    for (int64_t s = 0; s < slabs; ++s)
+    for (int64_t sl = 0; sl < slabs; ++sl)
       for (int64_t t = 0; t < nl; ++t)
-        out[s * nelems_global_ + dec_.sub.elem_l2g[t]] +=
-            local[s * nl + t];
+        out[sl * nelems_global_ + dec_.sub.elem_l2g[t]] +=
+            local[sl * nl + t];
     if (world_ > 1)
       comm_->allreduce_sum(out.data(), (int64_t)out.size());
     return out;
@@ -1172,10 +1421,220 @@ private:
     if (n != n_) throw std::runtime_error("global particle count mismatch");
   }
 
+  // Phase A + admission for one flying resident particle: relocation /
+  // reroute / eject resolution, then either queue a walk Item or emit a
+  // departure record.
+  void enter_particle(int64_t g, const double *origin3, const double *dest3,
+                      double w, uint16_t grp, const double *resp,
+                      bool resp_used, std::vector<Item> &items,
+                      std::vector<double> &dep) {
+    const Mesh &lm = dec_.sub.local;
+    if (origin3 && !esc_[g]) {
+      const Vec3 q{origin3[0], origin3[1], origin3[2]};
+      if (q.x != pos_[g * 3] || q.y != pos_[g * 3 + 1] ||
+          q.z != pos_[g * 3 + 2]) {
+        stats_.relocated++;
+        bool lo = false;
+        int32_t le = lm.locate(q, loc_tol_, &lo);
+        if (lo) stats_.loose_localizations++;
+        int64_t tgid = -1;
+        int towner = -1;
+        if (le >= 0 && dec_.lowner[le] != rank_) {
+          tgid = dec_.l2g32[le];
+          towner = dec_.lowner[le];
+        } else if (le < 0) {
+          bool lo2 = false;
+          const int32_t ge =
+              full_locate_ ? full_locate_(q, loc_tol_, &lo2) : -1;
+          if (lo2) stats_.loose_localizations++;
+          if (ge >= 0) {
+            tgid = ge;
+            towner = dec_.owners[ge];
+          } else {
+            elem_[g] = -1;
+            pos_[g * 3] = q.x;
+            pos_[g * 3 + 1] = q.y;
+            pos_[g * 3 + 2] = q.z;
+            return;
+          }
+        }
+        if (towner >= 0) {
+          emit_dep(dep, g, q, tgid,
+                   Vec3{dest3[0], dest3[1], dest3[2]}, w, grp, resp,
+                   resp_used, towner);
+          res_[g] = 0;
+          return;
+        }
+        elem_[g] = le;
+        pos_[g * 3] = q.x;
+        pos_[g * 3 + 1] = q.y;
+        pos_[g * 3 + 2] = q.z;
+      }
+    }
+    if (elem_[g] < 0) return; // outside mesh
+    Item it;
+    it.gid = g;
+    it.pos = Vec3{pos_[g * 3], pos_[g * 3 + 1], pos_[g * 3 + 2]};
+    it.dest = Vec3{dest3[0], dest3[1], dest3[2]};
+    it.w = w;
+    it.grp = grp;
+    if (resp_used) it.resp.assign(resp, resp + nscores_);
+    it.lelem = elem_[g];
+    items.push_back(std::move(it));
+  }
+
+  void emit_dep(std::vector<double> &dep, int64_t g, Vec3 p, int64_t tgid,
+                Vec3 d, double w, uint16_t grp, const double *resp,
+                bool resp_used, int owner) {
+    const size_t base = dep.size();
+    dep.resize(base + rec_w_ + 1, 0.0);
+    double *e = dep.data() + base;
+    e[0] = (double)g;
+    e[1] = p.x;
+    e[2] = p.y;
+    e[3] = p.z;
+    e[4] = (double)tgid;
+    e[5] = d.x;
+    e[6] = d.y;
+    e[7] = d.z;
+    e[8] = w;
+    int at = 9;
+    if (carry_grp_) e[at++] = (double)grp;
+    for (int k = 0; k < nscores_; ++k)
+      e[at + k] = (resp_used && resp) ? resp[k] : 1.0;
+    e[rec_w_] = (double)owner;
+  }
+
+  void run_items(std::vector<Item> items, std::vector<double> dep,
+                 bool resp_used) {
+    const bool groups_used = carry_grp_;
+    const int dw = rec_w_ + 1;
+    for (int round = 0; round <= max_rounds_; ++round) {
+      if (!items.empty()) {
+        const int64_t m = (int64_t)items.size();
+        std::vector<double> wpos(m * 3), wdest(m * 3), ww(m),
+            wout_pos(m * 3), wout_dest(m * 3);
+        std::vector<int32_t> welem(m), wout_elem(m);
+        std::vector<int8_t> wstatus(m);
+        std::vector<uint16_t> wgrp(groups_used ? m : 0);
+        std::vector<double> wresp(resp_used ? m * nscores_ : 0);
+        for (int64_t j = 0; j < m; ++j) {
+          const Item &it = items[j];
+          wpos[j * 3] = it.pos.x;
+          wpos[j * 3 + 1] = it.pos.y;
+          wpos[j * 3 + 2] = it.pos.z;
+          wdest[j * 3] = it.dest.x;
+          wdest[j * 3 + 1] = it.dest.y;
+          wdest[j * 3 + 2] = it.dest.z;
+          welem[j] = it.lelem;
+          ww[j] = it.w;
+          if (groups_used) wgrp[j] = it.grp;
+          if (resp_used)
+            for (int k = 0; k < nscores_; ++k)
+              wresp[j * nscores_ + k] = it.resp[k];
+        }
+        eng_->walk_raw(m, wpos.data(), wdest.data(), welem.data(), ww.data(),
+                       wout_pos.data(), wout_elem.data(), wstatus.data(),
+                       groups_used ? wgrp.data() : nullptr,
+                       resp_used ? wresp.data() : nullptr,
+                       wout_dest.data());
+        for (int64_t j = 0; j < m; ++j) {
+          const int64_t g = items[j].gid;
+          if (wstatus[j] == 2) {
+            const int32_t k = -(wout_elem[j] + 2);
+            emit_dep(dep, g,
+                     Vec3{wout_pos[j * 3], wout_pos[j * 3 + 1],
+                          wout_pos[j * 3 + 2]},
+                     dec_.sub.foreign_gid[k],
+                     Vec3{wout_dest[j * 3], wout_dest[j * 3 + 1],
+                          wout_dest[j * 3 + 2]},
+                     ww[j], groups_used ? wgrp[j] : (uint16_t)0,
+                     resp_used ? &wresp[j * nscores_] : nullptr, resp_used,
+                     dec_.sub.foreign_owner[k]);
+            res_[g] = 0;
+          } else {
+            pos_[g * 3] = wout_pos[j * 3];
+            pos_[g * 3 + 1] = wout_pos[j * 3 + 1];
+            pos_[g * 3 + 2] = wout_pos[j * 3 + 2];
+            elem_[g] = wout_elem[j];
+            esc_[g] = (wstatus[j] == 1) ? 1 : 0;
+          }
+        }
+      }
+      items.clear();
+
+      // bucket by owner, exchange, unpack into next-round items
+      const int64_t m = (int64_t)dep.size() / dw;
+      std::vector<int64_t> scounts(world_, 0);
+      std::vector<double> send(m * rec_w_);
+      {
+        std::vector<int64_t> offs(world_, 0), cur(world_, 0);
+        for (int64_t i = 0; i < m; ++i)
+          scounts[(int)dep[i * dw + rec_w_]]++;
+        int64_t acc = 0;
+        for (int r = 0; r < world_; ++r) {
+          offs[r] = acc;
+          acc += scounts[r];
+        }
+        for (int64_t i = 0; i < m; ++i) {
+          const int o = (int)dep[i * dw + rec_w_];
+          const int64_t sl = offs[o] + cur[o]++;
+          for (int k = 0; k < rec_w_; ++k)
+            send[sl * rec_w_ + k] = dep[i * dw + k];
+        }
+      }
+      dep.clear();
+
+      std::vector<double> recv;
+      if (world_ == 1) {
+        if (m == 0) break;
+        recv = std::move(send);
+      } else {
+        std::vector<int64_t> flat((int64_t)world_ * world_, 0);
+        for (int r = 0; r < world_; ++r)
+          flat[(int64_t)rank_ * world_ + r] = scounts[r];
+        comm_->allreduce_sum(flat.data(), (int64_t)world_ * world_);
+        int64_t total = 0;
+        for (int64_t c : flat) total += c;
+        if (total == 0) break;
+        std::vector<int64_t> sc(world_);
+        for (int r = 0; r < world_; ++r) sc[r] = scounts[r] * rec_w_;
+        recv = comm_->alltoallv(send.data(), sc);
+      }
+      const int64_t nr = (int64_t)recv.size() / rec_w_;
+      for (int64_t i = 0; i < nr; ++i) {
+        const double *e = recv.data() + i * rec_w_;
+        const int64_t g = (int64_t)e[0];
+        res_[g] = 1;
+        esc_[g] = 0;
+        pos_[g * 3] = e[1];
+        pos_[g * 3 + 1] = e[2];
+        pos_[g * 3 + 2] = e[3];
+        elem_[g] = dec_.g2l[(int64_t)e[4]];
+        Item it;
+        it.gid = g;
+        it.pos = Vec3{e[1], e[2], e[3]};
+        it.dest = Vec3{e[5], e[6], e[7]};
+        it.w = e[8];
+        int at = 9;
+        it.grp = carry_grp_ ? (uint16_t)e[at++] : (uint16_t)0;
+        if (resp_used) it.resp.assign(e + at, e + at + nscores_);
+        it.lelem = elem_[g];
+        items.push_back(std::move(it));
+      }
+      if (round == max_rounds_)
+        throw std::runtime_error("partitioned step did not converge in " +
+                                 std::to_string(max_rounds_) +
+                                 " handoff rounds");
+    }
+  }
+
   int64_t n_;
   Comm *comm_;
   int rank_, world_;
   int ngroups_, nscores_ = 1;
+  int rec_w_ = 10;
+  bool carry_grp_ = false;
   int64_t nelems_global_;
   Decomp dec_;
   std::unique_ptr<Engine> eng_;
@@ -1184,6 +1643,8 @@ private:
   std::vector<double> pos_;
   std::vector<int32_t> elem_;
   std::vector<uint8_t> res_, esc_;
+  std::vector<int64_t> frame_;
+  bool have_frame_ = false;
 };
 
 } // namespace

@@ -447,6 +447,67 @@ PYBIND11_MODULE(_core, m) {
              return out;
            })
       .def("synchronize", [](PyPartEngine &s) { s.pe->synchronize(); })
+      // coupled-host path: resident_list() snapshots this rank's
+      // particles; step_local consumes arrays in exactly that order
+      .def("resident_list",
+           [](PyPartEngine &s) {
+             std::vector<int64_t> v;
+             {
+               py::gil_scoped_release ng;
+               v = s.pe->resident_list();
+             }
+             auto out = py::array_t<int64_t>(v.size());
+             std::memcpy(out.mutable_data(), v.data(), v.size() * 8);
+             return out;
+           })
+      .def("step_local",
+           [](PyPartEngine &s,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  dest,
+              py::array_t<int8_t, py::array::c_style | py::array::forcecast>
+                  flying,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  weights,
+              py::object origin, py::object groups, py::object responses) {
+             const int64_t nl = (int64_t)flying.size();
+             if ((int64_t)dest.size() != nl * 3 ||
+                 (int64_t)weights.size() != nl)
+               throw std::runtime_error("step_local: array size mismatch");
+             const double *op = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast>
+                 oarr;
+             if (!origin.is_none()) {
+               oarr = origin.cast<py::array_t<
+                   double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)oarr.size() != nl * 3)
+                 throw std::runtime_error("step_local: origin size mismatch");
+               op = oarr.data();
+             }
+             const uint16_t *gp = nullptr;
+             py::array_t<uint16_t, py::array::c_style | py::array::forcecast>
+                 garr;
+             if (!groups.is_none()) {
+               garr = groups.cast<py::array_t<
+                   uint16_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)garr.size() != nl)
+                 throw std::runtime_error("step_local: groups size mismatch");
+               gp = garr.data();
+             }
+             const double *rp = nullptr;
+             py::array_t<double, py::array::c_style | py::array::forcecast>
+                 rarr;
+             if (!responses.is_none()) {
+               rarr = responses.cast<py::array_t<
+                   double, py::array::c_style | py::array::forcecast>>();
+               rp = rarr.data();
+             }
+             py::gil_scoped_release ng;
+             s.pe->step_local(dest.data(), flying.data(), weights.data(), nl,
+                              op, gp, rp);
+           },
+           py::arg("dest"), py::arg("flying"), py::arg("weights"),
+           py::arg("origin") = py::none(), py::arg("groups") = py::none(),
+           py::arg("responses") = py::none())
       // comm passthroughs so a driver (bench.py --partitioned) needs no
       // second communication stack
       .def("barrier",

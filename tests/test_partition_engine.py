@@ -248,3 +248,132 @@ def test_partition_engine_scored_world1():
     eng.move(p0.ravel(), p1.ravel(), fly.copy(), w, responses=resp)
     assert np.allclose(pe.flux_global(),
                        np.asarray(eng.flux()).ravel(), atol=1e-12)
+
+
+def test_step_local_world1_matches_step():
+    """Coupled-host path at world 1: resident_list + frame-ordered
+    step_local must reproduce the global-array step exactly (same
+    segments, same flux, same committed state)."""
+    mesh = pt.build_box(4, 4, 4)
+    n = 250
+    rng = np.random.default_rng(31)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    steps = []
+    pos = p0.copy()
+    for _ in range(3):
+        d = np.clip(pos + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+        w = rng.uniform(0.1, 1.0, n)
+        fly = (rng.random(n) > 0.1).astype(np.int8)
+        steps.append((d, fly, w))
+        pos = np.where(fly[:, None] == 1, d, pos)
+
+    a = pt._core.PartitionedEngine(mesh, n, device="cpu", ngroups=2,
+                                   nscores=2)
+    b = pt._core.PartitionedEngine(mesh, n, device="cpu", ngroups=2,
+                                   nscores=2)
+    grp = rng.integers(0, 2, n).astype(np.uint16)
+    rsp = rng.uniform(0.5, 2.0, size=(n, 2))
+    a.localize(p0.ravel())
+    b.localize(p0.ravel())
+    for d, fly, w in steps:
+        a.step(d.ravel(), fly, w, groups=grp, responses=rsp)
+        gids = b.resident_list()
+        # world 1: all particles resident, frame order == gid order
+        b.step_local(d[gids].ravel(), fly[gids], w[gids],
+                     groups=grp[gids], responses=rsp[gids])
+    fa, fb = a.flux_global(), b.flux_global()
+    assert np.allclose(fa, fb, atol=1e-12), np.abs(fa - fb).max()
+    assert np.allclose(a.positions(), b.positions(), atol=1e-15)
+
+
+LOCAL_WORKER = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+device = os.environ.get("PT_DEVICE", "cpu")
+
+mesh = pt.build_box(4, 4, 4)
+n = 400
+steps = 4
+rng = np.random.default_rng(47)  # same stream on all ranks
+p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+hist = []
+pos = p0.copy()
+for _ in range(steps):
+    d = np.clip(pos + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    fly = (rng.random(n) > 0.1).astype(np.int8)
+    w = rng.uniform(0.1, 1.0, n)
+    hist.append((d, fly, w))
+    pos = np.where(fly[:, None] == 1, d, pos)
+
+pe = pt._core.PartitionedEngine(mesh, n, device=device)
+pe.localize(p0.ravel())
+for d, fly, w in hist:
+    gids = pe.resident_list()
+    # each rank feeds ONLY its residents' inputs, in frame order -- the
+    # coupled-host contract (nothing global-sized crosses to the engine)
+    pe.step_local(d[gids].ravel(), fly[gids], w[gids])
+got = pe.flux_global()
+
+if rank == 0:
+    ref = pt.TallyEngine(mesh, n, device="cpu")
+    ref.copy_initial_position(p0.ravel())
+    for d, fly, w in hist:
+        ref.move_continue(d.ravel(), fly.copy(), w)
+    want = ref.flux()
+    assert np.allclose(got, want, atol=1e-11), np.abs(got - want).max()
+    print("STEP_LOCAL_WORLD2_OK resident=", pe.resident)
+"""
+
+
+def test_step_local_world2_cpu(tmp_path):
+    """Coupled-host world-2: per-rank frame-ordered inputs, records carry
+    weight+destination across ranks, flux == replicated oracle."""
+    script = tmp_path / "w.py"
+    script.write_text(LOCAL_WORKER)
+    env = dict(os.environ)
+    env.update({
+        "WORLD_SIZE": "2",
+        "MASTER_ADDR": "127.0.0.1",
+        "PUMITALLY_PORT": str(26000 + (os.getpid() + 77) % 14000),
+        "PUMITALLY_NO_TORCH": "1",
+        "PT_DEVICE": "cpu",
+        "PYTHONPATH": ROOT,
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = "0"
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "STEP_LOCAL_WORLD2_OK" in outs[0]
+
+
+@pytest.mark.gpu
+def test_step_local_world1_gpu():
+    mesh = pt.build_box(4, 4, 4)
+    n = 300
+    rng = np.random.default_rng(53)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = np.clip(p0 + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    w = rng.uniform(0.1, 1.0, n)
+    fly = np.ones(n, np.int8)
+
+    pe = pt._core.PartitionedEngine(mesh, n, device="cuda:0")
+    pe.localize(p0.ravel())
+    gids = pe.resident_list()
+    assert len(gids) == n
+    pe.step_local(d[gids].ravel(), fly[gids], w[gids])
+
+    ref = pt.TallyEngine(mesh, n, device="cpu")
+    ref.copy_initial_position(p0.ravel())
+    ref.move_continue(d.ravel(), fly.copy(), w)
+    assert np.allclose(pe.flux_global(), ref.flux(), atol=1e-11)
