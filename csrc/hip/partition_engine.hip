@@ -767,7 +767,9 @@ public:
     // host-resolve ejected relocations (rare: resampled origins that
     // missed the local submesh grid); their reroute records join the dep
     // list before the first exchange below
-    if (hctr[2] > 0) host_resolve_ejects((int64_t)hctr[2], origin, dest);
+    if (hctr[2] > 0)
+      host_resolve_ejects((int64_t)hctr[2], origin, dest, weights, groups,
+                          responses);
     stats_.relocated += (int64_t)hctr[3];
     stats_.loose_localizations += (int64_t)hctr[4];
 
@@ -1143,11 +1145,14 @@ private:
   // into the departure list (self-routes included -- they come back
   // through the exchange uniformly).
   void host_resolve_ejects(int64_t ne, const double *origin,
-                           const double *dest_host) {
+                           const double *dest_host, const double *w_host,
+                           const uint16_t *grp_host,
+                           const double *resp_host) {
     std::vector<int32_t> gids(ne);
     PT_HIP_CHECK(
         hipMemcpy(gids.data(), d_eject_, ne * 4, hipMemcpyDeviceToHost));
-    std::vector<double> dep9;
+    const int dw = rec_w_ + 1;
+    std::vector<double> depx;
     std::vector<int32_t> outside;
     for (int64_t i = 0; i < ne; ++i) {
       const int64_t g = gids[i];
@@ -1156,28 +1161,40 @@ private:
       const int32_t ge = full_locate_ ? full_locate_(q, loc_tol_, &lo) : -1;
       if (lo) stats_.loose_localizations++;
       if (ge >= 0) {
-        dep9.insert(dep9.end(),
-                    {(double)g, q.x, q.y, q.z, (double)ge,
-                     (double)dec_.owners[ge], dest_host[g * 3],
-                     dest_host[g * 3 + 1], dest_host[g * 3 + 2]});
+        std::vector<double> e(dw, 0.0);
+        e[0] = (double)g;
+        e[1] = q.x;
+        e[2] = q.y;
+        e[3] = q.z;
+        e[4] = (double)ge;
+        e[5] = dest_host[g * 3];
+        e[6] = dest_host[g * 3 + 1];
+        e[7] = dest_host[g * 3 + 2];
+        e[8] = w_host[g];
+        int at = 9;
+        if (carry_grp_) e[at++] = grp_host ? (double)grp_host[g] : 0.0;
+        for (int k = 0; k < nscores_; ++k)
+          e[at + k] = resp_host ? resp_host[g * nscores_ + k] : 1.0;
+        e[rec_w_] = (double)dec_.owners[ge];
+        depx.insert(depx.end(), e.begin(), e.end());
       } else {
         outside.push_back((int32_t)g);
       }
     }
-    if (!dep9.empty()) {
-      // append to the device dep list (capacity n_*9 is plenty: ejects
-      // are a subset of residents)
+    if (!depx.empty()) {
+      // append to the device dep list (capacity n_*(rec_w_+1) is plenty:
+      // ejects are a subset of residents)
       unsigned long long ndep = 0;
       PT_HIP_CHECK(hipMemcpy(&ndep, &d_ctr_[1], 8, hipMemcpyDeviceToHost));
-      const int64_t m = (int64_t)dep9.size() / 9;
-      PT_HIP_CHECK(hipMemcpy(d_dep_ + (int64_t)ndep * 9, dep9.data(),
-                             dep9.size() * 8, hipMemcpyHostToDevice));
+      const int64_t m = (int64_t)depx.size() / dw;
+      PT_HIP_CHECK(hipMemcpy(d_dep_ + (int64_t)ndep * dw, depx.data(),
+                             depx.size() * 8, hipMemcpyHostToDevice));
       ndep += (unsigned long long)m;
       PT_HIP_CHECK(hipMemcpy(&d_ctr_[1], &ndep, 8, hipMemcpyHostToDevice));
       // the particle left this rank
       std::vector<uint8_t> zero(1, 0);
       for (int64_t i = 0; i < m; ++i) {
-        const int64_t g = (int64_t)dep9[i * 9];
+        const int64_t g = (int64_t)depx[(int64_t)i * dw];
         PT_HIP_CHECK(
             hipMemcpy(d_res_ + g, zero.data(), 1, hipMemcpyHostToDevice));
       }
