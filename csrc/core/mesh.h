@@ -68,7 +68,10 @@ struct Mesh {
   // every face in `a`, translated by T, must coincide with a face in `b`
   // (matched by face centroid to tol); pairing is installed both ways
   // (b gets -T).  Call AFTER finalize() and BEFORE engine construction.
-  // Not supported in partitioned submeshes (PartitionedTally raises).
+  // Partitioned support: extract_submesh wires pairs whose partner is in
+  // the submesh into the local tables and turns cross-part pairs into
+  // translation-carrying handoff entries (SubMesh::foreign_shift); the
+  // stateful PartitionedEngine accepts periodic only at world 1.
   std::vector<int32_t> periodic_idx;   // nelems*4, -1 = no pair; empty = none
   std::vector<int32_t> periodic_elem;  // pair entry -> entry element
   std::vector<double> periodic_shift;  // pair entry -> translation (x,y,z)
