@@ -100,6 +100,21 @@ public:
   virtual std::vector<uint8_t> resident_mask() const = 0;
   virtual std::vector<double> positions() const = 0;  // n_global*3
   virtual std::vector<int32_t> elem_ids() const = 0;  // LOCAL elem ids
+
+  // Decomposition-independent state transfer (checkpoint/resume and
+  // dynamic repartitioning).  elem_ids_global() maps this rank's
+  // resident entries to GLOBAL element ids (-1 for out-of-mesh).
+  // set_state() installs a full global snapshot: every rank passes the
+  // SAME arrays and claims the particles whose element it owns under
+  // THIS engine's decomposition (out-of-mesh particles go to rank 0),
+  // so a snapshot taken under one decomposition restores under any
+  // other -- repartitioning is "build a new engine with new owners,
+  // set_state(old snapshot)".  The flux tally is NOT transferred: drain
+  // it with flux_global() first and sum host-side.
+  virtual std::vector<int32_t> elem_ids_global() const = 0;
+  virtual std::vector<uint8_t> escaped_mask() const = 0;
+  virtual void set_state(const double *pos, const int32_t *gelem,
+                         const uint8_t *escaped, int64_t n_global) = 0;
 };
 
 // comm may be null only when world == 1.  device: "cpu" or a HIP

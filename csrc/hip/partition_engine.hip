@@ -1127,6 +1127,58 @@ public:
     return e;
   }
 
+  std::vector<int32_t> elem_ids_global() const override {
+    eng_->synchronize();
+    std::vector<int32_t> le(n_);
+    std::vector<uint8_t> r(n_);
+    PT_HIP_CHECK(hipMemcpy(le.data(), d_elem_, n_ * 4,
+                           hipMemcpyDeviceToHost));
+    PT_HIP_CHECK(hipMemcpy(r.data(), d_res_, n_, hipMemcpyDeviceToHost));
+    std::vector<int32_t> out(n_, -1);
+    for (int64_t g = 0; g < n_; ++g)
+      if (r[g] && le[g] >= 0) out[g] = dec_.l2g32[le[g]];
+    return out;
+  }
+
+  std::vector<uint8_t> escaped_mask() const override {
+    eng_->synchronize();
+    std::vector<uint8_t> e(n_);
+    PT_HIP_CHECK(hipMemcpy(e.data(), d_esc_, n_, hipMemcpyDeviceToHost));
+    return e;
+  }
+
+  void set_state(const double *pos, const int32_t *gelem,
+                 const uint8_t *escaped, int64_t n) override {
+    check_n(n);
+    PT_HIP_CHECK(hipSetDevice(device_));
+    eng_->synchronize();
+    // ownership claims under THIS decomposition (host-side: once per
+    // restore/repartition, not a hot path)
+    std::vector<uint8_t> res(n_, 0);
+    std::vector<int32_t> lel(n_, -1);
+    for (int64_t g = 0; g < n_; ++g) {
+      const int32_t ge = gelem[g];
+      if (ge >= 0) {
+        if (dec_.owners[ge] == rank_) {
+          res[g] = 1;
+          lel[g] = dec_.g2l[ge];
+          if (lel[g] < 0)
+            throw std::runtime_error(
+                "set_state: owned element missing from this submesh");
+        }
+      } else if (rank_ == 0) {
+        res[g] = 1; // out-of-mesh particles live on rank 0
+      }
+    }
+    PT_HIP_CHECK(hipMemcpy(d_pos_, pos, n_ * 24, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(
+        hipMemcpy(d_elem_, lel.data(), n_ * 4, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(hipMemcpy(d_res_, res.data(), n_, hipMemcpyHostToDevice));
+    PT_HIP_CHECK(
+        hipMemcpy(d_esc_, escaped, n_, hipMemcpyHostToDevice));
+    n_frame_ = -1;
+  }
+
 private:
   void check_n(int64_t n) const {
     if (n != n_) throw std::runtime_error("global particle count mismatch");
@@ -1428,6 +1480,41 @@ public:
   std::vector<uint8_t> resident_mask() const override { return res_; }
   std::vector<double> positions() const override { return pos_; }
   std::vector<int32_t> elem_ids() const override { return elem_; }
+
+  std::vector<int32_t> elem_ids_global() const override {
+    std::vector<int32_t> out(n_, -1);
+    for (int64_t g = 0; g < n_; ++g)
+      if (res_[g] && elem_[g] >= 0) out[g] = dec_.l2g32[elem_[g]];
+    return out;
+  }
+
+  std::vector<uint8_t> escaped_mask() const override { return esc_; }
+
+  void set_state(const double *pos, const int32_t *gelem,
+                 const uint8_t *escaped, int64_t n) override {
+    check_n(n);
+    for (int64_t g = 0; g < n_; ++g) {
+      res_[g] = 0;
+      elem_[g] = -1;
+      const int32_t ge = gelem[g];
+      if (ge >= 0) {
+        if (dec_.owners[ge] == rank_) {
+          res_[g] = 1;
+          elem_[g] = dec_.g2l[ge];
+          if (elem_[g] < 0)
+            throw std::runtime_error(
+                "set_state: owned element missing from this submesh");
+        }
+      } else if (rank_ == 0) {
+        res_[g] = 1;
+      }
+      pos_[g * 3] = pos[g * 3];
+      pos_[g * 3 + 1] = pos[g * 3 + 1];
+      pos_[g * 3 + 2] = pos[g * 3 + 2];
+      esc_[g] = escaped[g];
+    }
+    have_frame_ = false;
+  }
 
 public:
   std::function<int32_t(Vec3, double, bool *)> full_locate_;
