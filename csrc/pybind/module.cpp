@@ -446,6 +446,38 @@ PYBIND11_MODULE(_core, m) {
              std::memcpy(out.mutable_data(), v.data(), v.size() * 4);
              return out;
            })
+      // decomposition-independent state transfer (checkpoint/resume and
+      // repartitioning -- see partition_engine.h)
+      .def("elem_ids_global",
+           [](const PyPartEngine &s) {
+             auto v = s.pe->elem_ids_global();
+             py::array_t<int32_t> out(v.size());
+             std::memcpy(out.mutable_data(), v.data(), v.size() * 4);
+             return out;
+           })
+      .def("escaped_mask",
+           [](const PyPartEngine &s) {
+             auto v = s.pe->escaped_mask();
+             py::array_t<uint8_t> out(v.size());
+             std::memcpy(out.mutable_data(), v.data(), v.size());
+             return out;
+           })
+      .def("set_state",
+           [](PyPartEngine &s,
+              py::array_t<double, py::array::c_style | py::array::forcecast>
+                  pos,
+              py::array_t<int32_t, py::array::c_style | py::array::forcecast>
+                  gelem,
+              py::array_t<uint8_t, py::array::c_style | py::array::forcecast>
+                  escaped) {
+             const int64_t n = s.pe->num_particles();
+             if ((int64_t)pos.size() != n * 3 ||
+                 (int64_t)gelem.size() != n ||
+                 (int64_t)escaped.size() != n)
+               throw std::runtime_error("set_state: size mismatch");
+             py::gil_scoped_release ng;
+             s.pe->set_state(pos.data(), gelem.data(), escaped.data(), n);
+           })
       .def("synchronize", [](PyPartEngine &s) { s.pe->synchronize(); })
       // coupled-host path: resident_list() snapshots this rank's
       // particles; step_local consumes arrays in exactly that order

@@ -377,3 +377,72 @@ def test_step_local_world1_gpu():
     ref.copy_initial_position(p0.ravel())
     ref.move_continue(d.ravel(), fly.copy(), w)
     assert np.allclose(pe.flux_global(), ref.flux(), atol=1e-11)
+
+
+def test_state_transfer_checkpoint_roundtrip():
+    """Decomposition-independent snapshot: walk, snapshot, rebuild a
+    FRESH engine, restore, continue -- total flux equals the
+    uninterrupted run (mid-batch flux is drained host-side)."""
+    mesh = pt.build_box(4, 4, 4)
+    n = 200
+    rng = np.random.default_rng(61)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    d1 = np.clip(p0 + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    d2 = np.clip(d1 + rng.normal(0, 0.3, size=(n, 3)), -0.1, 1.1)
+    w = rng.uniform(0.1, 1.0, n)
+    fly = np.ones(n, np.int8)
+
+    # uninterrupted
+    a = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    a.localize(p0.ravel())
+    a.step(d1.ravel(), fly, w)
+    a.step(d2.ravel(), fly.copy(), w)
+    want = a.flux_global()
+
+    # interrupted after step 1: snapshot -> fresh engine -> restore
+    b = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    b.localize(p0.ravel())
+    b.step(d1.ravel(), fly, w)
+    snap = (b.positions().copy(), b.elem_ids_global().copy(),
+            b.escaped_mask().copy())
+    flux_so_far = np.asarray(b.flux_global()).copy()
+    del b
+    c = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    c.set_state(*snap)
+    c.step(d2.ravel(), fly.copy(), w)
+    got = flux_so_far + np.asarray(c.flux_global())
+    assert np.allclose(got, want, atol=1e-12), np.abs(got - want).max()
+
+
+def test_repartition_by_reconstruction():
+    """Dynamic load balance for the stateful engine: build a NEW engine
+    with work-weighted owners and transfer the state; the continued walk
+    matches the un-repartitioned run."""
+    mesh = pt.build_box(4, 4, 4)
+    n = 200
+    rng = np.random.default_rng(67)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    d1 = np.clip(p0 + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    d2 = np.clip(d1 + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    w = rng.uniform(0.1, 1.0, n)
+    fly = np.ones(n, np.int8)
+
+    a = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    a.localize(p0.ravel())
+    a.step(d1.ravel(), fly, w)
+    flux1 = np.asarray(a.flux_global()).copy()
+    snap = (a.positions(), a.elem_ids_global(), a.escaped_mask())
+    # "measured work" = the first step's flux; weighted Morton owners
+    owners = pt._core.partition_morton(mesh, 1, flux1 + 1e-12)
+    b = pt._core.PartitionedEngine(mesh, n, device="cpu",
+                                   owners=np.asarray(owners, np.int32))
+    b.set_state(*snap)
+    b.step(d2.ravel(), fly.copy(), w)
+    got = flux1 + np.asarray(b.flux_global())
+
+    ref = pt._core.PartitionedEngine(mesh, n, device="cpu")
+    ref.localize(p0.ravel())
+    ref.step(d1.ravel(), fly.copy(), w)
+    ref.step(d2.ravel(), fly.copy(), w)
+    want = ref.flux_global()
+    assert np.allclose(got, want, atol=1e-12), np.abs(got - want).max()
