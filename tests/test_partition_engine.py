@@ -446,3 +446,28 @@ def test_repartition_by_reconstruction():
     ref.step(d2.ravel(), fly.copy(), w)
     want = ref.flux_global()
     assert np.allclose(got, want, atol=1e-12), np.abs(got - want).max()
+
+
+@pytest.mark.gpu
+def test_state_transfer_gpu_matches_cpu():
+    mesh = pt.build_box(4, 4, 4)
+    n = 200
+    rng = np.random.default_rng(71)
+    p0 = rng.uniform(0.05, 0.95, size=(n, 3))
+    d1 = np.clip(p0 + rng.normal(0, 0.3, size=(n, 3)), -0.05, 1.05)
+    d2 = np.clip(d1 + rng.normal(0, 0.3, size=(n, 3)), 0.02, 0.98)
+    w = rng.uniform(0.1, 1.0, n)
+    fly = np.ones(n, np.int8)
+
+    outs = {}
+    for dev in ("cpu", "cuda:0"):
+        a = pt._core.PartitionedEngine(mesh, n, device=dev)
+        a.localize(p0.ravel())
+        a.step(d1.ravel(), fly.copy(), w)
+        f1 = np.asarray(a.flux_global()).copy()
+        snap = (a.positions(), a.elem_ids_global(), a.escaped_mask())
+        b = pt._core.PartitionedEngine(mesh, n, device=dev)
+        b.set_state(*snap)
+        b.step(d2.ravel(), fly.copy(), w)
+        outs[dev] = f1 + np.asarray(b.flux_global())
+    assert np.allclose(outs["cpu"], outs["cuda:0"], atol=1e-10)
