@@ -14,6 +14,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <chrono>
 #include <cstdlib>
 #include <cstring>
@@ -40,6 +41,7 @@ void send_all(int fd, const void *buf, int64_t n) {
   const char *p = (const char *)buf;
   while (n > 0) {
     const ssize_t k = ::send(fd, p, (size_t)n, MSG_NOSIGNAL);
+    if (k < 0 && errno == EINTR) continue;
     if (k <= 0) die("send failed");
     p += k;
     n -= k;
@@ -50,11 +52,19 @@ void recv_all(int fd, void *buf, int64_t n) {
   char *p = (char *)buf;
   while (n > 0) {
     const ssize_t k = ::recv(fd, p, (size_t)n, 0);
+    if (k < 0 && errno == EINTR) continue;
     if (k <= 0) die("recv failed (peer closed?)");
     p += k;
     n -= k;
   }
 }
+
+// Every comm a process creates rendezvouses on its own derived port:
+// ranks construct comms in identical program order, so the Nth comm of
+// every rank agrees on port+13N and two back-to-back comms can never
+// cross-wire their handshakes (a fast rank's connection for comm N+1
+// arriving at comm N's listener).
+std::atomic<int> g_comm_seq{0};
 
 int connect_retry(const std::string &addr, int port, double timeout_s) {
   addrinfo hints{};
@@ -92,6 +102,7 @@ public:
   TcpComm(int rank, int world, const std::string &addr, int port)
       : rank_(rank), world_(world) {
     if (world_ < 2) return; // degenerate; all ops become no-ops
+    port += 13 * (g_comm_seq.fetch_add(1) % 97);
     if (rank_ == 0) {
       const int lfd = socket(AF_INET, SOCK_STREAM, 0);
       if (lfd < 0) die("socket failed");
