@@ -471,3 +471,36 @@ def test_state_transfer_gpu_matches_cpu():
         b.step(d2.ravel(), fly.copy(), w)
         outs[dev] = f1 + np.asarray(b.flux_global())
     assert np.allclose(outs["cpu"], outs["cuda:0"], atol=1e-10)
+
+
+def test_edge_cases_small_and_idle():
+    """Degenerate shapes: 1 particle, all-stopped steps, empty frames,
+    step after everyone escaped."""
+    mesh = pt.build_box(2, 2, 2)
+    pe = pt._core.PartitionedEngine(mesh, 1, device="cpu")
+    pe.localize(np.array([0.5, 0.6, 0.7]))
+    assert pe.resident == 1
+    # all-stopped step: nothing moves, nothing tallies
+    pe.step(np.array([0.9, 0.6, 0.7]), np.zeros(1, np.int8), np.ones(1))
+    assert pe.flux_global().sum() == 0.0
+    # escape everyone
+    pe.step(np.array([5.0, 0.6, 0.7]), np.ones(1, np.int8), np.ones(1))
+    f1 = pe.flux_global().sum()
+    assert f1 > 0
+    # step an escaped particle again (continue): walks from clipped pos
+    pe.step(np.array([0.2, 0.6, 0.7]), np.ones(1, np.int8), np.ones(1))
+    assert pe.flux_global().sum() > f1
+    # coupled-host with zero flying residents
+    gids = pe.resident_list()
+    pe.step_local(np.zeros((len(gids), 3)).ravel(),
+                  np.zeros(len(gids), np.int8), np.ones(len(gids)))
+    # n_local mismatch must throw
+    pe.resident_list()
+    with pytest.raises(RuntimeError, match="resident_list"):
+        pe.step_local(np.zeros(6), np.zeros(2, np.int8), np.ones(2))
+    # step_local without a snapshot must throw
+    pe2 = pt._core.PartitionedEngine(mesh, 1, device="cpu")
+    pe2.localize(np.array([0.5, 0.5, 0.25]))
+    pe2.step(np.array([0.5, 0.5, 0.3]), np.ones(1, np.int8), np.ones(1))
+    with pytest.raises(RuntimeError, match="resident_list"):
+        pe2.step_local(np.zeros(3), np.ones(1, np.int8), np.ones(1))
