@@ -49,10 +49,13 @@ namespace pumitally {
 namespace {
 
 struct OshStream {
-  std::vector<unsigned char> bytes;
+  // non-owning view: the prober restarts several candidate layouts over
+  // one in-memory copy of the file (no per-candidate duplication)
+  const std::vector<unsigned char> *buf = nullptr;
   size_t at = 0;
   std::string path;
 
+  const std::vector<unsigned char> &bytes() const { return *buf; }
   [[noreturn]] void fail(const std::string &why) const {
     throw std::runtime_error(
         "Omega_h .osh parse failed at byte " + std::to_string(at) + " of " +
@@ -60,17 +63,17 @@ struct OshStream {
         " (the layout is probed best-effort; convert the mesh to Gmsh .msh "
         "— ASCII or binary — for a fully supported path)");
   }
-  size_t remaining() const { return bytes.size() - at; }
+  size_t remaining() const { return bytes().size() - at; }
   template <class T> T peek(size_t off = 0) const {
     T v{};
-    if (at + off + sizeof(T) > bytes.size()) return v;
-    memcpy(&v, bytes.data() + at + off, sizeof v);
+    if (at + off + sizeof(T) > bytes().size()) return v;
+    memcpy(&v, bytes().data() + at + off, sizeof v);
     return v;
   }
   template <class T> T take() {
-    if (at + sizeof(T) > bytes.size()) fail("truncated value");
+    if (at + sizeof(T) > bytes().size()) fail("truncated value");
     T v{};
-    memcpy(&v, bytes.data() + at, sizeof v);
+    memcpy(&v, bytes().data() + at, sizeof v);
     at += sizeof v;
     return v;
   }
@@ -89,7 +92,7 @@ std::vector<unsigned char> take_payload(OshStream &s, int64_t count,
       s.at += 8;
       std::vector<unsigned char> out(raw);
       uLongf dlen = (uLongf)raw;
-      const int rc = uncompress(out.data(), &dlen, s.bytes.data() + s.at,
+      const int rc = uncompress(out.data(), &dlen, s.bytes().data() + s.at,
                                 (uLong)cb);
       if (rc != Z_OK || dlen != raw)
         s.fail("zlib payload did not decode to the expected " +
@@ -100,8 +103,8 @@ std::vector<unsigned char> take_payload(OshStream &s, int64_t count,
     }
   }
   if (s.remaining() < raw) s.fail("raw payload truncated");
-  std::vector<unsigned char> out(s.bytes.begin() + s.at,
-                                 s.bytes.begin() + s.at + raw);
+  std::vector<unsigned char> out(s.bytes().begin() + s.at,
+                                 s.bytes().begin() + s.at + raw);
   s.at += raw;
   return out;
 }
@@ -131,17 +134,17 @@ void skip_i8_array(OshStream &s, int64_t expect_count) {
 std::vector<double> find_coordinates(OshStream &s, int64_t nverts) {
   static const char kName[] = "coordinates";
   const size_t nl = sizeof(kName) - 1;
-  for (size_t p = s.at; p + nl < s.bytes.size(); ++p) {
-    if (memcmp(s.bytes.data() + p, kName, nl) != 0) continue;
+  for (size_t p = s.at; p + nl < s.bytes().size(); ++p) {
+    if (memcmp(s.bytes().data() + p, kName, nl) != 0) continue;
     // after the name: some small header fields (ncomps, class ids...),
     // then [int32 count == 3*nverts][payload of doubles]
-    for (size_t q = p + nl; q <= p + nl + 32 && q + 4 <= s.bytes.size();
+    for (size_t q = p + nl; q <= p + nl + 32 && q + 4 <= s.bytes().size();
          ++q) {
       int32_t cnt = 0;
-      memcpy(&cnt, s.bytes.data() + q, 4);
+      memcpy(&cnt, s.bytes().data() + q, 4);
       if ((int64_t)cnt != nverts * 3) continue;
       OshStream sub;
-      sub.bytes = s.bytes;
+      sub.buf = s.buf;
       sub.path = s.path;
       sub.at = q + 4;
       try {
@@ -161,7 +164,8 @@ std::vector<double> find_coordinates(OshStream &s, int64_t nverts) {
   s.fail("no decodable 'coordinates' vertex tag (3*nverts doubles) found");
 }
 
-Mesh parse_with_meta_skip(OshStream s, int meta_i32s) {
+Mesh parse_with_meta_skip(OshStream s /* cursor copy; shared buffer */,
+                          int meta_i32s) {
   // meta block: `meta_i32s` int32 fields we do not interpret beyond
   // requiring the dim field (3) to appear among them
   bool saw3 = false;
@@ -218,7 +222,7 @@ Mesh parse_with_meta_skip(OshStream s, int meta_i32s) {
         tri_v[t * 3 + m++] = u[k];
       }
     if (m != 3)
-      OshStream{{}, 0, s.path}.fail(
+      OshStream{nullptr, 0, s.path}.fail(
           "triangle " + std::to_string(t) +
           " edge-union does not have exactly 3 vertices");
   }
@@ -236,7 +240,7 @@ Mesh parse_with_meta_skip(OshStream s, int meta_i32s) {
         tet_v[t * 4 + m++] = u[k];
       }
     if (m != 4)
-      OshStream{{}, 0, s.path}.fail(
+      OshStream{nullptr, 0, s.path}.fail(
           "tet " + std::to_string(t) +
           " face-union does not have exactly 4 vertices");
   }
@@ -258,8 +262,8 @@ Mesh parse_with_meta_skip(OshStream s, int meta_i32s) {
                        ab[1] * (ac[0] * ad[2] - ac[2] * ad[0]) +
                        ab[2] * (ac[0] * ad[1] - ac[1] * ad[0]);
     if (det == 0.0)
-      OshStream{{}, 0, s.path}.fail("degenerate tet volume during "
-                                    "orientation restore");
+      OshStream{nullptr, 0, s.path}.fail("degenerate tet volume during "
+                                         "orientation restore");
     if (det < 0.0) {
       std::swap(tet_v[t * 4], tet_v[t * 4 + 1]);
       flipped++;
@@ -281,11 +285,13 @@ Mesh parse_with_meta_skip(OshStream s, int meta_i32s) {
 Mesh read_osh_omegah_stream(const std::string &stream_path) {
   std::ifstream f(stream_path, std::ios::binary);
   if (!f) throw std::runtime_error("cannot open " + stream_path);
+  const std::vector<unsigned char> bytes(
+      (std::istreambuf_iterator<char>(f)),
+      std::istreambuf_iterator<char>());
   OshStream s;
+  s.buf = &bytes;
   s.path = stream_path;
-  s.bytes.assign(std::istreambuf_iterator<char>(f),
-                 std::istreambuf_iterator<char>());
-  if (s.bytes.size() < 16 || s.bytes[0] != 0xa1 || s.bytes[1] != 0x1a)
+  if (bytes.size() < 16 || bytes[0] != 0xa1 || bytes[1] != 0x1a)
     throw std::runtime_error(stream_path + ": not an Omega_h stream");
   s.at = 2;
   const int32_t version = s.take<int32_t>();
@@ -297,8 +303,8 @@ Mesh read_osh_omegah_stream(const std::string &stream_path) {
   std::string errors;
   for (int meta : {2, 3, 4, 5, 6, 1, 0, 7, 8}) {
     try {
-      OshStream probe = s; // copy (cheap enough once per candidate)
-      return parse_with_meta_skip(std::move(probe), meta);
+      OshStream probe = s; // cursor copy; the byte buffer is shared
+      return parse_with_meta_skip(probe, meta);
     } catch (const std::exception &e) {
       errors += std::string("\n  [meta=") + std::to_string(meta) + "] " +
                 e.what();
