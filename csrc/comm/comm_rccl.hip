@@ -116,6 +116,8 @@ public:
     return out;
   }
 
+  bool has_device_collectives() const override { return true; }
+
   void allreduce_sum_device(double *d_data, int64_t n) override {
     PT_HIP_CK(hipSetDevice(device_));
     PT_NCCL_CK(ncclAllReduce(d_data, d_data, (size_t)n, ncclDouble, ncclSum,
