@@ -600,7 +600,33 @@ PYBIND11_MODULE(_core, m) {
            [](Comm &c, uintptr_t ptr, int64_t n) {
              py::gil_scoped_release ng;
              c.allreduce_sum_device((double *)ptr, n);
+           })
+      // device all-to-all-v: returns (device pointer, total doubles);
+      // the buffer is comm-owned and valid until the next device call
+      .def("alltoallv_device",
+           [](Comm &c, uintptr_t d_send,
+              const std::vector<int64_t> &send_counts,
+              const std::vector<int64_t> &recv_counts) {
+             double *d_recv = nullptr;
+             int64_t tot = 0;
+             {
+               py::gil_scoped_release ng;
+               tot = c.alltoallv_device((const double *)d_send, send_counts,
+                                        recv_counts, &d_recv);
+             }
+             return py::make_tuple((uintptr_t)d_recv, tot);
            });
+  // debug/test helper: copy n doubles from a device pointer to numpy
+  m.def("d2h_doubles", [](uintptr_t ptr, int64_t n) {
+    auto out = py::array_t<double>(n);
+    if (n) {
+      py::gil_scoped_release ng;
+      if (hipMemcpy(out.mutable_data(), (const void *)ptr, n * 8,
+                    hipMemcpyDeviceToHost) != hipSuccess)
+        throw std::runtime_error("d2h_doubles: hipMemcpy failed");
+    }
+    return out;
+  });
   m.def(
       "make_native_comm",
       [](bool want_gpu, int device) {

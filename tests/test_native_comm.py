@@ -237,3 +237,22 @@ def test_facade_world2_tcp_gpu_engines(tmp_path):
             "PUMITALLY_COMM": "tcp",     # comm stays off the GPU
         })
     assert "FACADE_WORLD2_OK" in outs[0]
+
+
+@pytest.mark.gpu
+def test_rccl_alltoallv_device_world1():
+    """The raw device all-to-all-v (ncclSend/ncclRecv group) with torch
+    device tensors, the exact call shape the partitioned engine uses for
+    its record exchange (has_device_collectives path)."""
+    import torch
+
+    import pumiumtally_amd as pt
+
+    comm = pt._core.make_rccl_comm(0, 1, "127.0.0.1", 29877, 0)
+    assert comm is not None
+    send = torch.arange(40, dtype=torch.float64, device="cuda:0") * 0.5
+    torch.cuda.synchronize()
+    ptr, tot = comm.alltoallv_device(send.data_ptr(), [40], [40])
+    assert tot == 40 and ptr != 0
+    got = pt._core.d2h_doubles(ptr, tot)
+    assert np.array_equal(got, send.cpu().numpy())
