@@ -63,3 +63,10 @@ print(f"flux mean total {mean[0].sum():.2f}, "
 eng.set_flux(mean.reshape(-1) * eng._eng.num_batches)
 eng.write_tally_results("features_flux.vtu")
 print("wrote features_flux.vtu (fields: flux, flux_g*, score1, score1_g*)")
+
+# -- failure-path observability ------------------------------------------
+# loose-tolerance localizations and lost walks are counted and the first
+# lost walks are captured with drop positions (the reference only
+# printfs "Not all particles are found")
+stats = eng.stats()
+print(f"stats: {stats}; lost records shape {eng.lost_records().shape}")
