@@ -343,6 +343,32 @@ class PartitionedTally:
             return out[:, 0]
         return out
 
+    def write_tally_pvtu(self, basename="fluxresult"):
+        """Parallel VTK output: rank-owned .vtu pieces + a .pvtu master
+        (the Omega_h vtk::write_parallel analog for a real
+        decomposition).  Writes basename.pvtu + basename_p<r>.vtu;
+        fields are the volume-normalized reduced flux (total + per-group
+        / per-score when enabled)."""
+        from .. import _core
+        from ..mesh import write_pvtu
+
+        f = np.asarray(self.flux_global()).reshape(
+            self.nscores, self.ngroups, self.mesh.nelems)
+        fields = []
+        for k in range(self.nscores):
+            name = "flux" if k == 0 else f"score{k}"
+            fields.append((name, _core.normalize_flux(
+                self.mesh, f[k].sum(axis=0))))
+            if self.ngroups > 1:
+                fields += [(f"{name}_g{g}", _core.normalize_flux(
+                                self.mesh, f[k, g]))
+                           for g in range(self.ngroups)]
+        write_pvtu(basename, self.mesh, self.owners, self.rank,
+                   self.world, fields)
+        if self.world > 1:
+            import torch.distributed as dist
+            dist.barrier()
+
     def write_tally_results(self, filename="fluxresult.vtk"):
         from .. import _core, write_tally_vtk
 
