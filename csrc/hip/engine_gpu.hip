@@ -445,7 +445,7 @@ int flux_slices(int64_t flux_doubles) {
   const char *s = getenv("PUMITALLY_FLUX_SLICES");
   int k = s ? atoi(s) : 64;
   if (k < 1) k = 1;
-  if (k > 64) k = 64;
+  if (k > 256) k = 256;
   while (k & (k - 1)) k--; // power of two for the cheap in-kernel mask
   if (!s) {
     // adaptive: keep the slice copies under ~1 GiB
