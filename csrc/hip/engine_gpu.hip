@@ -437,19 +437,21 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
 // blockIdx & (S-1), reduced at readout.  The hot-element (point-source)
 // mitigation: measured on MI355X (profiles/README.md round 2), the
 // config-4 contention stress at 10M particles through a 2%-cube source
-// runs 74.2 ms/step with 1 slice and 7.8 ms with 64 -- 9.5x -- while the
-// spread-source case is unchanged (5.48 -> 5.58 ms at 16).  Default is
-// adaptive: 64 slices when the extra copies fit a modest HBM budget
-// (trivial against 288 GB for 1M-tet meshes), halved until they do.
+// runs 74.2 ms/step with 1 slice, 7.8 ms with 64 and 6.48 ms with 256
+// (11.4x, within 18% of the uncontended walk), while spread sources and
+// the headline are unchanged to the noise floor at any slice count.
+// Default is adaptive: 256 slices when the copies fit a 2 GiB HBM
+// budget (trivial against 288 GB for 1M-tet meshes), halved until they
+// do.
 int flux_slices(int64_t flux_doubles) {
   const char *s = getenv("PUMITALLY_FLUX_SLICES");
-  int k = s ? atoi(s) : 64;
+  int k = s ? atoi(s) : 256;
   if (k < 1) k = 1;
   if (k > 256) k = 256;
   while (k & (k - 1)) k--; // power of two for the cheap in-kernel mask
   if (!s) {
-    // adaptive: keep the slice copies under ~1 GiB
-    const int64_t budget = (int64_t)1 << 30;
+    // adaptive: keep the slice copies under ~2 GiB
+    const int64_t budget = (int64_t)2 << 30;
     while (k > 1 && flux_doubles * k * 8 > budget) k >>= 1;
   }
   return k;
