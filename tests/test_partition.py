@@ -648,3 +648,62 @@ def test_per_face_reflective_carried_into_submesh():
     pe.localize(o.ravel())
     pe.step(d.ravel(), np.ones(n, np.int8), w)
     assert np.allclose(pe.flux_global(), ref.flux(), atol=1e-12)
+
+
+PVTU_WORKER = r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+from pumiumtally_amd.parallel.partition import PartitionedTally
+
+rank = int(os.environ["RANK"])
+mesh = pt.build_box(3, 3, 3)
+n = 100
+rng = np.random.default_rng(23)
+o = rng.uniform(0.05, 0.95, size=(n, 3))
+d = rng.uniform(0.05, 0.95, size=(n, 3))
+w = rng.uniform(0.1, 1.0, n)
+ptal = PartitionedTally(mesh, device="cpu", backend="gloo")
+ptal.run_segments(o, d, w)
+ptal.write_tally_pvtu(os.environ["PT_BASE"])
+if rank == 0:
+    print("PVTU_OK")
+import torch.distributed as dist
+dist.destroy_process_group()
+"""
+
+
+def test_pvtu_world2(tmp_path):
+    pytest.importorskip("torch")
+    script = tmp_path / "worker.py"
+    script.write_text(PVTU_WORKER)
+    env = dict(os.environ)
+    env.update({
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(20000 + (os.getpid() + 311) % 20000),
+        "WORLD_SIZE": "2",
+        "PT_BASE": str(tmp_path / "flux"),
+        "PYTHONPATH": os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        e["LOCAL_RANK"] = str(r)
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=180)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {r} failed:\n{out}"
+    assert "PVTU_OK" in outs[0]
+    assert (tmp_path / "flux.pvtu").exists()
+    assert (tmp_path / "flux_p0.vtu").exists()
+    assert (tmp_path / "flux_p1.vtu").exists()
+    import re
+    total = 0
+    for r in range(2):
+        txt = (tmp_path / f"flux_p{r}.vtu").read_bytes().decode("latin1")
+        total += int(re.search(r'NumberOfCells="(\d+)"', txt).group(1))
+    m = pt.build_box(3, 3, 3)
+    assert total == m.nelems
