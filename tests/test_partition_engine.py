@@ -504,3 +504,62 @@ def test_edge_cases_small_and_idle():
     pe2.step(np.array([0.5, 0.5, 0.3]), np.ones(1, np.int8), np.ones(1))
     with pytest.raises(RuntimeError, match="resident_list"):
         pe2.step_local(np.zeros(3), np.ones(1, np.int8), np.ones(1))
+
+
+def test_periodic_world1_supported_and_worldN_documented(tmp_path):
+    """Stateful engine + periodic: world-1 works (all pairs local);
+    world>1 is a documented throw pointing at the stateless driver."""
+    m = pt.build_box(3, 3, 3)
+    fid, cen, nor = m.boundary_faces()
+    hi = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+    lo = fid[np.abs(cen[:, 0] - 0.0) < 1e-12]
+    m.set_periodic_faces(hi, lo, np.array([-1.0, 0.0, 0.0]))
+
+    n = 40
+    rng = np.random.default_rng(83)
+    o = rng.uniform(0.05, 0.95, size=(n, 3))
+    d = o.copy()
+    d[:, 0] += rng.uniform(0.3, 1.2, n)  # wraps
+    d[:, 1:] = np.clip(d[:, 1:], 0.05, 0.95)
+    w = rng.uniform(0.5, 1.5, n)
+    pe = pt._core.PartitionedEngine(m, n, device="cpu")
+    pe.localize(o.ravel())
+    pe.step(d.ravel(), np.ones(n, np.int8), w)
+    ref = pt.TallyEngine(m, n, device="cpu")
+    ref.copy_initial_position(o.ravel())
+    ref.move(o.ravel(), d.ravel(), np.ones(n, np.int8), w)
+    assert np.allclose(pe.flux_global(), ref.flux(), atol=1e-12)
+
+    # world>1 rejection: spawn two ranks and expect the documented error
+    script = tmp_path / "w.py"
+    script.write_text(r"""
+import os
+import numpy as np
+import pumiumtally_amd as pt
+m = pt.build_box(3, 3, 3)
+fid, cen, nor = m.boundary_faces()
+hi = fid[np.abs(cen[:, 0] - 1.0) < 1e-12]
+lo = fid[np.abs(cen[:, 0] - 0.0) < 1e-12]
+m.set_periodic_faces(hi, lo, np.array([-1.0, 0.0, 0.0]))
+try:
+    pt._core.PartitionedEngine(m, 10, device="cpu")
+except RuntimeError as e:
+    assert "PartitionedTally" in str(e), e
+    print("PERIODIC_REJECT_OK")
+""")
+    env = dict(os.environ)
+    env.update({
+        "WORLD_SIZE": "2", "MASTER_ADDR": "127.0.0.1",
+        "PUMITALLY_PORT": str(27000 + (os.getpid() + 3) % 12000),
+        "PUMITALLY_NO_TORCH": "1", "PYTHONPATH": ROOT,
+    })
+    procs = []
+    for r in range(2):
+        e = dict(env)
+        e["RANK"] = str(r)
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = [p.communicate(timeout=120)[0].decode() for p in procs]
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert "PERIODIC_REJECT_OK" in out, f"rank {r}:\n{out}"
