@@ -181,7 +181,7 @@ def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
                 # replicated headline metric pays the reference-API origin
                 # leg (57 B) -- metric names differ accordingly
                 "step_mode": "continue",
-                "comm": "native-rccl",
+                "comm": "native-" + pe.comm_kind,
                 "ngroups": args.ngroups,
                 "nscores": args.nscores,
                 "device": "gpu" if device != "cpu" else "cpu",

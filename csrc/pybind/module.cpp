@@ -549,6 +549,12 @@ PYBIND11_MODULE(_core, m) {
                s.comm->barrier();
              }
            })
+      .def_property_readonly(
+          "comm_kind",
+          [](PyPartEngine &s) -> std::string {
+            if (!s.comm) return "local";
+            return s.comm->has_device_collectives() ? "rccl" : "tcp";
+          })
       .def("allreduce_max",
            [](PyPartEngine &s,
               py::array_t<double, py::array::c_style | py::array::forcecast>
