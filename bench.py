@@ -86,6 +86,15 @@ class CommShim:
         self._comm = None
 
 
+def _alloc(pt, shape, dtype, pin):
+    """Pinned host array when the engine is on a GPU (pageable sources
+    demote the chunked hipMemcpyAsync pipeline), plain numpy otherwise."""
+    import numpy as np
+    if pin:
+        return pt.pinned_array(shape, dtype)
+    return np.empty(shape, dtype)
+
+
 def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
     """BASELINE config 3 (default partitioned mode): Morton element
     partition across ranks, ghost rings, cross-rank handoff over RCCL,
@@ -111,15 +120,20 @@ def run_partitioned_stateful(args, mesh, cells, rank, world, local, device):
     d = np.asarray(p1).reshape(-1)
     w = np.asarray(weights)
     fly = np.asarray(flying)
+    # groups/responses ride the same chunked H2D pipeline as dest/weights,
+    # so they must be pinned too (pageable numpy arrays demote
+    # hipMemcpyAsync to staged synchronous copies and serialize the step)
     groups = None
     if args.ngroups > 1:
         rng_g = np.random.default_rng(1234)
-        groups = rng_g.integers(0, args.ngroups, n_global).astype(np.uint16)
+        groups = _alloc(pt, (n_global,), "uint16", pin)
+        groups[:] = rng_g.integers(0, args.ngroups, n_global)
 
     responses = None
     if args.nscores > 1:
         rng_r = np.random.default_rng(4321)
-        responses = rng_r.uniform(0.5, 2.0, size=(n_global, args.nscores))
+        responses = _alloc(pt, (n_global, args.nscores), "float64", pin)
+        responses[:] = rng_r.uniform(0.5, 2.0, size=(n_global, args.nscores))
     pe = pt._core.PartitionedEngine(mesh, n_global, device=device,
                                     ngroups=args.ngroups,
                                     nscores=args.nscores)
@@ -369,12 +383,15 @@ def main():
     groups = None
     if args.ngroups > 1:
         rng_g = np.random.default_rng(1234 + rank)
-        groups = rng_g.integers(0, args.ngroups, args.particles).astype(np.uint16)
+        groups = _alloc(pt, (args.particles,), "uint16", eng.is_gpu)
+        groups[:] = rng_g.integers(0, args.ngroups, args.particles)
     responses = None
     if args.nscores > 1:
         rng_r = np.random.default_rng(4321 + rank)
-        responses = rng_r.uniform(0.5, 2.0,
-                                  size=(args.particles, args.nscores))
+        responses = _alloc(pt, (args.particles, args.nscores), "float64",
+                           eng.is_gpu)
+        responses[:] = rng_r.uniform(0.5, 2.0,
+                                     size=(args.particles, args.nscores))
     eng.copy_initial_position(p0.reshape(-1))
     eng.synchronize()
 
