@@ -121,13 +121,26 @@ public:
   // termination -- reflective restarts mirror it, periodic restarts
   // translate it, so a handoff (status 2) must resume toward out_dest,
   // not the original dest.
+  // Bitwise-exact handoff resume (walk.h walk_segment doc): in_t/in_prev
+  // (optional, n) seed the walk's segment progress and entry-exclusion
+  // element for resumed particles; out_o (n*3) / out_t (n) / out_prev
+  // (n) export the wrap-segment origin, the progress t at termination,
+  // and the element a handoff exited from.  A handoff record carrying
+  // (out_o, out_dest, out_t, out_prev) lets the receiving rank replay
+  // the remaining crossings with fp decisions identical to an uncut
+  // walk, so partitioned flux attribution matches the replicated
+  // engine elementwise.
   virtual void walk_raw(int64_t n, const double *pos, const double *dest,
                         const int32_t *elem, const double *weights,
                         double *out_pos, int32_t *out_elem,
                         int8_t *out_status,
                         const uint16_t *groups = nullptr,
                         const double *responses = nullptr,
-                        double *out_dest = nullptr) = 0;
+                        double *out_dest = nullptr,
+                        const double *in_t = nullptr,
+                        const int32_t *in_prev = nullptr,
+                        double *out_o = nullptr, double *out_t = nullptr,
+                        int32_t *out_prev = nullptr) = 0;
 
   // Device-resident walk_raw: every array already lives in this engine's
   // device memory (no staging at all -- the partitioned driver keeps its
@@ -139,10 +152,16 @@ public:
                                int32_t *d_out_elem, int8_t *d_out_status,
                                const uint16_t *d_groups = nullptr,
                                const double *d_responses = nullptr,
-                               double *d_out_dest = nullptr) {
+                               double *d_out_dest = nullptr,
+                               const double *d_in_t = nullptr,
+                               const int32_t *d_in_prev = nullptr,
+                               double *d_out_o = nullptr,
+                               double *d_out_t = nullptr,
+                               int32_t *d_out_prev = nullptr) {
     (void)n; (void)d_pos; (void)d_dest; (void)d_elem; (void)d_weights;
     (void)d_out_pos; (void)d_out_elem; (void)d_out_status; (void)d_groups;
-    (void)d_responses; (void)d_out_dest;
+    (void)d_responses; (void)d_out_dest; (void)d_in_t; (void)d_in_prev;
+    (void)d_out_o; (void)d_out_t; (void)d_out_prev;
     throw std::runtime_error("walk_raw_device requires the GPU engine");
   }
 

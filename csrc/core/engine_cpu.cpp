@@ -210,7 +210,11 @@ public:
                 int32_t *out_elem, int8_t *out_status,
                 const uint16_t *groups = nullptr,
                 const double *responses = nullptr,
-                double *out_dest = nullptr) override {
+                double *out_dest = nullptr,
+                const double *in_t = nullptr,
+                const int32_t *in_prev = nullptr, double *out_o = nullptr,
+                double *out_t = nullptr,
+                int32_t *out_prev = nullptr) override {
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     const unsigned hw = std::thread::hardware_concurrency();
     if (n >= 65536 && hw > 1) {
@@ -226,7 +230,8 @@ public:
           int64_t my_lost = 0;
           for (int64_t i = lo; i < hi; ++i)
             walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
-                         out_elem, out_status, out_dest, i, steps,
+                         out_elem, out_status, out_dest, in_t, in_prev,
+                         out_o, out_t, out_prev, i, steps,
                          partial[t].data(), my_lost);
           lost += my_lost;
         });
@@ -240,8 +245,8 @@ public:
     int64_t lost = 0;
     for (int64_t i = 0; i < n; ++i)
       walk_raw_one(pos, dest, elem, weights, groups, responses, out_pos,
-                   out_elem, out_status, out_dest, i, steps, flux_.data(),
-                   lost);
+                   out_elem, out_status, out_dest, in_t, in_prev, out_o,
+                   out_t, out_prev, i, steps, flux_.data(), lost);
     stats_.lost_particles += lost;
   }
 
@@ -249,8 +254,10 @@ public:
                     const int32_t *elem, const double *weights,
                     const uint16_t *groups, const double *responses,
                     double *out_pos, int32_t *out_elem, int8_t *out_status,
-                    double *out_dest, int64_t i, int steps, double *flux_out,
-                    int64_t &lost) {
+                    double *out_dest, const double *in_t,
+                    const int32_t *in_prev, double *out_o, double *out_t,
+                    int32_t *out_prev, int64_t i, int steps,
+                    double *flux_out, int64_t &lost) {
     {
       const Vec3 o{pos[i * 3], pos[i * 3 + 1], pos[i * 3 + 2]};
       const Vec3 d{dest[i * 3], dest[i * 3 + 1], dest[i * 3 + 2]};
@@ -274,17 +281,24 @@ public:
       const int32_t *pix =
           mesh_.periodic_idx.empty() ? nullptr : mesh_.periodic_idx.data();
       Vec3 od{d.x, d.y, d.z};
+      const double rt = in_t ? in_t[i] : 0.0;
+      const int32_t rp = in_prev ? in_prev[i] : -1;
+      Vec3 oo{o.x, o.y, o.z};
+      double ot = 0.0;
+      int32_t opv = -1;
       if (walk_fp32)
         walk_segment32<true>(mesh_.planes.data(), mesh_.planes32.data(),
                              mesh_.nbr.data(), elem[i], o, d, weights[i],
                              steps, add, &oe, &op, &esc, reflective, bc, pix,
                              mesh_.periodic_elem.data(),
-                             mesh_.periodic_shift.data(), &od);
+                             mesh_.periodic_shift.data(), &od, rt, rp, &oo,
+                             &ot, &opv);
       else
         walk_segment<true>(mesh_.planes.data(), mesh_.nbr.data(), elem[i], o,
                            d, weights[i], steps, add, &oe, &op, &esc,
                            reflective, bc, pix, mesh_.periodic_elem.data(),
-                           mesh_.periodic_shift.data(), &od);
+                           mesh_.periodic_shift.data(), &od, rt, rp, &oo,
+                           &ot, &opv);
       int8_t st = 0;
       if (oe == kWalkLost) {
         st = 3;
@@ -306,6 +320,13 @@ public:
         out_dest[i * 3 + 1] = od.y;
         out_dest[i * 3 + 2] = od.z;
       }
+      if (out_o) {
+        out_o[i * 3] = oo.x;
+        out_o[i * 3 + 1] = oo.y;
+        out_o[i * 3 + 2] = oo.z;
+      }
+      if (out_t) out_t[i] = ot;
+      if (out_prev) out_prev[i] = opv;
     }
   }
 

@@ -211,7 +211,13 @@ PT_HD bool walk_advance(const Plane *__restrict__ planes,
   if (next < -1) {
     // Partitioned submesh: the face crosses into an element owned by
     // another rank.  Stop at the crossing; the caller decodes the
-    // foreign reference (k = -(next+2)) and ships the particle.
+    // foreign reference (k = -(next+2)) and ships the particle.  The
+    // state keeps (s.o, s.t_cur, s.elem) so the handoff record can carry
+    // the walk's t-parametrization and the receiver replays the
+    // remaining crossings with the SAME fp decisions as an uncut walk
+    // (bitwise-identical element attribution; see walk_segment's
+    // resume_t/resume_prev).
+    s.t_cur = t_clamped;
     *out_elem = next;
     *out_pos = s.o + t_clamped * (s.d - s.o);
     *out_escaped = false;
@@ -327,6 +333,7 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
     return true;
   }
   if (next < -1) {
+    s.t_cur = t_clamped; // export resume state (see fp64 variant)
     *out_elem = next;
     *out_pos = s.o + t_clamped * (s.d - s.o);
     *out_escaped = false;
@@ -345,6 +352,21 @@ PT_HD bool walk_advance32(const Plane *__restrict__ planes,
 // translate it, so after either a handoff record must ship s.d -- not
 // the caller's original d -- or the receiving rank walks to a stale
 // point (the round-2 periodic-partitioned ping-pong bug).
+//
+// Bitwise-exact handoff resume (resume_t / resume_prev in, out_o / out_t
+// / out_prev out): the walk is t-parametrized over (o, d), so a
+// cut-crossing record that carries the CURRENT wrap-segment origin s.o
+// (as its position), the progress t at the crossing, and the element
+// exited from lets the receiving rank continue with s.t_cur = t and
+// s.prev_elem = that element -- every subsequent plane evaluation,
+// monotone-t clamp and exit-face choice then reproduces the uncut
+// walk's fp decisions bit for bit, and partitioned flux attribution is
+// ELEMENTWISE identical to the replicated engine's (not merely
+// conservative).  Without this, a resumed walk re-based at the crossing
+// point computes crossings from perturbed endpoints, and a track
+// passing within fp noise of a face-edge junction can attribute its
+// final sliver to the neighboring tet (found by tools/part_world2_soak
+// at ~1 per 10^5 handoffs; conservation still held).
 template <bool Periodic = false, class FluxAdd>
 PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ nbr, int32_t elem, Vec3 o,
@@ -355,14 +377,22 @@ PT_HD void walk_segment(const Plane *__restrict__ planes,
                         const int32_t *__restrict__ pidx = nullptr,
                         const int32_t *__restrict__ pelem = nullptr,
                         const double *__restrict__ pshift = nullptr,
-                        Vec3 *out_dest = nullptr) {
+                        Vec3 *out_dest = nullptr, double resume_t = 0.0,
+                        int32_t resume_prev = -1, Vec3 *out_o = nullptr,
+                        double *out_t = nullptr,
+                        int32_t *out_prev = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
+  s.t_cur = resume_t;
+  s.prev_elem = resume_prev;
   while (!walk_advance<Periodic>(planes, nbr, s, max_steps, add, out_elem,
                                  out_pos, out_escaped, reflective, face_bc,
                                  pidx, pelem, pshift)) {
   }
   if (out_dest) *out_dest = s.d;
+  if (out_o) *out_o = s.o;
+  if (out_t) *out_t = s.t_cur;
+  if (out_prev) *out_prev = s.elem; // element exited from, on a handoff
 }
 
 template <bool Periodic = false, class FluxAdd>
@@ -376,14 +406,22 @@ PT_HD void walk_segment32(const Plane *__restrict__ planes,
                           const int32_t *__restrict__ pidx = nullptr,
                           const int32_t *__restrict__ pelem = nullptr,
                           const double *__restrict__ pshift = nullptr,
-                          Vec3 *out_dest = nullptr) {
+                          Vec3 *out_dest = nullptr, double resume_t = 0.0,
+                          int32_t resume_prev = -1, Vec3 *out_o = nullptr,
+                          double *out_t = nullptr,
+                          int32_t *out_prev = nullptr) {
   WalkState s;
   walk_init(s, elem, o, d, weight);
+  s.t_cur = resume_t;
+  s.prev_elem = resume_prev;
   while (!walk_advance32<Periodic>(planes, planes32, nbr, s, max_steps, add,
                                    out_elem, out_pos, out_escaped, reflective,
                                    face_bc, pidx, pelem, pshift)) {
   }
   if (out_dest) *out_dest = s.d;
+  if (out_o) *out_o = s.o;
+  if (out_t) *out_t = s.t_cur;
+  if (out_prev) *out_prev = s.elem;
 }
 
 // Point-in-tet test against the 4 planes (signed distances, unit normals).

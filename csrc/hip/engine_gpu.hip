@@ -368,7 +368,12 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
                            int64_t nelems, int nscores,
                            const int32_t *__restrict__ pidx = nullptr,
                            const int32_t *__restrict__ pelem = nullptr,
-                           const double *__restrict__ pshift = nullptr) {
+                           const double *__restrict__ pshift = nullptr,
+                           const double *__restrict__ in_t = nullptr,
+                           const int32_t *__restrict__ in_prev = nullptr,
+                           double *__restrict__ out_o = nullptr,
+                           double *__restrict__ out_t = nullptr,
+                           int32_t *__restrict__ out_prev = nullptr) {
   // XCD-aware block->range remap, same as k_move: MI355X dispatches
   // block b to XCD b%8; giving each XCD one contiguous (spatially
   // compact, since callers keep lists near-Morton-ordered) index range
@@ -396,14 +401,20 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
       }
     };
     Vec3 od{d.x, d.y, d.z};
+    const double rt = in_t ? in_t[i] : 0.0;
+    const int32_t rp = in_prev ? in_prev[i] : -1;
+    Vec3 oo{o.x, o.y, o.z};
+    double ot = 0.0;
+    int32_t opv = -1;
     if constexpr (F32)
       walk_segment32<true>(planes, planes32, nbr, elem[i], o, d, weights[i],
                            max_steps, add, &oe, &op, &esc, reflective,
-                           face_bc, pidx, pelem, pshift, &od);
+                           face_bc, pidx, pelem, pshift, &od, rt, rp, &oo,
+                           &ot, &opv);
     else
       walk_segment<true>(planes, nbr, elem[i], o, d, weights[i], max_steps,
                          add, &oe, &op, &esc, reflective, face_bc, pidx,
-                         pelem, pshift, &od);
+                         pelem, pshift, &od, rt, rp, &oo, &ot, &opv);
     int8_t st = 0;
     if (oe == kWalkLost) {
       st = 3;
@@ -430,6 +441,13 @@ __global__ void k_walk_raw(const Plane *__restrict__ planes,
       out_dest[i * 3 + 1] = od.y;
       out_dest[i * 3 + 2] = od.z;
     }
+    if (out_o) {
+      out_o[i * 3] = oo.x;
+      out_o[i * 3 + 1] = oo.y;
+      out_o[i * 3 + 2] = oo.z;
+    }
+    if (out_t) out_t[i] = ot;
+    if (out_prev) out_prev[i] = opv;
   }
 }
 
@@ -726,6 +744,8 @@ public:
     int32_t *elem = nullptr, *out_elem = nullptr;
     int8_t *status = nullptr;
     uint16_t *groups = nullptr;
+    double *in_t = nullptr, *out_o = nullptr, *out_t = nullptr;
+    int32_t *in_prev = nullptr, *out_prev = nullptr;
   };
 
   void ensure_wr_cap(int64_t n, bool need_groups, bool need_resp) {
@@ -749,7 +769,9 @@ public:
                     (void *)wr_.out_pos, (void *)wr_.elem,
                     (void *)wr_.out_elem, (void *)wr_.status,
                     (void *)wr_.groups, (void *)wr_.resp,
-                    (void *)wr_.out_dest})
+                    (void *)wr_.out_dest, (void *)wr_.in_t,
+                    (void *)wr_.out_o, (void *)wr_.out_t,
+                    (void *)wr_.in_prev, (void *)wr_.out_prev})
       if (q) (void)hipFree(q);
     wr_ = WalkRawScratch{};
   }
@@ -759,11 +781,27 @@ public:
                 int32_t *out_elem, int8_t *out_status,
                 const uint16_t *groups = nullptr,
                 const double *responses = nullptr,
-                double *out_dest = nullptr) override {
+                double *out_dest = nullptr, const double *in_t = nullptr,
+                const int32_t *in_prev = nullptr, double *out_o = nullptr,
+                double *out_t = nullptr,
+                int32_t *out_prev = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
     ensure_wr_cap(n, groups != nullptr, responses != nullptr);
+    const bool resume = in_t || in_prev || out_o || out_t || out_prev;
+    if (resume && !wr_.in_t) {
+      wr_.in_t = dmalloc<double>(wr_.cap);
+      wr_.in_prev = dmalloc<int32_t>(wr_.cap);
+      wr_.out_o = dmalloc<double>(wr_.cap * 3);
+      wr_.out_t = dmalloc<double>(wr_.cap);
+      wr_.out_prev = dmalloc<int32_t>(wr_.cap);
+    }
+    if (in_t)
+      PT_HIP_CHECK(hipMemcpy(wr_.in_t, in_t, n * 8, hipMemcpyHostToDevice));
+    if (in_prev)
+      PT_HIP_CHECK(hipMemcpy(wr_.in_prev, in_prev, n * 4,
+                             hipMemcpyHostToDevice));
     // Previous round's kernel has been synchronized below before this
     // call returns, so the scratch is free for reuse here.
     PT_HIP_CHECK(hipMemcpy(wr_.pos, pos, n * 3 * 8, hipMemcpyHostToDevice));
@@ -785,7 +823,9 @@ public:
           out_dest ? wr_.out_dest : nullptr, d_flux_, d_lost_,
           d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
-          d_pidx_, d_pelem_, d_pshift_);
+          d_pidx_, d_pelem_, d_pshift_, in_t ? wr_.in_t : nullptr,
+          in_prev ? wr_.in_prev : nullptr, out_o ? wr_.out_o : nullptr,
+          out_t ? wr_.out_t : nullptr, out_prev ? wr_.out_prev : nullptr);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, wr_.pos, wr_.dest, wr_.elem, wr_.w,
@@ -793,7 +833,9 @@ public:
           out_dest ? wr_.out_dest : nullptr, d_flux_, d_lost_,
           d_lostrec_, n,
           steps, reflective, d_face_bc_, ngroups, mesh_.nelems, nscores,
-          d_pidx_, d_pelem_, d_pshift_);
+          d_pidx_, d_pelem_, d_pshift_, in_t ? wr_.in_t : nullptr,
+          in_prev ? wr_.in_prev : nullptr, out_o ? wr_.out_o : nullptr,
+          out_t ? wr_.out_t : nullptr, out_prev ? wr_.out_prev : nullptr);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
     PT_HIP_CHECK(hipMemcpy(out_pos, wr_.out_pos, n * 3 * 8,
@@ -804,6 +846,15 @@ public:
     if (out_dest)
       PT_HIP_CHECK(hipMemcpy(out_dest, wr_.out_dest, n * 3 * 8,
                              hipMemcpyDeviceToHost));
+    if (out_o)
+      PT_HIP_CHECK(hipMemcpy(out_o, wr_.out_o, n * 3 * 8,
+                             hipMemcpyDeviceToHost));
+    if (out_t)
+      PT_HIP_CHECK(hipMemcpy(out_t, wr_.out_t, n * 8,
+                             hipMemcpyDeviceToHost));
+    if (out_prev)
+      PT_HIP_CHECK(hipMemcpy(out_prev, wr_.out_prev, n * 4,
+                             hipMemcpyDeviceToHost));
   }
 
   void walk_raw_device(int64_t n, const double *d_pos, const double *d_dest,
@@ -812,7 +863,11 @@ public:
                        int8_t *d_out_status,
                        const uint16_t *d_groups = nullptr,
                        const double *d_responses = nullptr,
-                       double *d_out_dest = nullptr) override {
+                       double *d_out_dest = nullptr,
+                       const double *d_in_t = nullptr,
+                       const int32_t *d_in_prev = nullptr,
+                       double *d_out_o = nullptr, double *d_out_t = nullptr,
+                       int32_t *d_out_prev = nullptr) override {
     if (n == 0) return;
     PT_HIP_CHECK(hipSetDevice(device_));
     const int steps = max_steps > 0 ? max_steps : default_max_steps(mesh_);
@@ -822,14 +877,14 @@ public:
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
           d_out_dest, d_flux_, d_lost_, d_lostrec_, n, steps, reflective,
           d_face_bc_, ngroups, mesh_.nelems, nscores, d_pidx_, d_pelem_,
-          d_pshift_);
+          d_pshift_, d_in_t, d_in_prev, d_out_o, d_out_t, d_out_prev);
     else
       k_walk_raw<false><<<grid_blocks(n), kBlock, 0, s_comp_>>>(
           d_planes_, d_planes32_, d_nbr_, d_pos, d_dest, d_elem, d_weights,
           d_groups, d_responses, d_out_pos, d_out_elem, d_out_status,
           d_out_dest, d_flux_, d_lost_, d_lostrec_, n, steps, reflective,
           d_face_bc_, ngroups, mesh_.nelems, nscores, d_pidx_, d_pelem_,
-          d_pshift_);
+          d_pshift_, d_in_t, d_in_prev, d_out_o, d_out_t, d_out_prev);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipStreamSynchronize(s_comp_));
   }

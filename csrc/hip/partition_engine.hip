@@ -142,7 +142,15 @@ __global__ void k_part_claim_rest(const unsigned long long *__restrict__ claim,
   }
 }
 
-// dep entry: 9 doubles [gid, px, py, pz, target_gid, owner, dx, dy, dz]
+// dep entry (rec_w doubles + owner): [gid, ox, oy, oz, target_gid,
+// dx, dy, dz, weight, t, prev_gid (, grp)(, resp x nscores)] [owner].
+// (ox,oy,oz) is the walk's CURRENT wrap-segment origin and t its
+// progress at the cut crossing; prev_gid is the global id of the
+// element exited from.  Carrying the t-parametrization (instead of the
+// crossing point) lets the receiving rank resume with the sender's
+// exact fp state, so partitioned flux attribution is elementwise
+// identical to the replicated engine (walk.h walk_segment doc).
+// Fresh entries (phase-A reroutes, host ejects) use t=0, prev=-1.
 // Operates on the gid range [g_lo, g_hi): step() pipelines chunks of the
 // batch so chunk c's prepare+walk overlap chunk c+1's H2D copies.  The
 // chunk's walk list lives in list[g_lo ...] with its own counter
@@ -151,7 +159,7 @@ __device__ __forceinline__ void dep_fill_tail(
     double *e, int rec_w, double weight, bool carry_grp, uint16_t grp,
     const double *resp, int nscores) {
   e[8] = weight;
-  int at = 9;
+  int at = 11; // [9]=t and [10]=prev_gid are written by the caller
   if (carry_grp) e[at++] = (double)grp;
   for (int k = 0; k < nscores; ++k) e[at + k] = resp ? resp[k] : 1.0;
   (void)rec_w;
@@ -201,6 +209,8 @@ __global__ void k_part_prepare(
             e[5] = dest[g * 3];
             e[6] = dest[g * 3 + 1];
             e[7] = dest[g * 3 + 2];
+            e[9] = 0.0;   // fresh walk: no resume state
+            e[10] = -1.0;
             dep_fill_tail(e, rec_w, w[g], carry_grp,
                           grp ? grp[g] : (uint16_t)0,
                           resp ? resp + (int64_t)g * nscores : nullptr,
@@ -239,7 +249,11 @@ __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
                               int32_t *__restrict__ welem,
                               double *__restrict__ ww,
                               uint16_t *__restrict__ wgrp,
-                              double *__restrict__ wresp) {
+                              double *__restrict__ wresp,
+                              const double *__restrict__ t0 = nullptr,
+                              const int32_t *__restrict__ prev = nullptr,
+                              double *__restrict__ wt0 = nullptr,
+                              int32_t *__restrict__ wprev = nullptr) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t j = blockIdx.x * blockDim.x + threadIdx.x; j < m; j += stride) {
     const int64_t g = list[j];
@@ -256,6 +270,8 @@ __global__ void k_part_gather(const int32_t *__restrict__ list, int64_t m,
     if (wresp)
       for (int k = 0; k < nscores; ++k)
         wresp[j * nscores + k] = resp[g * nscores + k];
+    if (wt0) wt0[j] = t0 ? t0[g] : 0.0;
+    if (wprev) wprev[j] = prev ? prev[g] : -1;
   }
 }
 
@@ -264,6 +280,10 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
                                const int32_t *__restrict__ wout_elem,
                                const int8_t *__restrict__ wstatus,
                                const double *__restrict__ wout_dest,
+                               const double *__restrict__ wout_o,
+                               const double *__restrict__ wout_t,
+                               const int32_t *__restrict__ wout_prev,
+                               const int32_t *__restrict__ l2g,
                                const double *__restrict__ ww,
                                const uint16_t *__restrict__ wgrp,
                                const double *__restrict__ wresp,
@@ -285,13 +305,16 @@ __global__ void k_part_collect(const int32_t *__restrict__ list, int64_t m,
       const unsigned long long d = atomicAdd(&ctr[1], 1ull);
       double *e = dep + d * (rec_w + 1);
       e[0] = (double)g;
-      e[1] = wout_pos[j * 3];
-      e[2] = wout_pos[j * 3 + 1];
-      e[3] = wout_pos[j * 3 + 2];
+      // resume state: wrap-segment origin, progress t, exited-from elem
+      e[1] = wout_o[j * 3];
+      e[2] = wout_o[j * 3 + 1];
+      e[3] = wout_o[j * 3 + 2];
       e[4] = (double)fgid[k];
       e[5] = wout_dest[j * 3];
       e[6] = wout_dest[j * 3 + 1];
       e[7] = wout_dest[j * 3 + 2];
+      e[9] = wout_t[j];
+      e[10] = wout_prev[j] >= 0 ? (double)l2g[wout_prev[j]] : -1.0;
       dep_fill_tail(e, rec_w, ww[j], carry_grp,
                     wgrp ? wgrp[j] : (uint16_t)0,
                     wresp ? wresp + (int64_t)j * nscores : nullptr, nscores);
@@ -339,6 +362,8 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
                               double *__restrict__ w_out,
                               uint16_t *__restrict__ grp_out,
                               double *__restrict__ resp_out,
+                              double *__restrict__ t0_out,
+                              int32_t *__restrict__ prev_out,
                               int32_t *__restrict__ list,
                               unsigned long long *__restrict__ ctr) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -357,7 +382,10 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
     // scatter the carried step inputs so later rounds (and the shared
     // gather-by-gid path) see them regardless of which mode sent them
     w_out[g] = e[8];
-    int at = 9;
+    t0_out[g] = e[9];
+    const int64_t pg = (int64_t)e[10];
+    prev_out[g] = pg >= 0 ? g2l[pg] : -1; // -1: outside submesh (no rings)
+    int at = 11;
     if (carry_grp && grp_out) grp_out[g] = (uint16_t)e[at];
     at += carry_grp ? 1 : 0;
     if (resp_out)
@@ -444,6 +472,8 @@ __global__ void k_part_prepare_local(
             e[5] = dest[j * 3];
             e[6] = dest[j * 3 + 1];
             e[7] = dest[j * 3 + 2];
+            e[9] = 0.0;   // fresh walk: no resume state
+            e[10] = -1.0;
             dep_fill_tail(e, rec_w, w[j], carry_grp,
                           grp ? grp[j] : (uint16_t)0,
                           resp ? resp + j * nscores : nullptr, nscores);
@@ -586,7 +616,7 @@ public:
 
     // record widths (see the layout comment at the top)
     carry_grp_ = ngroups_ > 1;
-    rec_w_ = 9 + (carry_grp_ ? 1 : 0) + nscores_;
+    rec_w_ = 11 + (carry_grp_ ? 1 : 0) + nscores_;
 
     // work buffers
     d_list_ = pdmalloc<int32_t>(n_);
@@ -606,6 +636,14 @@ public:
     d_wout_elem_ = pdmalloc<int32_t>(n_);
     d_wstatus_ = pdmalloc<int8_t>(n_);
     d_offs_ = pdmalloc<int64_t>(world_);
+    // bitwise handoff-resume state (record layout comment above)
+    d_t0_ = pdmalloc<double>(n_);
+    d_prev_ = pdmalloc<int32_t>(n_);
+    d_wt0_ = pdmalloc<double>(n_);
+    d_wprev_ = pdmalloc<int32_t>(n_);
+    d_wout_o_ = pdmalloc<double>(n_ * 3);
+    d_wout_t_ = pdmalloc<double>(n_);
+    d_wout_prev_ = pdmalloc<int32_t>(n_);
   }
 
   ~GpuPartitionedEngine() override {
@@ -625,7 +663,10 @@ public:
           (void *)d_offs_, (void *)d_send_, (void *)d_recv_,
           (void *)d_resp_, (void *)d_wresp_, (void *)d_frame_,
           (void *)d_ldest_, (void *)d_lw_, (void *)d_lorig_,
-          (void *)d_lresp_, (void *)d_lfly_, (void *)d_lgrp_})
+          (void *)d_lresp_, (void *)d_lfly_, (void *)d_lgrp_,
+          (void *)d_t0_, (void *)d_prev_, (void *)d_wt0_,
+          (void *)d_wprev_, (void *)d_wout_o_, (void *)d_wout_t_,
+          (void *)d_wout_prev_})
       if (p) (void)hipFree(p);
   }
 
@@ -750,10 +791,13 @@ public:
       eng_->walk_raw_device((int64_t)cw, d_wpos_, d_wdest_, d_welem_,
                             d_ww_, d_wout_pos_, d_wout_elem_, d_wstatus_,
                             groups ? d_wgrp_ : nullptr,
-                            responses ? d_wresp_ : nullptr, d_wout_dest_);
+                            responses ? d_wresp_ : nullptr, d_wout_dest_,
+                            nullptr, nullptr, d_wout_o_, d_wout_t_,
+                            d_wout_prev_);
       k_part_collect<<<pgrid_flat((int64_t)cw), kPBlock, 0, cs_>>>(
           d_list_ + lo, (int64_t)cw, d_wout_pos_, d_wout_elem_, d_wstatus_,
-          d_wout_dest_, d_ww_, groups ? d_wgrp_ : nullptr,
+          d_wout_dest_, d_wout_o_, d_wout_t_, d_wout_prev_, d_l2g_, d_ww_,
+          groups ? d_wgrp_ : nullptr,
           responses ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
           d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
           d_fowner_, d_dep_, d_ctr_);
@@ -787,15 +831,19 @@ public:
             d_list_, nwalk_r, d_pos_, d_elem_, d_dest_, d_dest_ovr_,
             /*use_ovr=*/true, d_w_, d_grp_, d_resp_,
             nscores_, d_wpos_, d_wdest_, d_welem_, d_ww_,
-            groups_used ? d_wgrp_ : nullptr, resp_used ? d_wresp_ : nullptr);
+            groups_used ? d_wgrp_ : nullptr, resp_used ? d_wresp_ : nullptr,
+            d_t0_, d_prev_, d_wt0_, d_wprev_);
         PT_HIP_CHECK(hipGetLastError());
         eng_->walk_raw_device(nwalk_r, d_wpos_, d_wdest_, d_welem_, d_ww_,
                               d_wout_pos_, d_wout_elem_, d_wstatus_,
                               groups_used ? d_wgrp_ : nullptr,
-                              resp_used ? d_wresp_ : nullptr, d_wout_dest_);
+                              resp_used ? d_wresp_ : nullptr, d_wout_dest_,
+                              d_wt0_, d_wprev_, d_wout_o_, d_wout_t_,
+                              d_wout_prev_);
         k_part_collect<<<pgrid_flat(nwalk_r), kPBlock, 0, cs_>>>(
             d_list_, nwalk_r, d_wout_pos_, d_wout_elem_, d_wstatus_,
-            d_wout_dest_, d_ww_, groups_used ? d_wgrp_ : nullptr,
+            d_wout_dest_, d_wout_o_, d_wout_t_, d_wout_prev_, d_l2g_, d_ww_,
+            groups_used ? d_wgrp_ : nullptr,
             resp_used ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
             d_pos_, d_elem_, d_esc_, d_res_, d_fgid_,
             d_fowner_, d_dep_, d_ctr_);
@@ -886,7 +934,7 @@ public:
         k_part_unpack<<<pgrid_flat(nrecv), kPBlock, 0, cs_>>>(
             recv_ptr, nrecv, rec_w_, carry_grp_, nscores_, d_g2l_, d_pos_,
             d_elem_, d_res_, d_esc_, d_dest_ovr_, d_w_, d_grp_, d_resp_,
-            d_list_, d_ctr_);
+            d_t0_, d_prev_, d_list_, d_ctr_);
         PT_HIP_CHECK(hipGetLastError());
       }
       PT_HIP_CHECK(hipStreamSynchronize(cs_));
@@ -997,10 +1045,13 @@ public:
       eng_->walk_raw_device((int64_t)cw, d_wpos_, d_wdest_, d_welem_, d_ww_,
                             d_wout_pos_, d_wout_elem_, d_wstatus_,
                             groups ? d_wgrp_ : nullptr,
-                            responses ? d_wresp_ : nullptr, d_wout_dest_);
+                            responses ? d_wresp_ : nullptr, d_wout_dest_,
+                            nullptr, nullptr, d_wout_o_, d_wout_t_,
+                            d_wout_prev_);
       k_part_collect<<<pgrid_flat((int64_t)cw), kPBlock, 0, cs_>>>(
           d_list_, (int64_t)cw, d_wout_pos_, d_wout_elem_, d_wstatus_,
-          d_wout_dest_, d_ww_, groups ? d_wgrp_ : nullptr,
+          d_wout_dest_, d_wout_o_, d_wout_t_, d_wout_prev_, d_l2g_, d_ww_,
+          groups ? d_wgrp_ : nullptr,
           responses ? d_wresp_ : nullptr, nscores_, carry_grp_, rec_w_,
           d_pos_, d_elem_, d_esc_, d_res_, d_fgid_, d_fowner_, d_dep_,
           d_ctr_);
@@ -1041,7 +1092,9 @@ public:
         e[6] = dest[j * 3 + 1];
         e[7] = dest[j * 3 + 2];
         e[8] = weights[j];
-        int at = 9;
+        e[9] = 0.0; // fresh walk: no resume state
+        e[10] = -1.0;
+        int at = 11;
         if (carry_grp_) e[at++] = groups ? (double)groups[j] : 0.0;
         for (int k = 0; k < nscores_; ++k)
           e[at + k] = responses ? responses[j * nscores_ + k] : 1.0;
@@ -1223,7 +1276,9 @@ private:
         e[6] = dest_host[g * 3 + 1];
         e[7] = dest_host[g * 3 + 2];
         e[8] = w_host[g];
-        int at = 9;
+        e[9] = 0.0; // fresh walk: no resume state
+        e[10] = -1.0;
+        int at = 11;
         if (carry_grp_) e[at++] = grp_host ? (double)grp_host[g] : 0.0;
         for (int k = 0; k < nscores_; ++k)
           e[at + k] = resp_host ? resp_host[g * nscores_ + k] : 1.0;
@@ -1313,6 +1368,11 @@ private:
   int64_t *d_offs_ = nullptr;
   double *d_send_ = nullptr, *d_recv_ = nullptr;
   int64_t cap_send_ = 0, cap_recv_ = 0;
+  // bitwise handoff-resume state (dep layout comment at the top)
+  double *d_t0_ = nullptr, *d_wt0_ = nullptr;
+  int32_t *d_prev_ = nullptr, *d_wprev_ = nullptr;
+  double *d_wout_o_ = nullptr, *d_wout_t_ = nullptr;
+  int32_t *d_wout_prev_ = nullptr;
 };
 
 // ---------------------------------------------------------------------------
@@ -1333,7 +1393,7 @@ public:
     eng_ = make_cpu_engine(dec_.sub.local, 1, ngroups_, nscores_);
     loc_tol_ = loc_tol_rel() * norm(full.bbox_hi - full.bbox_lo);
     carry_grp_ = ngroups_ > 1;
-    rec_w_ = 9 + (carry_grp_ ? 1 : 0) + nscores_;
+    rec_w_ = 11 + (carry_grp_ ? 1 : 0) + nscores_;
     pos_.assign(n_ * 3, 0.0);
     elem_.assign(n_, -1);
     res_.assign(n_, 0);
@@ -1390,6 +1450,10 @@ public:
     uint16_t grp;
     std::vector<double> resp; // nscores entries when used, else empty
     int32_t lelem;
+    // bitwise handoff resume (walk.h walk_segment doc): pos carries the
+    // wrap-segment origin for exchanged arrivals; t0/prev seed the walk
+    double t0 = 0.0;
+    int32_t prev = -1;
   };
 
   void step(const double *dest, const int8_t *flying, const double *weights,
@@ -1589,7 +1653,8 @@ private:
 
   void emit_dep(std::vector<double> &dep, int64_t g, Vec3 p, int64_t tgid,
                 Vec3 d, double w, uint16_t grp, const double *resp,
-                bool resp_used, int owner) {
+                bool resp_used, int owner, double t = 0.0,
+                int64_t prev_gid = -1) {
     const size_t base = dep.size();
     dep.resize(base + rec_w_ + 1, 0.0);
     double *e = dep.data() + base;
@@ -1602,7 +1667,9 @@ private:
     e[6] = d.y;
     e[7] = d.z;
     e[8] = w;
-    int at = 9;
+    e[9] = t;
+    e[10] = (double)prev_gid;
+    int at = 11;
     if (carry_grp_) e[at++] = (double)grp;
     for (int k = 0; k < nscores_; ++k)
       e[at + k] = (resp_used && resp) ? resp[k] : 1.0;
@@ -1622,6 +1689,8 @@ private:
         std::vector<int8_t> wstatus(m);
         std::vector<uint16_t> wgrp(groups_used ? m : 0);
         std::vector<double> wresp(resp_used ? m * nscores_ : 0);
+        std::vector<double> wt0(m), wout_o(m * 3), wout_t(m);
+        std::vector<int32_t> wprev(m), wout_prev(m);
         for (int64_t j = 0; j < m; ++j) {
           const Item &it = items[j];
           wpos[j * 3] = it.pos.x;
@@ -1632,6 +1701,8 @@ private:
           wdest[j * 3 + 2] = it.dest.z;
           welem[j] = it.lelem;
           ww[j] = it.w;
+          wt0[j] = it.t0;
+          wprev[j] = it.prev;
           if (groups_used) wgrp[j] = it.grp;
           if (resp_used)
             for (int k = 0; k < nscores_; ++k)
@@ -1641,20 +1712,23 @@ private:
                        wout_pos.data(), wout_elem.data(), wstatus.data(),
                        groups_used ? wgrp.data() : nullptr,
                        resp_used ? wresp.data() : nullptr,
-                       wout_dest.data());
+                       wout_dest.data(), wt0.data(), wprev.data(),
+                       wout_o.data(), wout_t.data(), wout_prev.data());
         for (int64_t j = 0; j < m; ++j) {
           const int64_t g = items[j].gid;
           if (wstatus[j] == 2) {
             const int32_t k = -(wout_elem[j] + 2);
             emit_dep(dep, g,
-                     Vec3{wout_pos[j * 3], wout_pos[j * 3 + 1],
-                          wout_pos[j * 3 + 2]},
+                     Vec3{wout_o[j * 3], wout_o[j * 3 + 1],
+                          wout_o[j * 3 + 2]},
                      dec_.sub.foreign_gid[k],
                      Vec3{wout_dest[j * 3], wout_dest[j * 3 + 1],
                           wout_dest[j * 3 + 2]},
                      ww[j], groups_used ? wgrp[j] : (uint16_t)0,
                      resp_used ? &wresp[j * nscores_] : nullptr, resp_used,
-                     dec_.sub.foreign_owner[k]);
+                     dec_.sub.foreign_owner[k], wout_t[j],
+                     wout_prev[j] >= 0 ? (int64_t)dec_.l2g32[wout_prev[j]]
+                                       : (int64_t)-1);
             res_[g] = 0;
           } else {
             pos_[g * 3] = wout_pos[j * 3];
@@ -1720,7 +1794,10 @@ private:
         it.pos = Vec3{e[1], e[2], e[3]};
         it.dest = Vec3{e[5], e[6], e[7]};
         it.w = e[8];
-        int at = 9;
+        it.t0 = e[9];
+        const int64_t pg = (int64_t)e[10];
+        it.prev = pg >= 0 ? dec_.g2l[pg] : -1;
+        int at = 11;
         it.grp = carry_grp_ ? (uint16_t)e[at++] : (uint16_t)0;
         if (resp_used) it.resp.assign(e + at, e + at + nscores_);
         it.lelem = elem_[g];

@@ -1,0 +1,130 @@
+"""World-2 stateful-partitioned soak: hundreds of cross-rank exchange
+rounds against a replicated oracle, on one shared GPU or on CPU.
+
+Spawns 2 ranks (the library's own TCP comm via RANK/WORLD_SIZE env; on a
+GPU both ranks share cuda:0 -- the same shape the world-2 GPU tests
+use).  Each step every particle gets a random destination, a slice is
+resampled to a new origin, and ~half the segments cross the Morton cut,
+so the phase-A claim, ghost-reroute, host-eject and unpack paths run
+continuously.  Rank 0 also drives a replicated TallyEngine fed the
+identical arrays and asserts elementwise agreement every 25 steps.
+
+Usage: python tools/part_world2_soak.py [--steps 200] [--particles 400000]
+       [--mesh-tets 100000] [--device auto]
+"""
+import argparse
+import os
+import subprocess
+import sys
+import tempfile
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+WORKER = r"""
+import os, sys, time
+import numpy as np
+sys.path.insert(0, os.environ["PT_ROOT"])
+import pumiumtally_amd as pt
+from pumiumtally_amd.mesh import box_mesh_with_tets
+
+dev = os.environ["PT_DEVICE"]
+steps = int(os.environ["PT_STEPS"])
+n = int(os.environ["PT_PARTICLES"])
+rank = int(os.environ["RANK"])
+
+mesh, cells = box_mesh_with_tets(int(os.environ["PT_TETS"]), extent=1.0)
+rng = np.random.default_rng(5)
+pos = rng.uniform(0.05, 0.95, size=(n, 3))
+
+pe = pt._core.PartitionedEngine(mesh, n, device=dev)
+pe.localize(pos.ravel())
+oracle = None
+if rank == 0:
+    oracle = pt.TallyEngine(mesh, n, device=dev)
+    oracle.copy_initial_position(pos.ravel())
+
+t0 = time.time()
+for s in range(steps):
+    dest = np.clip(pos + rng.normal(0, 0.2, size=(n, 3)), 0.001, 0.999)
+    fly = (rng.random(n) > 0.03).astype(np.int8)
+    w = rng.uniform(0.2, 1.0, n)
+    res = rng.random(n) < 0.15
+    origin = pos.copy()
+    origin[res] = rng.uniform(0.01, 0.99, size=(int(res.sum()), 3))
+    pe.step(dest.ravel(), fly, w, origin=origin.ravel())
+    if oracle is not None:
+        oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w)
+    pos = np.where(fly[:, None] == 1, dest, origin)
+    if (s + 1) % 25 == 0 or s + 1 == steps:
+        f1 = np.asarray(pe.flux_global())
+        if oracle is not None:
+            f2 = np.asarray(oracle.flux())
+            err = np.abs(f1 - f2).max() / max(f2.max(), 1e-30)
+            st = pe.stats()
+            print(f"step {s+1:4d}: rel err {err:.3e}, resident "
+                  f"{pe.resident}, lost {st['lost_particles']}", flush=True)
+            assert err < float(os.environ.get("PT_TOL", "1e-9")), "DIVERGED"
+if rank == 0:
+    print(f"PART_WORLD2_SOAK_OK: {steps} steps x {n} particles x 2 ranks "
+          f"in {time.time()-t0:.1f}s on {dev}", flush=True)
+"""
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--particles", type=int, default=400_000)
+    ap.add_argument("--mesh-tets", type=int, default=100_000)
+    ap.add_argument("--device", default="auto")
+    ap.add_argument("--tol", type=float, default=1e-9,
+                    help="oracle rel-err gate (handoff resume is bitwise, "
+                         "so 1e-12 holds; default leaves margin)")
+    args = ap.parse_args()
+
+    sys.path.insert(0, ROOT)
+    import pumiumtally_amd as pt
+    dev = args.device
+    if dev == "auto":
+        dev = "cuda:0" if pt.have_gpu() else "cpu"
+    if dev == "cpu":
+        args.particles = min(args.particles, 20_000)
+        args.steps = min(args.steps, 30)
+
+    with tempfile.TemporaryDirectory() as td:
+        script = os.path.join(td, "w.py")
+        with open(script, "w") as f:
+            f.write(WORKER)
+        env = dict(os.environ)
+        env.update({
+            "WORLD_SIZE": "2",
+            "MASTER_ADDR": "127.0.0.1",
+            "PUMITALLY_PORT": str(24000 + (os.getpid() + 17) % 15000),
+            "PUMITALLY_NO_TORCH": "1",
+            "PT_DEVICE": dev,
+            "PT_ROOT": ROOT,
+            "PT_STEPS": str(args.steps),
+            "PT_PARTICLES": str(args.particles),
+            "PT_TETS": str(args.mesh_tets),
+            "PT_TOL": repr(args.tol),
+        })
+        if dev != "cpu":
+            env["PUMITALLY_COMM"] = "tcp"  # two ranks share one device
+        procs = []
+        for r in range(2):
+            e = dict(env)
+            e["RANK"] = str(r)
+            e["LOCAL_RANK"] = "0"
+            procs.append(subprocess.Popen(
+                [sys.executable, script], env=e,
+                stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+        outs = [p.communicate(timeout=3000)[0].decode() for p in procs]
+        for r, (p, out) in enumerate(zip(procs, outs)):
+            if p.returncode != 0:
+                print(f"rank {r} FAILED:\n{out}")
+                sys.exit(1)
+        print(outs[0].rstrip())
+        assert "PART_WORLD2_SOAK_OK" in outs[0]
+
+
+if __name__ == "__main__":
+    main()
