@@ -107,8 +107,13 @@ __global__ void k_part_localize(const Plane *__restrict__ planes,
     const Vec3 q{origins[g * 3], origins[g * 3 + 1], origins[g * 3 + 2]};
     bool lo = false;
     const int32_t le = grid_locate(grid, planes, q, tol, &lo);
-    if (le >= 0 && lowner[le] == myrank) {
-      if (lo) atomicAdd(loose, 1ull);
+    // claim STRICT hits only: a loose (tol*1e4) hit on the SUBMESH can
+    // steal a point whose true element is absent from this rank's ghost
+    // ring (found at ~7e-7 depth by tools/part_world2_soak); such points
+    // are resolved on the host against the FULL mesh -- the same
+    // decision the replicated oracle makes -- so the starting element
+    // (and hence the first tally sliver) matches bitwise.
+    if (le >= 0 && !lo && lowner[le] == myrank) {
       res[g] = 1;
       esc[g] = 0;
       elem[g] = le;
@@ -189,8 +194,11 @@ __global__ void k_part_prepare(
       if (q.x != p.x || q.y != p.y || q.z != p.z) {
         atomicAdd(&ctr[3], 1ull);
         bool lo = false;
-        const int32_t le = grid_locate(grid, planes, q, tol, &lo);
-        if (lo) atomicAdd(&ctr[4], 1ull);
+        const int32_t le0 = grid_locate(grid, planes, q, tol, &lo);
+        // loose submesh hits are NOT trusted (see k_part_localize): the
+        // host resolves them on the full mesh, bitwise-matching the
+        // replicated oracle's relocation
+        const int32_t le = lo ? -1 : le0;
         if (le >= 0) {
           if (lowner[le] == myrank) {
             elem[g] = le;
@@ -397,6 +405,29 @@ __global__ void k_part_unpack(const double *__restrict__ recv, int64_t m,
 
 // apply host-resolved out-of-mesh relocations: particle stays resident
 // here with elem=-1 at its requested origin
+// apply host-resolved full-mesh claims from localize(): particle g
+// becomes resident here in local element lel[i] at its origin (already
+// on device in d_dest_)
+__global__ void k_part_apply_claims(const int32_t *__restrict__ gids,
+                                    const int32_t *__restrict__ lels,
+                                    int64_t m,
+                                    const double *__restrict__ orig,
+                                    double *__restrict__ pos,
+                                    int32_t *__restrict__ elem,
+                                    uint8_t *__restrict__ res,
+                                    uint8_t *__restrict__ esc) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < m; i += stride) {
+    const int64_t g = gids[i];
+    res[g] = 1;
+    esc[g] = 0;
+    elem[g] = lels[i];
+    pos[g * 3] = orig[g * 3];
+    pos[g * 3 + 1] = orig[g * 3 + 1];
+    pos[g * 3 + 2] = orig[g * 3 + 2];
+  }
+}
+
 __global__ void k_part_apply_outside(const int32_t *__restrict__ gids,
                                      int64_t m,
                                      const double *__restrict__ orig,
@@ -452,8 +483,8 @@ __global__ void k_part_prepare_local(
       if (q.x != p.x || q.y != p.y || q.z != p.z) {
         atomicAdd(&ctr[3], 1ull);
         bool lo = false;
-        const int32_t le = grid_locate(grid, planes, q, tol, &lo);
-        if (lo) atomicAdd(&ctr[4], 1ull);
+        const int32_t le0 = grid_locate(grid, planes, q, tol, &lo);
+        const int32_t le = lo ? -1 : le0; // see k_part_prepare
         if (le >= 0) {
           if (lowner[le] == myrank) {
             elem[g] = le;
@@ -689,16 +720,51 @@ public:
         d_pos_, d_elem_, d_res_, d_esc_, d_claim, &d_ctr_[4]);
     PT_HIP_CHECK(hipGetLastError());
     PT_HIP_CHECK(hipDeviceSynchronize());
-    if (world_ > 1) {
-      // claims are disjoint (only the owner claims), so bitwise OR of the
-      // claim masks == integer sum of the words
-      std::vector<int64_t> words(nwords);
-      PT_HIP_CHECK(hipMemcpy(words.data(), d_claim, nwords * 8,
-                             hipMemcpyDeviceToHost));
-      comm_->allreduce_sum(words.data(), nwords);
-      PT_HIP_CHECK(hipMemcpy(d_claim, words.data(), nwords * 8,
-                             hipMemcpyHostToDevice));
+    // claims are disjoint (only the owner claims), so bitwise OR of the
+    // claim masks == integer sum of the words
+    std::vector<int64_t> words(nwords);
+    PT_HIP_CHECK(hipMemcpy(words.data(), d_claim, nwords * 8,
+                           hipMemcpyDeviceToHost));
+    if (world_ > 1) comm_->allreduce_sum(words.data(), nwords);
+    // Unclaimed = outside the mesh OR only loosely localizable on a
+    // submesh (true element absent from the ghost ring, or the point is
+    // within tolerance of a cut face).  Resolve against the FULL mesh on
+    // the host -- every rank computes the same answer deterministically,
+    // so the owner claims without extra communication, and the chosen
+    // element matches the replicated oracle's bitwise.
+    if (full_locate_) {
+      std::vector<int32_t> cg, cl;
+      for (int64_t g = 0; g < n_; ++g) {
+        if ((words[g >> 6] >> (g & 63)) & 1) continue;
+        const Vec3 q{origins[g * 3], origins[g * 3 + 1], origins[g * 3 + 2]};
+        bool lo = false;
+        const int32_t ge = full_locate_(q, loc_tol_, &lo);
+        if (ge < 0) continue; // truly outside: rank 0 parks it below
+        if (lo) stats_.loose_localizations++;
+        words[g >> 6] |= (int64_t)(1ull << (g & 63));
+        if (dec_.owners[ge] == rank_) {
+          cg.push_back((int32_t)g);
+          cl.push_back(dec_.g2l[ge]);
+        }
+      }
+      if (!cg.empty()) {
+        int32_t *d_cg = pdmalloc<int32_t>((int64_t)cg.size());
+        int32_t *d_cl = pdmalloc<int32_t>((int64_t)cl.size());
+        PT_HIP_CHECK(hipMemcpy(d_cg, cg.data(), cg.size() * 4,
+                               hipMemcpyHostToDevice));
+        PT_HIP_CHECK(hipMemcpy(d_cl, cl.data(), cl.size() * 4,
+                               hipMemcpyHostToDevice));
+        k_part_apply_claims<<<pgrid((int64_t)cg.size()), kPBlock>>>(
+            d_cg, d_cl, (int64_t)cg.size(), d_dest_, d_pos_, d_elem_,
+            d_res_, d_esc_);
+        PT_HIP_CHECK(hipGetLastError());
+        PT_HIP_CHECK(hipDeviceSynchronize());
+        PT_HIP_CHECK(hipFree(d_cg));
+        PT_HIP_CHECK(hipFree(d_cl));
+      }
     }
+    PT_HIP_CHECK(hipMemcpy(d_claim, words.data(), nwords * 8,
+                           hipMemcpyHostToDevice));
     if (rank_ == 0) {
       k_part_claim_rest<<<pgrid_flat(n_), kPBlock>>>(d_claim, d_dest_, n_, d_pos_,
                                                 d_elem_, d_res_, d_esc_);
@@ -1413,8 +1479,10 @@ public:
       const Vec3 q{origins[g * 3], origins[g * 3 + 1], origins[g * 3 + 2]};
       bool lo = false;
       const int32_t le = lm.locate(q, loc_tol_, &lo);
-      if (le >= 0 && dec_.lowner[le] == rank_) {
-        if (lo) stats_.loose_localizations++;
+      // strict hits only -- loose submesh hits can steal points whose
+      // true element is absent from the ghost ring (GPU k_part_localize
+      // comment); the full-mesh pass below resolves them
+      if (le >= 0 && !lo && dec_.lowner[le] == rank_) {
         res_[g] = 1;
         esc_[g] = 0;
         elem_[g] = le;
@@ -1426,6 +1494,25 @@ public:
     }
     if (world_ > 1)
       comm_->allreduce_sum(claim.data(), (int64_t)claim.size());
+    if (full_locate_) {
+      for (int64_t g = 0; g < n_; ++g) {
+        if ((claim[g >> 6] >> (g & 63)) & 1) continue;
+        const Vec3 q{origins[g * 3], origins[g * 3 + 1], origins[g * 3 + 2]};
+        bool lo = false;
+        const int32_t ge = full_locate_(q, loc_tol_, &lo);
+        if (ge < 0) continue;
+        if (lo) stats_.loose_localizations++;
+        claim[g >> 6] |= (int64_t)(1ull << (g & 63));
+        if (dec_.owners[ge] == rank_) {
+          res_[g] = 1;
+          esc_[g] = 0;
+          elem_[g] = dec_.g2l[ge];
+          pos_[g * 3] = q.x;
+          pos_[g * 3 + 1] = q.y;
+          pos_[g * 3 + 2] = q.z;
+        }
+      }
+    }
     if (rank_ == 0) {
       for (int64_t g = 0; g < n_; ++g)
         if (!((claim[g >> 6] >> (g & 63)) & 1)) {
@@ -1604,7 +1691,7 @@ private:
         stats_.relocated++;
         bool lo = false;
         int32_t le = lm.locate(q, loc_tol_, &lo);
-        if (lo) stats_.loose_localizations++;
+        if (lo) le = -1; // loose submesh hit: resolve on the full mesh
         int64_t tgid = -1;
         int towner = -1;
         if (le >= 0 && dec_.lowner[le] != rank_) {
