@@ -31,3 +31,19 @@ def test_world2_handoff_is_bitwise():
         capture_output=True, text=True, timeout=600)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "PART_WORLD2_SOAK_OK" in r.stdout
+
+
+def test_world2_localization_matches_full_mesh_at_scale():
+    """Second divergence source (2.1e-03 after 25 steps at 400k
+    particles): a resampled origin ~7e-7 inside an element absent from
+    the rank's ghost ring was loose-claimed into the adjacent ghost
+    element, starting the walk one element off the oracle.  Submesh
+    localization now claims strict hits only; loose hits resolve on the
+    full mesh.  Replays the exact failing scale at 1e-12."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "part_world2_soak.py"),
+         "--steps", "25", "--particles", "400000", "--mesh-tets", "100000",
+         "--device", "cpu", "--full-size", "--tol", "1e-12"],
+        capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "PART_WORLD2_SOAK_OK" in r.stdout

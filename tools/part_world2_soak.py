@@ -76,6 +76,9 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--full-size", action="store_true",
+                    help="run the requested size even on CPU (default "
+                         "clamps to 20k particles / 30 steps)")
     ap.add_argument("--tol", type=float, default=1e-9,
                     help="oracle rel-err gate (handoff resume is bitwise, "
                          "so 1e-12 holds; default leaves margin)")
@@ -86,7 +89,7 @@ def main():
     dev = args.device
     if dev == "auto":
         dev = "cuda:0" if pt.have_gpu() else "cpu"
-    if dev == "cpu":
+    if dev == "cpu" and not args.full_size:
         args.particles = min(args.particles, 20_000)
         args.steps = min(args.steps, 30)
 
