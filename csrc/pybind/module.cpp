@@ -774,7 +774,9 @@ PYBIND11_MODULE(_core, m) {
               py::array_t<double, py::array::c_style | py::array::forcecast> dest,
               py::array_t<int32_t, py::array::c_style | py::array::forcecast> elem,
               py::array_t<double, py::array::c_style | py::array::forcecast> weights,
-              py::object groups_obj, py::object responses_obj) {
+              py::object groups_obj, py::object responses_obj,
+              py::object in_t_obj, py::object in_prev_obj,
+              bool resume) -> py::tuple {
              const int64_t n = (int64_t)elem.size();
              if ((int64_t)pos.size() != n * 3 || (int64_t)dest.size() != n * 3 ||
                  (int64_t)weights.size() != n)
@@ -797,44 +799,83 @@ PYBIND11_MODULE(_core, m) {
                  throw std::runtime_error("walk_raw: responses size must be n*nscores");
                rp = resp.data();
              }
+             py::array_t<double, py::array::c_style | py::array::forcecast> in_t;
+             const double *tp = nullptr;
+             if (!in_t_obj.is_none()) {
+               in_t = in_t_obj.cast<
+                   py::array_t<double, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)in_t.size() != n)
+                 throw std::runtime_error("walk_raw: in_t size mismatch");
+               tp = in_t.data();
+             }
+             py::array_t<int32_t, py::array::c_style | py::array::forcecast> in_prev;
+             const int32_t *pp = nullptr;
+             if (!in_prev_obj.is_none()) {
+               in_prev = in_prev_obj.cast<
+                   py::array_t<int32_t, py::array::c_style | py::array::forcecast>>();
+               if ((int64_t)in_prev.size() != n)
+                 throw std::runtime_error("walk_raw: in_prev size mismatch");
+               pp = in_prev.data();
+             }
              auto out_pos = py::array_t<double>({n, (int64_t)3});
              auto out_elem = py::array_t<int32_t>(n);
              auto out_status = py::array_t<int8_t>(n);
              auto out_dest = py::array_t<double>({n, (int64_t)3});
+             const bool want_res = resume || tp || pp;
+             auto out_o = py::array_t<double>({want_res ? n : 0, (int64_t)3});
+             auto out_t = py::array_t<double>(want_res ? n : 0);
+             auto out_prev = py::array_t<int32_t>(want_res ? n : 0);
              {
                py::gil_scoped_release nogil;
                e.eng->walk_raw(n, pos.data(), dest.data(), elem.data(),
                                weights.data(), out_pos.mutable_data(),
                                out_elem.mutable_data(), out_status.mutable_data(),
-                               gp, rp, out_dest.mutable_data());
+                               gp, rp, out_dest.mutable_data(), tp, pp,
+                               want_res ? out_o.mutable_data() : nullptr,
+                               want_res ? out_t.mutable_data() : nullptr,
+                               want_res ? out_prev.mutable_data() : nullptr);
              }
              // out_dest: the walk's final destination (mutated by
              // reflective/periodic restarts); handoffs must resume
-             // toward it, not the original dest
+             // toward it, not the original dest.  With resume=True the
+             // tuple additionally carries (out_o, out_t, out_prev): the
+             // bitwise handoff-resume state (walk.h walk_segment doc);
+             // feed them back via in_t/in_prev (with pos = out_o row)
+             // so the receiving rank replays the walk's fp decisions.
+             if (want_res)
+               return py::make_tuple(out_pos, out_elem, out_status, out_dest,
+                                     out_o, out_t, out_prev);
              return py::make_tuple(out_pos, out_elem, out_status, out_dest);
            },
            py::arg("pos"), py::arg("dest"), py::arg("elem"),
            py::arg("weights"), py::arg("groups") = py::none(),
-           py::arg("responses") = py::none())
+           py::arg("responses") = py::none(), py::arg("in_t") = py::none(),
+           py::arg("in_prev") = py::none(), py::arg("resume") = false)
       .def("walk_raw_device",
            // Raw device-pointer variant for the device-resident partitioned
            // round loop (pointers as from torch.Tensor.data_ptr()).
            [](PyEngine &e, int64_t n, uintptr_t pos, uintptr_t dest,
               uintptr_t elem, uintptr_t weights, uintptr_t out_pos,
               uintptr_t out_elem, uintptr_t out_status, uintptr_t groups,
-              uintptr_t responses, uintptr_t out_dest) {
+              uintptr_t responses, uintptr_t out_dest, uintptr_t in_t,
+              uintptr_t in_prev, uintptr_t out_o, uintptr_t out_t,
+              uintptr_t out_prev) {
              py::gil_scoped_release nogil;
              e.eng->walk_raw_device(
                  n, (const double *)pos, (const double *)dest,
                  (const int32_t *)elem, (const double *)weights,
                  (double *)out_pos, (int32_t *)out_elem, (int8_t *)out_status,
                  (const uint16_t *)groups, (const double *)responses,
-                 (double *)out_dest);
+                 (double *)out_dest, (const double *)in_t,
+                 (const int32_t *)in_prev, (double *)out_o, (double *)out_t,
+                 (int32_t *)out_prev);
            },
            py::arg("n"), py::arg("pos"), py::arg("dest"), py::arg("elem"),
            py::arg("weights"), py::arg("out_pos"), py::arg("out_elem"),
            py::arg("out_status"), py::arg("groups") = 0,
-           py::arg("responses") = 0, py::arg("out_dest") = 0)
+           py::arg("responses") = 0, py::arg("out_dest") = 0,
+           py::arg("in_t") = 0, py::arg("in_prev") = 0, py::arg("out_o") = 0,
+           py::arg("out_t") = 0, py::arg("out_prev") = 0)
       .def("synchronize", [](PyEngine &e) { py::gil_scoped_release nogil; e.eng->synchronize(); })
       .def("flux", [](const PyEngine &e) { return vec_to_np(e.eng->flux()); })
       .def("elem_ids",

@@ -158,7 +158,7 @@ class TallyEngine:
             ptr(groups, "<u2", n), ptr(responses, "<f8", n * self.nscores))
 
     def walk_raw(self, pos, dest, elem, weights, groups=None,
-                 responses=None):
+                 responses=None, in_t=None, in_prev=None, resume=False):
         """Batched raw segment walk (domain-decomposition support): returns
         (out_pos, out_elem, status, out_dest) with status 0=done 1=escaped
         2=handoff 3=lost; tallies into this engine's flux.  out_dest is the
@@ -166,9 +166,13 @@ class TallyEngine:
         and a handoff must resume toward out_dest, not the original dest.
         groups: optional uint16 per-segment energy-group indices (flux row
         group*nelems+elem); responses: optional n x nscores score
-        multipliers."""
+        multipliers.  With resume=True (or in_t/in_prev given) the tuple
+        additionally carries (out_o, out_t, out_prev) -- the bitwise
+        handoff-resume state (csrc/core/walk.h): ship them in the handoff
+        record (pos column = out_o row) and seed the receiving walk via
+        in_t/in_prev so it replays the sender's fp decisions exactly."""
         return self._eng.walk_raw(pos, dest, elem, weights, groups,
-                                  responses)
+                                  responses, in_t, in_prev, resume)
 
     def synchronize(self):
         self._eng.synchronize()

@@ -17,9 +17,15 @@ mesh on every rank, owners all rank 0) -- with a real decomposition.
 Exchange collective: torch.distributed all_to_all_single over RCCL/xGMI
 when on GPU (nccl backend), all_gather_object on gloo (CPU tests).
 
-Exchange record (9+nscores float64 per handed-off particle when
-responses are used, else 9): position[3], destination[3], weight,
-target global element id, energy group[, response multipliers].
+Exchange record (11+nscores float64 per handed-off particle when
+responses are used, else 11): resume origin[3], destination[3], weight,
+target global element id, energy group, walk progress t, previous
+global element id[, response multipliers].  The (origin, t, prev)
+triple is the bitwise handoff-resume state (csrc/core/walk.h
+walk_segment doc): the receiving rank seeds its walk with the sender's
+t-parametrization and replays the remaining crossings with identical
+fp decisions, so partitioned flux attribution matches a single-mesh
+walk elementwise.
 
 Load balance: pass elem_weights to the constructor (per-element work
 estimates -- e.g. a previous batch's raw flux) to split the Morton curve
@@ -34,7 +40,7 @@ import numpy as np
 
 from .dist import init_distributed
 
-_BASE_REC = 9  # floats per exchange record (before response columns)
+_BASE_REC = 11  # floats per exchange record (before response columns)
 
 
 class PartitionedTally:
@@ -150,16 +156,20 @@ class PartitionedTally:
         grp = groups[mine] if groups is not None else None
         rsp = responses[mine] if responses is not None else None
         elem = self.g2l[gids[mine]].astype(np.int32)
+        in_t = in_prev = None  # first round: fresh walks
 
         if self._use_device_rounds():
             return self._run_rounds_device(pos, dst, wgt, grp, rsp, elem,
                                            rec_w)
 
+        l2g_arr = np.asarray(self.l2g, np.int64)
         for _round in range(self.max_rounds):
             outbound = [np.zeros((0, rec_w)) for _ in range(self.world)]
             if len(elem):
-                out_pos, out_elem, status, out_dest = self.engine.walk_raw(
-                    pos.ravel(), dst.ravel(), elem, wgt, grp, rsp)
+                (out_pos, out_elem, status, out_dest, out_o, out_t,
+                 out_prev) = self.engine.walk_raw(
+                    pos.ravel(), dst.ravel(), elem, wgt, grp, rsp,
+                    in_t=in_t, in_prev=in_prev, resume=True)
                 hand = status == 2
                 if hand.any():
                     k = -(out_elem[hand].astype(np.int64) + 2)
@@ -168,11 +178,17 @@ class PartitionedTally:
                     shift = self.foreign_shift[k]
                     g_col = (grp[hand] if grp is not None
                              else np.zeros(int(hand.sum()), np.uint16))
-                    # out_dest, not dst: reflective/periodic restarts
-                    # inside the walk mutate the destination
-                    cols = [out_pos[hand] + shift, out_dest[hand] + shift,
+                    ph = out_prev[hand].astype(np.int64)
+                    prev_gid = np.where(ph >= 0, l2g_arr[np.maximum(ph, 0)],
+                                        -1).astype(np.float64)
+                    # resume state rides the record: wrap-segment origin
+                    # (out_o) + progress t + exited-from element; out_dest,
+                    # not dst, because in-walk reflective/periodic restarts
+                    # mutate the destination
+                    cols = [out_o[hand] + shift, out_dest[hand] + shift,
                             wgt[hand, None], tgt_gid[:, None].astype(np.float64),
-                            g_col[:, None].astype(np.float64)]
+                            g_col[:, None].astype(np.float64),
+                            out_t[hand, None], prev_gid[:, None]]
                     if rsp is not None:
                         cols.append(rsp[hand])
                     rec = np.concatenate(cols, axis=1)
@@ -202,7 +218,11 @@ class PartitionedTally:
             elem = self.g2l[inbound[:, 7].astype(np.int64)].astype(np.int32)
             grp = (inbound[:, 8].astype(np.uint16)
                    if groups is not None else None)
-            rsp = (np.ascontiguousarray(inbound[:, 9:9 + self.nscores])
+            in_t = np.ascontiguousarray(inbound[:, 9])
+            pg = inbound[:, 10].astype(np.int64)
+            in_prev = np.where(pg >= 0, self.g2l[np.maximum(pg, 0)],
+                               -1).astype(np.int32)
+            rsp = (np.ascontiguousarray(inbound[:, 11:11 + self.nscores])
                    if responses is not None else None)
         else:
             raise RuntimeError("partitioned walk did not converge "
@@ -242,8 +262,10 @@ class PartitionedTally:
                 to(np.asarray(self.foreign_gid, np.int64)),
                 to(np.asarray(self.foreign_owner, np.int64)),
                 to(self.foreign_shift),
+                to(np.asarray(self.l2g, np.int64)),
             )
-        g2l_t, fg_t, fo_t, fs_t = self._dev_tables
+        g2l_t, fg_t, fo_t, fs_t, l2g_t = self._dev_tables
+        in_t_t = in_prev_t = None  # first round: fresh walks
 
         for _round in range(self.max_rounds):
             k = int(elem_t.numel())
@@ -253,6 +275,9 @@ class PartitionedTally:
                 status = torch.empty(k, dtype=torch.int8, device=dev)
                 out_dest = torch.empty((k, 3), dtype=torch.float64,
                                        device=dev)
+                out_o = torch.empty((k, 3), dtype=torch.float64, device=dev)
+                out_t = torch.empty(k, dtype=torch.float64, device=dev)
+                out_prev = torch.empty(k, dtype=torch.int32, device=dev)
                 torch.cuda.synchronize()  # inputs/outputs materialized
                 self.engine._eng.walk_raw_device(
                     k, pos_t.data_ptr(), dst_t.data_ptr(), elem_t.data_ptr(),
@@ -260,7 +285,10 @@ class PartitionedTally:
                     status.data_ptr(),
                     grp_t.data_ptr() if grp_t is not None else 0,
                     rsp_t.data_ptr() if rsp_t is not None else 0,
-                    out_dest.data_ptr())
+                    out_dest.data_ptr(),
+                    in_t_t.data_ptr() if in_t_t is not None else 0,
+                    in_prev_t.data_ptr() if in_prev_t is not None else 0,
+                    out_o.data_ptr(), out_t.data_ptr(), out_prev.data_ptr())
                 hand = status == 2
                 nh = int(hand.sum())
             else:
@@ -272,12 +300,16 @@ class PartitionedTally:
                 order = torch.argsort(tgt_owner)
                 # out_dest, not dst: in-walk reflective/periodic restarts
                 # mutate the destination
-                cols = [out_pos[hand] + shift, out_dest[hand] + shift,
+                ph = out_prev[hand].long()
+                prev_gid = torch.where(ph >= 0, l2g_t[ph.clamp(min=0)],
+                                       torch.full_like(ph, -1)).double()
+                cols = [out_o[hand] + shift, out_dest[hand] + shift,
                         wgt_t[hand, None],
                         fg_t[idx][:, None].double(),
                         (grp_t[hand][:, None].double() if grp_t is not None
                          else torch.zeros((nh, 1), dtype=torch.float64,
-                                          device=dev))]
+                                          device=dev)),
+                        out_t[hand, None], prev_gid[:, None]]
                 if rsp_t is not None:
                     cols.append(rsp_t[hand])
                 rec = torch.cat(cols, dim=1)[order].contiguous()
@@ -312,8 +344,12 @@ class PartitionedTally:
             elem_t = g2l_t[inbound[:, 7].long()].int().contiguous()
             if grp_t is not None:
                 grp_t = inbound[:, 8].to(torch.int16).contiguous()
+            in_t_t = inbound[:, 9].contiguous()
+            pg = inbound[:, 10].long()
+            in_prev_t = torch.where(pg >= 0, g2l_t[pg.clamp(min=0)],
+                                    torch.full_like(pg, -1)).int().contiguous()
             if rsp_t is not None:
-                rsp_t = inbound[:, 9:9 + self.nscores].contiguous()
+                rsp_t = inbound[:, 11:11 + self.nscores].contiguous()
         else:
             raise RuntimeError("partitioned walk did not converge "
                                f"in {self.max_rounds} handoff rounds")
