@@ -33,6 +33,9 @@ n = int(os.environ["PT_PARTICLES"])
 rank = int(os.environ["RANK"])
 
 mesh, cells = box_mesh_with_tets(int(os.environ["PT_TETS"]), extent=1.0)
+if os.environ.get("PT_REFLECTIVE") == "1":
+    fid, _, _ = mesh.boundary_faces()
+    mesh.set_reflective_faces(fid)
 rng = np.random.default_rng(5)
 pos = rng.uniform(0.05, 0.95, size=(n, 3))
 
@@ -46,6 +49,12 @@ if rank == 0:
 t0 = time.time()
 for s in range(steps):
     dest = np.clip(pos + rng.normal(0, 0.2, size=(n, 3)), 0.001, 0.999)
+    esc_f = float(os.environ.get("PT_ESCAPE_FRAC", "0"))
+    if esc_f > 0.0:
+        # push a slice of destinations outside the box: their walks clip
+        # at the vacuum boundary (escape) -- exercised against the cut
+        sel = rng.random(n) < esc_f
+        dest[sel] = pos[sel] + rng.normal(0, 0.6, size=(int(sel.sum()), 3))
     fly = (rng.random(n) > 0.03).astype(np.int8)
     w = rng.uniform(0.2, 1.0, n)
     res = rng.random(n) < 0.15
@@ -54,7 +63,7 @@ for s in range(steps):
     pe.step(dest.ravel(), fly, w, origin=origin.ravel())
     if oracle is not None:
         oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w)
-    pos = np.where(fly[:, None] == 1, dest, origin)
+    pos = np.where(fly[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
     if (s + 1) % 25 == 0 or s + 1 == steps:
         f1 = np.asarray(pe.flux_global())
         if oracle is not None:
@@ -76,6 +85,13 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--reflective", action="store_true",
+                    help="mark every boundary face reflective (in-walk "
+                         "restarts re-base the segment; stresses the "
+                         "resume origin against the cut)")
+    ap.add_argument("--escape-frac", type=float, default=0.0,
+                    help="fraction of destinations pushed outside the box "
+                         "(vacuum-escape stress against the cut)")
     ap.add_argument("--full-size", action="store_true",
                     help="run the requested size even on CPU (default "
                          "clamps to 20k particles / 30 steps)")
@@ -109,6 +125,8 @@ def main():
             "PT_PARTICLES": str(args.particles),
             "PT_TETS": str(args.mesh_tets),
             "PT_TOL": repr(args.tol),
+            "PT_ESCAPE_FRAC": repr(args.escape_frac),
+            "PT_REFLECTIVE": "1" if args.reflective else "0",
         })
         if dev != "cpu":
             env["PUMITALLY_COMM"] = "tcp"  # two ranks share one device
