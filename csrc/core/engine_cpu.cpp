@@ -102,8 +102,12 @@ public:
     const int nthreads =
         (n >= 65536 && hw > 1) ? (int)std::min<unsigned>(hw, 64) : 1;
     if (nthreads > 1) {
+      // full tally shape INCLUDING the score dimension: the scored add
+      // writes k*ngroups*nelems + group*nelems + elem for k < nscores
+      // (an nelems*ngroups-sized partial overflows the heap -- found by
+      // tools/part_world2_soak at 400k particles with nscores=2)
       std::vector<std::vector<double>> partial(
-          nthreads, std::vector<double>(mesh_.nelems * ngroups, 0.0));
+          nthreads, std::vector<double>(flux_.size(), 0.0));
       std::atomic<int64_t> lost{0}, reloc{0};
       std::vector<std::thread> workers;
       const int64_t per = (n + nthreads - 1) / nthreads;
@@ -120,7 +124,7 @@ public:
       }
       for (auto &w : workers) w.join();
       for (int t = 0; t < nthreads; ++t)
-        for (int64_t e = 0; e < mesh_.nelems * ngroups; ++e)
+        for (size_t e = 0; e < flux_.size(); ++e)
           flux_[e] += partial[t][e];
       stats_.lost_particles += lost.load();
       stats_.relocated += reloc.load();

@@ -39,12 +39,16 @@ if os.environ.get("PT_REFLECTIVE") == "1":
 rng = np.random.default_rng(5)
 pos = rng.uniform(0.05, 0.95, size=(n, 3))
 
-pe = pt._core.PartitionedEngine(mesh, n, device=dev)
+G = int(os.environ.get("PT_NGROUPS", "1"))
+S = int(os.environ.get("PT_NSCORES", "1"))
+pe = pt._core.PartitionedEngine(mesh, n, device=dev, ngroups=G, nscores=S)
 pe.localize(pos.ravel())
 oracle = None
 if rank == 0:
-    oracle = pt.TallyEngine(mesh, n, device=dev)
+    oracle = pt.TallyEngine(mesh, n, device=dev, ngroups=G, nscores=S)
     oracle.copy_initial_position(pos.ravel())
+grp_all = rng.integers(0, G, n).astype(np.uint16) if G > 1 else None
+rsp_all = rng.uniform(0.5, 2.0, (n, S)) if S > 1 else None
 
 t0 = time.time()
 for s in range(steps):
@@ -60,14 +64,16 @@ for s in range(steps):
     res = rng.random(n) < 0.15
     origin = pos.copy()
     origin[res] = rng.uniform(0.01, 0.99, size=(int(res.sum()), 3))
-    pe.step(dest.ravel(), fly, w, origin=origin.ravel())
+    pe.step(dest.ravel(), fly, w, origin=origin.ravel(), groups=grp_all,
+            responses=rsp_all)
     if oracle is not None:
-        oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w)
+        oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w,
+                    groups=grp_all, responses=rsp_all)
     pos = np.where(fly[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
     if (s + 1) % 25 == 0 or s + 1 == steps:
-        f1 = np.asarray(pe.flux_global())
+        f1 = np.asarray(pe.flux_global()).ravel()
         if oracle is not None:
-            f2 = np.asarray(oracle.flux())
+            f2 = np.asarray(oracle.flux()).ravel()
             err = np.abs(f1 - f2).max() / max(f2.max(), 1e-30)
             st = pe.stats()
             print(f"step {s+1:4d}: rel err {err:.3e}, resident "
@@ -85,6 +91,8 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--ngroups", type=int, default=1)
+    ap.add_argument("--nscores", type=int, default=1)
     ap.add_argument("--reflective", action="store_true",
                     help="mark every boundary face reflective (in-walk "
                          "restarts re-base the segment; stresses the "
@@ -127,6 +135,8 @@ def main():
             "PT_TOL": repr(args.tol),
             "PT_ESCAPE_FRAC": repr(args.escape_frac),
             "PT_REFLECTIVE": "1" if args.reflective else "0",
+            "PT_NGROUPS": str(args.ngroups),
+            "PT_NSCORES": str(args.nscores),
         })
         if dev != "cpu":
             env["PUMITALLY_COMM"] = "tcp"  # two ranks share one device
