@@ -64,8 +64,17 @@ for s in range(steps):
     res = rng.random(n) < 0.15
     origin = pos.copy()
     origin[res] = rng.uniform(0.01, 0.99, size=(int(res.sum()), 3))
-    pe.step(dest.ravel(), fly, w, origin=origin.ravel(), groups=grp_all,
-            responses=rsp_all)
+    if os.environ.get("PT_LOCAL") == "1":
+        # coupled-host form: per-resident frame-ordered inputs only
+        frame = np.asarray(pe.resident_list(), np.int64)
+        pe.step_local(dest[frame].ravel(), fly[frame], w[frame],
+                      origin=origin[frame].ravel(),
+                      groups=grp_all[frame] if grp_all is not None else None,
+                      responses=rsp_all[frame] if rsp_all is not None
+                      else None)
+    else:
+        pe.step(dest.ravel(), fly, w, origin=origin.ravel(), groups=grp_all,
+                responses=rsp_all)
     if oracle is not None:
         oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w,
                     groups=grp_all, responses=rsp_all)
@@ -91,6 +100,9 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--local", action="store_true",
+                    help="drive via resident_list()/step_local() (the "
+                         "coupled-host input form) instead of global arrays")
     ap.add_argument("--ngroups", type=int, default=1)
     ap.add_argument("--nscores", type=int, default=1)
     ap.add_argument("--reflective", action="store_true",
@@ -135,6 +147,7 @@ def main():
             "PT_TOL": repr(args.tol),
             "PT_ESCAPE_FRAC": repr(args.escape_frac),
             "PT_REFLECTIVE": "1" if args.reflective else "0",
+            "PT_LOCAL": "1" if args.local else "0",
             "PT_NGROUPS": str(args.ngroups),
             "PT_NSCORES": str(args.nscores),
         })
