@@ -79,6 +79,24 @@ for s in range(steps):
         oracle.move(origin.ravel(), dest.ravel(), fly.copy(), w,
                     groups=grp_all, responses=rsp_all)
     pos = np.where(fly[:, None] == 1, np.clip(dest, 0.0, 1.0), origin)
+    rt = int(os.environ.get("PT_STATE_RT", "0"))
+    if rt and (s + 1) % rt == 0 and s + 1 < steps:
+        # checkpoint/restore roundtrip IN-SOAK: merge the per-rank state
+        # views into the global snapshot (resident rows win; ranks agree
+        # because residency is disjoint) and re-install it via
+        # set_state -- ownership reclaims must reproduce the exact
+        # pre-snapshot state or the oracle comparison below diverges
+        rm = np.asarray(pe.resident_mask()).astype(bool)
+        ge = np.asarray(pe.elem_ids_global()).astype(np.float64)
+        ev = np.asarray(pe.escaped_mask()).astype(np.float64)
+        pp = np.asarray(pe.positions()).reshape(-1, 3).copy()
+        pp[~rm] = 0.0
+        ev[~rm] = 0.0
+        ge_m = np.asarray(pe.allreduce_max(ge))
+        pp_m = np.asarray(pe.allreduce_max(pp.ravel()))
+        ev_m = np.asarray(pe.allreduce_max(ev))
+        pe.set_state(pp_m, ge_m.astype(np.int32),
+                     ev_m.astype(np.uint8))
     if (s + 1) % 25 == 0 or s + 1 == steps:
         f1 = np.asarray(pe.flux_global()).ravel()
         if oracle is not None:
@@ -100,6 +118,10 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--state-roundtrip-every", type=int, default=0,
+                    help="every N steps, snapshot the decomposition-"
+                         "independent state and re-install it via "
+                         "set_state (checkpoint/restore stress)")
     ap.add_argument("--local", action="store_true",
                     help="drive via resident_list()/step_local() (the "
                          "coupled-host input form) instead of global arrays")
@@ -148,6 +170,7 @@ def main():
             "PT_ESCAPE_FRAC": repr(args.escape_frac),
             "PT_REFLECTIVE": "1" if args.reflective else "0",
             "PT_LOCAL": "1" if args.local else "0",
+            "PT_STATE_RT": str(args.state_roundtrip_every),
             "PT_NGROUPS": str(args.ngroups),
             "PT_NSCORES": str(args.nscores),
         })
