@@ -1,9 +1,9 @@
 """World-2 stateful-partitioned soak: hundreds of cross-rank exchange
 rounds against a replicated oracle, on one shared GPU or on CPU.
 
-Spawns 2 ranks (the library's own TCP comm via RANK/WORLD_SIZE env; on a
-GPU both ranks share cuda:0 -- the same shape the world-2 GPU tests
-use).  Each step every particle gets a random destination, a slice is
+Spawns N ranks (--ranks, default 2; the library's own TCP comm via
+RANK/WORLD_SIZE env; on a GPU all ranks share cuda:0 -- the same shape
+the world-2 GPU tests use).  Each step every particle gets a random destination, a slice is
 resampled to a new origin, and ~half the segments cross the Morton cut,
 so the phase-A claim, ghost-reroute, host-eject and unpack paths run
 continuously.  Rank 0 also drives a replicated TallyEngine fed the
@@ -107,7 +107,8 @@ for s in range(steps):
                   f"{pe.resident}, lost {st['lost_particles']}", flush=True)
             assert err < float(os.environ.get("PT_TOL", "1e-9")), "DIVERGED"
 if rank == 0:
-    print(f"PART_WORLD2_SOAK_OK: {steps} steps x {n} particles x 2 ranks "
+    print(f"PART_WORLD2_SOAK_OK: {steps} steps x {n} particles x "
+          f"{os.environ['WORLD_SIZE']} ranks "
           f"in {time.time()-t0:.1f}s on {dev}", flush=True)
 """
 
@@ -118,6 +119,7 @@ def main():
     ap.add_argument("--particles", type=int, default=400_000)
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--ranks", type=int, default=2)
     ap.add_argument("--state-roundtrip-every", type=int, default=0,
                     help="every N steps, snapshot the decomposition-"
                          "independent state and re-install it via "
@@ -157,7 +159,7 @@ def main():
             f.write(WORKER)
         env = dict(os.environ)
         env.update({
-            "WORLD_SIZE": "2",
+            "WORLD_SIZE": str(args.ranks),
             "MASTER_ADDR": "127.0.0.1",
             "PUMITALLY_PORT": str(24000 + (os.getpid() + 17) % 15000),
             "PUMITALLY_NO_TORCH": "1",
@@ -177,7 +179,7 @@ def main():
         if dev != "cpu":
             env["PUMITALLY_COMM"] = "tcp"  # two ranks share one device
         procs = []
-        for r in range(2):
+        for r in range(args.ranks):
             e = dict(env)
             e["RANK"] = str(r)
             e["LOCAL_RANK"] = "0"
