@@ -128,9 +128,32 @@ static void comm_world2() {
   if (!WIFEXITED(st) || WEXITSTATUS(st) != 0) abort();
 }
 
+// the multithreaded move path (n >= 65536) with every tally dimension
+// on: per-thread partials must span the FULL flux shape (a partial
+// sized nelems*ngroups overflowed under nscores > 1 -- the bug this
+// harness should have caught; see tests/test_threaded_scored_move.py)
+void threaded_scored_move() {
+  Mesh m = build_box(8, 8, 8, 1.0, 1.0, 1.0);
+  const int64_t n = 70000;
+  const int G = 2, S = 2;
+  auto e = make_cpu_engine(m, n, G, S);
+  std::mt19937_64 rng(17);
+  std::uniform_real_distribution<double> u(0.02, 0.98);
+  std::vector<double> o(n * 3), d(n * 3), w(n), resp(n * S);
+  std::vector<int8_t> fly(n, 1);
+  std::vector<uint16_t> grp(n);
+  for (int64_t i = 0; i < n * 3; ++i) { o[i] = u(rng); d[i] = u(rng); }
+  for (int64_t i = 0; i < n; ++i) { w[i] = u(rng); grp[i] = rng() % G; }
+  for (int64_t i = 0; i < n * S; ++i) resp[i] = u(rng);
+  e->copy_initial_position(o.data(), n);
+  e->move(o.data(), d.data(), fly.data(), w.data(), n, grp.data(),
+          resp.data());
+}
+
 int main() {
   golden();
   fuzz();
+  threaded_scored_move();
   comm_world2();
   printf("asan_check: PASS\n");
   return 0;
