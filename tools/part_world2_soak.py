@@ -36,7 +36,7 @@ mesh, cells = box_mesh_with_tets(int(os.environ["PT_TETS"]), extent=1.0)
 if os.environ.get("PT_REFLECTIVE") == "1":
     fid, _, _ = mesh.boundary_faces()
     mesh.set_reflective_faces(fid)
-rng = np.random.default_rng(5)
+rng = np.random.default_rng(int(os.environ.get("PT_SEED", "5")))
 pos = rng.uniform(0.05, 0.95, size=(n, 3))
 
 G = int(os.environ.get("PT_NGROUPS", "1"))
@@ -90,7 +90,11 @@ for s in range(steps):
         ge = np.asarray(pe.elem_ids_global()).astype(np.float64)
         ev = np.asarray(pe.escaped_mask()).astype(np.float64)
         pp = np.asarray(pe.positions()).reshape(-1, 3).copy()
-        pp[~rm] = 0.0
+        # -inf fill, NOT 0: an escaped particle's boundary clip can sit
+        # at exactly 0.0 or a few ulps NEGATIVE, and max(0, x) would
+        # perturb the restored position by ~1e-17 -- enough to flip a
+        # face containment and compound (found at seed 202, world 4)
+        pp[~rm] = -1e300
         ev[~rm] = 0.0
         ge_m = np.asarray(pe.allreduce_max(ge))
         pp_m = np.asarray(pe.allreduce_max(pp.ravel()))
@@ -120,6 +124,7 @@ def main():
     ap.add_argument("--mesh-tets", type=int, default=100_000)
     ap.add_argument("--device", default="auto")
     ap.add_argument("--ranks", type=int, default=2)
+    ap.add_argument("--seed", type=int, default=5)
     ap.add_argument("--state-roundtrip-every", type=int, default=0,
                     help="every N steps, snapshot the decomposition-"
                          "independent state and re-install it via "
@@ -173,6 +178,7 @@ def main():
             "PT_REFLECTIVE": "1" if args.reflective else "0",
             "PT_LOCAL": "1" if args.local else "0",
             "PT_STATE_RT": str(args.state_roundtrip_every),
+            "PT_SEED": str(args.seed),
             "PT_NGROUPS": str(args.ngroups),
             "PT_NSCORES": str(args.nscores),
         })
