@@ -47,3 +47,19 @@ def test_world2_localization_matches_full_mesh_at_scale():
         capture_output=True, text=True, timeout=600)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "PART_WORLD2_SOAK_OK" in r.stdout
+
+
+def test_world4_roundtrip_grouped_scored():
+    """World-4 with groups, scores, escapes and a mid-run checkpoint
+    roundtrip -- the full parity matrix in one configuration (also
+    regression for the harness's own merge fill: max(0, x) corrupted
+    escaped particles' boundary-clip positions)."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "part_world2_soak.py"),
+         "--ranks", "4", "--seed", "202", "--steps", "20", "--particles",
+         "200000", "--mesh-tets", "100000", "--device", "cpu", "--full-size",
+         "--ngroups", "2", "--nscores", "2", "--escape-frac", "0.08",
+         "--state-roundtrip-every", "7", "--tol", "1e-12"],
+        capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "PART_WORLD2_SOAK_OK" in r.stdout
